@@ -1,0 +1,14 @@
+"""ORACLE package — CPU restatement of Paimon's merge-on-read hot path.
+
+TEST INFRASTRUCTURE ONLY: only tests/, __graft_entry__.smoke() and bench.py's
+cpu_baseline leg may import this package. The product path (paimon_amd +
+libpaimon_hip.so) must never route through it.
+"""
+from .oracle import (  # noqa: F401
+    merge_order,
+    merge_dedup,
+    merge_dedup_model,
+    partial_update_model,
+    rle_bp_decode,
+    lib_path,
+)

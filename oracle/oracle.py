@@ -1,0 +1,208 @@
+"""ctypes driver for the C oracle plus independent numpy models.
+
+The C functions are the faithful loser-tree restatement (merge_oracle.c,
+citing LoserTree.java / SortMergeReaderWithLoserTree.java). The numpy models
+here are the *independent* expected-output models, restating the reference's
+own test models (MergeFunctionTestUtils.java:35-85): sort by (key, seq,
+isAdd), group by key, reduce. Tests require C restatement == numpy model ==
+pypaimon golden vectors.
+"""
+
+import ctypes
+import os
+
+import numpy as np
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+
+
+def lib_path():
+    return os.path.join(_HERE, "libpaimon_oracle.so")
+
+
+_lib = None
+
+
+def _get_lib():
+    global _lib
+    if _lib is None:
+        p = lib_path()
+        if not os.path.exists(p):
+            import subprocess
+            subprocess.run(["make", "-C", _HERE], check=True,
+                           capture_output=True)
+        _lib = ctypes.CDLL(p)
+        i64p = ctypes.POINTER(ctypes.c_int64)
+        i8p = ctypes.POINTER(ctypes.c_int8)
+        _lib.pmo_merge_order.restype = ctypes.c_int64
+        _lib.pmo_merge_dedup.restype = ctypes.c_int64
+        _lib.pmo_merge_dedup_count.restype = ctypes.c_int64
+        _lib.pmo_rle_bp_decode.restype = ctypes.c_int64
+        _lib.pmo_rle_bp_decode.argtypes = [
+            ctypes.c_char_p, ctypes.c_int64, ctypes.c_int, ctypes.c_int64,
+            ctypes.POINTER(ctypes.c_int32)]
+        del i64p, i8p
+    return _lib
+
+
+def _ptr_arrays(runs):
+    """runs: list of dicts with contiguous numpy arrays key(i64)/seq(i64)/kind(i8)."""
+    n = len(runs)
+    keys = (ctypes.POINTER(ctypes.c_int64) * n)()
+    seqs = (ctypes.POINTER(ctypes.c_int64) * n)()
+    kinds = (ctypes.POINTER(ctypes.c_int8) * n)()
+    lens = (ctypes.c_int64 * n)()
+    holders = []
+    for i, r in enumerate(runs):
+        k = np.ascontiguousarray(r["key"], dtype=np.int64)
+        s = np.ascontiguousarray(r["seq"], dtype=np.int64)
+        kd = np.ascontiguousarray(r["kind"], dtype=np.int8)
+        holders += [k, s, kd]
+        keys[i] = k.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))
+        seqs[i] = s.ctypes.data_as(ctypes.POINTER(ctypes.c_int64))
+        kinds[i] = kd.ctypes.data_as(ctypes.POINTER(ctypes.c_int8))
+        lens[i] = len(k)
+    return keys, seqs, kinds, lens, holders
+
+
+def merge_order(runs):
+    """Full merged stream: returns (run_idx i32[N], row_idx i64[N], head u8[N])."""
+    lib = _get_lib()
+    total = sum(len(r["key"]) for r in runs)
+    out_run = np.empty(total, dtype=np.int32)
+    out_row = np.empty(total, dtype=np.int64)
+    out_head = np.empty(total, dtype=np.uint8)
+    keys, seqs, kinds, lens, hold = _ptr_arrays(runs)
+    n = lib.pmo_merge_order(
+        len(runs), keys, seqs, kinds, lens,
+        out_run.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+        out_row.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)),
+        out_head.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)))
+    assert n == total, (n, total)
+    return out_run, out_row, out_head
+
+
+def merge_dedup(runs, ignore_delete=False, drop_delete=True):
+    """C restatement of the Deduplicate merge-on-read. Returns (run, row)."""
+    lib = _get_lib()
+    total = sum(len(r["key"]) for r in runs)
+    out_run = np.empty(max(total, 1), dtype=np.int32)
+    out_row = np.empty(max(total, 1), dtype=np.int64)
+    keys, seqs, kinds, lens, hold = _ptr_arrays(runs)
+    n = lib.pmo_merge_dedup(
+        len(runs), keys, seqs, kinds, lens,
+        int(ignore_delete), int(drop_delete),
+        out_run.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)),
+        out_row.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)))
+    return out_run[:n].copy(), out_row[:n].copy()
+
+
+def _kind_is_add(kind):
+    return (kind == 0) | (kind == 2)
+
+
+def _sorted_stream(runs):
+    """Independent model: flatten and lexsort ascending (key, seq, isAdd) —
+    the total merge order of SortMergeReaderWithLoserTree.java:48-75."""
+    key = np.concatenate([r["key"] for r in runs])
+    seq = np.concatenate([r["seq"] for r in runs])
+    kind = np.concatenate([r["kind"] for r in runs])
+    run = np.concatenate([np.full(len(r["key"]), i, dtype=np.int32)
+                          for i, r in enumerate(runs)])
+    row = np.concatenate([np.arange(len(r["key"]), dtype=np.int64)
+                          for r in runs])
+    order = np.lexsort((_kind_is_add(kind).astype(np.int8), seq, key))
+    return key[order], seq[order], kind[order], run[order], row[order]
+
+
+def merge_dedup_model(runs, ignore_delete=False, drop_delete=True):
+    """Numpy restatement of MergeFunctionTestUtils.getExpectedForDeduplicate
+    (:35-47) extended with ignore-delete and DropDeleteReader semantics.
+    Returns (run, row) of surviving records in key order."""
+    key, seq, kind, run, row = _sorted_stream(runs)
+    n = len(key)
+    if n == 0:
+        return (np.empty(0, dtype=np.int32), np.empty(0, dtype=np.int64))
+    head = np.empty(n, dtype=bool)
+    head[0] = True
+    head[1:] = key[1:] != key[:-1]
+    group_id = np.cumsum(head) - 1
+    n_groups = group_id[-1] + 1
+    group_size = np.bincount(group_id, minlength=n_groups)
+    idx = np.arange(n)
+    if ignore_delete:
+        eligible = _kind_is_add(kind)
+    else:
+        eligible = np.ones(n, dtype=bool)
+    # last eligible index per group
+    last = np.full(n_groups, -1, dtype=np.int64)
+    np.maximum.at(last, group_id[eligible], idx[eligible])
+    # singleton groups bypass the merge function (ReducerMergeFunctionWrapper)
+    single_first = np.full(n_groups, -1, dtype=np.int64)
+    np.maximum.at(single_first, group_id, idx)
+    singles = group_size == 1
+    last[singles] = single_first[singles]
+    sel = last[last >= 0]
+    if drop_delete:
+        sel = sel[_kind_is_add(kind[sel])]
+    return run[sel], row[sel]
+
+
+def partial_update_model(runs, drop_delete=True):
+    """Numpy model of PartialUpdateMergeFunction for INSERT-only streams
+    (PartialUpdateMergeFunction.java:188-215 updateNonNullFields +
+    ReducerMergeFunctionWrapper singleton bypass). For each key group in
+    ascending (seq) order, each value column takes the value of the last
+    record where that column is non-null; result seq = last record's seq,
+    kind = INSERT. Returns dict of output columns.
+
+    runs entries carry 'values' (list of arrays, col 0 = pk) and 'valid'
+    (parallel list of boolean masks)."""
+    key, seq, kind, run, row = _sorted_stream(runs)
+    assert _kind_is_add(kind).all(), "partial_update_model: INSERT-only streams"
+    n = len(key)
+    head = np.empty(n, dtype=bool)
+    if n == 0:
+        return {"key": key, "seq": seq, "kind": kind, "values": [], "valid": []}
+    head[0] = True
+    head[1:] = key[1:] != key[:-1]
+    group_id = np.cumsum(head) - 1
+    n_groups = group_id[-1] + 1
+    idx = np.arange(n)
+    last_all = np.zeros(n_groups, dtype=np.int64)
+    np.maximum.at(last_all, group_id, idx)
+    n_cols = len(runs[0]["values"])
+    out_vals, out_valid = [], []
+    # merged order over the flattened records, same as _sorted_stream
+    key0 = np.concatenate([r["key"] for r in runs])
+    seq0 = np.concatenate([r["seq"] for r in runs])
+    kind0 = np.concatenate([r["kind"] for r in runs])
+    order = np.lexsort((_kind_is_add(kind0).astype(np.int8), seq0, key0))
+    for c in range(n_cols):
+        col = np.concatenate([r["values"][c] for r in runs])
+        msk = np.concatenate([r["valid"][c] for r in runs]) if "valid" in runs[0] \
+            else np.ones(len(col), dtype=bool)
+        col = col[order]
+        msk = msk[order]
+        lastv = np.full(n_groups, -1, dtype=np.int64)
+        np.maximum.at(lastv, group_id[msk], idx[msk])
+        vals = np.where(lastv >= 0, col[np.clip(lastv, 0, None)], 0)
+        out_vals.append(vals.astype(col.dtype))
+        out_valid.append(lastv >= 0)
+    return {
+        "key": key[last_all], "seq": seq[last_all],
+        "kind": np.zeros(n_groups, dtype=np.int8),
+        "values": out_vals, "valid": out_valid,
+    }
+
+
+def rle_bp_decode(data: bytes, bit_width: int, num_values: int):
+    """C restatement of the Parquet RLE/bit-packed hybrid decoder."""
+    lib = _get_lib()
+    out = np.empty(num_values, dtype=np.int32)
+    n = lib.pmo_rle_bp_decode(
+        data, len(data), bit_width, num_values,
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_int32)))
+    if n != num_values:
+        raise ValueError(f"rle decode overrun: {n} != {num_values}")
+    return out

@@ -1,0 +1,152 @@
+"""Seeded synthetic sorted-run generator for the merge-on-read hot path.
+
+Writes LSM sorted runs as Parquet files using Paimon's on-disk KeyValue schema
+(reference: KeyValue.createKeyValueFields, paimon-core/src/main/java/org/apache/
+paimon/KeyValue.java:135-143; key-field prefix per paimon-api/src/main/java/org/
+apache/paimon/table/SpecialFields.java:76-83):
+
+    _KEY_<pk> ... , _SEQUENCE_NUMBER: int64 not null, _VALUE_KIND: int8 not null,
+    <value cols> ...
+
+Run invariants (reference: SortedRun, paimon-core/.../mergetree/SortedRun.java;
+MergeTreeWriter flush): each run is sorted ascending by key and contains no
+duplicate keys; sequence numbers are unique across the whole bucket.
+
+RowKind byte encoding (paimon-api/.../types/RowKind.java:35-56):
+    INSERT=0, UPDATE_BEFORE=1, UPDATE_AFTER=2, DELETE=3; isAdd = {0, 2}.
+"""
+
+import json
+import os
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.parquet as pq
+
+KIND_INSERT = 0
+KIND_UPDATE_BEFORE = 1
+KIND_UPDATE_AFTER = 2
+KIND_DELETE = 3
+
+
+def _key_space(total_rows: int) -> int:
+    # ~50% cross-run key collision at the C2 shape: keys uniform over
+    # [0, 1.5 * total_rows).
+    return max(int(total_rows * 3 // 2), 16)
+
+
+def gen_runs_dedup(n_runs: int, rows_per_run: int, n_value_cols: int = 8,
+                   seed: int = 42, delete_frac: float = 0.05):
+    """Generate in-memory sorted runs for the Deduplicate configs (C1/C2).
+
+    Schema: int64 PK + n_value_cols x int32 values (value row also contains the
+    pk column first, mirroring Paimon where primary-key columns are part of the
+    value row). Returns a list of dicts with numpy arrays:
+      {key: i64[n], seq: i64[n], kind: i8[n], values: [i64 pk, i32 x n_value_cols]}
+    """
+    rng = np.random.default_rng(seed)
+    total = n_runs * rows_per_run
+    space = _key_space(total)
+    # Globally unique sequence numbers: a shuffled range, sliced per run.
+    seqs_all = rng.permutation(total).astype(np.int64)
+    runs = []
+    for r in range(n_runs):
+        keys = rng.choice(space, size=rows_per_run, replace=False).astype(np.int64)
+        keys.sort()
+        seq = seqs_all[r * rows_per_run:(r + 1) * rows_per_run]
+        kind = np.where(rng.random(rows_per_run) < delete_frac,
+                        KIND_DELETE, KIND_INSERT).astype(np.int8)
+        values = [keys.copy()]
+        for _ in range(n_value_cols):
+            values.append(rng.integers(-2**31, 2**31, size=rows_per_run,
+                                       dtype=np.int64).astype(np.int32))
+        runs.append({"key": keys, "seq": seq, "kind": kind, "values": values})
+    return runs
+
+
+def gen_runs_partial_update(n_runs: int, rows_per_run: int, n_value_cols: int = 20,
+                            seed: int = 43, update_frac: float = 0.3,
+                            update_cols: int = 6):
+    """Runs for the PartialUpdate config (C3): wide int32 rows where update
+    records carry a random subset of non-null columns (rest null). All records
+    are INSERTs (default partial-update rejects retracts,
+    PartialUpdateMergeFunction.java:170-186)."""
+    rng = np.random.default_rng(seed)
+    total = n_runs * rows_per_run
+    space = _key_space(total)
+    seqs_all = rng.permutation(total).astype(np.int64)
+    runs = []
+    for r in range(n_runs):
+        keys = rng.choice(space, size=rows_per_run, replace=False).astype(np.int64)
+        keys.sort()
+        seq = seqs_all[r * rows_per_run:(r + 1) * rows_per_run]
+        kind = np.full(rows_per_run, KIND_INSERT, dtype=np.int8)
+        is_update = rng.random(rows_per_run) < update_frac
+        values = [keys.copy()]  # pk col, never null
+        masks = [np.ones(rows_per_run, dtype=bool)]
+        # choose update column subsets: for update rows, update_cols random cols set
+        col_set = rng.random((rows_per_run, n_value_cols)).argsort(axis=1) < update_cols
+        for c in range(n_value_cols):
+            vals = rng.integers(-2**31, 2**31, size=rows_per_run,
+                                dtype=np.int64).astype(np.int32)
+            valid = np.where(is_update, col_set[:, c], True)
+            values.append(vals)
+            masks.append(valid)
+        runs.append({"key": keys, "seq": seq, "kind": kind,
+                     "values": values, "valid": masks})
+    return runs
+
+
+def run_to_arrow(run, value_names=None):
+    """Convert one in-memory run to an Arrow table in paimon on-disk layout."""
+    n_vals = len(run["values"])
+    if value_names is None:
+        value_names = ["v_k"] + [f"v_c{i}" for i in range(n_vals - 1)]
+    fields = [pa.field("_KEY_k", pa.int64(), nullable=False),
+              pa.field("_SEQUENCE_NUMBER", pa.int64(), nullable=False),
+              pa.field("_VALUE_KIND", pa.int8(), nullable=False)]
+    cols = [pa.array(run["key"]), pa.array(run["seq"]), pa.array(run["kind"])]
+    masks = run.get("valid")
+    for i, v in enumerate(run["values"]):
+        fields.append(pa.field(value_names[i], pa.from_numpy_dtype(v.dtype),
+                               nullable=True))
+        if masks is not None:
+            cols.append(pa.array(v, mask=~masks[i]))
+        else:
+            cols.append(pa.array(v))
+    return pa.Table.from_arrays(cols, schema=pa.schema(fields))
+
+
+def write_runs(runs, out_dir, compression="NONE", row_group_rows=1 << 20,
+               data_page_rows=1 << 16):
+    """Write runs as Parquet data files; returns list of per-file metadata
+    dicts shaped like DataFileMeta (io/DataFileMeta.java:124-190): fileName,
+    rowCount, minKey, maxKey, minSequenceNumber, maxSequenceNumber, level."""
+    os.makedirs(out_dir, exist_ok=True)
+    metas = []
+    for r, run in enumerate(runs):
+        tbl = run_to_arrow(run)
+        path = os.path.join(out_dir, f"run-{r}.parquet")
+        pq.write_table(
+            tbl, path,
+            compression=None if compression == "NONE" else compression,
+            use_dictionary=False,
+            data_page_version="1.0",
+            write_statistics=False,
+            row_group_size=row_group_rows,
+            data_page_size=data_page_rows * 8,
+            store_schema=False,
+        )
+        metas.append({
+            "path": path,
+            "rowCount": int(len(run["key"])),
+            "minKey": int(run["key"][0]),
+            "maxKey": int(run["key"][-1]),
+            "minSequenceNumber": int(run["seq"].min()),
+            "maxSequenceNumber": int(run["seq"].max()),
+            "level": 0 if r == 0 else 1,
+        })
+    manifest = os.path.join(out_dir, "files.json")
+    with open(manifest, "w") as f:
+        json.dump(metas, f, indent=1)
+    return metas
