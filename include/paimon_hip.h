@@ -1,0 +1,134 @@
+/* paimon_hip.h — C-ABI of libpaimon_hip.so: the MI355X-native drop-in for
+ * Apache Paimon's merge-on-read hot path.
+ *
+ * Replaced reference surfaces (apache/paimon @ 2026-08-21), per entry point:
+ *  - pmh_plan_create + pmh_read_next replace
+ *      SplitRead<KeyValue>.createReader(Split) -> RecordReader<KeyValue>
+ *      (paimon-core/.../operation/SplitRead.java:39-63) as implemented by
+ *      MergeFileSplitRead.createReader / createMergeReader
+ *      (paimon-core/.../operation/MergeFileSplitRead.java:242-270), i.e. the
+ *      IntervalPartition sectioning (mergetree/compact/IntervalPartition.java:
+ *      67-125), per-section sort-merge (mergetree/compact/SortMergeReader.java:
+ *      41-57 with the default LOSER_TREE engine), merge functions
+ *      (DeduplicateMergeFunction.java:48-62, PartialUpdateMergeFunction.java:
+ *      148-397) wrapped by ReducerMergeFunctionWrapper.java:53-73, and
+ *      DropDeleteReader.java:53-61.
+ *  - The batch contract mirrors RecordReader<T>
+ *      (paimon-common/.../reader/RecordReader.java:40-72): pmh_read_next
+ *      returns one batch per section (the loser-tree engine yields one batch
+ *      per section, SortMergeReaderWithLoserTree.java:77-83); returned
+ *      buffers stay valid until the next pmh_read_next / pmh_plan_close on
+ *      the same plan (releaseBatch reuse contract). One plan = one HIP
+ *      stream; concurrent plans are allowed (one per bucket).
+ *  - a compaction entry point — roadmap in DESIGN.md — will replace
+ *      CompactRewriter.rewrite, mergetree/compact/CompactRewriter.java:29-56.
+ *
+ * JNI precedent in the reference: the Vortex JNI reader
+ * (paimon-vortex/.../dev/vortex/jni/NativeRuntime.java:30-34) — a Java-side
+ * binding of exactly this shape is sketched in INTEGRATION.md.
+ *
+ * Error model: functions returning pointers return NULL on error; functions
+ * returning int/int64 return negative on error. pmh_last_error() returns a
+ * thread-local message (mirrors IOException propagation of RecordReader).
+ */
+#ifndef PAIMON_HIP_H
+#define PAIMON_HIP_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct pmh_session_t pmh_session_t;
+typedef struct pmh_plan_t pmh_plan_t;
+
+/* Column dtypes (paimon logical types on the hot path; TINYINT is stored as
+ * parquet INT32 and returned here as INT8 — KeyValueSerializer.java:58-66
+ * row layout: key cols | _SEQUENCE_NUMBER | _VALUE_KIND | value cols). */
+enum pmh_dtype {
+    PMH_DT_INT8 = 1,
+    PMH_DT_INT16 = 2,
+    PMH_DT_INT32 = 3,
+    PMH_DT_INT64 = 4,
+    PMH_DT_FLOAT32 = 5,
+    PMH_DT_FLOAT64 = 6,
+};
+
+typedef struct pmh_col {
+    const char *name;
+    int32_t dtype;         /* enum pmh_dtype */
+    const void *data;      /* columnar values; device ptr unless host batch */
+    const uint8_t *valid;  /* byte-per-row validity (1=non-null); NULL = all
+                              valid. (Arrow bitmap export: later round.) */
+} pmh_col;
+
+typedef struct pmh_batch {
+    int64_t n_rows;
+    int32_t n_cols;        /* key cols, then seq, kind, then value cols */
+    int32_t device;        /* HIP device ordinal; -1 = host-resident copy */
+    const pmh_col *cols;
+} pmh_batch;
+
+typedef struct pmh_stats {
+    int64_t rows_in;         /* records fed to the merge */
+    int64_t rows_out;        /* records emitted */
+    int64_t hbm_bytes_algo;  /* algorithmic bytes (encoded in + merged out) */
+    double decode_ms;        /* device time in decode kernels */
+    double partition_ms;     /* device time in merge-path partition */
+    double merge_ms;         /* device time in merge/winner kernels */
+    double emit_ms;          /* device time in gather/emit kernels */
+    double total_device_ms;  /* end-to-end device time of last read_next */
+    double h2d_ms;           /* untimed-region staging cost, informational */
+} pmh_stats;
+
+/* Session: owns the device + stream pool. device < 0 opens a host-only
+ * session usable for metadata entry points (footer parse, planning). */
+pmh_session_t *pmh_open_session(int device);
+void pmh_close_session(pmh_session_t *s);
+
+/* plan_json (one bucket's DataSplit, table/source/DataSplit.java:63-75):
+ * {
+ *   "key_cols":   [{"name": "_KEY_k", "type": "int64"}],
+ *   "value_cols": [{"name": "v_k", "type": "int64"}, ...],   // read type
+ *   "merge_engine": "deduplicate" | "partial-update",
+ *   "drop_delete": true,         // !forceKeepDelete
+ *   "ignore_delete": false,      // CoreOptions.IGNORE_DELETE
+ *   "output": "device" | "host",
+ *   "files": [{"path": "...", "rowCount": N, "minKey": x, "maxKey": y,
+ *              "level": L}, ...]                              // DataFileMeta
+ * }
+ * Staging (file read + H2D of encoded column chunks + page tables) happens
+ * here; pmh_read_next launches only device work. */
+pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json);
+
+/* Returns rows in the batch (one section per call), 0 at end of input,
+ * negative on error. Batch buffers valid until next call / plan close. */
+int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out);
+
+int pmh_plan_close(pmh_plan_t *p);
+
+int pmh_stats_get(pmh_plan_t *p, pmh_stats *out);
+
+const char *pmh_last_error(void);
+
+/* ---- host-only debug/metadata entry points (no GPU required) ---- */
+
+/* Parse a parquet footer + page headers; returns a malloc'd JSON string
+ * (caller frees with pmh_free_string). Used by CPU tests to pin the native
+ * thrift parser against pyarrow metadata. */
+char *pmh_debug_footer_json(const char *path);
+void pmh_free_string(char *s);
+
+/* Restatement of IntervalPartition.partition() for int64 keys
+ * (mergetree/compact/IntervalPartition.java:67-125): given n files'
+ * (minKey, maxKey), writes section id and run-within-section id per file
+ * (by input order). Returns number of sections, or negative on error. */
+int pmh_debug_interval_partition(int n, const int64_t *min_keys,
+                                 const int64_t *max_keys,
+                                 int32_t *out_section, int32_t *out_run);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* PAIMON_HIP_H */

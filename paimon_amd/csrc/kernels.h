@@ -1,0 +1,80 @@
+// Device structures + kernel launcher declarations shared between
+// kernels.hip (device) and plan.cpp (host).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace pmh {
+
+// Hard capacity limits (validated host-side at plan creation):
+//  - PMH_MAX_RUNS sorted runs per section (spill path is a later round,
+//    MergeSorter.java:112-125);
+//  - rows per run < 2^28 (winner packing run:4 | row:28).
+constexpr int PMH_MAX_RUNS = 16;
+constexpr int PMH_TILE_THREADS = 256;
+constexpr int64_t PMH_TILE_ROWS = 2048;
+constexpr int PMH_TILE_MAX = PMH_TILE_ROWS + PMH_MAX_RUNS;
+constexpr int PMH_TILE_ITER =
+    (PMH_TILE_MAX + PMH_TILE_THREADS - 1) / PMH_TILE_THREADS;
+
+// A decoded or raw-PLAIN region of one column of one run. Access is by
+// global row within the run: binary search pages by start_row. A fully
+// materialized (decoded) column is a single-page DevCol.
+struct DevPage {
+    uint64_t addr;      // device pointer to first element
+    int64_t start_row;  // first run-row covered by this page
+};
+
+struct DevCol {
+    const DevPage *pages;  // device array, sorted by start_row
+    int32_t n_pages;
+    int32_t esize;  // element size in bytes (stored width)
+};
+
+// Host-prescanned RLE/bit-packed work unit (split at group boundaries).
+struct RleChunk {
+    uint64_t src;       // device pointer to first packed group (kind 1)
+    int64_t out_start;  // output element index
+    int32_t count;
+    int32_t kind;       // 0 = RLE run, 1 = bit-packed groups
+    uint32_t value;     // RLE literal (kind 0)
+    int32_t bit_width;  // per-page bit width (dict-id streams vary per page)
+};
+
+extern "C" {
+
+hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
+                                int64_t tile_rows, int64_t n_bounds,
+                                int64_t total_rows, int32_t *cuts,
+                                hipStream_t stream);
+
+hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
+                                  const DevCol *kinds, const int64_t *lens,
+                                  int k, const int32_t *cuts, int64_t n_tiles,
+                                  int64_t tile_rows, int flags,
+                                  uint32_t *winners, int32_t *tile_counts,
+                                  hipStream_t stream);
+
+hipError_t pmh_launch_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
+                                 int64_t *tile_offsets, int64_t *total_out,
+                                 hipStream_t stream);
+
+hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
+                           int n_cols, int k, const uint32_t *winners,
+                           const int32_t *tile_counts,
+                           const int64_t *tile_offsets, int64_t n_tiles,
+                           int64_t tile_rows, void *const *out_ptrs,
+                           hipStream_t stream);
+
+hipError_t pmh_launch_rle_decode(const RleChunk *chunks, int64_t n_chunks,
+                                 int32_t *out, hipStream_t stream);
+
+hipError_t pmh_launch_dict_gather(const int32_t *ids, const void *dict,
+                                  int64_t n, void *out, int esize,
+                                  hipStream_t stream);
+
+}  // extern "C"
+
+}  // namespace pmh

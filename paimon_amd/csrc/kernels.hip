@@ -1,0 +1,589 @@
+// CDNA4 (gfx950) kernels for Paimon's merge-on-read hot path.
+//
+// MI355X-first design (NOT a translation of the reference's Java loops):
+// the reference's LoserTree (mergetree/compact/LoserTree.java) stalls per
+// key on a pointer-chasing tournament; here the same total order
+// (ascending userKey, sequenceNumber, isAdd —
+// SortMergeReaderWithLoserTree.java:48-75) is produced by a merge-path
+// partition over k sorted runs into independent tiles, an LDS-staged
+// pairwise-stable merge inside each tile, and a segmented winner reduction
+// per equal-key group (DeduplicateMergeFunction.java:48-62 +
+// ReducerMergeFunctionWrapper.java:53-73 + DropDeleteReader.java:53-61).
+// This path is HBM-bandwidth-bound integer work: no MFMA, wide coalesced
+// loads, LDS staging, grid-stride launches.
+//
+// Capacity limits (checked host-side): runs per section <= PMH_MAX_RUNS,
+// rows per run < 2^28 (packed winner format run:4 | row:28).
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+#include "kernels.h"
+
+#define DEV __device__ __forceinline__
+
+namespace pmh {
+
+// ---------------------------------------------------------------- paged cols
+
+template <typename T>
+DEV T col_load(const DevCol &c, int64_t row) {
+    // last page with start_row <= row; n_pages is small (L2-resident table)
+    int lo = 0, hi = c.n_pages - 1;
+    while (lo < hi) {
+        int mid = (lo + hi + 1) >> 1;
+        if (c.pages[mid].start_row <= row) lo = mid;
+        else hi = mid - 1;
+    }
+    const DevPage &pg = c.pages[lo];
+    return *reinterpret_cast<const T *>(pg.addr + (uint64_t)(row - pg.start_row) * sizeof(T));
+}
+
+DEV uint64_t ukey(int64_t k) { return (uint64_t)k ^ 0x8000000000000000ull; }
+
+// #elements in run col (sorted unique keys) with ukey(e) <= v, searched in
+// window [lo, hi).
+DEV int64_t count_le(const DevCol &c, int64_t lo, int64_t hi, uint64_t v) {
+    while (lo < hi) {
+        int64_t mid = lo + ((hi - lo) >> 1);
+        if (ukey(col_load<int64_t>(c, mid)) <= v) lo = mid + 1;
+        else hi = mid;
+    }
+    return lo;
+}
+
+DEV int64_t count_lt(const DevCol &c, int64_t lo, int64_t hi, uint64_t v) {
+    while (lo < hi) {
+        int64_t mid = lo + ((hi - lo) >> 1);
+        if (ukey(col_load<int64_t>(c, mid)) < v) lo = mid + 1;
+        else hi = mid;
+    }
+    return lo;
+}
+
+// ------------------------------------------------------------ k_partition
+//
+// For each tile boundary b (rank D = b*T clamped), find cuts c_r such that
+// sum c_r = D and the (key, run) total order is two-sided partitioned
+// (GPU merge-path generalized to k runs via key-domain bisection with
+// per-run shrinking windows).
+__global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
+                            int64_t tile_rows, int64_t n_bounds,
+                            int64_t total_rows, int32_t *cuts /* n_bounds*k */) {
+    int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= n_bounds) return;
+    int64_t D = b * tile_rows;
+    if (D > total_rows) D = total_rows;
+
+    int64_t wlo[PMH_MAX_RUNS], whi[PMH_MAX_RUNS];
+    for (int r = 0; r < k; r++) { wlo[r] = 0; whi[r] = lens[r]; }
+
+    if (D == 0) {
+        for (int r = 0; r < k; r++) cuts[b * k + r] = 0;
+        return;
+    }
+    if (D >= total_rows) {
+        for (int r = 0; r < k; r++) cuts[b * k + r] = (int32_t)lens[r];
+        return;
+    }
+    // bisect for the smallest v with (# ukey <= v) >= D  (v = pivot key of
+    // the element at rank D-1... we need cuts after D elements: the first D
+    // elements in (key, run) order). Let v* = key of the D-th smallest
+    // element boundary: smallest v such that cnt_le(v) >= D.
+    uint64_t klo = 0, khi = 0xffffffffffffffffull;
+    while (klo < khi) {
+        uint64_t mid = klo + ((khi - klo) >> 1);
+        int64_t cnt = 0;
+        for (int r = 0; r < k; r++)
+            cnt += count_le(keys[r], wlo[r], whi[r], mid);
+        if (cnt >= D) {
+            khi = mid;
+            for (int r = 0; r < k; r++)
+                whi[r] = count_le(keys[r], wlo[r], whi[r], mid);
+        } else {
+            klo = mid + 1;
+            for (int r = 0; r < k; r++)
+                wlo[r] = count_le(keys[r], wlo[r], whi[r], mid);
+        }
+    }
+    // klo == v*: all elements with key < v* are taken; among key == v*
+    // (<=1 per run), take the first t in run order. Searching within the
+    // final windows yields absolute positions: everything below wlo has
+    // key < v*, everything at/above whi has key > v*.
+    int64_t base = 0;
+    int64_t lb[PMH_MAX_RUNS], has[PMH_MAX_RUNS];
+    for (int r = 0; r < k; r++) {
+        int64_t l = count_lt(keys[r], wlo[r], whi[r], klo);
+        lb[r] = l;
+        has[r] = (l < lens[r]) &&
+                 (ukey(col_load<int64_t>(keys[r], l)) == klo);
+        base += l;
+    }
+    int64_t t = D - base;
+    for (int r = 0; r < k; r++) {
+        int64_t c = lb[r];
+        if (t > 0 && has[r]) { c++; t--; }
+        cuts[b * k + r] = (int32_t)c;
+    }
+}
+
+// ------------------------------------------------------------ k_merge_tiles
+//
+// Phase 1: per tile, stage (key, seq, kind) segments in LDS, stable
+// pairwise-merge into (key, run) order, mark equal-key group heads, run a
+// segmented (seq, isAdd) argmax per group (groups have <= k records: one
+// per run — SortedRun invariant, SortedRun.java), apply Deduplicate +
+// wrapper + drop-delete rules, and emit packed winners per owned group.
+
+struct TileSmem {
+    int64_t skey[PMH_TILE_MAX];
+    int64_t sseq[PMH_TILE_MAX];
+    uint16_t perm[2][PMH_TILE_MAX];
+    uint16_t win[PMH_TILE_MAX];
+    uint8_t skind[PMH_TILE_MAX];
+    uint8_t head[PMH_TILE_MAX];    // group head flags (merged order)
+    uint8_t fdone[PMH_TILE_MAX];   // segmented-scan carry flags
+    int32_t segoff[PMH_MAX_RUNS + 1];
+    int32_t seglen[PMH_MAX_RUNS];  // extended (incl. extra) lengths
+    int32_t paircnt[PMH_MAX_RUNS + 1];
+    int64_t predkey;
+    int32_t haspred;
+    int32_t mtotal;    // extended element count
+    int32_t mreal;     // real element count (rank width of this tile)
+    int32_t nemit;
+};
+
+// stable 2-way co-rank: position in A of the split for output rank d of
+// merge(A, B), ties take A first. A/B are perm-index sequences; key lookup
+// through skey[.].
+DEV int32_t corank(int64_t d, const uint16_t *pa, int32_t la,
+                   const uint16_t *pb, int32_t lb, const int64_t *skey,
+                   const uint16_t *, int32_t) {
+    int64_t ilo = d > lb ? d - lb : 0;
+    int64_t ihi = d < la ? d : la;
+    while (ilo < ihi) {
+        int64_t i = ilo + ((ihi - ilo) >> 1);
+        int64_t j = d - i - 1;  // candidate: a[i] vs b[j]
+        // a[i] goes after b[j] iff key(a[i]) > key(b[j])  (ties: A first)
+        if (j >= 0 && j < lb && skey[pa[i]] > skey[pb[j]]) ilo = i + 1;
+        else ihi = i;
+    }
+    return (int32_t)ilo;
+}
+
+DEV bool kind_is_add(uint8_t k) { return k == 0 || k == 2; }
+
+// winner preference order for dedup: later (seq, isAdd) wins; eligibility
+// (ignore-delete) dominates.
+DEV bool better(int64_t seq_a, uint8_t kind_a, bool elig_a, int64_t seq_b,
+                uint8_t kind_b, bool elig_b) {
+    // true if A strictly preferred over B as the group's result
+    if (elig_a != elig_b) return elig_a;
+    if (seq_a != seq_b) return seq_a > seq_b;
+    return kind_is_add(kind_a) && !kind_is_add(kind_b);
+}
+
+__launch_bounds__(PMH_TILE_THREADS) __global__
+void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
+                   const int64_t *lens, int k, const int32_t *cuts,
+                   int64_t n_tiles, int64_t tile_rows, int flags,
+                   uint32_t *winners, int32_t *tile_counts) {
+    const bool drop_delete = flags & 1;
+    const bool ignore_delete = flags & 2;
+    __shared__ TileSmem sm;
+    for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+        const int tid = threadIdx.x;
+        const int32_t *c0 = &cuts[tile * k];
+        const int32_t *c1 = &cuts[(tile + 1) * k];
+
+        // --- segment setup (extended: +1 element per run when available)
+        if (tid == 0) {
+            int32_t off = 0;
+            int32_t real = 0;
+            int64_t pred = 0;
+            int hp = 0;
+            for (int r = 0; r < k; r++) {
+                int32_t len = c1[r] - c0[r];
+                real += len;
+                int32_t ext = (c1[r] < (int32_t)lens[r]) ? 1 : 0;
+                sm.segoff[r] = off;
+                sm.seglen[r] = len + ext;
+                off += len + ext;
+                if (c0[r] > 0) {
+                    int64_t kk = col_load<int64_t>(keys[r], c0[r] - 1);
+                    if (!hp || kk > pred) pred = kk;
+                    hp = 1;
+                }
+            }
+            sm.segoff[k] = off;
+            sm.mtotal = off;
+            sm.mreal = real;
+            sm.predkey = pred;
+            sm.haspred = hp;
+            sm.nemit = 0;
+        }
+        __syncthreads();
+        const int32_t M = sm.mtotal;
+        const int32_t Mreal = sm.mreal;
+        if (Mreal == 0) {
+            if (tid == 0) tile_counts[tile] = 0;
+            __syncthreads();
+            continue;
+        }
+
+        // --- stage key/seq/kind segments into LDS (coalesced per run)
+        for (int r = 0; r < k; r++) {
+            int32_t off = sm.segoff[r], len = sm.seglen[r];
+            int64_t base = c0[r];
+            for (int32_t i = tid; i < len; i += blockDim.x) {
+                sm.skey[off + i] = col_load<int64_t>(keys[r], base + i);
+                sm.sseq[off + i] = col_load<int64_t>(seqs[r], base + i);
+                sm.skind[off + i] =
+                    (uint8_t)col_load<int32_t>(kinds[r], base + i);
+                sm.perm[0][off + i] = (uint16_t)(off + i);
+            }
+        }
+        __syncthreads();
+
+        // --- pairwise stable merge, ceil(log2(k)) levels
+        int cur = 0;
+        for (int width = 1; width < k; width <<= 1) {
+            // sequences at this level: [segoff[q*width], segoff[min((q+1)*width,k)])
+            // merge pairs (2q, 2q+1)
+            const int nxt = cur ^ 1;
+            // per-thread chunks of 8 outputs, grid-stride over all chunks
+            const int CH = 8;
+            int n_chunks = (M + CH - 1) / CH;
+            for (int ch = tid; ch < n_chunks; ch += blockDim.x) {
+                int64_t o = (int64_t)ch * CH;  // global output rank
+                // locate pair: pair p covers [segoff[a0], segoff[b1])
+                // where a0 = p*2*width, b1 = min(a0+2*width, k)
+                int p = 0;
+                // small loop over <= k/(2*width)+1 pairs
+                while ((p + 1) * 2 * width < k &&
+                       sm.segoff[(p + 1) * 2 * width] <= o)
+                    p++;
+                int a0 = p * 2 * width;
+                int amid = a0 + width < k ? a0 + width : k;
+                int b1 = a0 + 2 * width < k ? a0 + 2 * width : k;
+                int32_t abase = sm.segoff[a0];
+                int32_t la = sm.segoff[amid] - abase;
+                int32_t lb = sm.segoff[b1] - sm.segoff[amid];
+                int64_t d = o - abase;  // rank within pair
+                int32_t lim = la + lb - (int32_t)d;
+                if (lim <= 0) continue;
+                int n_out = lim < CH ? lim : CH;
+                const uint16_t *pa = &sm.perm[cur][abase];
+                const uint16_t *pb = &sm.perm[cur][abase + la];
+                int32_t ai = corank(d, pa, la, pb, lb, sm.skey, nullptr, 0);
+                int32_t bi = (int32_t)d - ai;
+                uint16_t *out = &sm.perm[nxt][abase + d];
+                for (int x = 0; x < n_out; x++) {
+                    bool takeA;
+                    if (ai >= la) takeA = false;
+                    else if (bi >= lb) takeA = true;
+                    else takeA = !(sm.skey[pa[ai]] > sm.skey[pb[bi]]);
+                    out[x] = takeA ? pa[ai++] : pb[bi++];
+                }
+            }
+            cur = nxt;
+            __syncthreads();
+        }
+        const uint16_t *mo = sm.perm[cur];
+
+        // --- group heads + previous-tile continuation skip
+        for (int32_t i = tid; i < M; i += blockDim.x) {
+            int64_t kk = sm.skey[mo[i]];
+            uint8_t h = (i == 0) ? 1 : (kk != sm.skey[mo[i - 1]]);
+            // records continuing the previous tile's last group are not
+            // heads here (that tile consumed them as its extras)
+            if (sm.haspred && kk == sm.predkey) h = 0;
+            sm.head[i] = h;
+            sm.fdone[i] = h;
+            sm.win[i] = (uint16_t)i;
+        }
+        __syncthreads();
+
+        // --- segmented argmax by (eligible, seq, isAdd): groups <= k.
+        // Double-buffered through registers (static indices: rule-of-thumb
+        // from the CDNA guide — runtime-indexed locals spill to scratch).
+        for (int d = 1; d < k; d <<= 1) {
+            uint16_t nw[PMH_TILE_ITER];
+            uint8_t nf[PMH_TILE_ITER];
+#pragma unroll
+            for (int j = 0; j < PMH_TILE_ITER; j++) {
+                int32_t i = tid + j * PMH_TILE_THREADS;
+                if (i >= M) continue;
+                nw[j] = sm.win[i];
+                nf[j] = sm.fdone[i];
+                if (i >= d && !sm.fdone[i]) {
+                    uint16_t a = sm.win[i - d];
+                    uint16_t sa = mo[a], sb = mo[nw[j]];
+                    bool elig_a = !ignore_delete || kind_is_add(sm.skind[sa]);
+                    bool elig_b = !ignore_delete || kind_is_add(sm.skind[sb]);
+                    if (better(sm.sseq[sa], sm.skind[sa], elig_a,
+                               sm.sseq[sb], sm.skind[sb], elig_b))
+                        nw[j] = a;
+                    nf[j] = sm.fdone[i] | sm.fdone[i - d];
+                }
+            }
+            __syncthreads();
+#pragma unroll
+            for (int j = 0; j < PMH_TILE_ITER; j++) {
+                int32_t i = tid + j * PMH_TILE_THREADS;
+                if (i >= M) continue;
+                sm.win[i] = nw[j];
+                sm.fdone[i] = nf[j];
+            }
+            __syncthreads();
+        }
+
+        // --- emit winners of owned groups, in key order.
+        // A group is owned iff its head is a real (rank < Mreal) head; the
+        // winner is the segmented-argmax value at the group's tail. Two
+        // passes per thread over a contiguous head range (count, then
+        // write at the scanned offset) — no per-thread buffering.
+        __shared__ int32_t s_scan[PMH_TILE_THREADS];
+        const int32_t per = (Mreal + (int32_t)blockDim.x - 1) / blockDim.x;
+        int32_t my_lo = tid * per;
+        int32_t my_hi = my_lo + per < Mreal ? my_lo + per : Mreal;
+        uint32_t *wout = &winners[tile * tile_rows];
+        int32_t total = 0;
+        for (int pass = 0; pass < 2; pass++) {
+            int32_t nloc = 0;
+            int32_t my_off = pass == 1 ? s_scan[tid] : 0;
+            for (int32_t i = my_lo; i < my_hi; i++) {
+                if (!sm.head[i]) continue;
+                int32_t tail = i;  // group tail: scan forward (<= k steps)
+                while (tail + 1 < M && !sm.head[tail + 1]) tail++;
+                uint16_t s = mo[sm.win[tail]];
+                int32_t gsize = tail - i + 1;
+                bool elig_w = !ignore_delete || kind_is_add(sm.skind[s]);
+                if (!elig_w && gsize > 1) continue;  // all records ignored
+                if (drop_delete && !kind_is_add(sm.skind[s])) continue;
+                if (pass == 1) {
+                    int r = 0;  // map seg index -> (run, global row)
+                    while (r + 1 <= k - 1 && sm.segoff[r + 1] <= (int32_t)s)
+                        r++;
+                    uint32_t grow =
+                        (uint32_t)(c0[r] + ((int32_t)s - sm.segoff[r]));
+                    wout[my_off + nloc] = ((uint32_t)r << 28) | grow;
+                }
+                nloc++;
+            }
+            if (pass == 0) {
+                s_scan[tid] = nloc;
+                __syncthreads();
+                // block exclusive scan (Hillis-Steele)
+                for (int d = 1; d < (int)blockDim.x; d <<= 1) {
+                    int32_t v = s_scan[tid];
+                    int32_t add = tid >= d ? s_scan[tid - d] : 0;
+                    __syncthreads();
+                    s_scan[tid] = v + add;
+                    __syncthreads();
+                }
+                total = s_scan[blockDim.x - 1];
+                int32_t incl = s_scan[tid];
+                __syncthreads();
+                s_scan[tid] = incl - nloc;  // exclusive
+                __syncthreads();
+            }
+        }
+        if (tid == 0) tile_counts[tile] = total;
+        __syncthreads();
+    }
+}
+
+// ------------------------------------------------------------ k_scan_tiles
+// Exclusive scan of tile_counts (single workgroup, chunked).
+__global__ void k_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
+                             int64_t *tile_offsets, int64_t *total_out) {
+    __shared__ int64_t s[PMH_TILE_THREADS];
+    __shared__ int64_t running;
+    if (threadIdx.x == 0) running = 0;
+    __syncthreads();
+    for (int64_t base = 0; base < n_tiles; base += blockDim.x) {
+        int64_t i = base + threadIdx.x;
+        int64_t v = i < n_tiles ? tile_counts[i] : 0;
+        s[threadIdx.x] = v;
+        __syncthreads();
+        for (int d = 1; d < (int)blockDim.x; d <<= 1) {
+            int64_t add = threadIdx.x >= d ? s[threadIdx.x - d] : 0;
+            __syncthreads();
+            s[threadIdx.x] += add;
+            __syncthreads();
+        }
+        if (i < n_tiles) tile_offsets[i] = running + s[threadIdx.x] - v;
+        __syncthreads();
+        if (threadIdx.x == 0) running += s[blockDim.x - 1];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) *total_out = running;
+}
+
+// ------------------------------------------------------------ k_emit
+// Phase 2: gather winner rows into contiguous output columns.
+// One workgroup per tile (grid-stride); thread per output row; inner loop
+// over columns. Output writes are coalesced per column.
+__global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
+                       const uint8_t *col_dtype, int n_cols, int k,
+                       const uint32_t *winners, const int32_t *tile_counts,
+                       const int64_t *tile_offsets, int64_t n_tiles,
+                       int64_t tile_rows, void *const *out_ptrs) {
+    for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+        int32_t cnt = tile_counts[tile];
+        int64_t base = tile_offsets[tile];
+        const uint32_t *w = &winners[tile * tile_rows];
+        for (int32_t j = threadIdx.x; j < cnt; j += blockDim.x) {
+            uint32_t packed = w[j];
+            int run = packed >> 28;
+            int64_t row = packed & 0x0fffffff;
+            int64_t o = base + j;
+            for (int c = 0; c < n_cols; c++) {
+                const DevCol &dc = cols[run * n_cols + c];
+                switch (col_dtype[c]) {
+                case 1:  // INT8 output from INT32-stored parquet TINYINT
+                    ((int8_t *)out_ptrs[c])[o] =
+                        (int8_t)col_load<int32_t>(dc, row);
+                    break;
+                case 3:
+                    ((int32_t *)out_ptrs[c])[o] = col_load<int32_t>(dc, row);
+                    break;
+                case 4:
+                    ((int64_t *)out_ptrs[c])[o] = col_load<int64_t>(dc, row);
+                    break;
+                case 5:
+                    ((float *)out_ptrs[c])[o] = col_load<float>(dc, row);
+                    break;
+                case 6:
+                    ((double *)out_ptrs[c])[o] = col_load<double>(dc, row);
+                    break;
+                default: break;
+                }
+            }
+        }
+    }
+}
+
+// ------------------------------------------------------------ k_rle_decode
+// Decode Parquet RLE/bit-packed hybrid streams (dictionary ids, def levels)
+// from host-prescanned run chunks (VectorizedRleValuesReader.java:977-1018
+// wire format; the sequential varint-header walk happens on the host at
+// staging time, the bulk expansion here).
+__global__ void k_rle_decode(const RleChunk *chunks, int64_t n_chunks,
+                             int32_t *out) {
+    for (int64_t cidx = blockIdx.x; cidx < n_chunks; cidx += gridDim.x) {
+        RleChunk ch = chunks[cidx];
+        int bit_width = ch.bit_width;
+        uint32_t mask =
+            bit_width >= 32 ? 0xffffffffu : ((1u << bit_width) - 1u);
+        if (ch.kind == 0) {
+            for (int32_t i = threadIdx.x; i < ch.count; i += blockDim.x)
+                out[ch.out_start + i] = (int32_t)ch.value;
+        } else {
+            const uint8_t *src = (const uint8_t *)ch.src;
+            for (int32_t i = threadIdx.x; i < ch.count; i += blockDim.x) {
+                int64_t g = i >> 3;
+                int lane = i & 7;
+                const uint8_t *gp = src + g * bit_width;
+                int64_t bit_off = (int64_t)lane * bit_width;
+                int64_t byte_off = bit_off >> 3;
+                int shift = (int)(bit_off & 7);
+                uint64_t word = 0;
+                for (int b = 0; b < 5 && byte_off + b < bit_width; b++)
+                    word |= (uint64_t)gp[byte_off + b] << (8 * b);
+                out[ch.out_start + i] = (int32_t)((word >> shift) & mask);
+            }
+        }
+    }
+}
+
+// ------------------------------------------------------------ k_dict_gather
+template <typename T>
+__global__ void k_dict_gather_t(const int32_t *ids, const T *dict, int64_t n,
+                                T *out) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) out[i] = dict[ids[i]];
+}
+
+// ---------------------------------------------------------------- launchers
+
+extern "C" {
+
+hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
+                                int64_t tile_rows, int64_t n_bounds,
+                                int64_t total_rows, int32_t *cuts,
+                                hipStream_t stream) {
+    int threads = 128;
+    int blocks = (int)((n_bounds + threads - 1) / threads);
+    hipLaunchKernelGGL(k_partition, dim3(blocks), dim3(threads), 0, stream,
+                       keys, lens, k, tile_rows, n_bounds, total_rows, cuts);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
+                                  const DevCol *kinds, const int64_t *lens,
+                                  int k, const int32_t *cuts, int64_t n_tiles,
+                                  int64_t tile_rows, int flags,
+                                  uint32_t *winners, int32_t *tile_counts,
+                                  hipStream_t stream) {
+    int blocks = n_tiles < 4096 ? (int)n_tiles : 4096;
+    hipLaunchKernelGGL(k_merge_tiles, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
+                       stream, keys, seqs, kinds, lens, k, cuts, n_tiles,
+                       tile_rows, flags, winners, tile_counts);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
+                                 int64_t *tile_offsets, int64_t *total_out,
+                                 hipStream_t stream) {
+    hipLaunchKernelGGL(k_scan_tiles, dim3(1), dim3(PMH_TILE_THREADS), 0,
+                       stream, tile_counts, n_tiles, tile_offsets, total_out);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
+                           int n_cols, int k, const uint32_t *winners,
+                           const int32_t *tile_counts,
+                           const int64_t *tile_offsets, int64_t n_tiles,
+                           int64_t tile_rows, void *const *out_ptrs,
+                           hipStream_t stream) {
+    int blocks = n_tiles < 4096 ? (int)n_tiles : 4096;
+    hipLaunchKernelGGL(k_emit, dim3(blocks), dim3(256), 0, stream, cols,
+                       col_dtype, n_cols, k, winners, tile_counts,
+                       tile_offsets, n_tiles, tile_rows, out_ptrs);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_rle_decode(const RleChunk *chunks, int64_t n_chunks,
+                                 int32_t *out, hipStream_t stream) {
+    int blocks = n_chunks < 4096 ? (int)(n_chunks ? n_chunks : 1) : 4096;
+    hipLaunchKernelGGL(k_rle_decode, dim3(blocks), dim3(256), 0, stream,
+                       chunks, n_chunks, out);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_dict_gather(const int32_t *ids, const void *dict,
+                                  int64_t n, void *out, int esize,
+                                  hipStream_t stream) {
+    int threads = 256;
+    int64_t want = (n + threads - 1) / threads;
+    int blocks = want < 4096 ? (int)(want ? want : 1) : 4096;
+    if (esize == 4)
+        hipLaunchKernelGGL(k_dict_gather_t<int32_t>, dim3(blocks),
+                           dim3(threads), 0, stream, ids,
+                           (const int32_t *)dict, n, (int32_t *)out);
+    else if (esize == 8)
+        hipLaunchKernelGGL(k_dict_gather_t<int64_t>, dim3(blocks),
+                           dim3(threads), 0, stream, ids,
+                           (const int64_t *)dict, n, (int64_t *)out);
+    else
+        return hipErrorInvalidValue;
+    return hipGetLastError();
+}
+
+}  // extern "C"
+
+}  // namespace pmh

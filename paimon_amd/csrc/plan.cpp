@@ -1,0 +1,988 @@
+// libpaimon_hip host side: session/plan lifecycle, planning
+// (IntervalPartition restatement), file staging (footer parse, page tables,
+// H2D of encoded column chunks), and the read_next device pipeline.
+//
+// Replaces, for this path, MergeFileSplitRead.createReader
+// (paimon-core/.../operation/MergeFileSplitRead.java:242-270) +
+// MergeTreeReaders (mergetree/MergeTreeReaders.java:40-102) +
+// KeyValueFileReaderFactory (io/KeyValueFileReaderFactory.java:127-209).
+// The KeyValue row layout contract (key cols | _SEQUENCE_NUMBER |
+// _VALUE_KIND | value cols) follows KeyValueSerializer.java:34-99.
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <chrono>
+#include <cstring>
+#include <fstream>
+#include <memory>
+#include <queue>
+#include <string>
+#include <vector>
+
+#include "../../include/paimon_hip.h"
+#include "common.h"
+#include "json.h"
+#include "kernels.h"
+#include "parquet_meta.h"
+
+#include <dlfcn.h>
+
+namespace pmh {
+
+std::string &last_error() {
+    thread_local std::string e;
+    return e;
+}
+
+#define HIP_TRY(expr)                                                     \
+    do {                                                                  \
+        hipError_t _e = (expr);                                           \
+        if (_e != hipSuccess) {                                           \
+            set_error("HIP error %s at %s:%d: %s", hipGetErrorString(_e), \
+                      __FILE__, __LINE__, #expr);                         \
+            return false;                                                 \
+        }                                                                 \
+    } while (0)
+
+// ------------------------------------------------------------------ zstd
+
+typedef size_t (*zstd_decompress_fn)(void *, size_t, const void *, size_t);
+typedef unsigned (*zstd_iserror_fn)(size_t);
+
+static bool zstd_decompress(const uint8_t *src, size_t src_n, uint8_t *dst,
+                            size_t dst_n) {
+    static zstd_decompress_fn fn = nullptr;
+    static zstd_iserror_fn err_fn = nullptr;
+    if (!fn) {
+        void *h = dlopen("libzstd.so.1", RTLD_NOW | RTLD_GLOBAL);
+        if (!h) {
+            set_error("libzstd.so.1 not found for zstd-compressed pages");
+            return false;
+        }
+        fn = (zstd_decompress_fn)dlsym(h, "ZSTD_decompress");
+        err_fn = (zstd_iserror_fn)dlsym(h, "ZSTD_isError");
+    }
+    size_t r = fn(dst, dst_n, src, src_n);
+    if (err_fn(r) || r != dst_n) {
+        set_error("zstd decompress failed (%zu != %zu)", r, dst_n);
+        return false;
+    }
+    return true;
+}
+
+// ------------------------------------------- IntervalPartition (restated)
+
+struct FileDesc {
+    std::string path;
+    int64_t row_count = 0;
+    int64_t min_key = 0;
+    int64_t max_key = 0;
+    int level = 0;
+    int input_index = 0;
+};
+
+// Restatement of IntervalPartition (mergetree/compact/IntervalPartition.java
+// :40-126) for int64 keys: sort files by (minKey, maxKey); cut sections
+// where minKey exceeds the running right bound; within a section, greedily
+// pack files into the fewest sorted runs via a min-heap on each run's last
+// maxKey.
+static std::vector<std::vector<std::vector<FileDesc>>> interval_partition(
+    std::vector<FileDesc> files) {
+    std::sort(files.begin(), files.end(), [](const FileDesc &a, const FileDesc &b) {
+        if (a.min_key != b.min_key) return a.min_key < b.min_key;
+        return a.max_key < b.max_key;
+    });
+    std::vector<std::vector<std::vector<FileDesc>>> result;
+    std::vector<FileDesc> section;
+    int64_t bound = 0;
+    bool has_bound = false;
+
+    auto pack = [](const std::vector<FileDesc> &metas) {
+        // min-heap of runs keyed by last file's maxKey (IntervalPartition.java:93-125)
+        auto cmp = [](const std::vector<FileDesc> &a, const std::vector<FileDesc> &b) {
+            return a.back().max_key > b.back().max_key;  // min-heap
+        };
+        std::priority_queue<std::vector<FileDesc>, std::vector<std::vector<FileDesc>>,
+                            decltype(cmp)>
+            q(cmp);
+        q.push({metas[0]});
+        for (size_t i = 1; i < metas.size(); i++) {
+            auto top = q.top();
+            q.pop();
+            if (metas[i].min_key > top.back().max_key) {
+                top.push_back(metas[i]);
+            } else {
+                q.push({metas[i]});
+            }
+            q.push(top);
+        }
+        std::vector<std::vector<FileDesc>> runs;
+        while (!q.empty()) {
+            runs.push_back(q.top());
+            q.pop();
+        }
+        return runs;
+    };
+
+    for (const auto &meta : files) {
+        if (!section.empty() && meta.min_key > bound) {
+            result.push_back(pack(section));
+            section.clear();
+            has_bound = false;
+        }
+        section.push_back(meta);
+        if (!has_bound || meta.max_key > bound) {
+            bound = meta.max_key;
+            has_bound = true;
+        }
+    }
+    if (!section.empty()) result.push_back(pack(section));
+    return result;
+}
+
+// ------------------------------------------------------------- data model
+
+struct ColSpec {
+    std::string name;
+    int dtype;  // pmh_dtype
+    int out_esize;
+    int stored_esize;  // parquet physical width (TINYINT stored as INT32)
+};
+
+struct DeviceBufs {
+    std::vector<void *> bufs;
+    void *alloc(size_t n) {
+        void *p = nullptr;
+        if (hipMalloc(&p, n ? n : 8) != hipSuccess) return nullptr;
+        bufs.push_back(p);
+        return p;
+    }
+    ~DeviceBufs() {
+        for (void *p : bufs) hipFree(p);
+    }
+};
+
+// one (run, column): page table + optional dictionary-decode work
+struct RunCol {
+    std::vector<DevPage> pages_host;
+    DevPage *pages_dev = nullptr;
+    int n_pages = 0;
+    // dictionary path (materialized at read_next into `contig`)
+    bool dict_encoded = false;
+    std::vector<RleChunk> rle_host;
+    RleChunk *rle_dev = nullptr;
+    int32_t *ids_dev = nullptr;
+    void *dict_dev = nullptr;
+    void *contig = nullptr;  // materialized column (single-page DevCol)
+    int64_t n_rows = 0;
+};
+
+struct Run {
+    int64_t length = 0;
+    std::vector<RunCol> cols;
+};
+
+struct Section {
+    std::vector<Run> runs;
+    int64_t total_rows = 0;
+    int64_t n_tiles = 0;
+    // device-side descriptors
+    DevCol *key_cols = nullptr;   // [k]
+    DevCol *seq_cols = nullptr;   // [k]
+    DevCol *kind_cols = nullptr;  // [k]
+    DevCol *all_cols = nullptr;   // [k * n_cols] run-major
+    int64_t *lens_dev = nullptr;
+    int32_t *cuts = nullptr;
+    uint32_t *winners = nullptr;
+    int32_t *tile_counts = nullptr;
+    int64_t *tile_offsets = nullptr;
+    int64_t *total_dev = nullptr;
+    bool any_dict = false;
+};
+
+}  // namespace pmh
+
+struct pmh_session_t {
+    int device = -1;
+};
+
+struct pmh_plan_t {
+    pmh_session_t *session = nullptr;
+    hipStream_t stream = nullptr;
+    pmh::DeviceBufs bufs;
+    std::vector<pmh::ColSpec> cols;  // key..., seq, kind, value...
+    int n_key_cols = 1;
+    bool drop_delete = true;
+    bool ignore_delete = false;
+    bool host_output = false;
+    std::vector<pmh::Section> sections;
+    size_t cur_section = 0;
+    int64_t rows_in_total = 0;
+    int64_t encoded_bytes_total = 0;
+    // output buffers (reused per section)
+    std::vector<void *> out_dev;
+    void **out_ptrs_dev = nullptr;
+    uint8_t *col_dtype_dev = nullptr;
+    std::vector<std::vector<uint8_t>> out_host;
+    std::vector<pmh_col> batch_cols;
+    std::vector<std::string> col_names;
+    pmh_stats stats{};
+    double h2d_ms = 0;
+};
+
+namespace pmh {
+
+static int dtype_from_str(const std::string &s) {
+    if (s == "int8" || s == "tinyint") return PMH_DT_INT8;
+    if (s == "int16" || s == "smallint") return PMH_DT_INT16;
+    if (s == "int32" || s == "int") return PMH_DT_INT32;
+    if (s == "int64" || s == "bigint") return PMH_DT_INT64;
+    if (s == "float" || s == "float32") return PMH_DT_FLOAT32;
+    if (s == "double" || s == "float64") return PMH_DT_FLOAT64;
+    return -1;
+}
+
+static int dtype_out_esize(int dt) {
+    switch (dt) {
+    case PMH_DT_INT8: return 1;
+    case PMH_DT_INT16: return 2;
+    case PMH_DT_INT32: return 4;
+    case PMH_DT_INT64: return 8;
+    case PMH_DT_FLOAT32: return 4;
+    case PMH_DT_FLOAT64: return 8;
+    }
+    return 0;
+}
+
+static int dtype_stored_esize(int dt) {
+    // parquet physical widths: INT8/16/32 stored as INT32
+    switch (dt) {
+    case PMH_DT_INT8:
+    case PMH_DT_INT16:
+    case PMH_DT_INT32: return 4;
+    case PMH_DT_INT64: return 8;
+    case PMH_DT_FLOAT32: return 4;
+    case PMH_DT_FLOAT64: return 8;
+    }
+    return 0;
+}
+
+// Walk an RLE/bit-packed def-level stream (bit width 1) and report whether
+// any zero (null) occurs among the first n values.
+static bool def_levels_have_nulls(const uint8_t *s, int64_t len, int64_t n) {
+    int64_t p = 0, cnt = 0;
+    while (cnt < n && p < len) {
+        uint64_t header = 0;
+        int shift = 0;
+        for (;;) {
+            if (p >= len) return true;  // malformed: treat as nulls
+            uint8_t b = s[p++];
+            header |= (uint64_t)(b & 0x7F) << shift;
+            if (!(b & 0x80)) break;
+            shift += 7;
+        }
+        if ((header & 1) == 0) {
+            int64_t count = (int64_t)(header >> 1);
+            if (p >= len) return true;
+            uint8_t v = s[p++];
+            if (count > n - cnt) count = n - cnt;
+            if (v == 0 && count > 0) return true;
+            cnt += count;
+        } else {
+            int64_t groups = (int64_t)(header >> 1);
+            for (int64_t g = 0; g < groups && cnt < n; g++) {
+                if (p >= len) return true;
+                uint8_t v = s[p++];
+                int64_t take = std::min<int64_t>(8, n - cnt);
+                uint8_t mask = take >= 8 ? 0xFF : (uint8_t)((1u << take) - 1);
+                if ((v & mask) != mask) return true;
+                cnt += take;
+            }
+        }
+    }
+    return cnt < n ? true : false;
+}
+
+// Prescan an RLE stream into device work chunks (splitting long runs).
+static bool prescan_rle(const uint8_t *s, int64_t len, int bit_width,
+                        int64_t n, int64_t out_base, uint64_t dev_base,
+                        int64_t host_off, std::vector<RleChunk> &out) {
+    int64_t p = 0, cnt = 0;
+    int byte_width = (bit_width + 7) / 8;
+    const int64_t MAX_CHUNK = 16384;
+    while (cnt < n) {
+        if (p >= len) {
+            set_error("RLE stream overrun");
+            return false;
+        }
+        uint64_t header = 0;
+        int shift = 0;
+        for (;;) {
+            uint8_t b = s[p++];
+            header |= (uint64_t)(b & 0x7F) << shift;
+            if (!(b & 0x80)) break;
+            shift += 7;
+        }
+        if ((header & 1) == 0) {
+            int64_t count = std::min<int64_t>((int64_t)(header >> 1), n - cnt);
+            uint32_t v = 0;
+            for (int i = 0; i < byte_width; i++) v |= (uint32_t)s[p + i] << (8 * i);
+            p += byte_width;
+            for (int64_t off = 0; off < count; off += MAX_CHUNK) {
+                RleChunk c{};
+                c.kind = 0;
+                c.value = v;
+                c.out_start = out_base + cnt + off;
+                c.count = (int32_t)std::min<int64_t>(MAX_CHUNK, count - off);
+                c.bit_width = bit_width;
+                out.push_back(c);
+            }
+            cnt += count;
+        } else {
+            int64_t groups = (int64_t)(header >> 1);
+            int64_t vals = std::min<int64_t>(groups * 8, n - cnt);
+            // split at group boundaries
+            for (int64_t voff = 0; voff < vals; voff += MAX_CHUNK) {
+                RleChunk c{};
+                c.kind = 1;
+                c.src = dev_base + (uint64_t)(host_off + p) +
+                        (uint64_t)((voff / 8) * bit_width);
+                c.out_start = out_base + cnt + voff;
+                c.count = (int32_t)std::min<int64_t>(MAX_CHUNK, vals - voff);
+                c.bit_width = bit_width;
+                out.push_back(c);
+            }
+            p += groups * bit_width;
+            cnt += vals;
+        }
+    }
+    return true;
+}
+
+struct StagedFile {
+    std::vector<uint8_t> data;  // whole file (or decompressed payload view)
+    ParquetFileMeta meta;
+};
+
+static bool load_file(const std::string &path, StagedFile &sf) {
+    std::ifstream f(path, std::ios::binary | std::ios::ate);
+    if (!f) {
+        set_error("cannot open %s", path.c_str());
+        return false;
+    }
+    int64_t n = f.tellg();
+    f.seekg(0);
+    sf.data.resize(n);
+    f.read((char *)sf.data.data(), n);
+    sf.meta = parse_parquet_footer(sf.data.data(), n);
+    if (!sf.meta.ok()) {
+        set_error("%s: %s", path.c_str(), sf.meta.error.c_str());
+        return false;
+    }
+    std::string err;
+    for (auto &rg : sf.meta.row_groups)
+        for (auto &cc : rg.columns)
+            if (!scan_chunk_pages(sf.data.data(), n, cc, err)) {
+                set_error("%s: %s", path.c_str(), err.c_str());
+                return false;
+            }
+    return true;
+}
+
+// Stage one run (list of files) for the required columns onto the device.
+static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
+                      Run &run) {
+    const auto &cols = plan->cols;
+    run.cols.resize(cols.size());
+    int64_t row_base = 0;
+    for (const auto &fd : files) {
+        StagedFile sf;
+        if (!load_file(fd.path, sf)) return false;
+        // map required column -> leaf index
+        std::vector<int> leaf(cols.size(), -1);
+        for (size_t c = 0; c < cols.size(); c++) {
+            for (size_t i = 0; i < sf.meta.schema_names.size(); i++)
+                if (sf.meta.schema_names[i] == cols[c].name) leaf[c] = (int)i;
+            if (leaf[c] < 0) {
+                set_error("%s: column %s not found", fd.path.c_str(),
+                          cols[c].name.c_str());
+                return false;
+            }
+        }
+        int64_t rg_row = 0;
+        for (auto &rg : sf.meta.row_groups) {
+            for (size_t c = 0; c < cols.size(); c++) {
+                auto &cc = rg.columns[leaf[c]];
+                RunCol &rc = run.cols[c];
+                int stored = cols[c].stored_esize;
+                int max_def = sf.meta.max_def_levels[leaf[c]];
+                // Build an uncompressed payload image of this chunk on host:
+                // for UNCOMPRESSED files this is a view into the raw chunk
+                // (uploaded verbatim); for ZSTD we decompress page payloads
+                // into a packed buffer.
+                int64_t chunk_start = cc.dictionary_page_offset
+                                          ? cc.dictionary_page_offset
+                                          : cc.data_page_offset;
+                std::vector<uint8_t> packed;  // zstd only
+                std::vector<int64_t> page_payload_off(cc.pages.size());
+                const uint8_t *payload_base = sf.data.data() + chunk_start;
+                int64_t payload_len = cc.total_compressed_size;
+                if (cc.codec == CODEC_ZSTD) {
+                    int64_t total_unc = 0;
+                    for (auto &pg : cc.pages) total_unc += pg.uncompressed_size;
+                    packed.resize(total_unc);
+                    int64_t off = 0;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                        auto &pg = cc.pages[pi];
+                        if (!zstd_decompress(sf.data.data() + pg.data_off,
+                                             pg.compressed_size,
+                                             packed.data() + off,
+                                             pg.uncompressed_size))
+                            return false;
+                        page_payload_off[pi] = off;
+                        off += pg.uncompressed_size;
+                    }
+                    payload_base = packed.data();
+                    payload_len = packed.size();
+                } else if (cc.codec == CODEC_UNCOMPRESSED) {
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++)
+                        page_payload_off[pi] =
+                            cc.pages[pi].data_off - chunk_start;
+                } else {
+                    set_error("%s: unsupported codec %d", fd.path.c_str(),
+                              cc.codec);
+                    return false;
+                }
+                // upload the payload image
+                void *dev = plan->bufs.alloc(payload_len);
+                if (!dev) {
+                    set_error("hipMalloc failed (%lld bytes)",
+                              (long long)payload_len);
+                    return false;
+                }
+                if (hipMemcpy(dev, payload_base, payload_len,
+                              hipMemcpyHostToDevice) != hipSuccess) {
+                    set_error("H2D failed");
+                    return false;
+                }
+                plan->encoded_bytes_total += payload_len;
+                // walk pages: compute per-page value offsets and encodings
+                const uint8_t *dict_host = nullptr;
+                int64_t dict_count = 0;
+                bool chunk_dict = false;
+                for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                    auto &pg = cc.pages[pi];
+                    const uint8_t *pp = payload_base + page_payload_off[pi];
+                    if (pg.page_type == 2) {  // dictionary page (PLAIN)
+                        dict_host = pp;
+                        dict_count = pg.num_values;
+                        continue;
+                    }
+                    int64_t pos = 0;
+                    if (max_def > 0) {
+                        uint32_t dl_len;
+                        memcpy(&dl_len, pp, 4);
+                        if (def_levels_have_nulls(pp + 4, dl_len,
+                                                  pg.num_values)) {
+                            set_error(
+                                "%s col %s: null values present — the GPU "
+                                "dedup path currently requires non-null "
+                                "columns (partial-update/null support is a "
+                                "later round; see DESIGN.md)",
+                                fd.path.c_str(), cols[c].name.c_str());
+                            return false;
+                        }
+                        pos = 4 + dl_len;
+                    }
+                    if (pg.encoding == ENC_PLAIN) {
+                        DevPage dp;
+                        dp.addr = (uint64_t)dev + page_payload_off[pi] + pos;
+                        dp.start_row = row_base + rg_row + pg.first_row;
+                        rc.pages_host.push_back(dp);
+                    } else if (pg.encoding == ENC_RLE_DICTIONARY ||
+                               pg.encoding == ENC_PLAIN_DICTIONARY) {
+                        chunk_dict = true;
+                        int bw = pp[pos];
+                        if (!prescan_rle(pp + pos + 1,
+                                         (pg.uncompressed_size ? pg.uncompressed_size
+                                                               : payload_len) -
+                                             pos - 1,
+                                         bw, pg.num_values,
+                                         row_base + rg_row + pg.first_row,
+                                         (uint64_t)dev,
+                                         page_payload_off[pi] + pos + 1,
+                                         rc.rle_host))
+                            return false;
+                    } else {
+                        set_error("%s: unsupported encoding %d",
+                                  fd.path.c_str(), pg.encoding);
+                        return false;
+                    }
+                }
+                if (chunk_dict) {
+                    rc.dict_encoded = true;
+                    if (!dict_host) {
+                        set_error("%s: dictionary page missing",
+                                  fd.path.c_str());
+                        return false;
+                    }
+                    if (!rc.dict_dev) {
+                        // v1: one dictionary per run-col (single file, single
+                        // row group chunks share dictionaries per chunk only)
+                        rc.dict_dev = plan->bufs.alloc(dict_count * stored);
+                        if (!rc.dict_dev) return false;
+                        if (hipMemcpy(rc.dict_dev, dict_host,
+                                      dict_count * stored,
+                                      hipMemcpyHostToDevice) != hipSuccess) {
+                            set_error("H2D dict failed");
+                            return false;
+                        }
+                    } else {
+                        set_error(
+                            "%s: multiple dictionary chunks per run column "
+                            "not supported yet",
+                            fd.path.c_str());
+                        return false;
+                    }
+                }
+                (void)stored;
+            }
+            rg_row += rg.num_rows;
+        }
+        if (rg_row != fd.row_count) {
+            set_error("%s: rowCount %lld != file rows %lld", fd.path.c_str(),
+                      (long long)fd.row_count, (long long)rg_row);
+            return false;
+        }
+        row_base += rg_row;
+    }
+    run.length = row_base;
+    // finalize device page tables / dictionary materialization buffers
+    for (size_t c = 0; c < run.cols.size(); c++) {
+        RunCol &rc = run.cols[c];
+        rc.n_rows = run.length;
+        if (rc.dict_encoded) {
+            int stored = plan->cols[c].stored_esize;
+            rc.ids_dev = (int32_t *)plan->bufs.alloc(run.length * 4);
+            rc.contig = plan->bufs.alloc(run.length * stored);
+            rc.rle_dev = (RleChunk *)plan->bufs.alloc(rc.rle_host.size() *
+                                                      sizeof(RleChunk));
+            if (!rc.ids_dev || !rc.contig || !rc.rle_dev) return false;
+            if (hipMemcpy(rc.rle_dev, rc.rle_host.data(),
+                          rc.rle_host.size() * sizeof(RleChunk),
+                          hipMemcpyHostToDevice) != hipSuccess)
+                return false;
+            DevPage dp{(uint64_t)rc.contig, 0};
+            rc.pages_host.assign(1, dp);
+        }
+        rc.n_pages = (int)rc.pages_host.size();
+        rc.pages_dev =
+            (DevPage *)plan->bufs.alloc(rc.pages_host.size() * sizeof(DevPage));
+        if (!rc.pages_dev) return false;
+        if (hipMemcpy(rc.pages_dev, rc.pages_host.data(),
+                      rc.pages_host.size() * sizeof(DevPage),
+                      hipMemcpyHostToDevice) != hipSuccess)
+            return false;
+    }
+    return true;
+}
+
+static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
+    int k = (int)sec.runs.size();
+    int n_cols = (int)plan->cols.size();
+    std::vector<DevCol> keyv(k), seqv(k), kindv(k), allv(k * n_cols);
+    std::vector<int64_t> lens(k);
+    int seq_idx = plan->n_key_cols;
+    int kind_idx = plan->n_key_cols + 1;
+    for (int r = 0; r < k; r++) {
+        Run &run = sec.runs[r];
+        lens[r] = run.length;
+        sec.total_rows += run.length;
+        for (int c = 0; c < n_cols; c++) {
+            DevCol dc{run.cols[c].pages_dev, run.cols[c].n_pages,
+                      plan->cols[c].stored_esize};
+            allv[r * n_cols + c] = dc;
+            if (c == 0) keyv[r] = dc;  // v1: single int64 key column
+            if (c == seq_idx) seqv[r] = dc;
+            if (c == kind_idx) kindv[r] = dc;
+        }
+    }
+    sec.n_tiles = (sec.total_rows + PMH_TILE_ROWS - 1) / PMH_TILE_ROWS;
+    if (sec.n_tiles == 0) sec.n_tiles = 1;
+    auto up = [&](const void *host, size_t n) -> void * {
+        void *d = plan->bufs.alloc(n);
+        if (d && host) hipMemcpy(d, host, n, hipMemcpyHostToDevice);
+        return d;
+    };
+    sec.key_cols = (DevCol *)up(keyv.data(), k * sizeof(DevCol));
+    sec.seq_cols = (DevCol *)up(seqv.data(), k * sizeof(DevCol));
+    sec.kind_cols = (DevCol *)up(kindv.data(), k * sizeof(DevCol));
+    sec.all_cols = (DevCol *)up(allv.data(), allv.size() * sizeof(DevCol));
+    sec.lens_dev = (int64_t *)up(lens.data(), k * sizeof(int64_t));
+    sec.cuts = (int32_t *)plan->bufs.alloc((sec.n_tiles + 1) * k * 4);
+    sec.winners =
+        (uint32_t *)plan->bufs.alloc(sec.n_tiles * PMH_TILE_ROWS * 4);
+    sec.tile_counts = (int32_t *)plan->bufs.alloc(sec.n_tiles * 4);
+    sec.tile_offsets = (int64_t *)plan->bufs.alloc(sec.n_tiles * 8);
+    sec.total_dev = (int64_t *)plan->bufs.alloc(8);
+    for (auto &run : sec.runs)
+        for (auto &rc : run.cols) sec.any_dict |= rc.dict_encoded;
+    return sec.key_cols && sec.seq_cols && sec.kind_cols && sec.all_cols &&
+           sec.lens_dev && sec.cuts && sec.winners && sec.tile_counts &&
+           sec.tile_offsets && sec.total_dev;
+}
+
+}  // namespace pmh
+
+// ==================================================================== C ABI
+
+using namespace pmh;
+
+extern "C" {
+
+const char *pmh_last_error(void) { return last_error().c_str(); }
+
+pmh_session_t *pmh_open_session(int device) {
+    auto *s = new pmh_session_t();
+    s->device = device;
+    if (device >= 0) {
+        if (hipSetDevice(device) != hipSuccess) {
+            set_error("hipSetDevice(%d) failed", device);
+            delete s;
+            return nullptr;
+        }
+    }
+    return s;
+}
+
+void pmh_close_session(pmh_session_t *s) { delete s; }
+
+pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
+    if (!s) {
+        set_error("null session");
+        return nullptr;
+    }
+    if (s->device < 0) {
+        set_error("pmh_plan_create requires a GPU session (device >= 0)");
+        return nullptr;
+    }
+    hipSetDevice(s->device);
+    std::unique_ptr<pmh_plan_t> plan(new pmh_plan_t());
+    plan->session = s;
+    try {
+        Json j = JsonParser(plan_json).parse();
+        // columns: key cols, then seq, kind, then value cols
+        auto add_col = [&](const Json &cj) -> bool {
+            ColSpec cs;
+            cs.name = cj["name"].as_str();
+            cs.dtype = dtype_from_str(cj["type"].as_str());
+            if (cs.dtype < 0 || cs.name.empty()) {
+                set_error("bad column spec");
+                return false;
+            }
+            cs.out_esize = dtype_out_esize(cs.dtype);
+            cs.stored_esize = dtype_stored_esize(cs.dtype);
+            plan->cols.push_back(cs);
+            return true;
+        };
+        const Json &kc = j["key_cols"];
+        if (kc.arr.size() != 1 ||
+            dtype_from_str(kc.arr[0]["type"].as_str()) != PMH_DT_INT64) {
+            set_error("v1 supports exactly one int64 key column (got %zu)",
+                      kc.arr.size());
+            return nullptr;
+        }
+        plan->n_key_cols = 1;
+        for (const auto &c : kc.arr)
+            if (!add_col(c)) return nullptr;
+        {  // _SEQUENCE_NUMBER, _VALUE_KIND (SpecialFields.java:79-83)
+            ColSpec seq{"_SEQUENCE_NUMBER", PMH_DT_INT64, 8, 8};
+            ColSpec kind{"_VALUE_KIND", PMH_DT_INT8, 1, 4};
+            plan->cols.push_back(seq);
+            plan->cols.push_back(kind);
+        }
+        for (const auto &c : j["value_cols"].arr)
+            if (!add_col(c)) return nullptr;
+        std::string engine = j["merge_engine"].as_str("deduplicate");
+        if (engine != "deduplicate") {
+            set_error("merge engine '%s' not yet on the GPU path (v1: "
+                      "deduplicate; partial-update next)",
+                      engine.c_str());
+            return nullptr;
+        }
+        plan->drop_delete = j["drop_delete"].as_bool(true);
+        plan->ignore_delete = j["ignore_delete"].as_bool(false);
+        plan->host_output = j["output"].as_str("device") == "host";
+
+        std::vector<FileDesc> files;
+        int idx = 0;
+        for (const auto &fj : j["files"].arr) {
+            FileDesc fd;
+            fd.path = fj["path"].as_str();
+            fd.row_count = fj["rowCount"].as_i64();
+            fd.min_key = fj["minKey"].as_i64();
+            fd.max_key = fj["maxKey"].as_i64();
+            fd.level = (int)fj["level"].as_i64();
+            fd.input_index = idx++;
+            if (fd.row_count >= (1 << 28)) {
+                set_error("file %s exceeds 2^28 rows per run limit",
+                          fd.path.c_str());
+                return nullptr;
+            }
+            files.push_back(fd);
+        }
+        if (files.empty()) {
+            set_error("no files in plan");
+            return nullptr;
+        }
+
+        hipStreamCreate(&plan->stream);
+
+        auto sections = interval_partition(files);
+        auto t0 = std::chrono::steady_clock::now();
+        for (auto &sec_files : sections) {
+            Section sec;
+            if ((int)sec_files.size() > PMH_MAX_RUNS) {
+                set_error("section has %zu runs > %d (spill path is a later "
+                          "round, MergeSorter.java:112-125)",
+                          sec_files.size(), PMH_MAX_RUNS);
+                return nullptr;
+            }
+            sec.runs.resize(sec_files.size());
+            int64_t run_rows = 0;
+            for (size_t r = 0; r < sec_files.size(); r++) {
+                if (!stage_run(plan.get(), sec_files[r], sec.runs[r]))
+                    return nullptr;
+                run_rows += sec.runs[r].length;
+            }
+            if (!build_section_descriptors(plan.get(), sec)) {
+                set_error("device allocation failed for section");
+                return nullptr;
+            }
+            plan->rows_in_total += run_rows;
+            plan->sections.push_back(std::move(sec));
+        }
+        auto t1 = std::chrono::steady_clock::now();
+        plan->h2d_ms =
+            std::chrono::duration<double, std::milli>(t1 - t0).count();
+
+        // output buffers: sized for the largest section
+        int64_t max_rows = 0;
+        for (auto &sec : plan->sections)
+            max_rows = std::max(max_rows, sec.total_rows);
+        int n_cols = (int)plan->cols.size();
+        std::vector<void *> outs(n_cols);
+        std::vector<uint8_t> dts(n_cols);
+        for (int c = 0; c < n_cols; c++) {
+            outs[c] = plan->bufs.alloc(max_rows * plan->cols[c].out_esize);
+            if (!outs[c]) {
+                set_error("output allocation failed");
+                return nullptr;
+            }
+            dts[c] = (uint8_t)plan->cols[c].dtype;
+        }
+        plan->out_dev = outs;
+        plan->out_ptrs_dev = (void **)plan->bufs.alloc(n_cols * sizeof(void *));
+        plan->col_dtype_dev = (uint8_t *)plan->bufs.alloc(n_cols);
+        hipMemcpy(plan->out_ptrs_dev, outs.data(), n_cols * sizeof(void *),
+                  hipMemcpyHostToDevice);
+        hipMemcpy(plan->col_dtype_dev, dts.data(), n_cols,
+                  hipMemcpyHostToDevice);
+        for (auto &cs : plan->cols) plan->col_names.push_back(cs.name);
+    } catch (const std::exception &e) {
+        set_error("plan parse: %s", e.what());
+        return nullptr;
+    }
+    plan->stats.h2d_ms = plan->h2d_ms;
+    return plan.release();
+}
+
+int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
+    if (!p) {
+        set_error("null plan");
+        return -1;
+    }
+    if (p->cur_section >= p->sections.size()) return 0;
+    hipSetDevice(p->session->device);
+    Section &sec = p->sections[p->cur_section++];
+    hipStream_t st = p->stream;
+    int k = (int)sec.runs.size();
+    int n_cols = (int)p->cols.size();
+
+    hipEvent_t ev[6];
+    for (auto &e : ev) hipEventCreate(&e);
+
+    auto fail = [&](const char *what, hipError_t e) -> int64_t {
+        set_error("%s: %s", what, hipGetErrorString(e));
+        for (auto &evv : ev) hipEventDestroy(evv);
+        return -1;
+    };
+
+    hipEventRecord(ev[0], st);
+    // decode (dictionary materialization)
+    if (sec.any_dict) {
+        for (auto &run : sec.runs) {
+            for (size_t c = 0; c < run.cols.size(); c++) {
+                RunCol &rc = run.cols[c];
+                if (!rc.dict_encoded) continue;
+                hipError_t e = pmh_launch_rle_decode(
+                    rc.rle_dev, (int64_t)rc.rle_host.size(), rc.ids_dev, st);
+                if (e != hipSuccess) return fail("rle_decode", e);
+                e = pmh_launch_dict_gather(rc.ids_dev, rc.dict_dev, rc.n_rows,
+                                           rc.contig,
+                                           p->cols[c].stored_esize, st);
+                if (e != hipSuccess) return fail("dict_gather", e);
+            }
+        }
+    }
+    hipEventRecord(ev[1], st);
+    hipError_t e = pmh_launch_partition(sec.key_cols, sec.lens_dev, k,
+                                        PMH_TILE_ROWS, sec.n_tiles + 1,
+                                        sec.total_rows, sec.cuts, st);
+    if (e != hipSuccess) return fail("partition", e);
+    hipEventRecord(ev[2], st);
+    int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0);
+    e = pmh_launch_merge_tiles(sec.key_cols, sec.seq_cols, sec.kind_cols,
+                               sec.lens_dev, k, sec.cuts, sec.n_tiles,
+                               PMH_TILE_ROWS, flags, sec.winners,
+                               sec.tile_counts, st);
+    if (e != hipSuccess) return fail("merge_tiles", e);
+    e = pmh_launch_scan_tiles(sec.tile_counts, sec.n_tiles, sec.tile_offsets,
+                              sec.total_dev, st);
+    if (e != hipSuccess) return fail("scan_tiles", e);
+    hipEventRecord(ev[3], st);
+    e = pmh_launch_emit(sec.all_cols, p->col_dtype_dev, n_cols, k, sec.winners,
+                        sec.tile_counts, sec.tile_offsets, sec.n_tiles,
+                        PMH_TILE_ROWS, p->out_ptrs_dev, st);
+    if (e != hipSuccess) return fail("emit", e);
+    hipEventRecord(ev[4], st);
+
+    int64_t total = 0;
+    e = hipMemcpyAsync(&total, sec.total_dev, 8, hipMemcpyDeviceToHost, st);
+    if (e != hipSuccess) return fail("total D2H", e);
+    e = hipStreamSynchronize(st);
+    if (e != hipSuccess) return fail("stream sync", e);
+
+    float ms;
+    hipEventElapsedTime(&ms, ev[0], ev[1]);
+    p->stats.decode_ms += ms;
+    hipEventElapsedTime(&ms, ev[1], ev[2]);
+    p->stats.partition_ms += ms;
+    hipEventElapsedTime(&ms, ev[2], ev[3]);
+    p->stats.merge_ms += ms;
+    hipEventElapsedTime(&ms, ev[3], ev[4]);
+    p->stats.emit_ms += ms;
+    hipEventElapsedTime(&ms, ev[0], ev[4]);
+    p->stats.total_device_ms += ms;
+    for (auto &evv : ev) hipEventDestroy(evv);
+
+    p->stats.rows_in += sec.total_rows;
+    p->stats.rows_out += total;
+
+    // assemble batch
+    p->batch_cols.resize(n_cols);
+    if (p->host_output) {
+        p->out_host.resize(n_cols);
+        for (int c = 0; c < n_cols; c++) {
+            p->out_host[c].resize(total * p->cols[c].out_esize);
+            if (total > 0 &&
+                hipMemcpy(p->out_host[c].data(), p->out_dev[c],
+                          total * p->cols[c].out_esize,
+                          hipMemcpyDeviceToHost) != hipSuccess) {
+                set_error("D2H output failed");
+                return -1;
+            }
+        }
+    }
+    for (int c = 0; c < n_cols; c++) {
+        pmh_col &pc = p->batch_cols[c];
+        pc.name = p->col_names[c].c_str();
+        pc.dtype = p->cols[c].dtype;
+        pc.data = p->host_output ? (const void *)p->out_host[c].data()
+                                 : (const void *)p->out_dev[c];
+        pc.valid = nullptr;
+    }
+    if (out) {
+        out->n_rows = total;
+        out->n_cols = n_cols;
+        out->device = p->host_output ? -1 : p->session->device;
+        out->cols = p->batch_cols.data();
+    }
+    return total;
+}
+
+int pmh_plan_close(pmh_plan_t *p) {
+    if (!p) return 0;
+    if (p->stream) hipStreamDestroy(p->stream);
+    delete p;
+    return 0;
+}
+
+int pmh_stats_get(pmh_plan_t *p, pmh_stats *out) {
+    if (!p || !out) return -1;
+    p->stats.hbm_bytes_algo = 0;  // filled by bench from encoded+output sizes
+    *out = p->stats;
+    out->hbm_bytes_algo = p->encoded_bytes_total;
+    return 0;
+}
+
+void pmh_free_string(char *s) { free(s); }
+
+char *pmh_debug_footer_json(const char *path) {
+    StagedFile sf;
+    if (!load_file(path, sf)) return nullptr;
+    std::string out = "{\"num_rows\": " + std::to_string(sf.meta.num_rows) +
+                      ", \"columns\": [";
+    for (size_t i = 0; i < sf.meta.schema_names.size(); i++) {
+        if (i) out += ",";
+        out += "{\"name\": \"" + json_escape(sf.meta.schema_names[i]) +
+               "\", \"max_def\": " + std::to_string(sf.meta.max_def_levels[i]) +
+               ", \"phys\": " + std::to_string(sf.meta.phys_types[i]) + "}";
+    }
+    out += "], \"row_groups\": [";
+    for (size_t g = 0; g < sf.meta.row_groups.size(); g++) {
+        auto &rg = sf.meta.row_groups[g];
+        if (g) out += ",";
+        out += "{\"num_rows\": " + std::to_string(rg.num_rows) + ", \"chunks\": [";
+        for (size_t c = 0; c < rg.columns.size(); c++) {
+            auto &cc = rg.columns[c];
+            if (c) out += ",";
+            int64_t n_data_pages = 0;
+            for (auto &pg : cc.pages)
+                if (pg.page_type == 0) n_data_pages++;
+            out += "{\"name\": \"" + json_escape(cc.name) +
+                   "\", \"num_values\": " + std::to_string(cc.num_values) +
+                   ", \"codec\": " + std::to_string(cc.codec) +
+                   ", \"data_page_offset\": " +
+                   std::to_string(cc.data_page_offset) +
+                   ", \"dict_page_offset\": " +
+                   std::to_string(cc.dictionary_page_offset) +
+                   ", \"n_data_pages\": " + std::to_string(n_data_pages) + "}";
+        }
+        out += "]}";
+    }
+    out += "]}";
+    return strdup(out.c_str());
+}
+
+int pmh_debug_interval_partition(int n, const int64_t *min_keys,
+                                 const int64_t *max_keys, int32_t *out_section,
+                                 int32_t *out_run) {
+    std::vector<FileDesc> files(n);
+    for (int i = 0; i < n; i++) {
+        files[i].min_key = min_keys[i];
+        files[i].max_key = max_keys[i];
+        files[i].input_index = i;
+    }
+    auto sections = interval_partition(files);
+    for (size_t s = 0; s < sections.size(); s++)
+        for (size_t r = 0; r < sections[s].size(); r++)
+            for (const auto &fd : sections[s][r]) {
+                out_section[fd.input_index] = (int32_t)s;
+                out_run[fd.input_index] = (int32_t)r;
+            }
+    return (int)sections.size();
+}
+
+}  // extern "C"

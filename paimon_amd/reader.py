@@ -1,0 +1,205 @@
+"""Python mirror of the reader surface over the C-ABI (ctypes).
+
+This is plumbing for tests/bench — the host side of the product is the C++
+in libpaimon_hip.so (the reference's host side is compiled (Java) code, so
+ours is C++ behind a C ABI; INTEGRATION.md shows the JNI binding a Java
+deployment would use instead of this module).
+
+Mirrors SplitRead<KeyValue>.createReader (operation/SplitRead.java:39-63) /
+RecordReader (paimon-common/.../reader/RecordReader.java:40-72): a
+MergeReadPlan yields one batch per section; batches reuse buffers
+(releaseBatch contract) — copy out if you keep them.
+
+The product path FAILS LOUDLY when the HIP library or a GPU is missing:
+there is no CPU fallback here (the CPU restatement lives in oracle/ and is
+test infrastructure only).
+"""
+
+import ctypes
+import json
+import os
+
+import numpy as np
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+LIB_PATH = os.path.join(_HERE, "libpaimon_hip.so")
+
+_DT_NP = {1: np.int8, 2: np.int16, 3: np.int32, 4: np.int64,
+          5: np.float32, 6: np.float64}
+
+
+class _Col(ctypes.Structure):
+    _fields_ = [("name", ctypes.c_char_p), ("dtype", ctypes.c_int32),
+                ("data", ctypes.c_void_p), ("valid", ctypes.c_void_p)]
+
+
+class _Batch(ctypes.Structure):
+    _fields_ = [("n_rows", ctypes.c_int64), ("n_cols", ctypes.c_int32),
+                ("device", ctypes.c_int32), ("cols", ctypes.POINTER(_Col))]
+
+
+class _Stats(ctypes.Structure):
+    _fields_ = [("rows_in", ctypes.c_int64), ("rows_out", ctypes.c_int64),
+                ("hbm_bytes_algo", ctypes.c_int64),
+                ("decode_ms", ctypes.c_double),
+                ("partition_ms", ctypes.c_double),
+                ("merge_ms", ctypes.c_double), ("emit_ms", ctypes.c_double),
+                ("total_device_ms", ctypes.c_double),
+                ("h2d_ms", ctypes.c_double)]
+
+
+_lib = None
+
+
+def load_lib(required=True):
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(LIB_PATH):
+        if required:
+            raise RuntimeError(
+                f"libpaimon_hip.so missing at {LIB_PATH} — build it with "
+                "`make -C paimon_amd/csrc` (no CPU fallback exists; the HIP "
+                "path is the product).")
+        return None
+    lib = ctypes.CDLL(LIB_PATH)
+    lib.pmh_open_session.restype = ctypes.c_void_p
+    lib.pmh_open_session.argtypes = [ctypes.c_int]
+    lib.pmh_close_session.argtypes = [ctypes.c_void_p]
+    lib.pmh_plan_create.restype = ctypes.c_void_p
+    lib.pmh_plan_create.argtypes = [ctypes.c_void_p, ctypes.c_char_p]
+    lib.pmh_read_next.restype = ctypes.c_int64
+    lib.pmh_read_next.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Batch)]
+    lib.pmh_plan_close.argtypes = [ctypes.c_void_p]
+    lib.pmh_stats_get.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Stats)]
+    lib.pmh_last_error.restype = ctypes.c_char_p
+    lib.pmh_debug_footer_json.restype = ctypes.c_void_p
+    lib.pmh_debug_footer_json.argtypes = [ctypes.c_char_p]
+    lib.pmh_free_string.argtypes = [ctypes.c_void_p]
+    lib.pmh_debug_interval_partition.restype = ctypes.c_int
+    lib.pmh_debug_interval_partition.argtypes = [
+        ctypes.c_int, ctypes.POINTER(ctypes.c_int64),
+        ctypes.POINTER(ctypes.c_int64), ctypes.POINTER(ctypes.c_int32),
+        ctypes.POINTER(ctypes.c_int32)]
+    _lib = lib
+    return lib
+
+
+def last_error():
+    return load_lib().pmh_last_error().decode()
+
+
+def debug_footer(path):
+    lib = load_lib()
+    p = lib.pmh_debug_footer_json(path.encode())
+    if not p:
+        raise RuntimeError(last_error())
+    s = ctypes.string_at(p).decode()
+    lib.pmh_free_string(p)
+    return json.loads(s)
+
+
+def interval_partition(min_keys, max_keys):
+    """Returns (section_id, run_id) per input file, per the IntervalPartition
+    restatement in libpaimon_hip (plan.cpp)."""
+    lib = load_lib()
+    n = len(min_keys)
+    mn = (ctypes.c_int64 * n)(*[int(x) for x in min_keys])
+    mx = (ctypes.c_int64 * n)(*[int(x) for x in max_keys])
+    sec = (ctypes.c_int32 * n)()
+    run = (ctypes.c_int32 * n)()
+    ns = lib.pmh_debug_interval_partition(n, mn, mx, sec, run)
+    if ns < 0:
+        raise RuntimeError(last_error())
+    return list(sec), list(run), ns
+
+
+class Session:
+    def __init__(self, device=0):
+        self.lib = load_lib()
+        self.h = self.lib.pmh_open_session(device)
+        if not self.h:
+            raise RuntimeError(f"pmh_open_session: {last_error()}")
+        self.device = device
+
+    def close(self):
+        if self.h:
+            self.lib.pmh_close_session(self.h)
+            self.h = None
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+
+class MergeReadPlan:
+    """One bucket's merge-on-read plan (MergeFileSplitRead.createReader)."""
+
+    def __init__(self, session: Session, files, key_cols, value_cols,
+                 merge_engine="deduplicate", drop_delete=True,
+                 ignore_delete=False, output="host"):
+        self.lib = session.lib
+        desc = {
+            "key_cols": key_cols,
+            "value_cols": value_cols,
+            "merge_engine": merge_engine,
+            "drop_delete": drop_delete,
+            "ignore_delete": ignore_delete,
+            "output": output,
+            "files": files,
+        }
+        self.h = self.lib.pmh_plan_create(session.h,
+                                          json.dumps(desc).encode())
+        if not self.h:
+            raise RuntimeError(f"pmh_plan_create: {last_error()}")
+        self.output = output
+
+    def read_next(self):
+        """Returns dict name -> numpy array (host output mode), or a raw
+        _Batch with device pointers (device mode); None at end of input."""
+        b = _Batch()
+        n = self.lib.pmh_read_next(self.h, ctypes.byref(b))
+        if n < 0:
+            raise RuntimeError(f"pmh_read_next: {last_error()}")
+        if n == 0 and b.n_cols == 0:
+            return None
+        if self.output == "device":
+            return b
+        out = {}
+        for c in range(b.n_cols):
+            col = b.cols[c]
+            dt = np.dtype(_DT_NP[col.dtype])
+            if b.n_rows and col.data:
+                buf = ctypes.cast(
+                    col.data,
+                    ctypes.POINTER(ctypes.c_uint8 * (b.n_rows * dt.itemsize)))
+                arr = np.frombuffer(bytes(buf.contents), dtype=dt)
+            else:
+                arr = np.empty(0, dtype=dt)
+            out[col.name.decode()] = arr
+        return out
+
+    def stats(self):
+        s = _Stats()
+        if self.lib.pmh_stats_get(self.h, ctypes.byref(s)) != 0:
+            raise RuntimeError(last_error())
+        return {f: getattr(s, f) for f, _ in s._fields_}
+
+    def close(self):
+        if self.h:
+            self.lib.pmh_plan_close(self.h)
+            self.h = None
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()
+
+
+def file_descs_from_metas(metas):
+    return [{"path": m["path"], "rowCount": m["rowCount"],
+             "minKey": m["minKey"], "maxKey": m["maxKey"],
+             "level": m.get("level", 0)} for m in metas]
