@@ -73,12 +73,13 @@ typedef struct pmh_batch {
 typedef struct pmh_stats {
     int64_t rows_in;         /* records fed to the merge */
     int64_t rows_out;        /* records emitted */
-    int64_t hbm_bytes_algo;  /* algorithmic bytes (encoded in + merged out) */
+    int64_t hbm_bytes_algo;  /* encoded input bytes staged in HBM */
     double decode_ms;        /* device time in decode kernels */
     double partition_ms;     /* device time in merge-path partition */
-    double merge_ms;         /* device time in merge/winner kernels */
+    double merge_ms;         /* device time in the tile merge kernel */
+    double scan_ms;          /* device time in the tile-offset scan */
     double emit_ms;          /* device time in gather/emit kernels */
-    double total_device_ms;  /* end-to-end device time of last read_next */
+    double total_device_ms;  /* end-to-end device time of read_next calls */
     double h2d_ms;           /* untimed-region staging cost, informational */
 } pmh_stats;
 
@@ -107,6 +108,11 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json);
 int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out);
 
 int pmh_plan_close(pmh_plan_t *p);
+
+/* Rewind the plan to its first section without restaging: the next
+ * pmh_read_next re-runs the full device pipeline on the resident encoded
+ * data. Used by benchmarks to repeat the timed region; stats accumulate. */
+int pmh_plan_reset(pmh_plan_t *p);
 
 int pmh_stats_get(pmh_plan_t *p, pmh_stats *out);
 

@@ -154,19 +154,18 @@ struct TileSmem {
     int32_t nemit;
 };
 
-// stable 2-way co-rank: position in A of the split for output rank d of
+// stable 2-way co-rank: number of A elements among the first d outputs of
 // merge(A, B), ties take A first. A/B are perm-index sequences; key lookup
-// through skey[.].
+// through skey[.]. For i in [max(0,d-lb), min(d,la)), j = d-1-i is always
+// in [0, lb). Advance while A[i] <= B[j] (A[i] belongs in the first d).
 DEV int32_t corank(int64_t d, const uint16_t *pa, int32_t la,
-                   const uint16_t *pb, int32_t lb, const int64_t *skey,
-                   const uint16_t *, int32_t) {
+                   const uint16_t *pb, int32_t lb, const int64_t *skey) {
     int64_t ilo = d > lb ? d - lb : 0;
     int64_t ihi = d < la ? d : la;
     while (ilo < ihi) {
         int64_t i = ilo + ((ihi - ilo) >> 1);
-        int64_t j = d - i - 1;  // candidate: a[i] vs b[j]
-        // a[i] goes after b[j] iff key(a[i]) > key(b[j])  (ties: A first)
-        if (j >= 0 && j < lb && skey[pa[i]] > skey[pb[j]]) ilo = i + 1;
+        int64_t j = d - 1 - i;
+        if (skey[pa[i]] <= skey[pb[j]]) ilo = i + 1;
         else ihi = i;
     }
     return (int32_t)ilo;
@@ -276,7 +275,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                 int n_out = lim < CH ? lim : CH;
                 const uint16_t *pa = &sm.perm[cur][abase];
                 const uint16_t *pb = &sm.perm[cur][abase + la];
-                int32_t ai = corank(d, pa, la, pb, lb, sm.skey, nullptr, 0);
+                int32_t ai = corank(d, pa, la, pb, lb, sm.skey);
                 int32_t bi = (int32_t)d - ai;
                 uint16_t *out = &sm.perm[nxt][abase + d];
                 for (int x = 0; x < n_out; x++) {

@@ -812,6 +812,8 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
 
     hipEvent_t ev[6];
     for (auto &e : ev) hipEventCreate(&e);
+    // ev: 0 start, 1 decode done, 2 partition done, 3 merge done,
+    //     4 scan done, 5 emit done
 
     auto fail = [&](const char *what, hipError_t e) -> int64_t {
         set_error("%s: %s", what, hipGetErrorString(e));
@@ -848,15 +850,16 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                                PMH_TILE_ROWS, flags, sec.winners,
                                sec.tile_counts, st);
     if (e != hipSuccess) return fail("merge_tiles", e);
+    hipEventRecord(ev[3], st);
     e = pmh_launch_scan_tiles(sec.tile_counts, sec.n_tiles, sec.tile_offsets,
                               sec.total_dev, st);
     if (e != hipSuccess) return fail("scan_tiles", e);
-    hipEventRecord(ev[3], st);
+    hipEventRecord(ev[4], st);
     e = pmh_launch_emit(sec.all_cols, p->col_dtype_dev, n_cols, k, sec.winners,
                         sec.tile_counts, sec.tile_offsets, sec.n_tiles,
                         PMH_TILE_ROWS, p->out_ptrs_dev, st);
     if (e != hipSuccess) return fail("emit", e);
-    hipEventRecord(ev[4], st);
+    hipEventRecord(ev[5], st);
 
     int64_t total = 0;
     e = hipMemcpyAsync(&total, sec.total_dev, 8, hipMemcpyDeviceToHost, st);
@@ -872,8 +875,10 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     hipEventElapsedTime(&ms, ev[2], ev[3]);
     p->stats.merge_ms += ms;
     hipEventElapsedTime(&ms, ev[3], ev[4]);
+    p->stats.scan_ms += ms;
+    hipEventElapsedTime(&ms, ev[4], ev[5]);
     p->stats.emit_ms += ms;
-    hipEventElapsedTime(&ms, ev[0], ev[4]);
+    hipEventElapsedTime(&ms, ev[0], ev[5]);
     p->stats.total_device_ms += ms;
     for (auto &evv : ev) hipEventDestroy(evv);
 
@@ -910,6 +915,12 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         out->cols = p->batch_cols.data();
     }
     return total;
+}
+
+int pmh_plan_reset(pmh_plan_t *p) {
+    if (!p) return -1;
+    p->cur_section = 0;
+    return 0;
 }
 
 int pmh_plan_close(pmh_plan_t *p) {

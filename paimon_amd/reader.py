@@ -43,7 +43,8 @@ class _Stats(ctypes.Structure):
                 ("hbm_bytes_algo", ctypes.c_int64),
                 ("decode_ms", ctypes.c_double),
                 ("partition_ms", ctypes.c_double),
-                ("merge_ms", ctypes.c_double), ("emit_ms", ctypes.c_double),
+                ("merge_ms", ctypes.c_double), ("scan_ms", ctypes.c_double),
+                ("emit_ms", ctypes.c_double),
                 ("total_device_ms", ctypes.c_double),
                 ("h2d_ms", ctypes.c_double)]
 
@@ -180,6 +181,11 @@ class MergeReadPlan:
                 arr = np.empty(0, dtype=dt)
             out[col.name.decode()] = arr
         return out
+
+    def reset(self):
+        """Rewind to the first section without restaging (bench repeats)."""
+        if self.lib.pmh_plan_reset(self.h) != 0:
+            raise RuntimeError(last_error())
 
     def stats(self):
         s = _Stats()
