@@ -256,34 +256,39 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             int n_chunks = (M + CH - 1) / CH;
             for (int ch = tid; ch < n_chunks; ch += blockDim.x) {
                 int64_t o = (int64_t)ch * CH;  // global output rank
-                // locate pair: pair p covers [segoff[a0], segoff[b1])
-                // where a0 = p*2*width, b1 = min(a0+2*width, k)
+                int remaining = (int)(M - o < CH ? M - o : CH);
                 int p = 0;
-                // small loop over <= k/(2*width)+1 pairs
-                while ((p + 1) * 2 * width < k &&
-                       sm.segoff[(p + 1) * 2 * width] <= o)
-                    p++;
-                int a0 = p * 2 * width;
-                int amid = a0 + width < k ? a0 + width : k;
-                int b1 = a0 + 2 * width < k ? a0 + 2 * width : k;
-                int32_t abase = sm.segoff[a0];
-                int32_t la = sm.segoff[amid] - abase;
-                int32_t lb = sm.segoff[b1] - sm.segoff[amid];
-                int64_t d = o - abase;  // rank within pair
-                int32_t lim = la + lb - (int32_t)d;
-                if (lim <= 0) continue;
-                int n_out = lim < CH ? lim : CH;
-                const uint16_t *pa = &sm.perm[cur][abase];
-                const uint16_t *pb = &sm.perm[cur][abase + la];
-                int32_t ai = corank(d, pa, la, pb, lb, sm.skey);
-                int32_t bi = (int32_t)d - ai;
-                uint16_t *out = &sm.perm[nxt][abase + d];
-                for (int x = 0; x < n_out; x++) {
-                    bool takeA;
-                    if (ai >= la) takeA = false;
-                    else if (bi >= lb) takeA = true;
-                    else takeA = !(sm.skey[pa[ai]] > sm.skey[pb[bi]]);
-                    out[x] = takeA ? pa[ai++] : pb[bi++];
+                // a chunk may span several pairs (pairs can be tiny):
+                // walk pairs until the chunk's outputs are all produced
+                while (remaining > 0) {
+                    // pair p covers output ranks [segoff[a0], segoff[b1])
+                    while ((p + 1) * 2 * width < k &&
+                           sm.segoff[(p + 1) * 2 * width] <= o)
+                        p++;
+                    int a0 = p * 2 * width;
+                    int amid = a0 + width < k ? a0 + width : k;
+                    int b1 = a0 + 2 * width < k ? a0 + 2 * width : k;
+                    int32_t abase = sm.segoff[a0];
+                    int32_t la = sm.segoff[amid] - abase;
+                    int32_t lb = sm.segoff[b1] - sm.segoff[amid];
+                    int64_t d = o - abase;  // rank within pair
+                    int32_t lim = la + lb - (int32_t)d;
+                    if (lim <= 0) break;  // past the last pair
+                    int n_out = lim < remaining ? lim : remaining;
+                    const uint16_t *pa = &sm.perm[cur][abase];
+                    const uint16_t *pb = &sm.perm[cur][abase + la];
+                    int32_t ai = corank(d, pa, la, pb, lb, sm.skey);
+                    int32_t bi = (int32_t)d - ai;
+                    uint16_t *out = &sm.perm[nxt][abase + d];
+                    for (int x = 0; x < n_out; x++) {
+                        bool takeA;
+                        if (ai >= la) takeA = false;
+                        else if (bi >= lb) takeA = true;
+                        else takeA = !(sm.skey[pa[ai]] > sm.skey[pb[bi]]);
+                        out[x] = takeA ? pa[ai++] : pb[bi++];
+                    }
+                    o += n_out;
+                    remaining -= n_out;
                 }
             }
             cur = nxt;
