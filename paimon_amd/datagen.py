@@ -35,6 +35,22 @@ def _key_space(total_rows: int) -> int:
     return max(int(total_rows * 3 // 2), 16)
 
 
+def _sorted_unique_keys(rng, n, space):
+    """Sorted n unique keys ~uniform over [0, space). Fast path for large n:
+    draw with replacement, unique, random-subset — each key has a symmetric
+    inclusion probability; exact choice() for small n."""
+    if n < 500_000 or n * 2 > space:
+        return np.sort(rng.choice(space, size=n, replace=False)).astype(np.int64)
+    over = int(n * 1.15)
+    while True:
+        u = np.unique(rng.integers(0, space, size=over, dtype=np.int64))
+        if len(u) >= n:
+            break
+        over = int(over * 1.2)
+    sel = np.sort(np.argpartition(rng.random(len(u)), n)[:n])
+    return u[sel]
+
+
 def gen_runs_dedup(n_runs: int, rows_per_run: int, n_value_cols: int = 8,
                    seed: int = 42, delete_frac: float = 0.05):
     """Generate in-memory sorted runs for the Deduplicate configs (C1/C2).
@@ -51,8 +67,7 @@ def gen_runs_dedup(n_runs: int, rows_per_run: int, n_value_cols: int = 8,
     seqs_all = rng.permutation(total).astype(np.int64)
     runs = []
     for r in range(n_runs):
-        keys = rng.choice(space, size=rows_per_run, replace=False).astype(np.int64)
-        keys.sort()
+        keys = _sorted_unique_keys(rng, rows_per_run, space)
         seq = seqs_all[r * rows_per_run:(r + 1) * rows_per_run]
         kind = np.where(rng.random(rows_per_run) < delete_frac,
                         KIND_DELETE, KIND_INSERT).astype(np.int8)
