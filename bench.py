@@ -1,0 +1,256 @@
+#!/usr/bin/env python3
+"""Benchmark: Paimon merge-on-read hot path on MI355X (BASELINE.json metric:
+merged rows/sec at 8-run x 10M-row merge-on-read, Parquet, Deduplicate).
+
+A "step" is one full merge-on-read pass (partition + tiled k-way merge +
+dedup + emit) over one bucket's sorted runs, with the encoded column chunks
+already resident in HBM (staged once at plan creation, untimed; the
+PCIe-inclusive staging rate is reported separately and noted in DESIGN.md).
+
+Workload (default = BASELINE.json configs[1], the single-GPU metric config):
+8 sorted runs x 10M rows, int64 PK + 8 int32 value columns, Parquet v1
+pages, PLAIN (dictionary off), uncompressed, Deduplicate merge with
+drop-delete — synthetic seeded data with ~50% cross-run key collisions.
+
+Multi-GPU: buckets shard one-per-GPU (weak scaling, no collective —
+SURVEY.md §8e); each rank merges its own bucket; value aggregates all ranks.
+
+cpu_baseline: the CPU oracle (faithful single-thread C restatement of
+SortMergeReaderWithLoserTree + Deduplicate, oracle/merge_oracle.c) timed on
+this box's host cores over a bounded prefix sample of the same workload —
+a reported baseline, not the optimization target.
+"""
+
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+
+
+def get_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--runs", type=int, default=8)
+    ap.add_argument("--rows", type=int, default=10_000_000)
+    ap.add_argument("--vals", type=int, default=8)
+    ap.add_argument("--compression", default="NONE")
+    ap.add_argument("--seed", type=int, default=42)
+    ap.add_argument("--data-dir", default=os.path.join(REPO, "data"))
+    ap.add_argument("--cpu-baseline-rows", type=int, default=2_000_000,
+                    help="prefix rows per run for the CPU-oracle baseline; "
+                    "0 disables")
+    return ap.parse_args()
+
+
+def ensure_data(args, rank):
+    from paimon_amd.datagen import gen_runs_dedup, write_runs
+    tag = (f"c2_{args.runs}x{args.rows}v{args.vals}_"
+           f"{args.compression}_seed{args.seed}_rank{rank}")
+    out_dir = os.path.join(args.data_dir, tag)
+    manifest = os.path.join(out_dir, "files.json")
+    if os.path.exists(manifest):
+        with open(manifest) as f:
+            return json.load(f), out_dir
+    runs = gen_runs_dedup(args.runs, args.rows, n_value_cols=args.vals,
+                          seed=args.seed + rank)
+    metas = write_runs(runs, out_dir, compression=args.compression)
+    return metas, out_dir
+
+
+def cpu_baseline(args, metas_dir):
+    """Time the C oracle loser-tree dedup on a bounded prefix sample."""
+    from oracle import merge_dedup
+    from paimon_amd.datagen import gen_runs_dedup
+    rows = min(args.cpu_baseline_rows, args.rows)
+    runs = gen_runs_dedup(args.runs, args.rows, n_value_cols=0,
+                          seed=args.seed)
+    sample = [{"key": r["key"][:rows], "seq": r["seq"][:rows],
+               "kind": r["kind"][:rows]} for r in runs]
+    n = sum(len(r["key"]) for r in sample)
+    t0 = time.perf_counter()
+    merge_dedup(sample, drop_delete=True)
+    dt = time.perf_counter() - t0
+    return {
+        "value": n / dt,
+        "unit": "rows/s",
+        "cores": 1,
+        "kind": "port",
+        "sample": (f"{args.runs} runs x {rows}-row prefixes of the same "
+                   f"workload ({n} rows, {dt:.1f}s single-thread C "
+                   "loser-tree restatement)"),
+    }
+
+
+def main():
+    args = get_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = world if world > 1 else args.gpus
+    if world == 1 and args.gpus > 1:
+        # standalone multi-GPU invocation: run sequential replicas is wrong;
+        # the driver launches via torch.distributed.run. Run rank 0 only.
+        print("WARNING: --gpus>1 without torchrun; running 1 rank",
+              file=sys.stderr)
+        n_gpus = 1
+
+    import torch
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        tdist.init_process_group("nccl")
+        dist = tdist
+        torch.cuda.set_device(local_rank)
+
+    from paimon_amd import Session, MergeReadPlan, file_descs_from_metas
+
+    t_gen0 = time.perf_counter()
+    metas, out_dir = ensure_data(args, rank)
+    t_gen = time.perf_counter() - t_gen0
+
+    sess = Session(local_rank)
+    key_cols = [{"name": "_KEY_k", "type": "int64"}]
+    value_cols = ([{"name": "v_k", "type": "int64"}] +
+                  [{"name": f"v_c{i}", "type": "int32"}
+                   for i in range(args.vals)])
+    t_stage0 = time.perf_counter()
+    plan = MergeReadPlan(sess, file_descs_from_metas(metas), key_cols,
+                         value_cols, merge_engine="deduplicate",
+                         drop_delete=True, output="device")
+    t_stage = time.perf_counter() - t_stage0
+
+    def one_step():
+        plan.reset()
+        rows = 0
+        while True:
+            b = plan.read_next()
+            if b is None:
+                break
+            rows += b.n_rows
+        return rows
+
+    # warmup
+    rows_out = 0
+    for _ in range(args.warmup):
+        rows_out = one_step()
+    stats0 = plan.stats()
+
+    if dist:
+        dist.barrier()
+    torch.cuda.synchronize(local_rank)
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        rows_out = one_step()
+    torch.cuda.synchronize(local_rank)
+    if dist:
+        dist.barrier()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=f"cuda:{local_rank}")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    stats1 = plan.stats()
+    d = {k: stats1[k] - stats0[k] for k in stats1}
+    K = args.steps
+    rows_in = args.runs * args.rows  # per step per rank, by construction
+    U = rows_out
+
+    # per-step kernel times (HIP events accumulated in libpaimon_hip)
+    kms = {k: d[k] / K for k in
+           ("decode_ms", "partition_ms", "merge_ms", "scan_ms", "emit_ms",
+            "total_device_ms")}
+
+    # algorithmic bytes per kernel launch (DESIGN.md "Measurement"):
+    #   merge:  N*(key 8 + seq 8 + kind-as-stored 4) read
+    #           + U*4 winners + tiles*4 counts written
+    #   emit:   U*4 winners read + U*(8+8+4 + 8 + vals*4) gathered
+    #           + U*(8+8+1 + 8 + vals*4) written
+    n_tiles = (rows_in + 2047) // 2048
+    merge_bytes = rows_in * 20 + U * 4 + n_tiles * 4
+    emit_bytes = U * (4 + (28 + args.vals * 4) + (25 + args.vals * 4))
+    kernels = {
+        "merge": (kms["merge_ms"], merge_bytes),
+        "emit": (kms["emit_ms"], emit_bytes),
+    }
+    dom = max(kernels, key=lambda k: kernels[k][0])
+    dom_ms, dom_bytes = kernels[dom]
+    achieved = dom_bytes / (dom_ms / 1e3) / 1e9 if dom_ms > 0 else 0.0
+
+    # whole-pipeline algorithmic rate (encoded input + merged output over
+    # total device time) — the SURVEY §8d fused-kernel denominator
+    in_bytes = rows_in * (8 + 8 + 4 + 8 + args.vals * 4)
+    out_bytes = U * (8 + 8 + 1 + 8 + args.vals * 4)
+    pipe_gbs = ((in_bytes + out_bytes) / (kms["total_device_ms"] / 1e3) / 1e9
+                if kms["total_device_ms"] > 0 else 0.0)
+
+    value_per_rank = rows_in * K / elapsed
+    value = value_per_rank * n_gpus if world > 1 else value_per_rank
+
+    result = {
+        "metric": "merged rows/sec",
+        "value": value,
+        "unit": "rows/s",
+        "n_gpus": n_gpus,
+        "steps": K,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed / K * 1e3,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "int64",
+        "data": "synthetic",
+        "config": {
+            "workload": (f"merge-on-read: {args.runs} sorted runs x "
+                         f"{args.rows} rows, int64 PK + {args.vals} int32, "
+                         f"Parquet {args.compression}, Deduplicate, "
+                         "drop-delete (BASELINE.json configs[1])"),
+            "n_runs": args.runs,
+            "rows_per_run": args.rows,
+            "value_cols": args.vals,
+            "compression": args.compression,
+            "seed": args.seed,
+            "rows_out_per_step": int(U),
+            "parallelism": f"bucket-parallel dp{n_gpus}, no collectives",
+        },
+        "roofline": {
+            "bound": "hbm",
+            "kernel": dom,
+            "achieved": round(achieved, 1),
+            "peak": HBM_PEAK_GBS,
+            "unit": "GB/s",
+            "frac": round(achieved / HBM_PEAK_GBS, 4),
+            "traffic": None,
+        },
+        "kernels_ms_per_step": {k: round(v, 3) for k, v in kms.items()},
+        "pipeline_algorithmic_GBs": round(pipe_gbs, 1),
+        "stage_seconds": round(t_stage, 2),
+        "gen_seconds": round(t_gen, 2),
+    }
+
+    if rank == 0 and n_gpus == 1 and args.cpu_baseline_rows > 0:
+        result["cpu_baseline"] = cpu_baseline(args, out_dir)
+
+    plan.close()
+    sess.close()
+    if dist:
+        dist.destroy_process_group()
+    if rank == 0:
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()
