@@ -72,6 +72,8 @@ def load_lib(required=True):
     lib.pmh_read_next.restype = ctypes.c_int64
     lib.pmh_read_next.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Batch)]
     lib.pmh_plan_close.argtypes = [ctypes.c_void_p]
+    lib.pmh_plan_reset.argtypes = [ctypes.c_void_p]
+    lib.pmh_plan_reset.restype = ctypes.c_int
     lib.pmh_stats_get.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Stats)]
     lib.pmh_last_error.restype = ctypes.c_char_p
     lib.pmh_debug_footer_json.restype = ctypes.c_void_p
