@@ -19,15 +19,17 @@ constexpr int PMH_TILE_MAX = PMH_TILE_ROWS + PMH_MAX_RUNS;
 constexpr int PMH_TILE_ITER =
     (PMH_TILE_MAX + PMH_TILE_THREADS - 1) / PMH_TILE_THREADS;
 
-// A decoded or raw-PLAIN region of one column of one run. Access is by
-// global row within the run: binary search pages by start_row. A fully
-// materialized (decoded) column is a single-page DevCol.
+// A decoded or raw-PLAIN region of one column of one run. Columns staged by
+// plan.cpp are CONTIGUOUS (n_pages == 1, direct addressing via addr0 — the
+// PLAIN page payloads are packed per column at H2D time); the paged path
+// (binary search by start_row) remains for future non-contiguous encodings.
 struct DevPage {
     uint64_t addr;      // device pointer to first element
     int64_t start_row;  // first run-row covered by this page
 };
 
 struct DevCol {
+    uint64_t addr0;        // n_pages == 1 fast path: element 0 address
     const DevPage *pages;  // device array, sorted by start_row
     int32_t n_pages;
     int32_t esize;  // element size in bytes (stored width)

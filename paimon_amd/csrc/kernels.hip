@@ -29,7 +29,9 @@ namespace pmh {
 
 template <typename T>
 DEV T col_load(const DevCol &c, int64_t row) {
-    // last page with start_row <= row; n_pages is small (L2-resident table)
+    if (c.n_pages == 1)  // staged columns are contiguous: direct addressing
+        return *reinterpret_cast<const T *>(c.addr0 + (uint64_t)row * sizeof(T));
+    // general paged path: last page with start_row <= row
     int lo = 0, hi = c.n_pages - 1;
     while (lo < hi) {
         int mid = (lo + hi + 1) >> 1;
@@ -87,11 +89,17 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
         for (int r = 0; r < k; r++) cuts[b * k + r] = (int32_t)lens[r];
         return;
     }
-    // bisect for the smallest v with (# ukey <= v) >= D  (v = pivot key of
-    // the element at rank D-1... we need cuts after D elements: the first D
-    // elements in (key, run) order). Let v* = key of the D-th smallest
-    // element boundary: smallest v such that cnt_le(v) >= D.
-    uint64_t klo = 0, khi = 0xffffffffffffffffull;
+    // bisect for the smallest v with (# ukey <= v) >= D: the cut falls at
+    // the D-th smallest element's key. Domain initialized from the runs'
+    // actual min/max keys (v* is an existing key; cnt_le(max) = total >= D).
+    uint64_t klo = ~0ull, khi = 0;
+    for (int r = 0; r < k; r++) {
+        if (lens[r] == 0) continue;
+        uint64_t lo_k = ukey(col_load<int64_t>(keys[r], 0));
+        uint64_t hi_k = ukey(col_load<int64_t>(keys[r], lens[r] - 1));
+        if (lo_k < klo) klo = lo_k;
+        if (hi_k > khi) khi = hi_k;
+    }
     while (klo < khi) {
         uint64_t mid = klo + ((khi - klo) >> 1);
         int64_t cnt = 0;

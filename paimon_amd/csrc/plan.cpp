@@ -163,18 +163,29 @@ struct DeviceBufs {
     }
 };
 
-// one (run, column): page table + optional dictionary-decode work
+// one (run, column), staged as ONE contiguous device column:
+//  - PLAIN chunks: page value-payloads packed host-side, H2D straight into
+//    `contig` at the chunk's row offset (the encoded PLAIN bytes ARE the
+//    column values; packing is layout, not decode);
+//  - dictionary chunks: the RLE id streams stay encoded in HBM; at read
+//    time k_rle_decode + k_dict_gather materialize the chunk's row range of
+//    `contig` (decode_ms, inside the timed region).
+struct GatherTask {
+    int64_t start = 0;  // run-row offset of the chunk
+    int64_t n = 0;
+    void *dict_dev = nullptr;
+};
+
 struct RunCol {
+    void *contig = nullptr;
     std::vector<DevPage> pages_host;
     DevPage *pages_dev = nullptr;
     int n_pages = 0;
-    // dictionary path (materialized at read_next into `contig`)
-    bool dict_encoded = false;
+    bool dict_encoded = false;  // any dictionary-encoded chunk
     std::vector<RleChunk> rle_host;
     RleChunk *rle_dev = nullptr;
     int32_t *ids_dev = nullptr;
-    void *dict_dev = nullptr;
-    void *contig = nullptr;  // materialized column (single-page DevCol)
+    std::vector<GatherTask> gathers;
     int64_t n_rows = 0;
 };
 
@@ -395,11 +406,22 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                       Run &run) {
     const auto &cols = plan->cols;
     run.cols.resize(cols.size());
+    int64_t total_rows = 0;
+    for (const auto &fd : files) total_rows += fd.row_count;
+    for (size_t c = 0; c < cols.size(); c++) {
+        RunCol &rc = run.cols[c];
+        rc.contig = plan->bufs.alloc(total_rows * cols[c].stored_esize);
+        if (!rc.contig) {
+            set_error("hipMalloc failed (%lld rows col %s)",
+                      (long long)total_rows, cols[c].name.c_str());
+            return false;
+        }
+        plan->encoded_bytes_total += total_rows * cols[c].stored_esize;
+    }
     int64_t row_base = 0;
     for (const auto &fd : files) {
         StagedFile sf;
         if (!load_file(fd.path, sf)) return false;
-        // map required column -> leaf index
         std::vector<int> leaf(cols.size(), -1);
         for (size_t c = 0; c < cols.size(); c++) {
             for (size_t i = 0; i < sf.meta.schema_names.size(); i++)
@@ -415,19 +437,16 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
             for (size_t c = 0; c < cols.size(); c++) {
                 auto &cc = rg.columns[leaf[c]];
                 RunCol &rc = run.cols[c];
-                int stored = cols[c].stored_esize;
-                int max_def = sf.meta.max_def_levels[leaf[c]];
-                // Build an uncompressed payload image of this chunk on host:
-                // for UNCOMPRESSED files this is a view into the raw chunk
-                // (uploaded verbatim); for ZSTD we decompress page payloads
-                // into a packed buffer.
+                const int stored = cols[c].stored_esize;
+                const int max_def = sf.meta.max_def_levels[leaf[c]];
+                const int64_t chunk_row0 = row_base + rg_row;
+                // uncompressed payload image of this chunk
                 int64_t chunk_start = cc.dictionary_page_offset
                                           ? cc.dictionary_page_offset
                                           : cc.data_page_offset;
                 std::vector<uint8_t> packed;  // zstd only
-                std::vector<int64_t> page_payload_off(cc.pages.size());
+                std::vector<int64_t> ppo(cc.pages.size());
                 const uint8_t *payload_base = sf.data.data() + chunk_start;
-                int64_t payload_len = cc.total_compressed_size;
                 if (cc.codec == CODEC_ZSTD) {
                     int64_t total_unc = 0;
                     for (auto &pg : cc.pages) total_unc += pg.uncompressed_size;
@@ -440,45 +459,50 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                                              packed.data() + off,
                                              pg.uncompressed_size))
                             return false;
-                        page_payload_off[pi] = off;
+                        ppo[pi] = off;
                         off += pg.uncompressed_size;
                     }
                     payload_base = packed.data();
-                    payload_len = packed.size();
                 } else if (cc.codec == CODEC_UNCOMPRESSED) {
                     for (size_t pi = 0; pi < cc.pages.size(); pi++)
-                        page_payload_off[pi] =
-                            cc.pages[pi].data_off - chunk_start;
+                        ppo[pi] = cc.pages[pi].data_off - chunk_start;
                 } else {
                     set_error("%s: unsupported codec %d", fd.path.c_str(),
                               cc.codec);
                     return false;
                 }
-                // upload the payload image
-                void *dev = plan->bufs.alloc(payload_len);
-                if (!dev) {
-                    set_error("hipMalloc failed (%lld bytes)",
-                              (long long)payload_len);
-                    return false;
-                }
-                if (hipMemcpy(dev, payload_base, payload_len,
-                              hipMemcpyHostToDevice) != hipSuccess) {
-                    set_error("H2D failed");
-                    return false;
-                }
-                plan->encoded_bytes_total += payload_len;
-                // walk pages: compute per-page value offsets and encodings
+                // classify data pages
+                bool has_plain = false, has_dict = false;
                 const uint8_t *dict_host = nullptr;
                 int64_t dict_count = 0;
-                bool chunk_dict = false;
                 for (size_t pi = 0; pi < cc.pages.size(); pi++) {
                     auto &pg = cc.pages[pi];
-                    const uint8_t *pp = payload_base + page_payload_off[pi];
-                    if (pg.page_type == 2) {  // dictionary page (PLAIN)
-                        dict_host = pp;
+                    if (pg.page_type == 2) {
+                        dict_host = payload_base + ppo[pi];
                         dict_count = pg.num_values;
-                        continue;
+                    } else if (pg.encoding == ENC_PLAIN) {
+                        has_plain = true;
+                    } else if (pg.encoding == ENC_RLE_DICTIONARY ||
+                               pg.encoding == ENC_PLAIN_DICTIONARY) {
+                        has_dict = true;
+                    } else {
+                        set_error("%s: unsupported encoding %d",
+                                  fd.path.c_str(), pg.encoding);
+                        return false;
                     }
+                }
+                if (has_plain && has_dict) {
+                    set_error("%s col %s: mixed PLAIN/dictionary pages in one "
+                              "chunk not supported yet",
+                              fd.path.c_str(), cols[c].name.c_str());
+                    return false;
+                }
+                // per-page value offsets past def levels (+ null check)
+                std::vector<int64_t> vpos(cc.pages.size(), 0);
+                for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                    auto &pg = cc.pages[pi];
+                    if (pg.page_type != 0) continue;
+                    const uint8_t *pp = payload_base + ppo[pi];
                     int64_t pos = 0;
                     if (max_def > 0) {
                         uint32_t dl_len;
@@ -487,66 +511,91 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                                                   pg.num_values)) {
                             set_error(
                                 "%s col %s: null values present — the GPU "
-                                "dedup path currently requires non-null "
-                                "columns (partial-update/null support is a "
-                                "later round; see DESIGN.md)",
+                                "dedup path requires non-null columns "
+                                "(partial-update/null support is a later "
+                                "round; see DESIGN.md)",
                                 fd.path.c_str(), cols[c].name.c_str());
                             return false;
                         }
                         pos = 4 + dl_len;
                     }
-                    if (pg.encoding == ENC_PLAIN) {
-                        DevPage dp;
-                        dp.addr = (uint64_t)dev + page_payload_off[pi] + pos;
-                        dp.start_row = row_base + rg_row + pg.first_row;
-                        rc.pages_host.push_back(dp);
-                    } else if (pg.encoding == ENC_RLE_DICTIONARY ||
-                               pg.encoding == ENC_PLAIN_DICTIONARY) {
-                        chunk_dict = true;
-                        int bw = pp[pos];
-                        if (!prescan_rle(pp + pos + 1,
-                                         (pg.uncompressed_size ? pg.uncompressed_size
-                                                               : payload_len) -
-                                             pos - 1,
-                                         bw, pg.num_values,
-                                         row_base + rg_row + pg.first_row,
-                                         (uint64_t)dev,
-                                         page_payload_off[pi] + pos + 1,
-                                         rc.rle_host))
-                            return false;
-                    } else {
-                        set_error("%s: unsupported encoding %d",
-                                  fd.path.c_str(), pg.encoding);
+                    vpos[pi] = pos;
+                }
+                if (has_plain || (!has_dict && cc.num_values > 0)) {
+                    // pack PLAIN value payloads and copy into contig
+                    std::vector<uint8_t> pack(cc.num_values * stored);
+                    int64_t off = 0;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                        auto &pg = cc.pages[pi];
+                        if (pg.page_type != 0) continue;
+                        int64_t nbytes = (int64_t)pg.num_values * stored;
+                        memcpy(pack.data() + off,
+                               payload_base + ppo[pi] + vpos[pi], nbytes);
+                        off += nbytes;
+                    }
+                    if (off != (int64_t)pack.size()) {
+                        set_error("%s: page value counts disagree with chunk",
+                                  fd.path.c_str());
                         return false;
                     }
-                }
-                if (chunk_dict) {
-                    rc.dict_encoded = true;
+                    if (hipMemcpy((uint8_t *)rc.contig + chunk_row0 * stored,
+                                  pack.data(), off,
+                                  hipMemcpyHostToDevice) != hipSuccess) {
+                        set_error("H2D failed");
+                        return false;
+                    }
+                } else if (has_dict) {
                     if (!dict_host) {
                         set_error("%s: dictionary page missing",
                                   fd.path.c_str());
                         return false;
                     }
-                    if (!rc.dict_dev) {
-                        // v1: one dictionary per run-col (single file, single
-                        // row group chunks share dictionaries per chunk only)
-                        rc.dict_dev = plan->bufs.alloc(dict_count * stored);
-                        if (!rc.dict_dev) return false;
-                        if (hipMemcpy(rc.dict_dev, dict_host,
-                                      dict_count * stored,
-                                      hipMemcpyHostToDevice) != hipSuccess) {
-                            set_error("H2D dict failed");
-                            return false;
-                        }
-                    } else {
-                        set_error(
-                            "%s: multiple dictionary chunks per run column "
-                            "not supported yet",
-                            fd.path.c_str());
+                    rc.dict_encoded = true;
+                    // upload the id-stream payload image
+                    int64_t payload_len = 0;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++)
+                        payload_len =
+                            std::max(payload_len,
+                                     ppo[pi] + (cc.codec == CODEC_ZSTD
+                                                    ? cc.pages[pi].uncompressed_size
+                                                    : cc.pages[pi].compressed_size));
+                    void *dev = plan->bufs.alloc(payload_len);
+                    if (!dev) return false;
+                    if (hipMemcpy(dev, payload_base, payload_len,
+                                  hipMemcpyHostToDevice) != hipSuccess) {
+                        set_error("H2D failed");
                         return false;
                     }
+                    plan->encoded_bytes_total += payload_len;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                        auto &pg = cc.pages[pi];
+                        if (pg.page_type != 0) continue;
+                        const uint8_t *pp = payload_base + ppo[pi];
+                        int64_t pos = vpos[pi];
+                        int bw = pp[pos];
+                        int64_t plen = (cc.codec == CODEC_ZSTD
+                                            ? pg.uncompressed_size
+                                            : pg.compressed_size);
+                        if (!prescan_rle(pp + pos + 1, plen - pos - 1, bw,
+                                         pg.num_values,
+                                         chunk_row0 + pg.first_row,
+                                         (uint64_t)dev, ppo[pi] + pos + 1,
+                                         rc.rle_host))
+                            return false;
+                    }
+                    void *dict_dev = plan->bufs.alloc(dict_count * stored);
+                    if (!dict_dev) return false;
+                    if (hipMemcpy(dict_dev, dict_host, dict_count * stored,
+                                  hipMemcpyHostToDevice) != hipSuccess) {
+                        set_error("H2D dict failed");
+                        return false;
+                    }
+                    GatherTask gt;
+                    gt.start = chunk_row0;
+                    gt.n = cc.num_values;
+                    gt.dict_dev = dict_dev;
+                    rc.gathers.push_back(gt);
                 }
-                (void)stored;
             }
             rg_row += rg.num_rows;
         }
@@ -558,30 +607,25 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
         row_base += rg_row;
     }
     run.length = row_base;
-    // finalize device page tables / dictionary materialization buffers
     for (size_t c = 0; c < run.cols.size(); c++) {
         RunCol &rc = run.cols[c];
         rc.n_rows = run.length;
         if (rc.dict_encoded) {
-            int stored = plan->cols[c].stored_esize;
             rc.ids_dev = (int32_t *)plan->bufs.alloc(run.length * 4);
-            rc.contig = plan->bufs.alloc(run.length * stored);
             rc.rle_dev = (RleChunk *)plan->bufs.alloc(rc.rle_host.size() *
                                                       sizeof(RleChunk));
-            if (!rc.ids_dev || !rc.contig || !rc.rle_dev) return false;
+            if (!rc.ids_dev || !rc.rle_dev) return false;
             if (hipMemcpy(rc.rle_dev, rc.rle_host.data(),
                           rc.rle_host.size() * sizeof(RleChunk),
                           hipMemcpyHostToDevice) != hipSuccess)
                 return false;
-            DevPage dp{(uint64_t)rc.contig, 0};
-            rc.pages_host.assign(1, dp);
         }
-        rc.n_pages = (int)rc.pages_host.size();
-        rc.pages_dev =
-            (DevPage *)plan->bufs.alloc(rc.pages_host.size() * sizeof(DevPage));
+        DevPage dp{(uint64_t)rc.contig, 0};
+        rc.pages_host.assign(1, dp);
+        rc.n_pages = 1;
+        rc.pages_dev = (DevPage *)plan->bufs.alloc(sizeof(DevPage));
         if (!rc.pages_dev) return false;
-        if (hipMemcpy(rc.pages_dev, rc.pages_host.data(),
-                      rc.pages_host.size() * sizeof(DevPage),
+        if (hipMemcpy(rc.pages_dev, rc.pages_host.data(), sizeof(DevPage),
                       hipMemcpyHostToDevice) != hipSuccess)
             return false;
     }
@@ -600,8 +644,8 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
         lens[r] = run.length;
         sec.total_rows += run.length;
         for (int c = 0; c < n_cols; c++) {
-            DevCol dc{run.cols[c].pages_dev, run.cols[c].n_pages,
-                      plan->cols[c].stored_esize};
+            DevCol dc{(uint64_t)run.cols[c].contig, run.cols[c].pages_dev,
+                      run.cols[c].n_pages, plan->cols[c].stored_esize};
             allv[r * n_cols + c] = dc;
             if (c == 0) keyv[r] = dc;  // v1: single int64 key column
             if (c == seq_idx) seqv[r] = dc;
@@ -831,10 +875,13 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                 hipError_t e = pmh_launch_rle_decode(
                     rc.rle_dev, (int64_t)rc.rle_host.size(), rc.ids_dev, st);
                 if (e != hipSuccess) return fail("rle_decode", e);
-                e = pmh_launch_dict_gather(rc.ids_dev, rc.dict_dev, rc.n_rows,
-                                           rc.contig,
-                                           p->cols[c].stored_esize, st);
-                if (e != hipSuccess) return fail("dict_gather", e);
+                int es = p->cols[c].stored_esize;
+                for (const GatherTask &gt : rc.gathers) {
+                    e = pmh_launch_dict_gather(
+                        rc.ids_dev + gt.start, gt.dict_dev, gt.n,
+                        (uint8_t *)rc.contig + gt.start * es, es, st);
+                    if (e != hipSuccess) return fail("dict_gather", e);
+                }
             }
         }
     }
