@@ -44,24 +44,44 @@ DEV T col_load(const DevCol &c, int64_t row) {
 
 DEV uint64_t ukey(int64_t k) { return (uint64_t)k ^ 0x8000000000000000ull; }
 
-// #elements in run col (sorted unique keys) with ukey(e) <= v, searched in
-// window [lo, hi).
-DEV int64_t count_le(const DevCol &c, int64_t lo, int64_t hi, uint64_t v) {
-    while (lo < hi) {
-        int64_t mid = lo + ((hi - lo) >> 1);
-        if (ukey(col_load<int64_t>(c, mid)) <= v) lo = mid + 1;
-        else hi = mid;
+// Lockstep multi-run lower/upper bound: runs the k binary searches together
+// so their probe loads issue back-to-back each step (8-16x memory-level
+// parallelism vs sequential searches — the partition kernel was 83%
+// latency-parked on serial probes, profiles/r01_c2_pmc.md). Arrays are
+// indexed only by the unrolled compile-time r, so they stay in registers.
+// le=true: first index with ukey > v; le=false: first index with ukey >= v.
+template <bool LE>
+DEV void bound_multi(const uint64_t *addr, const int64_t *lo_in,
+                     const int64_t *hi_in, int k, uint64_t v, int64_t *out) {
+    int64_t lo[PMH_MAX_RUNS], hi[PMH_MAX_RUNS];
+#pragma unroll
+    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+        if (r >= k) continue;
+        lo[r] = lo_in[r];
+        hi[r] = hi_in[r];
     }
-    return lo;
-}
-
-DEV int64_t count_lt(const DevCol &c, int64_t lo, int64_t hi, uint64_t v) {
-    while (lo < hi) {
-        int64_t mid = lo + ((hi - lo) >> 1);
-        if (ukey(col_load<int64_t>(c, mid)) < v) lo = mid + 1;
-        else hi = mid;
+    bool any = true;
+    while (any) {
+        any = false;
+#pragma unroll
+        for (int r = 0; r < PMH_MAX_RUNS; r++) {
+            if (r >= k) continue;
+            if (lo[r] < hi[r]) {
+                int64_t mid = lo[r] + ((hi[r] - lo[r]) >> 1);
+                uint64_t kk = ukey(
+                    *reinterpret_cast<const int64_t *>(addr[r] + mid * 8));
+                bool go = LE ? (kk <= v) : (kk < v);
+                if (go) lo[r] = mid + 1;
+                else hi[r] = mid;
+                any |= lo[r] < hi[r];
+            }
+        }
     }
-    return lo;
+#pragma unroll
+    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+        if (r >= k) continue;
+        out[r] = lo[r];
+    }
 }
 
 // ------------------------------------------------------------ k_partition
@@ -78,60 +98,83 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
     int64_t D = b * tile_rows;
     if (D > total_rows) D = total_rows;
 
-    int64_t wlo[PMH_MAX_RUNS], whi[PMH_MAX_RUNS];
-    for (int r = 0; r < k; r++) { wlo[r] = 0; whi[r] = lens[r]; }
-
-    if (D == 0) {
-        for (int r = 0; r < k; r++) cuts[b * k + r] = 0;
-        return;
+    uint64_t addr[PMH_MAX_RUNS];
+    int64_t len[PMH_MAX_RUNS], wlo[PMH_MAX_RUNS], whi[PMH_MAX_RUNS];
+#pragma unroll
+    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+        if (r >= k) continue;
+        addr[r] = keys[r].addr0;  // staged columns are contiguous
+        len[r] = lens[r];
+        wlo[r] = 0;
+        whi[r] = len[r];
     }
-    if (D >= total_rows) {
-        for (int r = 0; r < k; r++) cuts[b * k + r] = (int32_t)lens[r];
+    if (D == 0 || D >= total_rows) {
+#pragma unroll
+        for (int r = 0; r < PMH_MAX_RUNS; r++) {
+            if (r >= k) continue;
+            cuts[b * k + r] = D == 0 ? 0 : (int32_t)len[r];
+        }
         return;
     }
     // bisect for the smallest v with (# ukey <= v) >= D: the cut falls at
     // the D-th smallest element's key. Domain initialized from the runs'
     // actual min/max keys (v* is an existing key; cnt_le(max) = total >= D).
     uint64_t klo = ~0ull, khi = 0;
-    for (int r = 0; r < k; r++) {
-        if (lens[r] == 0) continue;
-        uint64_t lo_k = ukey(col_load<int64_t>(keys[r], 0));
-        uint64_t hi_k = ukey(col_load<int64_t>(keys[r], lens[r] - 1));
+#pragma unroll
+    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+        if (r >= k || len[r] == 0) continue;
+        uint64_t lo_k = ukey(*reinterpret_cast<const int64_t *>(addr[r]));
+        uint64_t hi_k = ukey(
+            *reinterpret_cast<const int64_t *>(addr[r] + (len[r] - 1) * 8));
         if (lo_k < klo) klo = lo_k;
         if (hi_k > khi) khi = hi_k;
     }
+    int64_t pos[PMH_MAX_RUNS];
     while (klo < khi) {
         uint64_t mid = klo + ((khi - klo) >> 1);
+        bound_multi<true>(addr, wlo, whi, k, mid, pos);
         int64_t cnt = 0;
-        for (int r = 0; r < k; r++)
-            cnt += count_le(keys[r], wlo[r], whi[r], mid);
+#pragma unroll
+        for (int r = 0; r < PMH_MAX_RUNS; r++) {
+            if (r >= k) continue;
+            cnt += pos[r];
+        }
         if (cnt >= D) {
             khi = mid;
-            for (int r = 0; r < k; r++)
-                whi[r] = count_le(keys[r], wlo[r], whi[r], mid);
+#pragma unroll
+            for (int r = 0; r < PMH_MAX_RUNS; r++) {
+                if (r >= k) continue;
+                whi[r] = pos[r];
+            }
         } else {
             klo = mid + 1;
-            for (int r = 0; r < k; r++)
-                wlo[r] = count_le(keys[r], wlo[r], whi[r], mid);
+#pragma unroll
+            for (int r = 0; r < PMH_MAX_RUNS; r++) {
+                if (r >= k) continue;
+                wlo[r] = pos[r];
+            }
         }
     }
     // klo == v*: all elements with key < v* are taken; among key == v*
     // (<=1 per run), take the first t in run order. Searching within the
     // final windows yields absolute positions: everything below wlo has
     // key < v*, everything at/above whi has key > v*.
+    bound_multi<false>(addr, wlo, whi, k, klo, pos);
     int64_t base = 0;
-    int64_t lb[PMH_MAX_RUNS], has[PMH_MAX_RUNS];
-    for (int r = 0; r < k; r++) {
-        int64_t l = count_lt(keys[r], wlo[r], whi[r], klo);
-        lb[r] = l;
-        has[r] = (l < lens[r]) &&
-                 (ukey(col_load<int64_t>(keys[r], l)) == klo);
-        base += l;
+#pragma unroll
+    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+        if (r >= k) continue;
+        base += pos[r];
     }
     int64_t t = D - base;
-    for (int r = 0; r < k; r++) {
-        int64_t c = lb[r];
-        if (t > 0 && has[r]) { c++; t--; }
+#pragma unroll
+    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+        if (r >= k) continue;
+        int64_t c = pos[r];
+        bool has = (c < len[r]) &&
+                   (ukey(*reinterpret_cast<const int64_t *>(addr[r] + c * 8)) ==
+                    klo);
+        if (t > 0 && has) { c++; t--; }
         cuts[b * k + r] = (int32_t)c;
     }
 }
@@ -443,34 +486,63 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                        const uint32_t *winners, const int32_t *tile_counts,
                        const int64_t *tile_offsets, int64_t n_tiles,
                        int64_t tile_rows, void *const *out_ptrs) {
+    // 4 output rows per thread iteration: the 4 gathers per column are
+    // independent, quadrupling loads in flight (emit was 96% latency-parked,
+    // profiles/r01_c2_pmc.md).
+    constexpr int R = 4;
     for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
         int32_t cnt = tile_counts[tile];
         int64_t base = tile_offsets[tile];
         const uint32_t *w = &winners[tile * tile_rows];
-        for (int32_t j = threadIdx.x; j < cnt; j += blockDim.x) {
-            uint32_t packed = w[j];
-            int run = packed >> 28;
-            int64_t row = packed & 0x0fffffff;
-            int64_t o = base + j;
+        for (int32_t j0 = threadIdx.x * R; j0 < cnt;
+             j0 += (int32_t)blockDim.x * R) {
+            int run[R];
+            int64_t row[R];
+            int nr = cnt - j0 < R ? cnt - j0 : R;
+#pragma unroll
+            for (int x = 0; x < R; x++) {
+                uint32_t packed = x < nr ? w[j0 + x] : w[j0];
+                run[x] = packed >> 28;
+                row[x] = packed & 0x0fffffff;
+            }
+            int64_t o = base + j0;
             for (int c = 0; c < n_cols; c++) {
-                const DevCol &dc = cols[run * n_cols + c];
                 switch (col_dtype[c]) {
-                case 1:  // INT8 output from INT32-stored parquet TINYINT
-                    ((int8_t *)out_ptrs[c])[o] =
-                        (int8_t)col_load<int32_t>(dc, row);
+                case 1: {  // INT8 output from INT32-stored parquet TINYINT
+                    int32_t v[R];
+#pragma unroll
+                    for (int x = 0; x < R; x++)
+                        v[x] = col_load<int32_t>(cols[run[x] * n_cols + c],
+                                                 row[x]);
+#pragma unroll
+                    for (int x = 0; x < R; x++)
+                        if (x < nr) ((int8_t *)out_ptrs[c])[o + x] = (int8_t)v[x];
                     break;
+                }
                 case 3:
-                    ((int32_t *)out_ptrs[c])[o] = col_load<int32_t>(dc, row);
+                case 5: {
+                    int32_t v[R];
+#pragma unroll
+                    for (int x = 0; x < R; x++)
+                        v[x] = col_load<int32_t>(cols[run[x] * n_cols + c],
+                                                 row[x]);
+#pragma unroll
+                    for (int x = 0; x < R; x++)
+                        if (x < nr) ((int32_t *)out_ptrs[c])[o + x] = v[x];
                     break;
+                }
                 case 4:
-                    ((int64_t *)out_ptrs[c])[o] = col_load<int64_t>(dc, row);
+                case 6: {
+                    int64_t v[R];
+#pragma unroll
+                    for (int x = 0; x < R; x++)
+                        v[x] = col_load<int64_t>(cols[run[x] * n_cols + c],
+                                                 row[x]);
+#pragma unroll
+                    for (int x = 0; x < R; x++)
+                        if (x < nr) ((int64_t *)out_ptrs[c])[o + x] = v[x];
                     break;
-                case 5:
-                    ((float *)out_ptrs[c])[o] = col_load<float>(dc, row);
-                    break;
-                case 6:
-                    ((double *)out_ptrs[c])[o] = col_load<double>(dc, row);
-                    break;
+                }
                 default: break;
                 }
             }
