@@ -67,8 +67,8 @@ hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
                            int n_cols, int k, const uint32_t *winners,
                            const int32_t *tile_counts,
                            const int64_t *tile_offsets, int64_t n_tiles,
-                           int64_t tile_rows, void *const *out_ptrs,
-                           hipStream_t stream);
+                           int64_t tile_rows, const int64_t *total_out,
+                           void *const *out_ptrs, hipStream_t stream);
 
 hipError_t pmh_launch_rle_decode(const RleChunk *chunks, int64_t n_chunks,
                                  int32_t *out, hipStream_t stream);

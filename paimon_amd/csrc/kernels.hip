@@ -241,6 +241,10 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    uint32_t *winners, int32_t *tile_counts) {
     const bool drop_delete = flags & 1;
     const bool ignore_delete = flags & 2;
+    // ablation levels (profiling only, flags bits 8..): 1=stage,2=+merge,
+    // 3=+scan, 0/absent=full. Partial levels publish a checksum so the
+    // compiler cannot dead-code the ablated phases' inputs.
+    const int ablate = (flags >> 8) & 0xf;
     __shared__ TileSmem sm;
     for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
         const int tid = threadIdx.x;
@@ -295,6 +299,13 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             }
         }
         __syncthreads();
+        if (ablate == 1) {
+            if (tid == 0)
+                tile_counts[tile] =
+                    (int32_t)(sm.skey[M - 1] ^ sm.sseq[M - 1] ^ sm.skind[0]);
+            __syncthreads();
+            continue;
+        }
 
         // --- pairwise stable merge, ceil(log2(k)) levels
         int cur = 0;
@@ -346,6 +357,11 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             __syncthreads();
         }
         const uint16_t *mo = sm.perm[cur];
+        if (ablate == 2) {
+            if (tid == 0) tile_counts[tile] = (int32_t)mo[M - 1] ^ (int32_t)mo[0];
+            __syncthreads();
+            continue;
+        }
 
         // --- group heads + previous-tile continuation skip
         for (int32_t i = tid; i < M; i += blockDim.x) {
@@ -394,6 +410,12 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             __syncthreads();
         }
 
+        if (ablate == 3) {
+            if (tid == 0)
+                tile_counts[tile] = (int32_t)sm.win[M - 1] ^ (int32_t)sm.fdone[0];
+            __syncthreads();
+            continue;
+        }
         // --- emit winners of owned groups, in key order.
         // A group is owned iff its head is a real (rank < Mreal) head; the
         // winner is the segmented-argmax value at the group's tail. Two
@@ -485,66 +507,77 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                        const uint8_t *col_dtype, int n_cols, int k,
                        const uint32_t *winners, const int32_t *tile_counts,
                        const int64_t *tile_offsets, int64_t n_tiles,
-                       int64_t tile_rows, void *const *out_ptrs) {
-    // 4 output rows per thread iteration: the 4 gathers per column are
-    // independent, quadrupling loads in flight (emit was 96% latency-parked,
-    // profiles/r01_c2_pmc.md).
+                       int64_t tile_rows, const int64_t *total_out,
+                       void *const *out_ptrs) {
+    // Flat grid-stride over the DENSE output index space (tile_offsets is an
+    // exclusive scan of tile_counts, so tile t owns output ranks
+    // [off[t], off[t]+cnt[t]) contiguously). R rows per thread iteration:
+    // independent gathers per column keep R x n_cols loads in flight (emit
+    // was 96% latency-parked, profiles/r01_c2_pmc.md) without idling tile
+    // remainders.
     constexpr int R = 4;
-    for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
-        int32_t cnt = tile_counts[tile];
-        int64_t base = tile_offsets[tile];
-        const uint32_t *w = &winners[tile * tile_rows];
-        for (int32_t j0 = threadIdx.x * R; j0 < cnt;
-             j0 += (int32_t)blockDim.x * R) {
-            int run[R];
-            int64_t row[R];
-            int nr = cnt - j0 < R ? cnt - j0 : R;
+    const int64_t total = *total_out;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x * R;
+    for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * R;
+         i0 < total; i0 += stride) {
+        // locate owning tile of i0 (binary search, L2-resident table)
+        int64_t lo = 0, hi = n_tiles - 1;
+        while (lo < hi) {
+            int64_t mid = (lo + hi + 1) >> 1;
+            if (tile_offsets[mid] <= i0) lo = mid;
+            else hi = mid - 1;
+        }
+        int64_t tile[R];
+        int run[R];
+        int64_t row[R];
+        int nr = total - i0 < R ? (int)(total - i0) : R;
+        int64_t t = lo;
 #pragma unroll
-            for (int x = 0; x < R; x++) {
-                uint32_t packed = x < nr ? w[j0 + x] : w[j0];
-                run[x] = packed >> 28;
-                row[x] = packed & 0x0fffffff;
+        for (int x = 0; x < R; x++) {
+            int64_t i = i0 + (x < nr ? x : 0);
+            // dense ranks: advance tile while i falls past its count
+            while (t + 1 < n_tiles && tile_offsets[t + 1] <= i) t++;
+            tile[x] = t;
+            uint32_t packed = winners[t * tile_rows + (i - tile_offsets[t])];
+            run[x] = packed >> 28;
+            row[x] = packed & 0x0fffffff;
+        }
+        for (int c = 0; c < n_cols; c++) {
+            switch (col_dtype[c]) {
+            case 1: {  // INT8 output from INT32-stored parquet TINYINT
+                int32_t v[R];
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    v[x] = col_load<int32_t>(cols[run[x] * n_cols + c], row[x]);
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    if (x < nr)
+                        ((int8_t *)out_ptrs[c])[i0 + x] = (int8_t)v[x];
+                break;
             }
-            int64_t o = base + j0;
-            for (int c = 0; c < n_cols; c++) {
-                switch (col_dtype[c]) {
-                case 1: {  // INT8 output from INT32-stored parquet TINYINT
-                    int32_t v[R];
+            case 3:
+            case 5: {
+                int32_t v[R];
 #pragma unroll
-                    for (int x = 0; x < R; x++)
-                        v[x] = col_load<int32_t>(cols[run[x] * n_cols + c],
-                                                 row[x]);
+                for (int x = 0; x < R; x++)
+                    v[x] = col_load<int32_t>(cols[run[x] * n_cols + c], row[x]);
 #pragma unroll
-                    for (int x = 0; x < R; x++)
-                        if (x < nr) ((int8_t *)out_ptrs[c])[o + x] = (int8_t)v[x];
-                    break;
-                }
-                case 3:
-                case 5: {
-                    int32_t v[R];
+                for (int x = 0; x < R; x++)
+                    if (x < nr) ((int32_t *)out_ptrs[c])[i0 + x] = v[x];
+                break;
+            }
+            case 4:
+            case 6: {
+                int64_t v[R];
 #pragma unroll
-                    for (int x = 0; x < R; x++)
-                        v[x] = col_load<int32_t>(cols[run[x] * n_cols + c],
-                                                 row[x]);
+                for (int x = 0; x < R; x++)
+                    v[x] = col_load<int64_t>(cols[run[x] * n_cols + c], row[x]);
 #pragma unroll
-                    for (int x = 0; x < R; x++)
-                        if (x < nr) ((int32_t *)out_ptrs[c])[o + x] = v[x];
-                    break;
-                }
-                case 4:
-                case 6: {
-                    int64_t v[R];
-#pragma unroll
-                    for (int x = 0; x < R; x++)
-                        v[x] = col_load<int64_t>(cols[run[x] * n_cols + c],
-                                                 row[x]);
-#pragma unroll
-                    for (int x = 0; x < R; x++)
-                        if (x < nr) ((int64_t *)out_ptrs[c])[o + x] = v[x];
-                    break;
-                }
-                default: break;
-                }
+                for (int x = 0; x < R; x++)
+                    if (x < nr) ((int64_t *)out_ptrs[c])[i0 + x] = v[x];
+                break;
+            }
+            default: break;
             }
         }
     }
@@ -632,12 +665,11 @@ hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
                            int n_cols, int k, const uint32_t *winners,
                            const int32_t *tile_counts,
                            const int64_t *tile_offsets, int64_t n_tiles,
-                           int64_t tile_rows, void *const *out_ptrs,
-                           hipStream_t stream) {
-    int blocks = n_tiles < 4096 ? (int)n_tiles : 4096;
-    hipLaunchKernelGGL(k_emit, dim3(blocks), dim3(256), 0, stream, cols,
+                           int64_t tile_rows, const int64_t *total_out,
+                           void *const *out_ptrs, hipStream_t stream) {
+    hipLaunchKernelGGL(k_emit, dim3(2048), dim3(256), 0, stream, cols,
                        col_dtype, n_cols, k, winners, tile_counts,
-                       tile_offsets, n_tiles, tile_rows, out_ptrs);
+                       tile_offsets, n_tiles, tile_rows, total_out, out_ptrs);
     return hipGetLastError();
 }
 

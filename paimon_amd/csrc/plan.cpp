@@ -892,6 +892,8 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     if (e != hipSuccess) return fail("partition", e);
     hipEventRecord(ev[2], st);
     int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0);
+    if (const char *ab = getenv("PMH_ABLATE"))  // profiling-only phase knob
+        flags |= (atoi(ab) & 0xf) << 8;
     e = pmh_launch_merge_tiles(sec.key_cols, sec.seq_cols, sec.kind_cols,
                                sec.lens_dev, k, sec.cuts, sec.n_tiles,
                                PMH_TILE_ROWS, flags, sec.winners,
@@ -904,7 +906,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     hipEventRecord(ev[4], st);
     e = pmh_launch_emit(sec.all_cols, p->col_dtype_dev, n_cols, k, sec.winners,
                         sec.tile_counts, sec.tile_offsets, sec.n_tiles,
-                        PMH_TILE_ROWS, p->out_ptrs_dev, st);
+                        PMH_TILE_ROWS, sec.total_dev, p->out_ptrs_dev, st);
     if (e != hipSuccess) return fail("emit", e);
     hipEventRecord(ev[5], st);
 
