@@ -300,9 +300,12 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         }
         __syncthreads();
         if (ablate == 1) {
-            if (tid == 0)
-                tile_counts[tile] =
-                    (int32_t)(sm.skey[M - 1] ^ sm.sseq[M - 1] ^ sm.skind[0]);
+            // keep staged data live; counts stay in {0,1} so the downstream
+            // scan/emit of an ablated (profiling-only) run never goes OOB
+            if (tid == 0) {
+                int64_t x = sm.skey[M - 1] ^ sm.sseq[M - 1] ^ sm.skind[0];
+                tile_counts[tile] = (int32_t)((x ^ (x >> 32)) & 1);
+            }
             __syncthreads();
             continue;
         }
@@ -358,7 +361,8 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         }
         const uint16_t *mo = sm.perm[cur];
         if (ablate == 2) {
-            if (tid == 0) tile_counts[tile] = (int32_t)mo[M - 1] ^ (int32_t)mo[0];
+            if (tid == 0)
+                tile_counts[tile] = ((int32_t)mo[M - 1] ^ (int32_t)mo[0]) & 1;
             __syncthreads();
             continue;
         }
@@ -412,7 +416,8 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
 
         if (ablate == 3) {
             if (tid == 0)
-                tile_counts[tile] = (int32_t)sm.win[M - 1] ^ (int32_t)sm.fdone[0];
+                tile_counts[tile] =
+                    ((int32_t)sm.win[M - 1] ^ (int32_t)sm.fdone[0]) & 1;
             __syncthreads();
             continue;
         }
