@@ -30,6 +30,7 @@ struct DevPage {
 
 struct DevCol {
     uint64_t addr0;        // n_pages == 1 fast path: element 0 address
+    uint64_t valid0;       // byte-per-row validity array; 0 = all valid
     const DevPage *pages;  // device array, sorted by start_row
     int32_t n_pages;
     int32_t esize;  // element size in bytes (stored width)
@@ -43,6 +44,8 @@ struct RleChunk {
     int32_t kind;       // 0 = RLE run, 1 = bit-packed groups
     uint32_t value;     // RLE literal (kind 0)
     int32_t bit_width;  // per-page bit width (dict-id streams vary per page)
+    int64_t aux;        // def-level streams: dense (non-null) values before
+                        // this chunk within its (run, column)
 };
 
 extern "C" {
@@ -72,6 +75,27 @@ hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
 
 hipError_t pmh_launch_rle_decode(const RleChunk *chunks, int64_t n_chunks,
                                  int32_t *out, hipStream_t stream);
+
+// Decode def-level streams (bit width 1) and position the dense PLAIN
+// values: valid[row] = level; out[row] = dense[aux + prefix] for valid rows
+// (VectorizedColumnReader null handling, VectorizedColumnReader.java:143-241).
+hipError_t pmh_launch_level_scatter(const RleChunk *chunks, int64_t n_chunks,
+                                    const void *dense, void *out,
+                                    uint8_t *valid, int esize,
+                                    hipStream_t stream);
+
+// PartialUpdate emit: per owned group, overlay non-null fields in ascending
+// (seq, isAdd) order (PartialUpdateMergeFunction.java:188-215 + Reducer
+// wrapper bypass). members/group_start written by k_merge_tiles in PU mode.
+hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
+                              const uint8_t *col_nullable, int n_cols, int k,
+                              int seq_col, int kind_col,
+                              const uint32_t *members,
+                              const uint16_t *group_start,
+                              const int64_t *tile_offsets, int64_t n_tiles,
+                              int64_t tile_rows, const int64_t *total_out,
+                              void *const *out_ptrs,
+                              uint8_t *const *out_valid, hipStream_t stream);
 
 hipError_t pmh_launch_dict_gather(const int32_t *ids, const void *dict,
                                   int64_t n, void *out, int esize,

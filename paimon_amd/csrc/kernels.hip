@@ -286,15 +286,23 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             continue;
         }
 
-        // --- stage key/seq/kind segments into LDS (coalesced per run)
+        // --- stage key/seq/kind segments into LDS (coalesced per run).
+        // Column base addresses hoisted out of the element loop: the staged
+        // columns are contiguous, and re-reading the DevCol descriptor per
+        // element serialized the loads (ablation: staging was 1.5 of 3.6 ms).
         for (int r = 0; r < k; r++) {
             int32_t off = sm.segoff[r], len = sm.seglen[r];
             int64_t base = c0[r];
+            const int64_t *kaddr =
+                reinterpret_cast<const int64_t *>(keys[r].addr0) + base;
+            const int64_t *saddr =
+                reinterpret_cast<const int64_t *>(seqs[r].addr0) + base;
+            const int32_t *daddr =
+                reinterpret_cast<const int32_t *>(kinds[r].addr0) + base;
             for (int32_t i = tid; i < len; i += blockDim.x) {
-                sm.skey[off + i] = col_load<int64_t>(keys[r], base + i);
-                sm.sseq[off + i] = col_load<int64_t>(seqs[r], base + i);
-                sm.skind[off + i] =
-                    (uint8_t)col_load<int32_t>(kinds[r], base + i);
+                sm.skey[off + i] = kaddr[i];
+                sm.sseq[off + i] = saddr[i];
+                sm.skind[off + i] = (uint8_t)daddr[i];
                 sm.perm[0][off + i] = (uint16_t)(off + i);
             }
         }
@@ -520,11 +528,18 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
     // independent gathers per column keep R x n_cols loads in flight (emit
     // was 96% latency-parked, profiles/r01_c2_pmc.md) without idling tile
     // remainders.
+    // Each block owns a CONTIGUOUS output slice so its gathers stay within a
+    // few tiles' source segments (~130 KB working set -> XCD-L2 resident;
+    // grid-striding spread every block over the whole output and thrashed L2).
     constexpr int R = 4;
     const int64_t total = *total_out;
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x * R;
-    for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * R;
-         i0 < total; i0 += stride) {
+    const int64_t per_block =
+        (total + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
+    const int64_t slice_lo = (int64_t)blockIdx.x * per_block;
+    const int64_t slice_hi = slice_lo + per_block < total
+                                 ? slice_lo + per_block : total;
+    for (int64_t i0 = slice_lo + (int64_t)threadIdx.x * R; i0 < slice_hi;
+         i0 += (int64_t)blockDim.x * R) {
         // locate owning tile of i0 (binary search, L2-resident table)
         int64_t lo = 0, hi = n_tiles - 1;
         while (lo < hi) {
