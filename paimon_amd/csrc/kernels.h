@@ -60,6 +60,7 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
                                   int k, const int32_t *cuts, int64_t n_tiles,
                                   int64_t tile_rows, int flags,
                                   uint32_t *winners, int32_t *tile_counts,
+                                  uint16_t *group_start, uint32_t *err_flag,
                                   hipStream_t stream);
 
 hipError_t pmh_launch_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
@@ -67,11 +68,13 @@ hipError_t pmh_launch_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
                                  hipStream_t stream);
 
 hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
-                           int n_cols, int k, const uint32_t *winners,
+                           const uint8_t *col_nullable, int n_cols, int k,
+                           const uint32_t *winners,
                            const int32_t *tile_counts,
                            const int64_t *tile_offsets, int64_t n_tiles,
                            int64_t tile_rows, const int64_t *total_out,
-                           void *const *out_ptrs, hipStream_t stream);
+                           void *const *out_ptrs, uint8_t *const *out_valid,
+                           hipStream_t stream);
 
 hipError_t pmh_launch_rle_decode(const RleChunk *chunks, int64_t n_chunks,
                                  int32_t *out, hipStream_t stream);

@@ -238,9 +238,16 @@ __launch_bounds__(PMH_TILE_THREADS) __global__
 void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    const int64_t *lens, int k, const int32_t *cuts,
                    int64_t n_tiles, int64_t tile_rows, int flags,
-                   uint32_t *winners, int32_t *tile_counts) {
+                   uint32_t *winners, int32_t *tile_counts,
+                   uint16_t *group_start, uint32_t *err_flag) {
     const bool drop_delete = flags & 1;
     const bool ignore_delete = flags & 2;
+    // PartialUpdate mode: no winner reduction; emit every owned group's
+    // member list (ascending (seq, isAdd) order within the group) for the
+    // per-field overlay in k_emit_pu. v1 accepts INSERT-only streams (the
+    // reference's default partial-update rejects retracts,
+    // PartialUpdateMergeFunction.java:170-186) — retracts set err_flag.
+    const bool pu_mode = flags & 4;
     // ablation levels (profiling only, flags bits 8..): 1=stage,2=+merge,
     // 3=+scan, 0/absent=full. Partial levels publish a checksum so the
     // compiler cannot dead-code the ablated phases' inputs.
@@ -388,6 +395,75 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         }
         __syncthreads();
 
+        if (pu_mode) {
+            // --- PartialUpdate: emit owned groups' member lists
+            __shared__ int32_t s_scan[PMH_TILE_THREADS];
+            __shared__ int32_t s_scan2[PMH_TILE_THREADS];
+            const int32_t per =
+                (Mreal + (int32_t)blockDim.x - 1) / blockDim.x;
+            int32_t my_lo = tid * per;
+            int32_t my_hi = my_lo + per < Mreal ? my_lo + per : Mreal;
+            uint32_t *mout = &winners[tile * (tile_rows + PMH_MAX_RUNS)];
+            uint16_t *gout = &group_start[tile * (tile_rows + 1)];
+            int32_t total_g = 0, total_m = 0;
+            bool bad_kind = false;
+            for (int pass = 0; pass < 2; pass++) {
+                int32_t ng = 0, nm = 0;
+                int32_t g_off = pass == 1 ? s_scan[tid] : 0;
+                int32_t m_off = pass == 1 ? s_scan2[tid] : 0;
+                for (int32_t i = my_lo; i < my_hi; i++) {
+                    if (!sm.head[i]) continue;
+                    int32_t tail = i;
+                    while (tail + 1 < M && !sm.head[tail + 1]) tail++;
+                    if (pass == 1) {
+                        gout[g_off + ng] = (uint16_t)(m_off + nm);
+                        for (int32_t x = i; x <= tail; x++) {
+                            uint16_t s = mo[x];
+                            if (!kind_is_add(sm.skind[s])) bad_kind = true;
+                            int r = 0;
+                            while (r + 1 <= k - 1 &&
+                                   sm.segoff[r + 1] <= (int32_t)s)
+                                r++;
+                            uint32_t grow = (uint32_t)(
+                                c0[r] + ((int32_t)s - sm.segoff[r]));
+                            mout[m_off + nm + (x - i)] =
+                                ((uint32_t)r << 28) | grow;
+                        }
+                    }
+                    ng++;
+                    nm += tail - i + 1;
+                }
+                if (pass == 0) {
+                    s_scan[tid] = ng;
+                    s_scan2[tid] = nm;
+                    __syncthreads();
+                    for (int d = 1; d < (int)blockDim.x; d <<= 1) {
+                        int32_t v = s_scan[tid], v2 = s_scan2[tid];
+                        int32_t a = tid >= d ? s_scan[tid - d] : 0;
+                        int32_t a2 = tid >= d ? s_scan2[tid - d] : 0;
+                        __syncthreads();
+                        s_scan[tid] = v + a;
+                        s_scan2[tid] = v2 + a2;
+                        __syncthreads();
+                    }
+                    total_g = s_scan[blockDim.x - 1];
+                    total_m = s_scan2[blockDim.x - 1];
+                    int32_t incl = s_scan[tid], incl2 = s_scan2[tid];
+                    __syncthreads();
+                    s_scan[tid] = incl - ng;
+                    s_scan2[tid] = incl2 - nm;
+                    __syncthreads();
+                }
+            }
+            if (bad_kind && err_flag) atomicOr(err_flag, 1u);
+            if (tid == 0) {
+                tile_counts[tile] = total_g;
+                gout[total_g] = (uint16_t)total_m;  // sentinel
+            }
+            __syncthreads();
+            continue;
+        }
+
         // --- segmented argmax by (eligible, seq, isAdd): groups <= k.
         // Double-buffered through registers (static indices: rule-of-thumb
         // from the CDNA guide — runtime-indexed locals spill to scratch).
@@ -438,7 +514,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         const int32_t per = (Mreal + (int32_t)blockDim.x - 1) / blockDim.x;
         int32_t my_lo = tid * per;
         int32_t my_hi = my_lo + per < Mreal ? my_lo + per : Mreal;
-        uint32_t *wout = &winners[tile * tile_rows];
+        uint32_t *wout = &winners[tile * (tile_rows + PMH_MAX_RUNS)];
         int32_t total = 0;
         for (int pass = 0; pass < 2; pass++) {
             int32_t nloc = 0;
@@ -517,11 +593,12 @@ __global__ void k_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
 // One workgroup per tile (grid-stride); thread per output row; inner loop
 // over columns. Output writes are coalesced per column.
 __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
-                       const uint8_t *col_dtype, int n_cols, int k,
-                       const uint32_t *winners, const int32_t *tile_counts,
+                       const uint8_t *col_dtype, const uint8_t *col_nullable,
+                       int n_cols, int k, const uint32_t *winners,
+                       const int32_t *tile_counts,
                        const int64_t *tile_offsets, int64_t n_tiles,
                        int64_t tile_rows, const int64_t *total_out,
-                       void *const *out_ptrs) {
+                       void *const *out_ptrs, uint8_t *const *out_valid) {
     // Flat grid-stride over the DENSE output index space (tile_offsets is an
     // exclusive scan of tile_counts, so tile t owns output ranks
     // [off[t], off[t]+cnt[t]) contiguously). R rows per thread iteration:
@@ -558,11 +635,22 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
             // dense ranks: advance tile while i falls past its count
             while (t + 1 < n_tiles && tile_offsets[t + 1] <= i) t++;
             tile[x] = t;
-            uint32_t packed = winners[t * tile_rows + (i - tile_offsets[t])];
+            uint32_t packed =
+                winners[t * (tile_rows + PMH_MAX_RUNS) + (i - tile_offsets[t])];
             run[x] = packed >> 28;
             row[x] = packed & 0x0fffffff;
         }
         for (int c = 0; c < n_cols; c++) {
+            if (col_nullable[c] && out_valid[c]) {
+                // winner's validity byte (dedup keeps the record as-is)
+#pragma unroll
+                for (int x = 0; x < R; x++) {
+                    if (x >= nr) continue;
+                    const DevCol &dc = cols[run[x] * n_cols + c];
+                    out_valid[c][i0 + x] =
+                        dc.valid0 ? ((const uint8_t *)dc.valid0)[row[x]] : 1;
+                }
+            }
             switch (col_dtype[c]) {
             case 1: {  // INT8 output from INT32-stored parquet TINYINT
                 int32_t v[R];
@@ -599,6 +687,170 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
             }
             default: break;
             }
+        }
+    }
+}
+
+// ------------------------------------------------------------ k_emit_pu
+//
+// PartialUpdate emit: one output row per owned group; for each value column
+// take the LAST member whose field is non-null (updateNonNullFields,
+// PartialUpdateMergeFunction.java:188-215; members arrive in ascending
+// (seq, isAdd) order). seq = last member's sequenceNumber
+// (latestSequenceNumber, :188); kind = INSERT (insert-only streams, v1).
+// Singleton groups return the record as-is (ReducerMergeFunctionWrapper
+// bypass) — identical to the overlay for INSERT records.
+__global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
+                          const uint8_t *col_nullable, int n_cols, int k,
+                          int seq_col, int kind_col, const uint32_t *members,
+                          const uint16_t *group_start,
+                          const int64_t *tile_offsets, int64_t n_tiles,
+                          int64_t tile_rows, const int64_t *total_out,
+                          void *const *out_ptrs, uint8_t *const *out_valid) {
+    const int64_t total = *total_out;
+    const int64_t per_block =
+        (total + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
+    const int64_t slice_lo = (int64_t)blockIdx.x * per_block;
+    const int64_t slice_hi =
+        slice_lo + per_block < total ? slice_lo + per_block : total;
+    for (int64_t i = slice_lo + threadIdx.x; i < slice_hi; i += blockDim.x) {
+        int64_t lo = 0, hi = n_tiles - 1;
+        while (lo < hi) {
+            int64_t mid = (lo + hi + 1) >> 1;
+            if (tile_offsets[mid] <= i) lo = mid;
+            else hi = mid - 1;
+        }
+        int64_t t = lo;
+        int64_t g = i - tile_offsets[t];
+        const uint16_t *gs = &group_start[t * (tile_rows + 1)];
+        const uint32_t *mem = &members[t * (tile_rows + PMH_MAX_RUNS)];
+        int32_t ms = gs[g], me = gs[g + 1];
+        uint32_t last = mem[me - 1];
+        int lrun = last >> 28;
+        int64_t lrow = last & 0x0fffffff;
+        for (int c = 0; c < n_cols; c++) {
+            if (c == kind_col) {
+                ((int8_t *)out_ptrs[c])[i] = 0;  // RowKind.INSERT
+                continue;
+            }
+            int64_t run = lrun, row = lrow;
+            uint8_t ok = 1;
+            if (col_nullable[c]) {
+                ok = 0;
+                for (int32_t x = me - 1; x >= ms; x--) {
+                    uint32_t m = mem[x];
+                    const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                    uint8_t v = dc.valid0
+                                    ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                                    : 1;
+                    if (v) {
+                        run = m >> 28;
+                        row = m & 0x0fffffff;
+                        ok = 1;
+                        break;
+                    }
+                }
+            }
+            const DevCol &dc = cols[run * n_cols + c];
+            switch (col_dtype[c]) {
+            case 1:
+                ((int8_t *)out_ptrs[c])[i] =
+                    ok ? (int8_t)col_load<int32_t>(dc, row) : 0;
+                break;
+            case 3:
+            case 5:
+                ((int32_t *)out_ptrs[c])[i] =
+                    ok ? col_load<int32_t>(dc, row) : 0;
+                break;
+            case 4:
+            case 6:
+                ((int64_t *)out_ptrs[c])[i] =
+                    ok ? col_load<int64_t>(dc, row) : 0;
+                break;
+            default: break;
+            }
+            if (out_valid[c]) out_valid[c][i] = ok;
+        }
+    }
+}
+
+// --------------------------------------------------------- k_level_scatter
+//
+// Decode def-level streams (RLE/bit-packed hybrid, bit width 1) and position
+// the dense PLAIN values: valid[row] = level, out[row] = dense[aux + prefix]
+// (the null handling of VectorizedColumnReader.readBatch,
+// paimon-format/.../reader/VectorizedColumnReader.java:143-241). One
+// workgroup per chunk; packed chunks compute the dense prefix with an
+// in-LDS running scan over 256-element blocks.
+__global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks,
+                                const uint8_t *dense, uint8_t *out,
+                                uint8_t *valid, int esize) {
+    __shared__ int32_t s_scan[PMH_TILE_THREADS];
+    __shared__ int32_t s_running;
+    for (int64_t cidx = blockIdx.x; cidx < n_chunks; cidx += gridDim.x) {
+        RleChunk ch = chunks[cidx];
+        if (ch.kind == 0) {
+            if (ch.value) {  // run of non-nulls: dense block copy
+                for (int32_t i = threadIdx.x; i < ch.count; i += blockDim.x) {
+                    valid[ch.out_start + i] = 1;
+                    const uint8_t *s = dense + (ch.aux + i) * esize;
+                    uint8_t *d = out + (ch.out_start + i) * esize;
+                    if (esize == 4)
+                        *(int32_t *)d = *(const int32_t *)s;
+                    else
+                        *(int64_t *)d = *(const int64_t *)s;
+                }
+            } else {  // run of nulls
+                for (int32_t i = threadIdx.x; i < ch.count; i += blockDim.x) {
+                    valid[ch.out_start + i] = 0;
+                    uint8_t *d = out + (ch.out_start + i) * esize;
+                    if (esize == 4)
+                        *(int32_t *)d = 0;
+                    else
+                        *(int64_t *)d = 0;
+                }
+            }
+            __syncthreads();
+            continue;
+        }
+        // bit-packed: 1 byte per 8 levels
+        const uint8_t *src = (const uint8_t *)ch.src;
+        if (threadIdx.x == 0) s_running = 0;
+        __syncthreads();
+        for (int32_t b = 0; b < ch.count; b += (int32_t)blockDim.x) {
+            int32_t i = b + threadIdx.x;
+            int bit = 0;
+            if (i < ch.count) bit = (src[i >> 3] >> (i & 7)) & 1;
+            s_scan[threadIdx.x] = bit;
+            __syncthreads();
+            for (int d = 1; d < (int)blockDim.x; d <<= 1) {
+                int32_t v = s_scan[threadIdx.x];
+                int32_t a = threadIdx.x >= d ? s_scan[threadIdx.x - d] : 0;
+                __syncthreads();
+                s_scan[threadIdx.x] = v + a;
+                __syncthreads();
+            }
+            if (i < ch.count) {
+                valid[ch.out_start + i] = (uint8_t)bit;
+                uint8_t *dp = out + (ch.out_start + i) * esize;
+                if (bit) {
+                    int64_t didx =
+                        ch.aux + s_running + s_scan[threadIdx.x] - 1;
+                    const uint8_t *sp = dense + didx * esize;
+                    if (esize == 4)
+                        *(int32_t *)dp = *(const int32_t *)sp;
+                    else
+                        *(int64_t *)dp = *(const int64_t *)sp;
+                } else {
+                    if (esize == 4)
+                        *(int32_t *)dp = 0;
+                    else
+                        *(int64_t *)dp = 0;
+                }
+            }
+            __syncthreads();
+            if (threadIdx.x == 0) s_running += s_scan[blockDim.x - 1];
+            __syncthreads();
         }
     }
 }
@@ -665,11 +917,40 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
                                   int k, const int32_t *cuts, int64_t n_tiles,
                                   int64_t tile_rows, int flags,
                                   uint32_t *winners, int32_t *tile_counts,
+                                  uint16_t *group_start, uint32_t *err_flag,
                                   hipStream_t stream) {
     int blocks = n_tiles < 4096 ? (int)n_tiles : 4096;
     hipLaunchKernelGGL(k_merge_tiles, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
                        stream, keys, seqs, kinds, lens, k, cuts, n_tiles,
-                       tile_rows, flags, winners, tile_counts);
+                       tile_rows, flags, winners, tile_counts, group_start,
+                       err_flag);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
+                              const uint8_t *col_nullable, int n_cols, int k,
+                              int seq_col, int kind_col,
+                              const uint32_t *members,
+                              const uint16_t *group_start,
+                              const int64_t *tile_offsets, int64_t n_tiles,
+                              int64_t tile_rows, const int64_t *total_out,
+                              void *const *out_ptrs,
+                              uint8_t *const *out_valid, hipStream_t stream) {
+    hipLaunchKernelGGL(k_emit_pu, dim3(2048), dim3(256), 0, stream, cols,
+                       col_dtype, col_nullable, n_cols, k, seq_col, kind_col,
+                       members, group_start, tile_offsets, n_tiles, tile_rows,
+                       total_out, out_ptrs, out_valid);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_level_scatter(const RleChunk *chunks, int64_t n_chunks,
+                                    const void *dense, void *out,
+                                    uint8_t *valid, int esize,
+                                    hipStream_t stream) {
+    int blocks = n_chunks < 4096 ? (int)(n_chunks ? n_chunks : 1) : 4096;
+    hipLaunchKernelGGL(k_level_scatter, dim3(blocks), dim3(PMH_TILE_THREADS),
+                       0, stream, chunks, n_chunks, (const uint8_t *)dense,
+                       (uint8_t *)out, valid, esize);
     return hipGetLastError();
 }
 
@@ -682,14 +963,17 @@ hipError_t pmh_launch_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
 }
 
 hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
-                           int n_cols, int k, const uint32_t *winners,
+                           const uint8_t *col_nullable, int n_cols, int k,
+                           const uint32_t *winners,
                            const int32_t *tile_counts,
                            const int64_t *tile_offsets, int64_t n_tiles,
                            int64_t tile_rows, const int64_t *total_out,
-                           void *const *out_ptrs, hipStream_t stream) {
+                           void *const *out_ptrs, uint8_t *const *out_valid,
+                           hipStream_t stream) {
     hipLaunchKernelGGL(k_emit, dim3(2048), dim3(256), 0, stream, cols,
-                       col_dtype, n_cols, k, winners, tile_counts,
-                       tile_offsets, n_tiles, tile_rows, total_out, out_ptrs);
+                       col_dtype, col_nullable, n_cols, k, winners,
+                       tile_counts, tile_offsets, n_tiles, tile_rows,
+                       total_out, out_ptrs, out_valid);
     return hipGetLastError();
 }
 

@@ -187,6 +187,17 @@ struct RunCol {
     int32_t *ids_dev = nullptr;
     std::vector<GatherTask> gathers;
     int64_t n_rows = 0;
+    // nullable columns with actual nulls: dense PLAIN values + def-level
+    // streams stay encoded in HBM; k_level_scatter positions them at read
+    // time (timed decode) and fills the byte-validity array.
+    bool has_nulls = false;
+    std::vector<uint8_t> dense_host;   // packed non-null values (staging)
+    std::vector<uint8_t> levels_host;  // packed def-level streams (staging)
+    std::vector<RleChunk> def_host;    // src = RELATIVE offset until upload
+    int64_t dense_before = 0;          // running non-null count
+    void *dense_dev = nullptr;
+    RleChunk *def_dev = nullptr;
+    uint8_t *valid_dev = nullptr;
 };
 
 struct Run {
@@ -209,7 +220,10 @@ struct Section {
     int32_t *tile_counts = nullptr;
     int64_t *tile_offsets = nullptr;
     int64_t *total_dev = nullptr;
+    uint16_t *group_start = nullptr;  // partial-update member offsets
+    uint32_t *err_dev = nullptr;
     bool any_dict = false;
+    bool any_decode = false;  // dict or null-scatter work at read time
 };
 
 }  // namespace pmh
@@ -227,6 +241,7 @@ struct pmh_plan_t {
     bool drop_delete = true;
     bool ignore_delete = false;
     bool host_output = false;
+    bool pu = false;  // partial-update merge engine
     std::vector<pmh::Section> sections;
     size_t cur_section = 0;
     int64_t rows_in_total = 0;
@@ -235,6 +250,11 @@ struct pmh_plan_t {
     std::vector<void *> out_dev;
     void **out_ptrs_dev = nullptr;
     uint8_t *col_dtype_dev = nullptr;
+    std::vector<bool> col_nullable;        // any nulls staged for this col
+    uint8_t *col_nullable_dev = nullptr;
+    std::vector<uint8_t *> out_valid;      // per col; nullptr if never null
+    uint8_t **out_valid_dev = nullptr;     // device array of the above
+    std::vector<std::vector<uint8_t>> out_valid_host;
     std::vector<std::vector<uint8_t>> out_host;
     std::vector<pmh_col> batch_cols;
     std::vector<std::string> col_names;
@@ -277,6 +297,76 @@ static int dtype_stored_esize(int dt) {
     case PMH_DT_FLOAT64: return 8;
     }
     return 0;
+}
+
+static inline int popcount8(uint8_t b) { return __builtin_popcount(b); }
+
+// Walk a def-level stream (bit width 1): emit device work chunks with
+// running dense offsets (aux) and report the number of non-null values.
+// src offsets are RELATIVE to the (run,col) levels buffer (rel_base =
+// offset of this stream within it); patched to device addresses at upload.
+static bool prescan_def(const uint8_t *s, int64_t len, int64_t n,
+                        int64_t out_row0, int64_t rel_base,
+                        int64_t *dense_before, std::vector<RleChunk> &out) {
+    int64_t p = 0, cnt = 0;
+    const int64_t MAX_CHUNK = 16384;
+    while (cnt < n) {
+        if (p >= len) {
+            set_error("def-level stream overrun");
+            return false;
+        }
+        uint64_t header = 0;
+        int shift = 0;
+        for (;;) {
+            uint8_t b = s[p++];
+            header |= (uint64_t)(b & 0x7F) << shift;
+            if (!(b & 0x80)) break;
+            shift += 7;
+        }
+        if ((header & 1) == 0) {
+            int64_t count = std::min<int64_t>((int64_t)(header >> 1), n - cnt);
+            uint8_t v = s[p++] & 1;
+            for (int64_t off = 0; off < count; off += MAX_CHUNK) {
+                RleChunk c{};
+                c.kind = 0;
+                c.value = v;
+                c.out_start = out_row0 + cnt + off;
+                c.count = (int32_t)std::min<int64_t>(MAX_CHUNK, count - off);
+                c.bit_width = 1;
+                c.aux = *dense_before + (v ? off : 0);
+                out.push_back(c);
+            }
+            if (v) *dense_before += count;
+            cnt += count;
+        } else {
+            int64_t groups = (int64_t)(header >> 1);
+            int64_t vals = std::min<int64_t>(groups * 8, n - cnt);
+            int64_t voff = 0;
+            while (voff < vals) {
+                int64_t take = std::min<int64_t>(MAX_CHUNK, vals - voff);
+                RleChunk c{};
+                c.kind = 1;
+                c.src = (uint64_t)(rel_base + p + voff / 8);
+                c.out_start = out_row0 + cnt + voff;
+                c.count = (int32_t)take;
+                c.bit_width = 1;
+                c.aux = *dense_before;
+                out.push_back(c);
+                // advance dense count by the set bits in this chunk
+                for (int64_t i = 0; i < take; i += 8) {
+                    uint8_t b = s[p + (voff + i) / 8];
+                    int64_t rem = take - i;
+                    uint8_t mask =
+                        rem >= 8 ? 0xFF : (uint8_t)((1u << rem) - 1);
+                    *dense_before += popcount8(b & mask);
+                }
+                voff += take;
+            }
+            p += groups;
+            cnt += vals;
+        }
+    }
+    return true;
 }
 
 // Walk an RLE/bit-packed def-level stream (bit width 1) and report whether
@@ -497,8 +587,9 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                               fd.path.c_str(), cols[c].name.c_str());
                     return false;
                 }
-                // per-page value offsets past def levels (+ null check)
+                // per-page value offsets past def levels (+ null detection)
                 std::vector<int64_t> vpos(cc.pages.size(), 0);
+                bool chunk_nulls = false;
                 for (size_t pi = 0; pi < cc.pages.size(); pi++) {
                     auto &pg = cc.pages[pi];
                     if (pg.page_type != 0) continue;
@@ -507,21 +598,42 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     if (max_def > 0) {
                         uint32_t dl_len;
                         memcpy(&dl_len, pp, 4);
-                        if (def_levels_have_nulls(pp + 4, dl_len,
-                                                  pg.num_values)) {
-                            set_error(
-                                "%s col %s: null values present — the GPU "
-                                "dedup path requires non-null columns "
-                                "(partial-update/null support is a later "
-                                "round; see DESIGN.md)",
-                                fd.path.c_str(), cols[c].name.c_str());
-                            return false;
-                        }
+                        chunk_nulls |= def_levels_have_nulls(pp + 4, dl_len,
+                                                             pg.num_values);
                         pos = 4 + dl_len;
                     }
                     vpos[pi] = pos;
                 }
-                if (has_plain || (!has_dict && cc.num_values > 0)) {
+                if (chunk_nulls && has_dict) {
+                    set_error("%s col %s: nulls in dictionary-encoded chunks "
+                              "not supported yet",
+                              fd.path.c_str(), cols[c].name.c_str());
+                    return false;
+                }
+                if (chunk_nulls) {
+                    // dense PLAIN values + def-level streams stay encoded;
+                    // k_level_scatter positions them at read time
+                    rc.has_nulls = true;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                        auto &pg = cc.pages[pi];
+                        if (pg.page_type != 0) continue;
+                        const uint8_t *pp = payload_base + ppo[pi];
+                        uint32_t dl_len;
+                        memcpy(&dl_len, pp, 4);
+                        int64_t rel = (int64_t)rc.levels_host.size();
+                        rc.levels_host.insert(rc.levels_host.end(), pp + 4,
+                                              pp + 4 + dl_len);
+                        int64_t before = rc.dense_before;
+                        if (!prescan_def(pp + 4, dl_len, pg.num_values,
+                                         chunk_row0 + pg.first_row, rel,
+                                         &rc.dense_before, rc.def_host))
+                            return false;
+                        int64_t nvalid = rc.dense_before - before;
+                        rc.dense_host.insert(
+                            rc.dense_host.end(), pp + vpos[pi],
+                            pp + vpos[pi] + nvalid * stored);
+                    }
+                } else if (has_plain || (!has_dict && cc.num_values > 0)) {
                     // pack PLAIN value payloads and copy into contig
                     std::vector<uint8_t> pack(cc.num_values * stored);
                     int64_t off = 0;
@@ -620,6 +732,36 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                           hipMemcpyHostToDevice) != hipSuccess)
                 return false;
         }
+        if (rc.has_nulls) {
+            rc.valid_dev = (uint8_t *)plan->bufs.alloc(run.length);
+            rc.dense_dev = plan->bufs.alloc(rc.dense_host.size());
+            void *levels_dev = plan->bufs.alloc(rc.levels_host.size());
+            rc.def_dev = (RleChunk *)plan->bufs.alloc(rc.def_host.size() *
+                                                      sizeof(RleChunk));
+            if (!rc.valid_dev || !rc.dense_dev || !levels_dev || !rc.def_dev)
+                return false;
+            // rows in non-null chunks of this column keep validity 1
+            if (hipMemset(rc.valid_dev, 1, run.length) != hipSuccess)
+                return false;
+            if (hipMemcpy(rc.dense_dev, rc.dense_host.data(),
+                          rc.dense_host.size(),
+                          hipMemcpyHostToDevice) != hipSuccess ||
+                hipMemcpy(levels_dev, rc.levels_host.data(),
+                          rc.levels_host.size(),
+                          hipMemcpyHostToDevice) != hipSuccess)
+                return false;
+            for (auto &dc : rc.def_host)
+                if (dc.kind == 1) dc.src += (uint64_t)levels_dev;
+            if (hipMemcpy(rc.def_dev, rc.def_host.data(),
+                          rc.def_host.size() * sizeof(RleChunk),
+                          hipMemcpyHostToDevice) != hipSuccess)
+                return false;
+            plan->encoded_bytes_total += rc.levels_host.size();
+            rc.dense_host.clear();
+            rc.dense_host.shrink_to_fit();
+            rc.levels_host.clear();
+            rc.levels_host.shrink_to_fit();
+        }
         DevPage dp{(uint64_t)rc.contig, 0};
         rc.pages_host.assign(1, dp);
         rc.n_pages = 1;
@@ -644,8 +786,10 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
         lens[r] = run.length;
         sec.total_rows += run.length;
         for (int c = 0; c < n_cols; c++) {
-            DevCol dc{(uint64_t)run.cols[c].contig, run.cols[c].pages_dev,
-                      run.cols[c].n_pages, plan->cols[c].stored_esize};
+            DevCol dc{(uint64_t)run.cols[c].contig,
+                      (uint64_t)run.cols[c].valid_dev,
+                      run.cols[c].pages_dev, run.cols[c].n_pages,
+                      plan->cols[c].stored_esize};
             allv[r * n_cols + c] = dc;
             if (c == 0) keyv[r] = dc;  // v1: single int64 key column
             if (c == seq_idx) seqv[r] = dc;
@@ -665,16 +809,26 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
     sec.all_cols = (DevCol *)up(allv.data(), allv.size() * sizeof(DevCol));
     sec.lens_dev = (int64_t *)up(lens.data(), k * sizeof(int64_t));
     sec.cuts = (int32_t *)plan->bufs.alloc((sec.n_tiles + 1) * k * 4);
-    sec.winners =
-        (uint32_t *)plan->bufs.alloc(sec.n_tiles * PMH_TILE_ROWS * 4);
+    sec.winners = (uint32_t *)plan->bufs.alloc(
+        sec.n_tiles * (PMH_TILE_ROWS + PMH_MAX_RUNS) * 4);
     sec.tile_counts = (int32_t *)plan->bufs.alloc(sec.n_tiles * 4);
     sec.tile_offsets = (int64_t *)plan->bufs.alloc(sec.n_tiles * 8);
     sec.total_dev = (int64_t *)plan->bufs.alloc(8);
+    sec.err_dev = (uint32_t *)plan->bufs.alloc(4);
+    if (sec.err_dev) hipMemset(sec.err_dev, 0, 4);
+    if (plan->pu) {
+        sec.group_start = (uint16_t *)plan->bufs.alloc(
+            sec.n_tiles * (PMH_TILE_ROWS + 1) * 2);
+        if (!sec.group_start) return false;
+    }
     for (auto &run : sec.runs)
-        for (auto &rc : run.cols) sec.any_dict |= rc.dict_encoded;
+        for (auto &rc : run.cols) {
+            sec.any_dict |= rc.dict_encoded;
+            sec.any_decode |= rc.dict_encoded || rc.has_nulls;
+        }
     return sec.key_cols && sec.seq_cols && sec.kind_cols && sec.all_cols &&
            sec.lens_dev && sec.cuts && sec.winners && sec.tile_counts &&
-           sec.tile_offsets && sec.total_dev;
+           sec.tile_offsets && sec.total_dev && sec.err_dev;
 }
 
 }  // namespace pmh
@@ -749,9 +903,13 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         for (const auto &c : j["value_cols"].arr)
             if (!add_col(c)) return nullptr;
         std::string engine = j["merge_engine"].as_str("deduplicate");
-        if (engine != "deduplicate") {
-            set_error("merge engine '%s' not yet on the GPU path (v1: "
-                      "deduplicate; partial-update next)",
+        if (engine == "partial-update") {
+            plan->pu = true;  // INSERT-only v1 (PartialUpdateMergeFunction
+                              // rejects retracts by default, :170-186)
+        } else if (engine != "deduplicate") {
+            set_error("merge engine '%s' not on the GPU path (deduplicate | "
+                      "partial-update; aggregation/first-row are later "
+                      "rounds)",
                       engine.c_str());
             return nullptr;
         }
@@ -833,6 +991,32 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                   hipMemcpyHostToDevice);
         hipMemcpy(plan->col_dtype_dev, dts.data(), n_cols,
                   hipMemcpyHostToDevice);
+        // per-column nullability (any staged nulls in any section/run) +
+        // output validity buffers
+        plan->col_nullable.assign(n_cols, false);
+        for (auto &sec : plan->sections)
+            for (auto &run : sec.runs)
+                for (int c = 0; c < n_cols; c++)
+                    if (run.cols[c].has_nulls) plan->col_nullable[c] = true;
+        plan->out_valid.assign(n_cols, nullptr);
+        std::vector<uint8_t> nul(n_cols, 0);
+        for (int c = 0; c < n_cols; c++) {
+            nul[c] = plan->col_nullable[c] ? 1 : 0;
+            if (plan->col_nullable[c]) {
+                plan->out_valid[c] = (uint8_t *)plan->bufs.alloc(max_rows);
+                if (!plan->out_valid[c]) {
+                    set_error("validity allocation failed");
+                    return nullptr;
+                }
+            }
+        }
+        plan->col_nullable_dev = (uint8_t *)plan->bufs.alloc(n_cols);
+        plan->out_valid_dev =
+            (uint8_t **)plan->bufs.alloc(n_cols * sizeof(void *));
+        hipMemcpy(plan->col_nullable_dev, nul.data(), n_cols,
+                  hipMemcpyHostToDevice);
+        hipMemcpy(plan->out_valid_dev, plan->out_valid.data(),
+                  n_cols * sizeof(void *), hipMemcpyHostToDevice);
         for (auto &cs : plan->cols) plan->col_names.push_back(cs.name);
     } catch (const std::exception &e) {
         set_error("plan parse: %s", e.what());
@@ -866,21 +1050,29 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     };
 
     hipEventRecord(ev[0], st);
-    // decode (dictionary materialization)
-    if (sec.any_dict) {
+    // decode: dictionary materialization + null-column level scatter
+    if (sec.any_decode) {
         for (auto &run : sec.runs) {
             for (size_t c = 0; c < run.cols.size(); c++) {
                 RunCol &rc = run.cols[c];
-                if (!rc.dict_encoded) continue;
-                hipError_t e = pmh_launch_rle_decode(
-                    rc.rle_dev, (int64_t)rc.rle_host.size(), rc.ids_dev, st);
-                if (e != hipSuccess) return fail("rle_decode", e);
                 int es = p->cols[c].stored_esize;
-                for (const GatherTask &gt : rc.gathers) {
-                    e = pmh_launch_dict_gather(
-                        rc.ids_dev + gt.start, gt.dict_dev, gt.n,
-                        (uint8_t *)rc.contig + gt.start * es, es, st);
-                    if (e != hipSuccess) return fail("dict_gather", e);
+                if (rc.dict_encoded) {
+                    hipError_t e = pmh_launch_rle_decode(
+                        rc.rle_dev, (int64_t)rc.rle_host.size(), rc.ids_dev,
+                        st);
+                    if (e != hipSuccess) return fail("rle_decode", e);
+                    for (const GatherTask &gt : rc.gathers) {
+                        e = pmh_launch_dict_gather(
+                            rc.ids_dev + gt.start, gt.dict_dev, gt.n,
+                            (uint8_t *)rc.contig + gt.start * es, es, st);
+                        if (e != hipSuccess) return fail("dict_gather", e);
+                    }
+                }
+                if (rc.has_nulls) {
+                    hipError_t e = pmh_launch_level_scatter(
+                        rc.def_dev, (int64_t)rc.def_host.size(), rc.dense_dev,
+                        rc.contig, rc.valid_dev, es, st);
+                    if (e != hipSuccess) return fail("level_scatter", e);
                 }
             }
         }
@@ -891,30 +1083,52 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                                         sec.total_rows, sec.cuts, st);
     if (e != hipSuccess) return fail("partition", e);
     hipEventRecord(ev[2], st);
-    int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0);
+    int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0) |
+                (p->pu ? 4 : 0);
     if (const char *ab = getenv("PMH_ABLATE"))  // profiling-only phase knob
         flags |= (atoi(ab) & 0xf) << 8;
     e = pmh_launch_merge_tiles(sec.key_cols, sec.seq_cols, sec.kind_cols,
                                sec.lens_dev, k, sec.cuts, sec.n_tiles,
                                PMH_TILE_ROWS, flags, sec.winners,
-                               sec.tile_counts, st);
+                               sec.tile_counts, sec.group_start, sec.err_dev,
+                               st);
     if (e != hipSuccess) return fail("merge_tiles", e);
     hipEventRecord(ev[3], st);
     e = pmh_launch_scan_tiles(sec.tile_counts, sec.n_tiles, sec.tile_offsets,
                               sec.total_dev, st);
     if (e != hipSuccess) return fail("scan_tiles", e);
     hipEventRecord(ev[4], st);
-    e = pmh_launch_emit(sec.all_cols, p->col_dtype_dev, n_cols, k, sec.winners,
-                        sec.tile_counts, sec.tile_offsets, sec.n_tiles,
-                        PMH_TILE_ROWS, sec.total_dev, p->out_ptrs_dev, st);
+    if (p->pu) {
+        e = pmh_launch_emit_pu(
+            sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,
+            p->n_key_cols, p->n_key_cols + 1, sec.winners, sec.group_start,
+            sec.tile_offsets, sec.n_tiles, PMH_TILE_ROWS, sec.total_dev,
+            p->out_ptrs_dev, p->out_valid_dev, st);
+    } else {
+        e = pmh_launch_emit(sec.all_cols, p->col_dtype_dev,
+                            p->col_nullable_dev, n_cols, k, sec.winners,
+                            sec.tile_counts, sec.tile_offsets, sec.n_tiles,
+                            PMH_TILE_ROWS, sec.total_dev, p->out_ptrs_dev,
+                            p->out_valid_dev, st);
+    }
     if (e != hipSuccess) return fail("emit", e);
     hipEventRecord(ev[5], st);
 
     int64_t total = 0;
+    uint32_t err_word = 0;
     e = hipMemcpyAsync(&total, sec.total_dev, 8, hipMemcpyDeviceToHost, st);
     if (e != hipSuccess) return fail("total D2H", e);
+    e = hipMemcpyAsync(&err_word, sec.err_dev, 4, hipMemcpyDeviceToHost, st);
+    if (e != hipSuccess) return fail("err D2H", e);
     e = hipStreamSynchronize(st);
     if (e != hipSuccess) return fail("stream sync", e);
+    if (err_word & 1) {
+        set_error("partial-update with retract records is not supported "
+                  "(reference default also rejects them, "
+                  "PartialUpdateMergeFunction.java:170-186); configure "
+                  "ignore-delete/sequence-groups paths are later rounds");
+        return -1;
+    }
 
     float ms;
     hipEventElapsedTime(&ms, ev[0], ev[1]);
@@ -949,6 +1163,19 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             }
         }
     }
+    if (p->host_output) {
+        p->out_valid_host.resize(n_cols);
+        for (int c = 0; c < n_cols; c++) {
+            if (!p->out_valid[c]) continue;
+            p->out_valid_host[c].resize(total);
+            if (total > 0 &&
+                hipMemcpy(p->out_valid_host[c].data(), p->out_valid[c], total,
+                          hipMemcpyDeviceToHost) != hipSuccess) {
+                set_error("D2H validity failed");
+                return -1;
+            }
+        }
+    }
     for (int c = 0; c < n_cols; c++) {
         pmh_col &pc = p->batch_cols[c];
         pc.name = p->col_names[c].c_str();
@@ -956,6 +1183,10 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         pc.data = p->host_output ? (const void *)p->out_host[c].data()
                                  : (const void *)p->out_dev[c];
         pc.valid = nullptr;
+        if (p->out_valid[c])
+            pc.valid = p->host_output
+                           ? (const uint8_t *)p->out_valid_host[c].data()
+                           : (const uint8_t *)p->out_valid[c];
     }
     if (out) {
         out->n_rows = total;

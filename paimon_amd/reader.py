@@ -174,6 +174,7 @@ class MergeReadPlan:
         for c in range(b.n_cols):
             col = b.cols[c]
             dt = np.dtype(_DT_NP[col.dtype])
+            name = col.name.decode()
             if b.n_rows and col.data:
                 buf = ctypes.cast(
                     col.data,
@@ -181,7 +182,12 @@ class MergeReadPlan:
                 arr = np.frombuffer(bytes(buf.contents), dtype=dt)
             else:
                 arr = np.empty(0, dtype=dt)
-            out[col.name.decode()] = arr
+            out[name] = arr
+            if col.valid and b.n_rows:
+                vbuf = ctypes.cast(col.valid,
+                                   ctypes.POINTER(ctypes.c_uint8 * b.n_rows))
+                out[name + "#valid"] = np.frombuffer(
+                    bytes(vbuf.contents), dtype=np.uint8).astype(bool)
         return out
 
     def reset(self):
