@@ -417,8 +417,30 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                     while (tail + 1 < M && !sm.head[tail + 1]) tail++;
                     if (pass == 1) {
                         gout[g_off + ng] = (uint16_t)(m_off + nm);
-                        for (int32_t x = i; x <= tail; x++) {
-                            uint16_t s = mo[x];
+                        // merged order within a group is (key, run); the
+                        // overlay consumes records in ascending (seq, isAdd)
+                        // add order (SortMergeReaderWithLoserTree.java:53-63)
+                        // — insertion-sort the <= k members by seq.
+                        uint16_t gm[PMH_MAX_RUNS];
+                        int gn = tail - i + 1;
+                        for (int x = 0; x < gn; x++) {
+                            uint16_t s = mo[i + x];
+                            int y = x;
+                            while (y > 0) {
+                                uint16_t prv = gm[y - 1];
+                                int64_t s1 = sm.sseq[prv], s2 = sm.sseq[s];
+                                bool gt = s1 > s2 ||
+                                          (s1 == s2 &&
+                                           kind_is_add(sm.skind[prv]) &&
+                                           !kind_is_add(sm.skind[s]));
+                                if (!gt) break;
+                                gm[y] = gm[y - 1];
+                                y--;
+                            }
+                            gm[y] = s;
+                        }
+                        for (int x = 0; x < gn; x++) {
+                            uint16_t s = gm[x];
                             if (!kind_is_add(sm.skind[s])) bad_kind = true;
                             int r = 0;
                             while (r + 1 <= k - 1 &&
@@ -426,8 +448,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                                 r++;
                             uint32_t grow = (uint32_t)(
                                 c0[r] + ((int32_t)s - sm.segoff[r]));
-                            mout[m_off + nm + (x - i)] =
-                                ((uint32_t)r << 28) | grow;
+                            mout[m_off + nm + x] = ((uint32_t)r << 28) | grow;
                         }
                     }
                     ng++;
