@@ -258,22 +258,34 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         const int32_t *c0 = &cuts[tile * k];
         const int32_t *c1 = &cuts[(tile + 1) * k];
 
-        // --- segment setup (extended: +1 element per run when available)
+        // --- segment setup (extended: +1 element per run when available).
+        // Parallel across the first k threads — a serial tid-0 loop of ~30
+        // dependent global loads stalled the whole tile (ablation,
+        // profiles/r01_c2_pmc.md); only the tiny k-entry prefix stays serial.
+        __shared__ int64_t s_predcand[PMH_MAX_RUNS];
+        if (tid < k) {
+            int32_t a = c0[tid], b = c1[tid];
+            int32_t ext = (b < (int32_t)lens[tid]) ? 1 : 0;
+            sm.seglen[tid] = (b - a) + ext;
+            s_predcand[tid] =
+                a > 0 ? col_load<int64_t>(keys[tid], a - 1) : INT64_MIN;
+            // reuse head[] as a tiny flag channel for "has predecessor"
+            sm.head[tid] = a > 0;
+        }
+        __syncthreads();
         if (tid == 0) {
             int32_t off = 0;
             int32_t real = 0;
             int64_t pred = 0;
             int hp = 0;
             for (int r = 0; r < k; r++) {
-                int32_t len = c1[r] - c0[r];
-                real += len;
-                int32_t ext = (c1[r] < (int32_t)lens[r]) ? 1 : 0;
                 sm.segoff[r] = off;
-                sm.seglen[r] = len + ext;
-                off += len + ext;
-                if (c0[r] > 0) {
-                    int64_t kk = col_load<int64_t>(keys[r], c0[r] - 1);
-                    if (!hp || kk > pred) pred = kk;
+                int32_t ext_len = sm.seglen[r];
+                int32_t real_len = c1[r] - c0[r];
+                real += real_len;
+                off += ext_len;
+                if (sm.head[r]) {
+                    if (!hp || s_predcand[r] > pred) pred = s_predcand[r];
                     hp = 1;
                 }
             }
