@@ -234,6 +234,14 @@ DEV bool better(int64_t seq_a, uint8_t kind_a, bool elig_a, int64_t seq_b,
     return kind_is_add(kind_a) && !kind_is_add(kind_b);
 }
 
+// PU=true: PartialUpdate mode — no winner reduction; emit every owned
+// group's member list (ascending (seq, isAdd) order within the group) for
+// the per-field overlay in k_emit_pu. v1 accepts INSERT-only streams (the
+// reference's default partial-update rejects retracts,
+// PartialUpdateMergeFunction.java:170-186) — retracts set err_flag.
+// Separate template instantiations keep the PU path's extra registers out
+// of the deduplicate kernel.
+template <bool PU>
 __launch_bounds__(PMH_TILE_THREADS) __global__
 void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    const int64_t *lens, int k, const int32_t *cuts,
@@ -242,12 +250,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    uint16_t *group_start, uint32_t *err_flag) {
     const bool drop_delete = flags & 1;
     const bool ignore_delete = flags & 2;
-    // PartialUpdate mode: no winner reduction; emit every owned group's
-    // member list (ascending (seq, isAdd) order within the group) for the
-    // per-field overlay in k_emit_pu. v1 accepts INSERT-only streams (the
-    // reference's default partial-update rejects retracts,
-    // PartialUpdateMergeFunction.java:170-186) — retracts set err_flag.
-    const bool pu_mode = flags & 4;
+    constexpr bool pu_mode = PU;
     // ablation levels (profiling only, flags bits 8..): 1=stage,2=+merge,
     // 3=+scan, 0/absent=full. Partial levels publish a checksum so the
     // compiler cannot dead-code the ablated phases' inputs.
@@ -407,7 +410,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         }
         __syncthreads();
 
-        if (pu_mode) {
+        if constexpr (PU) {
             // --- PartialUpdate: emit owned groups' member lists
             __shared__ int32_t s_scan[PMH_TILE_THREADS];
             __shared__ int32_t s_scan2[PMH_TILE_THREADS];
@@ -953,10 +956,16 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
                                   uint16_t *group_start, uint32_t *err_flag,
                                   hipStream_t stream) {
     int blocks = n_tiles < 4096 ? (int)n_tiles : 4096;
-    hipLaunchKernelGGL(k_merge_tiles, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
-                       stream, keys, seqs, kinds, lens, k, cuts, n_tiles,
-                       tile_rows, flags, winners, tile_counts, group_start,
-                       err_flag);
+    if (flags & 4)
+        hipLaunchKernelGGL(k_merge_tiles<true>, dim3(blocks),
+                           dim3(PMH_TILE_THREADS), 0, stream, keys, seqs,
+                           kinds, lens, k, cuts, n_tiles, tile_rows, flags,
+                           winners, tile_counts, group_start, err_flag);
+    else
+        hipLaunchKernelGGL(k_merge_tiles<false>, dim3(blocks),
+                           dim3(PMH_TILE_THREADS), 0, stream, keys, seqs,
+                           kinds, lens, k, cuts, n_tiles, tile_rows, flags,
+                           winners, tile_counts, group_start, err_flag);
     return hipGetLastError();
 }
 
