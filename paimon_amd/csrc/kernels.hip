@@ -651,20 +651,24 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
     const int64_t slice_lo = (int64_t)blockIdx.x * per_block;
     const int64_t slice_hi = slice_lo + per_block < total
                                  ? slice_lo + per_block : total;
+    // one tile binary search per thread; ranks then advance monotonically
+    // within the slice, so the tile cursor just walks forward
+    int64_t t = -1;
     for (int64_t i0 = slice_lo + (int64_t)threadIdx.x * R; i0 < slice_hi;
          i0 += (int64_t)blockDim.x * R) {
-        // locate owning tile of i0 (binary search, L2-resident table)
-        int64_t lo = 0, hi = n_tiles - 1;
-        while (lo < hi) {
-            int64_t mid = (lo + hi + 1) >> 1;
-            if (tile_offsets[mid] <= i0) lo = mid;
-            else hi = mid - 1;
+        if (t < 0) {  // first iteration: locate owning tile of i0
+            int64_t lo = 0, hi = n_tiles - 1;
+            while (lo < hi) {
+                int64_t mid = (lo + hi + 1) >> 1;
+                if (tile_offsets[mid] <= i0) lo = mid;
+                else hi = mid - 1;
+            }
+            t = lo;
         }
         int64_t tile[R];
         int run[R];
         int64_t row[R];
         int nr = total - i0 < R ? (int)(total - i0) : R;
-        int64_t t = lo;
 #pragma unroll
         for (int x = 0; x < R; x++) {
             int64_t i = i0 + (x < nr ? x : 0);
