@@ -45,6 +45,11 @@ def get_args():
     ap.add_argument("--rows", type=int, default=10_000_000)
     ap.add_argument("--vals", type=int, default=8)
     ap.add_argument("--compression", default="NONE")
+    ap.add_argument("--engine", default="deduplicate",
+                    choices=["deduplicate", "partial-update"],
+                    help="partial-update = the C3 merge-function half on "
+                    "Parquet (supplementary; the default metric line is "
+                    "deduplicate per BASELINE.json configs[1])")
     ap.add_argument("--seed", type=int, default=42)
     ap.add_argument("--data-dir", default=os.path.join(REPO, "data"))
     ap.add_argument("--cpu-baseline-rows", type=int, default=2_000_000,
@@ -54,16 +59,24 @@ def get_args():
 
 
 def ensure_data(args, rank):
-    from paimon_amd.datagen import gen_runs_dedup, write_runs
-    tag = (f"c2_{args.runs}x{args.rows}v{args.vals}_"
+    from paimon_amd.datagen import (gen_runs_dedup, gen_runs_partial_update,
+                                    write_runs)
+    pu = args.engine == "partial-update"
+    tag = (f"{'c3pu' if pu else 'c2'}_{args.runs}x{args.rows}v{args.vals}_"
            f"{args.compression}_seed{args.seed}_rank{rank}")
     out_dir = os.path.join(args.data_dir, tag)
     manifest = os.path.join(out_dir, "files.json")
     if os.path.exists(manifest):
         with open(manifest) as f:
             return json.load(f), out_dir
-    runs = gen_runs_dedup(args.runs, args.rows, n_value_cols=args.vals,
-                          seed=args.seed + rank)
+    if pu:
+        runs = gen_runs_partial_update(args.runs, args.rows,
+                                       n_value_cols=args.vals,
+                                       seed=args.seed + rank,
+                                       update_frac=0.3, update_cols=6)
+    else:
+        runs = gen_runs_dedup(args.runs, args.rows, n_value_cols=args.vals,
+                              seed=args.seed + rank)
     metas = write_runs(runs, out_dir, compression=args.compression)
     return metas, out_dir
 
@@ -126,7 +139,7 @@ def main():
                    for i in range(args.vals)])
     t_stage0 = time.perf_counter()
     plan = MergeReadPlan(sess, file_descs_from_metas(metas), key_cols,
-                         value_cols, merge_engine="deduplicate",
+                         value_cols, merge_engine=args.engine,
                          drop_delete=True, output="device")
     t_stage = time.perf_counter() - t_stage0
 
@@ -216,8 +229,12 @@ def main():
         "config": {
             "workload": (f"merge-on-read: {args.runs} sorted runs x "
                          f"{args.rows} rows, int64 PK + {args.vals} int32, "
-                         f"Parquet {args.compression}, Deduplicate, "
-                         "drop-delete (BASELINE.json configs[1])"),
+                         f"Parquet {args.compression}, "
+                         + ("PartialUpdate (30% updates, 6-col subsets; "
+                            "parquet variant of configs[2])"
+                            if args.engine == "partial-update" else
+                            "Deduplicate, drop-delete "
+                            "(BASELINE.json configs[1])")),
             "n_runs": args.runs,
             "rows_per_run": args.rows,
             "value_cols": args.vals,
