@@ -652,33 +652,46 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
     const int64_t slice_hi = slice_lo + per_block < total
                                  ? slice_lo + per_block : total;
     // one tile binary search per thread; ranks then advance monotonically
-    // within the slice, so the tile cursor just walks forward
+    // within the slice, so the tile cursor just walks forward. The R rows
+    // per thread are WAVE-STRIDED (i = base + tid + x*blockDim), so every
+    // access stays fully coalesced while keeping R loads in flight —
+    // thread-contiguous R strides the lanes by R and breaks coalescing.
     int64_t t = -1;
-    for (int64_t i0 = slice_lo + (int64_t)threadIdx.x * R; i0 < slice_hi;
-         i0 += (int64_t)blockDim.x * R) {
-        if (t < 0) {  // first iteration: locate owning tile of i0
+    for (int64_t base = slice_lo; base < slice_hi;
+         base += (int64_t)blockDim.x * R) {
+        int64_t i_first = base + threadIdx.x;
+        if (i_first >= slice_hi) break;
+        if (t < 0) {  // first iteration: locate owning tile
             int64_t lo = 0, hi = n_tiles - 1;
             while (lo < hi) {
                 int64_t mid = (lo + hi + 1) >> 1;
-                if (tile_offsets[mid] <= i0) lo = mid;
+                if (tile_offsets[mid] <= i_first) lo = mid;
                 else hi = mid - 1;
             }
             t = lo;
         }
-        int64_t tile[R];
+        int64_t idx[R];
         int run[R];
         int64_t row[R];
-        int nr = total - i0 < R ? (int)(total - i0) : R;
+        int nr = 0;
 #pragma unroll
         for (int x = 0; x < R; x++) {
-            int64_t i = i0 + (x < nr ? x : 0);
-            // dense ranks: advance tile while i falls past its count
-            while (t + 1 < n_tiles && tile_offsets[t + 1] <= i) t++;
-            tile[x] = t;
-            uint32_t packed =
-                winners[t * (tile_rows + PMH_MAX_RUNS) + (i - tile_offsets[t])];
-            run[x] = packed >> 28;
-            row[x] = packed & 0x0fffffff;
+            int64_t i = base + threadIdx.x + (int64_t)x * blockDim.x;
+            bool live = i < slice_hi;
+            idx[x] = i;
+            if (live) {
+                nr = x + 1;
+                // dense ranks: advance tile while i falls past its count
+                while (t + 1 < n_tiles && tile_offsets[t + 1] <= i) t++;
+                uint32_t packed =
+                    winners[t * (tile_rows + PMH_MAX_RUNS) +
+                            (i - tile_offsets[t])];
+                run[x] = packed >> 28;
+                row[x] = packed & 0x0fffffff;
+            } else {  // dead lane: duplicate x=0's gather, stores are guarded
+                run[x] = run[0];
+                row[x] = row[0];
+            }
         }
         for (int c = 0; c < n_cols; c++) {
             if (col_nullable[c] && out_valid[c]) {
@@ -687,7 +700,7 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                 for (int x = 0; x < R; x++) {
                     if (x >= nr) continue;
                     const DevCol &dc = cols[run[x] * n_cols + c];
-                    out_valid[c][i0 + x] =
+                    out_valid[c][idx[x]] =
                         dc.valid0 ? ((const uint8_t *)dc.valid0)[row[x]] : 1;
                 }
             }
@@ -700,7 +713,7 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
 #pragma unroll
                 for (int x = 0; x < R; x++)
                     if (x < nr)
-                        ((int8_t *)out_ptrs[c])[i0 + x] = (int8_t)v[x];
+                        ((int8_t *)out_ptrs[c])[idx[x]] = (int8_t)v[x];
                 break;
             }
             case 3:
@@ -711,7 +724,7 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                     v[x] = col_load<int32_t>(cols[run[x] * n_cols + c], row[x]);
 #pragma unroll
                 for (int x = 0; x < R; x++)
-                    if (x < nr) ((int32_t *)out_ptrs[c])[i0 + x] = v[x];
+                    if (x < nr) ((int32_t *)out_ptrs[c])[idx[x]] = v[x];
                 break;
             }
             case 4:
@@ -722,7 +735,7 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                     v[x] = col_load<int64_t>(cols[run[x] * n_cols + c], row[x]);
 #pragma unroll
                 for (int x = 0; x < R; x++)
-                    if (x < nr) ((int64_t *)out_ptrs[c])[i0 + x] = v[x];
+                    if (x < nr) ((int64_t *)out_ptrs[c])[idx[x]] = v[x];
                 break;
             }
             default: break;
