@@ -183,3 +183,54 @@ class TestDedupParity:
                 assert st["rows_in"] == 20_000
                 assert st["rows_out"] > 0
                 assert st["total_device_ms"] > 0
+
+
+class TestMultiFileRuns:
+    def test_files_chain_into_runs(self, tmp_path):
+        # disjoint-range files chain into one SortedRun (IntervalPartition
+        # min-heap packing); an overlapping file forms its own run. The
+        # staged run is the concatenation of its files (SortedRun.fromSorted).
+        import pyarrow.parquet as pq
+        from paimon_amd.datagen import run_to_arrow
+        rng = np.random.default_rng(60)
+        seqs = rng.permutation(10_000).astype(np.int64)
+
+        def mk(keys, seq):
+            keys = np.sort(np.asarray(keys, np.int64))
+            return {"key": keys, "seq": seq[:len(keys)],
+                    "kind": np.zeros(len(keys), np.int8),
+                    "values": [keys.copy()]}
+
+        fa = mk(rng.choice(1000, 800, replace=False), seqs[:800])
+        fb = mk(rng.choice(1000, 800, replace=False) + 2000, seqs[800:1600])
+        fc = mk(rng.choice(2500, 1500, replace=False) + 400, seqs[1600:3100])
+        metas = []
+        for name, f in (("a", fa), ("b", fb), ("c", fc)):
+            path = str(tmp_path / f"{name}.parquet")
+            pq.write_table(run_to_arrow(f), path, compression=None,
+                           use_dictionary=False, data_page_version="1.0",
+                           store_schema=False)
+            metas.append({"path": path, "rowCount": len(f["key"]),
+                          "minKey": int(f["key"][0]),
+                          "maxKey": int(f["key"][-1]), "level": 0})
+        # oracle: run1 = concat(A, B) (chained), run2 = C
+        runs = [{k: np.concatenate([fa[k], fb[k]]) for k in ("key", "seq", "kind")},
+                {k: fc[k] for k in ("key", "seq", "kind")}]
+        runs[0]["values"] = [np.concatenate([fa["values"][0], fb["values"][0]])]
+        runs[1]["values"] = [fc["values"][0]]
+        r, w = merge_dedup(runs, drop_delete=True)
+        exp_key = np.array([runs[a]["key"][b] for a, b in zip(r, w)], np.int64)
+        exp_seq = np.array([runs[a]["seq"][b] for a, b in zip(r, w)], np.int64)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               [{"name": "v_k", "type": "int64"}]) as plan:
+                got = {}
+                while True:
+                    b = plan.read_next()
+                    if b is None:
+                        break
+                    for kk, v in b.items():
+                        got.setdefault(kk, []).append(v.copy())
+                got = {kk: np.concatenate(v) for kk, v in got.items()}
+        assert (got["_KEY_k"] == exp_key).all()
+        assert (got["_SEQUENCE_NUMBER"] == exp_seq).all()
