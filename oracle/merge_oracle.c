@@ -431,3 +431,212 @@ int64_t pmo_merge_dedup_count(int n_runs, const int64_t **keys,
     free(orow);
     return n;
 }
+
+/* ================================================================ *
+ * ORC RLEv2 integer decoder — restatement of the published ORC v1
+ * specification ("Run Length Encoding version 2") as implemented by the
+ * orc-core 1.9.8 dependency of the reference (RunLengthIntegerReaderV2;
+ * the dep is NOT vendored under /root/reference — SURVEY.md §8c — so
+ * parity is pinned via pyarrow.orc round trips + synthetic KATs).
+ * Sub-encodings by the top 2 bits of the first byte:
+ *   00 SHORT_REPEAT, 01 DIRECT, 10 PATCHED_BASE, 11 DELTA.
+ * ================================================================ */
+
+static const int pmo_fbs[32] = {1,  2,  3,  4,  5,  6,  7,  8,
+                                9,  10, 11, 12, 13, 14, 15, 16,
+                                17, 18, 19, 20, 21, 22, 23, 24,
+                                26, 28, 30, 32, 40, 48, 56, 64};
+
+static int pmo_closest_fixed_bits(int n) {
+    for (int i = 0; i < 32; i++)
+        if (pmo_fbs[i] >= n) return pmo_fbs[i];
+    return 64;
+}
+
+typedef struct {
+    const uint8_t *p;
+    const uint8_t *end;
+    int err;
+} pmo_bs;
+
+static uint64_t bs_byte(pmo_bs *b) {
+    if (b->p >= b->end) {
+        b->err = 1;
+        return 0;
+    }
+    return *b->p++;
+}
+
+static uint64_t bs_be(pmo_bs *b, int nbytes) {
+    uint64_t v = 0;
+    for (int i = 0; i < nbytes; i++) v = (v << 8) | bs_byte(b);
+    return v;
+}
+
+static uint64_t bs_uvarint(pmo_bs *b) {
+    uint64_t v = 0;
+    int shift = 0;
+    for (;;) {
+        uint64_t x = bs_byte(b);
+        v |= (x & 0x7F) << shift;
+        if (!(x & 0x80)) return v;
+        shift += 7;
+    }
+}
+
+static int64_t bs_svarint(pmo_bs *b) {
+    uint64_t v = bs_uvarint(b);
+    return (int64_t)(v >> 1) ^ -(int64_t)(v & 1);
+}
+
+/* big-endian bit-packed reader */
+typedef struct {
+    pmo_bs *b;
+    uint64_t cur;
+    int bits_left;
+} pmo_br;
+
+static uint64_t br_read(pmo_br *r, int width) {
+    uint64_t v = 0;
+    int need = width;
+    while (need > 0) {
+        if (r->bits_left == 0) {
+            r->cur = bs_byte(r->b);
+            r->bits_left = 8;
+        }
+        int take = need < r->bits_left ? need : r->bits_left;
+        v = (v << take) |
+            ((r->cur >> (r->bits_left - take)) & ((1ull << take) - 1));
+        r->bits_left -= take;
+        need -= take;
+    }
+    return v;
+}
+
+static int64_t pmo_zz(uint64_t v) {
+    return (int64_t)(v >> 1) ^ -(int64_t)(v & 1);
+}
+
+/* Decode exactly n values; returns n or -1 on malformed input. */
+int64_t pmo_orc_rlev2_decode(const uint8_t *in, int64_t len, int64_t n,
+                             int is_signed, int64_t *out) {
+    pmo_bs b = {in, in + len, 0};
+    int64_t cnt = 0;
+    while (cnt < n && !b.err) {
+        uint64_t first = bs_byte(&b);
+        int enc = (int)(first >> 6) & 3;
+        if (enc == 0) { /* SHORT_REPEAT */
+            int w = ((int)(first >> 3) & 7) + 1;
+            int rep = ((int)first & 7) + 3;
+            uint64_t raw = bs_be(&b, w);
+            int64_t v = is_signed ? pmo_zz(raw) : (int64_t)raw;
+            for (int i = 0; i < rep && cnt < n; i++) out[cnt++] = v;
+        } else if (enc == 1) { /* DIRECT */
+            int width = pmo_fbs[(first >> 1) & 0x1f];
+            int count = (int)(((first & 1) << 8) | bs_byte(&b)) + 1;
+            pmo_br r = {&b, 0, 0};
+            for (int i = 0; i < count; i++) {
+                uint64_t raw = br_read(&r, width);
+                if (cnt < n)
+                    out[cnt++] = is_signed ? pmo_zz(raw) : (int64_t)raw;
+            }
+        } else if (enc == 2) { /* PATCHED_BASE */
+            int width = pmo_fbs[(first >> 1) & 0x1f];
+            int count = (int)(((first & 1) << 8) | bs_byte(&b)) + 1;
+            uint64_t third = bs_byte(&b), fourth = bs_byte(&b);
+            int bw = ((int)(third >> 5) & 7) + 1;
+            int pw = pmo_fbs[third & 0x1f];
+            int pgw = ((int)(fourth >> 5) & 7) + 1;
+            int pl = (int)fourth & 0x1f;
+            uint64_t braw = bs_be(&b, bw);
+            uint64_t smask = 1ull << (bw * 8 - 1);
+            int64_t base = (braw & smask) ? -(int64_t)(braw & ~smask)
+                                          : (int64_t)braw;
+            if (count > 512 || pl > 32) return -1;
+            uint64_t vals[512];
+            pmo_br r = {&b, 0, 0};
+            for (int i = 0; i < count; i++) vals[i] = br_read(&r, width);
+            int cfb = pmo_closest_fixed_bits(pw + pgw);
+            uint64_t patches[32];
+            pmo_br r2 = {&b, 0, 0};
+            for (int i = 0; i < pl; i++) patches[i] = br_read(&r2, cfb);
+            uint64_t pmask = (pw == 64) ? ~0ull : ((1ull << pw) - 1);
+            int64_t gap = 0;
+            int pidx = 0;
+            /* apply patches: gap accumulates; (gap=max,patch=0) chains */
+            int64_t actual = -1;
+            while (pidx < pl) {
+                uint64_t g = patches[pidx] >> pw;
+                uint64_t pv = patches[pidx] & pmask;
+                pidx++;
+                gap += (int64_t)g;
+                if (pv == 0 && g == ((1ull << pgw) - 1)) continue;
+                actual += gap + 1;
+                gap = 0;
+                if (actual >= 0 && actual < count)
+                    vals[actual] |= pv << width;
+                else
+                    return -1;
+            }
+            for (int i = 0; i < count && cnt < n; i++)
+                out[cnt++] = base + (int64_t)vals[i];
+        } else { /* DELTA */
+            int wcode = (first >> 1) & 0x1f;
+            int width = wcode == 0 ? 0 : pmo_fbs[wcode];
+            int count = (int)(((first & 1) << 8) | bs_byte(&b)) + 1;
+            int64_t base = is_signed ? bs_svarint(&b)
+                                     : (int64_t)bs_uvarint(&b);
+            int64_t delta = bs_svarint(&b);
+            if (cnt < n) out[cnt] = base;
+            cnt++;
+            int64_t prev = base;
+            if (width == 0) {
+                for (int i = 1; i < count; i++) {
+                    prev += delta;
+                    if (cnt < n) out[cnt] = prev;
+                    cnt++;
+                }
+            } else {
+                prev = base + delta;
+                if (count > 1) {
+                    if (cnt < n) out[cnt] = prev;
+                    cnt++;
+                }
+                pmo_br r = {&b, 0, 0};
+                for (int i = 2; i < count; i++) {
+                    uint64_t d = br_read(&r, width);
+                    prev += delta < 0 ? -(int64_t)d : (int64_t)d;
+                    if (cnt < n) out[cnt] = prev;
+                    cnt++;
+                }
+            }
+        }
+    }
+    return b.err ? -1 : cnt;
+}
+
+/* ORC boolean run: outer byte-RLE (header h: h>=0 -> h+3 copies of next
+ * byte; h<0 -> -h literal bytes) over MSB-first bit-packed bytes. Used by
+ * PRESENT streams. Emits one byte (0/1) per value. */
+int64_t pmo_orc_boolrle_decode(const uint8_t *in, int64_t len, int64_t n,
+                               uint8_t *out) {
+    pmo_bs b = {in, in + len, 0};
+    int64_t cnt = 0;
+    while (cnt < n && !b.err) {
+        int8_t h = (int8_t)bs_byte(&b);
+        if (b.err) break;
+        if (h >= 0) { /* run of h+3 repeated bytes */
+            uint8_t v = (uint8_t)bs_byte(&b);
+            for (int i = 0; i < h + 3; i++)
+                for (int bit = 7; bit >= 0 && cnt < n; bit--)
+                    out[cnt++] = (v >> bit) & 1;
+        } else {
+            for (int i = 0; i < -(int)h; i++) {
+                uint8_t v = (uint8_t)bs_byte(&b);
+                for (int bit = 7; bit >= 0 && cnt < n; bit--)
+                    out[cnt++] = (v >> bit) & 1;
+            }
+        }
+    }
+    return b.err ? -1 : cnt;
+}
