@@ -561,22 +561,21 @@ int64_t pmo_orc_rlev2_decode(const uint8_t *in, int64_t len, int64_t n,
             pmo_br r2 = {&b, 0, 0};
             for (int i = 0; i < pl; i++) patches[i] = br_read(&r2, cfb);
             uint64_t pmask = (pw == 64) ? ~0ull : ((1ull << pw) - 1);
-            int64_t gap = 0;
-            int pidx = 0;
-            /* apply patches: gap accumulates; (gap=max,patch=0) chains */
-            int64_t actual = -1;
-            while (pidx < pl) {
+            /* apply patches: positions are cumulative gaps from position 0;
+             * an entry with gap == 2^pgw-1 and patch == 0 only extends the
+             * gap (verified against pyarrow-written streams) */
+            int64_t gap = 0, pos = 0;
+            int started = 0;
+            for (int pidx = 0; pidx < pl; pidx++) {
                 uint64_t g = patches[pidx] >> pw;
                 uint64_t pv = patches[pidx] & pmask;
-                pidx++;
                 gap += (int64_t)g;
                 if (pv == 0 && g == ((1ull << pgw) - 1)) continue;
-                actual += gap + 1;
+                pos = started ? pos + gap : gap;
+                started = 1;
                 gap = 0;
-                if (actual >= 0 && actual < count)
-                    vals[actual] |= pv << width;
-                else
-                    return -1;
+                if (pos < 0 || pos >= count) return -1;
+                vals[pos] |= pv << width;
             }
             for (int i = 0; i < count && cnt < n; i++)
                 out[cnt++] = base + (int64_t)vals[i];

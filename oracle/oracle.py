@@ -206,3 +206,35 @@ def rle_bp_decode(data: bytes, bit_width: int, num_values: int):
     if n != num_values:
         raise ValueError(f"rle decode overrun: {n} != {num_values}")
     return out
+
+
+def orc_rlev2_decode(data: bytes, num_values: int, signed: bool = True):
+    """C restatement of ORC RLEv2 integer decode (SHORT_REPEAT / DIRECT /
+    PATCHED_BASE / DELTA)."""
+    lib = _get_lib()
+    lib.pmo_orc_rlev2_decode.restype = ctypes.c_int64
+    lib.pmo_orc_rlev2_decode.argtypes = [
+        ctypes.c_char_p, ctypes.c_int64, ctypes.c_int64, ctypes.c_int,
+        ctypes.POINTER(ctypes.c_int64)]
+    out = np.empty(num_values, dtype=np.int64)
+    n = lib.pmo_orc_rlev2_decode(
+        data, len(data), num_values, int(signed),
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)))
+    if n != num_values:
+        raise ValueError(f"rlev2 decode: {n} != {num_values}")
+    return out
+
+
+def orc_boolrle_decode(data: bytes, num_values: int):
+    lib = _get_lib()
+    lib.pmo_orc_boolrle_decode.restype = ctypes.c_int64
+    lib.pmo_orc_boolrle_decode.argtypes = [
+        ctypes.c_char_p, ctypes.c_int64, ctypes.c_int64,
+        ctypes.POINTER(ctypes.c_uint8)]
+    out = np.empty(num_values, dtype=np.uint8)
+    n = lib.pmo_orc_boolrle_decode(
+        data, len(data), num_values,
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_uint8)))
+    if n != num_values:
+        raise ValueError(f"bool rle decode: {n} != {num_values}")
+    return out.astype(bool)

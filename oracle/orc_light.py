@@ -108,9 +108,22 @@ def parse_orc(path) -> OrcFileInfo:
     types = [
         _fields(t) for t in footer.get(4, [])
     ]
+    def _ints(vals):
+        # repeated uint32 may arrive packed (length-delimited varint blob)
+        out = []
+        for v in vals:
+            if isinstance(v, int):
+                out.append(v)
+            else:
+                p = 0
+                while p < len(v):
+                    x, p = _uvarint(v, p)
+                    out.append(x)
+        return out
+
     root = types[0]
     names = [x.decode() for x in root.get(3, [])]
-    kinds = [types[i].get(1, [0])[0] for i in root.get(2, [])]
+    kinds = [types[i].get(1, [0])[0] for i in _ints(root.get(2, []))]
     fi = OrcFileInfo(num_rows=footer.get(6, [0])[0], compression=compression,
                      column_names=names, column_kinds=kinds)
     for s in footer.get(3, []):
