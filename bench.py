@@ -45,6 +45,8 @@ def get_args():
     ap.add_argument("--rows", type=int, default=10_000_000)
     ap.add_argument("--vals", type=int, default=8)
     ap.add_argument("--compression", default="NONE")
+    ap.add_argument("--format", dest="file_format", default="parquet",
+                    choices=["parquet", "orc"])
     ap.add_argument("--engine", default="deduplicate",
                     choices=["deduplicate", "partial-update"],
                     help="partial-update = the C3 merge-function half on "
@@ -63,7 +65,7 @@ def ensure_data(args, rank):
                                     write_runs)
     pu = args.engine == "partial-update"
     tag = (f"{'c3pu' if pu else 'c2'}_{args.runs}x{args.rows}v{args.vals}_"
-           f"{args.compression}_seed{args.seed}_rank{rank}")
+           f"{args.file_format}_{args.compression}_seed{args.seed}_rank{rank}")
     out_dir = os.path.join(args.data_dir, tag)
     manifest = os.path.join(out_dir, "files.json")
     if os.path.exists(manifest):
@@ -77,7 +79,8 @@ def ensure_data(args, rank):
     else:
         runs = gen_runs_dedup(args.runs, args.rows, n_value_cols=args.vals,
                               seed=args.seed + rank)
-    metas = write_runs(runs, out_dir, compression=args.compression)
+    metas = write_runs(runs, out_dir, compression=args.compression,
+                       file_format=args.file_format)
     return metas, out_dir
 
 
@@ -229,7 +232,7 @@ def main():
         "config": {
             "workload": (f"merge-on-read: {args.runs} sorted runs x "
                          f"{args.rows} rows, int64 PK + {args.vals} int32, "
-                         f"Parquet {args.compression}, "
+                         f"{args.file_format} {args.compression}, "
                          + ("PartialUpdate (30% updates, 6-col subsets; "
                             "parquet variant of configs[2])"
                             if args.engine == "partial-update" else

@@ -48,7 +48,37 @@ struct RleChunk {
                         // this chunk within its (run, column)
 };
 
+// Host-prescanned ORC RLEv2 / byte-RLE work unit (one run per chunk; runs
+// are <= 512 values and byte-aligned). Restates the ORC v1 spec RLEv2 as
+// consumed by the reference through orc-core 1.9.8.
+struct Rlev2Chunk {
+    uint64_t src;        // packed values (byte-aligned within DATA stream)
+    int64_t out_start;   // dense output element index
+    int32_t count;
+    uint8_t kind;        // 0 SHORT_REPEAT, 1 DIRECT, 2 PATCHED_BASE,
+                         // 3 DELTA, 4 BYTE_RUN, 5 BYTE_LITERAL
+    uint8_t width;       // packed bit width (0 = fixed-delta / none)
+    uint8_t is_signed;   // zigzag-decode DIRECT/SHORT_REPEAT values
+    uint8_t out_esize;   // 4 or 8
+    int64_t base;        // SR/BYTE_RUN: value; DELTA/PATCHED: base
+    int64_t delta;       // DELTA: delta base (sign = direction)
+    uint64_t patch_src;  // PATCHED: packed patch entries
+    uint16_t patch_pl;   // PATCHED: number of patch entries
+    uint8_t patch_pw;    // PATCHED: patch value bits (decoded)
+    uint8_t patch_pgw;   // PATCHED: gap bits
+    uint8_t patch_cfb;   // PATCHED: packed entry bits
+    uint8_t dense_target;  // 0: write contig (positioned); 1: write the
+                           // dense buffer (PRESENT scatter follows)
+    uint8_t _pad[2];
+};
+
 extern "C" {
+
+// Decode ORC RLEv2 / byte-RLE work chunks into a dense typed column
+// (int32 or int64 elements per chunk.out_esize). One wave per chunk.
+hipError_t pmh_launch_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks,
+                            void *out_contig, void *out_dense,
+                            hipStream_t stream);
 
 hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
                                 int64_t tile_rows, int64_t n_bounds,

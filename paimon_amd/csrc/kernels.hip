@@ -908,6 +908,151 @@ __global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks,
     }
 }
 
+// ------------------------------------------------------------- k_rlev2
+//
+// ORC RLEv2 / byte-RLE decode from host-prescanned run chunks (ORC v1 spec;
+// the reference consumes this via orc-core 1.9.8 — parity pinned by the
+// oracle restatement + pyarrow). One 64-lane wave per chunk
+// (runs are <= 512 values); bit order is big-endian MSB-first.
+
+DEV uint64_t be_bits(const uint8_t *src, int64_t bit_off, int width) {
+    // read `width` bits (<= 64) starting at bit_off, MSB-first
+    uint64_t v = 0;
+    int need = width;
+    int64_t byte = bit_off >> 3;
+    int used = (int)(bit_off & 7);
+    while (need > 0) {
+        int avail = 8 - used;
+        int take = need < avail ? need : avail;
+        uint64_t b = src[byte];
+        v = (v << take) | ((b >> (avail - take)) & ((1ull << take) - 1));
+        need -= take;
+        used += take;
+        if (used == 8) {
+            used = 0;
+            byte++;
+        }
+    }
+    return v;
+}
+
+DEV int64_t zz_dec(uint64_t v) {
+    return (int64_t)(v >> 1) ^ -(int64_t)(v & 1);
+}
+
+DEV void dense_store(void *out, int esize, int64_t idx, int64_t v) {
+    if (esize == 4)
+        ((int32_t *)out)[idx] = (int32_t)v;
+    else
+        ((int64_t *)out)[idx] = v;
+}
+
+__global__ void k_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks,
+                        void *out_contig, void *out_dense) {
+    const int wave = (int)(threadIdx.x >> 6);
+    const int lane = (int)(threadIdx.x & 63);
+    const int waves = (int)(blockDim.x >> 6);
+    for (int64_t cidx = (int64_t)blockIdx.x * waves + wave; cidx < n_chunks;
+         cidx += (int64_t)gridDim.x * waves) {
+        Rlev2Chunk ch = chunks[cidx];
+        void *out = ch.dense_target ? out_dense : out_contig;
+        const uint8_t *src = (const uint8_t *)ch.src;
+        switch (ch.kind) {
+        case 0:  // SHORT_REPEAT (value already sign-decoded by the host)
+        case 4:  // BYTE_RUN
+            for (int i = lane; i < ch.count; i += 64)
+                dense_store(out, ch.out_esize, ch.out_start + i, ch.base);
+            break;
+        case 5:  // BYTE_LITERAL (ORC tinyint byte stream, sign-extended)
+            for (int i = lane; i < ch.count; i += 64)
+                dense_store(out, ch.out_esize, ch.out_start + i,
+                            (int8_t)src[i]);
+            break;
+        case 1: {  // DIRECT
+            for (int i = lane; i < ch.count; i += 64) {
+                uint64_t raw = be_bits(src, (int64_t)i * ch.width, ch.width);
+                int64_t v = ch.is_signed ? zz_dec(raw) : (int64_t)raw;
+                dense_store(out, ch.out_esize, ch.out_start + i, v);
+            }
+            break;
+        }
+        case 2: {  // PATCHED_BASE: unsigned packed + base; patches on lane 0
+            for (int i = lane; i < ch.count; i += 64) {
+                uint64_t raw = be_bits(src, (int64_t)i * ch.width, ch.width);
+                dense_store(out, ch.out_esize, ch.out_start + i,
+                            ch.base + (int64_t)raw);
+            }
+            __builtin_amdgcn_s_waitcnt(0);  // drain this wave's stores
+            if (lane == 0) {
+                const uint8_t *psrc = (const uint8_t *)ch.patch_src;
+                uint64_t pmask = ch.patch_pw >= 64
+                                     ? ~0ull
+                                     : ((1ull << ch.patch_pw) - 1);
+                int64_t pos = 0, gap = 0;
+                int started = 0;
+                for (int pidx = 0; pidx < ch.patch_pl; pidx++) {
+                    uint64_t e = be_bits(psrc, (int64_t)pidx * ch.patch_cfb,
+                                         ch.patch_cfb);
+                    uint64_t g = e >> ch.patch_pw;
+                    uint64_t pv = e & pmask;
+                    gap += (int64_t)g;
+                    if (pv == 0 && g == ((1ull << ch.patch_pgw) - 1))
+                        continue;
+                    pos = started ? pos + gap : gap;
+                    started = 1;
+                    gap = 0;
+                    if (pos >= 0 && pos < ch.count) {
+                        uint64_t raw =
+                            be_bits(src, pos * (int64_t)ch.width, ch.width) |
+                            (pv << ch.width);
+                        dense_store(out, ch.out_esize, ch.out_start + pos,
+                                    ch.base + (int64_t)raw);
+                    }
+                }
+            }
+            break;
+        }
+        case 3: {  // DELTA
+            if (ch.width == 0) {  // fixed delta
+                for (int i = lane; i < ch.count; i += 64)
+                    dense_store(out, ch.out_esize, ch.out_start + i,
+                                ch.base + (int64_t)i * ch.delta);
+            } else {
+                // packed |deltas| for elements 2..count-1; direction =
+                // sign(delta). Wave-parallel prefix over 64-lane rounds.
+                if (lane == 0) {
+                    dense_store(out, ch.out_esize, ch.out_start, ch.base);
+                    if (ch.count > 1)
+                        dense_store(out, ch.out_esize, ch.out_start + 1,
+                                    ch.base + ch.delta);
+                }
+                int64_t run = ch.base + ch.delta;  // value at index 1
+                int sign = ch.delta < 0 ? -1 : 1;
+                for (int b = 0; b < ch.count - 2; b += 64) {
+                    int i = b + lane;  // delta index (element 2 + i)
+                    int64_t d = 0;
+                    if (i < ch.count - 2)
+                        d = (int64_t)be_bits(src, (int64_t)i * ch.width,
+                                             ch.width);
+                    // inclusive wave scan of deltas
+                    int64_t acc = d;
+                    for (int off = 1; off < 64; off <<= 1) {
+                        int64_t up = __shfl_up(acc, off, 64);
+                        if (lane >= off) acc += up;
+                    }
+                    if (i < ch.count - 2)
+                        dense_store(out, ch.out_esize, ch.out_start + 2 + i,
+                                    run + sign * acc);
+                    run += sign * __shfl(acc, 63, 64);
+                }
+            }
+            break;
+        }
+        default: break;
+        }
+    }
+}
+
 // ------------------------------------------------------------ k_rle_decode
 // Decode Parquet RLE/bit-packed hybrid streams (dictionary ids, def levels)
 // from host-prescanned run chunks (VectorizedRleValuesReader.java:977-1018
@@ -1033,6 +1178,17 @@ hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
                        col_dtype, col_nullable, n_cols, k, winners,
                        tile_counts, tile_offsets, n_tiles, tile_rows,
                        total_out, out_ptrs, out_valid);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks,
+                            void *out_contig, void *out_dense,
+                            hipStream_t stream) {
+    int waves_per_block = 4;  // 256 threads
+    int64_t want = (n_chunks + waves_per_block - 1) / waves_per_block;
+    int blocks = want < 4096 ? (int)(want ? want : 1) : 4096;
+    hipLaunchKernelGGL(k_rlev2, dim3(blocks), dim3(256), 0, stream, chunks,
+                       n_chunks, out_contig, out_dense);
     return hipGetLastError();
 }
 

@@ -24,6 +24,7 @@
 #include "common.h"
 #include "json.h"
 #include "kernels.h"
+#include "orc_meta.h"
 #include "parquet_meta.h"
 
 #include <dlfcn.h>
@@ -198,6 +199,10 @@ struct RunCol {
     void *dense_dev = nullptr;
     RleChunk *def_dev = nullptr;
     uint8_t *valid_dev = nullptr;
+    // ORC: RLEv2 / byte-RLE work chunks decode on the GPU at read time
+    bool orc_encoded = false;
+    std::vector<Rlev2Chunk> rlev2_host;
+    Rlev2Chunk *rlev2_dev = nullptr;
 };
 
 struct Run {
@@ -465,7 +470,9 @@ static bool prescan_rle(const uint8_t *s, int64_t len, int bit_width,
 
 struct StagedFile {
     std::vector<uint8_t> data;  // whole file (or decompressed payload view)
+    bool orc = false;
     ParquetFileMeta meta;
+    OrcFileMeta orc_meta;
 };
 
 static bool load_file(const std::string &path, StagedFile &sf) {
@@ -478,6 +485,15 @@ static bool load_file(const std::string &path, StagedFile &sf) {
     f.seekg(0);
     sf.data.resize(n);
     f.read((char *)sf.data.data(), n);
+    if (is_orc_file(sf.data.data(), n)) {
+        sf.orc = true;
+        sf.orc_meta = parse_orc_meta(sf.data.data(), n);
+        if (!sf.orc_meta.ok()) {
+            set_error("%s: %s", path.c_str(), sf.orc_meta.error.c_str());
+            return false;
+        }
+        return true;
+    }
     sf.meta = parse_parquet_footer(sf.data.data(), n);
     if (!sf.meta.ok()) {
         set_error("%s: %s", path.c_str(), sf.meta.error.c_str());
@@ -490,6 +506,327 @@ static bool load_file(const std::string &path, StagedFile &sf) {
                 set_error("%s: %s", path.c_str(), err.c_str());
                 return false;
             }
+    return true;
+}
+
+// ------------------------------------------------- ORC staging (RLEv2)
+
+// Walk an ORC RLEv2 stream (runs <= 512 values, byte-aligned) and emit one
+// device work chunk per run. dev_base = device address of the uploaded
+// stream bytes; dense0 = dense output offset of this stream's first value.
+static bool prescan_rlev2(const uint8_t *s, int64_t len, int64_t n_values,
+                          int is_signed, int out_esize, int dense_target,
+                          uint64_t dev_base, int64_t dense0,
+                          std::vector<Rlev2Chunk> &out) {
+    static const int fbs[32] = {1,  2,  3,  4,  5,  6,  7,  8,
+                                9,  10, 11, 12, 13, 14, 15, 16,
+                                17, 18, 19, 20, 21, 22, 23, 24,
+                                26, 28, 30, 32, 40, 48, 56, 64};
+    int64_t p = 0, cnt = 0;
+    auto uvarint = [&](bool &ok) -> uint64_t {
+        uint64_t v = 0;
+        int shift = 0;
+        for (;;) {
+            if (p >= len) { ok = false; return 0; }
+            uint8_t b = s[p++];
+            v |= (uint64_t)(b & 0x7F) << shift;
+            if (!(b & 0x80)) return v;
+            shift += 7;
+        }
+    };
+    while (cnt < n_values) {
+        if (p >= len) {
+            set_error("RLEv2 stream overrun");
+            return false;
+        }
+        uint8_t first = s[p++];
+        int enc = (first >> 6) & 3;
+        Rlev2Chunk c{};
+        c.is_signed = (uint8_t)is_signed;
+        c.out_esize = (uint8_t)out_esize;
+        c.dense_target = (uint8_t)dense_target;
+        c.out_start = dense0 + cnt;
+        if (enc == 0) {  // SHORT_REPEAT: host decodes the literal
+            int w = ((first >> 3) & 7) + 1;
+            int rep = (first & 7) + 3;
+            uint64_t raw = 0;
+            for (int i = 0; i < w; i++) raw = (raw << 8) | s[p++];
+            c.kind = 0;
+            c.base = is_signed
+                         ? ((int64_t)(raw >> 1) ^ -(int64_t)(raw & 1))
+                         : (int64_t)raw;
+            c.count = rep;
+            cnt += rep;
+        } else if (enc == 1) {  // DIRECT
+            int width = fbs[(first >> 1) & 0x1f];
+            int count = (int)(((first & 1) << 8) | s[p++]) + 1;
+            c.kind = 1;
+            c.width = (uint8_t)width;
+            c.count = count;
+            c.src = dev_base + p;
+            p += ((int64_t)count * width + 7) / 8;
+            cnt += count;
+        } else if (enc == 2) {  // PATCHED_BASE
+            int width = fbs[(first >> 1) & 0x1f];
+            int count = (int)(((first & 1) << 8) | s[p++]) + 1;
+            uint8_t third = s[p++], fourth = s[p++];
+            int bw = ((third >> 5) & 7) + 1;
+            int pw = fbs[third & 0x1f];
+            int pgw = ((fourth >> 5) & 7) + 1;
+            int pl = fourth & 0x1f;
+            uint64_t braw = 0;
+            for (int i = 0; i < bw; i++) braw = (braw << 8) | s[p++];
+            uint64_t smask = 1ull << (bw * 8 - 1);
+            c.kind = 2;
+            c.width = (uint8_t)width;
+            c.count = count;
+            c.base = (braw & smask) ? -(int64_t)(braw & ~smask)
+                                    : (int64_t)braw;
+            c.src = dev_base + p;
+            p += ((int64_t)count * width + 7) / 8;
+            int cfb = 64;
+            for (int i = 0; i < 32; i++)
+                if (fbs[i] >= pw + pgw) { cfb = fbs[i]; break; }
+            c.patch_src = dev_base + p;
+            c.patch_pl = (uint16_t)pl;
+            c.patch_pw = (uint8_t)pw;
+            c.patch_pgw = (uint8_t)pgw;
+            c.patch_cfb = (uint8_t)cfb;
+            p += ((int64_t)pl * cfb + 7) / 8;
+            cnt += count;
+        } else {  // DELTA
+            int wcode = (first >> 1) & 0x1f;
+            int width = wcode == 0 ? 0 : fbs[wcode];
+            int count = (int)(((first & 1) << 8) | s[p++]) + 1;
+            bool ok = true;
+            uint64_t braw = uvarint(ok);
+            int64_t base = is_signed
+                               ? ((int64_t)(braw >> 1) ^ -(int64_t)(braw & 1))
+                               : (int64_t)braw;
+            uint64_t draw = uvarint(ok);
+            int64_t delta = (int64_t)(draw >> 1) ^ -(int64_t)(draw & 1);
+            if (!ok) {
+                set_error("RLEv2 delta varint overrun");
+                return false;
+            }
+            c.kind = 3;
+            c.width = (uint8_t)width;
+            c.count = count;
+            c.base = base;
+            c.delta = delta;
+            c.src = dev_base + p;
+            if (width > 0 && count > 2)
+                p += ((int64_t)(count - 2) * width + 7) / 8;
+            cnt += count;
+        }
+        out.push_back(c);
+        if (p > len) {
+            set_error("RLEv2 stream overran its length");
+            return false;
+        }
+    }
+    return true;
+}
+
+// ORC byte-RLE (tinyint columns, e.g. _VALUE_KIND): header h >= 0 -> h+3
+// copies of the next byte; h < 0 -> -h literal bytes.
+static bool prescan_byterle(const uint8_t *s, int64_t len, int64_t n_values,
+                            int out_esize, int dense_target,
+                            uint64_t dev_base, int64_t dense0,
+                            std::vector<Rlev2Chunk> &out) {
+    int64_t p = 0, cnt = 0;
+    while (cnt < n_values) {
+        if (p >= len) {
+            set_error("byte-RLE stream overrun");
+            return false;
+        }
+        int8_t h = (int8_t)s[p++];
+        Rlev2Chunk c{};
+        c.out_esize = (uint8_t)out_esize;
+        c.dense_target = (uint8_t)dense_target;
+        c.out_start = dense0 + cnt;
+        if (h >= 0) {
+            c.kind = 4;
+            c.base = (int8_t)s[p++];
+            c.count = (int32_t)std::min<int64_t>(h + 3, n_values - cnt);
+        } else {
+            c.kind = 5;
+            c.src = dev_base + p;
+            c.count = (int32_t)std::min<int64_t>(-(int64_t)h, n_values - cnt);
+            p += -(int64_t)h;
+        }
+        cnt += c.count;
+        out.push_back(c);
+    }
+    return true;
+}
+
+static const uint8_t BITREV[256] = {
+#define R2(n) n, n + 2 * 64, n + 1 * 64, n + 3 * 64
+#define R4(n) R2(n), R2(n + 2 * 16), R2(n + 1 * 16), R2(n + 3 * 16)
+#define R6(n) R4(n), R4(n + 2 * 4), R4(n + 1 * 4), R4(n + 3 * 4)
+    R6(0), R6(2), R6(1), R6(3)
+#undef R2
+#undef R4
+#undef R6
+};
+
+// ORC PRESENT stream (boolean RLE = byte-RLE over MSB-first bit-packed
+// bytes) -> the parquet-style def-level RleChunks consumed by
+// k_level_scatter (which reads LSB-first: bytes are bit-reversed into the
+// (run,col) levels buffer). Tracks the dense (non-null) cursor.
+static bool prescan_present(const uint8_t *s, int64_t len, int64_t n_rows,
+                            int64_t out_row0, RunCol &rc) {
+    int64_t p = 0, cnt = 0;
+    while (cnt < n_rows) {
+        if (p >= len) {
+            set_error("PRESENT stream overrun");
+            return false;
+        }
+        int8_t h = (int8_t)s[p++];
+        if (h >= 0) {
+            uint8_t v = s[p++];
+            int64_t vals = std::min<int64_t>((int64_t)(h + 3) * 8,
+                                             n_rows - cnt);
+            if (v == 0x00 || v == 0xFF) {
+                RleChunk c{};
+                c.kind = 0;
+                c.value = v ? 1 : 0;
+                c.out_start = out_row0 + cnt;
+                c.count = (int32_t)vals;
+                c.bit_width = 1;
+                c.aux = rc.dense_before;
+                rc.def_host.push_back(c);
+                if (v) rc.dense_before += vals;
+            } else {
+                RleChunk c{};
+                c.kind = 1;
+                c.src = (uint64_t)rc.levels_host.size();
+                c.out_start = out_row0 + cnt;
+                c.count = (int32_t)vals;
+                c.bit_width = 1;
+                c.aux = rc.dense_before;
+                for (int i = 0; i < h + 3; i++)
+                    rc.levels_host.push_back(BITREV[v]);
+                rc.def_host.push_back(c);
+                for (int64_t i = 0; i < vals; i += 8) {
+                    int64_t rem = vals - i;
+                    uint8_t mask = rem >= 8 ? 0xFF : (uint8_t)((1u << rem) - 1);
+                    rc.dense_before += __builtin_popcount(BITREV[v] & mask);
+                }
+            }
+            cnt += vals;
+        } else {
+            int lit = -(int)h;
+            int64_t vals = std::min<int64_t>((int64_t)lit * 8, n_rows - cnt);
+            RleChunk c{};
+            c.kind = 1;
+            c.src = (uint64_t)rc.levels_host.size();
+            c.out_start = out_row0 + cnt;
+            c.count = (int32_t)vals;
+            c.bit_width = 1;
+            c.aux = rc.dense_before;
+            for (int i = 0; i < lit; i++)
+                rc.levels_host.push_back(BITREV[s[p + i]]);
+            rc.def_host.push_back(c);
+            for (int64_t i = 0; i < vals; i++) {
+                uint8_t b = BITREV[s[p + (i >> 3)]];
+                rc.dense_before += (b >> (i & 7)) & 1;
+            }
+            p += lit;
+            cnt += vals;
+        }
+    }
+    return true;
+}
+
+// Stage one ORC file's stripes for the required columns.
+static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
+                           const StagedFile &sf, Run &run, int64_t row_base) {
+    const auto &cols = plan->cols;
+    const OrcFileMeta &om = sf.orc_meta;
+    std::vector<int> col_id(cols.size(), -1);
+    for (size_t c = 0; c < cols.size(); c++) {
+        for (size_t i = 0; i < om.column_names.size(); i++)
+            if (om.column_names[i] == cols[c].name) col_id[c] = (int)i + 1;
+        if (col_id[c] < 0) {
+            set_error("%s: column %s not found", fd.path.c_str(),
+                      cols[c].name.c_str());
+            return false;
+        }
+    }
+    int64_t stripe_row = 0;
+    for (const auto &st : om.stripes) {
+        for (size_t c = 0; c < cols.size(); c++) {
+            RunCol &rc = run.cols[c];
+            rc.orc_encoded = true;
+            int cid = col_id[c];
+            int ckind = om.column_kinds[cid - 1];
+            const int stored = cols[c].stored_esize;
+            const OrcStream *data = orc_find_stream(st, cid, ORC_STREAM_DATA);
+            const OrcStream *present =
+                orc_find_stream(st, cid, ORC_STREAM_PRESENT);
+            if (!data) {
+                set_error("%s col %s: DATA stream missing", fd.path.c_str(),
+                          cols[c].name.c_str());
+                return false;
+            }
+            if ((int)st.encodings.size() > cid &&
+                st.encodings[cid] != 0 && st.encodings[cid] != 2) {
+                set_error("%s col %s: ORC dictionary encodings not "
+                          "supported yet",
+                          fd.path.c_str(), cols[c].name.c_str());
+                return false;
+            }
+            // upload the encoded DATA stream
+            void *dev = plan->bufs.alloc(data->length);
+            if (!dev) return false;
+            if (hipMemcpy(dev, sf.data.data() + data->offset, data->length,
+                          hipMemcpyHostToDevice) != hipSuccess) {
+                set_error("H2D failed");
+                return false;
+            }
+            plan->encoded_bytes_total += data->length;
+            int64_t row0 = row_base + stripe_row;
+            int64_t n_dense = st.num_rows;
+            int64_t dense0 = row0;
+            int dense_target = 0;
+            if (present) {
+                rc.has_nulls = true;
+                int64_t before = rc.dense_before;
+                if (!prescan_present(sf.data.data() + present->offset,
+                                     present->length, st.num_rows, row0, rc))
+                    return false;
+                n_dense = rc.dense_before - before;
+                dense0 = before;
+                dense_target = 1;
+            }
+            bool ok;
+            if (ckind == ORC_BYTE) {
+                ok = prescan_byterle(sf.data.data() + data->offset,
+                                     data->length, n_dense, stored,
+                                     dense_target, (uint64_t)dev, dense0,
+                                     rc.rlev2_host);
+            } else if (ckind == ORC_SHORT || ckind == ORC_INT ||
+                       ckind == ORC_LONG) {
+                ok = prescan_rlev2(sf.data.data() + data->offset,
+                                   data->length, n_dense, 1, stored,
+                                   dense_target, (uint64_t)dev, dense0,
+                                   rc.rlev2_host);
+            } else {
+                set_error("%s col %s: ORC type kind %d not supported yet",
+                          fd.path.c_str(), cols[c].name.c_str(), ckind);
+                return false;
+            }
+            if (!ok) return false;
+        }
+        stripe_row += st.num_rows;
+    }
+    if (stripe_row != fd.row_count) {
+        set_error("%s: rowCount %lld != file rows %lld", fd.path.c_str(),
+                  (long long)fd.row_count, (long long)stripe_row);
+        return false;
+    }
     return true;
 }
 
@@ -514,6 +851,11 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
     for (const auto &fd : files) {
         StagedFile sf;
         if (!load_file(fd.path, sf)) return false;
+        if (sf.orc) {
+            if (!stage_orc_file(plan, fd, sf, run, row_base)) return false;
+            row_base += fd.row_count;
+            continue;
+        }
         std::vector<int> leaf(cols.size(), -1);
         for (size_t c = 0; c < cols.size(); c++) {
             for (size_t i = 0; i < sf.meta.schema_names.size(); i++)
@@ -734,9 +1076,24 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                           hipMemcpyHostToDevice) != hipSuccess)
                 return false;
         }
+        if (!rc.rlev2_host.empty()) {
+            rc.rlev2_dev = (Rlev2Chunk *)plan->bufs.alloc(
+                rc.rlev2_host.size() * sizeof(Rlev2Chunk));
+            if (!rc.rlev2_dev) return false;
+            if (hipMemcpy(rc.rlev2_dev, rc.rlev2_host.data(),
+                          rc.rlev2_host.size() * sizeof(Rlev2Chunk),
+                          hipMemcpyHostToDevice) != hipSuccess)
+                return false;
+        }
         if (rc.has_nulls) {
+            // parquet: dense values packed host-side (dense_host);
+            // ORC: dense values are produced on-device by k_rlev2
+            size_t dense_bytes = rc.orc_encoded
+                                     ? (size_t)rc.dense_before *
+                                           plan->cols[c].stored_esize
+                                     : rc.dense_host.size();
             rc.valid_dev = (uint8_t *)plan->bufs.alloc(run.length);
-            rc.dense_dev = plan->bufs.alloc(rc.dense_host.size());
+            rc.dense_dev = plan->bufs.alloc(dense_bytes);
             void *levels_dev = plan->bufs.alloc(rc.levels_host.size());
             rc.def_dev = (RleChunk *)plan->bufs.alloc(rc.def_host.size() *
                                                       sizeof(RleChunk));
@@ -745,10 +1102,12 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
             // rows in non-null chunks of this column keep validity 1
             if (hipMemset(rc.valid_dev, 1, run.length) != hipSuccess)
                 return false;
-            if (hipMemcpy(rc.dense_dev, rc.dense_host.data(),
+            if (!rc.dense_host.empty() &&
+                hipMemcpy(rc.dense_dev, rc.dense_host.data(),
                           rc.dense_host.size(),
-                          hipMemcpyHostToDevice) != hipSuccess ||
-                hipMemcpy(levels_dev, rc.levels_host.data(),
+                          hipMemcpyHostToDevice) != hipSuccess)
+                return false;
+            if (hipMemcpy(levels_dev, rc.levels_host.data(),
                           rc.levels_host.size(),
                           hipMemcpyHostToDevice) != hipSuccess)
                 return false;
@@ -826,7 +1185,8 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
     for (auto &run : sec.runs)
         for (auto &rc : run.cols) {
             sec.any_dict |= rc.dict_encoded;
-            sec.any_decode |= rc.dict_encoded || rc.has_nulls;
+            sec.any_decode |=
+                rc.dict_encoded || rc.has_nulls || !rc.rlev2_host.empty();
         }
     return sec.key_cols && sec.seq_cols && sec.kind_cols && sec.all_cols &&
            sec.lens_dev && sec.cuts && sec.winners && sec.tile_counts &&
@@ -1069,6 +1429,12 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                             (uint8_t *)rc.contig + gt.start * es, es, st);
                         if (e != hipSuccess) return fail("dict_gather", e);
                     }
+                }
+                if (!rc.rlev2_host.empty()) {
+                    hipError_t e = pmh_launch_rlev2(
+                        rc.rlev2_dev, (int64_t)rc.rlev2_host.size(),
+                        rc.contig, rc.dense_dev, st);
+                    if (e != hipSuccess) return fail("rlev2", e);
                 }
                 if (rc.has_nulls) {
                     hipError_t e = pmh_launch_level_scatter(

@@ -133,25 +133,30 @@ def run_to_arrow(run, value_names=None):
 
 
 def write_runs(runs, out_dir, compression="NONE", row_group_rows=1 << 20,
-               data_page_rows=1 << 16):
-    """Write runs as Parquet data files; returns list of per-file metadata
-    dicts shaped like DataFileMeta (io/DataFileMeta.java:124-190): fileName,
-    rowCount, minKey, maxKey, minSequenceNumber, maxSequenceNumber, level."""
+               data_page_rows=1 << 16, file_format="parquet"):
+    """Write runs as Parquet or ORC data files; returns list of per-file
+    metadata dicts shaped like DataFileMeta (io/DataFileMeta.java:124-190):
+    fileName, rowCount, minKey, maxKey, minSequenceNumber,
+    maxSequenceNumber, level."""
     os.makedirs(out_dir, exist_ok=True)
     metas = []
     for r, run in enumerate(runs):
         tbl = run_to_arrow(run)
-        path = os.path.join(out_dir, f"run-{r}.parquet")
-        pq.write_table(
-            tbl, path,
-            compression=None if compression == "NONE" else compression,
-            use_dictionary=False,
-            data_page_version="1.0",
-            write_statistics=False,
-            row_group_size=row_group_rows,
-            data_page_size=data_page_rows * 8,
-            store_schema=False,
-        )
+        path = os.path.join(out_dir, f"run-{r}.{file_format}")
+        if file_format == "orc":
+            from pyarrow import orc as pa_orc
+            pa_orc.write_table(tbl, path, compression="uncompressed")
+        else:
+            pq.write_table(
+                tbl, path,
+                compression=None if compression == "NONE" else compression,
+                use_dictionary=False,
+                data_page_version="1.0",
+                write_statistics=False,
+                row_group_size=row_group_rows,
+                data_page_size=data_page_rows * 8,
+                store_schema=False,
+            )
         metas.append({
             "path": path,
             "rowCount": int(len(run["key"])),
