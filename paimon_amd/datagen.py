@@ -92,15 +92,19 @@ def gen_runs_partial_update(n_runs: int, rows_per_run: int, n_value_cols: int = 
     seqs_all = rng.permutation(total).astype(np.int64)
     runs = []
     for r in range(n_runs):
-        keys = rng.choice(space, size=rows_per_run, replace=False).astype(np.int64)
-        keys.sort()
+        keys = _sorted_unique_keys(rng, rows_per_run, space)
         seq = seqs_all[r * rows_per_run:(r + 1) * rows_per_run]
         kind = np.full(rows_per_run, KIND_INSERT, dtype=np.int8)
         is_update = rng.random(rows_per_run) < update_frac
         values = [keys.copy()]  # pk col, never null
         masks = [np.ones(rows_per_run, dtype=bool)]
-        # choose update column subsets: for update rows, update_cols random cols set
-        col_set = rng.random((rows_per_run, n_value_cols)).argsort(axis=1) < update_cols
+        # update rows set a random update_cols-subset of columns: per-row
+        # random threshold rank (argpartition-free approximation: each col
+        # kept with the row's top-k cut via random keys)
+        uc = min(update_cols, n_value_cols)
+        rk = rng.random((rows_per_run, n_value_cols))
+        cut = np.partition(rk, uc - 1, axis=1)[:, uc - 1:uc]
+        col_set = rk <= cut
         for c in range(n_value_cols):
             vals = rng.integers(-2**31, 2**31, size=rows_per_run,
                                 dtype=np.int64).astype(np.int32)
