@@ -190,7 +190,7 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
 struct TileSmem {
     int64_t skey[PMH_TILE_MAX];
     int64_t sseq[PMH_TILE_MAX];
-    uint16_t perm[PMH_TILE_MAX];
+    uint16_t perm[2][PMH_TILE_MAX];
     uint16_t win[PMH_TILE_MAX];
     uint8_t skind[PMH_TILE_MAX];
     uint8_t head[PMH_TILE_MAX];    // group head flags (merged order)
@@ -204,6 +204,23 @@ struct TileSmem {
     int32_t mreal;     // real element count (rank width of this tile)
     int32_t nemit;
 };
+
+// stable 2-way co-rank: number of A elements among the first d outputs of
+// merge(A, B), ties take A first. A/B are perm-index sequences; key lookup
+// through skey[.]. For i in [max(0,d-lb), min(d,la)), j = d-1-i is always
+// in [0, lb). Advance while A[i] <= B[j] (A[i] belongs in the first d).
+DEV int32_t corank(int64_t d, const uint16_t *pa, int32_t la,
+                   const uint16_t *pb, int32_t lb, const int64_t *skey) {
+    int64_t ilo = d > lb ? d - lb : 0;
+    int64_t ihi = d < la ? d : la;
+    while (ilo < ihi) {
+        int64_t i = ilo + ((ihi - ilo) >> 1);
+        int64_t j = d - 1 - i;
+        if (skey[pa[i]] <= skey[pb[j]]) ilo = i + 1;
+        else ihi = i;
+    }
+    return (int32_t)ilo;
+}
 
 DEV bool kind_is_add(uint8_t k) { return k == 0 || k == 2; }
 
@@ -233,6 +250,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    uint16_t *group_start, uint32_t *err_flag) {
     const bool drop_delete = flags & 1;
     const bool ignore_delete = flags & 2;
+    constexpr bool pu_mode = PU;
     // ablation levels (profiling only, flags bits 8..): 1=stage,2=+merge,
     // 3=+scan, 0/absent=full. Partial levels publish a checksum so the
     // compiler cannot dead-code the ablated phases' inputs.
@@ -307,6 +325,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                 sm.skey[off + i] = kaddr[i];
                 sm.sseq[off + i] = saddr[i];
                 sm.skind[off + i] = (uint8_t)daddr[i];
+                sm.perm[0][off + i] = (uint16_t)(off + i);
             }
         }
         __syncthreads();
@@ -321,181 +340,56 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             continue;
         }
 
-        // --- direct k-way register merge: each thread selects its output
-        // rank's cuts across the k LDS segments (same key-domain bisection
-        // as k_partition, with lockstep probes), then serially merges CH
-        // outputs with the candidate keys held in registers. One phase, no
-        // level barriers, no perm indirection (replaced the 3-level
-        // pairwise merge: ~40% less LDS-latency chain).
-        {
-            constexpr int CH = 8;
-            const int n_chunks = (M + CH - 1) / CH;
-            for (int ch = tid; ch < n_chunks; ch += (int)blockDim.x) {
-                const int32_t d0 = ch * CH;
-                int32_t cur[PMH_MAX_RUNS];
-                int32_t seg_end[PMH_MAX_RUNS];
-#pragma unroll
-                for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                    if (r >= k) continue;
-                    cur[r] = 0;
-                    seg_end[r] = sm.seglen[r];
-                }
-                if (d0 > 0) {
-                    // bisect (key, run) for rank d0 within the LDS segments
-                    uint64_t klo = ~0ull, khi = 0;
-#pragma unroll
-                    for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                        if (r >= k || sm.seglen[r] == 0) continue;
-                        uint64_t lo_k = ukey(sm.skey[sm.segoff[r]]);
-                        uint64_t hi_k = ukey(
-                            sm.skey[sm.segoff[r] + sm.seglen[r] - 1]);
-                        if (lo_k < klo) klo = lo_k;
-                        if (hi_k > khi) khi = hi_k;
+        // --- pairwise stable merge, ceil(log2(k)) levels
+        int cur = 0;
+        for (int width = 1; width < k; width <<= 1) {
+            // sequences at this level: [segoff[q*width], segoff[min((q+1)*width,k)])
+            // merge pairs (2q, 2q+1)
+            const int nxt = cur ^ 1;
+            // per-thread chunks of 8 outputs, grid-stride over all chunks
+            const int CH = 8;
+            int n_chunks = (M + CH - 1) / CH;
+            for (int ch = tid; ch < n_chunks; ch += blockDim.x) {
+                int64_t o = (int64_t)ch * CH;  // global output rank
+                int remaining = (int)(M - o < CH ? M - o : CH);
+                int p = 0;
+                // a chunk may span several pairs (pairs can be tiny):
+                // walk pairs until the chunk's outputs are all produced
+                while (remaining > 0) {
+                    // pair p covers output ranks [segoff[a0], segoff[b1])
+                    while ((p + 1) * 2 * width < k &&
+                           sm.segoff[(p + 1) * 2 * width] <= o)
+                        p++;
+                    int a0 = p * 2 * width;
+                    int amid = a0 + width < k ? a0 + width : k;
+                    int b1 = a0 + 2 * width < k ? a0 + 2 * width : k;
+                    int32_t abase = sm.segoff[a0];
+                    int32_t la = sm.segoff[amid] - abase;
+                    int32_t lb = sm.segoff[b1] - sm.segoff[amid];
+                    int64_t d = o - abase;  // rank within pair
+                    int32_t lim = la + lb - (int32_t)d;
+                    if (lim <= 0) break;  // past the last pair
+                    int n_out = lim < remaining ? lim : remaining;
+                    const uint16_t *pa = &sm.perm[cur][abase];
+                    const uint16_t *pb = &sm.perm[cur][abase + la];
+                    int32_t ai = corank(d, pa, la, pb, lb, sm.skey);
+                    int32_t bi = (int32_t)d - ai;
+                    uint16_t *out = &sm.perm[nxt][abase + d];
+                    for (int x = 0; x < n_out; x++) {
+                        bool takeA;
+                        if (ai >= la) takeA = false;
+                        else if (bi >= lb) takeA = true;
+                        else takeA = !(sm.skey[pa[ai]] > sm.skey[pb[bi]]);
+                        out[x] = takeA ? pa[ai++] : pb[bi++];
                     }
-                    int32_t wlo[PMH_MAX_RUNS], whi[PMH_MAX_RUNS];
-#pragma unroll
-                    for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                        if (r >= k) continue;
-                        wlo[r] = 0;
-                        whi[r] = sm.seglen[r];
-                    }
-                    while (klo < khi) {
-                        uint64_t mid = klo + ((khi - klo) >> 1);
-                        // lockstep windowed upper-bounds across segments
-                        int32_t lo[PMH_MAX_RUNS], hi[PMH_MAX_RUNS];
-#pragma unroll
-                        for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                            if (r >= k) continue;
-                            lo[r] = wlo[r];
-                            hi[r] = whi[r];
-                        }
-                        bool any = true;
-                        while (any) {
-                            any = false;
-#pragma unroll
-                            for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                                if (r >= k) continue;
-                                if (lo[r] < hi[r]) {
-                                    int32_t m =
-                                        lo[r] + ((hi[r] - lo[r]) >> 1);
-                                    if (ukey(sm.skey[sm.segoff[r] + m]) <=
-                                        mid)
-                                        lo[r] = m + 1;
-                                    else
-                                        hi[r] = m;
-                                    any |= lo[r] < hi[r];
-                                }
-                            }
-                        }
-                        int32_t cnt = 0;
-#pragma unroll
-                        for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                            if (r >= k) continue;
-                            cnt += lo[r];
-                        }
-                        if (cnt >= d0) {
-                            khi = mid;
-#pragma unroll
-                            for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                                if (r >= k) continue;
-                                whi[r] = lo[r];
-                            }
-                        } else {
-                            klo = mid + 1;
-#pragma unroll
-                            for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                                if (r >= k) continue;
-                                wlo[r] = lo[r];
-                            }
-                        }
-                    }
-                    // v* = klo; lower bounds within the final windows, then
-                    // run-order tie allocation (as in k_partition)
-                    int32_t lb[PMH_MAX_RUNS];
-                    {
-                        int32_t lo[PMH_MAX_RUNS], hi[PMH_MAX_RUNS];
-#pragma unroll
-                        for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                            if (r >= k) continue;
-                            lo[r] = wlo[r];
-                            hi[r] = whi[r];
-                        }
-                        bool any = true;
-                        while (any) {
-                            any = false;
-#pragma unroll
-                            for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                                if (r >= k) continue;
-                                if (lo[r] < hi[r]) {
-                                    int32_t m =
-                                        lo[r] + ((hi[r] - lo[r]) >> 1);
-                                    if (ukey(sm.skey[sm.segoff[r] + m]) < klo)
-                                        lo[r] = m + 1;
-                                    else
-                                        hi[r] = m;
-                                    any |= lo[r] < hi[r];
-                                }
-                            }
-                        }
-#pragma unroll
-                        for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                            if (r >= k) continue;
-                            lb[r] = lo[r];
-                        }
-                    }
-                    int32_t base = 0;
-#pragma unroll
-                    for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                        if (r >= k) continue;
-                        base += lb[r];
-                    }
-                    int32_t t = d0 - base;
-#pragma unroll
-                    for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                        if (r >= k) continue;
-                        int32_t c = lb[r];
-                        bool has =
-                            (c < sm.seglen[r]) &&
-                            (ukey(sm.skey[sm.segoff[r] + c]) == klo);
-                        if (t > 0 && has) {
-                            c++;
-                            t--;
-                        }
-                        cur[r] = c;
-                    }
-                }
-                // serial CH-way register merge from the cuts
-                int64_t ck[PMH_MAX_RUNS];
-#pragma unroll
-                for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                    if (r >= k) continue;
-                    ck[r] = cur[r] < seg_end[r]
-                                ? sm.skey[sm.segoff[r] + cur[r]]
-                                : 0;
-                }
-                const int n_out = M - d0 < CH ? M - d0 : CH;
-                for (int x = 0; x < n_out; x++) {
-                    int best = -1;
-                    int64_t bkey = 0;
-#pragma unroll
-                    for (int r = 0; r < PMH_MAX_RUNS; r++) {
-                        if (r >= k) continue;
-                        if (cur[r] < seg_end[r] &&
-                            (best < 0 || ck[r] < bkey)) {
-                            best = r;
-                            bkey = ck[r];
-                        }
-                    }
-                    sm.perm[d0 + x] =
-                        (uint16_t)(sm.segoff[best] + cur[best]);
-                    cur[best]++;
-                    if (cur[best] < seg_end[best])
-                        ck[best] = sm.skey[sm.segoff[best] + cur[best]];
+                    o += n_out;
+                    remaining -= n_out;
                 }
             }
+            cur = nxt;
             __syncthreads();
         }
-        const uint16_t *mo = sm.perm;
+        const uint16_t *mo = sm.perm[cur];
         if (ablate == 2) {
             if (tid == 0)
                 tile_counts[tile] = ((int32_t)mo[M - 1] ^ (int32_t)mo[0]) & 1;
