@@ -234,3 +234,79 @@ class TestMultiFileRuns:
                 got = {kk: np.concatenate(v) for kk, v in got.items()}
         assert (got["_KEY_k"] == exp_key).all()
         assert (got["_SEQUENCE_NUMBER"] == exp_seq).all()
+
+
+class TestScaleAndTypes:
+    def test_size_representative_8x1M(self, tmp_path):
+        # 1/10th of the C2 shape, bit-exact against the oracle
+        runs = gen_runs_dedup(8, 1_000_000, n_value_cols=4, seed=1042,
+                              delete_frac=0.05)
+        _run_and_compare(tmp_path, runs)
+
+    def test_float_double_columns(self, tmp_path):
+        import pyarrow as pa
+        import pyarrow.parquet as pq
+        rng = np.random.default_rng(53)
+        n = 30_000
+        runs = []
+        metas = []
+        seqs = rng.permutation(3 * n).astype(np.int64)
+        for r in range(3):
+            keys = np.sort(rng.choice(3 * n, n, replace=False)).astype(np.int64)
+            f32 = rng.standard_normal(n).astype(np.float32)
+            f64 = rng.standard_normal(n)
+            runs.append({"key": keys, "seq": seqs[r * n:(r + 1) * n],
+                         "kind": np.zeros(n, np.int8),
+                         "values": [keys.copy(), f32, f64]})
+            fields = [pa.field("_KEY_k", pa.int64(), nullable=False),
+                      pa.field("_SEQUENCE_NUMBER", pa.int64(), nullable=False),
+                      pa.field("_VALUE_KIND", pa.int8(), nullable=False),
+                      pa.field("v_k", pa.int64()),
+                      pa.field("f", pa.float32()),
+                      pa.field("d", pa.float64())]
+            tbl = pa.Table.from_arrays(
+                [pa.array(keys), pa.array(runs[r]["seq"]),
+                 pa.array(runs[r]["kind"]), pa.array(keys), pa.array(f32),
+                 pa.array(f64)], schema=pa.schema(fields))
+            path = str(tmp_path / f"run-{r}.parquet")
+            pq.write_table(tbl, path, compression=None, use_dictionary=False,
+                           data_page_version="1.0", store_schema=False)
+            metas.append({"path": path, "rowCount": n,
+                          "minKey": int(keys[0]), "maxKey": int(keys[-1]),
+                          "level": 0})
+        r_, w_ = merge_dedup(runs)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               [{"name": "v_k", "type": "int64"},
+                                {"name": "f", "type": "float32"},
+                                {"name": "d", "type": "float64"}]) as plan:
+                got = plan.read_next()
+        # floats pass through the gather bit-exactly
+        ef = np.array([runs[a]["values"][1][b] for a, b in zip(r_, w_)])
+        ed = np.array([runs[a]["values"][2][b] for a, b in zip(r_, w_)])
+        assert got["f"].dtype == np.float32
+        assert (got["f"].view(np.int32) == ef.view(np.int32)).all()
+        assert (got["d"].view(np.int64) == ed.view(np.int64)).all()
+
+    def test_multi_row_group_files(self, tmp_path):
+        runs = gen_runs_dedup(4, 50_000, n_value_cols=3, seed=54)
+        metas = write_runs(runs, str(tmp_path), compression="NONE",
+                           row_group_rows=8_192)
+        exp = _expected_dedup(runs)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(3)) as plan:
+                got = _read_all_batches(plan)
+        assert (got["_KEY_k"] == exp["_KEY_k"]).all()
+        assert (got["v_c0"] == exp["v_c0"]).all()
+
+
+def _read_all_batches(plan):
+    got = {}
+    while True:
+        b = plan.read_next()
+        if b is None:
+            break
+        for k, v in b.items():
+            got.setdefault(k, []).append(v.copy())
+    return {k: np.concatenate(v) for k, v in got.items()}
