@@ -14,7 +14,7 @@ namespace pmh {
 //  - rows per run < 2^28 (winner packing run:4 | row:28).
 constexpr int PMH_MAX_RUNS = 16;
 constexpr int PMH_TILE_THREADS = 256;
-constexpr int64_t PMH_TILE_ROWS = 2048;
+constexpr int64_t PMH_TILE_ROWS = 1792;
 constexpr int PMH_TILE_MAX = PMH_TILE_ROWS + PMH_MAX_RUNS;
 constexpr int PMH_TILE_ITER =
     (PMH_TILE_MAX + PMH_TILE_THREADS - 1) / PMH_TILE_THREADS;

@@ -50,12 +50,12 @@ DEV uint64_t ukey(int64_t k) { return (uint64_t)k ^ 0x8000000000000000ull; }
 // latency-parked on serial probes, profiles/r01_c2_pmc.md). Arrays are
 // indexed only by the unrolled compile-time r, so they stay in registers.
 // le=true: first index with ukey > v; le=false: first index with ukey >= v.
-template <bool LE>
+template <bool LE, int KM>
 DEV void bound_multi(const uint64_t *addr, const int64_t *lo_in,
                      const int64_t *hi_in, int k, uint64_t v, int64_t *out) {
-    int64_t lo[PMH_MAX_RUNS], hi[PMH_MAX_RUNS];
+    int64_t lo[KM], hi[KM];
 #pragma unroll
-    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+    for (int r = 0; r < KM; r++) {
         if (r >= k) continue;
         lo[r] = lo_in[r];
         hi[r] = hi_in[r];
@@ -64,7 +64,7 @@ DEV void bound_multi(const uint64_t *addr, const int64_t *lo_in,
     while (any) {
         any = false;
 #pragma unroll
-        for (int r = 0; r < PMH_MAX_RUNS; r++) {
+        for (int r = 0; r < KM; r++) {
             if (r >= k) continue;
             if (lo[r] < hi[r]) {
                 int64_t mid = lo[r] + ((hi[r] - lo[r]) >> 1);
@@ -78,7 +78,7 @@ DEV void bound_multi(const uint64_t *addr, const int64_t *lo_in,
         }
     }
 #pragma unroll
-    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+    for (int r = 0; r < KM; r++) {
         if (r >= k) continue;
         out[r] = lo[r];
     }
@@ -90,6 +90,8 @@ DEV void bound_multi(const uint64_t *addr, const int64_t *lo_in,
 // sum c_r = D and the (key, run) total order is two-sided partitioned
 // (GPU merge-path generalized to k runs via key-domain bisection with
 // per-run shrinking windows).
+template <int KM>  // compile-time run-count bound: keeps the per-thread
+                   // window arrays in registers (KM=16 spilled 80 VGPRs)
 __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
                             int64_t tile_rows, int64_t n_bounds,
                             int64_t total_rows, int32_t *cuts /* n_bounds*k */) {
@@ -98,10 +100,10 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
     int64_t D = b * tile_rows;
     if (D > total_rows) D = total_rows;
 
-    uint64_t addr[PMH_MAX_RUNS];
-    int64_t len[PMH_MAX_RUNS], wlo[PMH_MAX_RUNS], whi[PMH_MAX_RUNS];
+    uint64_t addr[KM];
+    int64_t len[KM], wlo[KM], whi[KM];
 #pragma unroll
-    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+    for (int r = 0; r < KM; r++) {
         if (r >= k) continue;
         addr[r] = keys[r].addr0;  // staged columns are contiguous
         len[r] = lens[r];
@@ -110,7 +112,7 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
     }
     if (D == 0 || D >= total_rows) {
 #pragma unroll
-        for (int r = 0; r < PMH_MAX_RUNS; r++) {
+        for (int r = 0; r < KM; r++) {
             if (r >= k) continue;
             cuts[b * k + r] = D == 0 ? 0 : (int32_t)len[r];
         }
@@ -121,7 +123,7 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
     // actual min/max keys (v* is an existing key; cnt_le(max) = total >= D).
     uint64_t klo = ~0ull, khi = 0;
 #pragma unroll
-    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+    for (int r = 0; r < KM; r++) {
         if (r >= k || len[r] == 0) continue;
         uint64_t lo_k = ukey(*reinterpret_cast<const int64_t *>(addr[r]));
         uint64_t hi_k = ukey(
@@ -129,27 +131,27 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
         if (lo_k < klo) klo = lo_k;
         if (hi_k > khi) khi = hi_k;
     }
-    int64_t pos[PMH_MAX_RUNS];
+    int64_t pos[KM];
     while (klo < khi) {
         uint64_t mid = klo + ((khi - klo) >> 1);
-        bound_multi<true>(addr, wlo, whi, k, mid, pos);
+        bound_multi<true, KM>(addr, wlo, whi, k, mid, pos);
         int64_t cnt = 0;
 #pragma unroll
-        for (int r = 0; r < PMH_MAX_RUNS; r++) {
+        for (int r = 0; r < KM; r++) {
             if (r >= k) continue;
             cnt += pos[r];
         }
         if (cnt >= D) {
             khi = mid;
 #pragma unroll
-            for (int r = 0; r < PMH_MAX_RUNS; r++) {
+            for (int r = 0; r < KM; r++) {
                 if (r >= k) continue;
                 whi[r] = pos[r];
             }
         } else {
             klo = mid + 1;
 #pragma unroll
-            for (int r = 0; r < PMH_MAX_RUNS; r++) {
+            for (int r = 0; r < KM; r++) {
                 if (r >= k) continue;
                 wlo[r] = pos[r];
             }
@@ -159,16 +161,16 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
     // (<=1 per run), take the first t in run order. Searching within the
     // final windows yields absolute positions: everything below wlo has
     // key < v*, everything at/above whi has key > v*.
-    bound_multi<false>(addr, wlo, whi, k, klo, pos);
+    bound_multi<false, KM>(addr, wlo, whi, k, klo, pos);
     int64_t base = 0;
 #pragma unroll
-    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+    for (int r = 0; r < KM; r++) {
         if (r >= k) continue;
         base += pos[r];
     }
     int64_t t = D - base;
 #pragma unroll
-    for (int r = 0; r < PMH_MAX_RUNS; r++) {
+    for (int r = 0; r < KM; r++) {
         if (r >= k) continue;
         int64_t c = pos[r];
         bool has = (c < len[r]) &&
@@ -189,12 +191,15 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
 
 struct TileSmem {
     int64_t skey[PMH_TILE_MAX];
-    int64_t sseq[PMH_TILE_MAX];
+    int64_t sseq[PMH_TILE_MAX];  // packed (sequenceNumber << 1) | isAdd:
+                                 // ascending == ascending (seq, isAdd) —
+                                 // the merge-order tie-break of
+                                 // SortMergeReaderWithLoserTree.java:53-63.
+                                 // Requires |seq| < 2^62 (Paimon sequence
+                                 // numbers are non-negative counters).
     uint16_t perm[2][PMH_TILE_MAX];
-    uint16_t win[PMH_TILE_MAX];
-    uint8_t skind[PMH_TILE_MAX];
     uint8_t head[PMH_TILE_MAX];    // group head flags (merged order)
-    uint8_t fdone[PMH_TILE_MAX];   // segmented-scan carry flags
+    int32_t wave_tot[2 * (PMH_TILE_THREADS / 64)];
     int32_t segoff[PMH_MAX_RUNS + 1];
     int32_t seglen[PMH_MAX_RUNS];  // extended (incl. extra) lengths
     int32_t paircnt[PMH_MAX_RUNS + 1];
@@ -224,16 +229,6 @@ DEV int32_t corank(int64_t d, const uint16_t *pa, int32_t la,
 
 DEV bool kind_is_add(uint8_t k) { return k == 0 || k == 2; }
 
-// winner preference order for dedup: later (seq, isAdd) wins; eligibility
-// (ignore-delete) dominates.
-DEV bool better(int64_t seq_a, uint8_t kind_a, bool elig_a, int64_t seq_b,
-                uint8_t kind_b, bool elig_b) {
-    // true if A strictly preferred over B as the group's result
-    if (elig_a != elig_b) return elig_a;
-    if (seq_a != seq_b) return seq_a > seq_b;
-    return kind_is_add(kind_a) && !kind_is_add(kind_b);
-}
-
 // PU=true: PartialUpdate mode — no winner reduction; emit every owned
 // group's member list (ascending (seq, isAdd) order within the group) for
 // the per-field overlay in k_emit_pu. v1 accepts INSERT-only streams (the
@@ -250,7 +245,6 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    uint16_t *group_start, uint32_t *err_flag) {
     const bool drop_delete = flags & 1;
     const bool ignore_delete = flags & 2;
-    constexpr bool pu_mode = PU;
     // ablation levels (profiling only, flags bits 8..): 1=stage,2=+merge,
     // 3=+scan, 0/absent=full. Partial levels publish a checksum so the
     // compiler cannot dead-code the ablated phases' inputs.
@@ -323,8 +317,9 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                 reinterpret_cast<const int32_t *>(kinds[r].addr0) + base;
             for (int32_t i = tid; i < len; i += blockDim.x) {
                 sm.skey[off + i] = kaddr[i];
-                sm.sseq[off + i] = saddr[i];
-                sm.skind[off + i] = (uint8_t)daddr[i];
+                int32_t kd = daddr[i];
+                sm.sseq[off + i] =
+                    (saddr[i] << 1) | (int64_t)(kd == 0 || kd == 2);
                 sm.perm[0][off + i] = (uint16_t)(off + i);
             }
         }
@@ -333,7 +328,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             // keep staged data live; counts stay in {0,1} so the downstream
             // scan/emit of an ablated (profiling-only) run never goes OOB
             if (tid == 0) {
-                int64_t x = sm.skey[M - 1] ^ sm.sseq[M - 1] ^ sm.skind[0];
+                int64_t x = sm.skey[M - 1] ^ sm.sseq[M - 1];
                 tile_counts[tile] = (int32_t)((x ^ (x >> 32)) & 1);
             }
             __syncthreads();
@@ -405,27 +400,28 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             // heads here (that tile consumed them as its extras)
             if (sm.haspred && kk == sm.predkey) h = 0;
             sm.head[i] = h;
-            sm.fdone[i] = h;
-            sm.win[i] = (uint16_t)i;
         }
         __syncthreads();
 
+        // block-wide exclusive scan (wave shuffle scans + one barrier):
+        // the former Hillis-Steele block scans cost 16 __syncthreads per
+        // tile and dominated the kernel's wave-parked time.
+        const int lane = tid & 63;
+        const int wv = tid >> 6;
+        constexpr int NW = PMH_TILE_THREADS / 64;
+
         if constexpr (PU) {
             // --- PartialUpdate: emit owned groups' member lists
-            __shared__ int32_t s_scan[PMH_TILE_THREADS];
-            __shared__ int32_t s_scan2[PMH_TILE_THREADS];
             const int32_t per =
                 (Mreal + (int32_t)blockDim.x - 1) / blockDim.x;
             int32_t my_lo = tid * per;
             int32_t my_hi = my_lo + per < Mreal ? my_lo + per : Mreal;
             uint32_t *mout = &winners[tile * (tile_rows + PMH_MAX_RUNS)];
             uint16_t *gout = &group_start[tile * (tile_rows + 1)];
-            int32_t total_g = 0, total_m = 0;
             bool bad_kind = false;
+            int32_t g_off = 0, m_off = 0, total_g = 0, total_m = 0;
             for (int pass = 0; pass < 2; pass++) {
                 int32_t ng = 0, nm = 0;
-                int32_t g_off = pass == 1 ? s_scan[tid] : 0;
-                int32_t m_off = pass == 1 ? s_scan2[tid] : 0;
                 for (int32_t i = my_lo; i < my_hi; i++) {
                     if (!sm.head[i]) continue;
                     int32_t tail = i;
@@ -434,21 +430,15 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                         gout[g_off + ng] = (uint16_t)(m_off + nm);
                         // merged order within a group is (key, run); the
                         // overlay consumes records in ascending (seq, isAdd)
-                        // add order (SortMergeReaderWithLoserTree.java:53-63)
-                        // — insertion-sort the <= k members by seq.
+                        // order — insertion-sort the <= k members by the
+                        // packed sseq.
                         uint16_t gm[PMH_MAX_RUNS];
                         int gn = tail - i + 1;
                         for (int x = 0; x < gn; x++) {
                             uint16_t s = mo[i + x];
                             int y = x;
-                            while (y > 0) {
-                                uint16_t prv = gm[y - 1];
-                                int64_t s1 = sm.sseq[prv], s2 = sm.sseq[s];
-                                bool gt = s1 > s2 ||
-                                          (s1 == s2 &&
-                                           kind_is_add(sm.skind[prv]) &&
-                                           !kind_is_add(sm.skind[s]));
-                                if (!gt) break;
+                            while (y > 0 &&
+                                   sm.sseq[gm[y - 1]] > sm.sseq[s]) {
                                 gm[y] = gm[y - 1];
                                 y--;
                             }
@@ -456,7 +446,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                         }
                         for (int x = 0; x < gn; x++) {
                             uint16_t s = gm[x];
-                            if (!kind_is_add(sm.skind[s])) bad_kind = true;
+                            if (!(sm.sseq[s] & 1)) bad_kind = true;
                             int r = 0;
                             while (r + 1 <= k - 1 &&
                                    sm.segoff[r + 1] <= (int32_t)s)
@@ -470,25 +460,35 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                     nm += tail - i + 1;
                 }
                 if (pass == 0) {
-                    s_scan[tid] = ng;
-                    s_scan2[tid] = nm;
-                    __syncthreads();
-                    for (int d = 1; d < (int)blockDim.x; d <<= 1) {
-                        int32_t v = s_scan[tid], v2 = s_scan2[tid];
-                        int32_t a = tid >= d ? s_scan[tid - d] : 0;
-                        int32_t a2 = tid >= d ? s_scan2[tid - d] : 0;
-                        __syncthreads();
-                        s_scan[tid] = v + a;
-                        s_scan2[tid] = v2 + a2;
-                        __syncthreads();
+                    // dual wave scan of (ng, nm)
+                    int32_t ig = ng, im = nm;
+                    for (int off = 1; off < 64; off <<= 1) {
+                        int32_t ug = __shfl_up(ig, off, 64);
+                        int32_t um = __shfl_up(im, off, 64);
+                        if (lane >= off) {
+                            ig += ug;
+                            im += um;
+                        }
                     }
-                    total_g = s_scan[blockDim.x - 1];
-                    total_m = s_scan2[blockDim.x - 1];
-                    int32_t incl = s_scan[tid], incl2 = s_scan2[tid];
+                    if (lane == 63) {
+                        sm.wave_tot[wv] = ig;
+                        sm.wave_tot[NW + wv] = im;
+                    }
                     __syncthreads();
-                    s_scan[tid] = incl - ng;
-                    s_scan2[tid] = incl2 - nm;
-                    __syncthreads();
+                    int32_t addg = 0, addm = 0;
+                    total_g = 0;
+                    total_m = 0;
+#pragma unroll
+                    for (int w = 0; w < NW; w++) {
+                        if (w < wv) {
+                            addg += sm.wave_tot[w];
+                            addm += sm.wave_tot[NW + w];
+                        }
+                        total_g += sm.wave_tot[w];
+                        total_m += sm.wave_tot[NW + w];
+                    }
+                    g_off = addg + ig - ng;
+                    m_off = addm + im - nm;
                 }
             }
             if (bad_kind && err_flag) atomicOr(err_flag, 1u);
@@ -500,96 +500,74 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             continue;
         }
 
-        // --- segmented argmax by (eligible, seq, isAdd): groups <= k.
-        // Double-buffered through registers (static indices: rule-of-thumb
-        // from the CDNA guide — runtime-indexed locals spill to scratch).
-        for (int d = 1; d < k; d <<= 1) {
-            uint16_t nw[PMH_TILE_ITER];
-            uint8_t nf[PMH_TILE_ITER];
-#pragma unroll
-            for (int j = 0; j < PMH_TILE_ITER; j++) {
-                int32_t i = tid + j * PMH_TILE_THREADS;
-                if (i >= M) continue;
-                nw[j] = sm.win[i];
-                nf[j] = sm.fdone[i];
-                if (i >= d && !sm.fdone[i]) {
-                    uint16_t a = sm.win[i - d];
-                    uint16_t sa = mo[a], sb = mo[nw[j]];
-                    bool elig_a = !ignore_delete || kind_is_add(sm.skind[sa]);
-                    bool elig_b = !ignore_delete || kind_is_add(sm.skind[sb]);
-                    if (better(sm.sseq[sa], sm.skind[sa], elig_a,
-                               sm.sseq[sb], sm.skind[sb], elig_b))
-                        nw[j] = a;
-                    nf[j] = sm.fdone[i] | sm.fdone[i - d];
-                }
-            }
-            __syncthreads();
-#pragma unroll
-            for (int j = 0; j < PMH_TILE_ITER; j++) {
-                int32_t i = tid + j * PMH_TILE_THREADS;
-                if (i >= M) continue;
-                sm.win[i] = nw[j];
-                sm.fdone[i] = nf[j];
-            }
-            __syncthreads();
-        }
-
         if (ablate == 3) {
             if (tid == 0)
-                tile_counts[tile] =
-                    ((int32_t)sm.win[M - 1] ^ (int32_t)sm.fdone[0]) & 1;
+                tile_counts[tile] = ((int32_t)mo[0] ^ (int32_t)sm.head[0]) & 1;
             __syncthreads();
             continue;
         }
         // --- emit winners of owned groups, in key order.
         // A group is owned iff its head is a real (rank < Mreal) head; the
-        // winner is the segmented-argmax value at the group's tail. Two
-        // passes per thread over a contiguous head range (count, then
-        // write at the scanned offset) — no per-thread buffering.
-        __shared__ int32_t s_scan[PMH_TILE_THREADS];
+        // winner — DeduplicateMergeFunction's surviving record — is found
+        // inline while walking the group's <= k members (max by packed
+        // (eligible, seq, isAdd)). Two passes per thread over a contiguous
+        // head range (count, then write at the scanned offset).
         const int32_t per = (Mreal + (int32_t)blockDim.x - 1) / blockDim.x;
         int32_t my_lo = tid * per;
         int32_t my_hi = my_lo + per < Mreal ? my_lo + per : Mreal;
         uint32_t *wout = &winners[tile * (tile_rows + PMH_MAX_RUNS)];
         int32_t total = 0;
+        int32_t my_off = 0;
         for (int pass = 0; pass < 2; pass++) {
             int32_t nloc = 0;
-            int32_t my_off = pass == 1 ? s_scan[tid] : 0;
             for (int32_t i = my_lo; i < my_hi; i++) {
                 if (!sm.head[i]) continue;
-                int32_t tail = i;  // group tail: scan forward (<= k steps)
-                while (tail + 1 < M && !sm.head[tail + 1]) tail++;
-                uint16_t s = mo[sm.win[tail]];
+                // walk the group, tracking the winner by (elig, sseq)
+                int32_t tail = i;
+                uint16_t s_best = mo[i];
+                int64_t v_best = sm.sseq[s_best];
+                bool e_best = !ignore_delete || (v_best & 1);
+                while (tail + 1 < M && !sm.head[tail + 1]) {
+                    tail++;
+                    uint16_t s = mo[tail];
+                    int64_t v = sm.sseq[s];
+                    bool e = !ignore_delete || (v & 1);
+                    if ((e && !e_best) || (e == e_best && v > v_best)) {
+                        s_best = s;
+                        v_best = v;
+                        e_best = e;
+                    }
+                }
                 int32_t gsize = tail - i + 1;
-                bool elig_w = !ignore_delete || kind_is_add(sm.skind[s]);
-                if (!elig_w && gsize > 1) continue;  // all records ignored
-                if (drop_delete && !kind_is_add(sm.skind[s])) continue;
+                if (!e_best && gsize > 1) continue;  // all records ignored
+                if (drop_delete && !(v_best & 1)) continue;
                 if (pass == 1) {
                     int r = 0;  // map seg index -> (run, global row)
-                    while (r + 1 <= k - 1 && sm.segoff[r + 1] <= (int32_t)s)
+                    while (r + 1 <= k - 1 &&
+                           sm.segoff[r + 1] <= (int32_t)s_best)
                         r++;
-                    uint32_t grow =
-                        (uint32_t)(c0[r] + ((int32_t)s - sm.segoff[r]));
+                    uint32_t grow = (uint32_t)(
+                        c0[r] + ((int32_t)s_best - sm.segoff[r]));
                     wout[my_off + nloc] = ((uint32_t)r << 28) | grow;
                 }
                 nloc++;
             }
             if (pass == 0) {
-                s_scan[tid] = nloc;
-                __syncthreads();
-                // block exclusive scan (Hillis-Steele)
-                for (int d = 1; d < (int)blockDim.x; d <<= 1) {
-                    int32_t v = s_scan[tid];
-                    int32_t add = tid >= d ? s_scan[tid - d] : 0;
-                    __syncthreads();
-                    s_scan[tid] = v + add;
-                    __syncthreads();
+                int32_t incl = nloc;
+                for (int off = 1; off < 64; off <<= 1) {
+                    int32_t up = __shfl_up(incl, off, 64);
+                    if (lane >= off) incl += up;
                 }
-                total = s_scan[blockDim.x - 1];
-                int32_t incl = s_scan[tid];
+                if (lane == 63) sm.wave_tot[wv] = incl;
                 __syncthreads();
-                s_scan[tid] = incl - nloc;  // exclusive
-                __syncthreads();
+                int32_t add = 0;
+                total = 0;
+#pragma unroll
+                for (int w = 0; w < NW; w++) {
+                    if (w < wv) add += sm.wave_tot[w];
+                    total += sm.wave_tot[w];
+                }
+                my_off = add + incl - nloc;
             }
         }
         if (tid == 0) tile_counts[tile] = total;
@@ -1105,8 +1083,18 @@ hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
                                 hipStream_t stream) {
     int threads = 128;
     int blocks = (int)((n_bounds + threads - 1) / threads);
-    hipLaunchKernelGGL(k_partition, dim3(blocks), dim3(threads), 0, stream,
-                       keys, lens, k, tile_rows, n_bounds, total_rows, cuts);
+    if (k <= 4)
+        hipLaunchKernelGGL(k_partition<4>, dim3(blocks), dim3(threads), 0,
+                           stream, keys, lens, k, tile_rows, n_bounds,
+                           total_rows, cuts);
+    else if (k <= 8)
+        hipLaunchKernelGGL(k_partition<8>, dim3(blocks), dim3(threads), 0,
+                           stream, keys, lens, k, tile_rows, n_bounds,
+                           total_rows, cuts);
+    else
+        hipLaunchKernelGGL(k_partition<PMH_MAX_RUNS>, dim3(blocks),
+                           dim3(threads), 0, stream, keys, lens, k, tile_rows,
+                           n_bounds, total_rows, cuts);
     return hipGetLastError();
 }
 
