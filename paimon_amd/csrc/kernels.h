@@ -83,7 +83,7 @@ hipError_t pmh_launch_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks,
 hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
                                 int64_t tile_rows, int64_t n_bounds,
                                 int64_t total_rows, int32_t *cuts,
-                                int32_t *coarse_cuts, hipStream_t stream);
+                                hipStream_t stream);
 
 hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
                                   const DevCol *kinds, const int64_t *lens,

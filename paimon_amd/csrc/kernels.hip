@@ -94,9 +94,7 @@ template <int KM>  // compile-time run-count bound: keeps the per-thread
                    // window arrays in registers (KM=16 spilled 80 VGPRs)
 __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
                             int64_t tile_rows, int64_t n_bounds,
-                            int64_t total_rows, int32_t *cuts /* n_bounds*k */,
-                            const int32_t *coarse /* may be null */,
-                            int64_t coarse_step) {
+                            int64_t total_rows, int32_t *cuts /* n_bounds*k */) {
     int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (b >= n_bounds) return;
     int64_t D = b * tile_rows;
@@ -120,26 +118,16 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
         }
         return;
     }
-    if (coarse) {  // two-level: windows seeded from surrounding coarse cuts
-        int64_t cb = b / coarse_step;
-#pragma unroll
-        for (int r = 0; r < KM; r++) {
-            if (r >= k) continue;
-            wlo[r] = coarse[cb * k + r];
-            whi[r] = coarse[(cb + 1) * k + r];
-        }
-    }
     // bisect for the smallest v with (# ukey <= v) >= D: the cut falls at
-    // the D-th smallest element's key. Key domain from the windows' actual
-    // min/max keys (v* is an existing key; cnt_le(max key) covers D).
+    // the D-th smallest element's key. Domain initialized from the runs'
+    // actual min/max keys (v* is an existing key; cnt_le(max) = total >= D).
     uint64_t klo = ~0ull, khi = 0;
 #pragma unroll
     for (int r = 0; r < KM; r++) {
-        if (r >= k || wlo[r] >= whi[r]) continue;
-        uint64_t lo_k = ukey(
-            *reinterpret_cast<const int64_t *>(addr[r] + wlo[r] * 8));
+        if (r >= k || len[r] == 0) continue;
+        uint64_t lo_k = ukey(*reinterpret_cast<const int64_t *>(addr[r]));
         uint64_t hi_k = ukey(
-            *reinterpret_cast<const int64_t *>(addr[r] + (whi[r] - 1) * 8));
+            *reinterpret_cast<const int64_t *>(addr[r] + (len[r] - 1) * 8));
         if (lo_k < klo) klo = lo_k;
         if (hi_k > khi) khi = hi_k;
     }
@@ -1092,28 +1080,21 @@ extern "C" {
 hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
                                 int64_t tile_rows, int64_t n_bounds,
                                 int64_t total_rows, int32_t *cuts,
-                                int32_t *coarse_cuts, hipStream_t stream) {
-    const int threads = 128;
-    // two-level: coarse boundaries every COARSE tiles seed the fine windows
-    const int64_t COARSE = 64;
-    bool two = coarse_cuts && n_bounds > 2 * COARSE;
-    int64_t nbc = (n_bounds + COARSE - 1) / COARSE + 1;
-    auto go = [&](auto kern) {
-        if (two) {
-            int blks = (int)((nbc + threads - 1) / threads);
-            hipLaunchKernelGGL(kern, dim3(blks), dim3(threads), 0, stream,
-                               keys, lens, k, tile_rows * COARSE, nbc,
-                               total_rows, coarse_cuts, (const int32_t *)nullptr,
-                               (int64_t)0);
-        }
-        int blks = (int)((n_bounds + threads - 1) / threads);
-        hipLaunchKernelGGL(kern, dim3(blks), dim3(threads), 0, stream, keys,
-                           lens, k, tile_rows, n_bounds, total_rows, cuts,
-                           two ? coarse_cuts : nullptr, COARSE);
-    };
-    if (k <= 4) go(k_partition<4>);
-    else if (k <= 8) go(k_partition<8>);
-    else go(k_partition<PMH_MAX_RUNS>);
+                                hipStream_t stream) {
+    int threads = 128;
+    int blocks = (int)((n_bounds + threads - 1) / threads);
+    if (k <= 4)
+        hipLaunchKernelGGL(k_partition<4>, dim3(blocks), dim3(threads), 0,
+                           stream, keys, lens, k, tile_rows, n_bounds,
+                           total_rows, cuts);
+    else if (k <= 8)
+        hipLaunchKernelGGL(k_partition<8>, dim3(blocks), dim3(threads), 0,
+                           stream, keys, lens, k, tile_rows, n_bounds,
+                           total_rows, cuts);
+    else
+        hipLaunchKernelGGL(k_partition<PMH_MAX_RUNS>, dim3(blocks),
+                           dim3(threads), 0, stream, keys, lens, k, tile_rows,
+                           n_bounds, total_rows, cuts);
     return hipGetLastError();
 }
 

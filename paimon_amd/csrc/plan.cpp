@@ -221,7 +221,6 @@ struct Section {
     DevCol *all_cols = nullptr;   // [k * n_cols] run-major
     int64_t *lens_dev = nullptr;
     int32_t *cuts = nullptr;
-    int32_t *coarse_cuts = nullptr;
     uint32_t *winners = nullptr;
     int32_t *tile_counts = nullptr;
     int64_t *tile_offsets = nullptr;
@@ -1171,8 +1170,6 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
     sec.all_cols = (DevCol *)up(allv.data(), allv.size() * sizeof(DevCol));
     sec.lens_dev = (int64_t *)up(lens.data(), k * sizeof(int64_t));
     sec.cuts = (int32_t *)plan->bufs.alloc((sec.n_tiles + 1) * k * 4);
-    sec.coarse_cuts =
-        (int32_t *)plan->bufs.alloc(((sec.n_tiles + 1) / 64 + 2) * k * 4);
     sec.winners = (uint32_t *)plan->bufs.alloc(
         sec.n_tiles * (PMH_TILE_ROWS + PMH_MAX_RUNS) * 4);
     sec.tile_counts = (int32_t *)plan->bufs.alloc(sec.n_tiles * 4);
@@ -1451,8 +1448,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     hipEventRecord(ev[1], st);
     hipError_t e = pmh_launch_partition(sec.key_cols, sec.lens_dev, k,
                                         PMH_TILE_ROWS, sec.n_tiles + 1,
-                                        sec.total_rows, sec.cuts,
-                                        sec.coarse_cuts, st);
+                                        sec.total_rows, sec.cuts, st);
     if (e != hipSuccess) return fail("partition", e);
     hipEventRecord(ev[2], st);
     int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0) |
