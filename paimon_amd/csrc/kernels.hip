@@ -844,29 +844,38 @@ __global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks,
             __syncthreads();
             continue;
         }
-        // bit-packed: 1 byte per 8 levels
+        // bit-packed: 1 byte per 8 levels. Dense prefixes come from wave
+        // ballots (popcount of lower lanes) + a tiny cross-wave LDS combine
+        // — the former Hillis-Steele block scan cost 16 barriers per 256
+        // values.
         const uint8_t *src = (const uint8_t *)ch.src;
+        const int lane = (int)(threadIdx.x & 63);
+        const int wv = (int)(threadIdx.x >> 6);
+        constexpr int NW = PMH_TILE_THREADS / 64;
         if (threadIdx.x == 0) s_running = 0;
         __syncthreads();
         for (int32_t b = 0; b < ch.count; b += (int32_t)blockDim.x) {
             int32_t i = b + threadIdx.x;
             int bit = 0;
             if (i < ch.count) bit = (src[i >> 3] >> (i & 7)) & 1;
-            s_scan[threadIdx.x] = bit;
+            uint64_t mask = __ballot(bit != 0);
+            uint64_t lt = lane == 0 ? 0 : (mask << (64 - lane));
+            int32_t my_before = __popcll(lt);
+            int32_t wave_total = __popcll(mask);
+            if (lane == 0) s_scan[wv] = wave_total;
             __syncthreads();
-            for (int d = 1; d < (int)blockDim.x; d <<= 1) {
-                int32_t v = s_scan[threadIdx.x];
-                int32_t a = threadIdx.x >= d ? s_scan[threadIdx.x - d] : 0;
-                __syncthreads();
-                s_scan[threadIdx.x] = v + a;
-                __syncthreads();
+            int32_t wave_base = 0, round_total = 0;
+#pragma unroll
+            for (int w = 0; w < NW; w++) {
+                if (w < wv) wave_base += s_scan[w];
+                round_total += s_scan[w];
             }
             if (i < ch.count) {
                 valid[ch.out_start + i] = (uint8_t)bit;
                 uint8_t *dp = out + (ch.out_start + i) * esize;
                 if (bit) {
                     int64_t didx =
-                        ch.aux + s_running + s_scan[threadIdx.x] - 1;
+                        ch.aux + s_running + wave_base + my_before;
                     const uint8_t *sp = dense + didx * esize;
                     if (esize == 4)
                         *(int32_t *)dp = *(const int32_t *)sp;
@@ -880,7 +889,7 @@ __global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks,
                 }
             }
             __syncthreads();
-            if (threadIdx.x == 0) s_running += s_scan[blockDim.x - 1];
+            if (threadIdx.x == 0) s_running += round_total;
             __syncthreads();
         }
     }

@@ -1275,6 +1275,17 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                       engine.c_str());
             return nullptr;
         }
+        // sort-engine (CoreOptions.java:733-736, default LOSER_TREE):
+        // both engines share one output contract (SortMergeReader.java:41-57
+        // — identical merged stream), so either value maps to the same GPU
+        // merge; accepted for drop-in config compatibility.
+        {
+            std::string se = j["sort_engine"].as_str("loser-tree");
+            if (se != "loser-tree" && se != "min-heap") {
+                set_error("unknown sort-engine '%s'", se.c_str());
+                return nullptr;
+            }
+        }
         plan->drop_delete = j["drop_delete"].as_bool(true);
         plan->ignore_delete = j["ignore_delete"].as_bool(false);
         plan->host_output = j["output"].as_str("device") == "host";
