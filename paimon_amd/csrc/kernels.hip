@@ -756,7 +756,16 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
         const uint16_t *gs = &group_start[t * (tile_rows + 1)];
         const uint32_t *mem = &members[t * (tile_rows + PMH_MAX_RUNS)];
         int32_t ms = gs[g], me = gs[g + 1];
-        uint32_t last = mem[me - 1];
+        // cache the newest members in registers: the per-column non-null
+        // walk starts at the newest record and nearly always ends within a
+        // few members (group size averages ~1.3), and re-reading the member
+        // list from HBM per column dominated this kernel
+        uint32_t mc[4];
+        const int gn = me - ms;
+#pragma unroll
+        for (int x = 0; x < 4; x++)
+            mc[x] = x < gn ? mem[me - 1 - x] : mem[me - 1];
+        uint32_t last = mc[0];
         int lrun = last >> 28;
         int64_t lrow = last & 0x0fffffff;
         for (int c = 0; c < n_cols; c++) {
@@ -766,9 +775,23 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
             }
             int64_t run = lrun, row = lrow;
             uint8_t ok = 1;
-            if (col_nullable[c]) {
+            if (col_nullable[c] && gn > 1) {
                 ok = 0;
-                for (int32_t x = me - 1; x >= ms; x--) {
+#pragma unroll
+                for (int x = 0; x < 4; x++) {
+                    if (ok || x >= gn) continue;
+                    uint32_t m = mc[x];
+                    const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                    uint8_t v = dc.valid0
+                                    ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                                    : 1;
+                    if (v) {
+                        run = m >> 28;
+                        row = m & 0x0fffffff;
+                        ok = 1;
+                    }
+                }
+                for (int32_t x = me - 5; !ok && x >= ms; x--) {
                     uint32_t m = mem[x];
                     const DevCol &dc = cols[(m >> 28) * n_cols + c];
                     uint8_t v = dc.valid0
@@ -778,9 +801,12 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                         run = m >> 28;
                         row = m & 0x0fffffff;
                         ok = 1;
-                        break;
                     }
                 }
+            } else if (col_nullable[c]) {
+                // singleton: the record passes through with its own validity
+                const DevCol &dc = cols[lrun * n_cols + c];
+                ok = dc.valid0 ? ((const uint8_t *)dc.valid0)[lrow] : 1;
             }
             const DevCol &dc = cols[run * n_cols + c];
             switch (col_dtype[c]) {
