@@ -160,7 +160,7 @@ struct DeviceBufs {
         return p;
     }
     ~DeviceBufs() {
-        for (void *p : bufs) hipFree(p);
+        for (void *p : bufs) (void)hipFree(p);
     }
 };
 
@@ -1161,7 +1161,7 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
     if (sec.n_tiles == 0) sec.n_tiles = 1;
     auto up = [&](const void *host, size_t n) -> void * {
         void *d = plan->bufs.alloc(n);
-        if (d && host) hipMemcpy(d, host, n, hipMemcpyHostToDevice);
+        if (d && host) (void)hipMemcpy(d, host, n, hipMemcpyHostToDevice);
         return d;
     };
     sec.key_cols = (DevCol *)up(keyv.data(), k * sizeof(DevCol));
@@ -1176,7 +1176,7 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
     sec.tile_offsets = (int64_t *)plan->bufs.alloc(sec.n_tiles * 8);
     sec.total_dev = (int64_t *)plan->bufs.alloc(8);
     sec.err_dev = (uint32_t *)plan->bufs.alloc(4);
-    if (sec.err_dev) hipMemset(sec.err_dev, 0, 4);
+    if (sec.err_dev) (void)hipMemset(sec.err_dev, 0, 4);
     if (plan->pu) {
         sec.group_start = (uint16_t *)plan->bufs.alloc(
             sec.n_tiles * (PMH_TILE_ROWS + 1) * 2);
@@ -1227,7 +1227,7 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         set_error("pmh_plan_create requires a GPU session (device >= 0)");
         return nullptr;
     }
-    hipSetDevice(s->device);
+    (void)hipSetDevice(s->device);
     std::unique_ptr<pmh_plan_t> plan(new pmh_plan_t());
     plan->session = s;
     try {
@@ -1312,7 +1312,10 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             return nullptr;
         }
 
-        hipStreamCreate(&plan->stream);
+        if (hipStreamCreate(&plan->stream) != hipSuccess) {
+            set_error("hipStreamCreate failed");
+            return nullptr;
+        }
 
         auto sections = interval_partition(files);
         auto t0 = std::chrono::steady_clock::now();
@@ -1360,10 +1363,14 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         plan->out_dev = outs;
         plan->out_ptrs_dev = (void **)plan->bufs.alloc(n_cols * sizeof(void *));
         plan->col_dtype_dev = (uint8_t *)plan->bufs.alloc(n_cols);
-        hipMemcpy(plan->out_ptrs_dev, outs.data(), n_cols * sizeof(void *),
-                  hipMemcpyHostToDevice);
-        hipMemcpy(plan->col_dtype_dev, dts.data(), n_cols,
-                  hipMemcpyHostToDevice);
+        if (hipMemcpy(plan->out_ptrs_dev, outs.data(),
+                      n_cols * sizeof(void *),
+                      hipMemcpyHostToDevice) != hipSuccess ||
+            hipMemcpy(plan->col_dtype_dev, dts.data(), n_cols,
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_error("H2D of output descriptors failed");
+            return nullptr;
+        }
         // per-column nullability (any staged nulls in any section/run) +
         // output validity buffers
         plan->col_nullable.assign(n_cols, false);
@@ -1386,10 +1393,14 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         plan->col_nullable_dev = (uint8_t *)plan->bufs.alloc(n_cols);
         plan->out_valid_dev =
             (uint8_t **)plan->bufs.alloc(n_cols * sizeof(void *));
-        hipMemcpy(plan->col_nullable_dev, nul.data(), n_cols,
-                  hipMemcpyHostToDevice);
-        hipMemcpy(plan->out_valid_dev, plan->out_valid.data(),
-                  n_cols * sizeof(void *), hipMemcpyHostToDevice);
+        if (hipMemcpy(plan->col_nullable_dev, nul.data(), n_cols,
+                      hipMemcpyHostToDevice) != hipSuccess ||
+            hipMemcpy(plan->out_valid_dev, plan->out_valid.data(),
+                      n_cols * sizeof(void *),
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_error("H2D of validity descriptors failed");
+            return nullptr;
+        }
         for (auto &cs : plan->cols) plan->col_names.push_back(cs.name);
     } catch (const std::exception &e) {
         set_error("plan parse: %s", e.what());
@@ -1405,24 +1416,24 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         return -1;
     }
     if (p->cur_section >= p->sections.size()) return 0;
-    hipSetDevice(p->session->device);
+    (void)hipSetDevice(p->session->device);
     Section &sec = p->sections[p->cur_section++];
     hipStream_t st = p->stream;
     int k = (int)sec.runs.size();
     int n_cols = (int)p->cols.size();
 
     hipEvent_t ev[6];
-    for (auto &e : ev) hipEventCreate(&e);
+    for (auto &e : ev) (void)hipEventCreate(&e);
     // ev: 0 start, 1 decode done, 2 partition done, 3 merge done,
     //     4 scan done, 5 emit done
 
     auto fail = [&](const char *what, hipError_t e) -> int64_t {
         set_error("%s: %s", what, hipGetErrorString(e));
-        for (auto &evv : ev) hipEventDestroy(evv);
+        for (auto &evv : ev) (void)hipEventDestroy(evv);
         return -1;
     };
 
-    hipEventRecord(ev[0], st);
+    (void)hipEventRecord(ev[0], st);
     // decode: dictionary materialization + null-column level scatter
     if (sec.any_decode) {
         for (auto &run : sec.runs) {
@@ -1456,12 +1467,12 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             }
         }
     }
-    hipEventRecord(ev[1], st);
+    (void)hipEventRecord(ev[1], st);
     hipError_t e = pmh_launch_partition(sec.key_cols, sec.lens_dev, k,
                                         PMH_TILE_ROWS, sec.n_tiles + 1,
                                         sec.total_rows, sec.cuts, st);
     if (e != hipSuccess) return fail("partition", e);
-    hipEventRecord(ev[2], st);
+    (void)hipEventRecord(ev[2], st);
     int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0) |
                 (p->pu ? 4 : 0);
     if (const char *ab = getenv("PMH_ABLATE"))  // profiling-only phase knob
@@ -1472,11 +1483,11 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                                sec.tile_counts, sec.group_start, sec.err_dev,
                                st);
     if (e != hipSuccess) return fail("merge_tiles", e);
-    hipEventRecord(ev[3], st);
+    (void)hipEventRecord(ev[3], st);
     e = pmh_launch_scan_tiles(sec.tile_counts, sec.n_tiles, sec.tile_offsets,
                               sec.total_dev, st);
     if (e != hipSuccess) return fail("scan_tiles", e);
-    hipEventRecord(ev[4], st);
+    (void)hipEventRecord(ev[4], st);
     if (p->pu) {
         e = pmh_launch_emit_pu(
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,
@@ -1491,7 +1502,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                             p->out_valid_dev, st);
     }
     if (e != hipSuccess) return fail("emit", e);
-    hipEventRecord(ev[5], st);
+    (void)hipEventRecord(ev[5], st);
 
     int64_t total = 0;
     uint32_t err_word = 0;
@@ -1510,19 +1521,19 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     }
 
     float ms;
-    hipEventElapsedTime(&ms, ev[0], ev[1]);
+    (void)hipEventElapsedTime(&ms, ev[0], ev[1]);
     p->stats.decode_ms += ms;
-    hipEventElapsedTime(&ms, ev[1], ev[2]);
+    (void)hipEventElapsedTime(&ms, ev[1], ev[2]);
     p->stats.partition_ms += ms;
-    hipEventElapsedTime(&ms, ev[2], ev[3]);
+    (void)hipEventElapsedTime(&ms, ev[2], ev[3]);
     p->stats.merge_ms += ms;
-    hipEventElapsedTime(&ms, ev[3], ev[4]);
+    (void)hipEventElapsedTime(&ms, ev[3], ev[4]);
     p->stats.scan_ms += ms;
-    hipEventElapsedTime(&ms, ev[4], ev[5]);
+    (void)hipEventElapsedTime(&ms, ev[4], ev[5]);
     p->stats.emit_ms += ms;
-    hipEventElapsedTime(&ms, ev[0], ev[5]);
+    (void)hipEventElapsedTime(&ms, ev[0], ev[5]);
     p->stats.total_device_ms += ms;
-    for (auto &evv : ev) hipEventDestroy(evv);
+    for (auto &evv : ev) (void)hipEventDestroy(evv);
 
     p->stats.rows_in += sec.total_rows;
     p->stats.rows_out += total;
@@ -1584,7 +1595,7 @@ int pmh_plan_reset(pmh_plan_t *p) {
 
 int pmh_plan_close(pmh_plan_t *p) {
     if (!p) return 0;
-    if (p->stream) hipStreamDestroy(p->stream);
+    if (p->stream) (void)hipStreamDestroy(p->stream);
     delete p;
     return 0;
 }
