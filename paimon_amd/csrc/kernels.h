@@ -13,8 +13,8 @@ namespace pmh {
 //    MergeSorter.java:112-125);
 //  - rows per run < 2^28 (winner packing run:4 | row:28).
 constexpr int PMH_MAX_RUNS = 16;
-constexpr int PMH_TILE_THREADS = 256;
-constexpr int64_t PMH_TILE_ROWS = 1792;
+constexpr int PMH_TILE_THREADS = 512;
+constexpr int64_t PMH_TILE_ROWS = 3584;
 constexpr int PMH_TILE_MAX = PMH_TILE_ROWS + PMH_MAX_RUNS;
 constexpr int PMH_TILE_ITER =
     (PMH_TILE_MAX + PMH_TILE_THREADS - 1) / PMH_TILE_THREADS;

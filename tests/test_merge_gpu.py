@@ -310,3 +310,28 @@ def _read_all_batches(plan):
         for k, v in b.items():
             got.setdefault(k, []).append(v.copy())
     return {k: np.concatenate(v) for k, v in got.items()}
+
+
+class TestConcurrentPlans:
+    def test_two_plans_interleaved(self, tmp_path):
+        # one plan per bucket, interleaved batches (the C-ABI threading
+        # contract: one plan = one HIP stream; concurrent plans allowed)
+        runs_a = gen_runs_dedup(4, 20_000, n_value_cols=2, seed=61)
+        runs_b = gen_runs_dedup(4, 20_000, n_value_cols=2, seed=62)
+        metas_a = write_runs(runs_a, str(tmp_path / "a"), compression="NONE")
+        metas_b = write_runs(runs_b, str(tmp_path / "b"), compression="NONE")
+        ra, wa = merge_dedup(runs_a)
+        rb, wb = merge_dedup(runs_b)
+        with Session(0) as s:
+            pa_ = MergeReadPlan(s, file_descs_from_metas(metas_a), KEY_COLS,
+                                _value_cols(2))
+            pb_ = MergeReadPlan(s, file_descs_from_metas(metas_b), KEY_COLS,
+                                _value_cols(2))
+            ga = pa_.read_next()
+            gb = pb_.read_next()
+            ea = np.array([runs_a[x]["key"][y] for x, y in zip(ra, wa)])
+            eb = np.array([runs_b[x]["key"][y] for x, y in zip(rb, wb)])
+            assert (ga["_KEY_k"] == ea).all()
+            assert (gb["_KEY_k"] == eb).all()
+            pa_.close()
+            pb_.close()
