@@ -778,8 +778,9 @@ static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
                           fd.path.c_str(), cols[c].name.c_str());
                 return false;
             }
-            // upload the encoded DATA stream
-            void *dev = plan->bufs.alloc(data->length);
+            // upload the encoded DATA stream (+16 B pad: the RLEv2
+            // bit reader uses an aligned 16-byte window)
+            void *dev = plan->bufs.alloc(data->length + 16);
             if (!dev) return false;
             if (hipMemcpy(dev, sf.data.data() + data->offset, data->length,
                           hipMemcpyHostToDevice) != hipSuccess) {
