@@ -22,13 +22,10 @@ a reported baseline, not the optimization target.
 """
 
 import argparse
-import ctypes
 import json
 import os
 import sys
 import time
-
-import numpy as np
 
 REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
