@@ -46,6 +46,13 @@ struct RleChunk {
     int32_t bit_width;  // per-page bit width (dict-id streams vary per page)
     int64_t aux;        // def-level streams: dense (non-null) values before
                         // this chunk within its (run, column)
+    // absolute device addresses (patched at staging finalize) so decode
+    // chunks from every (run, column) batch into ONE launch:
+    uint64_t out_addr;    // positioned column values
+    uint64_t valid_addr;  // byte validity
+    uint64_t dense_addr;  // dense (non-null) values
+    int32_t esize;
+    int32_t _pad2;
 };
 
 // Host-prescanned ORC RLEv2 / byte-RLE work unit (one run per chunk; runs
@@ -67,9 +74,10 @@ struct Rlev2Chunk {
     uint8_t patch_pw;    // PATCHED: patch value bits (decoded)
     uint8_t patch_pgw;   // PATCHED: gap bits
     uint8_t patch_cfb;   // PATCHED: packed entry bits
-    uint8_t dense_target;  // 0: write contig (positioned); 1: write the
-                           // dense buffer (PRESENT scatter follows)
+    uint8_t dense_target;  // 0: contig (positioned); 1: dense buffer
+                           // (PRESENT scatter follows)
     uint8_t _pad[2];
+    uint64_t out_addr;     // absolute output base (patched at finalize)
 };
 
 extern "C" {
@@ -77,7 +85,6 @@ extern "C" {
 // Decode ORC RLEv2 / byte-RLE work chunks into a dense typed column
 // (int32 or int64 elements per chunk.out_esize). One wave per chunk.
 hipError_t pmh_launch_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks,
-                            void *out_contig, void *out_dense,
                             hipStream_t stream);
 
 hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
@@ -113,8 +120,6 @@ hipError_t pmh_launch_rle_decode(const RleChunk *chunks, int64_t n_chunks,
 // values: valid[row] = level; out[row] = dense[aux + prefix] for valid rows
 // (VectorizedColumnReader null handling, VectorizedColumnReader.java:143-241).
 hipError_t pmh_launch_level_scatter(const RleChunk *chunks, int64_t n_chunks,
-                                    const void *dense, void *out,
-                                    uint8_t *valid, int esize,
                                     hipStream_t stream);
 
 // PartialUpdate emit: per owned group, overlay non-null fields in ascending

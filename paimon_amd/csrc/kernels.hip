@@ -839,13 +839,15 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
 // paimon-format/.../reader/VectorizedColumnReader.java:143-241). One
 // workgroup per chunk; packed chunks compute the dense prefix with an
 // in-LDS running scan over 256-element blocks.
-__global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks,
-                                const uint8_t *dense, uint8_t *out,
-                                uint8_t *valid, int esize) {
+__global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks) {
     __shared__ int32_t s_scan[PMH_TILE_THREADS];
     __shared__ int32_t s_running;
     for (int64_t cidx = blockIdx.x; cidx < n_chunks; cidx += gridDim.x) {
         RleChunk ch = chunks[cidx];
+        const uint8_t *dense = (const uint8_t *)ch.dense_addr;
+        uint8_t *out = (uint8_t *)ch.out_addr;
+        uint8_t *valid = (uint8_t *)ch.valid_addr;
+        const int esize = ch.esize;
         if (ch.kind == 0) {
             if (ch.value) {  // run of non-nulls: dense block copy
                 for (int32_t i = threadIdx.x; i < ch.count; i += blockDim.x) {
@@ -954,15 +956,14 @@ DEV void dense_store(void *out, int esize, int64_t idx, int64_t v) {
         ((int64_t *)out)[idx] = v;
 }
 
-__global__ void k_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks,
-                        void *out_contig, void *out_dense) {
+__global__ void k_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks) {
     const int wave = (int)(threadIdx.x >> 6);
     const int lane = (int)(threadIdx.x & 63);
     const int waves = (int)(blockDim.x >> 6);
     for (int64_t cidx = (int64_t)blockIdx.x * waves + wave; cidx < n_chunks;
          cidx += (int64_t)gridDim.x * waves) {
         Rlev2Chunk ch = chunks[cidx];
-        void *out = ch.dense_target ? out_dense : out_contig;
+        void *out = (void *)ch.out_addr;
         const uint8_t *src = (const uint8_t *)ch.src;
         switch (ch.kind) {
         case 0:  // SHORT_REPEAT (value already sign-decoded by the host)
@@ -1165,13 +1166,10 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
 }
 
 hipError_t pmh_launch_level_scatter(const RleChunk *chunks, int64_t n_chunks,
-                                    const void *dense, void *out,
-                                    uint8_t *valid, int esize,
                                     hipStream_t stream) {
     int blocks = n_chunks < 4096 ? (int)(n_chunks ? n_chunks : 1) : 4096;
     hipLaunchKernelGGL(k_level_scatter, dim3(blocks), dim3(PMH_TILE_THREADS),
-                       0, stream, chunks, n_chunks, (const uint8_t *)dense,
-                       (uint8_t *)out, valid, esize);
+                       0, stream, chunks, n_chunks);
     return hipGetLastError();
 }
 
@@ -1199,13 +1197,12 @@ hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
 }
 
 hipError_t pmh_launch_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks,
-                            void *out_contig, void *out_dense,
                             hipStream_t stream) {
     int waves_per_block = 4;  // 256 threads
     int64_t want = (n_chunks + waves_per_block - 1) / waves_per_block;
     int blocks = want < 4096 ? (int)(want ? want : 1) : 4096;
     hipLaunchKernelGGL(k_rlev2, dim3(blocks), dim3(256), 0, stream, chunks,
-                       n_chunks, out_contig, out_dense);
+                       n_chunks);
     return hipGetLastError();
 }
 

@@ -227,6 +227,11 @@ struct Section {
     int64_t *total_dev = nullptr;
     uint16_t *group_start = nullptr;  // partial-update member offsets
     uint32_t *err_dev = nullptr;
+    // batched decode work (all run-columns in ONE launch each)
+    Rlev2Chunk *rlev2_all = nullptr;
+    int64_t n_rlev2 = 0;
+    RleChunk *def_all = nullptr;
+    int64_t n_def = 0;
     bool any_dict = false;
     bool any_decode = false;  // dict or null-scatter work at read time
 };
@@ -1077,15 +1082,6 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                           hipMemcpyHostToDevice) != hipSuccess)
                 return false;
         }
-        if (!rc.rlev2_host.empty()) {
-            rc.rlev2_dev = (Rlev2Chunk *)plan->bufs.alloc(
-                rc.rlev2_host.size() * sizeof(Rlev2Chunk));
-            if (!rc.rlev2_dev) return false;
-            if (hipMemcpy(rc.rlev2_dev, rc.rlev2_host.data(),
-                          rc.rlev2_host.size() * sizeof(Rlev2Chunk),
-                          hipMemcpyHostToDevice) != hipSuccess)
-                return false;
-        }
         if (rc.has_nulls) {
             // parquet: dense values packed host-side (dense_host);
             // ORC: dense values are produced on-device by k_rlev2
@@ -1096,9 +1092,7 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
             rc.valid_dev = (uint8_t *)plan->bufs.alloc(run.length);
             rc.dense_dev = plan->bufs.alloc(dense_bytes);
             void *levels_dev = plan->bufs.alloc(rc.levels_host.size());
-            rc.def_dev = (RleChunk *)plan->bufs.alloc(rc.def_host.size() *
-                                                      sizeof(RleChunk));
-            if (!rc.valid_dev || !rc.dense_dev || !levels_dev || !rc.def_dev)
+            if (!rc.valid_dev || !rc.dense_dev || !levels_dev)
                 return false;
             // rows in non-null chunks of this column keep validity 1
             if (hipMemset(rc.valid_dev, 1, run.length) != hipSuccess)
@@ -1112,18 +1106,22 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                           rc.levels_host.size(),
                           hipMemcpyHostToDevice) != hipSuccess)
                 return false;
-            for (auto &dc : rc.def_host)
+            for (auto &dc : rc.def_host) {
                 if (dc.kind == 1) dc.src += (uint64_t)levels_dev;
-            if (hipMemcpy(rc.def_dev, rc.def_host.data(),
-                          rc.def_host.size() * sizeof(RleChunk),
-                          hipMemcpyHostToDevice) != hipSuccess)
-                return false;
+                dc.out_addr = (uint64_t)rc.contig;
+                dc.valid_addr = (uint64_t)rc.valid_dev;
+                dc.dense_addr = (uint64_t)rc.dense_dev;
+                dc.esize = plan->cols[c].stored_esize;
+            }
             plan->encoded_bytes_total += rc.levels_host.size();
             rc.dense_host.clear();
             rc.dense_host.shrink_to_fit();
             rc.levels_host.clear();
             rc.levels_host.shrink_to_fit();
         }
+        for (auto &vc : rc.rlev2_host)
+            vc.out_addr =
+                vc.dense_target ? (uint64_t)rc.dense_dev : (uint64_t)rc.contig;
         DevPage dp{(uint64_t)rc.contig, 0};
         rc.pages_host.assign(1, dp);
         rc.n_pages = 1;
@@ -1183,12 +1181,29 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
             sec.n_tiles * (PMH_TILE_ROWS + 1) * 2);
         if (!sec.group_start) return false;
     }
+    std::vector<Rlev2Chunk> all_v;
+    std::vector<RleChunk> all_d;
     for (auto &run : sec.runs)
         for (auto &rc : run.cols) {
             sec.any_dict |= rc.dict_encoded;
             sec.any_decode |=
                 rc.dict_encoded || rc.has_nulls || !rc.rlev2_host.empty();
+            all_v.insert(all_v.end(), rc.rlev2_host.begin(),
+                         rc.rlev2_host.end());
+            all_d.insert(all_d.end(), rc.def_host.begin(), rc.def_host.end());
         }
+    sec.n_rlev2 = (int64_t)all_v.size();
+    sec.n_def = (int64_t)all_d.size();
+    if (sec.n_rlev2) {
+        sec.rlev2_all =
+            (Rlev2Chunk *)up(all_v.data(), all_v.size() * sizeof(Rlev2Chunk));
+        if (!sec.rlev2_all) return false;
+    }
+    if (sec.n_def) {
+        sec.def_all =
+            (RleChunk *)up(all_d.data(), all_d.size() * sizeof(RleChunk));
+        if (!sec.def_all) return false;
+    }
     return sec.key_cols && sec.seq_cols && sec.kind_cols && sec.all_cols &&
            sec.lens_dev && sec.cuts && sec.winners && sec.tile_counts &&
            sec.tile_offsets && sec.total_dev && sec.err_dev;
@@ -1435,37 +1450,35 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     };
 
     (void)hipEventRecord(ev[0], st);
-    // decode: dictionary materialization + null-column level scatter
+    // decode: batched RLEv2/byte-RLE + PRESENT/def-level scatter (one
+    // launch each across every run-column — the per-run-column launches
+    // serialized ~350 small kernels per step on C3), plus parquet
+    // dictionary materialization per chunk
     if (sec.any_decode) {
+        if (sec.n_rlev2) {
+            hipError_t e = pmh_launch_rlev2(sec.rlev2_all, sec.n_rlev2, st);
+            if (e != hipSuccess) return fail("rlev2", e);
+        }
         for (auto &run : sec.runs) {
             for (size_t c = 0; c < run.cols.size(); c++) {
                 RunCol &rc = run.cols[c];
                 int es = p->cols[c].stored_esize;
-                if (rc.dict_encoded) {
-                    hipError_t e = pmh_launch_rle_decode(
-                        rc.rle_dev, (int64_t)rc.rle_host.size(), rc.ids_dev,
-                        st);
-                    if (e != hipSuccess) return fail("rle_decode", e);
-                    for (const GatherTask &gt : rc.gathers) {
-                        e = pmh_launch_dict_gather(
-                            rc.ids_dev + gt.start, gt.dict_dev, gt.n,
-                            (uint8_t *)rc.contig + gt.start * es, es, st);
-                        if (e != hipSuccess) return fail("dict_gather", e);
-                    }
-                }
-                if (!rc.rlev2_host.empty()) {
-                    hipError_t e = pmh_launch_rlev2(
-                        rc.rlev2_dev, (int64_t)rc.rlev2_host.size(),
-                        rc.contig, rc.dense_dev, st);
-                    if (e != hipSuccess) return fail("rlev2", e);
-                }
-                if (rc.has_nulls) {
-                    hipError_t e = pmh_launch_level_scatter(
-                        rc.def_dev, (int64_t)rc.def_host.size(), rc.dense_dev,
-                        rc.contig, rc.valid_dev, es, st);
-                    if (e != hipSuccess) return fail("level_scatter", e);
+                if (!rc.dict_encoded) continue;
+                hipError_t e = pmh_launch_rle_decode(
+                    rc.rle_dev, (int64_t)rc.rle_host.size(), rc.ids_dev, st);
+                if (e != hipSuccess) return fail("rle_decode", e);
+                for (const GatherTask &gt : rc.gathers) {
+                    e = pmh_launch_dict_gather(
+                        rc.ids_dev + gt.start, gt.dict_dev, gt.n,
+                        (uint8_t *)rc.contig + gt.start * es, es, st);
+                    if (e != hipSuccess) return fail("dict_gather", e);
                 }
             }
+        }
+        if (sec.n_def) {
+            hipError_t e =
+                pmh_launch_level_scatter(sec.def_all, sec.n_def, st);
+            if (e != hipSuccess) return fail("level_scatter", e);
         }
     }
     (void)hipEventRecord(ev[1], st);
