@@ -8,6 +8,7 @@ from .oracle import (  # noqa: F401
     merge_order,
     merge_dedup,
     merge_dedup_model,
+    merge_first_row_model,
     partial_update_model,
     rle_bp_decode,
     lib_path,

@@ -238,3 +238,35 @@ def orc_boolrle_decode(data: bytes, num_values: int):
     if n != num_values:
         raise ValueError(f"bool rle decode: {n} != {num_values}")
     return out.astype(bool)
+
+
+def merge_first_row_model(runs, ignore_delete=False, drop_delete=True):
+    """Numpy restatement of getExpectedForFirstRow
+    (MergeFunctionTestUtils.java:135-147) + FirstRowMergeFunction semantics
+    (FirstRowMergeFunction.java:32-77): FIRST eligible record per key group
+    in ascending (seq, isAdd) order; retracts are ineligible under
+    ignore-delete (and reject the stream otherwise — callers pass
+    insert-only or ignore_delete=True); singleton groups bypass the merge
+    function. Returns (run, row)."""
+    key, seq, kind, run, row = _sorted_stream(runs)
+    n = len(key)
+    if n == 0:
+        return (np.empty(0, dtype=np.int32), np.empty(0, dtype=np.int64))
+    head = np.empty(n, dtype=bool)
+    head[0] = True
+    head[1:] = key[1:] != key[:-1]
+    group_id = np.cumsum(head) - 1
+    n_groups = group_id[-1] + 1
+    group_size = np.bincount(group_id, minlength=n_groups)
+    idx = np.arange(n)
+    eligible = _kind_is_add(kind) if ignore_delete else np.ones(n, dtype=bool)
+    first = np.full(n_groups, n, dtype=np.int64)
+    np.minimum.at(first, group_id[eligible], idx[eligible])
+    single_first = np.full(n_groups, n, dtype=np.int64)
+    np.minimum.at(single_first, group_id, idx)
+    singles = group_size == 1
+    first[singles] = single_first[singles]
+    sel = first[first < n]
+    if drop_delete:
+        sel = sel[_kind_is_add(kind[sel])]
+    return run[sel], row[sel]

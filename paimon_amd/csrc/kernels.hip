@@ -245,6 +245,11 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    uint16_t *group_start, uint32_t *err_flag) {
     const bool drop_delete = flags & 1;
     const bool ignore_delete = flags & 2;
+    // first-row engine (FirstRowMergeFunction.java:32-77): keep the FIRST
+    // record per key in ascending (seq, isAdd) order; retracts throw unless
+    // ignore-delete (then skipped); singleton groups bypass the merge
+    // function (wrapper) exactly as deduplicate does.
+    const bool first_row = flags & 8;
     // ablation levels (profiling only, flags bits 8..): 1=stage,2=+merge,
     // 3=+scan, 0/absent=full. Partial levels publish a checksum so the
     // compiler cannot dead-code the ablated phases' inputs.
@@ -522,23 +527,32 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             int32_t nloc = 0;
             for (int32_t i = my_lo; i < my_hi; i++) {
                 if (!sm.head[i]) continue;
-                // walk the group, tracking the winner by (elig, sseq)
+                // walk the group, tracking the winner by (elig, sseq):
+                // deduplicate keeps the LAST record, first-row the FIRST
                 int32_t tail = i;
                 uint16_t s_best = mo[i];
                 int64_t v_best = sm.sseq[s_best];
                 bool e_best = !ignore_delete || (v_best & 1);
+                bool any_retract = !(v_best & 1);
                 while (tail + 1 < M && !sm.head[tail + 1]) {
                     tail++;
                     uint16_t s = mo[tail];
                     int64_t v = sm.sseq[s];
                     bool e = !ignore_delete || (v & 1);
-                    if ((e && !e_best) || (e == e_best && v > v_best)) {
+                    any_retract |= !(v & 1);
+                    bool take = (e && !e_best) ||
+                                (e == e_best &&
+                                 (first_row ? v < v_best : v > v_best));
+                    if (take) {
                         s_best = s;
                         v_best = v;
                         e_best = e;
                     }
                 }
                 int32_t gsize = tail - i + 1;
+                if (first_row && !ignore_delete && any_retract && gsize > 1 &&
+                    err_flag)
+                    atomicOr(err_flag, 2u);  // FirstRow rejects retracts
                 if (!e_best && gsize > 1) continue;  // all records ignored
                 if (drop_delete && !(v_best & 1)) continue;
                 if (pass == 1) {

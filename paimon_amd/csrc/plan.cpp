@@ -251,7 +251,8 @@ struct pmh_plan_t {
     bool drop_delete = true;
     bool ignore_delete = false;
     bool host_output = false;
-    bool pu = false;  // partial-update merge engine
+    bool pu = false;         // partial-update merge engine
+    bool first_row = false;  // first-row merge engine
     std::vector<pmh::Section> sections;
     size_t cur_section = 0;
     int64_t rows_in_total = 0;
@@ -1284,6 +1285,8 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         if (engine == "partial-update") {
             plan->pu = true;  // INSERT-only v1 (PartialUpdateMergeFunction
                               // rejects retracts by default, :170-186)
+        } else if (engine == "first-row") {
+            plan->first_row = true;  // FirstRowMergeFunction.java:32-77
         } else if (engine != "deduplicate") {
             set_error("merge engine '%s' not on the GPU path (deduplicate | "
                       "partial-update; aggregation/first-row are later "
@@ -1488,7 +1491,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     if (e != hipSuccess) return fail("partition", e);
     (void)hipEventRecord(ev[2], st);
     int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0) |
-                (p->pu ? 4 : 0);
+                (p->pu ? 4 : 0) | (p->first_row ? 8 : 0);
     if (const char *ab = getenv("PMH_ABLATE"))  // profiling-only phase knob
         flags |= (atoi(ab) & 0xf) << 8;
     e = pmh_launch_merge_tiles(sec.key_cols, sec.seq_cols, sec.kind_cols,
@@ -1531,6 +1534,12 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                   "(reference default also rejects them, "
                   "PartialUpdateMergeFunction.java:170-186); configure "
                   "ignore-delete/sequence-groups paths are later rounds");
+        return -1;
+    }
+    if (err_word & 2) {
+        set_error("first-row merge engine cannot accept DELETE/UPDATE_BEFORE "
+                  "records; configure 'ignore-delete' to skip them "
+                  "(FirstRowMergeFunction.java:49-59)");
         return -1;
     }
 

@@ -335,3 +335,40 @@ class TestConcurrentPlans:
             assert (gb["_KEY_k"] == eb).all()
             pa_.close()
             pb_.close()
+
+
+class TestFirstRowEngine:
+    def _run(self, tmp_path, runs, **kw):
+        from oracle import merge_first_row_model
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        r, w = merge_first_row_model(runs, **kw)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(len(runs[0]["values"]) - 1),
+                               merge_engine="first-row", **kw) as plan:
+                got = _read_all_batches(plan)
+        ek = np.array([runs[a]["key"][b] for a, b in zip(r, w)], np.int64)
+        es = np.array([runs[a]["seq"][b] for a, b in zip(r, w)], np.int64)
+        assert (got["_KEY_k"] == ek).all()
+        assert (got["_SEQUENCE_NUMBER"] == es).all()
+
+    def test_first_row_insert_only(self, tmp_path):
+        runs = gen_runs_dedup(6, 25_000, n_value_cols=2, seed=63,
+                              delete_frac=0.0)
+        self._run(tmp_path, runs)
+
+    def test_first_row_ignore_delete(self, tmp_path):
+        runs = gen_runs_dedup(5, 15_000, n_value_cols=2, seed=64,
+                              delete_frac=0.25)
+        self._run(tmp_path, runs, ignore_delete=True)
+
+    def test_first_row_rejects_retracts(self, tmp_path):
+        runs = gen_runs_dedup(3, 8_000, n_value_cols=1, seed=65,
+                              delete_frac=0.3)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(1),
+                               merge_engine="first-row") as plan:
+                with pytest.raises(RuntimeError, match="first-row"):
+                    plan.read_next()
