@@ -196,6 +196,97 @@ def partial_update_model(runs, drop_delete=True):
     }
 
 
+def aggregation_model(runs, aggs=None):
+    """Model of AggregateMergeFunction (AggregateMergeFunction.java:82-125)
+    for INSERT-only streams. aggs is a list of aggregate-function names, one
+    per value column (default last_non_null_value, :197-203). Per key group
+    in ascending (seq, isAdd) order each column folds through its
+    FieldAggregator; result seq = last member's, kind = INSERT. Singleton
+    groups bypass the merge function (ReducerMergeFunctionWrapper.java:53-73).
+
+    Follows Java numerics: int sums wrap at 64 bits here (Java's addExact
+    overflow check is not modelled); float/double sums accumulate in the
+    column's own precision in merge order; max/min compare like
+    Float.compare/Double.compare (IEEE total order)."""
+    key, seq, kind, run, row = _sorted_stream(runs)
+    assert _kind_is_add(kind).all(), "aggregation_model: INSERT-only streams"
+    n = len(key)
+    n_cols = len(runs[0]["values"]) if runs else 0
+    if aggs is None:
+        aggs = ["last_non_null_value"] * n_cols
+    if n == 0:
+        return {"key": key, "seq": seq, "kind": kind,
+                "values": [np.empty(0, r.dtype) for r in
+                           (runs[0]["values"] if runs else [])],
+                "valid": [np.empty(0, bool) for _ in range(n_cols)]}
+    has_valid = "valid" in runs[0]
+
+    def fold(agg, vals, msk):
+        # one group's members, merge order; returns (value, valid)
+        if len(vals) == 1:  # ReducerMergeFunctionWrapper bypass
+            return vals[-1], bool(msk[-1])
+        if agg == "last_value":
+            return vals[-1], bool(msk[-1])
+        if agg == "first_value":
+            return vals[0], bool(msk[0])
+        if agg in ("first_non_null_value", "first_not_null_value"):
+            nz = np.nonzero(msk)[0]
+            return (vals[nz[0]], True) if len(nz) else (vals[0], False)
+        if agg == "last_non_null_value":
+            nz = np.nonzero(msk)[0]
+            return (vals[nz[-1]], True) if len(nz) else (vals[0], False)
+        nz = np.nonzero(msk)[0]
+        if len(nz) == 0:
+            return vals[0], False
+        v = vals[nz]
+        if agg == "sum":
+            if v.dtype.kind == "f":
+                acc = v.dtype.type(0)
+                for x in v:  # sequential, column precision — as Java does
+                    acc = v.dtype.type(acc + x)
+                return acc, True
+            return np.sum(v.astype(np.int64)), True
+        if v.dtype.kind == "f":  # total-order max/min (Float.compare)
+            ib = v.astype(v.dtype).view(
+                np.int32 if v.dtype.itemsize == 4 else np.int64)
+            width = 8 * v.dtype.itemsize
+            ordv = np.where(ib < 0, ~ib.astype(np.uint64) & ((1 << width) - 1),
+                            ib.astype(np.uint64) | (1 << (width - 1)))
+            pick = np.argmax(ordv) if agg == "max" else np.argmin(ordv)
+        else:
+            pick = np.argmax(v) if agg == "max" else np.argmin(v)
+        return v[pick], True
+
+    head = np.empty(n, dtype=bool)
+    head[0] = True
+    head[1:] = key[1:] != key[:-1]
+    starts = np.nonzero(head)[0]
+    ends = np.append(starts[1:], n)
+    cols = [np.concatenate([r["values"][c] for r in runs]) for c in
+            range(n_cols)]
+    msks = [np.concatenate([r["valid"][c] for r in runs]) if has_valid
+            else np.ones(len(cols[c]), dtype=bool) for c in range(n_cols)]
+    order = np.lexsort((
+        _kind_is_add(np.concatenate([r["kind"] for r in runs])).astype(np.int8),
+        np.concatenate([r["seq"] for r in runs]),
+        np.concatenate([r["key"] for r in runs])))
+    cols = [c[order] for c in cols]
+    msks = [m[order] for m in msks]
+    out_vals = [np.zeros(len(starts), dtype=c.dtype) for c in cols]
+    out_valid = [np.zeros(len(starts), dtype=bool) for _ in cols]
+    for g, (s, e) in enumerate(zip(starts, ends)):
+        for c in range(n_cols):
+            v, ok = fold(aggs[c], cols[c][s:e], msks[c][s:e])
+            if ok:
+                out_vals[c][g] = v
+            out_valid[c][g] = ok
+    return {
+        "key": key[ends - 1], "seq": seq[ends - 1],
+        "kind": np.zeros(len(starts), dtype=np.int8),
+        "values": out_vals, "valid": out_valid,
+    }
+
+
 def rle_bp_decode(data: bytes, bit_width: int, num_values: int):
     """C restatement of the Parquet RLE/bit-packed hybrid decoder."""
     lib = _get_lib()

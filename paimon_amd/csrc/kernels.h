@@ -135,6 +135,31 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                               void *const *out_ptrs,
                               uint8_t *const *out_valid, hipStream_t stream);
 
+// Aggregation emit: per owned group, fold members in ascending (seq, isAdd)
+// order through per-column FieldAggregators (AggregateMergeFunction.java:
+// 82-125; default last_non_null_value, :201). col_agg holds one PMH_AGG_*
+// code per output column. INSERT-only streams in v1 (retracts are detected
+// by k_merge_tiles in PU mode and fail the read).
+enum {
+    PMH_AGG_LAST_NON_NULL = 0,  // FieldLastNonNullValueAgg (the default)
+    PMH_AGG_LAST_VALUE = 1,     // FieldLastValueAgg
+    PMH_AGG_FIRST_VALUE = 2,    // FieldFirstValueAgg
+    PMH_AGG_FIRST_NON_NULL = 3, // FieldFirstNonNullValueAgg
+    PMH_AGG_SUM = 4,            // FieldSumAgg
+    PMH_AGG_MAX = 5,            // FieldMaxAgg
+    PMH_AGG_MIN = 6,            // FieldMinAgg
+};
+hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
+                               const uint8_t *col_nullable,
+                               const uint8_t *col_agg, int n_cols, int k,
+                               int seq_col, int kind_col,
+                               const uint32_t *members,
+                               const uint16_t *group_start,
+                               const int64_t *tile_offsets, int64_t n_tiles,
+                               int64_t tile_rows, const int64_t *total_out,
+                               void *const *out_ptrs,
+                               uint8_t *const *out_valid, hipStream_t stream);
+
 hipError_t pmh_launch_dict_gather(const int32_t *ids, const void *dict,
                                   int64_t n, void *out, int esize,
                                   hipStream_t stream);

@@ -708,6 +708,17 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                         ((int8_t *)out_ptrs[c])[idx[x]] = (int8_t)v[x];
                 break;
             }
+            case 2: {  // INT16 output from INT32-stored parquet SMALLINT
+                int32_t v[R];
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    v[x] = col_load<int32_t>(cols[run[x] * n_cols + c], row[x]);
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    if (x < nr)
+                        ((int16_t *)out_ptrs[c])[idx[x]] = (int16_t)v[x];
+                break;
+            }
             case 3:
             case 5: {
                 int32_t v[R];
@@ -828,6 +839,10 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                 ((int8_t *)out_ptrs[c])[i] =
                     ok ? (int8_t)col_load<int32_t>(dc, row) : 0;
                 break;
+            case 2:
+                ((int16_t *)out_ptrs[c])[i] =
+                    ok ? (int16_t)col_load<int32_t>(dc, row) : 0;
+                break;
             case 3:
             case 5:
                 ((int32_t *)out_ptrs[c])[i] =
@@ -838,6 +853,206 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                 ((int64_t *)out_ptrs[c])[i] =
                     ok ? col_load<int64_t>(dc, row) : 0;
                 break;
+            default: break;
+            }
+            if (out_valid[c]) out_valid[c][i] = ok;
+        }
+    }
+}
+
+// ------------------------------------------------------------ k_emit_agg
+//
+// Aggregation emit (AggregateMergeFunction.java:82-125): one output row per
+// owned group; each value column folds the group's members in ascending
+// (seq, isAdd) order through its FieldAggregator:
+//   last_non_null_value (default, :201)  — FieldLastNonNullValueAgg
+//   last_value / first_value / first_non_null_value
+//   sum  — FieldSumAgg (null skipped; int accumulates in 64-bit then
+//          truncates to the column width — Java's addExact overflow check
+//          is not replicated, documented in DESIGN.md; float/double add in
+//          the column's own precision, in merge order, matching Java)
+//   max / min — FieldMaxAgg/FieldMinAgg; floats compare in IEEE total order
+//          on the stored bits (= Float.compare/Double.compare up to
+//          non-canonical negative NaNs)
+// Result: seq = last member's sequenceNumber, kind = INSERT (getResult,
+// :119-125). Singleton groups bypass the merge function entirely
+// (ReducerMergeFunctionWrapper.java:53-73) and pass through unchanged.
+// v1 accepts INSERT-only streams; retracts were flagged by k_merge_tiles.
+__device__ inline uint32_t f32_ord(int32_t b) {
+    return b < 0 ? ~(uint32_t)b : ((uint32_t)b | 0x80000000u);
+}
+__device__ inline uint64_t f64_ord(int64_t b) {
+    return b < 0 ? ~(uint64_t)b : ((uint64_t)b | 0x8000000000000000ull);
+}
+
+__global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
+                           const uint8_t *col_nullable,
+                           const uint8_t *col_agg, int n_cols, int k,
+                           int seq_col, int kind_col, const uint32_t *members,
+                           const uint16_t *group_start,
+                           const int64_t *tile_offsets, int64_t n_tiles,
+                           int64_t tile_rows, const int64_t *total_out,
+                           void *const *out_ptrs, uint8_t *const *out_valid) {
+    const int64_t total = *total_out;
+    const int64_t per_block =
+        (total + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
+    const int64_t slice_lo = (int64_t)blockIdx.x * per_block;
+    const int64_t slice_hi =
+        slice_lo + per_block < total ? slice_lo + per_block : total;
+    for (int64_t i = slice_lo + threadIdx.x; i < slice_hi; i += blockDim.x) {
+        int64_t lo = 0, hi = n_tiles - 1;
+        while (lo < hi) {
+            int64_t mid = (lo + hi + 1) >> 1;
+            if (tile_offsets[mid] <= i) lo = mid;
+            else hi = mid - 1;
+        }
+        int64_t t = lo;
+        int64_t g = i - tile_offsets[t];
+        const uint16_t *gs = &group_start[t * (tile_rows + 1)];
+        const uint32_t *mem = &members[t * (tile_rows + PMH_MAX_RUNS)];
+        const int32_t ms = gs[g], me = gs[g + 1];
+        const int gn = me - ms;
+        const uint32_t last = mem[me - 1];
+        const int lrun = last >> 28;
+        const int64_t lrow = last & 0x0fffffff;
+        for (int c = 0; c < n_cols; c++) {
+            if (c == kind_col) {
+                ((int8_t *)out_ptrs[c])[i] = 0;  // RowKind.INSERT
+                continue;
+            }
+            const int dt = col_dtype[c];
+            const int agg = gn == 1 ? PMH_AGG_LAST_VALUE : col_agg[c];
+            int64_t run = lrun, row = lrow;
+            uint8_t ok = 1;
+            int64_t bits = 0;  // result payload (raw stored bits / int value)
+            bool direct = true;  // bits not set; load from (run,row) at store
+            switch (agg) {
+            case PMH_AGG_LAST_VALUE:
+                // last member as-is, its own validity (also the singleton
+                // ReducerMergeFunctionWrapper bypass)
+                if (col_nullable[c]) {
+                    const DevCol &dc = cols[lrun * n_cols + c];
+                    ok = dc.valid0 ? ((const uint8_t *)dc.valid0)[lrow] : 1;
+                }
+                break;
+            case PMH_AGG_FIRST_VALUE: {
+                run = mem[ms] >> 28;
+                row = mem[ms] & 0x0fffffff;
+                if (col_nullable[c]) {
+                    const DevCol &dc = cols[run * n_cols + c];
+                    ok = dc.valid0 ? ((const uint8_t *)dc.valid0)[row] : 1;
+                }
+                break;
+            }
+            case PMH_AGG_LAST_NON_NULL:
+                if (col_nullable[c]) {
+                    ok = 0;
+                    for (int32_t x = me - 1; !ok && x >= ms; x--) {
+                        uint32_t m = mem[x];
+                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                        uint8_t v =
+                            dc.valid0
+                                ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                                : 1;
+                        if (v) {
+                            run = m >> 28;
+                            row = m & 0x0fffffff;
+                            ok = 1;
+                        }
+                    }
+                }
+                break;
+            case PMH_AGG_FIRST_NON_NULL:
+                if (col_nullable[c]) {
+                    ok = 0;
+                    for (int32_t x = ms; !ok && x < me; x++) {
+                        uint32_t m = mem[x];
+                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                        uint8_t v =
+                            dc.valid0
+                                ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                                : 1;
+                        if (v) {
+                            run = m >> 28;
+                            row = m & 0x0fffffff;
+                            ok = 1;
+                        }
+                    }
+                } else {
+                    run = mem[ms] >> 28;
+                    row = mem[ms] & 0x0fffffff;
+                }
+                break;
+            default: {  // SUM / MAX / MIN: full fold, null inputs skipped
+                direct = false;
+                ok = 0;
+                int64_t iacc = 0;
+                float facc = 0.f;
+                double dacc = 0.0;
+                for (int32_t x = ms; x < me; x++) {
+                    uint32_t m = mem[x];
+                    const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                    const int64_t r = m & 0x0fffffff;
+                    if (col_nullable[c] && dc.valid0 &&
+                        !((const uint8_t *)dc.valid0)[r])
+                        continue;
+                    int64_t vb = (dt == 4 || dt == 6)
+                                     ? col_load<int64_t>(dc, r)
+                                     : (int64_t)col_load<int32_t>(dc, r);
+                    if (!ok) {
+                        ok = 1;
+                        if (agg == PMH_AGG_SUM && dt == 5)
+                            facc = __int_as_float((int32_t)vb);
+                        else if (agg == PMH_AGG_SUM && dt == 6)
+                            dacc = __longlong_as_double(vb);
+                        else
+                            iacc = vb;  // raw bits for float max/min
+                        continue;
+                    }
+                    if (agg == PMH_AGG_SUM) {
+                        if (dt == 5) facc += __int_as_float((int32_t)vb);
+                        else if (dt == 6) dacc += __longlong_as_double(vb);
+                        else iacc += vb;
+                    } else {
+                        bool take;
+                        if (dt == 5) {
+                            uint32_t a = f32_ord((int32_t)iacc);
+                            uint32_t b = f32_ord((int32_t)vb);
+                            take = agg == PMH_AGG_MAX ? b > a : b < a;
+                        } else if (dt == 6) {
+                            uint64_t a = f64_ord(iacc);
+                            uint64_t b = f64_ord(vb);
+                            take = agg == PMH_AGG_MAX ? b > a : b < a;
+                        } else {
+                            take = agg == PMH_AGG_MAX ? vb > iacc : vb < iacc;
+                        }
+                        if (take) iacc = vb;
+                    }
+                }
+                if (agg == PMH_AGG_SUM && dt == 5)
+                    bits = (int64_t)__float_as_int(facc);
+                else if (agg == PMH_AGG_SUM && dt == 6)
+                    bits = __double_as_longlong(dacc);
+                else
+                    bits = iacc;
+                break;
+            }
+            }
+            if (direct) {
+                const DevCol &dc = cols[run * n_cols + c];
+                if (ok)
+                    bits = (dt == 4 || dt == 6)
+                               ? col_load<int64_t>(dc, row)
+                               : (int64_t)col_load<int32_t>(dc, row);
+            }
+            if (!ok) bits = 0;
+            switch (dt) {
+            case 1: ((int8_t *)out_ptrs[c])[i] = (int8_t)bits; break;
+            case 2: ((int16_t *)out_ptrs[c])[i] = (int16_t)bits; break;
+            case 3:
+            case 5: ((int32_t *)out_ptrs[c])[i] = (int32_t)bits; break;
+            case 4:
+            case 6: ((int64_t *)out_ptrs[c])[i] = bits; break;
             default: break;
             }
             if (out_valid[c]) out_valid[c][i] = ok;
@@ -1176,6 +1391,24 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                        col_dtype, col_nullable, n_cols, k, seq_col, kind_col,
                        members, group_start, tile_offsets, n_tiles, tile_rows,
                        total_out, out_ptrs, out_valid);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
+                               const uint8_t *col_nullable,
+                               const uint8_t *col_agg, int n_cols, int k,
+                               int seq_col, int kind_col,
+                               const uint32_t *members,
+                               const uint16_t *group_start,
+                               const int64_t *tile_offsets, int64_t n_tiles,
+                               int64_t tile_rows, const int64_t *total_out,
+                               void *const *out_ptrs,
+                               uint8_t *const *out_valid,
+                               hipStream_t stream) {
+    hipLaunchKernelGGL(k_emit_agg, dim3(2048), dim3(256), 0, stream, cols,
+                       col_dtype, col_nullable, col_agg, n_cols, k, seq_col,
+                       kind_col, members, group_start, tile_offsets, n_tiles,
+                       tile_rows, total_out, out_ptrs, out_valid);
     return hipGetLastError();
 }
 

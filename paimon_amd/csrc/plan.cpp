@@ -253,6 +253,8 @@ struct pmh_plan_t {
     bool host_output = false;
     bool pu = false;         // partial-update merge engine
     bool first_row = false;  // first-row merge engine
+    bool agg = false;        // aggregation merge engine (uses PU member lists)
+    uint8_t *col_agg_dev = nullptr;  // per-column PMH_AGG_* codes
     std::vector<pmh::Section> sections;
     size_t cur_section = 0;
     int64_t rows_in_total = 0;
@@ -1287,10 +1289,14 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                               // rejects retracts by default, :170-186)
         } else if (engine == "first-row") {
             plan->first_row = true;  // FirstRowMergeFunction.java:32-77
+        } else if (engine == "aggregation") {
+            // AggregateMergeFunction.java:50-125 — per-field aggregators
+            // over the same grouped stream as partial-update
+            plan->pu = true;
+            plan->agg = true;
         } else if (engine != "deduplicate") {
             set_error("merge engine '%s' not on the GPU path (deduplicate | "
-                      "partial-update; aggregation/first-row are later "
-                      "rounds)",
+                      "partial-update | aggregation | first-row)",
                       engine.c_str());
             return nullptr;
         }
@@ -1421,6 +1427,52 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             return nullptr;
         }
         for (auto &cs : plan->cols) plan->col_names.push_back(cs.name);
+        if (plan->agg) {
+            // "aggregations": {"col": "sum", ...}; unnamed value columns get
+            // last_non_null_value (AggregateMergeFunction.java:197-203);
+            // "first_not_null_value" is the legacy alias
+            // (FieldFirstNonNullValueAggLegacyFactory.java:29)
+            std::vector<uint8_t> ca(n_cols, PMH_AGG_LAST_NON_NULL);
+            auto code_of = [](const std::string &s) -> int {
+                if (s == "last_non_null_value") return PMH_AGG_LAST_NON_NULL;
+                if (s == "last_value") return PMH_AGG_LAST_VALUE;
+                if (s == "first_value") return PMH_AGG_FIRST_VALUE;
+                if (s == "first_non_null_value" ||
+                    s == "first_not_null_value")
+                    return PMH_AGG_FIRST_NON_NULL;
+                if (s == "sum") return PMH_AGG_SUM;
+                if (s == "max") return PMH_AGG_MAX;
+                if (s == "min") return PMH_AGG_MIN;
+                return -1;
+            };
+            const int first_val = plan->n_key_cols + 2;
+            for (const auto &kv : j["aggregations"].obj) {
+                int idx = -1;
+                for (int c = first_val; c < n_cols; c++)
+                    if (plan->cols[c].name == kv.first) idx = c;
+                if (idx < 0) {
+                    set_error("aggregations: '%s' is not a value column",
+                              kv.first.c_str());
+                    return nullptr;
+                }
+                int code = code_of(kv.second.as_str());
+                if (code < 0) {
+                    set_error("aggregate function '%s' not on the GPU path "
+                              "(v1: sum, max, min, last_value, first_value, "
+                              "last_non_null_value, first_non_null_value)",
+                              kv.second.as_str().c_str());
+                    return nullptr;
+                }
+                ca[idx] = (uint8_t)code;
+            }
+            plan->col_agg_dev = (uint8_t *)plan->bufs.alloc(n_cols);
+            if (!plan->col_agg_dev ||
+                hipMemcpy(plan->col_agg_dev, ca.data(), n_cols,
+                          hipMemcpyHostToDevice) != hipSuccess) {
+                set_error("H2D of aggregator codes failed");
+                return nullptr;
+            }
+        }
     } catch (const std::exception &e) {
         set_error("plan parse: %s", e.what());
         return nullptr;
@@ -1505,7 +1557,14 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                               sec.total_dev, st);
     if (e != hipSuccess) return fail("scan_tiles", e);
     (void)hipEventRecord(ev[4], st);
-    if (p->pu) {
+    if (p->agg) {
+        e = pmh_launch_emit_agg(
+            sec.all_cols, p->col_dtype_dev, p->col_nullable_dev,
+            p->col_agg_dev, n_cols, k, p->n_key_cols, p->n_key_cols + 1,
+            sec.winners, sec.group_start, sec.tile_offsets, sec.n_tiles,
+            PMH_TILE_ROWS, sec.total_dev, p->out_ptrs_dev, p->out_valid_dev,
+            st);
+    } else if (p->pu) {
         e = pmh_launch_emit_pu(
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,
             p->n_key_cols, p->n_key_cols + 1, sec.winners, sec.group_start,
@@ -1530,10 +1589,12 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     e = hipStreamSynchronize(st);
     if (e != hipSuccess) return fail("stream sync", e);
     if (err_word & 1) {
-        set_error("partial-update with retract records is not supported "
-                  "(reference default also rejects them, "
-                  "PartialUpdateMergeFunction.java:170-186); configure "
-                  "ignore-delete/sequence-groups paths are later rounds");
+        set_error("%s with retract records is not supported "
+                  "(the reference default also rejects them, "
+                  "PartialUpdateMergeFunction.java:170-186); "
+                  "ignore-delete/sequence-group/retract-aggregator paths "
+                  "are later rounds",
+                  p->agg ? "aggregation" : "partial-update");
         return -1;
     }
     if (err_word & 2) {

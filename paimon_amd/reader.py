@@ -142,7 +142,7 @@ class MergeReadPlan:
 
     def __init__(self, session: Session, files, key_cols, value_cols,
                  merge_engine="deduplicate", drop_delete=True,
-                 ignore_delete=False, output="host"):
+                 ignore_delete=False, output="host", aggregations=None):
         self.lib = session.lib
         desc = {
             "key_cols": key_cols,
@@ -153,6 +153,10 @@ class MergeReadPlan:
             "output": output,
             "files": files,
         }
+        if aggregations:
+            # fields.<name>.aggregate-function (CoreOptions FIELDS_PREFIX);
+            # unnamed columns default to last_non_null_value
+            desc["aggregations"] = dict(aggregations)
         self.h = self.lib.pmh_plan_create(session.h,
                                           json.dumps(desc).encode())
         if not self.h:

@@ -1,0 +1,141 @@
+"""GPU parity for the aggregation merge engine (AggregateMergeFunction,
+per-field FieldAggregators) against the oracle model — which is itself pinned
+to the reference's ported semantics in test_oracle_merge.TestAggregationModel."""
+
+import numpy as np
+import pytest
+
+from oracle import aggregation_model
+from paimon_amd import Session, MergeReadPlan, file_descs_from_metas
+from paimon_amd.datagen import gen_runs_partial_update, write_runs
+
+pytestmark = pytest.mark.gpu
+
+KEY_COLS = [{"name": "_KEY_k", "type": "int64"}]
+AGGS = ["sum", "max", "min", "last_value", "first_value",
+        "last_non_null_value", "first_non_null_value"]
+
+
+def _value_cols(n, t="int32"):
+    return ([{"name": "v_k", "type": "int64"}] +
+            [{"name": f"v_c{i}", "type": t} for i in range(n)])
+
+
+def _read_all(plan):
+    got = {}
+    while True:
+        b = plan.read_next()
+        if b is None:
+            break
+        for kk, v in b.items():
+            got.setdefault(kk, []).append(v.copy())
+    return {kk: np.concatenate(v) for kk, v in got.items()}
+
+
+def _check(got, exp, names):
+    assert (got["_KEY_k"] == exp["key"]).all()
+    assert (got["_SEQUENCE_NUMBER"] == exp["seq"]).all()
+    assert (got["_VALUE_KIND"] == exp["kind"]).all()
+    for c, nm in enumerate(names):
+        ev, evalid = exp["values"][c], exp["valid"][c]
+        gvalid = got.get(nm + "#valid")
+        if gvalid is None:
+            gvalid = np.ones(len(got[nm]), dtype=bool)
+        assert (gvalid == evalid).all(), nm
+        if ev.dtype.kind == "f":
+            # sums run in the same member order on both sides: bit-exact
+            assert (got[nm][evalid].view(np.uint8).reshape(evalid.sum(), -1)
+                    == ev[evalid].view(np.uint8).reshape(evalid.sum(), -1)
+                    ).all(), nm
+        else:
+            assert (got[nm][evalid] == ev[evalid]).all(), nm
+
+
+class TestAggregationEngine:
+    def _run(self, tmp_path, runs, aggs_by_col, file_format="parquet",
+             col_type="int32"):
+        metas = write_runs(runs, str(tmp_path), compression="NONE",
+                           file_format=file_format)
+        n_vals = len(runs[0]["values"])
+        names = ["v_k"] + [f"v_c{i}" for i in range(n_vals - 1)]
+        exp = aggregation_model(
+            runs, ["last_non_null_value"] +
+            [aggs_by_col.get(nm, "last_non_null_value") for nm in names[1:]])
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(n_vals - 1, col_type),
+                               merge_engine="aggregation",
+                               aggregations=aggs_by_col) as plan:
+                got = _read_all(plan)
+        _check(got, exp, names)
+
+    def test_all_aggregators_int32(self, tmp_path):
+        runs = gen_runs_partial_update(4, 30_000, n_value_cols=7, seed=91,
+                                       update_frac=0.4, update_cols=4)
+        for r in runs:  # bound so int32 sums cannot overflow
+            for c in range(1, 8):
+                r["values"][c] = (r["values"][c] % 10_000).astype(np.int32)
+        aggs = {f"v_c{i}": a for i, a in enumerate(AGGS)}
+        self._run(tmp_path, runs, aggs)
+
+    def test_default_is_last_non_null(self, tmp_path):
+        # no aggregations map at all -> every column last_non_null_value
+        runs = gen_runs_partial_update(3, 20_000, n_value_cols=4, seed=92)
+        self._run(tmp_path, runs, {})
+
+    def test_orc_aggregation(self, tmp_path):
+        runs = gen_runs_partial_update(4, 25_000, n_value_cols=4, seed=93,
+                                       update_frac=0.3, update_cols=2)
+        for r in runs:
+            for c in range(1, 5):
+                r["values"][c] = (r["values"][c] % 10_000).astype(np.int32)
+        aggs = {"v_c0": "sum", "v_c1": "max", "v_c2": "min",
+                "v_c3": "first_non_null_value"}
+        self._run(tmp_path, runs, aggs, file_format="orc")
+
+    def test_float_sum_max_min(self, tmp_path):
+        rng = np.random.default_rng(94)
+        runs = gen_runs_partial_update(4, 20_000, n_value_cols=3, seed=95,
+                                       update_frac=0.5, update_cols=2)
+        for r in runs:
+            n = len(r["key"])
+            for c in (1, 2, 3):
+                r["values"][c] = rng.standard_normal(n).astype(np.float64)
+        aggs = {"v_c0": "sum", "v_c1": "max", "v_c2": "min"}
+        self._run(tmp_path, runs, aggs, col_type="float64")
+
+    def test_float32_sum(self, tmp_path):
+        rng = np.random.default_rng(96)
+        runs = gen_runs_partial_update(3, 15_000, n_value_cols=2, seed=97,
+                                       update_frac=0.5, update_cols=1)
+        for r in runs:
+            n = len(r["key"])
+            for c in (1, 2):
+                r["values"][c] = rng.standard_normal(n).astype(np.float32)
+        self._run(tmp_path, runs, {"v_c0": "sum", "v_c1": "max"},
+                  col_type="float32")
+
+    def test_retract_rejected(self, tmp_path):
+        runs = gen_runs_partial_update(2, 5_000, n_value_cols=2, seed=98)
+        # turn some rows of run 1 into DELETEs
+        runs[1]["kind"][::7] = 3
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(2), merge_engine="aggregation",
+                               aggregations={"v_c0": "sum"}) as plan:
+                with pytest.raises(RuntimeError, match="aggregation"):
+                    _read_all(plan)
+
+    def test_unknown_aggregator_rejected(self, tmp_path):
+        runs = gen_runs_partial_update(1, 100, n_value_cols=2, seed=99)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="not on the GPU path"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(2), merge_engine="aggregation",
+                              aggregations={"v_c0": "hll_sketch"})
+            with pytest.raises(RuntimeError, match="not a value column"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(2), merge_engine="aggregation",
+                              aggregations={"nope": "sum"})

@@ -49,9 +49,11 @@ def load_runs_npz(d, prefix="in"):
     return runs
 
 
-def random_runs(rng, n_runs, max_rows, key_space, delete_p=0.2, n_value_cols=0):
+def random_runs(rng, n_runs, max_rows, key_space, delete_p=0.2, n_value_cols=0,
+                null_p=0.0):
     """Random sorted runs obeying LSM invariants (sorted, unique keys per
-    run, globally unique seqs)."""
+    run, globally unique seqs). null_p > 0 adds 'valid' masks (col 0, the
+    key copy, stays non-null)."""
     total_cap = n_runs * max_rows
     seqpool = rng.permutation(total_cap * 2).astype(np.int64)
     off = 0
@@ -67,6 +69,9 @@ def random_runs(rng, n_runs, max_rows, key_space, delete_p=0.2, n_value_cols=0):
             r["values"] = [keys.copy()] + [
                 rng.integers(-2**31, 2**31, n).astype(np.int32)
                 for _ in range(n_value_cols)]
+            if null_p > 0:
+                r["valid"] = [np.ones(n, dtype=bool)] + [
+                    rng.random(n) >= null_p for _ in range(n_value_cols)]
         off += n
         runs.append(r)
     return runs
