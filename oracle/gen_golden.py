@@ -98,6 +98,19 @@ def main():
     np.savez_compressed(os.path.join(GOLDEN_DIR, "dedup_edge.npz"), **d)
     print("edge golden:", len(d["dedup_dd1_key"]), "records")
 
+    # Case D: first-row, 5 runs x 1200 rows, INSERT-only (the pypaimon
+    # FirstRowMergeFunction raises on retracts, like the Java reference)
+    runs = gen_runs_dedup(5, 1200, n_value_cols=2, seed=23, delete_frac=0.0)
+    d = {}
+    _save_runs(d, "in", runs)
+    out = merge_with_pypaimon(runs, "first-row", drop_delete=True)
+    d["fr_key"] = np.array([o[0] for o in out], np.int64)
+    d["fr_seq"] = np.array([o[1] for o in out], np.int64)
+    d["fr_kind"] = np.array([o[2] for o in out], np.int8)
+    d["fr_vals"] = np.array([o[3] for o in out], np.int64)
+    np.savez_compressed(os.path.join(GOLDEN_DIR, "first_row_5x1200.npz"), **d)
+    print("first-row golden:", len(out), "records")
+
 
 if __name__ == "__main__":
     main()

@@ -146,6 +146,10 @@ def merge_with_pypaimon(runs, merge_function_name="deduplicate",
         from pypaimon.read.reader.partial_update_merge_function import \
             PartialUpdateMergeFunction
         mf = PartialUpdateMergeFunction(key_arity, n_value_cols)
+    elif merge_function_name == "first-row":
+        from pypaimon.read.reader.first_row_merge_function import \
+            FirstRowMergeFunction
+        mf = FirstRowMergeFunction(ignore_delete=False)
     else:
         raise ValueError(merge_function_name)
 
