@@ -1,0 +1,97 @@
+#!/usr/bin/env python3
+"""Collect + reduce rocprofv3 PMC passes into profiles/pmc_traffic.json.
+
+Runs ON THE GPU BOX (via scripts/pmc_traffic.sh). Two separate --pmc passes
+(FETCH_SIZE, WRITE_SIZE — never combined with trace domains) over the default
+bench command; per-dispatch averages per kernel are corrected per
+MI355X_MICROARCH.md §HBM:
+  - wide coalesced streaming reads are counted at 1/2 on gfx950 -> x2
+    (applies to k_merge_tiles' dwordx4 LDS staging; validated here against
+    its exact algorithmic byte count);
+  - 64B-granule gather reads are counted at request granularity -> raw
+    (applies to k_emit's column gathers; its WRITE_SIZE raw has been
+    validated against the exact algorithmic output bytes).
+The result is the `roofline.traffic` calibration bench.py reports for the
+default workload.
+"""
+import csv
+import glob
+import json
+import sys
+
+# kernel-name substring -> (bench kernel key, fetch correction factor, note)
+KERNELS = {
+    "k_merge_tiles": ("merge", 2.0,
+                      "FETCH x2 (wide coalesced staging, guide §HBM) + "
+                      "WRITE raw"),
+    "k_emit": ("emit", 1.0,
+               "FETCH raw (64B-granule gathers counted exactly) + WRITE raw "
+               "(validated = algorithmic output bytes)"),
+}
+
+
+def per_dispatch(pattern, counter):
+    tot, disp = {}, {}
+    for f in glob.glob(pattern):
+        with open(f) as fh:
+            for row in csv.DictReader(fh):
+                if row.get("Counter_Name") != counter:
+                    continue
+                kn = row["Kernel_Name"]
+                tot[kn] = tot.get(kn, 0.0) + float(row["Counter_Value"])
+                disp.setdefault(kn, set()).add(
+                    (row.get("Dispatch_Id"), row.get("Correlation_Id")))
+    return {k: tot[k] / len(disp[k]) for k in tot}
+
+
+def match(averages):
+    out = {}
+    for kn, v in averages.items():
+        for sub, (key, corr, note) in KERNELS.items():
+            if sub in kn and "_pu" not in kn and "_agg" not in kn:
+                out[key] = (v, corr, note)
+    return out
+
+
+def main():
+    prof_dir = sys.argv[1]
+    runs, rows, vals = (int(x) for x in sys.argv[2:5])
+    fetch = match(per_dispatch(f"{prof_dir}/*pmc_fetch*counter*.csv",
+                               "FETCH_SIZE"))
+    write = match(per_dispatch(f"{prof_dir}/*pmc_write*counter*.csv",
+                               "WRITE_SIZE"))
+    # FETCH_SIZE/WRITE_SIZE report kilobytes
+    KB = 1024.0
+    kernels = {}
+    for key in fetch:
+        fv, fcorr, note = fetch[key]
+        wv = write.get(key, (0.0,))[0]
+        kernels[key] = {
+            "fetch_raw_bytes": fv * KB,
+            "write_raw_bytes": wv * KB,
+            "fetch_correction": fcorr,
+            "bytes_per_launch": fv * KB * fcorr + wv * KB,
+            "note": note,
+        }
+    # sanity: merge algorithmic read bytes = rows*(8+8+4)
+    checks = {}
+    if "merge" in kernels:
+        alg = runs * rows * 20.0
+        got = (kernels["merge"]["fetch_raw_bytes"] *
+               kernels["merge"]["fetch_correction"])
+        checks["merge_fetch_x2_vs_algorithmic_read"] = round(got / alg, 3)
+    out = {
+        "workload": {"runs": runs, "rows": rows, "vals": vals,
+                     "engine": "deduplicate", "format": "parquet",
+                     "compression": "NONE"},
+        "collected": ("rocprofv3 --pmc FETCH_SIZE / --pmc WRITE_SIZE, "
+                      "separate passes, per-dispatch averages; corrections "
+                      "per MI355X_MICROARCH.md §HBM"),
+        "sanity": checks,
+        "kernels": kernels,
+    }
+    print(json.dumps(out, indent=1))
+
+
+if __name__ == "__main__":
+    main()
