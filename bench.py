@@ -33,6 +33,29 @@ sys.path.insert(0, REPO)
 HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
 
 
+def pmc_traffic(dom, args):
+    """Measured HBM bytes/launch for the dominant kernel, from the committed
+    PMC calibration (profiles/pmc_traffic.json — rocprofv3 --pmc FETCH_SIZE /
+    WRITE_SIZE passes over this same command, corrections per
+    MI355X_MICROARCH.md §HBM; scripts/pmc_traffic.sh regenerates it). Only
+    valid at the calibrated workload; None otherwise."""
+    try:
+        with open(os.path.join(REPO, "profiles", "pmc_traffic.json")) as f:
+            cal = json.load(f)
+    except (OSError, ValueError):
+        return None
+    w = cal.get("workload", {})
+    if (w.get("runs") == args.runs and w.get("rows") == args.rows
+            and w.get("vals") == args.vals
+            and w.get("engine") == args.engine
+            and w.get("format") == args.file_format
+            and w.get("compression") == args.compression):
+        e = cal.get("kernels", {}).get(dom)
+        if e:
+            return e["bytes_per_launch"]
+    return None
+
+
 def get_args():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -250,7 +273,7 @@ def main():
             "peak": HBM_PEAK_GBS,
             "unit": "GB/s",
             "frac": round(achieved / HBM_PEAK_GBS, 4),
-            "traffic": None,
+            "traffic": pmc_traffic(dom, args),
         },
         "kernels_ms_per_step": {k: round(v, 3) for k, v in kms.items()},
         "pipeline_algorithmic_GBs": round(pipe_gbs, 1),
