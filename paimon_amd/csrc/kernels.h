@@ -125,6 +125,8 @@ hipError_t pmh_launch_level_scatter(const RleChunk *chunks, int64_t n_chunks,
 // PartialUpdate emit: per owned group, overlay non-null fields in ascending
 // (seq, isAdd) order (PartialUpdateMergeFunction.java:188-215 + Reducer
 // wrapper bypass). members/group_start written by k_merge_tiles in PU mode.
+// run_masks: device array of k per-run packed-validity pointers
+// (k_pack_valid), or null for the legacy per-column byte walk (>64 cols).
 hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                               const uint8_t *col_nullable, int n_cols, int k,
                               int seq_col, int kind_col,
@@ -132,8 +134,14 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                               const uint16_t *group_start,
                               const int64_t *tile_offsets, int64_t n_tiles,
                               int64_t tile_rows, const int64_t *total_out,
+                              uint64_t *const *run_masks,
                               void *const *out_ptrs,
                               uint8_t *const *out_valid, hipStream_t stream);
+
+// Pack one run's per-column validity bytes into u64 row masks (bit c =
+// column c non-null; columns without staged nulls contribute 1).
+hipError_t pmh_launch_pack_valid(const DevCol *cols, int n_cols, int64_t rows,
+                                 uint64_t *mask, hipStream_t stream);
 
 // Aggregation emit: per owned group, fold members in ascending (seq, isAdd)
 // order through per-column FieldAggregators (AggregateMergeFunction.java:

@@ -747,6 +747,30 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
     }
 }
 
+// ------------------------------------------------------------ k_pack_valid
+//
+// Pack one run's per-column validity bytes into a u64 bitmask per row
+// (bit c = column c non-null; columns with no staged nulls contribute 1).
+// The PU/aggregation emit then resolves EVERY column's nullability from one
+// mask load per group member instead of one byte load per (member, column) —
+// the per-column walk was a serial dependent-load chain over up to n_cols
+// scattered arrays and dominated the C3 emit.
+__global__ void k_pack_valid(const DevCol *cols, int n_cols, int64_t rows,
+                             uint64_t *mask) {
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < rows;
+         i += stride) {
+        uint64_t m = 0;
+        for (int c = 0; c < n_cols; c++) {
+            const DevCol &dc = cols[c];
+            uint64_t v =
+                dc.valid0 ? (uint64_t)((const uint8_t *)dc.valid0)[i] : 1u;
+            m |= (v & 1) << c;
+        }
+        mask[i] = m;
+    }
+}
+
 // ------------------------------------------------------------ k_emit_pu
 //
 // PartialUpdate emit: one output row per owned group; for each value column
@@ -756,13 +780,18 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
 // (latestSequenceNumber, :188); kind = INSERT (insert-only streams, v1).
 // Singleton groups return the record as-is (ReducerMergeFunctionWrapper
 // bypass) — identical to the overlay for INSERT records.
+// MASKS: per-row packed validity (k_pack_valid) available — one u64 load per
+// member resolves all columns; false = legacy per-column byte walk (>64
+// columns).
+template <bool MASKS>
 __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                           const uint8_t *col_nullable, int n_cols, int k,
                           int seq_col, int kind_col, const uint32_t *members,
                           const uint16_t *group_start,
                           const int64_t *tile_offsets, int64_t n_tiles,
                           int64_t tile_rows, const int64_t *total_out,
-                          void *const *out_ptrs, uint8_t *const *out_valid) {
+                          uint64_t *const *run_masks, void *const *out_ptrs,
+                          uint8_t *const *out_valid) {
     const int64_t total = *total_out;
     const int64_t per_block =
         (total + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
@@ -786,10 +815,18 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
         // few members (group size averages ~1.3), and re-reading the member
         // list from HBM per column dominated this kernel
         uint32_t mc[4];
+        uint64_t vm[4];
         const int gn = me - ms;
 #pragma unroll
         for (int x = 0; x < 4; x++)
             mc[x] = x < gn ? mem[me - 1 - x] : mem[me - 1];
+        if (MASKS) {
+#pragma unroll
+            for (int x = 0; x < 4; x++)
+                vm[x] = x < gn
+                            ? run_masks[mc[x] >> 28][mc[x] & 0x0fffffff]
+                            : 0;
+        }
         uint32_t last = mc[0];
         int lrun = last >> 28;
         int64_t lrow = last & 0x0fffffff;
@@ -806,10 +843,15 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                 for (int x = 0; x < 4; x++) {
                     if (ok || x >= gn) continue;
                     uint32_t m = mc[x];
-                    const DevCol &dc = cols[(m >> 28) * n_cols + c];
-                    uint8_t v = dc.valid0
-                                    ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
-                                    : 1;
+                    uint8_t v;
+                    if (MASKS) {
+                        v = (uint8_t)((vm[x] >> c) & 1);
+                    } else {
+                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                        v = dc.valid0
+                                ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                                : 1;
+                    }
                     if (v) {
                         run = m >> 28;
                         row = m & 0x0fffffff;
@@ -818,10 +860,16 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                 }
                 for (int32_t x = me - 5; !ok && x >= ms; x--) {
                     uint32_t m = mem[x];
-                    const DevCol &dc = cols[(m >> 28) * n_cols + c];
-                    uint8_t v = dc.valid0
-                                    ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
-                                    : 1;
+                    uint8_t v;
+                    if (MASKS) {
+                        v = (uint8_t)(
+                            (run_masks[m >> 28][m & 0x0fffffff] >> c) & 1);
+                    } else {
+                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                        v = dc.valid0
+                                ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                                : 1;
+                    }
                     if (v) {
                         run = m >> 28;
                         row = m & 0x0fffffff;
@@ -830,8 +878,12 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                 }
             } else if (col_nullable[c]) {
                 // singleton: the record passes through with its own validity
-                const DevCol &dc = cols[lrun * n_cols + c];
-                ok = dc.valid0 ? ((const uint8_t *)dc.valid0)[lrow] : 1;
+                if (MASKS) {
+                    ok = (uint8_t)((vm[0] >> c) & 1);
+                } else {
+                    const DevCol &dc = cols[lrun * n_cols + c];
+                    ok = dc.valid0 ? ((const uint8_t *)dc.valid0)[lrow] : 1;
+                }
             }
             const DevCol &dc = cols[run * n_cols + c];
             switch (col_dtype[c]) {
@@ -1385,12 +1437,28 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                               const uint16_t *group_start,
                               const int64_t *tile_offsets, int64_t n_tiles,
                               int64_t tile_rows, const int64_t *total_out,
+                              uint64_t *const *run_masks,
                               void *const *out_ptrs,
                               uint8_t *const *out_valid, hipStream_t stream) {
-    hipLaunchKernelGGL(k_emit_pu, dim3(2048), dim3(256), 0, stream, cols,
-                       col_dtype, col_nullable, n_cols, k, seq_col, kind_col,
-                       members, group_start, tile_offsets, n_tiles, tile_rows,
-                       total_out, out_ptrs, out_valid);
+    if (run_masks)
+        hipLaunchKernelGGL(k_emit_pu<true>, dim3(2048), dim3(256), 0, stream,
+                           cols, col_dtype, col_nullable, n_cols, k, seq_col,
+                           kind_col, members, group_start, tile_offsets,
+                           n_tiles, tile_rows, total_out, run_masks, out_ptrs,
+                           out_valid);
+    else
+        hipLaunchKernelGGL(k_emit_pu<false>, dim3(2048), dim3(256), 0, stream,
+                           cols, col_dtype, col_nullable, n_cols, k, seq_col,
+                           kind_col, members, group_start, tile_offsets,
+                           n_tiles, tile_rows, total_out, run_masks, out_ptrs,
+                           out_valid);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_pack_valid(const DevCol *cols, int n_cols, int64_t rows,
+                                 uint64_t *mask, hipStream_t stream) {
+    hipLaunchKernelGGL(k_pack_valid, dim3(1024), dim3(256), 0, stream, cols,
+                       n_cols, rows, mask);
     return hipGetLastError();
 }
 

@@ -227,6 +227,11 @@ struct Section {
     int64_t *total_dev = nullptr;
     uint16_t *group_start = nullptr;  // partial-update member offsets
     uint32_t *err_dev = nullptr;
+    // packed per-row validity (PU/agg emit): one u64 per row per run,
+    // bit c = column c non-null; built once per section by k_pack_valid
+    std::vector<uint64_t *> row_masks;
+    uint64_t **row_masks_dev = nullptr;  // [k] device array of the above
+    bool masks_built = false;
     // batched decode work (all run-columns in ONE launch each)
     Rlev2Chunk *rlev2_all = nullptr;
     int64_t n_rlev2 = 0;
@@ -1183,6 +1188,21 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
         sec.group_start = (uint16_t *)plan->bufs.alloc(
             sec.n_tiles * (PMH_TILE_ROWS + 1) * 2);
         if (!sec.group_start) return false;
+        bool any_valid = false;
+        for (auto &run : sec.runs)
+            for (auto &rc : run.cols)
+                if (rc.valid_dev) any_valid = true;
+        if (any_valid && n_cols <= 64) {
+            sec.row_masks.assign(k, nullptr);
+            for (int r = 0; r < k; r++) {
+                int64_t n = sec.runs[r].length > 0 ? sec.runs[r].length : 1;
+                sec.row_masks[r] = (uint64_t *)plan->bufs.alloc(n * 8);
+                if (!sec.row_masks[r]) return false;
+            }
+            sec.row_masks_dev =
+                (uint64_t **)up(sec.row_masks.data(), k * sizeof(void *));
+            if (!sec.row_masks_dev) return false;
+        }
     }
     std::vector<Rlev2Chunk> all_v;
     std::vector<RleChunk> all_d;
@@ -1536,6 +1556,16 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             if (e != hipSuccess) return fail("level_scatter", e);
         }
     }
+    if (sec.row_masks_dev && !sec.masks_built) {
+        for (int r = 0; r < k; r++) {
+            if (sec.runs[r].length <= 0) continue;
+            hipError_t pe = pmh_launch_pack_valid(
+                sec.all_cols + (size_t)r * n_cols, n_cols,
+                sec.runs[r].length, sec.row_masks[r], st);
+            if (pe != hipSuccess) return fail("pack_valid", pe);
+        }
+        sec.masks_built = true;
+    }
     (void)hipEventRecord(ev[1], st);
     hipError_t e = pmh_launch_partition(sec.key_cols, sec.lens_dev, k,
                                         PMH_TILE_ROWS, sec.n_tiles + 1,
@@ -1569,7 +1599,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,
             p->n_key_cols, p->n_key_cols + 1, sec.winners, sec.group_start,
             sec.tile_offsets, sec.n_tiles, PMH_TILE_ROWS, sec.total_dev,
-            p->out_ptrs_dev, p->out_valid_dev, st);
+            sec.row_masks_dev, p->out_ptrs_dev, p->out_valid_dev, st);
     } else {
         e = pmh_launch_emit(sec.all_cols, p->col_dtype_dev,
                             p->col_nullable_dev, n_cols, k, sec.winners,
