@@ -165,6 +165,7 @@ hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                                const uint16_t *group_start,
                                const int64_t *tile_offsets, int64_t n_tiles,
                                int64_t tile_rows, const int64_t *total_out,
+                               uint64_t *const *run_masks,
                                void *const *out_ptrs,
                                uint8_t *const *out_valid, hipStream_t stream);
 

@@ -937,6 +937,7 @@ __device__ inline uint64_t f64_ord(int64_t b) {
     return b < 0 ? ~(uint64_t)b : ((uint64_t)b | 0x8000000000000000ull);
 }
 
+template <bool MASKS>
 __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                            const uint8_t *col_nullable,
                            const uint8_t *col_agg, int n_cols, int k,
@@ -944,7 +945,14 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                            const uint16_t *group_start,
                            const int64_t *tile_offsets, int64_t n_tiles,
                            int64_t tile_rows, const int64_t *total_out,
-                           void *const *out_ptrs, uint8_t *const *out_valid) {
+                           uint64_t *const *run_masks, void *const *out_ptrs,
+                           uint8_t *const *out_valid) {
+    auto valid_of = [&](uint32_t m, int c) -> uint8_t {
+        if (MASKS)
+            return (uint8_t)((run_masks[m >> 28][m & 0x0fffffff] >> c) & 1);
+        const DevCol &dc = cols[(m >> 28) * n_cols + c];
+        return dc.valid0 ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff] : 1;
+    };
     const int64_t total = *total_out;
     const int64_t per_block =
         (total + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
@@ -982,18 +990,12 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
             case PMH_AGG_LAST_VALUE:
                 // last member as-is, its own validity (also the singleton
                 // ReducerMergeFunctionWrapper bypass)
-                if (col_nullable[c]) {
-                    const DevCol &dc = cols[lrun * n_cols + c];
-                    ok = dc.valid0 ? ((const uint8_t *)dc.valid0)[lrow] : 1;
-                }
+                if (col_nullable[c]) ok = valid_of(last, c);
                 break;
             case PMH_AGG_FIRST_VALUE: {
                 run = mem[ms] >> 28;
                 row = mem[ms] & 0x0fffffff;
-                if (col_nullable[c]) {
-                    const DevCol &dc = cols[run * n_cols + c];
-                    ok = dc.valid0 ? ((const uint8_t *)dc.valid0)[row] : 1;
-                }
+                if (col_nullable[c]) ok = valid_of(mem[ms], c);
                 break;
             }
             case PMH_AGG_LAST_NON_NULL:
@@ -1001,12 +1003,7 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                     ok = 0;
                     for (int32_t x = me - 1; !ok && x >= ms; x--) {
                         uint32_t m = mem[x];
-                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
-                        uint8_t v =
-                            dc.valid0
-                                ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
-                                : 1;
-                        if (v) {
+                        if (valid_of(m, c)) {
                             run = m >> 28;
                             row = m & 0x0fffffff;
                             ok = 1;
@@ -1019,12 +1016,7 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                     ok = 0;
                     for (int32_t x = ms; !ok && x < me; x++) {
                         uint32_t m = mem[x];
-                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
-                        uint8_t v =
-                            dc.valid0
-                                ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
-                                : 1;
-                        if (v) {
+                        if (valid_of(m, c)) {
                             run = m >> 28;
                             row = m & 0x0fffffff;
                             ok = 1;
@@ -1045,9 +1037,7 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                     uint32_t m = mem[x];
                     const DevCol &dc = cols[(m >> 28) * n_cols + c];
                     const int64_t r = m & 0x0fffffff;
-                    if (col_nullable[c] && dc.valid0 &&
-                        !((const uint8_t *)dc.valid0)[r])
-                        continue;
+                    if (col_nullable[c] && !valid_of(m, c)) continue;
                     int64_t vb = (dt == 4 || dt == 6)
                                      ? col_load<int64_t>(dc, r)
                                      : (int64_t)col_load<int32_t>(dc, r);
@@ -1470,13 +1460,22 @@ hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                                const uint16_t *group_start,
                                const int64_t *tile_offsets, int64_t n_tiles,
                                int64_t tile_rows, const int64_t *total_out,
+                               uint64_t *const *run_masks,
                                void *const *out_ptrs,
                                uint8_t *const *out_valid,
                                hipStream_t stream) {
-    hipLaunchKernelGGL(k_emit_agg, dim3(2048), dim3(256), 0, stream, cols,
-                       col_dtype, col_nullable, col_agg, n_cols, k, seq_col,
-                       kind_col, members, group_start, tile_offsets, n_tiles,
-                       tile_rows, total_out, out_ptrs, out_valid);
+    if (run_masks)
+        hipLaunchKernelGGL(k_emit_agg<true>, dim3(2048), dim3(256), 0, stream,
+                           cols, col_dtype, col_nullable, col_agg, n_cols, k,
+                           seq_col, kind_col, members, group_start,
+                           tile_offsets, n_tiles, tile_rows, total_out,
+                           run_masks, out_ptrs, out_valid);
+    else
+        hipLaunchKernelGGL(k_emit_agg<false>, dim3(2048), dim3(256), 0,
+                           stream, cols, col_dtype, col_nullable, col_agg,
+                           n_cols, k, seq_col, kind_col, members, group_start,
+                           tile_offsets, n_tiles, tile_rows, total_out,
+                           run_masks, out_ptrs, out_valid);
     return hipGetLastError();
 }
 

@@ -1592,8 +1592,8 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev,
             p->col_agg_dev, n_cols, k, p->n_key_cols, p->n_key_cols + 1,
             sec.winners, sec.group_start, sec.tile_offsets, sec.n_tiles,
-            PMH_TILE_ROWS, sec.total_dev, p->out_ptrs_dev, p->out_valid_dev,
-            st);
+            PMH_TILE_ROWS, sec.total_dev, sec.row_masks_dev, p->out_ptrs_dev,
+            p->out_valid_dev, st);
     } else if (p->pu) {
         e = pmh_launch_emit_pu(
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,
