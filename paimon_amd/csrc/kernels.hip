@@ -1111,9 +1111,18 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
 // workgroup per chunk; packed chunks compute the dense prefix with an
 // in-LDS running scan over 256-element blocks.
 __global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks) {
-    __shared__ int32_t s_scan[PMH_TILE_THREADS];
-    __shared__ int32_t s_running;
-    for (int64_t cidx = blockIdx.x; cidx < n_chunks; cidx += gridDim.x) {
+    // one WAVE per chunk, barrier-free: the host prescan bounds packed
+    // chunks at <= 512 values (parquet) / <= 1040 (ORC byte-RLE groups) and
+    // precomputes each chunk's dense prefix (aux), so the only running state
+    // is a per-wave register popcount. The previous one-WORKGROUP-per-chunk
+    // version chained 3 __syncthreads per 512 values through an LDS running
+    // scan and was serialization-bound (62 ms at the 8x10M PU shape; 94 GB/s
+    // effective).
+    const int lane = (int)(threadIdx.x & 63);
+    const int64_t wave_id =
+        (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    const int64_t n_waves = (int64_t)gridDim.x * (blockDim.x >> 6);
+    for (int64_t cidx = wave_id; cidx < n_chunks; cidx += n_waves) {
         RleChunk ch = chunks[cidx];
         const uint8_t *dense = (const uint8_t *)ch.dense_addr;
         uint8_t *out = (uint8_t *)ch.out_addr;
@@ -1121,7 +1130,7 @@ __global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks) {
         const int esize = ch.esize;
         if (ch.kind == 0) {
             if (ch.value) {  // run of non-nulls: dense block copy
-                for (int32_t i = threadIdx.x; i < ch.count; i += blockDim.x) {
+                for (int32_t i = lane; i < ch.count; i += 64) {
                     valid[ch.out_start + i] = 1;
                     const uint8_t *s = dense + (ch.aux + i) * esize;
                     uint8_t *d = out + (ch.out_start + i) * esize;
@@ -1131,7 +1140,7 @@ __global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks) {
                         *(int64_t *)d = *(const int64_t *)s;
                 }
             } else {  // run of nulls
-                for (int32_t i = threadIdx.x; i < ch.count; i += blockDim.x) {
+                for (int32_t i = lane; i < ch.count; i += 64) {
                     valid[ch.out_start + i] = 0;
                     uint8_t *d = out + (ch.out_start + i) * esize;
                     if (esize == 4)
@@ -1140,41 +1149,24 @@ __global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks) {
                         *(int64_t *)d = 0;
                 }
             }
-            __syncthreads();
             continue;
         }
-        // bit-packed: 1 byte per 8 levels. Dense prefixes come from wave
-        // ballots (popcount of lower lanes) + a tiny cross-wave LDS combine
-        // — the former Hillis-Steele block scan cost 16 barriers per 256
-        // values.
+        // bit-packed: 1 byte per 8 levels; dense index = aux + running
+        // popcount (registers only) + lower-lane ballot prefix
         const uint8_t *src = (const uint8_t *)ch.src;
-        const int lane = (int)(threadIdx.x & 63);
-        const int wv = (int)(threadIdx.x >> 6);
-        constexpr int NW = PMH_TILE_THREADS / 64;
-        if (threadIdx.x == 0) s_running = 0;
-        __syncthreads();
-        for (int32_t b = 0; b < ch.count; b += (int32_t)blockDim.x) {
-            int32_t i = b + threadIdx.x;
+        int32_t running = 0;
+        for (int32_t b = 0; b < ch.count; b += 64) {
+            int32_t i = b + lane;
             int bit = 0;
             if (i < ch.count) bit = (src[i >> 3] >> (i & 7)) & 1;
             uint64_t mask = __ballot(bit != 0);
-            uint64_t lt = lane == 0 ? 0 : (mask << (64 - lane));
-            int32_t my_before = __popcll(lt);
-            int32_t wave_total = __popcll(mask);
-            if (lane == 0) s_scan[wv] = wave_total;
-            __syncthreads();
-            int32_t wave_base = 0, round_total = 0;
-#pragma unroll
-            for (int w = 0; w < NW; w++) {
-                if (w < wv) wave_base += s_scan[w];
-                round_total += s_scan[w];
-            }
+            int32_t my_before =
+                lane == 0 ? 0 : __popcll(mask << (64 - lane));
             if (i < ch.count) {
                 valid[ch.out_start + i] = (uint8_t)bit;
                 uint8_t *dp = out + (ch.out_start + i) * esize;
                 if (bit) {
-                    int64_t didx =
-                        ch.aux + s_running + wave_base + my_before;
+                    int64_t didx = ch.aux + running + my_before;
                     const uint8_t *sp = dense + didx * esize;
                     if (esize == 4)
                         *(int32_t *)dp = *(const int32_t *)sp;
@@ -1187,9 +1179,7 @@ __global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks) {
                         *(int64_t *)dp = 0;
                 }
             }
-            __syncthreads();
-            if (threadIdx.x == 0) s_running += round_total;
-            __syncthreads();
+            running += __popcll(mask);
         }
     }
 }

@@ -327,9 +327,9 @@ static bool prescan_def(const uint8_t *s, int64_t len, int64_t n,
                         int64_t out_row0, int64_t rel_base,
                         int64_t *dense_before, std::vector<RleChunk> &out) {
     int64_t p = 0, cnt = 0;
-    // small chunks: k_level_scatter scans a chunk serially in 256-element
-    // blocks, so chunk size bounds the serial depth per workgroup
-    const int64_t MAX_CHUNK = 2048;
+    // one wave per chunk in k_level_scatter: chunk size bounds the serial
+    // 64-value passes per wave
+    const int64_t MAX_CHUNK = 512;
     while (cnt < n) {
         if (p >= len) {
             set_error("def-level stream overrun");
