@@ -798,14 +798,19 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
     const int64_t slice_lo = (int64_t)blockIdx.x * per_block;
     const int64_t slice_hi =
         slice_lo + per_block < total ? slice_lo + per_block : total;
+    int64_t t = -1;
     for (int64_t i = slice_lo + threadIdx.x; i < slice_hi; i += blockDim.x) {
-        int64_t lo = 0, hi = n_tiles - 1;
-        while (lo < hi) {
-            int64_t mid = (lo + hi + 1) >> 1;
-            if (tile_offsets[mid] <= i) lo = mid;
-            else hi = mid - 1;
+        if (t < 0) {  // first iteration: binary-search the owning tile; i
+                      // then grows monotonically, so a forward cursor walks
+            int64_t lo = 0, hi = n_tiles - 1;
+            while (lo < hi) {
+                int64_t mid = (lo + hi + 1) >> 1;
+                if (tile_offsets[mid] <= i) lo = mid;
+                else hi = mid - 1;
+            }
+            t = lo;
         }
-        int64_t t = lo;
+        while (t + 1 < n_tiles && tile_offsets[t + 1] <= i) t++;
         int64_t g = i - tile_offsets[t];
         const uint16_t *gs = &group_start[t * (tile_rows + 1)];
         const uint32_t *mem = &members[t * (tile_rows + PMH_MAX_RUNS)];
@@ -959,14 +964,19 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
     const int64_t slice_lo = (int64_t)blockIdx.x * per_block;
     const int64_t slice_hi =
         slice_lo + per_block < total ? slice_lo + per_block : total;
+    int64_t t = -1;
     for (int64_t i = slice_lo + threadIdx.x; i < slice_hi; i += blockDim.x) {
-        int64_t lo = 0, hi = n_tiles - 1;
-        while (lo < hi) {
-            int64_t mid = (lo + hi + 1) >> 1;
-            if (tile_offsets[mid] <= i) lo = mid;
-            else hi = mid - 1;
+        if (t < 0) {  // first iteration: binary-search the owning tile; i
+                      // then grows monotonically, so a forward cursor walks
+            int64_t lo = 0, hi = n_tiles - 1;
+            while (lo < hi) {
+                int64_t mid = (lo + hi + 1) >> 1;
+                if (tile_offsets[mid] <= i) lo = mid;
+                else hi = mid - 1;
+            }
+            t = lo;
         }
-        int64_t t = lo;
+        while (t + 1 < n_tiles && tile_offsets[t + 1] <= i) t++;
         int64_t g = i - tile_offsets[t];
         const uint16_t *gs = &group_start[t * (tile_rows + 1)];
         const uint32_t *mem = &members[t * (tile_rows + PMH_MAX_RUNS)];
