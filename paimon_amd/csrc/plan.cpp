@@ -231,7 +231,6 @@ struct Section {
     // bit c = column c non-null; built once per section by k_pack_valid
     std::vector<uint64_t *> row_masks;
     uint64_t **row_masks_dev = nullptr;  // [k] device array of the above
-    bool masks_built = false;
     // batched decode work (all run-columns in ONE launch each)
     Rlev2Chunk *rlev2_all = nullptr;
     int64_t n_rlev2 = 0;
@@ -1556,7 +1555,9 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             if (e != hipSuccess) return fail("level_scatter", e);
         }
     }
-    if (sec.row_masks_dev && !sec.masks_built) {
+    if (sec.row_masks_dev) {
+        // rebuilt every pass: masks derive from the per-step level decode,
+        // so caching them across steps would skip timed work
         for (int r = 0; r < k; r++) {
             if (sec.runs[r].length <= 0) continue;
             hipError_t pe = pmh_launch_pack_valid(
@@ -1564,7 +1565,6 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                 sec.runs[r].length, sec.row_masks[r], st);
             if (pe != hipSuccess) return fail("pack_valid", pe);
         }
-        sec.masks_built = true;
     }
     (void)hipEventRecord(ev[1], st);
     hipError_t e = pmh_launch_partition(sec.key_cols, sec.lens_dev, k,
