@@ -68,10 +68,10 @@ def get_args():
     ap.add_argument("--format", dest="file_format", default="parquet",
                     choices=["parquet", "orc"])
     ap.add_argument("--engine", default="deduplicate",
-                    choices=["deduplicate", "partial-update"],
-                    help="partial-update = the C3 merge-function half on "
-                    "Parquet (supplementary; the default metric line is "
-                    "deduplicate per BASELINE.json configs[1])")
+                    choices=["deduplicate", "partial-update", "aggregation",
+                             "first-row"],
+                    help="supplementary merge-engine variants; the default "
+                    "metric line is deduplicate per BASELINE.json configs[1]")
     ap.add_argument("--seed", type=int, default=42)
     ap.add_argument("--data-dir", default=os.path.join(REPO, "data"))
     ap.add_argument("--cpu-baseline-rows", type=int, default=2_000_000,
@@ -83,8 +83,10 @@ def get_args():
 def ensure_data(args, rank):
     from paimon_amd.datagen import (gen_runs_dedup, gen_runs_partial_update,
                                     write_runs)
-    pu = args.engine == "partial-update"
-    tag = (f"{'c3pu' if pu else 'c2'}_{args.runs}x{args.rows}v{args.vals}_"
+    pu = args.engine in ("partial-update", "aggregation")
+    tagname = {"partial-update": "c3pu", "aggregation": "c3pu",
+               "first-row": "c2fr"}.get(args.engine, "c2")
+    tag = (f"{tagname}_{args.runs}x{args.rows}v{args.vals}_"
            f"{args.file_format}_{args.compression}_seed{args.seed}_rank{rank}")
     out_dir = os.path.join(args.data_dir, tag)
     manifest = os.path.join(out_dir, "files.json")
@@ -96,6 +98,10 @@ def ensure_data(args, rank):
                                        n_value_cols=args.vals,
                                        seed=args.seed + rank,
                                        update_frac=0.3, update_cols=6)
+    elif args.engine == "first-row":
+        # first-row rejects retracts (FirstRowMergeFunction.java:49-59)
+        runs = gen_runs_dedup(args.runs, args.rows, n_value_cols=args.vals,
+                              seed=args.seed + rank, delete_frac=0.0)
     else:
         runs = gen_runs_dedup(args.runs, args.rows, n_value_cols=args.vals,
                               seed=args.seed + rank)
@@ -160,10 +166,15 @@ def main():
     value_cols = ([{"name": "v_k", "type": "int64"}] +
                   [{"name": f"v_c{i}", "type": "int32"}
                    for i in range(args.vals)])
+    aggs = None
+    if args.engine == "aggregation":
+        cycle = ["sum", "max", "min", "last_non_null_value"]
+        aggs = {f"v_c{i}": cycle[i % 4] for i in range(args.vals)}
     t_stage0 = time.perf_counter()
     plan = MergeReadPlan(sess, file_descs_from_metas(metas), key_cols,
                          value_cols, merge_engine=args.engine,
-                         drop_delete=True, output="device")
+                         drop_delete=True, output="device",
+                         aggregations=aggs)
     t_stage = time.perf_counter() - t_stage0
 
     def one_step():
@@ -253,11 +264,16 @@ def main():
             "workload": (f"merge-on-read: {args.runs} sorted runs x "
                          f"{args.rows} rows, int64 PK + {args.vals} int32, "
                          f"{args.file_format} {args.compression}, "
-                         + ("PartialUpdate (30% updates, 6-col subsets; "
-                            "parquet variant of configs[2])"
-                            if args.engine == "partial-update" else
-                            "Deduplicate, drop-delete "
-                            "(BASELINE.json configs[1])")),
+                         + {"partial-update":
+                            "PartialUpdate (30% updates, 6-col subsets; "
+                            "variant of configs[2])",
+                            "aggregation":
+                            "Aggregation (sum/max/min/last_non_null cycle, "
+                            "30% update rows)",
+                            "first-row": "FirstRow (insert-only)",
+                            }.get(args.engine,
+                                  "Deduplicate, drop-delete "
+                                  "(BASELINE.json configs[1])")),
             "n_runs": args.runs,
             "rows_per_run": args.rows,
             "value_cols": args.vals,
