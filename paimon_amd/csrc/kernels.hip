@@ -236,7 +236,7 @@ DEV bool kind_is_add(uint8_t k) { return k == 0 || k == 2; }
 // PartialUpdateMergeFunction.java:170-186) — retracts set err_flag.
 // Separate template instantiations keep the PU path's extra registers out
 // of the deduplicate kernel.
-template <bool PU>
+template <bool PU, bool FR>
 __launch_bounds__(PMH_TILE_THREADS) __global__
 void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    const int64_t *lens, int k, const int32_t *cuts,
@@ -245,11 +245,12 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    uint16_t *group_start, uint32_t *err_flag) {
     const bool drop_delete = flags & 1;
     const bool ignore_delete = flags & 2;
-    // first-row engine (FirstRowMergeFunction.java:32-77): keep the FIRST
-    // record per key in ascending (seq, isAdd) order; retracts throw unless
-    // ignore-delete (then skipped); singleton groups bypass the merge
-    // function (wrapper) exactly as deduplicate does.
-    const bool first_row = flags & 8;
+    // FR = first-row engine (FirstRowMergeFunction.java:32-77): keep the
+    // FIRST record per key in ascending (seq, isAdd) order; retracts throw
+    // unless ignore-delete (then skipped); singleton groups bypass the merge
+    // function (wrapper) exactly as deduplicate does. A template parameter so
+    // the deduplicate winner walk carries no extra select.
+    constexpr bool first_row = FR;
     // ablation levels (profiling only, flags bits 8..): 1=stage,2=+merge,
     // 3=+scan, 0/absent=full. Partial levels publish a checksum so the
     // compiler cannot dead-code the ablated phases' inputs.
@@ -1407,16 +1408,19 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
                                   uint16_t *group_start, uint32_t *err_flag,
                                   hipStream_t stream) {
     int blocks = n_tiles < 4096 ? (int)n_tiles : 4096;
-    if (flags & 4)
-        hipLaunchKernelGGL(k_merge_tiles<true>, dim3(blocks),
-                           dim3(PMH_TILE_THREADS), 0, stream, keys, seqs,
-                           kinds, lens, k, cuts, n_tiles, tile_rows, flags,
-                           winners, tile_counts, group_start, err_flag);
+    const bool pu = flags & 4, fr = flags & 8;
+    auto launch = [&](auto kern) {
+        hipLaunchKernelGGL(kern, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
+                           stream, keys, seqs, kinds, lens, k, cuts, n_tiles,
+                           tile_rows, flags, winners, tile_counts,
+                           group_start, err_flag);
+    };
+    if (pu)
+        launch(k_merge_tiles<true, false>);  // PU/agg never first-row
+    else if (fr)
+        launch(k_merge_tiles<false, true>);
     else
-        hipLaunchKernelGGL(k_merge_tiles<false>, dim3(blocks),
-                           dim3(PMH_TILE_THREADS), 0, stream, keys, seqs,
-                           kinds, lens, k, cuts, n_tiles, tile_rows, flags,
-                           winners, tile_counts, group_start, err_flag);
+        launch(k_merge_tiles<false, false>);
     return hipGetLastError();
 }
 
