@@ -983,7 +983,26 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
         const uint32_t *mem = &members[t * (tile_rows + PMH_MAX_RUNS)];
         const int32_t ms = gs[g], me = gs[g + 1];
         const int gn = me - ms;
-        const uint32_t last = mem[me - 1];
+        // register-cache the first 4 members + their packed masks: the
+        // per-column folds below re-walk the group for EVERY column, and
+        // re-reading the member list / masks from HBM per column dominated
+        // this kernel (groups average ~1.3 members)
+        uint32_t ma[4];
+        uint64_t vma[4];
+#pragma unroll
+        for (int x = 0; x < 4; x++)
+            ma[x] = mem[ms + (x < gn ? x : gn - 1)];
+        if (MASKS) {
+#pragma unroll
+            for (int x = 0; x < 4; x++)
+                vma[x] = x < gn
+                             ? run_masks[ma[x] >> 28][ma[x] & 0x0fffffff]
+                             : 0;
+        }
+        auto member_at = [&](int32_t x) -> uint32_t {
+            return x < 4 ? ma[x] : mem[ms + x];
+        };
+        const uint32_t last = gn <= 4 ? ma[gn - 1] : mem[me - 1];
         const int lrun = last >> 28;
         const int64_t lrow = last & 0x0fffffff;
         for (int c = 0; c < n_cols; c++) {
@@ -991,6 +1010,10 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 ((int8_t *)out_ptrs[c])[i] = 0;  // RowKind.INSERT
                 continue;
             }
+            auto valid_at = [&](int32_t x) -> uint8_t {
+                if (MASKS && x < 4) return (uint8_t)((vma[x] >> c) & 1);
+                return valid_of(member_at(x), c);
+            };
             const int dt = col_dtype[c];
             const int agg = gn == 1 ? PMH_AGG_LAST_VALUE : col_agg[c];
             int64_t run = lrun, row = lrow;
@@ -1001,20 +1024,21 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
             case PMH_AGG_LAST_VALUE:
                 // last member as-is, its own validity (also the singleton
                 // ReducerMergeFunctionWrapper bypass)
-                if (col_nullable[c]) ok = valid_of(last, c);
+                if (col_nullable[c])
+                    ok = gn <= 4 ? valid_at(gn - 1) : valid_of(last, c);
                 break;
             case PMH_AGG_FIRST_VALUE: {
-                run = mem[ms] >> 28;
-                row = mem[ms] & 0x0fffffff;
-                if (col_nullable[c]) ok = valid_of(mem[ms], c);
+                run = ma[0] >> 28;
+                row = ma[0] & 0x0fffffff;
+                if (col_nullable[c]) ok = valid_at(0);
                 break;
             }
             case PMH_AGG_LAST_NON_NULL:
                 if (col_nullable[c]) {
                     ok = 0;
-                    for (int32_t x = me - 1; !ok && x >= ms; x--) {
-                        uint32_t m = mem[x];
-                        if (valid_of(m, c)) {
+                    for (int32_t x = gn - 1; !ok && x >= 0; x--) {
+                        if (valid_at(x)) {
+                            uint32_t m = member_at(x);
                             run = m >> 28;
                             row = m & 0x0fffffff;
                             ok = 1;
@@ -1025,17 +1049,17 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
             case PMH_AGG_FIRST_NON_NULL:
                 if (col_nullable[c]) {
                     ok = 0;
-                    for (int32_t x = ms; !ok && x < me; x++) {
-                        uint32_t m = mem[x];
-                        if (valid_of(m, c)) {
+                    for (int32_t x = 0; !ok && x < gn; x++) {
+                        if (valid_at(x)) {
+                            uint32_t m = member_at(x);
                             run = m >> 28;
                             row = m & 0x0fffffff;
                             ok = 1;
                         }
                     }
                 } else {
-                    run = mem[ms] >> 28;
-                    row = mem[ms] & 0x0fffffff;
+                    run = ma[0] >> 28;
+                    row = ma[0] & 0x0fffffff;
                 }
                 break;
             default: {  // SUM / MAX / MIN: full fold, null inputs skipped
@@ -1044,11 +1068,11 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 int64_t iacc = 0;
                 float facc = 0.f;
                 double dacc = 0.0;
-                for (int32_t x = ms; x < me; x++) {
-                    uint32_t m = mem[x];
+                for (int32_t x = 0; x < gn; x++) {
+                    uint32_t m = member_at(x);
                     const DevCol &dc = cols[(m >> 28) * n_cols + c];
                     const int64_t r = m & 0x0fffffff;
-                    if (col_nullable[c] && !valid_of(m, c)) continue;
+                    if (col_nullable[c] && !valid_at(x)) continue;
                     int64_t vb = (dt == 4 || dt == 6)
                                      ? col_load<int64_t>(dc, r)
                                      : (int64_t)col_load<int32_t>(dc, r);
