@@ -1142,9 +1142,7 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
 // Decode def-level streams (RLE/bit-packed hybrid, bit width 1) and position
 // the dense PLAIN values: valid[row] = level, out[row] = dense[aux + prefix]
 // (the null handling of VectorizedColumnReader.readBatch,
-// paimon-format/.../reader/VectorizedColumnReader.java:143-241). One
-// workgroup per chunk; packed chunks compute the dense prefix with an
-// in-LDS running scan over 256-element blocks.
+// paimon-format/.../reader/VectorizedColumnReader.java:143-241).
 __global__ void k_level_scatter(const RleChunk *chunks, int64_t n_chunks) {
     // one WAVE per chunk, barrier-free: the host prescan bounds packed
     // chunks at <= 512 values (parquet) / <= 1040 (ORC byte-RLE groups) and
