@@ -53,9 +53,10 @@ def _check(got, exp, names):
 
 class TestAggregationEngine:
     def _run(self, tmp_path, runs, aggs_by_col, file_format="parquet",
-             col_type="int32"):
-        metas = write_runs(runs, str(tmp_path), compression="NONE",
-                           file_format=file_format)
+             col_type="int32", **write_kwargs):
+        write_kwargs.setdefault("compression", "NONE")
+        metas = write_runs(runs, str(tmp_path), file_format=file_format,
+                           **write_kwargs)
         n_vals = len(runs[0]["values"])
         names = ["v_k"] + [f"v_c{i}" for i in range(n_vals - 1)]
         exp = aggregation_model(
@@ -114,6 +115,18 @@ class TestAggregationEngine:
                 r["values"][c] = rng.standard_normal(n).astype(np.float32)
         self._run(tmp_path, runs, {"v_c0": "sum", "v_c1": "max"},
                   col_type="float32")
+
+    def test_agg_zstd_multi_rowgroup(self, tmp_path):
+        # chunked pages + zstd staging + >1 row group under the member lists
+        runs = gen_runs_partial_update(3, 60_000, n_value_cols=5, seed=101,
+                                       update_frac=0.4, update_cols=3)
+        for r in runs:
+            for c in range(1, 6):
+                r["values"][c] = (r["values"][c] % 10_000).astype(np.int32)
+        aggs = {"v_c0": "sum", "v_c1": "max", "v_c2": "min",
+                "v_c3": "last_non_null_value", "v_c4": "first_value"}
+        self._run(tmp_path, runs, aggs, compression="zstd",
+                  row_group_rows=16_384, data_page_rows=4_096)
 
     def test_retract_rejected(self, tmp_path):
         runs = gen_runs_partial_update(2, 5_000, n_value_cols=2, seed=98)

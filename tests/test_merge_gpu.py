@@ -338,9 +338,10 @@ class TestConcurrentPlans:
 
 
 class TestFirstRowEngine:
-    def _run(self, tmp_path, runs, **kw):
+    def _run(self, tmp_path, runs, file_format="parquet", **kw):
         from oracle import merge_first_row_model
-        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        metas = write_runs(runs, str(tmp_path), compression="NONE",
+                           file_format=file_format)
         r, w = merge_first_row_model(runs, **kw)
         with Session(0) as s:
             with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
@@ -361,6 +362,11 @@ class TestFirstRowEngine:
         runs = gen_runs_dedup(5, 15_000, n_value_cols=2, seed=64,
                               delete_frac=0.25)
         self._run(tmp_path, runs, ignore_delete=True)
+
+    def test_first_row_orc(self, tmp_path):
+        runs = gen_runs_dedup(4, 20_000, n_value_cols=2, seed=66,
+                              delete_frac=0.0)
+        self._run(tmp_path, runs, file_format="orc")
 
     def test_first_row_rejects_retracts(self, tmp_path):
         runs = gen_runs_dedup(3, 8_000, n_value_cols=1, seed=65,
