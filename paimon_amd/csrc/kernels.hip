@@ -983,10 +983,10 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
         const uint32_t *mem = &members[t * (tile_rows + PMH_MAX_RUNS)];
         const int32_t ms = gs[g], me = gs[g + 1];
         const int gn = me - ms;
-        // register-cache the first 4 members + their packed masks: the
-        // per-column folds below re-walk the group for EVERY column, and
-        // re-reading the member list / masks from HBM per column dominated
-        // this kernel (groups average ~1.3 members)
+        // register-cache the first 4 members (oldest-first) + their packed
+        // masks; every access below uses CONSTANT indices after unrolling —
+        // a dynamically-indexed local array spills the whole array to
+        // scratch (64 B/lane measured before this shape)
         uint32_t ma[4];
         uint64_t vma[4];
 #pragma unroll
@@ -999,10 +999,18 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                              ? run_masks[ma[x] >> 28][ma[x] & 0x0fffffff]
                              : 0;
         }
-        auto member_at = [&](int32_t x) -> uint32_t {
-            return x < 4 ? ma[x] : mem[ms + x];
-        };
-        const uint32_t last = gn <= 4 ? ma[gn - 1] : mem[me - 1];
+        uint32_t last = ma[0];
+        uint64_t vlast = MASKS ? vma[0] : 0;
+#pragma unroll
+        for (int x = 1; x < 4; x++)
+            if (x < gn) {
+                last = ma[x];
+                if (MASKS) vlast = vma[x];
+            }
+        if (gn > 4) {
+            last = mem[me - 1];
+            if (MASKS) vlast = run_masks[last >> 28][last & 0x0fffffff];
+        }
         const int lrun = last >> 28;
         const int64_t lrow = last & 0x0fffffff;
         for (int c = 0; c < n_cols; c++) {
@@ -1010,10 +1018,6 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 ((int8_t *)out_ptrs[c])[i] = 0;  // RowKind.INSERT
                 continue;
             }
-            auto valid_at = [&](int32_t x) -> uint8_t {
-                if (MASKS && x < 4) return (uint8_t)((vma[x] >> c) & 1);
-                return valid_of(member_at(x), c);
-            };
             const int dt = col_dtype[c];
             const int agg = gn == 1 ? PMH_AGG_LAST_VALUE : col_agg[c];
             int64_t run = lrun, row = lrow;
@@ -1025,22 +1029,37 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 // last member as-is, its own validity (also the singleton
                 // ReducerMergeFunctionWrapper bypass)
                 if (col_nullable[c])
-                    ok = gn <= 4 ? valid_at(gn - 1) : valid_of(last, c);
+                    ok = MASKS ? (uint8_t)((vlast >> c) & 1)
+                               : valid_of(last, c);
                 break;
             case PMH_AGG_FIRST_VALUE: {
                 run = ma[0] >> 28;
                 row = ma[0] & 0x0fffffff;
-                if (col_nullable[c]) ok = valid_at(0);
+                if (col_nullable[c])
+                    ok = MASKS ? (uint8_t)((vma[0] >> c) & 1)
+                               : valid_of(ma[0], c);
                 break;
             }
             case PMH_AGG_LAST_NON_NULL:
                 if (col_nullable[c]) {
                     ok = 0;
-                    for (int32_t x = gn - 1; !ok && x >= 0; x--) {
-                        if (valid_at(x)) {
-                            uint32_t m = member_at(x);
+                    // newest first: uncached tail (x >= 4), then cached
+                    for (int32_t x = gn - 1; !ok && x >= 4; x--) {
+                        uint32_t m = mem[ms + x];
+                        if (valid_of(m, c)) {
                             run = m >> 28;
                             row = m & 0x0fffffff;
+                            ok = 1;
+                        }
+                    }
+#pragma unroll
+                    for (int x = 3; x >= 0; x--) {
+                        if (ok || x >= gn) continue;
+                        uint8_t v = MASKS ? (uint8_t)((vma[x] >> c) & 1)
+                                          : valid_of(ma[x], c);
+                        if (v) {
+                            run = ma[x] >> 28;
+                            row = ma[x] & 0x0fffffff;
                             ok = 1;
                         }
                     }
@@ -1049,9 +1068,20 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
             case PMH_AGG_FIRST_NON_NULL:
                 if (col_nullable[c]) {
                     ok = 0;
-                    for (int32_t x = 0; !ok && x < gn; x++) {
-                        if (valid_at(x)) {
-                            uint32_t m = member_at(x);
+#pragma unroll
+                    for (int x = 0; x < 4; x++) {
+                        if (ok || x >= gn) continue;
+                        uint8_t v = MASKS ? (uint8_t)((vma[x] >> c) & 1)
+                                          : valid_of(ma[x], c);
+                        if (v) {
+                            run = ma[x] >> 28;
+                            row = ma[x] & 0x0fffffff;
+                            ok = 1;
+                        }
+                    }
+                    for (int32_t x = 4; !ok && x < gn; x++) {
+                        uint32_t m = mem[ms + x];
+                        if (valid_of(m, c)) {
                             run = m >> 28;
                             row = m & 0x0fffffff;
                             ok = 1;
@@ -1068,11 +1098,9 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 int64_t iacc = 0;
                 float facc = 0.f;
                 double dacc = 0.0;
-                for (int32_t x = 0; x < gn; x++) {
-                    uint32_t m = member_at(x);
+                auto fold_one = [&](uint32_t m) {
                     const DevCol &dc = cols[(m >> 28) * n_cols + c];
                     const int64_t r = m & 0x0fffffff;
-                    if (col_nullable[c] && !valid_at(x)) continue;
                     int64_t vb = (dt == 4 || dt == 6)
                                      ? col_load<int64_t>(dc, r)
                                      : (int64_t)col_load<int32_t>(dc, r);
@@ -1084,7 +1112,7 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                             dacc = __longlong_as_double(vb);
                         else
                             iacc = vb;  // raw bits for float max/min
-                        continue;
+                        return;
                     }
                     if (agg == PMH_AGG_SUM) {
                         if (dt == 5) facc += __int_as_float((int32_t)vb);
@@ -1105,6 +1133,20 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                         }
                         if (take) iacc = vb;
                     }
+                };
+#pragma unroll
+                for (int x = 0; x < 4; x++) {
+                    if (x >= gn) continue;
+                    if (col_nullable[c] &&
+                        !(MASKS ? (uint8_t)((vma[x] >> c) & 1)
+                                : valid_of(ma[x], c)))
+                        continue;
+                    fold_one(ma[x]);
+                }
+                for (int32_t x = 4; x < gn; x++) {
+                    uint32_t m = mem[ms + x];
+                    if (col_nullable[c] && !valid_of(m, c)) continue;
+                    fold_one(m);
                 }
                 if (agg == PMH_AGG_SUM && dt == 5)
                     bits = (int64_t)__float_as_int(facc);
