@@ -686,25 +686,7 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                 row[x] = row[0];
             }
         }
-        // software-pipelined column loop: issue column c+1's gathers BEFORE
-        // column c's stores. Without this the stores sit between the loads in
-        // program order and (absent aliasing info) fence them, so only R
-        // loads were ever in flight; preloading doubles the outstanding
-        // gathers on this latency-bound kernel. Raw bits travel in int64;
-        // the store switch truncates to the output width.
-        auto load_col = [&](int c, int64_t v[R]) {
-            const bool wide = col_dtype[c] == 4 || col_dtype[c] == 6;
-#pragma unroll
-            for (int x = 0; x < R; x++) {
-                const DevCol &dc = cols[run[x] * n_cols + c];
-                v[x] = wide ? col_load<int64_t>(dc, row[x])
-                            : (int64_t)col_load<int32_t>(dc, row[x]);
-            }
-        };
-        int64_t cur[R], nxt[R];
-        load_col(0, cur);
         for (int c = 0; c < n_cols; c++) {
-            if (c + 1 < n_cols) load_col(c + 1, nxt);
             if (col_nullable[c] && out_valid[c]) {
                 // winner's validity byte (dedup keeps the record as-is)
 #pragma unroll
@@ -716,35 +698,52 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                 }
             }
             switch (col_dtype[c]) {
-            case 1:  // INT8 output from INT32-stored parquet TINYINT
+            case 1: {  // INT8 output from INT32-stored parquet TINYINT
+                int32_t v[R];
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    v[x] = col_load<int32_t>(cols[run[x] * n_cols + c], row[x]);
 #pragma unroll
                 for (int x = 0; x < R; x++)
                     if (x < nr)
-                        ((int8_t *)out_ptrs[c])[idx[x]] = (int8_t)cur[x];
+                        ((int8_t *)out_ptrs[c])[idx[x]] = (int8_t)v[x];
                 break;
-            case 2:  // INT16 output from INT32-stored parquet SMALLINT
+            }
+            case 2: {  // INT16 output from INT32-stored parquet SMALLINT
+                int32_t v[R];
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    v[x] = col_load<int32_t>(cols[run[x] * n_cols + c], row[x]);
 #pragma unroll
                 for (int x = 0; x < R; x++)
                     if (x < nr)
-                        ((int16_t *)out_ptrs[c])[idx[x]] = (int16_t)cur[x];
+                        ((int16_t *)out_ptrs[c])[idx[x]] = (int16_t)v[x];
                 break;
+            }
             case 3:
-            case 5:
+            case 5: {
+                int32_t v[R];
 #pragma unroll
                 for (int x = 0; x < R; x++)
-                    if (x < nr)
-                        ((int32_t *)out_ptrs[c])[idx[x]] = (int32_t)cur[x];
+                    v[x] = col_load<int32_t>(cols[run[x] * n_cols + c], row[x]);
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    if (x < nr) ((int32_t *)out_ptrs[c])[idx[x]] = v[x];
                 break;
+            }
             case 4:
-            case 6:
+            case 6: {
+                int64_t v[R];
 #pragma unroll
                 for (int x = 0; x < R; x++)
-                    if (x < nr) ((int64_t *)out_ptrs[c])[idx[x]] = cur[x];
+                    v[x] = col_load<int64_t>(cols[run[x] * n_cols + c], row[x]);
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    if (x < nr) ((int64_t *)out_ptrs[c])[idx[x]] = v[x];
                 break;
+            }
             default: break;
             }
-#pragma unroll
-            for (int x = 0; x < R; x++) cur[x] = nxt[x];
         }
     }
 }
