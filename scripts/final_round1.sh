@@ -27,6 +27,11 @@ python3 bench.py --steps 6 --warmup 2 --engine first-row \
     --cpu-baseline-rows 0 > gpurun_out/bench_r01_fr.json \
     2> gpurun_out/bench_r01_fr.err
 echo BENCH_FR_RC=$?
+# capacity: 16 runs x 20M rows (320M input rows, near the C5 shape)
+python3 bench.py --steps 3 --warmup 1 --runs 16 --rows 20000000 \
+    --cpu-baseline-rows 0 > gpurun_out/bench_r01_c5_16x20M.json \
+    2> gpurun_out/bench_r01_c5.err
+echo BENCH_C5_RC=$?
 export TMPDIR=/tmp
 cd /tmp
 timeout 420 rocprofv3 --kernel-trace --stats -d /root/repo/gpurun_out/prof \
