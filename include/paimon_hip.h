@@ -126,6 +126,18 @@ const char *pmh_last_error(void);
 char *pmh_debug_footer_json(const char *path);
 void pmh_free_string(char *s);
 
+/* Write a Parquet v1 data file (PLAIN, uncompressed) from HOST columnar
+ * buffers — the compaction write-back half of the CompactRewriter surface
+ * (what KeyValueDataFileWriter + the vendored parquet-mr writer do in the
+ * reference, io/KeyValueDataFileWriter.java:121-170). cols[i].data/valid
+ * are host pointers at the column's output width (TINYINT/SMALLINT widen to
+ * the INT32 physical type on write, matching the read path). No GPU
+ * required. row_group_rows/page_rows <= 0 pick defaults (1M / 64k).
+ * Returns 0, or -1 with pmh_last_error() set. */
+int pmh_write_parquet(const pmh_col *cols, int32_t n_cols, int64_t n_rows,
+                      const char *path, int64_t row_group_rows,
+                      int64_t page_rows);
+
 /* Restatement of IntervalPartition.partition() for int64 keys
  * (mergetree/compact/IntervalPartition.java:67-125): given n files'
  * (minKey, maxKey), writes section id and run-within-section id per file

@@ -10,6 +10,7 @@ from .reader import (  # noqa: F401
     debug_footer,
     interval_partition,
     file_descs_from_metas,
+    write_parquet,
     LIB_PATH,
 )
 

@@ -5,8 +5,9 @@ MergeTreeCompactTask.doCompact (mergetree/compact/MergeTreeCompactTask.java:
 83-162).
 
 v1 shape (SURVEY §8a "Write-back" row): the merge runs on the GPU through
-the same C-ABI plan as the read path; the Parquet *encode* side is CPU
-(pyarrow) — GPU encode is roadmap row §8f.1. Rolling files + per-file
+the same C-ABI plan as the read path; the Parquet *encode* side runs in
+libpaimon_hip's native writer (parquet_write.cpp: PLAIN v1 pages, thrift
+footer) — on-GPU page encode/compress is roadmap row §8f.1. Rolling files + per-file
 DataFileMeta stats follow KeyValueDataFileWriter (io/KeyValueDataFileWriter.
 java:121-170: rowCount, minKey/maxKey copies, min/max sequenceNumber,
 deleteRecordCount) and RollingFileWriterImpl (roll at target row count).
@@ -15,10 +16,9 @@ deleteRecordCount) and RollingFileWriterImpl (roll at target row count).
 import os
 
 import numpy as np
-import pyarrow as pa
-import pyarrow.parquet as pq
 
-from .reader import MergeReadPlan, Session, file_descs_from_metas
+from .reader import (MergeReadPlan, Session, file_descs_from_metas,
+                     write_parquet)
 
 KIND_IS_ADD = (0, 2)
 
@@ -60,26 +60,21 @@ def rewrite(session: Session, file_metas, key_cols, value_cols, out_dir,
             seq = batch["_SEQUENCE_NUMBER"]
             kind = batch["_VALUE_KIND"]
             for s, e in _roll_slices(n, target_file_rows):
-                fields = [pa.field("_KEY_k", pa.int64(), nullable=False),
-                          pa.field("_SEQUENCE_NUMBER", pa.int64(),
-                                   nullable=False),
-                          pa.field("_VALUE_KIND", pa.int8(), nullable=False)]
-                cols = [pa.array(key[s:e]), pa.array(seq[s:e]),
-                        pa.array(kind[s:e])]
+                cols = [("_KEY_k", key[s:e]), ("_SEQUENCE_NUMBER", seq[s:e]),
+                        ("_VALUE_KIND", kind[s:e])]
                 for name, arr in batch.items():
-                    if name in ("_KEY_k", "_SEQUENCE_NUMBER", "_VALUE_KIND"):
+                    if name in ("_KEY_k", "_SEQUENCE_NUMBER", "_VALUE_KIND") \
+                            or name.endswith("#valid"):
                         continue
-                    fields.append(pa.field(name, pa.from_numpy_dtype(arr.dtype),
-                                           nullable=True))
-                    cols.append(pa.array(arr[s:e]))
-                tbl = pa.Table.from_arrays(cols, schema=pa.schema(fields))
+                    valid = batch.get(name + "#valid")
+                    cols.append((name, arr[s:e],
+                                 valid[s:e] if valid is not None else None))
                 path = os.path.join(out_dir, f"{file_prefix}-{file_idx}.parquet")
                 file_idx += 1
-                pq.write_table(
-                    tbl, path,
-                    compression=None if compression == "NONE" else compression,
-                    use_dictionary=False, data_page_version="1.0",
-                    write_statistics=False, store_schema=False)
+                if compression != "NONE":
+                    raise ValueError("native compact write is uncompressed "
+                                     "v1 (compressed encode: roadmap §8f.1)")
+                write_parquet(path, cols)
                 ks = key[s:e]
                 sq = seq[s:e]
                 kd = kind[s:e]

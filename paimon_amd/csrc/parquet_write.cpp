@@ -1,0 +1,303 @@
+// Host-side Parquet v1 writer: PLAIN encoding, uncompressed, thrift-compact
+// footer/page headers — the write-back half of the compaction surface.
+//
+// Replaces, for this path, the reference's format writer stack
+// (paimon-format/src/main/java/org/apache/paimon/format/parquet/
+// ParquetWriterFactory.java + the vendored parquet-mr writer) as used by
+// KeyValueDataFileWriter (io/KeyValueDataFileWriter.java:121-170) under
+// CompactRewriter. v1 matrix: INT8/16 (stored INT32), INT32, INT64, FLOAT,
+// DOUBLE; REQUIRED or OPTIONAL (byte validity -> RLE/bit-packed def levels);
+// one PLAIN data page per `page_rows`; no dictionary, no compression
+// (roadmap §8f.1). Readable by parquet-mr/pyarrow and by this library's own
+// reader (tests pin both).
+
+#include "parquet_write.h"
+
+#include <cstdio>
+#include <cstring>
+
+#include "parquet_meta.h"
+
+namespace pmh {
+
+namespace {
+
+// ---- thrift compact protocol writer (parquet.thrift field ids) ----
+struct TC {
+    std::string out;
+    void byte(uint8_t b) { out.push_back((char)b); }
+    void uvarint(uint64_t v) {
+        while (v >= 0x80) {
+            byte((uint8_t)(v | 0x80));
+            v >>= 7;
+        }
+        byte((uint8_t)v);
+    }
+    static uint64_t zz(int64_t v) {
+        return ((uint64_t)v << 1) ^ (uint64_t)(v >> 63);
+    }
+    // field header; caller tracks the previous field id per struct level
+    void field(int &last, int id, int type) {
+        int delta = id - last;
+        if (delta >= 1 && delta <= 15) {
+            byte((uint8_t)((delta << 4) | type));
+        } else {
+            byte((uint8_t)type);
+            uvarint(zz(id));
+        }
+        last = id;
+    }
+    void stop() { byte(0); }
+    void i32(int &last, int id, int32_t v) {
+        field(last, id, 5);
+        uvarint(zz(v));
+    }
+    void i64(int &last, int id, int64_t v) {
+        field(last, id, 6);
+        uvarint(zz(v));
+    }
+    void str(int &last, int id, const std::string &s) {
+        field(last, id, 8);
+        uvarint(s.size());
+        out.append(s);
+    }
+    void list_begin(int &last, int id, int elem_type, size_t n) {
+        field(last, id, 9);
+        if (n < 15) byte((uint8_t)((n << 4) | elem_type));
+        else {
+            byte((uint8_t)(0xF0 | elem_type));
+            uvarint(n);
+        }
+    }
+};
+
+int phys_of(int dtype) {
+    switch (dtype) {
+    case 1:
+    case 2:
+    case 3: return PHYS_INT32;
+    case 4: return PHYS_INT64;
+    case 5: return PHYS_FLOAT;
+    case 6: return PHYS_DOUBLE;
+    }
+    return -1;
+}
+
+int converted_of(int dtype) {  // ConvertedType; -1 = none
+    switch (dtype) {
+    case 1: return 15;  // INT_8
+    case 2: return 16;  // INT_16
+    }
+    return -1;
+}
+
+// RLE/bit-packed hybrid def levels (bit width 1), with the 4-byte LE length
+// prefix of v1 data pages. All-valid pages use one RLE run; mixed pages use
+// one bit-packed literal (LSB-first, zero-padded).
+void encode_def_levels(const uint8_t *valid, int64_t n, std::string &out) {
+    bool all = true;
+    for (int64_t i = 0; i < n && all; i++) all = valid[i] != 0;
+    std::string body;
+    if (all) {
+        TC t;
+        t.uvarint((uint64_t)n << 1);
+        body = t.out;
+        body.push_back((char)1);
+    } else {
+        int64_t groups = (n + 7) / 8;
+        TC t;
+        t.uvarint(((uint64_t)groups << 1) | 1);
+        body = t.out;
+        for (int64_t g = 0; g < groups; g++) {
+            uint8_t b = 0;
+            for (int j = 0; j < 8; j++) {
+                int64_t i = g * 8 + j;
+                if (i < n && valid[i]) b |= (uint8_t)(1u << j);
+            }
+            body.push_back((char)b);
+        }
+    }
+    uint32_t len = (uint32_t)body.size();
+    out.push_back((char)(len & 0xFF));
+    out.push_back((char)((len >> 8) & 0xFF));
+    out.push_back((char)((len >> 16) & 0xFF));
+    out.push_back((char)((len >> 24) & 0xFF));
+    out.append(body);
+}
+
+// PLAIN payload for rows [s, e): non-null values only, int8/16 widened to
+// the INT32 physical type.
+void encode_values(const PwCol &c, int64_t s, int64_t e, std::string &out) {
+    for (int64_t i = s; i < e; i++) {
+        if (c.valid && !c.valid[i]) continue;
+        int32_t v32;
+        switch (c.dtype) {
+        case 1:
+            v32 = (int32_t)((const int8_t *)c.data)[i];
+            out.append((const char *)&v32, 4);
+            break;
+        case 2:
+            v32 = (int32_t)((const int16_t *)c.data)[i];
+            out.append((const char *)&v32, 4);
+            break;
+        case 3:
+        case 5:
+            out.append((const char *)c.data + i * 4, 4);
+            break;
+        case 4:
+        case 6:
+            out.append((const char *)c.data + i * 8, 8);
+            break;
+        }
+    }
+}
+
+void page_header(int64_t n_vals, int32_t payload, std::string &out) {
+    TC t;
+    int l0 = 0;
+    t.i32(l0, 1, 0);        // type = DATA_PAGE
+    t.i32(l0, 2, payload);  // uncompressed_page_size
+    t.i32(l0, 3, payload);  // compressed_page_size (uncompressed)
+    t.field(l0, 5, 12);     // data_page_header: struct
+    {
+        int l1 = 0;
+        t.i32(l1, 1, (int32_t)n_vals);
+        t.i32(l1, 2, ENC_PLAIN);  // encoding
+        t.i32(l1, 3, ENC_RLE);    // definition_level_encoding
+        t.i32(l1, 4, ENC_RLE);    // repetition_level_encoding
+        t.stop();
+    }
+    t.stop();
+    out.append(t.out);
+}
+
+}  // namespace
+
+bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
+                   const std::string &path, int64_t row_group_rows,
+                   int64_t page_rows, std::string &err) {
+    for (const auto &c : cols)
+        if (phys_of(c.dtype) < 0) {
+            err = "unsupported dtype " + std::to_string(c.dtype) +
+                  " for parquet write (v1 matrix: int8..int64/float/double)";
+            return false;
+        }
+    if (row_group_rows <= 0) row_group_rows = 1 << 20;
+    if (page_rows <= 0) page_rows = 1 << 16;
+    FILE *f = fopen(path.c_str(), "wb");
+    if (!f) {
+        err = "cannot open " + path + " for write";
+        return false;
+    }
+    std::string buf = "PAR1";
+
+    struct CcInfo {
+        int64_t data_page_offset;
+        int64_t total_size;
+        int64_t num_values;
+    };
+    struct RgInfo {
+        std::vector<CcInfo> ccs;
+        int64_t rows;
+        int64_t bytes;
+    };
+    std::vector<RgInfo> rgs;
+
+    for (int64_t rg0 = 0; rg0 < n_rows || (n_rows == 0 && rg0 == 0);
+         rg0 += row_group_rows) {
+        int64_t rg1 = rg0 + row_group_rows < n_rows ? rg0 + row_group_rows
+                                                    : n_rows;
+        RgInfo rg{};
+        rg.rows = rg1 - rg0;
+        for (const auto &c : cols) {
+            CcInfo cc{};
+            cc.data_page_offset = (int64_t)buf.size();
+            cc.num_values = rg1 - rg0;
+            for (int64_t p0 = rg0; p0 < rg1 || (rg1 == rg0 && p0 == rg0);
+                 p0 += page_rows) {
+                int64_t p1 = p0 + page_rows < rg1 ? p0 + page_rows : rg1;
+                std::string payload;
+                if (c.valid) encode_def_levels(c.valid + p0, p1 - p0, payload);
+                encode_values(c, p0, p1, payload);
+                page_header(p1 - p0, (int32_t)payload.size(), buf);
+                buf.append(payload);
+                if (rg1 == rg0) break;  // single empty page for 0 rows
+            }
+            cc.total_size = (int64_t)buf.size() - cc.data_page_offset;
+            rg.bytes += cc.total_size;
+            rg.ccs.push_back(cc);
+        }
+        rgs.push_back(rg);
+        if (n_rows == 0) break;
+    }
+
+    // footer: FileMetaData
+    TC t;
+    int l0 = 0;
+    t.i32(l0, 1, 1);  // version
+    t.list_begin(l0, 2, 12, cols.size() + 1);  // schema
+    {
+        int ls = 0;  // root group
+        TC &g = t;
+        g.i32(ls, 3, 0);  // repetition REQUIRED (ignored on root)
+        g.str(ls, 4, "schema");
+        g.i32(ls, 5, (int32_t)cols.size());  // num_children
+        g.stop();
+        for (const auto &c : cols) {
+            int lf = 0;
+            g.i32(lf, 1, phys_of(c.dtype));
+            g.i32(lf, 3, c.valid ? 1 : 0);  // OPTIONAL : REQUIRED
+            g.str(lf, 4, c.name);
+            int ct = converted_of(c.dtype);
+            if (ct >= 0) g.i32(lf, 6, ct);
+            g.stop();
+        }
+    }
+    t.i64(l0, 3, n_rows);
+    t.list_begin(l0, 4, 12, rgs.size());
+    for (const auto &rg : rgs) {
+        int lr = 0;
+        t.list_begin(lr, 1, 12, rg.ccs.size());
+        for (size_t ci = 0; ci < rg.ccs.size(); ci++) {
+            const CcInfo &cc = rg.ccs[ci];
+            const PwCol &c = cols[ci];
+            int lc = 0;
+            t.i64(lc, 2, cc.data_page_offset);  // file_offset
+            t.field(lc, 3, 12);                 // meta_data: struct
+            {
+                int lm = 0;
+                t.i32(lm, 1, phys_of(c.dtype));
+                t.list_begin(lm, 2, 5, 2);  // encodings
+                t.uvarint(TC::zz(ENC_PLAIN));
+                t.uvarint(TC::zz(ENC_RLE));
+                t.list_begin(lm, 3, 8, 1);  // path_in_schema
+                t.uvarint(c.name.size());
+                t.out.append(c.name);
+                t.i32(lm, 4, CODEC_UNCOMPRESSED);
+                t.i64(lm, 5, cc.num_values);
+                t.i64(lm, 6, cc.total_size);  // total_uncompressed_size
+                t.i64(lm, 7, cc.total_size);  // total_compressed_size
+                t.i64(lm, 9, cc.data_page_offset);
+                t.stop();
+            }
+            t.stop();
+        }
+        t.i64(lr, 2, rg.bytes);
+        t.i64(lr, 3, rg.rows);
+        t.stop();
+    }
+    t.str(l0, 6, "paimon-hip v1");
+    t.stop();
+
+    buf.append(t.out);
+    uint32_t flen = (uint32_t)t.out.size();
+    buf.append((const char *)&flen, 4);
+    buf.append("PAR1");
+
+    bool ok = fwrite(buf.data(), 1, buf.size(), f) == buf.size();
+    fclose(f);
+    if (!ok) err = "short write to " + path;
+    return ok;
+}
+
+}  // namespace pmh

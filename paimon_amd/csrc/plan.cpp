@@ -26,6 +26,7 @@
 #include "kernels.h"
 #include "orc_meta.h"
 #include "parquet_meta.h"
+#include "parquet_write.h"
 
 #include <dlfcn.h>
 
@@ -1723,6 +1724,32 @@ int pmh_stats_get(pmh_plan_t *p, pmh_stats *out) {
 }
 
 void pmh_free_string(char *s) { free(s); }
+
+int pmh_write_parquet(const pmh_col *cols, int32_t n_cols, int64_t n_rows,
+                      const char *path, int64_t row_group_rows,
+                      int64_t page_rows) {
+    if (!cols || n_cols <= 0 || !path) {
+        set_error("pmh_write_parquet: bad arguments");
+        return -1;
+    }
+    std::vector<PwCol> pc(n_cols);
+    for (int32_t i = 0; i < n_cols; i++) {
+        pc[i].name = cols[i].name ? cols[i].name : "";
+        pc[i].dtype = cols[i].dtype;
+        pc[i].data = cols[i].data;
+        pc[i].valid = cols[i].valid;
+        if (pc[i].name.empty() || (!pc[i].data && n_rows > 0)) {
+            set_error("pmh_write_parquet: column %d missing name/data", i);
+            return -1;
+        }
+    }
+    std::string err;
+    if (!write_parquet(pc, n_rows, path, row_group_rows, page_rows, err)) {
+        set_error("%s", err.c_str());
+        return -1;
+    }
+    return 0;
+}
 
 char *pmh_debug_footer_json(const char *path) {
     StagedFile sf;

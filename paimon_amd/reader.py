@@ -79,6 +79,10 @@ def load_lib(required=True):
     lib.pmh_debug_footer_json.restype = ctypes.c_void_p
     lib.pmh_debug_footer_json.argtypes = [ctypes.c_char_p]
     lib.pmh_free_string.argtypes = [ctypes.c_void_p]
+    lib.pmh_write_parquet.restype = ctypes.c_int
+    lib.pmh_write_parquet.argtypes = [
+        ctypes.POINTER(_Col), ctypes.c_int32, ctypes.c_int64,
+        ctypes.c_char_p, ctypes.c_int64, ctypes.c_int64]
     lib.pmh_debug_interval_partition.restype = ctypes.c_int
     lib.pmh_debug_interval_partition.argtypes = [
         ctypes.c_int, ctypes.POINTER(ctypes.c_int64),
@@ -100,6 +104,45 @@ def debug_footer(path):
     s = ctypes.string_at(p).decode()
     lib.pmh_free_string(p)
     return json.loads(s)
+
+
+_NP_DT = {np.dtype(np.int8): 1, np.dtype(np.int16): 2,
+          np.dtype(np.int32): 3, np.dtype(np.int64): 4,
+          np.dtype(np.float32): 5, np.dtype(np.float64): 6}
+
+
+def write_parquet(path, columns, row_group_rows=0, page_rows=0):
+    """Write a Parquet v1 data file (PLAIN, uncompressed) via the native
+    writer (pmh_write_parquet) — the compaction write-back path. `columns`
+    is an ordered list of (name, values[, valid]) with numpy arrays; valid
+    None/omitted = REQUIRED column."""
+    lib = load_lib()
+    n_rows = len(columns[0][1]) if columns else 0
+    arrs = []  # keep contiguous buffers alive
+    cols = (_Col * len(columns))()
+    for i, col in enumerate(columns):
+        name, vals = col[0], np.ascontiguousarray(col[1])
+        valid = col[2] if len(col) > 2 else None
+        if vals.dtype not in _NP_DT:
+            raise ValueError(f"unsupported dtype {vals.dtype} for {name}")
+        if len(vals) != n_rows:
+            raise ValueError("ragged columns")
+        arrs.append(vals)
+        cols[i].name = name.encode()
+        cols[i].dtype = _NP_DT[vals.dtype]
+        cols[i].data = vals.ctypes.data_as(ctypes.c_void_p)
+        if valid is not None:
+            v8 = np.ascontiguousarray(np.asarray(valid, dtype=np.uint8))
+            if len(v8) != n_rows:
+                raise ValueError("ragged validity")
+            arrs.append(v8)
+            cols[i].valid = v8.ctypes.data_as(ctypes.c_void_p)
+        else:
+            cols[i].valid = None
+    rc = lib.pmh_write_parquet(cols, len(columns), n_rows, path.encode(),
+                               row_group_rows, page_rows)
+    if rc != 0:
+        raise RuntimeError(f"pmh_write_parquet: {last_error()}")
 
 
 def interval_partition(min_keys, max_keys):

@@ -1,0 +1,103 @@
+"""Native Parquet writer (parquet_write.cpp — the CompactRewriter write-back
+half) pinned against TWO independent readers on CPU:
+  1. pyarrow (an independent Parquet implementation);
+  2. this repo's own reader stack — the C++ footer parser via
+     pmh_debug_footer_json, and the oracle's pure-Python page decoder
+     (oracle/parquet_decode.py, itself pinned to pyarrow) — which also proves
+     the GPU read path can consume compaction output, since both share the
+     same wire format.
+"""
+
+import os
+
+import numpy as np
+import pytest
+
+import pyarrow.parquet as pq
+
+from paimon_amd import debug_footer, write_parquet
+from oracle.parquet_decode import read_file
+
+
+def _rand_cols(rng, n):
+    msk = rng.random(n) > 0.3
+    return [
+        ("k", np.sort(rng.choice(10 ** 12, n, replace=False)).astype(np.int64)),
+        ("v32", rng.integers(-2 ** 31, 2 ** 31, n).astype(np.int32), msk),
+        ("v8", rng.integers(-128, 128, n).astype(np.int8)),
+        ("v16", rng.integers(-2 ** 15, 2 ** 15, n).astype(np.int16), msk),
+        ("f32", rng.standard_normal(n).astype(np.float32)),
+        ("f64", rng.standard_normal(n), msk),
+    ]
+
+
+def _check_pyarrow(path, cols, n):
+    t = pq.read_table(path)
+    assert t.num_rows == n
+    for col in cols:
+        name, vals = col[0], col[1]
+        valid = col[2] if len(col) > 2 else None
+        got_valid = t[name].is_valid().to_numpy(zero_copy_only=False)
+        if valid is None:
+            assert got_valid.all(), name
+            got = t[name].to_numpy(zero_copy_only=False)
+            assert got.dtype == vals.dtype, name
+            assert (got == vals).all(), name
+        else:
+            assert (got_valid == valid).all(), name
+            got = t[name].to_numpy(zero_copy_only=False)
+            assert (got[valid] == vals[valid]).all(), name
+
+
+class TestNativeParquetWriter:
+    def test_pyarrow_roundtrip(self, tmp_path):
+        rng = np.random.default_rng(7)
+        n = 120_000
+        cols = _rand_cols(rng, n)
+        p = str(tmp_path / "t.parquet")
+        write_parquet(p, cols, row_group_rows=50_000, page_rows=9_000)
+        assert pq.ParquetFile(p).num_row_groups == 3
+        _check_pyarrow(p, cols, n)
+
+    def test_own_reader_roundtrip(self, tmp_path):
+        rng = np.random.default_rng(8)
+        n = 40_000
+        cols = _rand_cols(rng, n)
+        p = str(tmp_path / "t.parquet")
+        write_parquet(p, cols, row_group_rows=16_000, page_rows=3_000)
+        meta = debug_footer(p)  # the C++ thrift parser reads it
+        assert meta["num_rows"] == n
+        assert [c["name"] for c in meta["columns"]] == [c[0] for c in cols]
+        decoded = read_file(p)  # the oracle's pure-python page decoder
+        for col in cols:
+            name, vals = col[0], col[1]
+            valid = col[2] if len(col) > 2 else np.ones(n, bool)
+            got, got_valid = decoded[name]
+            assert (np.asarray(got_valid, bool) == valid).all(), name
+            g = np.asarray(got)
+            v = vals.astype(g.dtype)
+            assert (g[valid] == v[valid]).all(), name
+
+    def test_edge_cases(self, tmp_path):
+        # zero rows, one row, all-null column, sub-page file
+        p0 = str(tmp_path / "zero.parquet")
+        write_parquet(p0, [("k", np.empty(0, np.int64))])
+        assert pq.read_table(p0).num_rows == 0
+        p1 = str(tmp_path / "one.parquet")
+        write_parquet(p1, [("k", np.array([42], np.int64)),
+                           ("v", np.array([7], np.int32),
+                            np.array([True]))])
+        t = pq.read_table(p1)
+        assert t["k"].to_pylist() == [42] and t["v"].to_pylist() == [7]
+        pn = str(tmp_path / "null.parquet")
+        n = 5_000
+        write_parquet(pn, [("k", np.arange(n, dtype=np.int64)),
+                           ("v", np.zeros(n, np.int32),
+                            np.zeros(n, bool))])
+        t = pq.read_table(pn)
+        assert t["v"].null_count == n
+
+    def test_unsupported_dtype_rejected(self, tmp_path):
+        with pytest.raises(ValueError, match="unsupported dtype"):
+            write_parquet(str(tmp_path / "x.parquet"),
+                          [("k", np.zeros(4, np.uint64))])
