@@ -1,0 +1,119 @@
+#include "codec.h"
+
+#include <dlfcn.h>
+#include <zlib.h>
+
+#include <cstring>
+
+namespace pmh {
+
+namespace {
+typedef size_t (*zstd_fn)(void *, size_t, const void *, size_t);
+typedef unsigned (*zstd_iserror_fn)(size_t);
+
+bool zstd_call(const uint8_t *src, size_t n, uint8_t *dst, size_t cap,
+               size_t &got, std::string &err) {
+    static zstd_fn fn = nullptr;
+    static zstd_iserror_fn err_fn = nullptr;
+    if (!fn) {
+        void *h = dlopen("libzstd.so.1", RTLD_NOW | RTLD_GLOBAL);
+        if (!h) {
+            err = "libzstd.so.1 not found for zstd-compressed input";
+            return false;
+        }
+        fn = (zstd_fn)dlsym(h, "ZSTD_decompress");
+        err_fn = (zstd_iserror_fn)dlsym(h, "ZSTD_isError");
+        if (!fn || !err_fn) {
+            err = "ZSTD_decompress symbol missing";
+            return false;
+        }
+    }
+    size_t r = fn(dst, cap, src, n);
+    if (err_fn(r)) {
+        err = "zstd decompress failed";
+        return false;
+    }
+    got = r;
+    return true;
+}
+
+bool inflate_raw(const uint8_t *src, size_t n, uint8_t *dst, size_t cap,
+                 size_t &got, std::string &err) {
+    z_stream zs;
+    std::memset(&zs, 0, sizeof(zs));
+    if (inflateInit2(&zs, -15) != Z_OK) {  // raw deflate (ORC ZLIB framing)
+        err = "inflateInit2 failed";
+        return false;
+    }
+    zs.next_in = const_cast<Bytef *>(src);
+    zs.avail_in = (uInt)n;
+    zs.next_out = dst;
+    zs.avail_out = (uInt)cap;
+    int rc = inflate(&zs, Z_FINISH);
+    got = zs.total_out;
+    inflateEnd(&zs);
+    if (rc != Z_STREAM_END) {
+        err = "raw deflate decode failed (rc " + std::to_string(rc) + ")";
+        return false;
+    }
+    return true;
+}
+}  // namespace
+
+bool zstd_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
+                           size_t dst_n, std::string &err) {
+    size_t got = 0;
+    if (!zstd_call(src, n, dst, dst_n, got, err)) return false;
+    if (got != dst_n) {
+        err = "zstd decompressed size mismatch (" + std::to_string(got) +
+              " != " + std::to_string(dst_n) + ")";
+        return false;
+    }
+    return true;
+}
+
+bool orc_decompress(const uint8_t *src, int64_t len, int kind,
+                    int64_t block_size, std::vector<uint8_t> &out,
+                    std::string &err) {
+    if (kind != 1 && kind != 5) {
+        err = "ORC compression kind " + std::to_string(kind) +
+              " not supported (v1: NONE, ZLIB, ZSTD)";
+        return false;
+    }
+    if (block_size <= 0) block_size = 256 * 1024;
+    out.clear();
+    int64_t p = 0;
+    while (p < len) {
+        if (p + 3 > len) {
+            err = "truncated ORC compression chunk header";
+            return false;
+        }
+        uint32_t h = (uint32_t)src[p] | ((uint32_t)src[p + 1] << 8) |
+                     ((uint32_t)src[p + 2] << 16);
+        bool original = h & 1;
+        int64_t clen = h >> 1;
+        p += 3;
+        if (p + clen > len) {
+            err = "ORC compression chunk overruns stream";
+            return false;
+        }
+        if (original) {
+            out.insert(out.end(), src + p, src + p + clen);
+        } else {
+            size_t old = out.size();
+            out.resize(old + block_size);
+            size_t got = 0;
+            bool ok = kind == 1
+                          ? inflate_raw(src + p, clen, out.data() + old,
+                                        block_size, got, err)
+                          : zstd_call(src + p, clen, out.data() + old,
+                                      block_size, got, err);
+            if (!ok) return false;
+            out.resize(old + got);
+        }
+        p += clen;
+    }
+    return true;
+}
+
+}  // namespace pmh

@@ -1,5 +1,7 @@
 #include "orc_meta.h"
 
+#include "codec.h"
+
 #include <cstring>
 #include <map>
 #include <stdexcept>
@@ -112,15 +114,25 @@ OrcFileMeta parse_orc_meta(const uint8_t *data, int64_t size) {
         PbMsg ps = pb_parse(data + size - 1 - ps_len, ps_len);
         int64_t footer_len = (int64_t)pb_int(ps, 1);
         out.compression = (int)pb_int(ps, 2);
-        if (out.compression != 0) {
-            out.error =
-                "compressed ORC not supported yet (write with "
-                "compression='uncompressed'; on-GPU/codec support is a later "
-                "round)";
-            return out;
-        }
+        out.compression_block_size = (int64_t)pb_int(ps, 3);
+        // file footer and stripe footers share the stream compression
+        // framing (ORC spec "Compression"); streams themselves decompress
+        // at staging (plan.cpp)
+        std::string cerr;
+        std::vector<uint8_t> fdec;
         int64_t fstart = size - 1 - ps_len - footer_len;
-        PbMsg footer = pb_parse(data + fstart, footer_len);
+        const uint8_t *fptr = data + fstart;
+        int64_t flen = footer_len;
+        if (out.compression != 0) {
+            if (!orc_decompress(fptr, flen, out.compression,
+                                out.compression_block_size, fdec, cerr)) {
+                out.error = cerr;
+                return out;
+            }
+            fptr = fdec.data();
+            flen = (int64_t)fdec.size();
+        }
+        PbMsg footer = pb_parse(fptr, flen);
         out.num_rows = (int64_t)pb_int(footer, 6);
         // types (field 4, repeated Type)
         std::vector<PbMsg> types;
@@ -156,7 +168,20 @@ OrcFileMeta parse_orc_meta(const uint8_t *data, int64_t size) {
                 st.footer_length = (int64_t)pb_int(sm, 4);
                 st.num_rows = (int64_t)pb_int(sm, 5);
                 int64_t sf_off = st.offset + st.index_length + st.data_length;
-                PbMsg spf = pb_parse(data + sf_off, st.footer_length);
+                std::vector<uint8_t> sdec;
+                const uint8_t *sptr = data + sf_off;
+                int64_t slen = st.footer_length;
+                if (out.compression != 0) {
+                    if (!orc_decompress(sptr, slen, out.compression,
+                                        out.compression_block_size, sdec,
+                                        cerr)) {
+                        out.error = cerr;
+                        return out;
+                    }
+                    sptr = sdec.data();
+                    slen = (int64_t)sdec.size();
+                }
+                PbMsg spf = pb_parse(sptr, slen);
                 int64_t pos = st.offset;
                 auto stit = spf.find(1);
                 if (stit != spf.end()) {

@@ -51,7 +51,8 @@ struct OrcStripe {
 
 struct OrcFileMeta {
     int64_t num_rows = 0;
-    int compression = 0;  // 0 = NONE (only supported)
+    int compression = 0;  // CompressionKind: 0 NONE, 1 ZLIB, 5 ZSTD (v1 set)
+    int64_t compression_block_size = 0;
     std::vector<std::string> column_names;  // flat struct; col id = idx + 1
     std::vector<int> column_kinds;
     std::vector<OrcStripe> stripes;

@@ -21,6 +21,7 @@
 #include <vector>
 
 #include "../../include/paimon_hip.h"
+#include "codec.h"
 #include "common.h"
 #include "json.h"
 #include "kernels.h"
@@ -47,27 +48,13 @@ std::string &last_error() {
         }                                                                 \
     } while (0)
 
-// ------------------------------------------------------------------ zstd
-
-typedef size_t (*zstd_decompress_fn)(void *, size_t, const void *, size_t);
-typedef unsigned (*zstd_iserror_fn)(size_t);
+// ---------------------------------------------------------------- codecs
 
 static bool zstd_decompress(const uint8_t *src, size_t src_n, uint8_t *dst,
                             size_t dst_n) {
-    static zstd_decompress_fn fn = nullptr;
-    static zstd_iserror_fn err_fn = nullptr;
-    if (!fn) {
-        void *h = dlopen("libzstd.so.1", RTLD_NOW | RTLD_GLOBAL);
-        if (!h) {
-            set_error("libzstd.so.1 not found for zstd-compressed pages");
-            return false;
-        }
-        fn = (zstd_decompress_fn)dlsym(h, "ZSTD_decompress");
-        err_fn = (zstd_iserror_fn)dlsym(h, "ZSTD_isError");
-    }
-    size_t r = fn(dst, dst_n, src, src_n);
-    if (err_fn(r) || r != dst_n) {
-        set_error("zstd decompress failed (%zu != %zu)", r, dst_n);
+    std::string err;
+    if (!zstd_decompress_exact(src, src_n, dst, dst_n, err)) {
+        set_error("%s", err.c_str());
         return false;
     }
     return true;
@@ -791,16 +778,48 @@ static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
                           fd.path.c_str(), cols[c].name.c_str());
                 return false;
             }
+            // compressed streams decompress on the host at staging, like
+            // the parquet zstd path (on-GPU codecs: §8f); the RLEv2/byte-RLE
+            // bytes then upload verbatim
+            const uint8_t *data_ptr = sf.data.data() + data->offset;
+            int64_t data_len = data->length;
+            const uint8_t *pres_ptr =
+                present ? sf.data.data() + present->offset : nullptr;
+            int64_t pres_len = present ? present->length : 0;
+            std::vector<uint8_t> data_dec, pres_dec;
+            if (om.compression != 0) {
+                std::string cerr;
+                if (!orc_decompress(data_ptr, data_len, om.compression,
+                                    om.compression_block_size, data_dec,
+                                    cerr)) {
+                    set_error("%s col %s: %s", fd.path.c_str(),
+                              cols[c].name.c_str(), cerr.c_str());
+                    return false;
+                }
+                data_ptr = data_dec.data();
+                data_len = (int64_t)data_dec.size();
+                if (present) {
+                    if (!orc_decompress(pres_ptr, pres_len, om.compression,
+                                        om.compression_block_size, pres_dec,
+                                        cerr)) {
+                        set_error("%s col %s PRESENT: %s", fd.path.c_str(),
+                                  cols[c].name.c_str(), cerr.c_str());
+                        return false;
+                    }
+                    pres_ptr = pres_dec.data();
+                    pres_len = (int64_t)pres_dec.size();
+                }
+            }
             // upload the encoded DATA stream (+16 B pad: the RLEv2
             // bit reader uses an aligned 16-byte window)
-            void *dev = plan->bufs.alloc(data->length + 16);
+            void *dev = plan->bufs.alloc(data_len + 16);
             if (!dev) return false;
-            if (hipMemcpy(dev, sf.data.data() + data->offset, data->length,
-                          hipMemcpyHostToDevice) != hipSuccess) {
+            if (hipMemcpy(dev, data_ptr, data_len, hipMemcpyHostToDevice) !=
+                hipSuccess) {
                 set_error("H2D failed");
                 return false;
             }
-            plan->encoded_bytes_total += data->length;
+            plan->encoded_bytes_total += data_len;
             int64_t row0 = row_base + stripe_row;
             int64_t n_dense = st.num_rows;
             int64_t dense0 = row0;
@@ -808,8 +827,8 @@ static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
             if (present) {
                 rc.has_nulls = true;
                 int64_t before = rc.dense_before;
-                if (!prescan_present(sf.data.data() + present->offset,
-                                     present->length, st.num_rows, row0, rc))
+                if (!prescan_present(pres_ptr, pres_len, st.num_rows, row0,
+                                     rc))
                     return false;
                 n_dense = rc.dense_before - before;
                 dense0 = before;
@@ -817,14 +836,12 @@ static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
             }
             bool ok;
             if (ckind == ORC_BYTE) {
-                ok = prescan_byterle(sf.data.data() + data->offset,
-                                     data->length, n_dense, stored,
+                ok = prescan_byterle(data_ptr, data_len, n_dense, stored,
                                      dense_target, (uint64_t)dev, dense0,
                                      rc.rlev2_host);
             } else if (ckind == ORC_SHORT || ckind == ORC_INT ||
                        ckind == ORC_LONG) {
-                ok = prescan_rlev2(sf.data.data() + data->offset,
-                                   data->length, n_dense, 1, stored,
+                ok = prescan_rlev2(data_ptr, data_len, n_dense, 1, stored,
                                    dense_target, (uint64_t)dev, dense0,
                                    rc.rlev2_host);
             } else {

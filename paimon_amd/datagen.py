@@ -149,7 +149,10 @@ def write_runs(runs, out_dir, compression="NONE", row_group_rows=1 << 20,
         path = os.path.join(out_dir, f"run-{r}.{file_format}")
         if file_format == "orc":
             from pyarrow import orc as pa_orc
-            pa_orc.write_table(tbl, path, compression="uncompressed")
+            pa_orc.write_table(
+                tbl, path,
+                compression="uncompressed" if compression == "NONE"
+                else compression)
         else:
             pq.write_table(
                 tbl, path,

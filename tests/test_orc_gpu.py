@@ -31,8 +31,9 @@ def _read_all(plan):
 
 
 class TestOrcDedup:
-    def _run(self, tmp_path, runs, **kw):
-        metas = write_runs(runs, str(tmp_path), file_format="orc")
+    def _run(self, tmp_path, runs, compression="NONE", **kw):
+        metas = write_runs(runs, str(tmp_path), file_format="orc",
+                           compression=compression)
         r, w = merge_dedup(runs, **kw)
         with Session(0) as s:
             with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
@@ -84,6 +85,18 @@ class TestOrcDedup:
                          "values": [keys.copy(), v]})
         self._run(tmp_path, runs)
 
+    def test_orc_zlib(self, tmp_path):
+        # ORC's default codec: streams + footers arrive chunk-compressed
+        # (host decompress at staging, ORC spec "Compression")
+        runs = gen_runs_dedup(6, 30_000, n_value_cols=3, seed=88,
+                              delete_frac=0.1)
+        self._run(tmp_path, runs, compression="zlib")
+
+    def test_orc_zstd(self, tmp_path):
+        runs = gen_runs_dedup(5, 25_000, n_value_cols=2, seed=89,
+                              delete_frac=0.15)
+        self._run(tmp_path, runs, compression="zstd")
+
     def test_orc_keep_delete(self, tmp_path):
         runs = gen_runs_dedup(4, 15_000, n_value_cols=2, seed=84,
                               delete_frac=0.3)
@@ -95,7 +108,8 @@ class TestOrcPartialUpdate:
         # C3 proper: ORC + PartialUpdate + nulls (PRESENT streams on device)
         runs = gen_runs_partial_update(4, 25_000, n_value_cols=8, seed=85,
                                        update_frac=0.3, update_cols=3)
-        metas = write_runs(runs, str(tmp_path), file_format="orc")
+        metas = write_runs(runs, str(tmp_path), file_format="orc",
+                           compression="zlib")
         exp = partial_update_model(runs)
         with Session(0) as s:
             with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
