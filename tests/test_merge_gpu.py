@@ -378,3 +378,28 @@ class TestFirstRowEngine:
                                merge_engine="first-row") as plan:
                 with pytest.raises(RuntimeError, match="first-row"):
                     plan.read_next()
+
+
+class TestNarrowIntColumns:
+    def test_int8_int16_columns(self, tmp_path):
+        # TINYINT/SMALLINT travel as INT32 physical and truncate on emit
+        runs = gen_runs_dedup(4, 20_000, n_value_cols=2, seed=67,
+                              delete_frac=0.1)
+        for r in runs:
+            r["values"][1] = (r["values"][1] % 127).astype(np.int8)
+            r["values"].append(
+                (r["values"][1].astype(np.int32) * 37 % 32_000)
+                .astype(np.int16))
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        r, w = merge_dedup(runs, drop_delete=True)
+        vcols = [{"name": "v_k", "type": "int64"},
+                 {"name": "v_c0", "type": "int8"},
+                 {"name": "v_c1", "type": "int16"}]
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               vcols) as plan:
+                got = _read_all_batches(plan)
+        for nm, c in (("v_c0", 1), ("v_c1", 2)):
+            e = np.array([runs[a]["values"][c][b] for a, b in zip(r, w)])
+            assert got[nm].dtype == e.dtype, nm
+            assert (got[nm] == e).all(), nm
