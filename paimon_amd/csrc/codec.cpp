@@ -72,6 +72,29 @@ bool zstd_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
     return true;
 }
 
+bool gzip_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
+                           size_t dst_n, std::string &err) {
+    z_stream zs;
+    std::memset(&zs, 0, sizeof(zs));
+    if (inflateInit2(&zs, 15 + 32) != Z_OK) {  // auto gzip/zlib wrapper
+        err = "inflateInit2 failed";
+        return false;
+    }
+    zs.next_in = const_cast<Bytef *>(src);
+    zs.avail_in = (uInt)n;
+    zs.next_out = dst;
+    zs.avail_out = (uInt)dst_n;
+    int rc = inflate(&zs, Z_FINISH);
+    size_t got = zs.total_out;
+    inflateEnd(&zs);
+    if (rc != Z_STREAM_END || got != dst_n) {
+        err = "gzip page decode failed (rc " + std::to_string(rc) + ", " +
+              std::to_string(got) + " != " + std::to_string(dst_n) + ")";
+        return false;
+    }
+    return true;
+}
+
 bool orc_decompress(const uint8_t *src, int64_t len, int kind,
                     int64_t block_size, std::vector<uint8_t> &out,
                     std::string &err) {

@@ -14,6 +14,11 @@ namespace pmh {
 bool zstd_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
                            size_t dst_n, std::string &err);
 
+// one-shot gzip/zlib-wrapped member with known decompressed size
+// (parquet GZIP pages)
+bool gzip_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
+                           size_t dst_n, std::string &err);
+
 // ORC chunked stream framing: 3-byte LE header (len << 1 | isOriginal) per
 // chunk, each decompressing to <= block_size bytes. kind: 1 = ZLIB (raw
 // deflate), 5 = ZSTD (ORC proto CompressionKind).

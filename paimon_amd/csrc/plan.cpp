@@ -160,9 +160,11 @@ struct DeviceBufs {
 //    time k_rle_decode + k_dict_gather materialize the chunk's row range of
 //    `contig` (decode_ms, inside the timed region).
 struct GatherTask {
-    int64_t start = 0;  // run-row offset of the chunk
+    int64_t start = 0;  // run-row offset (or dense offset, to_dense)
     int64_t n = 0;
     void *dict_dev = nullptr;
+    bool to_dense = false;  // nullable chunk: gather into the dense buffer,
+                            // k_level_scatter positions the rows after
 };
 
 struct RunCol {
@@ -181,6 +183,9 @@ struct RunCol {
     // time (timed decode) and fills the byte-validity array.
     bool has_nulls = false;
     std::vector<uint8_t> dense_host;   // packed non-null values (staging)
+    // (dense_start, nbytes) per dense_host append — PLAIN and dictionary
+    // null-chunks can interleave, so host segments land at their own offsets
+    std::vector<std::pair<int64_t, int64_t>> dense_segs;
     std::vector<uint8_t> levels_host;  // packed def-level streams (staging)
     std::vector<RleChunk> def_host;    // src = RELATIVE offset until upload
     int64_t dense_before = 0;          // running non-null count
@@ -909,21 +914,32 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                 int64_t chunk_start = cc.dictionary_page_offset
                                           ? cc.dictionary_page_offset
                                           : cc.data_page_offset;
-                std::vector<uint8_t> packed;  // zstd only
+                std::vector<uint8_t> packed;  // compressed codecs
                 std::vector<int64_t> ppo(cc.pages.size());
                 const uint8_t *payload_base = sf.data.data() + chunk_start;
-                if (cc.codec == CODEC_ZSTD) {
+                if (cc.codec == CODEC_ZSTD || cc.codec == CODEC_GZIP) {
                     int64_t total_unc = 0;
                     for (auto &pg : cc.pages) total_unc += pg.uncompressed_size;
                     packed.resize(total_unc);
                     int64_t off = 0;
                     for (size_t pi = 0; pi < cc.pages.size(); pi++) {
                         auto &pg = cc.pages[pi];
-                        if (!zstd_decompress(sf.data.data() + pg.data_off,
-                                             pg.compressed_size,
-                                             packed.data() + off,
-                                             pg.uncompressed_size))
+                        std::string cerr;
+                        bool ok = cc.codec == CODEC_ZSTD
+                                      ? zstd_decompress(
+                                            sf.data.data() + pg.data_off,
+                                            pg.compressed_size,
+                                            packed.data() + off,
+                                            pg.uncompressed_size)
+                                      : gzip_decompress_exact(
+                                            sf.data.data() + pg.data_off,
+                                            pg.compressed_size,
+                                            packed.data() + off,
+                                            pg.uncompressed_size, cerr);
+                        if (!ok) {
+                            if (!cerr.empty()) set_error("%s", cerr.c_str());
                             return false;
+                        }
                         ppo[pi] = off;
                         off += pg.uncompressed_size;
                     }
@@ -980,12 +996,72 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     vpos[pi] = pos;
                 }
                 if (chunk_nulls && has_dict) {
-                    set_error("%s col %s: nulls in dictionary-encoded chunks "
-                              "not supported yet",
-                              fd.path.c_str(), cols[c].name.c_str());
-                    return false;
-                }
-                if (chunk_nulls) {
+                    // dictionary chunk with nulls: def-level streams stay
+                    // encoded; the id stream decodes to DENSE positions on
+                    // the GPU (k_rle_decode), k_dict_gather fills the dense
+                    // buffer, and k_level_scatter positions the rows
+                    rc.has_nulls = true;
+                    if (!dict_host) {
+                        set_error("%s: dictionary page missing",
+                                  fd.path.c_str());
+                        return false;
+                    }
+                    rc.dict_encoded = true;
+                    int64_t payload_len = 0;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++)
+                        payload_len = std::max(
+                            payload_len,
+                            ppo[pi] + (cc.codec != CODEC_UNCOMPRESSED
+                                           ? cc.pages[pi].uncompressed_size
+                                           : cc.pages[pi].compressed_size));
+                    void *dev = plan->bufs.alloc(payload_len);
+                    if (!dev) return false;
+                    if (hipMemcpy(dev, payload_base, payload_len,
+                                  hipMemcpyHostToDevice) != hipSuccess) {
+                        set_error("H2D failed");
+                        return false;
+                    }
+                    plan->encoded_bytes_total += payload_len;
+                    int64_t dense_start = rc.dense_before;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                        auto &pg = cc.pages[pi];
+                        if (pg.page_type != 0) continue;
+                        const uint8_t *pp = payload_base + ppo[pi];
+                        uint32_t dl_len;
+                        memcpy(&dl_len, pp, 4);
+                        int64_t rel = (int64_t)rc.levels_host.size();
+                        rc.levels_host.insert(rc.levels_host.end(), pp + 4,
+                                              pp + 4 + dl_len);
+                        int64_t before = rc.dense_before;
+                        if (!prescan_def(pp + 4, dl_len, pg.num_values,
+                                         chunk_row0 + pg.first_row, rel,
+                                         &rc.dense_before, rc.def_host))
+                            return false;
+                        int64_t nvalid = rc.dense_before - before;
+                        int64_t pos = vpos[pi];
+                        int bw = pp[pos];
+                        int64_t plen = (cc.codec != CODEC_UNCOMPRESSED
+                                            ? pg.uncompressed_size
+                                            : pg.compressed_size);
+                        if (!prescan_rle(pp + pos + 1, plen - pos - 1, bw,
+                                         nvalid, before, (uint64_t)dev,
+                                         ppo[pi] + pos + 1, rc.rle_host))
+                            return false;
+                    }
+                    void *dict_dev = plan->bufs.alloc(dict_count * stored);
+                    if (!dict_dev) return false;
+                    if (hipMemcpy(dict_dev, dict_host, dict_count * stored,
+                                  hipMemcpyHostToDevice) != hipSuccess) {
+                        set_error("H2D dict failed");
+                        return false;
+                    }
+                    GatherTask gt;
+                    gt.start = dense_start;
+                    gt.n = rc.dense_before - dense_start;
+                    gt.dict_dev = dict_dev;
+                    gt.to_dense = true;
+                    rc.gathers.push_back(gt);
+                } else if (chunk_nulls) {
                     // dense PLAIN values + def-level streams stay encoded;
                     // k_level_scatter positions them at read time
                     rc.has_nulls = true;
@@ -1007,6 +1083,8 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                         rc.dense_host.insert(
                             rc.dense_host.end(), pp + vpos[pi],
                             pp + vpos[pi] + nvalid * stored);
+                        rc.dense_segs.emplace_back(before,
+                                                   nvalid * stored);
                     }
                 } else if (has_plain || (!has_dict && cc.num_values > 0)) {
                     // pack PLAIN value payloads and copy into contig
@@ -1043,7 +1121,7 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     for (size_t pi = 0; pi < cc.pages.size(); pi++)
                         payload_len =
                             std::max(payload_len,
-                                     ppo[pi] + (cc.codec == CODEC_ZSTD
+                                     ppo[pi] + (cc.codec != CODEC_UNCOMPRESSED
                                                     ? cc.pages[pi].uncompressed_size
                                                     : cc.pages[pi].compressed_size));
                     void *dev = plan->bufs.alloc(payload_len);
@@ -1060,7 +1138,7 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                         const uint8_t *pp = payload_base + ppo[pi];
                         int64_t pos = vpos[pi];
                         int bw = pp[pos];
-                        int64_t plen = (cc.codec == CODEC_ZSTD
+                        int64_t plen = (cc.codec != CODEC_UNCOMPRESSED
                                             ? pg.uncompressed_size
                                             : pg.compressed_size);
                         if (!prescan_rle(pp + pos + 1, plen - pos - 1, bw,
@@ -1108,12 +1186,11 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                 return false;
         }
         if (rc.has_nulls) {
-            // parquet: dense values packed host-side (dense_host);
-            // ORC: dense values are produced on-device by k_rlev2
-            size_t dense_bytes = rc.orc_encoded
-                                     ? (size_t)rc.dense_before *
-                                           plan->cols[c].stored_esize
-                                     : rc.dense_host.size();
+            // parquet PLAIN: dense values packed host-side (dense_host);
+            // parquet dictionary / ORC: dense values are produced on-device
+            // (k_dict_gather / k_rlev2) — size by the global dense count
+            size_t dense_bytes =
+                (size_t)rc.dense_before * plan->cols[c].stored_esize;
             rc.valid_dev = (uint8_t *)plan->bufs.alloc(run.length);
             rc.dense_dev = plan->bufs.alloc(dense_bytes);
             void *levels_dev = plan->bufs.alloc(rc.levels_host.size());
@@ -1122,11 +1199,16 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
             // rows in non-null chunks of this column keep validity 1
             if (hipMemset(rc.valid_dev, 1, run.length) != hipSuccess)
                 return false;
-            if (!rc.dense_host.empty() &&
-                hipMemcpy(rc.dense_dev, rc.dense_host.data(),
-                          rc.dense_host.size(),
-                          hipMemcpyHostToDevice) != hipSuccess)
-                return false;
+            size_t ho = 0;
+            for (const auto &sg : rc.dense_segs) {
+                if (sg.second > 0 &&
+                    hipMemcpy((uint8_t *)rc.dense_dev +
+                                  sg.first * plan->cols[c].stored_esize,
+                              rc.dense_host.data() + ho, sg.second,
+                              hipMemcpyHostToDevice) != hipSuccess)
+                    return false;
+                ho += sg.second;
+            }
             if (hipMemcpy(levels_dev, rc.levels_host.data(),
                           rc.levels_host.size(),
                           hipMemcpyHostToDevice) != hipSuccess)
@@ -1560,9 +1642,13 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                     rc.rle_dev, (int64_t)rc.rle_host.size(), rc.ids_dev, st);
                 if (e != hipSuccess) return fail("rle_decode", e);
                 for (const GatherTask &gt : rc.gathers) {
-                    e = pmh_launch_dict_gather(
-                        rc.ids_dev + gt.start, gt.dict_dev, gt.n,
-                        (uint8_t *)rc.contig + gt.start * es, es, st);
+                    if (!gt.n) continue;
+                    uint8_t *dst = gt.to_dense
+                                       ? (uint8_t *)rc.dense_dev
+                                       : (uint8_t *)rc.contig;
+                    e = pmh_launch_dict_gather(rc.ids_dev + gt.start,
+                                               gt.dict_dev, gt.n,
+                                               dst + gt.start * es, es, st);
                     if (e != hipSuccess) return fail("dict_gather", e);
                 }
             }

@@ -403,3 +403,7 @@ class TestNarrowIntColumns:
             e = np.array([runs[a]["values"][c][b] for a, b in zip(r, w)])
             assert got[nm].dtype == e.dtype, nm
             assert (got[nm] == e).all(), nm
+
+    def test_gzip_compressed(self, tmp_path):
+        runs = gen_runs_dedup(4, 25_000, n_value_cols=3, seed=54)
+        _run_and_compare(tmp_path, runs, compression="gzip")
