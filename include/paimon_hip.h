@@ -138,6 +138,12 @@ int pmh_write_parquet(const pmh_col *cols, int32_t n_cols, int64_t n_rows,
                       const char *path, int64_t row_group_rows,
                       int64_t page_rows);
 
+/* Raw snappy block decode (the library's from-scratch decoder, used for
+ * SNAPPY parquet pages / ORC chunks). Returns decompressed size or -1 with
+ * pmh_last_error() set. Exposed so CPU tests can pin the decoder against an
+ * independent compressor. */
+int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap);
+
 /* Restatement of IntervalPartition.partition() for int64 keys
  * (mergetree/compact/IntervalPartition.java:67-125): given n files'
  * (minKey, maxKey), writes section id and run-within-section id per file

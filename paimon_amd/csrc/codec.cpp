@@ -72,6 +72,97 @@ bool zstd_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
     return true;
 }
 
+// raw snappy block format (no framing): uvarint decompressed length, then
+// literal/copy tags — implemented from scratch (no libsnappy in this image).
+// Format per google/snappy format_description.txt.
+bool snappy_decompress(const uint8_t *src, size_t n, uint8_t *dst,
+                       size_t cap, size_t &got, std::string &err) {
+    size_t p = 0;
+    uint64_t ulen = 0;
+    int shift = 0;
+    for (;;) {
+        if (p >= n) {
+            err = "snappy: truncated length";
+            return false;
+        }
+        uint8_t b = src[p++];
+        ulen |= (uint64_t)(b & 0x7F) << shift;
+        if (!(b & 0x80)) break;
+        shift += 7;
+    }
+    if (ulen > cap) {
+        err = "snappy: output exceeds capacity";
+        return false;
+    }
+    size_t o = 0;
+    while (p < n) {
+        uint8_t tag = src[p++];
+        size_t len;
+        if ((tag & 3) == 0) {  // literal
+            len = (tag >> 2) + 1;
+            if (len > 60) {
+                int extra = (int)len - 60;
+                if (p + extra > n) {
+                    err = "snappy: bad literal length";
+                    return false;
+                }
+                len = 0;
+                for (int i = 0; i < extra; i++)
+                    len |= (size_t)src[p + i] << (8 * i);
+                len += 1;
+                p += extra;
+            }
+            if (p + len > n || o + len > ulen) {
+                err = "snappy: literal overrun";
+                return false;
+            }
+            std::memcpy(dst + o, src + p, len);
+            p += len;
+            o += len;
+        } else {  // copy
+            size_t off;
+            if ((tag & 3) == 1) {
+                if (p >= n) {
+                    err = "snappy: bad copy1";
+                    return false;
+                }
+                len = ((tag >> 2) & 7) + 4;
+                off = ((size_t)(tag >> 5) << 8) | src[p++];
+            } else if ((tag & 3) == 2) {
+                if (p + 2 > n) {
+                    err = "snappy: bad copy2";
+                    return false;
+                }
+                len = (tag >> 2) + 1;
+                off = (size_t)src[p] | ((size_t)src[p + 1] << 8);
+                p += 2;
+            } else {
+                if (p + 4 > n) {
+                    err = "snappy: bad copy4";
+                    return false;
+                }
+                len = (tag >> 2) + 1;
+                off = (size_t)src[p] | ((size_t)src[p + 1] << 8) |
+                      ((size_t)src[p + 2] << 16) |
+                      ((size_t)src[p + 3] << 24);
+                p += 4;
+            }
+            if (off == 0 || off > o || o + len > ulen) {
+                err = "snappy: bad copy offset/length";
+                return false;
+            }
+            // copies may overlap (off < len): byte-by-byte semantics
+            for (size_t i = 0; i < len; i++, o++) dst[o] = dst[o - off];
+        }
+    }
+    if (o != ulen) {
+        err = "snappy: decompressed size mismatch";
+        return false;
+    }
+    got = o;
+    return true;
+}
+
 bool gzip_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
                            size_t dst_n, std::string &err) {
     z_stream zs;
@@ -98,9 +189,9 @@ bool gzip_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
 bool orc_decompress(const uint8_t *src, int64_t len, int kind,
                     int64_t block_size, std::vector<uint8_t> &out,
                     std::string &err) {
-    if (kind != 1 && kind != 5) {
+    if (kind != 1 && kind != 2 && kind != 5) {
         err = "ORC compression kind " + std::to_string(kind) +
-              " not supported (v1: NONE, ZLIB, ZSTD)";
+              " not supported (v1: NONE, ZLIB, SNAPPY, ZSTD)";
         return false;
     }
     if (block_size <= 0) block_size = 256 * 1024;
@@ -126,11 +217,16 @@ bool orc_decompress(const uint8_t *src, int64_t len, int kind,
             size_t old = out.size();
             out.resize(old + block_size);
             size_t got = 0;
-            bool ok = kind == 1
-                          ? inflate_raw(src + p, clen, out.data() + old,
-                                        block_size, got, err)
-                          : zstd_call(src + p, clen, out.data() + old,
-                                      block_size, got, err);
+            bool ok;
+            if (kind == 1)
+                ok = inflate_raw(src + p, clen, out.data() + old, block_size,
+                                 got, err);
+            else if (kind == 2)
+                ok = snappy_decompress(src + p, clen, out.data() + old,
+                                       block_size, got, err);
+            else
+                ok = zstd_call(src + p, clen, out.data() + old, block_size,
+                               got, err);
             if (!ok) return false;
             out.resize(old + got);
         }

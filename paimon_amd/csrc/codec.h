@@ -19,6 +19,11 @@ bool zstd_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
 bool gzip_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
                            size_t dst_n, std::string &err);
 
+// raw snappy block (parquet SNAPPY pages, ORC SNAPPY chunks); from-scratch
+// decoder, no libsnappy dependency
+bool snappy_decompress(const uint8_t *src, size_t n, uint8_t *dst,
+                       size_t cap, size_t &got, std::string &err);
+
 // ORC chunked stream framing: 3-byte LE header (len << 1 | isOriginal) per
 // chunk, each decompressing to <= block_size bytes. kind: 1 = ZLIB (raw
 // deflate), 5 = ZSTD (ORC proto CompressionKind).

@@ -917,7 +917,8 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                 std::vector<uint8_t> packed;  // compressed codecs
                 std::vector<int64_t> ppo(cc.pages.size());
                 const uint8_t *payload_base = sf.data.data() + chunk_start;
-                if (cc.codec == CODEC_ZSTD || cc.codec == CODEC_GZIP) {
+                if (cc.codec == CODEC_ZSTD || cc.codec == CODEC_GZIP ||
+                    cc.codec == CODEC_SNAPPY) {
                     int64_t total_unc = 0;
                     for (auto &pg : cc.pages) total_unc += pg.uncompressed_size;
                     packed.resize(total_unc);
@@ -925,19 +926,28 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     for (size_t pi = 0; pi < cc.pages.size(); pi++) {
                         auto &pg = cc.pages[pi];
                         std::string cerr;
-                        bool ok = cc.codec == CODEC_ZSTD
-                                      ? zstd_decompress(
-                                            sf.data.data() + pg.data_off,
-                                            pg.compressed_size,
-                                            packed.data() + off,
-                                            pg.uncompressed_size)
-                                      : gzip_decompress_exact(
-                                            sf.data.data() + pg.data_off,
-                                            pg.compressed_size,
-                                            packed.data() + off,
-                                            pg.uncompressed_size, cerr);
+                        size_t got = 0;
+                        bool ok;
+                        if (cc.codec == CODEC_ZSTD)
+                            ok = zstd_decompress(
+                                sf.data.data() + pg.data_off,
+                                pg.compressed_size, packed.data() + off,
+                                pg.uncompressed_size);
+                        else if (cc.codec == CODEC_GZIP)
+                            ok = gzip_decompress_exact(
+                                sf.data.data() + pg.data_off,
+                                pg.compressed_size, packed.data() + off,
+                                pg.uncompressed_size, cerr);
+                        else
+                            ok = snappy_decompress(
+                                     sf.data.data() + pg.data_off,
+                                     pg.compressed_size, packed.data() + off,
+                                     pg.uncompressed_size, got, cerr) &&
+                                 got == (size_t)pg.uncompressed_size;
                         if (!ok) {
                             if (!cerr.empty()) set_error("%s", cerr.c_str());
+                            else if (cc.codec == CODEC_SNAPPY)
+                                set_error("snappy page size mismatch");
                             return false;
                         }
                         ppo[pi] = off;
@@ -1827,6 +1837,17 @@ int pmh_stats_get(pmh_plan_t *p, pmh_stats *out) {
 }
 
 void pmh_free_string(char *s) { free(s); }
+
+int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap) {
+    size_t got = 0;
+    std::string err;
+    if (!snappy_decompress((const uint8_t *)src, (size_t)n, (uint8_t *)dst,
+                           (size_t)cap, got, err)) {
+        set_error("%s", err.c_str());
+        return -1;
+    }
+    return (int64_t)got;
+}
 
 int pmh_write_parquet(const pmh_col *cols, int32_t n_cols, int64_t n_rows,
                       const char *path, int64_t row_group_rows,

@@ -387,9 +387,7 @@ class TestNarrowIntColumns:
                               delete_frac=0.1)
         for r in runs:
             r["values"][1] = (r["values"][1] % 127).astype(np.int8)
-            r["values"].append(
-                (r["values"][1].astype(np.int32) * 37 % 32_000)
-                .astype(np.int16))
+            r["values"][2] = (r["values"][2] % 32_000).astype(np.int16)
         metas = write_runs(runs, str(tmp_path), compression="NONE")
         r, w = merge_dedup(runs, drop_delete=True)
         vcols = [{"name": "v_k", "type": "int64"},
@@ -407,3 +405,7 @@ class TestNarrowIntColumns:
     def test_gzip_compressed(self, tmp_path):
         runs = gen_runs_dedup(4, 25_000, n_value_cols=3, seed=54)
         _run_and_compare(tmp_path, runs, compression="gzip")
+
+    def test_snappy_compressed(self, tmp_path):
+        runs = gen_runs_dedup(4, 25_000, n_value_cols=3, seed=55)
+        _run_and_compare(tmp_path, runs, compression="snappy")

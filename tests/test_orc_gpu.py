@@ -97,6 +97,11 @@ class TestOrcDedup:
                               delete_frac=0.15)
         self._run(tmp_path, runs, compression="zstd")
 
+    def test_orc_snappy(self, tmp_path):
+        runs = gen_runs_dedup(4, 20_000, n_value_cols=2, seed=90,
+                              delete_frac=0.1)
+        self._run(tmp_path, runs, compression="snappy")
+
     def test_orc_keep_delete(self, tmp_path):
         runs = gen_runs_dedup(4, 15_000, n_value_cols=2, seed=84,
                               delete_frac=0.3)
