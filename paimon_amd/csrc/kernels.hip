@@ -676,9 +676,9 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                 nr = x + 1;
                 // dense ranks: advance tile while i falls past its count
                 while (t + 1 < n_tiles && tile_offsets[t + 1] <= i) t++;
-                uint32_t packed = __builtin_nontemporal_load(
-                    &winners[t * (tile_rows + PMH_MAX_RUNS) +
-                             (i - tile_offsets[t])]);
+                uint32_t packed =
+                    winners[t * (tile_rows + PMH_MAX_RUNS) +
+                            (i - tile_offsets[t])];
                 run[x] = packed >> 28;
                 row[x] = packed & 0x0fffffff;
             } else {  // dead lane: duplicate x=0's gather, stores are guarded
@@ -706,8 +706,7 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
 #pragma unroll
                 for (int x = 0; x < R; x++)
                     if (x < nr)
-                        __builtin_nontemporal_store(
-                            (int8_t)v[x], &((int8_t *)out_ptrs[c])[idx[x]]);
+                        ((int8_t *)out_ptrs[c])[idx[x]] = (int8_t)v[x];
                 break;
             }
             case 2: {  // INT16 output from INT32-stored parquet SMALLINT
@@ -718,9 +717,7 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
 #pragma unroll
                 for (int x = 0; x < R; x++)
                     if (x < nr)
-                        __builtin_nontemporal_store(
-                            (int16_t)v[x],
-                            &((int16_t *)out_ptrs[c])[idx[x]]);
+                        ((int16_t *)out_ptrs[c])[idx[x]] = (int16_t)v[x];
                 break;
             }
             case 3:
@@ -731,9 +728,7 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                     v[x] = col_load<int32_t>(cols[run[x] * n_cols + c], row[x]);
 #pragma unroll
                 for (int x = 0; x < R; x++)
-                    if (x < nr)
-                        __builtin_nontemporal_store(
-                            v[x], &((int32_t *)out_ptrs[c])[idx[x]]);
+                    if (x < nr) ((int32_t *)out_ptrs[c])[idx[x]] = v[x];
                 break;
             }
             case 4:
@@ -744,9 +739,7 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                     v[x] = col_load<int64_t>(cols[run[x] * n_cols + c], row[x]);
 #pragma unroll
                 for (int x = 0; x < R; x++)
-                    if (x < nr)
-                        __builtin_nontemporal_store(
-                            v[x], &((int64_t *)out_ptrs[c])[idx[x]]);
+                    if (x < nr) ((int64_t *)out_ptrs[c])[idx[x]] = v[x];
                 break;
             }
             default: break;
