@@ -816,11 +816,10 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
         const uint16_t *gs = &group_start[t * (tile_rows + 1)];
         const uint32_t *mem = &members[t * (tile_rows + PMH_MAX_RUNS)];
         int32_t ms = gs[g], me = gs[g + 1];
-        // cache the newest members (+ packed masks) in registers: the
-        // per-column non-null walk starts at the newest record and nearly
-        // always ends within a few members (group size averages ~1.3), and
-        // re-reading the member list / validity from HBM per column
-        // dominated this kernel
+        // cache the newest members in registers: the per-column non-null
+        // walk starts at the newest record and nearly always ends within a
+        // few members (group size averages ~1.3), and re-reading the member
+        // list from HBM per column dominated this kernel
         uint32_t mc[4];
         uint64_t vm[4];
         const int gn = me - ms;
@@ -837,108 +836,84 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
         uint32_t last = mc[0];
         int lrun = last >> 28;
         int64_t lrow = last & 0x0fffffff;
-        // resolve + gather + store in BATCHES of 4 columns: resolution is
-        // register work (cached masks), so batching keeps 4 independent
-        // value gathers in flight per thread — the column-at-a-time loop
-        // had ONE load outstanding and sat 93% wait-parked (profiles/)
-        for (int cb = 0; cb < n_cols; cb += 4) {
-            int rrun[4];
-            int64_t rrow[4];
-            uint8_t rok[4];
-            int64_t v[4];
+        for (int c = 0; c < n_cols; c++) {
+            if (c == kind_col) {
+                ((int8_t *)out_ptrs[c])[i] = 0;  // RowKind.INSERT
+                continue;
+            }
+            int64_t run = lrun, row = lrow;
+            uint8_t ok = 1;
+            if (col_nullable[c] && gn > 1) {
+                ok = 0;
 #pragma unroll
-            for (int j = 0; j < 4; j++) {
-                const int c = cb + j;
-                rrun[j] = lrun;
-                rrow[j] = lrow;
-                rok[j] = 1;
-                if (c >= n_cols || c == kind_col) continue;
-                if (col_nullable[c] && gn > 1) {
-                    uint8_t ok = 0;
-#pragma unroll
-                    for (int x = 0; x < 4; x++) {
-                        if (ok || x >= gn) continue;
-                        uint8_t vb;
-                        if (MASKS) {
-                            vb = (uint8_t)((vm[x] >> c) & 1);
-                        } else {
-                            const DevCol &dc =
-                                cols[(mc[x] >> 28) * n_cols + c];
-                            vb = dc.valid0 ? ((const uint8_t *)
-                                                  dc.valid0)[mc[x] &
-                                                             0x0fffffff]
-                                           : 1;
-                        }
-                        if (vb) {
-                            rrun[j] = mc[x] >> 28;
-                            rrow[j] = mc[x] & 0x0fffffff;
-                            ok = 1;
-                        }
-                    }
-                    for (int32_t x = me - 5; !ok && x >= ms; x--) {
-                        uint32_t m = mem[x];
-                        uint8_t vb;
-                        if (MASKS) {
-                            vb = (uint8_t)(
-                                (run_masks[m >> 28][m & 0x0fffffff] >> c) &
-                                1);
-                        } else {
-                            const DevCol &dc = cols[(m >> 28) * n_cols + c];
-                            vb = dc.valid0
-                                     ? ((const uint8_t *)
-                                            dc.valid0)[m & 0x0fffffff]
-                                     : 1;
-                        }
-                        if (vb) {
-                            rrun[j] = m >> 28;
-                            rrow[j] = m & 0x0fffffff;
-                            ok = 1;
-                        }
-                    }
-                    rok[j] = ok;
-                } else if (col_nullable[c]) {
-                    // singleton: the record passes through with its own
-                    // validity
+                for (int x = 0; x < 4; x++) {
+                    if (ok || x >= gn) continue;
+                    uint32_t m = mc[x];
+                    uint8_t v;
                     if (MASKS) {
-                        rok[j] = (uint8_t)((vm[0] >> c) & 1);
+                        v = (uint8_t)((vm[x] >> c) & 1);
                     } else {
-                        const DevCol &dc = cols[lrun * n_cols + c];
-                        rok[j] =
-                            dc.valid0 ? ((const uint8_t *)dc.valid0)[lrow]
-                                      : 1;
+                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                        v = dc.valid0
+                                ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                                : 1;
+                    }
+                    if (v) {
+                        run = m >> 28;
+                        row = m & 0x0fffffff;
+                        ok = 1;
                     }
                 }
-            }
-#pragma unroll
-            for (int j = 0; j < 4; j++) {  // 4 independent gathers in flight
-                const int c = cb + j;
-                v[j] = 0;
-                if (c >= n_cols || c == kind_col || !rok[j]) continue;
-                const DevCol &dc = cols[rrun[j] * n_cols + c];
-                const int dt = col_dtype[c];
-                v[j] = (dt == 4 || dt == 6)
-                           ? col_load<int64_t>(dc, rrow[j])
-                           : (int64_t)col_load<int32_t>(dc, rrow[j]);
-            }
-#pragma unroll
-            for (int j = 0; j < 4; j++) {
-                const int c = cb + j;
-                if (c >= n_cols) continue;
-                if (c == kind_col) {
-                    ((int8_t *)out_ptrs[c])[i] = 0;  // RowKind.INSERT
-                    continue;
+                for (int32_t x = me - 5; !ok && x >= ms; x--) {
+                    uint32_t m = mem[x];
+                    uint8_t v;
+                    if (MASKS) {
+                        v = (uint8_t)(
+                            (run_masks[m >> 28][m & 0x0fffffff] >> c) & 1);
+                    } else {
+                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                        v = dc.valid0
+                                ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                                : 1;
+                    }
+                    if (v) {
+                        run = m >> 28;
+                        row = m & 0x0fffffff;
+                        ok = 1;
+                    }
                 }
-                switch (col_dtype[c]) {
-                case 1: ((int8_t *)out_ptrs[c])[i] = (int8_t)v[j]; break;
-                case 2: ((int16_t *)out_ptrs[c])[i] = (int16_t)v[j]; break;
-                case 3:
-                case 5: ((int32_t *)out_ptrs[c])[i] = (int32_t)v[j]; break;
-                case 4:
-                case 6: ((int64_t *)out_ptrs[c])[i] = v[j]; break;
-                default: break;
+            } else if (col_nullable[c]) {
+                // singleton: the record passes through with its own validity
+                if (MASKS) {
+                    ok = (uint8_t)((vm[0] >> c) & 1);
+                } else {
+                    const DevCol &dc = cols[lrun * n_cols + c];
+                    ok = dc.valid0 ? ((const uint8_t *)dc.valid0)[lrow] : 1;
                 }
-                if (out_valid[c]) out_valid[c][i] = rok[j];
             }
+            const DevCol &dc = cols[run * n_cols + c];
+            switch (col_dtype[c]) {
+            case 1:
+                ((int8_t *)out_ptrs[c])[i] =
+                    ok ? (int8_t)col_load<int32_t>(dc, row) : 0;
+                break;
+            case 2:
+                ((int16_t *)out_ptrs[c])[i] =
+                    ok ? (int16_t)col_load<int32_t>(dc, row) : 0;
+                break;
+            case 3:
+            case 5:
+                ((int32_t *)out_ptrs[c])[i] =
+                    ok ? col_load<int32_t>(dc, row) : 0;
+                break;
+            case 4:
+            case 6:
+                ((int64_t *)out_ptrs[c])[i] =
+                    ok ? col_load<int64_t>(dc, row) : 0;
+                break;
+            default: break;
+            }
+            if (out_valid[c]) out_valid[c][i] = ok;
         }
     }
 }
