@@ -1461,32 +1461,6 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         plan->drop_delete = j["drop_delete"].as_bool(true);
         plan->ignore_delete = j["ignore_delete"].as_bool(false);
         plan->host_output = j["output"].as_str("device") == "host";
-        if (plan->pu) {
-            // row-major blob layout (k_transpose): stored widths, 8-byte
-            // fields first for alignment, row padded to 16 bytes. Must be
-            // known BEFORE sections are built (they allocate the blobs).
-            const int nc = (int)plan->cols.size();
-            std::vector<uint32_t> off(nc, 0);
-            uint32_t w = 0;
-            for (int pass = 0; pass < 2; pass++)
-                for (int c = 0; c < nc; c++) {
-                    int es = plan->cols[c].stored_esize;
-                    if ((pass == 0) != (es == 8)) continue;
-                    off[c] = w;
-                    w += es;
-                }
-            w = (w + 15u) & ~15u;
-            if ((int)w <= PMH_MAX_BLOB_W) {
-                plan->blob_w = (int)w;
-                plan->col_off_dev = (uint32_t *)plan->bufs.alloc(nc * 4);
-                if (!plan->col_off_dev ||
-                    hipMemcpy(plan->col_off_dev, off.data(), nc * 4,
-                              hipMemcpyHostToDevice) != hipSuccess) {
-                    set_error("H2D of blob offsets failed");
-                    return nullptr;
-                }
-            }  // wider rows: fall back to columnar gathers
-        }
 
         std::vector<FileDesc> files;
         int idx = 0;
@@ -1559,6 +1533,30 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             dts[c] = (uint8_t)plan->cols[c].dtype;
         }
         plan->out_dev = outs;
+        if (plan->pu) {
+            // row-major blob layout (k_transpose): stored widths, 8-byte
+            // fields first for alignment, row padded to 16 bytes
+            std::vector<uint32_t> off(n_cols, 0);
+            uint32_t w = 0;
+            for (int pass = 0; pass < 2; pass++)
+                for (int c = 0; c < n_cols; c++) {
+                    int es = plan->cols[c].stored_esize;
+                    if ((pass == 0) != (es == 8)) continue;
+                    off[c] = w;
+                    w += es;
+                }
+            w = (w + 15u) & ~15u;
+            if ((int)w <= PMH_MAX_BLOB_W) {
+                plan->blob_w = (int)w;
+                plan->col_off_dev = (uint32_t *)plan->bufs.alloc(n_cols * 4);
+                if (!plan->col_off_dev ||
+                    hipMemcpy(plan->col_off_dev, off.data(), n_cols * 4,
+                              hipMemcpyHostToDevice) != hipSuccess) {
+                    set_error("H2D of blob offsets failed");
+                    return nullptr;
+                }
+            }  // wider rows: fall back to columnar gathers
+        }
         plan->out_ptrs_dev = (void **)plan->bufs.alloc(n_cols * sizeof(void *));
         plan->col_dtype_dev = (uint8_t *)plan->bufs.alloc(n_cols);
         if (hipMemcpy(plan->out_ptrs_dev, outs.data(),
