@@ -127,8 +127,6 @@ hipError_t pmh_launch_level_scatter(const RleChunk *chunks, int64_t n_chunks,
 // wrapper bypass). members/group_start written by k_merge_tiles in PU mode.
 // run_masks: device array of k per-run packed-validity pointers
 // (k_pack_valid), or null for the legacy per-column byte walk (>64 cols).
-// run_blobs: device array of k per-run row-major value blobs (k_transpose;
-// w_pad bytes per row, field offsets col_off), or null for columnar gathers.
 hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                               const uint8_t *col_nullable, int n_cols, int k,
                               int seq_col, int kind_col,
@@ -137,17 +135,8 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                               const int64_t *tile_offsets, int64_t n_tiles,
                               int64_t tile_rows, const int64_t *total_out,
                               uint64_t *const *run_masks,
-                              const uint8_t *const *run_blobs,
-                              const uint32_t *col_off, int w_pad,
                               void *const *out_ptrs,
                               uint8_t *const *out_valid, hipStream_t stream);
-
-// Decode-time row-major transpose of one run into a blob (w_pad-byte rows,
-// 16-byte aligned, w_pad <= PMH_MAX_BLOB_W).
-constexpr int PMH_MAX_BLOB_W = 240;  // 4 waves * 64 rows * w_pad <= 60 KB LDS
-hipError_t pmh_launch_transpose(const DevCol *cols, const uint32_t *col_off,
-                                int n_cols, int w_pad, int64_t rows,
-                                uint8_t *blob, hipStream_t stream);
 
 // Pack one run's per-column validity bytes into u64 row masks (bit c =
 // column c non-null; columns without staged nulls contribute 1).
@@ -177,8 +166,6 @@ hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                                const int64_t *tile_offsets, int64_t n_tiles,
                                int64_t tile_rows, const int64_t *total_out,
                                uint64_t *const *run_masks,
-                               const uint8_t *const *run_blobs,
-                               const uint32_t *col_off, int w_pad,
                                void *const *out_ptrs,
                                uint8_t *const *out_valid, hipStream_t stream);
 

@@ -748,50 +748,6 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
     }
 }
 
-// ------------------------------------------------------------ k_transpose
-//
-// Decode-time row-major transpose of one run: blob[row * w_pad + col_off[c]]
-// = column c's stored value. The PU/aggregation emits then read a resolved
-// member's ENTIRE row from 1-2 cache lines (all later column loads are
-// L1 hits) instead of n_cols scattered 64B-line gathers — the gather-latency
-// wall that scheduling changes could not move. One wave
-// stages 64 rows through LDS so the global writes are fully coalesced
-// 16-byte chunks; w_pad is 16-byte aligned and <= PMH_MAX_BLOB_W.
-__global__ void k_transpose(const DevCol *cols, const uint32_t *col_off,
-                            int n_cols, int w_pad, int64_t rows,
-                            uint8_t *blob) {
-    extern __shared__ uint8_t s_rows[];  // waves * 64 * w_pad
-    const int lane = (int)(threadIdx.x & 63);
-    const int wv = (int)(threadIdx.x >> 6);
-    uint8_t *ws = s_rows + (size_t)wv * 64 * w_pad;
-    const int64_t waves =
-        ((int64_t)gridDim.x * blockDim.x) >> 6;
-    const int64_t wid =
-        ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-    for (int64_t base = wid * 64; base < rows; base += waves * 64) {
-        const int nr = rows - base < 64 ? (int)(rows - base) : 64;
-        const int64_t r = base + (lane < nr ? lane : nr - 1);
-        for (int c = 0; c < n_cols; c++) {
-            const DevCol &dc = cols[c];
-            uint8_t *dst = ws + (size_t)lane * w_pad + col_off[c];
-            if (dc.esize == 8)
-                *(int64_t *)dst = col_load<int64_t>(dc, r);
-            else
-                *(int32_t *)dst = col_load<int32_t>(dc, r);
-        }
-        // wave-synchronous LDS -> global, 16B chunks (w_pad % 16 == 0, so a
-        // chunk never crosses a row boundary)
-        uint4 *out = (uint4 *)(blob + (uint64_t)base * w_pad);
-        const int qpr = w_pad >> 4;
-        const int total_q = nr * qpr;
-        for (int q = lane; q < total_q; q += 64) {
-            int row = q / qpr;
-            int within = q - row * qpr;
-            out[q] = *(uint4 *)(ws + (size_t)row * w_pad + within * 16);
-        }
-    }
-}
-
 // ------------------------------------------------------------ k_pack_valid
 //
 // Pack one run's per-column validity bytes into a u64 bitmask per row
@@ -828,20 +784,14 @@ __global__ void k_pack_valid(const DevCol *cols, int n_cols, int64_t rows,
 // MASKS: per-row packed validity (k_pack_valid) available — one u64 load per
 // member resolves all columns; false = legacy per-column byte walk (>64
 // columns).
-// BLOB: decode-time row-major blob available (k_transpose) — a member's
-// whole row sits in 1-2 cache lines, so per-column loads after the first
-// are L1 hits instead of independent 64B-line gathers.
-template <bool MASKS, bool BLOB>
+template <bool MASKS>
 __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                           const uint8_t *col_nullable, int n_cols, int k,
                           int seq_col, int kind_col, const uint32_t *members,
                           const uint16_t *group_start,
                           const int64_t *tile_offsets, int64_t n_tiles,
                           int64_t tile_rows, const int64_t *total_out,
-                          uint64_t *const *run_masks,
-                          const uint8_t *const *run_blobs,
-                          const uint32_t *col_off, int w_pad,
-                          void *const *out_ptrs,
+                          uint64_t *const *run_masks, void *const *out_ptrs,
                           uint8_t *const *out_valid) {
     const int64_t total = *total_out;
     const int64_t per_block =
@@ -941,53 +891,27 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                     ok = dc.valid0 ? ((const uint8_t *)dc.valid0)[lrow] : 1;
                 }
             }
-            if (BLOB) {
-                const uint8_t *br =
-                    run_blobs[run] + (uint64_t)row * w_pad + col_off[c];
-                switch (col_dtype[c]) {
-                case 1:
-                    ((int8_t *)out_ptrs[c])[i] =
-                        ok ? (int8_t)*(const int32_t *)br : 0;
-                    break;
-                case 2:
-                    ((int16_t *)out_ptrs[c])[i] =
-                        ok ? (int16_t)*(const int32_t *)br : 0;
-                    break;
-                case 3:
-                case 5:
-                    ((int32_t *)out_ptrs[c])[i] =
-                        ok ? *(const int32_t *)br : 0;
-                    break;
-                case 4:
-                case 6:
-                    ((int64_t *)out_ptrs[c])[i] =
-                        ok ? *(const int64_t *)br : 0;
-                    break;
-                default: break;
-                }
-            } else {
-                const DevCol &dc = cols[run * n_cols + c];
-                switch (col_dtype[c]) {
-                case 1:
-                    ((int8_t *)out_ptrs[c])[i] =
-                        ok ? (int8_t)col_load<int32_t>(dc, row) : 0;
-                    break;
-                case 2:
-                    ((int16_t *)out_ptrs[c])[i] =
-                        ok ? (int16_t)col_load<int32_t>(dc, row) : 0;
-                    break;
-                case 3:
-                case 5:
-                    ((int32_t *)out_ptrs[c])[i] =
-                        ok ? col_load<int32_t>(dc, row) : 0;
-                    break;
-                case 4:
-                case 6:
-                    ((int64_t *)out_ptrs[c])[i] =
-                        ok ? col_load<int64_t>(dc, row) : 0;
-                    break;
-                default: break;
-                }
+            const DevCol &dc = cols[run * n_cols + c];
+            switch (col_dtype[c]) {
+            case 1:
+                ((int8_t *)out_ptrs[c])[i] =
+                    ok ? (int8_t)col_load<int32_t>(dc, row) : 0;
+                break;
+            case 2:
+                ((int16_t *)out_ptrs[c])[i] =
+                    ok ? (int16_t)col_load<int32_t>(dc, row) : 0;
+                break;
+            case 3:
+            case 5:
+                ((int32_t *)out_ptrs[c])[i] =
+                    ok ? col_load<int32_t>(dc, row) : 0;
+                break;
+            case 4:
+            case 6:
+                ((int64_t *)out_ptrs[c])[i] =
+                    ok ? col_load<int64_t>(dc, row) : 0;
+                break;
+            default: break;
             }
             if (out_valid[c]) out_valid[c][i] = ok;
         }
@@ -1019,7 +943,7 @@ __device__ inline uint64_t f64_ord(int64_t b) {
     return b < 0 ? ~(uint64_t)b : ((uint64_t)b | 0x8000000000000000ull);
 }
 
-template <bool MASKS, bool BLOB>
+template <bool MASKS>
 __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                            const uint8_t *col_nullable,
                            const uint8_t *col_agg, int n_cols, int k,
@@ -1027,10 +951,7 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                            const uint16_t *group_start,
                            const int64_t *tile_offsets, int64_t n_tiles,
                            int64_t tile_rows, const int64_t *total_out,
-                           uint64_t *const *run_masks,
-                           const uint8_t *const *run_blobs,
-                           const uint32_t *col_off, int w_pad,
-                           void *const *out_ptrs,
+                           uint64_t *const *run_masks, void *const *out_ptrs,
                            uint8_t *const *out_valid) {
     auto valid_of = [&](uint32_t m, int c) -> uint8_t {
         if (MASKS)
@@ -1178,20 +1099,11 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 float facc = 0.f;
                 double dacc = 0.0;
                 auto fold_one = [&](uint32_t m) {
+                    const DevCol &dc = cols[(m >> 28) * n_cols + c];
                     const int64_t r = m & 0x0fffffff;
-                    int64_t vb;
-                    if (BLOB) {
-                        const uint8_t *br = run_blobs[m >> 28] +
-                                            (uint64_t)r * w_pad + col_off[c];
-                        vb = (dt == 4 || dt == 6)
-                                 ? *(const int64_t *)br
-                                 : (int64_t)*(const int32_t *)br;
-                    } else {
-                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
-                        vb = (dt == 4 || dt == 6)
-                                 ? col_load<int64_t>(dc, r)
-                                 : (int64_t)col_load<int32_t>(dc, r);
-                    }
+                    int64_t vb = (dt == 4 || dt == 6)
+                                     ? col_load<int64_t>(dc, r)
+                                     : (int64_t)col_load<int32_t>(dc, r);
                     if (!ok) {
                         ok = 1;
                         if (agg == PMH_AGG_SUM && dt == 5)
@@ -1245,19 +1157,12 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 break;
             }
             }
-            if (direct && ok) {
-                if (BLOB) {
-                    const uint8_t *br =
-                        run_blobs[run] + (uint64_t)row * w_pad + col_off[c];
-                    bits = (dt == 4 || dt == 6)
-                               ? *(const int64_t *)br
-                               : (int64_t)*(const int32_t *)br;
-                } else {
-                    const DevCol &dc = cols[run * n_cols + c];
+            if (direct) {
+                const DevCol &dc = cols[run * n_cols + c];
+                if (ok)
                     bits = (dt == 4 || dt == 6)
                                ? col_load<int64_t>(dc, row)
                                : (int64_t)col_load<int32_t>(dc, row);
-                }
             }
             if (!ok) bits = 0;
             switch (dt) {
@@ -1591,33 +1496,20 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                               const int64_t *tile_offsets, int64_t n_tiles,
                               int64_t tile_rows, const int64_t *total_out,
                               uint64_t *const *run_masks,
-                              const uint8_t *const *run_blobs,
-                              const uint32_t *col_off, int w_pad,
                               void *const *out_ptrs,
                               uint8_t *const *out_valid, hipStream_t stream) {
-    auto launch = [&](auto kern) {
-        hipLaunchKernelGGL(kern, dim3(2048), dim3(256), 0, stream, cols,
-                           col_dtype, col_nullable, n_cols, k, seq_col,
+    if (run_masks)
+        hipLaunchKernelGGL(k_emit_pu<true>, dim3(2048), dim3(256), 0, stream,
+                           cols, col_dtype, col_nullable, n_cols, k, seq_col,
                            kind_col, members, group_start, tile_offsets,
-                           n_tiles, tile_rows, total_out, run_masks,
-                           run_blobs, col_off, w_pad, out_ptrs, out_valid);
-    };
-    if (run_masks && run_blobs) launch(k_emit_pu<true, true>);
-    else if (run_masks) launch(k_emit_pu<true, false>);
-    else if (run_blobs) launch(k_emit_pu<false, true>);
-    else launch(k_emit_pu<false, false>);
-    return hipGetLastError();
-}
-
-hipError_t pmh_launch_transpose(const DevCol *cols, const uint32_t *col_off,
-                                int n_cols, int w_pad, int64_t rows,
-                                uint8_t *blob, hipStream_t stream) {
-    const int threads = 256;  // 4 waves -> 4 * 64 * w_pad LDS bytes
-    const size_t lds = (size_t)4 * 64 * w_pad;
-    int64_t blocks64 = (rows + 4 * 64 - 1) / (4 * 64);
-    int blocks = blocks64 < 8192 ? (int)(blocks64 ? blocks64 : 1) : 8192;
-    hipLaunchKernelGGL(k_transpose, dim3(blocks), dim3(threads), lds, stream,
-                       cols, col_off, n_cols, w_pad, rows, blob);
+                           n_tiles, tile_rows, total_out, run_masks, out_ptrs,
+                           out_valid);
+    else
+        hipLaunchKernelGGL(k_emit_pu<false>, dim3(2048), dim3(256), 0, stream,
+                           cols, col_dtype, col_nullable, n_cols, k, seq_col,
+                           kind_col, members, group_start, tile_offsets,
+                           n_tiles, tile_rows, total_out, run_masks, out_ptrs,
+                           out_valid);
     return hipGetLastError();
 }
 
@@ -1637,23 +1529,21 @@ hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                                const int64_t *tile_offsets, int64_t n_tiles,
                                int64_t tile_rows, const int64_t *total_out,
                                uint64_t *const *run_masks,
-                               const uint8_t *const *run_blobs,
-                               const uint32_t *col_off, int w_pad,
                                void *const *out_ptrs,
                                uint8_t *const *out_valid,
                                hipStream_t stream) {
-    auto launch = [&](auto kern) {
-        hipLaunchKernelGGL(kern, dim3(2048), dim3(256), 0, stream, cols,
-                           col_dtype, col_nullable, col_agg, n_cols, k,
+    if (run_masks)
+        hipLaunchKernelGGL(k_emit_agg<true>, dim3(2048), dim3(256), 0, stream,
+                           cols, col_dtype, col_nullable, col_agg, n_cols, k,
                            seq_col, kind_col, members, group_start,
                            tile_offsets, n_tiles, tile_rows, total_out,
-                           run_masks, run_blobs, col_off, w_pad, out_ptrs,
-                           out_valid);
-    };
-    if (run_masks && run_blobs) launch(k_emit_agg<true, true>);
-    else if (run_masks) launch(k_emit_agg<true, false>);
-    else if (run_blobs) launch(k_emit_agg<false, true>);
-    else launch(k_emit_agg<false, false>);
+                           run_masks, out_ptrs, out_valid);
+    else
+        hipLaunchKernelGGL(k_emit_agg<false>, dim3(2048), dim3(256), 0,
+                           stream, cols, col_dtype, col_nullable, col_agg,
+                           n_cols, k, seq_col, kind_col, members, group_start,
+                           tile_offsets, n_tiles, tile_rows, total_out,
+                           run_masks, out_ptrs, out_valid);
     return hipGetLastError();
 }
 

@@ -224,9 +224,6 @@ struct Section {
     // bit c = column c non-null; built once per section by k_pack_valid
     std::vector<uint64_t *> row_masks;
     uint64_t **row_masks_dev = nullptr;  // [k] device array of the above
-    // per-run row-major value blobs (k_transpose; PU/agg emits)
-    std::vector<uint8_t *> blobs;
-    uint8_t **blobs_dev = nullptr;  // [k] device array of the above
     // batched decode work (all run-columns in ONE launch each)
     Rlev2Chunk *rlev2_all = nullptr;
     int64_t n_rlev2 = 0;
@@ -255,9 +252,6 @@ struct pmh_plan_t {
     bool first_row = false;  // first-row merge engine
     bool agg = false;        // aggregation merge engine (uses PU member lists)
     uint8_t *col_agg_dev = nullptr;  // per-column PMH_AGG_* codes
-    // row-major blob layout for the PU/agg emits (k_transpose)
-    uint32_t *col_off_dev = nullptr;  // per-column byte offset within a row
-    int blob_w = 0;                   // padded row width; 0 = blobs disabled
     std::vector<pmh::Section> sections;
     size_t cur_section = 0;
     int64_t rows_in_total = 0;
@@ -1318,18 +1312,6 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
                 (uint64_t **)up(sec.row_masks.data(), k * sizeof(void *));
             if (!sec.row_masks_dev) return false;
         }
-        if (plan->blob_w > 0) {
-            sec.blobs.assign(k, nullptr);
-            for (int r = 0; r < k; r++) {
-                int64_t n = sec.runs[r].length > 0 ? sec.runs[r].length : 1;
-                sec.blobs[r] =
-                    (uint8_t *)plan->bufs.alloc(n * (int64_t)plan->blob_w);
-                if (!sec.blobs[r]) return false;
-            }
-            sec.blobs_dev =
-                (uint8_t **)up(sec.blobs.data(), k * sizeof(void *));
-            if (!sec.blobs_dev) return false;
-        }
     }
     std::vector<Rlev2Chunk> all_v;
     std::vector<RleChunk> all_d;
@@ -1533,30 +1515,6 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             dts[c] = (uint8_t)plan->cols[c].dtype;
         }
         plan->out_dev = outs;
-        if (plan->pu) {
-            // row-major blob layout (k_transpose): stored widths, 8-byte
-            // fields first for alignment, row padded to 16 bytes
-            std::vector<uint32_t> off(n_cols, 0);
-            uint32_t w = 0;
-            for (int pass = 0; pass < 2; pass++)
-                for (int c = 0; c < n_cols; c++) {
-                    int es = plan->cols[c].stored_esize;
-                    if ((pass == 0) != (es == 8)) continue;
-                    off[c] = w;
-                    w += es;
-                }
-            w = (w + 15u) & ~15u;
-            if ((int)w <= PMH_MAX_BLOB_W) {
-                plan->blob_w = (int)w;
-                plan->col_off_dev = (uint32_t *)plan->bufs.alloc(n_cols * 4);
-                if (!plan->col_off_dev ||
-                    hipMemcpy(plan->col_off_dev, off.data(), n_cols * 4,
-                              hipMemcpyHostToDevice) != hipSuccess) {
-                    set_error("H2D of blob offsets failed");
-                    return nullptr;
-                }
-            }  // wider rows: fall back to columnar gathers
-        }
         plan->out_ptrs_dev = (void **)plan->bufs.alloc(n_cols * sizeof(void *));
         plan->col_dtype_dev = (uint8_t *)plan->bufs.alloc(n_cols);
         if (hipMemcpy(plan->out_ptrs_dev, outs.data(),
@@ -1722,17 +1680,6 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             if (pe != hipSuccess) return fail("pack_valid", pe);
         }
     }
-    if (sec.blobs_dev) {
-        // row-major value blobs for the PU/agg emits, rebuilt per pass
-        // (decode-derived, like the masks)
-        for (int r = 0; r < k; r++) {
-            if (sec.runs[r].length <= 0) continue;
-            hipError_t te = pmh_launch_transpose(
-                sec.all_cols + (size_t)r * n_cols, p->col_off_dev, n_cols,
-                p->blob_w, sec.runs[r].length, sec.blobs[r], st);
-            if (te != hipSuccess) return fail("transpose", te);
-        }
-    }
     (void)hipEventRecord(ev[1], st);
     hipError_t e = pmh_launch_partition(sec.key_cols, sec.lens_dev, k,
                                         PMH_TILE_ROWS, sec.n_tiles + 1,
@@ -1759,17 +1706,14 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev,
             p->col_agg_dev, n_cols, k, p->n_key_cols, p->n_key_cols + 1,
             sec.winners, sec.group_start, sec.tile_offsets, sec.n_tiles,
-            PMH_TILE_ROWS, sec.total_dev, sec.row_masks_dev,
-            (const uint8_t *const *)sec.blobs_dev, p->col_off_dev, p->blob_w,
-            p->out_ptrs_dev, p->out_valid_dev, st);
+            PMH_TILE_ROWS, sec.total_dev, sec.row_masks_dev, p->out_ptrs_dev,
+            p->out_valid_dev, st);
     } else if (p->pu) {
         e = pmh_launch_emit_pu(
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,
             p->n_key_cols, p->n_key_cols + 1, sec.winners, sec.group_start,
             sec.tile_offsets, sec.n_tiles, PMH_TILE_ROWS, sec.total_dev,
-            sec.row_masks_dev, (const uint8_t *const *)sec.blobs_dev,
-            p->col_off_dev, p->blob_w, p->out_ptrs_dev, p->out_valid_dev,
-            st);
+            sec.row_masks_dev, p->out_ptrs_dev, p->out_valid_dev, st);
     } else {
         e = pmh_launch_emit(sec.all_cols, p->col_dtype_dev,
                             p->col_nullable_dev, n_cols, k, sec.winners,
