@@ -409,3 +409,50 @@ class TestNarrowIntColumns:
     def test_snappy_compressed(self, tmp_path):
         runs = gen_runs_dedup(4, 25_000, n_value_cols=3, seed=55)
         _run_and_compare(tmp_path, runs, compression="snappy")
+
+
+class TestErrorPaths:
+    # everything outside the matrix fails loudly at plan create with a
+    # descriptive error so a Java-side provider can fall
+    # back to the stock reader per split
+    def test_too_many_overlapping_runs(self, tmp_path):
+        runs = gen_runs_dedup(17, 500, n_value_cols=1, seed=70,
+                              delete_frac=0.0)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="16"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(1))
+
+    def test_missing_column(self, tmp_path):
+        runs = gen_runs_dedup(2, 500, n_value_cols=1, seed=71)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="not found"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              [{"name": "nope", "type": "int32"}])
+
+    def test_unsupported_parquet_codec(self, tmp_path):
+        runs = gen_runs_dedup(2, 500, n_value_cols=1, seed=72)
+        metas = write_runs(runs, str(tmp_path), compression="lz4")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="codec"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(1))
+
+    def test_unsupported_orc_codec(self, tmp_path):
+        runs = gen_runs_dedup(2, 500, n_value_cols=1, seed=73)
+        metas = write_runs(runs, str(tmp_path), compression="lz4",
+                           file_format="orc")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="compression kind"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(1))
+
+    def test_unknown_merge_engine(self, tmp_path):
+        runs = gen_runs_dedup(1, 100, n_value_cols=1, seed=74)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="merge engine"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(1), merge_engine="lookup")
