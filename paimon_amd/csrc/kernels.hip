@@ -44,6 +44,14 @@ DEV T col_load(const DevCol &c, int64_t row) {
 
 DEV uint64_t ukey(int64_t k) { return (uint64_t)k ^ 0x8000000000000000ull; }
 
+// key column element (stored width 4 or 8 bytes — TINYINT..INT stage as
+// INT32; the key TYPE is uniform across runs, so es is a scalar)
+DEV int64_t key_at(uint64_t addr, int64_t idx, int es) {
+    return es == 8 ? *reinterpret_cast<const int64_t *>(addr + idx * 8)
+                   : (int64_t)*reinterpret_cast<const int32_t *>(addr +
+                                                                 idx * 4);
+}
+
 // Lockstep multi-run lower/upper bound: runs the k binary searches together
 // so their probe loads issue back-to-back each step (8-16x memory-level
 // parallelism vs sequential searches — the partition kernel was 83%
@@ -52,7 +60,8 @@ DEV uint64_t ukey(int64_t k) { return (uint64_t)k ^ 0x8000000000000000ull; }
 // le=true: first index with ukey > v; le=false: first index with ukey >= v.
 template <bool LE, int KM>
 DEV void bound_multi(const uint64_t *addr, const int64_t *lo_in,
-                     const int64_t *hi_in, int k, uint64_t v, int64_t *out) {
+                     const int64_t *hi_in, int k, int kes, uint64_t v,
+                     int64_t *out) {
     int64_t lo[KM], hi[KM];
 #pragma unroll
     for (int r = 0; r < KM; r++) {
@@ -68,8 +77,7 @@ DEV void bound_multi(const uint64_t *addr, const int64_t *lo_in,
             if (r >= k) continue;
             if (lo[r] < hi[r]) {
                 int64_t mid = lo[r] + ((hi[r] - lo[r]) >> 1);
-                uint64_t kk = ukey(
-                    *reinterpret_cast<const int64_t *>(addr[r] + mid * 8));
+                uint64_t kk = ukey(key_at(addr[r], mid, kes));
                 bool go = LE ? (kk <= v) : (kk < v);
                 if (go) lo[r] = mid + 1;
                 else hi[r] = mid;
@@ -102,6 +110,7 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
 
     uint64_t addr[KM];
     int64_t len[KM], wlo[KM], whi[KM];
+    const int kes = keys[0].esize;
 #pragma unroll
     for (int r = 0; r < KM; r++) {
         if (r >= k) continue;
@@ -125,16 +134,15 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
 #pragma unroll
     for (int r = 0; r < KM; r++) {
         if (r >= k || len[r] == 0) continue;
-        uint64_t lo_k = ukey(*reinterpret_cast<const int64_t *>(addr[r]));
-        uint64_t hi_k = ukey(
-            *reinterpret_cast<const int64_t *>(addr[r] + (len[r] - 1) * 8));
+        uint64_t lo_k = ukey(key_at(addr[r], 0, kes));
+        uint64_t hi_k = ukey(key_at(addr[r], len[r] - 1, kes));
         if (lo_k < klo) klo = lo_k;
         if (hi_k > khi) khi = hi_k;
     }
     int64_t pos[KM];
     while (klo < khi) {
         uint64_t mid = klo + ((khi - klo) >> 1);
-        bound_multi<true, KM>(addr, wlo, whi, k, mid, pos);
+        bound_multi<true, KM>(addr, wlo, whi, k, kes, mid, pos);
         int64_t cnt = 0;
 #pragma unroll
         for (int r = 0; r < KM; r++) {
@@ -161,7 +169,7 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
     // (<=1 per run), take the first t in run order. Searching within the
     // final windows yields absolute positions: everything below wlo has
     // key < v*, everything at/above whi has key > v*.
-    bound_multi<false, KM>(addr, wlo, whi, k, klo, pos);
+    bound_multi<false, KM>(addr, wlo, whi, k, kes, klo, pos);
     int64_t base = 0;
 #pragma unroll
     for (int r = 0; r < KM; r++) {
@@ -173,9 +181,7 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
     for (int r = 0; r < KM; r++) {
         if (r >= k) continue;
         int64_t c = pos[r];
-        bool has = (c < len[r]) &&
-                   (ukey(*reinterpret_cast<const int64_t *>(addr[r] + c * 8)) ==
-                    klo);
+        bool has = (c < len[r]) && (ukey(key_at(addr[r], c, kes)) == klo);
         if (t > 0 && has) { c++; t--; }
         cuts[b * k + r] = (int32_t)c;
     }
@@ -271,7 +277,8 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             int32_t ext = (b < (int32_t)lens[tid]) ? 1 : 0;
             sm.seglen[tid] = (b - a) + ext;
             s_predcand[tid] =
-                a > 0 ? col_load<int64_t>(keys[tid], a - 1) : INT64_MIN;
+                a > 0 ? key_at(keys[tid].addr0, a - 1, keys[tid].esize)
+                      : INT64_MIN;
             // reuse head[] as a tiny flag channel for "has predecessor"
             sm.head[tid] = a > 0;
         }
@@ -315,14 +322,14 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         for (int r = 0; r < k; r++) {
             int32_t off = sm.segoff[r], len = sm.seglen[r];
             int64_t base = c0[r];
-            const int64_t *kaddr =
-                reinterpret_cast<const int64_t *>(keys[r].addr0) + base;
+            const int kes = keys[r].esize;
+            const uint64_t kaddr0 = keys[r].addr0 + (uint64_t)base * kes;
             const int64_t *saddr =
                 reinterpret_cast<const int64_t *>(seqs[r].addr0) + base;
             const int32_t *daddr =
                 reinterpret_cast<const int32_t *>(kinds[r].addr0) + base;
             for (int32_t i = tid; i < len; i += blockDim.x) {
-                sm.skey[off + i] = kaddr[i];
+                sm.skey[off + i] = key_at(kaddr0, i, kes);
                 int32_t kd = daddr[i];
                 sm.sseq[off + i] =
                     (saddr[i] << 1) | (int64_t)(kd == 0 || kd == 2);

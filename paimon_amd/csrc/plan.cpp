@@ -1395,9 +1395,12 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             return true;
         };
         const Json &kc = j["key_cols"];
-        if (kc.arr.size() != 1 ||
-            dtype_from_str(kc.arr[0]["type"].as_str()) != PMH_DT_INT64) {
-            set_error("v1 supports exactly one int64 key column (got %zu)",
+        int kdt = kc.arr.size() == 1
+                      ? dtype_from_str(kc.arr[0]["type"].as_str())
+                      : -1;
+        if (kdt < PMH_DT_INT8 || kdt > PMH_DT_INT64) {
+            set_error("v1 supports exactly one integer key column "
+                      "(TINYINT..BIGINT; got %zu column(s))",
                       kc.arr.size());
             return nullptr;
         }

@@ -121,7 +121,8 @@ def run_to_arrow(run, value_names=None):
     n_vals = len(run["values"])
     if value_names is None:
         value_names = ["v_k"] + [f"v_c{i}" for i in range(n_vals - 1)]
-    fields = [pa.field("_KEY_k", pa.int64(), nullable=False),
+    fields = [pa.field("_KEY_k", pa.from_numpy_dtype(run["key"].dtype),
+                       nullable=False),
               pa.field("_SEQUENCE_NUMBER", pa.int64(), nullable=False),
               pa.field("_VALUE_KIND", pa.int8(), nullable=False)]
     cols = [pa.array(run["key"]), pa.array(run["seq"]), pa.array(run["kind"])]

@@ -456,3 +456,51 @@ class TestErrorPaths:
             with pytest.raises(RuntimeError, match="merge engine"):
                 MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
                               _value_cols(1), merge_engine="lookup")
+
+    def test_int32_key_column(self, tmp_path):
+        # integer key widths: TINYINT..INT keys stage as INT32; the
+        # partition/merge key loads are width-aware
+        runs = gen_runs_dedup(4, 30_000, n_value_cols=2, seed=75,
+                              delete_frac=0.1)
+        for r in runs:
+            r["key"] = r["key"].astype(np.int32)
+            r["values"][0] = r["values"][0].astype(np.int32)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        oruns = [{**r, "key": r["key"].astype(np.int64)} for r in runs]
+        r, w = merge_dedup(oruns, drop_delete=True)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas),
+                               [{"name": "_KEY_k", "type": "int32"}],
+                               [{"name": "v_k", "type": "int32"},
+                                {"name": "v_c0", "type": "int32"},
+                                {"name": "v_c1", "type": "int32"}]) as plan:
+                got = _read_all_batches(plan)
+        ek = np.array([runs[a]["key"][b] for a, b in zip(r, w)], np.int32)
+        es = np.array([runs[a]["seq"][b] for a, b in zip(r, w)], np.int64)
+        assert got["_KEY_k"].dtype == np.int32
+        assert (got["_KEY_k"] == ek).all()
+        assert (got["_SEQUENCE_NUMBER"] == es).all()
+
+    def test_int16_key_column_orc(self, tmp_path):
+        rng = np.random.default_rng(76)
+        runs = []
+        seqs = rng.permutation(40_000).astype(np.int64)
+        for i in range(4):
+            keys = np.sort(rng.choice(20_000, 10_000,
+                                      replace=False)).astype(np.int16)
+            runs.append({
+                "key": keys, "seq": seqs[i * 10_000:(i + 1) * 10_000],
+                "kind": np.zeros(10_000, np.int8),
+                "values": [keys.astype(np.int32)]})
+        metas = write_runs(runs, str(tmp_path), compression="NONE",
+                           file_format="orc")
+        oruns = [{**r, "key": r["key"].astype(np.int64)} for r in runs]
+        r, w = merge_dedup(oruns, drop_delete=True)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas),
+                               [{"name": "_KEY_k", "type": "int16"}],
+                               [{"name": "v_k", "type": "int32"}]) as plan:
+                got = _read_all_batches(plan)
+        ek = np.array([runs[a]["key"][b] for a, b in zip(r, w)], np.int16)
+        assert got["_KEY_k"].dtype == np.int16
+        assert (got["_KEY_k"] == ek).all()
