@@ -143,6 +143,12 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
 hipError_t pmh_launch_pack_valid(const DevCol *cols, int n_cols, int64_t rows,
                                  uint64_t *mask, hipStream_t stream);
 
+// Build one run's order-preserving composite key from <= 8 integer key
+// columns whose widths sum to <= 64 bits (spec packed 8 bits per sub-key).
+hipError_t pmh_launch_composite(const DevCol *keys, int nk, uint64_t shifts,
+                                uint64_t bits, int64_t rows, int64_t *ckey,
+                                hipStream_t stream);
+
 // Aggregation emit: per owned group, fold members in ascending (seq, isAdd)
 // order through per-column FieldAggregators (AggregateMergeFunction.java:
 // 82-125; default last_non_null_value, :201). col_agg holds one PMH_AGG_*

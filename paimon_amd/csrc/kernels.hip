@@ -755,6 +755,36 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
     }
 }
 
+// ------------------------------------------------------------ k_composite
+//
+// Build one run's order-preserving composite key: each integer sub-key is
+// biased to unsigned at its declared bit width and concatenated MSB-first
+// (first key column highest), then the u64 is sign-flipped back to int64 so
+// ukey() in the partition/merge recovers the unsigned order. Paimon compares
+// multi-column keys lexicographically (keyComparator over the _KEY_ fields),
+// which this encoding preserves exactly for integer columns whose widths sum
+// to <= 64 bits. Rebuilt per pass (decode-derived).
+// spec packs (shift, bits) per sub-key, 8 bits each, sub-key i at byte i.
+__global__ void k_composite(const DevCol *keys, int nk, uint64_t shifts,
+                            uint64_t bits, int64_t rows, int64_t *ckey) {
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < rows;
+         i += stride) {
+        uint64_t u = 0;
+        for (int j = 0; j < nk; j++) {
+            const DevCol &dc = keys[j];
+            int64_t v = dc.esize == 8 ? col_load<int64_t>(dc, i)
+                                      : (int64_t)col_load<int32_t>(dc, i);
+            const int w = (int)((bits >> (8 * j)) & 0xff);
+            const int sh = (int)((shifts >> (8 * j)) & 0xff);
+            uint64_t b = ((uint64_t)v + (1ull << (w - 1))) &
+                         (w >= 64 ? ~0ull : ((1ull << w) - 1));
+            u |= b << sh;
+        }
+        ckey[i] = (int64_t)(u ^ 0x8000000000000000ull);
+    }
+}
+
 // ------------------------------------------------------------ k_pack_valid
 //
 // Pack one run's per-column validity bytes into a u64 bitmask per row
@@ -1517,6 +1547,14 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                            kind_col, members, group_start, tile_offsets,
                            n_tiles, tile_rows, total_out, run_masks, out_ptrs,
                            out_valid);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_composite(const DevCol *keys, int nk, uint64_t shifts,
+                                uint64_t bits, int64_t rows, int64_t *ckey,
+                                hipStream_t stream) {
+    hipLaunchKernelGGL(k_composite, dim3(1024), dim3(256), 0, stream, keys,
+                       nk, shifts, bits, rows, ckey);
     return hipGetLastError();
 }
 

@@ -224,6 +224,7 @@ struct Section {
     // bit c = column c non-null; built once per section by k_pack_valid
     std::vector<uint64_t *> row_masks;
     uint64_t **row_masks_dev = nullptr;  // [k] device array of the above
+    std::vector<int64_t *> ckeys;  // per-run composite keys (k_composite)
     // batched decode work (all run-columns in ONE launch each)
     Rlev2Chunk *rlev2_all = nullptr;
     int64_t n_rlev2 = 0;
@@ -252,6 +253,9 @@ struct pmh_plan_t {
     bool first_row = false;  // first-row merge engine
     bool agg = false;        // aggregation merge engine (uses PU member lists)
     uint8_t *col_agg_dev = nullptr;  // per-column PMH_AGG_* codes
+    // composite key (>1 key column): order-preserving packed comparand
+    bool composite_key = false;
+    uint64_t key_shifts = 0, key_bits = 0;  // 8 bits per sub-key
     std::vector<pmh::Section> sections;
     size_t cur_section = 0;
     int64_t rows_in_total = 0;
@@ -1258,6 +1262,7 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
     std::vector<int64_t> lens(k);
     int seq_idx = plan->n_key_cols;
     int kind_idx = plan->n_key_cols + 1;
+    if (plan->composite_key) sec.ckeys.assign(k, nullptr);
     for (int r = 0; r < k; r++) {
         Run &run = sec.runs[r];
         lens[r] = run.length;
@@ -1268,9 +1273,17 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
                       run.cols[c].pages_dev, run.cols[c].n_pages,
                       plan->cols[c].stored_esize};
             allv[r * n_cols + c] = dc;
-            if (c == 0) keyv[r] = dc;  // v1: single int64 key column
+            if (c == 0) keyv[r] = dc;  // single-column key: the column itself
             if (c == seq_idx) seqv[r] = dc;
             if (c == kind_idx) kindv[r] = dc;
+        }
+        if (plan->composite_key) {
+            // the partition/merge comparand is the packed composite key,
+            // built per pass by k_composite
+            int64_t n = run.length > 0 ? run.length : 1;
+            sec.ckeys[r] = (int64_t *)plan->bufs.alloc(n * 8);
+            if (!sec.ckeys[r]) return false;
+            keyv[r] = DevCol{(uint64_t)sec.ckeys[r], 0, nullptr, 1, 8};
         }
     }
     sec.n_tiles = (sec.total_rows + PMH_TILE_ROWS - 1) / PMH_TILE_ROWS;
@@ -1395,16 +1408,38 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             return true;
         };
         const Json &kc = j["key_cols"];
-        int kdt = kc.arr.size() == 1
-                      ? dtype_from_str(kc.arr[0]["type"].as_str())
-                      : -1;
-        if (kdt < PMH_DT_INT8 || kdt > PMH_DT_INT64) {
-            set_error("v1 supports exactly one integer key column "
-                      "(TINYINT..BIGINT; got %zu column(s))",
+        if (kc.arr.empty() || kc.arr.size() > 8) {
+            set_error("v1 supports 1..8 integer key columns (got %zu)",
                       kc.arr.size());
             return nullptr;
         }
-        plan->n_key_cols = 1;
+        int key_total_bits = 0;
+        {
+            int shift = 64;
+            for (size_t i = 0; i < kc.arr.size(); i++) {
+                int dt = dtype_from_str(kc.arr[i]["type"].as_str());
+                if (dt < PMH_DT_INT8 || dt > PMH_DT_INT64) {
+                    set_error("key column %zu: integer types only "
+                              "(TINYINT..BIGINT) in v1",
+                              i);
+                    return nullptr;
+                }
+                int bits = 8 << (dt - PMH_DT_INT8);
+                key_total_bits += bits;
+                shift -= bits;
+                if (key_total_bits > 64) {
+                    set_error("composite key widths sum to %d bits > 64 "
+                              "(the order-preserving packed comparand is one "
+                              "int64; wider keys are a later round)",
+                              key_total_bits);
+                    return nullptr;
+                }
+                plan->key_shifts |= (uint64_t)(uint8_t)shift << (8 * i);
+                plan->key_bits |= (uint64_t)(uint8_t)bits << (8 * i);
+            }
+        }
+        plan->n_key_cols = (int)kc.arr.size();
+        plan->composite_key = plan->n_key_cols > 1;
         for (const auto &c : kc.arr)
             if (!add_col(c)) return nullptr;
         {  // _SEQUENCE_NUMBER, _VALUE_KIND (SpecialFields.java:79-83)
@@ -1670,6 +1705,17 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             hipError_t e =
                 pmh_launch_level_scatter(sec.def_all, sec.n_def, st);
             if (e != hipSuccess) return fail("level_scatter", e);
+        }
+    }
+    if (p->composite_key) {
+        // rebuilt per pass (decode-derived, like the validity masks)
+        for (int r = 0; r < k; r++) {
+            if (sec.runs[r].length <= 0) continue;
+            hipError_t ce = pmh_launch_composite(
+                sec.all_cols + (size_t)r * n_cols, p->n_key_cols,
+                p->key_shifts, p->key_bits, sec.runs[r].length, sec.ckeys[r],
+                st);
+            if (ce != hipSuccess) return fail("composite", ce);
         }
     }
     if (sec.row_masks_dev) {

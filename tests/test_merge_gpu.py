@@ -504,3 +504,74 @@ class TestErrorPaths:
         ek = np.array([runs[a]["key"][b] for a, b in zip(r, w)], np.int16)
         assert got["_KEY_k"].dtype == np.int16
         assert (got["_KEY_k"] == ek).all()
+
+
+class TestCompositeKeys:
+    # multi-column integer PKs: the partition/merge comparand is an
+    # order-preserving pack of the biased sub-keys (<= 64 bits total);
+    # minKey/maxKey in the plan carry the same encoding
+    @staticmethod
+    def _comp(k1, k2):
+        u = ((int(k1) + 2**31) << 32) | (int(k2) + 2**31)
+        u ^= 2**63
+        return u - 2**64 if u >= 2**63 else u
+
+    def test_two_int32_key_columns(self, tmp_path):
+        import os
+        import pyarrow as pa
+        import pyarrow.parquet as pq
+        rng = np.random.default_rng(78)
+        runs = []
+        metas = []
+        seqs = rng.permutation(60_000).astype(np.int64)
+        for i in range(3):
+            key = np.sort(rng.choice(90_000, 20_000,
+                                     replace=False)).astype(np.int64)
+            k1 = (key >> 8).astype(np.int32) - 1000   # negatives included
+            k2 = (key & 0xFF).astype(np.int32)
+            vals = rng.integers(-2**31, 2**31, 20_000).astype(np.int32)
+            runs.append({"key": key, "seq": seqs[i*20_000:(i+1)*20_000],
+                         "kind": np.zeros(20_000, np.int8),
+                         "values": [vals], "k1": k1, "k2": k2})
+            fields = [pa.field("_KEY_k1", pa.int32(), nullable=False),
+                      pa.field("_KEY_k2", pa.int32(), nullable=False),
+                      pa.field("_SEQUENCE_NUMBER", pa.int64(),
+                               nullable=False),
+                      pa.field("_VALUE_KIND", pa.int8(), nullable=False),
+                      pa.field("v_c0", pa.int32())]
+            tbl = pa.Table.from_arrays(
+                [pa.array(k1), pa.array(k2), pa.array(runs[i]["seq"]),
+                 pa.array(runs[i]["kind"]), pa.array(vals)],
+                schema=pa.schema(fields))
+            path = os.path.join(str(tmp_path), f"run-{i}.parquet")
+            pq.write_table(tbl, path, compression=None, use_dictionary=False,
+                           data_page_version="1.0", store_schema=False)
+            metas.append({"path": path, "rowCount": 20_000,
+                          "minKey": self._comp(k1[0], k2[0]),
+                          "maxKey": self._comp(k1[-1], k2[-1]), "level": 0})
+        r, w = merge_dedup(runs, drop_delete=True)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas),
+                               [{"name": "_KEY_k1", "type": "int32"},
+                                {"name": "_KEY_k2", "type": "int32"}],
+                               [{"name": "v_c0", "type": "int32"}]) as plan:
+                got = _read_all_batches(plan)
+        e1 = np.array([runs[a]["k1"][b] for a, b in zip(r, w)], np.int32)
+        e2 = np.array([runs[a]["k2"][b] for a, b in zip(r, w)], np.int32)
+        es = np.array([runs[a]["seq"][b] for a, b in zip(r, w)], np.int64)
+        ev = np.array([runs[a]["values"][0][b] for a, b in zip(r, w)],
+                      np.int32)
+        assert (got["_KEY_k1"] == e1).all()
+        assert (got["_KEY_k2"] == e2).all()
+        assert (got["_SEQUENCE_NUMBER"] == es).all()
+        assert (got["v_c0"] == ev).all()
+
+    def test_key_bits_overflow_rejected(self, tmp_path):
+        runs = gen_runs_dedup(1, 100, n_value_cols=1, seed=79)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="64"):
+                MergeReadPlan(s, file_descs_from_metas(metas),
+                              [{"name": "_KEY_k", "type": "int64"},
+                               {"name": "x", "type": "int32"}],
+                              _value_cols(1))
