@@ -819,6 +819,44 @@ static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
                     pres_len = (int64_t)pres_dec.size();
                 }
             }
+            if (ckind == ORC_FLOAT || ckind == ORC_DOUBLE) {
+                // FLOAT/DOUBLE DATA streams are raw IEEE754 LE values — the
+                // stream IS the decoded representation, so staging is a copy
+                // (layout, not decode), exactly like parquet PLAIN packing.
+                // With PRESENT, the stream holds only the non-null values
+                // and feeds the dense buffer the level scatter consumes.
+                int64_t row0 = row_base + stripe_row;
+                if (present) {
+                    rc.has_nulls = true;
+                    int64_t before = rc.dense_before;
+                    if (!prescan_present(pres_ptr, pres_len, st.num_rows,
+                                         row0, rc))
+                        return false;
+                    int64_t n_dense = rc.dense_before - before;
+                    if (data_len < n_dense * stored) {
+                        set_error("%s col %s: FLOAT/DOUBLE stream short",
+                                  fd.path.c_str(), cols[c].name.c_str());
+                        return false;
+                    }
+                    rc.dense_host.insert(rc.dense_host.end(), data_ptr,
+                                         data_ptr + n_dense * stored);
+                    rc.dense_segs.emplace_back(before, n_dense * stored);
+                } else {
+                    if (data_len < st.num_rows * stored) {
+                        set_error("%s col %s: FLOAT/DOUBLE stream short",
+                                  fd.path.c_str(), cols[c].name.c_str());
+                        return false;
+                    }
+                    if (hipMemcpy((uint8_t *)rc.contig + row0 * stored,
+                                  data_ptr, st.num_rows * stored,
+                                  hipMemcpyHostToDevice) != hipSuccess) {
+                        set_error("H2D failed");
+                        return false;
+                    }
+                }
+                plan->encoded_bytes_total += data_len;
+                continue;
+            }
             // upload the encoded DATA stream (+16 B pad: the RLEv2
             // bit reader uses an aligned 16-byte window)
             void *dev = plan->bufs.alloc(data_len + 16);

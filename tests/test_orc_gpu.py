@@ -131,3 +131,60 @@ class TestOrcPartialUpdate:
                 gm = np.ones(len(got[nm]), dtype=bool)
             assert (gm == em).all(), nm
             assert (got[nm][em] == ev[em]).all(), nm
+
+
+class TestOrcFloatDouble:
+    def test_orc_float_double_columns(self, tmp_path):
+        # FLOAT/DOUBLE DATA streams are raw IEEE754: staged as a copy, merged
+        # bit-exactly
+        rng = np.random.default_rng(86)
+        runs = []
+        seqs = rng.permutation(45_000).astype(np.int64)
+        for i in range(3):
+            keys = np.sort(rng.choice(70_000, 15_000,
+                                      replace=False)).astype(np.int64)
+            runs.append({
+                "key": keys, "seq": seqs[i*15_000:(i+1)*15_000],
+                "kind": rng.choice([0, 3], 15_000, p=[.9, .1]).astype(np.int8),
+                "values": [keys.copy(),
+                           rng.standard_normal(15_000).astype(np.float32),
+                           rng.standard_normal(15_000)]})
+        metas = write_runs(runs, str(tmp_path), file_format="orc",
+                           compression="zlib")
+        r, w = merge_dedup(runs, drop_delete=True)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               [{"name": "v_k", "type": "int64"},
+                                {"name": "v_c0", "type": "float32"},
+                                {"name": "v_c1", "type": "float64"}]) as plan:
+                got = _read_all(plan)
+        for nm, c, dt in (("v_c0", 1, np.float32), ("v_c1", 2, np.float64)):
+            e = np.array([runs[a]["values"][c][b] for a, b in zip(r, w)], dt)
+            assert got[nm].dtype == dt
+            assert (got[nm].view(np.uint8) == e.view(np.uint8)).all(), nm
+
+    def test_orc_nullable_double_pu(self, tmp_path):
+        rng = np.random.default_rng(87)
+        runs = gen_runs_partial_update(3, 12_000, n_value_cols=2, seed=87,
+                                       update_frac=0.5, update_cols=1)
+        for r in runs:
+            n = len(r["key"])
+            r["values"][1] = rng.standard_normal(n)
+            r["values"][2] = rng.standard_normal(n).astype(np.float32)
+        metas = write_runs(runs, str(tmp_path), file_format="orc")
+        exp = partial_update_model(runs)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               [{"name": "v_k", "type": "int64"},
+                                {"name": "v_c0", "type": "float64"},
+                                {"name": "v_c1", "type": "float32"}],
+                               merge_engine="partial-update") as plan:
+                got = _read_all(plan)
+        assert (got["_KEY_k"] == exp["key"]).all()
+        for c, nm in ((1, "v_c0"), (2, "v_c1")):
+            em = exp["valid"][c]
+            gm = got.get(nm + "#valid")
+            if gm is None:
+                gm = np.ones(len(got[nm]), dtype=bool)
+            assert (gm == em).all(), nm
+            assert (got[nm][em] == exp["values"][c][em]).all(), nm
