@@ -10,6 +10,7 @@ from .oracle import (  # noqa: F401
     merge_dedup_model,
     merge_first_row_model,
     partial_update_model,
+    partial_update_rrod_model,
     aggregation_model,
     rle_bp_decode,
     lib_path,

@@ -148,6 +148,61 @@ def merge_dedup_model(runs, ignore_delete=False, drop_delete=True):
     return run[sel], row[sel]
 
 
+def partial_update_rrod_model(runs, drop_delete=True):
+    """PartialUpdateMergeFunction with remove-record-on-delete
+    (PartialUpdateMergeFunction.java:173-180, getResult :213-221) for
+    INSERT/DELETE streams: a DELETE re-initializes the row from the DELETE
+    record's own value fields; adds newer than the last DELETE overlay
+    non-null fields; result kind = DELETE when the last member is the
+    DELETE (dropped under drop_delete); seq = last member's."""
+    key, seq, kind, run, row = _sorted_stream(runs)
+    assert ((kind == 0) | (kind == 2) | (kind == 3)).all(), \
+        "rrod model: INSERT/UPDATE_AFTER/DELETE only"
+    n = len(key)
+    n_cols = len(runs[0]["values"]) if runs else 0
+    out = {"key": [], "seq": [], "kind": [],
+           "values": [[] for _ in range(n_cols)],
+           "valid": [[] for _ in range(n_cols)]}
+    i = 0
+    while i < n:
+        j = i
+        while j + 1 < n and key[j + 1] == key[i]:
+            j += 1
+        members = list(range(i, j + 1))  # ascending (seq, isAdd)
+        dels = [m for m in members if kind[m] == 3]
+        last_del = dels[-1] if dels else None
+        res_kind = 3 if kind[members[-1]] == 3 else 0
+        if not (drop_delete and res_kind == 3):
+            out["key"].append(key[i])
+            out["seq"].append(seq[members[-1]])
+            out["kind"].append(res_kind)
+            for c in range(n_cols):
+                val, ok = 0, False
+                for m in members:
+                    if last_del is not None and m < last_del:
+                        continue
+                    a, b = run[m], row[m]
+                    valid = runs[a].get("valid")
+                    mv = bool(valid[c][b]) if valid is not None else True
+                    if m == last_del:
+                        # initRow: the DELETE's field value, even null
+                        val = runs[a]["values"][c][b] if mv else 0
+                        ok = mv
+                    elif kind[m] != 3 and mv:
+                        val, ok = runs[a]["values"][c][b], True
+                out["values"][c].append(val if ok else 0)
+                out["valid"][c].append(ok)
+        i = j + 1
+    return {
+        "key": np.array(out["key"], np.int64),
+        "seq": np.array(out["seq"], np.int64),
+        "kind": np.array(out["kind"], np.int8),
+        "values": [np.array(v, runs[0]["values"][c].dtype)
+                   for c, v in enumerate(out["values"])],
+        "valid": [np.array(v, bool) for v in out["valid"]],
+    }
+
+
 def partial_update_model(runs, drop_delete=True):
     """Numpy model of PartialUpdateMergeFunction for INSERT-only streams
     (PartialUpdateMergeFunction.java:188-215 updateNonNullFields +

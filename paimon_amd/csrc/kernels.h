@@ -129,7 +129,7 @@ hipError_t pmh_launch_level_scatter(const RleChunk *chunks, int64_t n_chunks,
 // (k_pack_valid), or null for the legacy per-column byte walk (>64 cols).
 hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                               const uint8_t *col_nullable, int n_cols, int k,
-                              int seq_col, int kind_col,
+                              int seq_col, int kind_col, int flags,
                               const uint32_t *members,
                               const uint16_t *group_start,
                               const int64_t *tile_offsets, int64_t n_tiles,

@@ -257,6 +257,11 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
     // function (wrapper) exactly as deduplicate does. A template parameter so
     // the deduplicate winner walk carries no extra select.
     constexpr bool first_row = FR;
+    // partial-update.remove-record-on-delete (PU mode): DELETE members are
+    // legal; a group whose LAST member (max packed sseq) is a DELETE yields
+    // a DELETE result and is dropped here under drop_delete. UPDATE_BEFORE
+    // is rejected at staging (v1: INSERT/DELETE streams).
+    const bool rrod = (flags & 16) != 0;
     // ablation levels (profiling only, flags bits 8..): 1=stage,2=+merge,
     // 3=+scan, 0/absent=full. Partial levels publish a checksum so the
     // compiler cannot dead-code the ablated phases' inputs.
@@ -331,6 +336,8 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             for (int32_t i = tid; i < len; i += blockDim.x) {
                 sm.skey[off + i] = key_at(kaddr0, i, kes);
                 int32_t kd = daddr[i];
+                if (PU && rrod && kd == 1 && err_flag)
+                    atomicOr(err_flag, 1u);  // UPDATE_BEFORE: not in v1 RROD
                 sm.sseq[off + i] =
                     (saddr[i] << 1) | (int64_t)(kd == 0 || kd == 2);
                 sm.perm[0][off + i] = (uint16_t)(off + i);
@@ -439,6 +446,16 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                     if (!sm.head[i]) continue;
                     int32_t tail = i;
                     while (tail + 1 < M && !sm.head[tail + 1]) tail++;
+                    if (rrod && drop_delete) {
+                        // result kind = last member's kind (max packed sseq);
+                        // DELETE results drop here (DropDeleteReader)
+                        int64_t mx = INT64_MIN;
+                        for (int32_t x = i; x <= tail; x++) {
+                            int64_t v = sm.sseq[mo[x]];
+                            if (v > mx) mx = v;
+                        }
+                        if (!(mx & 1)) continue;
+                    }
                     if (pass == 1) {
                         gout[g_off + ng] = (uint16_t)(m_off + nm);
                         // merged order within a group is (key, run); the
@@ -459,7 +476,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                         }
                         for (int x = 0; x < gn; x++) {
                             uint16_t s = gm[x];
-                            if (!(sm.sseq[s] & 1)) bad_kind = true;
+                            if (!rrod && !(sm.sseq[s] & 1)) bad_kind = true;
                             int r = 0;
                             while (r + 1 <= k - 1 &&
                                    sm.segoff[r + 1] <= (int32_t)s)
@@ -824,12 +841,20 @@ __global__ void k_pack_valid(const DevCol *cols, int n_cols, int64_t rows,
 template <bool MASKS>
 __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                           const uint8_t *col_nullable, int n_cols, int k,
-                          int seq_col, int kind_col, const uint32_t *members,
+                          int seq_col, int kind_col, int flags,
+                          const uint32_t *members,
                           const uint16_t *group_start,
                           const int64_t *tile_offsets, int64_t n_tiles,
                           int64_t tile_rows, const int64_t *total_out,
                           uint64_t *const *run_masks, void *const *out_ptrs,
                           uint8_t *const *out_valid) {
+    // remove-record-on-delete (flags bit 16): a DELETE resets the row —
+    // per column, overlay only the adds NEWER than the last DELETE and fall
+    // back to the DELETE record's own field (initRow semantics,
+    // PartialUpdateMergeFunction.java:173-180). Result kind = DELETE when
+    // the last member is the DELETE (getResult, :213-221). Groups whose result
+    // DELETE were already dropped by the merge pass under drop_delete.
+    const bool rrod = (flags & 16) != 0;
     const int64_t total = *total_out;
     const int64_t per_block =
         (total + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
@@ -873,9 +898,26 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
         uint32_t last = mc[0];
         int lrun = last >> 28;
         int64_t lrow = last & 0x0fffffff;
+        // newest-first index of the last DELETE member; gn = none (also the
+        // value when RROD is off, which neutralizes every bound below)
+        int j_del = gn;
+        if (rrod) {
+#pragma unroll
+            for (int x = 0; x < 4; x++) {
+                if (x >= gn || j_del < gn) continue;
+                const DevCol &dc = cols[(mc[x] >> 28) * n_cols + kind_col];
+                if (col_load<int32_t>(dc, mc[x] & 0x0fffffff) == 3) j_del = x;
+            }
+            for (int x = 4; x < gn && j_del == gn; x++) {
+                uint32_t m = mem[me - 1 - x];
+                const DevCol &dc = cols[(m >> 28) * n_cols + kind_col];
+                if (col_load<int32_t>(dc, m & 0x0fffffff) == 3) j_del = x;
+            }
+        }
         for (int c = 0; c < n_cols; c++) {
             if (c == kind_col) {
-                ((int8_t *)out_ptrs[c])[i] = 0;  // RowKind.INSERT
+                ((int8_t *)out_ptrs[c])[i] =
+                    (rrod && j_del == 0) ? 3 : 0;  // DELETE : INSERT
                 continue;
             }
             int64_t run = lrun, row = lrow;
@@ -884,7 +926,7 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                 ok = 0;
 #pragma unroll
                 for (int x = 0; x < 4; x++) {
-                    if (ok || x >= gn) continue;
+                    if (ok || x >= gn || x >= j_del) continue;
                     uint32_t m = mc[x];
                     uint8_t v;
                     if (MASKS) {
@@ -902,6 +944,7 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                     }
                 }
                 for (int32_t x = me - 5; !ok && x >= ms; x--) {
+                    if (me - 1 - x >= j_del) break;  // not newer than DELETE
                     uint32_t m = mem[x];
                     uint8_t v;
                     if (MASKS) {
@@ -926,6 +969,26 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                 } else {
                     const DevCol &dc = cols[lrun * n_cols + c];
                     ok = dc.valid0 ? ((const uint8_t *)dc.valid0)[lrow] : 1;
+                }
+            }
+            if (!ok && j_del < gn && col_nullable[c] && gn > 1) {
+                // no add newer than the DELETE set this field: the row was
+                // re-initialized from the DELETE record's value (initRow)
+                uint32_t m = mem[me - 1 - j_del];
+                uint8_t v;
+                if (MASKS) {
+                    v = (uint8_t)(
+                        (run_masks[m >> 28][m & 0x0fffffff] >> c) & 1);
+                } else {
+                    const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                    v = dc.valid0
+                            ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                            : 1;
+                }
+                if (v) {
+                    run = m >> 28;
+                    row = m & 0x0fffffff;
+                    ok = 1;
                 }
             }
             const DevCol &dc = cols[run * n_cols + c];
@@ -1527,7 +1590,7 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
 
 hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                               const uint8_t *col_nullable, int n_cols, int k,
-                              int seq_col, int kind_col,
+                              int seq_col, int kind_col, int flags,
                               const uint32_t *members,
                               const uint16_t *group_start,
                               const int64_t *tile_offsets, int64_t n_tiles,
@@ -1538,15 +1601,15 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
     if (run_masks)
         hipLaunchKernelGGL(k_emit_pu<true>, dim3(2048), dim3(256), 0, stream,
                            cols, col_dtype, col_nullable, n_cols, k, seq_col,
-                           kind_col, members, group_start, tile_offsets,
-                           n_tiles, tile_rows, total_out, run_masks, out_ptrs,
-                           out_valid);
+                           kind_col, flags, members, group_start,
+                           tile_offsets, n_tiles, tile_rows, total_out,
+                           run_masks, out_ptrs, out_valid);
     else
         hipLaunchKernelGGL(k_emit_pu<false>, dim3(2048), dim3(256), 0, stream,
                            cols, col_dtype, col_nullable, n_cols, k, seq_col,
-                           kind_col, members, group_start, tile_offsets,
-                           n_tiles, tile_rows, total_out, run_masks, out_ptrs,
-                           out_valid);
+                           kind_col, flags, members, group_start,
+                           tile_offsets, n_tiles, tile_rows, total_out,
+                           run_masks, out_ptrs, out_valid);
     return hipGetLastError();
 }
 

@@ -256,6 +256,10 @@ struct pmh_plan_t {
     // composite key (>1 key column): order-preserving packed comparand
     bool composite_key = false;
     uint64_t key_shifts = 0, key_bits = 0;  // 8 bits per sub-key
+    // partial-update.remove-record-on-delete: a DELETE resets the row
+    // (PartialUpdateMergeFunction.java:173-180); v1 accepts INSERT/DELETE
+    // streams (UPDATE_BEFORE rejected)
+    bool rrod = false;
     std::vector<pmh::Section> sections;
     size_t cur_section = 0;
     int64_t rows_in_total = 0;
@@ -1490,8 +1494,10 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             if (!add_col(c)) return nullptr;
         std::string engine = j["merge_engine"].as_str("deduplicate");
         if (engine == "partial-update") {
-            plan->pu = true;  // INSERT-only v1 (PartialUpdateMergeFunction
-                              // rejects retracts by default, :170-186)
+            plan->pu = true;  // INSERT-only unless remove-record-on-delete
+                              // (PartialUpdateMergeFunction rejects retracts
+                              // by default, :170-186)
+            plan->rrod = j["remove_record_on_delete"].as_bool(false);
         } else if (engine == "first-row") {
             plan->first_row = true;  // FirstRowMergeFunction.java:32-77
         } else if (engine == "aggregation") {
@@ -1518,6 +1524,12 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         }
         plan->drop_delete = j["drop_delete"].as_bool(true);
         plan->ignore_delete = j["ignore_delete"].as_bool(false);
+        if (plan->rrod && plan->ignore_delete) {
+            set_error("remove-record-on-delete cannot be used with "
+                      "ignore-delete (PartialUpdateMergeFunction.java:"
+                      "491-495)");
+            return nullptr;
+        }
         plan->host_output = j["output"].as_str("device") == "host";
 
         std::vector<FileDesc> files;
@@ -1774,7 +1786,8 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     if (e != hipSuccess) return fail("partition", e);
     (void)hipEventRecord(ev[2], st);
     int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0) |
-                (p->pu ? 4 : 0) | (p->first_row ? 8 : 0);
+                (p->pu ? 4 : 0) | (p->first_row ? 8 : 0) |
+                (p->rrod ? 16 : 0);
     if (const char *ab = getenv("PMH_ABLATE"))  // profiling-only phase knob
         flags |= (atoi(ab) & 0xf) << 8;
     e = pmh_launch_merge_tiles(sec.key_cols, sec.seq_cols, sec.kind_cols,
@@ -1798,9 +1811,10 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     } else if (p->pu) {
         e = pmh_launch_emit_pu(
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,
-            p->n_key_cols, p->n_key_cols + 1, sec.winners, sec.group_start,
-            sec.tile_offsets, sec.n_tiles, PMH_TILE_ROWS, sec.total_dev,
-            sec.row_masks_dev, p->out_ptrs_dev, p->out_valid_dev, st);
+            p->n_key_cols, p->n_key_cols + 1, flags, sec.winners,
+            sec.group_start, sec.tile_offsets, sec.n_tiles, PMH_TILE_ROWS,
+            sec.total_dev, sec.row_masks_dev, p->out_ptrs_dev,
+            p->out_valid_dev, st);
     } else {
         e = pmh_launch_emit(sec.all_cols, p->col_dtype_dev,
                             p->col_nullable_dev, n_cols, k, sec.winners,
@@ -1820,12 +1834,17 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     e = hipStreamSynchronize(st);
     if (e != hipSuccess) return fail("stream sync", e);
     if (err_word & 1) {
-        set_error("%s with retract records is not supported "
-                  "(the reference default also rejects them, "
-                  "PartialUpdateMergeFunction.java:170-186); "
-                  "ignore-delete/sequence-group/retract-aggregator paths "
-                  "are later rounds",
-                  p->agg ? "aggregation" : "partial-update");
+        if (p->rrod)
+            set_error("remove-record-on-delete accepts INSERT/DELETE "
+                      "streams in v1; UPDATE_BEFORE records are not "
+                      "supported yet");
+        else
+            set_error("%s with retract records is not supported "
+                      "(the reference default also rejects them, "
+                      "PartialUpdateMergeFunction.java:170-186); configure "
+                      "'partial-update.remove-record-on-delete', or wait "
+                      "for the sequence-group/retract-aggregator rounds",
+                      p->agg ? "aggregation" : "partial-update");
         return -1;
     }
     if (err_word & 2) {

@@ -185,7 +185,8 @@ class MergeReadPlan:
 
     def __init__(self, session: Session, files, key_cols, value_cols,
                  merge_engine="deduplicate", drop_delete=True,
-                 ignore_delete=False, output="host", aggregations=None):
+                 ignore_delete=False, output="host", aggregations=None,
+                 remove_record_on_delete=False):
         self.lib = session.lib
         desc = {
             "key_cols": key_cols,
@@ -193,6 +194,7 @@ class MergeReadPlan:
             "merge_engine": merge_engine,
             "drop_delete": drop_delete,
             "ignore_delete": ignore_delete,
+            "remove_record_on_delete": remove_record_on_delete,
             "output": output,
             "files": files,
         }

@@ -143,3 +143,68 @@ class TestDictionaryWithNulls:
                 gm = np.ones(len(got[nm]), dtype=bool)
             assert (gm == em).all(), nm
             assert (got[nm][em] == ev[em]).all(), nm
+
+
+class TestRemoveRecordOnDelete:
+    def _run(self, tmp_path, runs, drop_delete=True):
+        from oracle import partial_update_rrod_model
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        exp = partial_update_rrod_model(runs, drop_delete=drop_delete)
+        n_vals = len(runs[0]["values"])
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(n_vals - 1),
+                               merge_engine="partial-update",
+                               drop_delete=drop_delete,
+                               remove_record_on_delete=True) as plan:
+                got = _read_all(plan)
+        assert (got["_KEY_k"] == exp["key"]).all()
+        assert (got["_SEQUENCE_NUMBER"] == exp["seq"]).all()
+        assert (got["_VALUE_KIND"] == exp["kind"]).all()
+        names = ["v_k"] + [f"v_c{i}" for i in range(n_vals - 1)]
+        for c, nm in enumerate(names):
+            ev, em = exp["values"][c], exp["valid"][c]
+            gm = got.get(nm + "#valid")
+            if gm is None:
+                gm = np.ones(len(got[nm]), dtype=bool)
+            assert (gm == em).all(), nm
+            assert (got[nm][em] == ev[em]).all(), nm
+
+    def test_rrod_basic(self, tmp_path):
+        rng = np.random.default_rng(88)
+        runs = gen_runs_partial_update(4, 25_000, n_value_cols=5, seed=88,
+                                       update_frac=0.4, update_cols=2)
+        for r in runs:  # ~15% deletes, fields per the existing masks
+            r["kind"] = np.where(rng.random(len(r["key"])) < 0.15, 3,
+                                 0).astype(np.int8)
+        self._run(tmp_path, runs, drop_delete=True)
+
+    def test_rrod_keep_delete(self, tmp_path):
+        rng = np.random.default_rng(89)
+        runs = gen_runs_partial_update(3, 15_000, n_value_cols=3, seed=89,
+                                       update_frac=0.5, update_cols=2)
+        for r in runs:
+            r["kind"] = np.where(rng.random(len(r["key"])) < 0.25, 3,
+                                 0).astype(np.int8)
+        self._run(tmp_path, runs, drop_delete=False)
+
+    def test_rrod_rejects_update_before(self, tmp_path):
+        runs = gen_runs_partial_update(2, 4_000, n_value_cols=2, seed=90)
+        runs[0]["kind"][::9] = 1  # UPDATE_BEFORE
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(2), merge_engine="partial-update",
+                               remove_record_on_delete=True) as plan:
+                with pytest.raises(RuntimeError, match="UPDATE_BEFORE"):
+                    _read_all(plan)
+
+    def test_rrod_conflicts_with_ignore_delete(self, tmp_path):
+        runs = gen_runs_partial_update(1, 100, n_value_cols=1, seed=91)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="ignore-delete"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(1), merge_engine="partial-update",
+                              remove_record_on_delete=True,
+                              ignore_delete=True)
