@@ -12,6 +12,7 @@ from .oracle import (  # noqa: F401
     partial_update_model,
     partial_update_rrod_model,
     aggregation_model,
+    aggregation_rrod_model,
     rle_bp_decode,
     lib_path,
 )

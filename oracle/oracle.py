@@ -342,6 +342,88 @@ def aggregation_model(runs, aggs=None):
     }
 
 
+def aggregation_rrod_model(runs, aggs=None, drop_delete=True):
+    """AggregateMergeFunction with remove-record-on-delete for INSERT/DELETE
+    streams (AggregateMergeFunction.java:85-91 currentDeleteRow + initRow;
+    expected-model shape: MergeFunctionTestUtils.getExpectedForAggSum's
+    removeRecordOndelete branch): a DELETE re-initializes the row from its
+    own value and the aggregators continue from those values; result kind =
+    DELETE iff the last member is the DELETE. first_* aggregators excluded
+    (their initialized-state does not reset on DELETE)."""
+    key, seq, kind, run, row = _sorted_stream(runs)
+    assert ((kind == 0) | (kind == 2) | (kind == 3)).all()
+    n = len(key)
+    n_cols = len(runs[0]["values"]) if runs else 0
+    if aggs is None:
+        aggs = ["last_non_null_value"] * n_cols
+    assert not any(a.startswith("first") for a in aggs)
+    out = {"key": [], "seq": [], "kind": [],
+           "values": [[] for _ in range(n_cols)],
+           "valid": [[] for _ in range(n_cols)]}
+    i = 0
+    while i < n:
+        j = i
+        while j + 1 < n and key[j + 1] == key[i]:
+            j += 1
+        members = list(range(i, j + 1))
+        dels = [m for m in members if kind[m] == 3]
+        d = dels[-1] if dels else None
+        res_kind = 3 if kind[members[-1]] == 3 else 0
+        if len(members) == 1:  # ReducerMergeFunctionWrapper bypass
+            res_kind = int(kind[members[0]] == 3) * 3
+        if not (drop_delete and res_kind == 3):
+            out["key"].append(key[i])
+            out["seq"].append(seq[members[-1]])
+            out["kind"].append(res_kind)
+            for c in range(n_cols):
+                def fv(m):
+                    a, b = run[m], row[m]
+                    valid = runs[a].get("valid")
+                    mv = bool(valid[c][b]) if valid is not None else True
+                    return runs[a]["values"][c][b], mv
+                if len(members) == 1:
+                    val, ok = fv(members[0])
+                    out["values"][c].append(val if ok else 0)
+                    out["valid"][c].append(ok)
+                    continue
+                agg = aggs[c]
+                seeds = []
+                if d is not None:
+                    v, mv = fv(d)
+                    if mv:
+                        seeds.append(v)
+                adds = [m for m in members
+                        if kind[m] != 3 and (d is None or m > d)]
+                vals = seeds + [fv(m)[0] for m in adds if fv(m)[1]]
+                if agg == "last_value":
+                    val, ok = fv(members[-1])
+                elif agg == "last_non_null_value":
+                    ok = len(vals) > 0
+                    val = vals[-1] if ok else 0
+                elif agg == "sum":
+                    ok = len(vals) > 0
+                    val = sum(int(v) for v in vals) if ok else 0
+                elif agg == "max":
+                    ok = len(vals) > 0
+                    val = max(vals) if ok else 0
+                elif agg == "min":
+                    ok = len(vals) > 0
+                    val = min(vals) if ok else 0
+                else:
+                    raise ValueError(agg)
+                out["values"][c].append(val if ok else 0)
+                out["valid"][c].append(ok)
+        i = j + 1
+    return {
+        "key": np.array(out["key"], np.int64),
+        "seq": np.array(out["seq"], np.int64),
+        "kind": np.array(out["kind"], np.int8),
+        "values": [np.array(v, runs[0]["values"][c].dtype)
+                   for c, v in enumerate(out["values"])],
+        "valid": [np.array(v, bool) for v in out["valid"]],
+    }
+
+
 def rle_bp_decode(data: bytes, bit_width: int, num_values: int):
     """C restatement of the Parquet RLE/bit-packed hybrid decoder."""
     lib = _get_lib()

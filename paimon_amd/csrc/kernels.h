@@ -166,7 +166,7 @@ enum {
 hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                                const uint8_t *col_nullable,
                                const uint8_t *col_agg, int n_cols, int k,
-                               int seq_col, int kind_col,
+                               int seq_col, int kind_col, int flags,
                                const uint32_t *members,
                                const uint16_t *group_start,
                                const int64_t *tile_offsets, int64_t n_tiles,

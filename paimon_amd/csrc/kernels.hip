@@ -1047,12 +1047,19 @@ template <bool MASKS>
 __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                            const uint8_t *col_nullable,
                            const uint8_t *col_agg, int n_cols, int k,
-                           int seq_col, int kind_col, const uint32_t *members,
+                           int seq_col, int kind_col, int flags,
+                           const uint32_t *members,
                            const uint16_t *group_start,
                            const int64_t *tile_offsets, int64_t n_tiles,
                            int64_t tile_rows, const int64_t *total_out,
                            uint64_t *const *run_masks, void *const *out_ptrs,
                            uint8_t *const *out_valid) {
+    // remove-record-on-delete (flags bit 16): a DELETE re-initializes the
+    // row from its own value and the aggregators continue FROM those values
+    // (AggregateMergeFunction.add, currentDeleteRow path) — so folds seed
+    // with the last DELETE's field and consume only newer adds. first_*
+    // aggregators are rejected at plan creation under this mode.
+    const bool rrod = (flags & 16) != 0;
     auto valid_of = [&](uint32_t m, int c) -> uint8_t {
         if (MASKS)
             return (uint8_t)((run_masks[m >> 28][m & 0x0fffffff] >> c) & 1);
@@ -1113,13 +1120,43 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
         }
         const int lrun = last >> 28;
         const int64_t lrow = last & 0x0fffffff;
+        // ascending index of the LAST DELETE member (-1 = none)
+        int d_del = -1;
+        uint32_t mdel = 0;
+        uint64_t vdel = 0;
+        if (rrod) {
+#pragma unroll
+            for (int x = 0; x < 4; x++) {
+                if (x >= gn) continue;
+                const DevCol &dc = cols[(ma[x] >> 28) * n_cols + kind_col];
+                if (col_load<int32_t>(dc, ma[x] & 0x0fffffff) == 3) d_del = x;
+            }
+            for (int x = 4; x < gn; x++) {
+                uint32_t m = mem[ms + x];
+                const DevCol &dc = cols[(m >> 28) * n_cols + kind_col];
+                if (col_load<int32_t>(dc, m & 0x0fffffff) == 3) d_del = x;
+            }
+            if (d_del >= 0) {
+                mdel = mem[ms + d_del];
+                if (MASKS) vdel = run_masks[mdel >> 28][mdel & 0x0fffffff];
+            }
+        }
         for (int c = 0; c < n_cols; c++) {
             if (c == kind_col) {
-                ((int8_t *)out_ptrs[c])[i] = 0;  // RowKind.INSERT
+                ((int8_t *)out_ptrs[c])[i] =
+                    (rrod && d_del == gn - 1) ? 3 : 0;  // DELETE : INSERT
                 continue;
             }
             const int dt = col_dtype[c];
             const int agg = gn == 1 ? PMH_AGG_LAST_VALUE : col_agg[c];
+            auto del_valid = [&]() -> uint8_t {
+                if (!col_nullable[c]) return 1;
+                if (MASKS) return (uint8_t)((vdel >> c) & 1);
+                const DevCol &dc = cols[(mdel >> 28) * n_cols + c];
+                return dc.valid0
+                           ? ((const uint8_t *)dc.valid0)[mdel & 0x0fffffff]
+                           : 1;
+            };
             int64_t run = lrun, row = lrow;
             uint8_t ok = 1;
             int64_t bits = 0;  // result payload (raw stored bits / int value)
@@ -1141,10 +1178,12 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 break;
             }
             case PMH_AGG_LAST_NON_NULL:
-                if (col_nullable[c]) {
+                if (col_nullable[c] || d_del >= 0) {
                     ok = 0;
-                    // newest first: uncached tail (x >= 4), then cached
+                    // newest first: uncached tail (x >= 4), then cached;
+                    // only members NEWER than the last DELETE participate
                     for (int32_t x = gn - 1; !ok && x >= 4; x--) {
+                        if (x <= d_del) break;
                         uint32_t m = mem[ms + x];
                         if (valid_of(m, c)) {
                             run = m >> 28;
@@ -1154,7 +1193,7 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                     }
 #pragma unroll
                     for (int x = 3; x >= 0; x--) {
-                        if (ok || x >= gn) continue;
+                        if (ok || x >= gn || x <= d_del) continue;
                         uint8_t v = MASKS ? (uint8_t)((vma[x] >> c) & 1)
                                           : valid_of(ma[x], c);
                         if (v) {
@@ -1162,6 +1201,11 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                             row = ma[x] & 0x0fffffff;
                             ok = 1;
                         }
+                    }
+                    if (!ok && d_del >= 0 && del_valid()) {
+                        run = mdel >> 28;  // initRow: the DELETE's field
+                        row = mdel & 0x0fffffff;
+                        ok = 1;
                     }
                 }
                 break;
@@ -1234,9 +1278,11 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                         if (take) iacc = vb;
                     }
                 };
+                if (d_del >= 0 && del_valid())
+                    fold_one(mdel);  // aggregators continue from initRow
 #pragma unroll
                 for (int x = 0; x < 4; x++) {
-                    if (x >= gn) continue;
+                    if (x >= gn || x <= d_del) continue;
                     if (col_nullable[c] &&
                         !(MASKS ? (uint8_t)((vma[x] >> c) & 1)
                                 : valid_of(ma[x], c)))
@@ -1244,6 +1290,7 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                     fold_one(ma[x]);
                 }
                 for (int32_t x = 4; x < gn; x++) {
+                    if (x <= d_del) continue;
                     uint32_t m = mem[ms + x];
                     if (col_nullable[c] && !valid_of(m, c)) continue;
                     fold_one(m);
@@ -1631,7 +1678,7 @@ hipError_t pmh_launch_pack_valid(const DevCol *cols, int n_cols, int64_t rows,
 hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                                const uint8_t *col_nullable,
                                const uint8_t *col_agg, int n_cols, int k,
-                               int seq_col, int kind_col,
+                               int seq_col, int kind_col, int flags,
                                const uint32_t *members,
                                const uint16_t *group_start,
                                const int64_t *tile_offsets, int64_t n_tiles,
@@ -1643,15 +1690,15 @@ hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
     if (run_masks)
         hipLaunchKernelGGL(k_emit_agg<true>, dim3(2048), dim3(256), 0, stream,
                            cols, col_dtype, col_nullable, col_agg, n_cols, k,
-                           seq_col, kind_col, members, group_start,
+                           seq_col, kind_col, flags, members, group_start,
                            tile_offsets, n_tiles, tile_rows, total_out,
                            run_masks, out_ptrs, out_valid);
     else
         hipLaunchKernelGGL(k_emit_agg<false>, dim3(2048), dim3(256), 0,
                            stream, cols, col_dtype, col_nullable, col_agg,
-                           n_cols, k, seq_col, kind_col, members, group_start,
-                           tile_offsets, n_tiles, tile_rows, total_out,
-                           run_masks, out_ptrs, out_valid);
+                           n_cols, k, seq_col, kind_col, flags, members,
+                           group_start, tile_offsets, n_tiles, tile_rows,
+                           total_out, run_masks, out_ptrs, out_valid);
     return hipGetLastError();
 }
 

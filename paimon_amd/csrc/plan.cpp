@@ -1505,6 +1505,7 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             // over the same grouped stream as partial-update
             plan->pu = true;
             plan->agg = true;
+            plan->rrod = j["remove_record_on_delete"].as_bool(false);
         } else if (engine != "deduplicate") {
             set_error("merge engine '%s' not on the GPU path (deduplicate | "
                       "partial-update | aggregation | first-row)",
@@ -1680,6 +1681,15 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                               kv.second.as_str().c_str());
                     return nullptr;
                 }
+                if (plan->rrod && (code == PMH_AGG_FIRST_VALUE ||
+                                   code == PMH_AGG_FIRST_NON_NULL)) {
+                    set_error("aggregate function '%s' is not supported "
+                              "with remove-record-on-delete in v1 (its "
+                              "initialized-state does not reset on DELETE "
+                              "in the reference; later round)",
+                              kv.second.as_str().c_str());
+                    return nullptr;
+                }
                 ca[idx] = (uint8_t)code;
             }
             plan->col_agg_dev = (uint8_t *)plan->bufs.alloc(n_cols);
@@ -1805,9 +1815,9 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         e = pmh_launch_emit_agg(
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev,
             p->col_agg_dev, n_cols, k, p->n_key_cols, p->n_key_cols + 1,
-            sec.winners, sec.group_start, sec.tile_offsets, sec.n_tiles,
-            PMH_TILE_ROWS, sec.total_dev, sec.row_masks_dev, p->out_ptrs_dev,
-            p->out_valid_dev, st);
+            flags, sec.winners, sec.group_start, sec.tile_offsets,
+            sec.n_tiles, PMH_TILE_ROWS, sec.total_dev, sec.row_masks_dev,
+            p->out_ptrs_dev, p->out_valid_dev, st);
     } else if (p->pu) {
         e = pmh_launch_emit_pu(
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,

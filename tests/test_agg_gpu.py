@@ -152,3 +152,51 @@ class TestAggregationEngine:
                 MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
                               _value_cols(2), merge_engine="aggregation",
                               aggregations={"nope": "sum"})
+
+
+class TestAggRemoveRecordOnDelete:
+    def test_agg_rrod(self, tmp_path):
+        from oracle import aggregation_rrod_model
+        rng = np.random.default_rng(102)
+        runs = gen_runs_partial_update(4, 20_000, n_value_cols=4, seed=102,
+                                       update_frac=0.4, update_cols=2)
+        for r in runs:
+            n = len(r["key"])
+            r["kind"] = np.where(rng.random(n) < 0.2, 3, 0).astype(np.int8)
+            for c in range(1, 5):
+                r["values"][c] = (r["values"][c] % 10_000).astype(np.int32)
+        aggs_map = {"v_c0": "sum", "v_c1": "max", "v_c2": "min",
+                    "v_c3": "last_non_null_value"}
+        aggs_list = ["last_non_null_value", "sum", "max", "min",
+                     "last_non_null_value"]
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        for dd in (True, False):
+            exp = aggregation_rrod_model(runs, aggs_list, drop_delete=dd)
+            with Session(0) as s:
+                with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                                   _value_cols(4),
+                                   merge_engine="aggregation",
+                                   aggregations=aggs_map, drop_delete=dd,
+                                   remove_record_on_delete=True) as plan:
+                    got = _read_all(plan)
+            assert (got["_KEY_k"] == exp["key"]).all(), dd
+            assert (got["_SEQUENCE_NUMBER"] == exp["seq"]).all(), dd
+            assert (got["_VALUE_KIND"] == exp["kind"]).all(), dd
+            names = ["v_k"] + [f"v_c{i}" for i in range(4)]
+            for c, nm in enumerate(names):
+                ev, em = exp["values"][c], exp["valid"][c]
+                gm = got.get(nm + "#valid")
+                if gm is None:
+                    gm = np.ones(len(got[nm]), dtype=bool)
+                assert (gm == em).all(), (dd, nm)
+                assert (got[nm][em] == ev[em]).all(), (dd, nm)
+
+    def test_first_agg_rejected_with_rrod(self, tmp_path):
+        runs = gen_runs_partial_update(1, 200, n_value_cols=2, seed=103)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="remove-record-on-delete"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(2), merge_engine="aggregation",
+                              aggregations={"v_c0": "first_value"},
+                              remove_record_on_delete=True)
