@@ -20,8 +20,10 @@
  *      buffers stay valid until the next pmh_read_next / pmh_plan_close on
  *      the same plan (releaseBatch reuse contract). One plan = one HIP
  *      stream; concurrent plans are allowed (one per bucket).
- *  - a compaction entry point — roadmap in DESIGN.md — will replace
- *      CompactRewriter.rewrite, mergetree/compact/CompactRewriter.java:29-56.
+ *  - pmh_write_parquet is the encode half of the compaction surface
+ *      (CompactRewriter.rewrite, mergetree/compact/CompactRewriter.java:
+ *      29-56): the merge half runs through a plan, the rolling write-back
+ *      through this writer (driven by paimon_amd/compact.py in-repo).
  *
  * JNI precedent in the reference: the Vortex JNI reader
  * (paimon-vortex/.../dev/vortex/jni/NativeRuntime.java:30-34) — a Java-side
@@ -126,17 +128,18 @@ const char *pmh_last_error(void);
 char *pmh_debug_footer_json(const char *path);
 void pmh_free_string(char *s);
 
-/* Write a Parquet v1 data file (PLAIN, uncompressed) from HOST columnar
+/* Write a Parquet v1 data file (PLAIN pages) from HOST columnar
  * buffers — the compaction write-back half of the CompactRewriter surface
  * (what KeyValueDataFileWriter + the vendored parquet-mr writer do in the
  * reference, io/KeyValueDataFileWriter.java:121-170). cols[i].data/valid
  * are host pointers at the column's output width (TINYINT/SMALLINT widen to
  * the INT32 physical type on write, matching the read path). No GPU
  * required. row_group_rows/page_rows <= 0 pick defaults (1M / 64k).
+ * compression: NULL or "NONE" for uncompressed, "zstd" for ZSTD pages.
  * Returns 0, or -1 with pmh_last_error() set. */
 int pmh_write_parquet(const pmh_col *cols, int32_t n_cols, int64_t n_rows,
                       const char *path, int64_t row_group_rows,
-                      int64_t page_rows);
+                      int64_t page_rows, const char *compression);
 
 /* Raw snappy block decode (the library's from-scratch decoder, used for
  * SNAPPY parquet pages / ORC chunks). Returns decompressed size or -1 with

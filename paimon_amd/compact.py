@@ -71,10 +71,7 @@ def rewrite(session: Session, file_metas, key_cols, value_cols, out_dir,
                                  valid[s:e] if valid is not None else None))
                 path = os.path.join(out_dir, f"{file_prefix}-{file_idx}.parquet")
                 file_idx += 1
-                if compression != "NONE":
-                    raise ValueError("native compact write is uncompressed "
-                                     "v1 (compressed encode: roadmap §8f.1)")
-                write_parquet(path, cols)
+                write_parquet(path, cols, compression=compression)
                 ks = key[s:e]
                 sq = seq[s:e]
                 kd = kind[s:e]

@@ -60,6 +60,37 @@ bool inflate_raw(const uint8_t *src, size_t n, uint8_t *dst, size_t cap,
 }
 }  // namespace
 
+bool zstd_compress_buf(const uint8_t *src, size_t n,
+                       std::vector<uint8_t> &out, std::string &err) {
+    typedef size_t (*compress_fn)(void *, size_t, const void *, size_t, int);
+    typedef size_t (*bound_fn)(size_t);
+    static compress_fn cfn = nullptr;
+    static bound_fn bfn = nullptr;
+    static zstd_iserror_fn efn = nullptr;
+    if (!cfn) {
+        void *h = dlopen("libzstd.so.1", RTLD_NOW | RTLD_GLOBAL);
+        if (!h) {
+            err = "libzstd.so.1 not found for zstd-compressed output";
+            return false;
+        }
+        cfn = (compress_fn)dlsym(h, "ZSTD_compress");
+        bfn = (bound_fn)dlsym(h, "ZSTD_compressBound");
+        efn = (zstd_iserror_fn)dlsym(h, "ZSTD_isError");
+        if (!cfn || !bfn || !efn) {
+            err = "ZSTD_compress symbols missing";
+            return false;
+        }
+    }
+    out.resize(bfn(n));
+    size_t r = cfn(out.data(), out.size(), src, n, 3);  // default level
+    if (efn(r)) {
+        err = "zstd compress failed";
+        return false;
+    }
+    out.resize(r);
+    return true;
+}
+
 bool zstd_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
                            size_t dst_n, std::string &err) {
     size_t got = 0;

@@ -10,6 +10,10 @@
 
 namespace pmh {
 
+// zstd-compress a buffer (parquet page write-back); level 3
+bool zstd_compress_buf(const uint8_t *src, size_t n,
+                       std::vector<uint8_t> &out, std::string &err);
+
 // one-shot zstd frame with known decompressed size (parquet pages)
 bool zstd_decompress_exact(const uint8_t *src, size_t n, uint8_t *dst,
                            size_t dst_n, std::string &err);

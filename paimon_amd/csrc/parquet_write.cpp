@@ -7,15 +7,16 @@
 // KeyValueDataFileWriter (io/KeyValueDataFileWriter.java:121-170) under
 // CompactRewriter. v1 matrix: INT8/16 (stored INT32), INT32, INT64, FLOAT,
 // DOUBLE; REQUIRED or OPTIONAL (byte validity -> RLE/bit-packed def levels);
-// one PLAIN data page per `page_rows`; no dictionary, no compression
-// (roadmap §8f.1). Readable by parquet-mr/pyarrow and by this library's own
-// reader (tests pin both).
+// one PLAIN data page per `page_rows`; UNCOMPRESSED or ZSTD pages (host
+// compress; no dictionary — GPU-side encode is roadmap §8f.1). Readable by
+// parquet-mr/pyarrow and by this library's own reader (tests pin both).
 
 #include "parquet_write.h"
 
 #include <cstdio>
 #include <cstring>
 
+#include "codec.h"
 #include "parquet_meta.h"
 
 namespace pmh {
@@ -152,12 +153,13 @@ void encode_values(const PwCol &c, int64_t s, int64_t e, std::string &out) {
     }
 }
 
-void page_header(int64_t n_vals, int32_t payload, std::string &out) {
+void page_header(int64_t n_vals, int32_t unc, int32_t comp,
+                 std::string &out) {
     TC t;
     int l0 = 0;
-    t.i32(l0, 1, 0);        // type = DATA_PAGE
-    t.i32(l0, 2, payload);  // uncompressed_page_size
-    t.i32(l0, 3, payload);  // compressed_page_size (uncompressed)
+    t.i32(l0, 1, 0);     // type = DATA_PAGE
+    t.i32(l0, 2, unc);   // uncompressed_page_size
+    t.i32(l0, 3, comp);  // compressed_page_size
     t.field(l0, 5, 12);     // data_page_header: struct
     {
         int l1 = 0;
@@ -175,7 +177,12 @@ void page_header(int64_t n_vals, int32_t payload, std::string &out) {
 
 bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
                    const std::string &path, int64_t row_group_rows,
-                   int64_t page_rows, std::string &err) {
+                   int64_t page_rows, int codec, std::string &err) {
+    if (codec != CODEC_UNCOMPRESSED && codec != CODEC_ZSTD) {
+        err = "write codec " + std::to_string(codec) +
+              " not supported (v1: UNCOMPRESSED, ZSTD)";
+        return false;
+    }
     for (const auto &c : cols)
         if (phys_of(c.dtype) < 0) {
             err = "unsupported dtype " + std::to_string(c.dtype) +
@@ -193,7 +200,8 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
 
     struct CcInfo {
         int64_t data_page_offset;
-        int64_t total_size;
+        int64_t total_size;       // compressed (on-file) bytes
+        int64_t total_unc_size;   // uncompressed payload + headers
         int64_t num_values;
     };
     struct RgInfo {
@@ -213,17 +221,38 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
             CcInfo cc{};
             cc.data_page_offset = (int64_t)buf.size();
             cc.num_values = rg1 - rg0;
+            int64_t unc_total = 0;
             for (int64_t p0 = rg0; p0 < rg1 || (rg1 == rg0 && p0 == rg0);
                  p0 += page_rows) {
                 int64_t p1 = p0 + page_rows < rg1 ? p0 + page_rows : rg1;
                 std::string payload;
                 if (c.valid) encode_def_levels(c.valid + p0, p1 - p0, payload);
                 encode_values(c, p0, p1, payload);
-                page_header(p1 - p0, (int32_t)payload.size(), buf);
-                buf.append(payload);
+                if (codec == CODEC_ZSTD) {
+                    std::vector<uint8_t> comp;
+                    if (!zstd_compress_buf((const uint8_t *)payload.data(),
+                                           payload.size(), comp, err)) {
+                        fclose(f);
+                        return false;
+                    }
+                    size_t h0 = buf.size();
+                    page_header(p1 - p0, (int32_t)payload.size(),
+                                (int32_t)comp.size(), buf);
+                    unc_total +=
+                        (int64_t)(buf.size() - h0) + (int64_t)payload.size();
+                    buf.append((const char *)comp.data(), comp.size());
+                } else {
+                    size_t h0 = buf.size();
+                    page_header(p1 - p0, (int32_t)payload.size(),
+                                (int32_t)payload.size(), buf);
+                    unc_total +=
+                        (int64_t)(buf.size() - h0) + (int64_t)payload.size();
+                    buf.append(payload);
+                }
                 if (rg1 == rg0) break;  // single empty page for 0 rows
             }
             cc.total_size = (int64_t)buf.size() - cc.data_page_offset;
+            cc.total_unc_size = unc_total;
             rg.bytes += cc.total_size;
             rg.ccs.push_back(cc);
         }
@@ -273,10 +302,10 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
                 t.list_begin(lm, 3, 8, 1);  // path_in_schema
                 t.uvarint(c.name.size());
                 t.out.append(c.name);
-                t.i32(lm, 4, CODEC_UNCOMPRESSED);
+                t.i32(lm, 4, codec);
                 t.i64(lm, 5, cc.num_values);
-                t.i64(lm, 6, cc.total_size);  // total_uncompressed_size
-                t.i64(lm, 7, cc.total_size);  // total_compressed_size
+                t.i64(lm, 6, cc.total_unc_size);  // total_uncompressed_size
+                t.i64(lm, 7, cc.total_size);      // total_compressed_size
                 t.i64(lm, 9, cc.data_page_offset);
                 t.stop();
             }

@@ -15,8 +15,9 @@ struct PwCol {
     const uint8_t *valid; // byte validity or null (REQUIRED)
 };
 
+// codec: CODEC_UNCOMPRESSED or CODEC_ZSTD (parquet_meta.h values)
 bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
                    const std::string &path, int64_t row_group_rows,
-                   int64_t page_rows, std::string &err);
+                   int64_t page_rows, int codec, std::string &err);
 
 }  // namespace pmh

@@ -1967,7 +1967,17 @@ int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap) {
 
 int pmh_write_parquet(const pmh_col *cols, int32_t n_cols, int64_t n_rows,
                       const char *path, int64_t row_group_rows,
-                      int64_t page_rows) {
+                      int64_t page_rows, const char *compression) {
+    int codec = CODEC_UNCOMPRESSED;
+    if (compression && *compression) {
+        std::string cs(compression);
+        if (cs == "zstd" || cs == "ZSTD") codec = CODEC_ZSTD;
+        else if (cs != "NONE" && cs != "none" && cs != "UNCOMPRESSED") {
+            set_error("write compression '%s' not supported (NONE | zstd)",
+                      compression);
+            return -1;
+        }
+    }
     if (!cols || n_cols <= 0 || !path) {
         set_error("pmh_write_parquet: bad arguments");
         return -1;
@@ -1984,7 +1994,8 @@ int pmh_write_parquet(const pmh_col *cols, int32_t n_cols, int64_t n_rows,
         }
     }
     std::string err;
-    if (!write_parquet(pc, n_rows, path, row_group_rows, page_rows, err)) {
+    if (!write_parquet(pc, n_rows, path, row_group_rows, page_rows, codec,
+                       err)) {
         set_error("%s", err.c_str());
         return -1;
     }

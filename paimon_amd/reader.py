@@ -82,7 +82,7 @@ def load_lib(required=True):
     lib.pmh_write_parquet.restype = ctypes.c_int
     lib.pmh_write_parquet.argtypes = [
         ctypes.POINTER(_Col), ctypes.c_int32, ctypes.c_int64,
-        ctypes.c_char_p, ctypes.c_int64, ctypes.c_int64]
+        ctypes.c_char_p, ctypes.c_int64, ctypes.c_int64, ctypes.c_char_p]
     lib.pmh_debug_interval_partition.restype = ctypes.c_int
     lib.pmh_debug_interval_partition.argtypes = [
         ctypes.c_int, ctypes.POINTER(ctypes.c_int64),
@@ -111,7 +111,8 @@ _NP_DT = {np.dtype(np.int8): 1, np.dtype(np.int16): 2,
           np.dtype(np.float32): 5, np.dtype(np.float64): 6}
 
 
-def write_parquet(path, columns, row_group_rows=0, page_rows=0):
+def write_parquet(path, columns, row_group_rows=0, page_rows=0,
+                  compression="NONE"):
     """Write a Parquet v1 data file (PLAIN, uncompressed) via the native
     writer (pmh_write_parquet) — the compaction write-back path. `columns`
     is an ordered list of (name, values[, valid]) with numpy arrays; valid
@@ -140,7 +141,8 @@ def write_parquet(path, columns, row_group_rows=0, page_rows=0):
         else:
             cols[i].valid = None
     rc = lib.pmh_write_parquet(cols, len(columns), n_rows, path.encode(),
-                               row_group_rows, page_rows)
+                               row_group_rows, page_rows,
+                               compression.encode())
     if rc != 0:
         raise RuntimeError(f"pmh_write_parquet: {last_error()}")
 
