@@ -127,3 +127,27 @@ def test_compact_partial_update_nullable(tmp_path):
                 if gv is None:
                     gv = np.ones(len(b[nm]), bool)
                 assert (gv.astype(bool) == exp["valid"][c]).all(), nm
+
+
+def test_compact_zstd_write(tmp_path):
+    # zstd write-back round-trips: pyarrow reads it AND the GPU reader
+    # re-merges the compacted level
+    runs = gen_runs_dedup(4, 10_000, n_value_cols=2, seed=94, delete_frac=0.1)
+    metas = write_runs(runs, str(tmp_path / "in"), compression="NONE")
+    with Session(0) as s:
+        res = rewrite(s, metas, KEY_COLS, _value_cols(2),
+                      str(tmp_path / "out"), output_level=5,
+                      drop_delete=True, compression="zstd",
+                      target_file_rows=1_000_000)
+    r, w = merge_dedup(runs, drop_delete=True)
+    exp_key = np.array([runs[a]["key"][b] for a, b in zip(r, w)], np.int64)
+    m = res["after"][0]
+    t = pq.read_table(m["path"])
+    assert pq.ParquetFile(m["path"]).metadata.row_group(0).column(0)\
+        .compression == "ZSTD"
+    assert (np.asarray(t.column("_KEY_k")) == exp_key).all()
+    with Session(0) as s:
+        with MergeReadPlan(s, file_descs_from_metas(res["after"]), KEY_COLS,
+                           _value_cols(2)) as plan:
+            b = plan.read_next()
+            assert (b["_KEY_k"] == exp_key).all()
