@@ -211,10 +211,14 @@ def partial_update_model(runs, drop_delete=True):
     record where that column is non-null; result seq = last record's seq,
     kind = INSERT. Returns dict of output columns.
 
+    Retract records are accepted ONLY in singleton groups: the wrapper
+    bypasses the merge function there (ReducerMergeFunctionWrapper.java:
+    53-73), so the lone record is served as-is with its own RowKind and a
+    retract result drops under drop_delete (DropDeleteReader.java:53-61).
+
     runs entries carry 'values' (list of arrays, col 0 = pk) and 'valid'
     (parallel list of boolean masks)."""
     key, seq, kind, run, row = _sorted_stream(runs)
-    assert _kind_is_add(kind).all(), "partial_update_model: INSERT-only streams"
     n = len(key)
     head = np.empty(n, dtype=bool)
     if n == 0:
@@ -223,9 +227,19 @@ def partial_update_model(runs, drop_delete=True):
     head[1:] = key[1:] != key[:-1]
     group_id = np.cumsum(head) - 1
     n_groups = group_id[-1] + 1
+    group_size = np.bincount(group_id, minlength=n_groups)
+    assert (_kind_is_add(kind) | (group_size[group_id] == 1)).all(), \
+        "partial_update_model: retracts only in singleton groups"
     idx = np.arange(n)
     last_all = np.zeros(n_groups, dtype=np.int64)
     np.maximum.at(last_all, group_id, idx)
+    # result kind: INSERT except singleton bypass (record's own kind)
+    out_kind = np.zeros(n_groups, dtype=np.int8)
+    singles = group_size == 1
+    out_kind[singles] = kind[last_all[singles]]
+    keep = np.ones(n_groups, dtype=bool)
+    if drop_delete:
+        keep = _kind_is_add(out_kind)
     n_cols = len(runs[0]["values"])
     out_vals, out_valid = [], []
     # merged order over the flattened records, same as _sorted_stream
@@ -242,29 +256,31 @@ def partial_update_model(runs, drop_delete=True):
         lastv = np.full(n_groups, -1, dtype=np.int64)
         np.maximum.at(lastv, group_id[msk], idx[msk])
         vals = np.where(lastv >= 0, col[np.clip(lastv, 0, None)], 0)
-        out_vals.append(vals.astype(col.dtype))
-        out_valid.append(lastv >= 0)
+        out_vals.append(vals.astype(col.dtype)[keep])
+        out_valid.append((lastv >= 0)[keep])
     return {
-        "key": key[last_all], "seq": seq[last_all],
-        "kind": np.zeros(n_groups, dtype=np.int8),
+        "key": key[last_all][keep], "seq": seq[last_all][keep],
+        "kind": out_kind[keep],
         "values": out_vals, "valid": out_valid,
     }
 
 
-def aggregation_model(runs, aggs=None):
+def aggregation_model(runs, aggs=None, drop_delete=True):
     """Model of AggregateMergeFunction (AggregateMergeFunction.java:82-125)
     for INSERT-only streams. aggs is a list of aggregate-function names, one
     per value column (default last_non_null_value, :197-203). Per key group
     in ascending (seq, isAdd) order each column folds through its
     FieldAggregator; result seq = last member's, kind = INSERT. Singleton
-    groups bypass the merge function (ReducerMergeFunctionWrapper.java:53-73).
+    groups bypass the merge function (ReducerMergeFunctionWrapper.java:53-73)
+    — so retracts are accepted in singleton groups only, served as-is with
+    their own RowKind (dropped under drop_delete, DropDeleteReader.java:
+    53-61).
 
     Follows Java numerics: int sums wrap at 64 bits here (Java's addExact
     overflow check is not modelled); float/double sums accumulate in the
     column's own precision in merge order; max/min compare like
     Float.compare/Double.compare (IEEE total order)."""
     key, seq, kind, run, row = _sorted_stream(runs)
-    assert _kind_is_add(kind).all(), "aggregation_model: INSERT-only streams"
     n = len(key)
     n_cols = len(runs[0]["values"]) if runs else 0
     if aggs is None:
@@ -317,6 +333,9 @@ def aggregation_model(runs, aggs=None):
     head[1:] = key[1:] != key[:-1]
     starts = np.nonzero(head)[0]
     ends = np.append(starts[1:], n)
+    sizes = ends - starts
+    assert (_kind_is_add(kind) | (sizes[np.cumsum(head) - 1] == 1)).all(), \
+        "aggregation_model: retracts only in singleton groups"
     cols = [np.concatenate([r["values"][c] for r in runs]) for c in
             range(n_cols)]
     msks = [np.concatenate([r["valid"][c] for r in runs]) if has_valid
@@ -335,10 +354,16 @@ def aggregation_model(runs, aggs=None):
             if ok:
                 out_vals[c][g] = v
             out_valid[c][g] = ok
+    out_kind = np.zeros(len(starts), dtype=np.int8)
+    singles = sizes == 1
+    out_kind[singles] = kind[starts[singles]]
+    keep = _kind_is_add(out_kind) if drop_delete \
+        else np.ones(len(starts), dtype=bool)
     return {
-        "key": key[ends - 1], "seq": seq[ends - 1],
-        "kind": np.zeros(len(starts), dtype=np.int8),
-        "values": out_vals, "valid": out_valid,
+        "key": key[ends - 1][keep], "seq": seq[ends - 1][keep],
+        "kind": out_kind[keep],
+        "values": [v[keep] for v in out_vals],
+        "valid": [v[keep] for v in out_valid],
     }
 
 

@@ -46,6 +46,11 @@ def rewrite(session: Session, file_metas, key_cols, value_cols, out_dir,
     plan = MergeReadPlan(session, file_descs_from_metas(file_metas), key_cols,
                          value_cols, merge_engine=merge_engine,
                          drop_delete=drop_delete, output="host")
+    # key/value split driven by the plan's declared key columns — the
+    # KeyValue column layout is key cols | _SEQUENCE_NUMBER | _VALUE_KIND |
+    # value cols (KeyValueSerializer.java:34-99)
+    key_names = [kc["name"] for kc in key_cols]
+    special = set(key_names) | {"_SEQUENCE_NUMBER", "_VALUE_KIND"}
     after = []
     file_idx = 0
     try:
@@ -53,18 +58,17 @@ def rewrite(session: Session, file_metas, key_cols, value_cols, out_dir,
             batch = plan.read_next()
             if batch is None:
                 break
-            n = len(batch["_KEY_k"])
+            n = len(batch[key_names[0]])
             if n == 0:
                 continue
-            key = batch["_KEY_k"]
             seq = batch["_SEQUENCE_NUMBER"]
             kind = batch["_VALUE_KIND"]
             for s, e in _roll_slices(n, target_file_rows):
-                cols = [("_KEY_k", key[s:e]), ("_SEQUENCE_NUMBER", seq[s:e]),
-                        ("_VALUE_KIND", kind[s:e])]
+                cols = [(kn, batch[kn][s:e]) for kn in key_names]
+                cols += [("_SEQUENCE_NUMBER", seq[s:e]),
+                         ("_VALUE_KIND", kind[s:e])]
                 for name, arr in batch.items():
-                    if name in ("_KEY_k", "_SEQUENCE_NUMBER", "_VALUE_KIND") \
-                            or name.endswith("#valid"):
+                    if name in special or name.endswith("#valid"):
                         continue
                     valid = batch.get(name + "#valid")
                     cols.append((name, arr[s:e],
@@ -72,16 +76,24 @@ def rewrite(session: Session, file_metas, key_cols, value_cols, out_dir,
                 path = os.path.join(out_dir, f"{file_prefix}-{file_idx}.parquet")
                 file_idx += 1
                 write_parquet(path, cols, compression=compression)
-                ks = key[s:e]
                 sq = seq[s:e]
                 kd = kind[s:e]
+                # minKey/maxKey are the first/last merged rows' full key
+                # tuples (DataFileMeta.java:124-190); scalar for the common
+                # single-column key, list for composite keys
+                if len(key_names) == 1:
+                    min_key = int(batch[key_names[0]][s])
+                    max_key = int(batch[key_names[0]][e - 1])
+                else:
+                    min_key = [int(batch[kn][s]) for kn in key_names]
+                    max_key = [int(batch[kn][e - 1]) for kn in key_names]
                 after.append({
                     "path": path,
                     "fileName": os.path.basename(path),
                     "fileSize": os.path.getsize(path),
                     "rowCount": int(e - s),
-                    "minKey": int(ks[0]),
-                    "maxKey": int(ks[-1]),
+                    "minKey": min_key,
+                    "maxKey": max_key,
                     "minSequenceNumber": int(sq.min()),
                     "maxSequenceNumber": int(sq.max()),
                     "deleteRowCount": int(np.count_nonzero(

@@ -9,10 +9,14 @@
 namespace pmh {
 
 // Hard capacity limits (validated host-side at plan creation):
-//  - PMH_MAX_RUNS sorted runs per section (spill path is a later round,
-//    MergeSorter.java:112-125);
-//  - rows per run < 2^28 (winner packing run:4 | row:28).
-constexpr int PMH_MAX_RUNS = 16;
+//  - PMH_MAX_RUNS sorted runs per section (the reference's disk-spill path,
+//    MergeSorter.java:112-125, stays out of scope while real compaction
+//    shapes fit — 32 covers C5's 16->1 with 2x headroom);
+//  - rows per run < 2^PMH_ROW_BITS (winner packing run:5 | row:27; a run is
+//    the concatenation of its files, the cap applies to the run TOTAL).
+constexpr int PMH_MAX_RUNS = 32;
+constexpr int PMH_ROW_BITS = 27;
+constexpr uint32_t PMH_ROW_MASK = ((uint32_t)1 << PMH_ROW_BITS) - 1;
 constexpr int PMH_TILE_THREADS = 512;
 constexpr int64_t PMH_TILE_ROWS = 3584;
 constexpr int PMH_TILE_MAX = PMH_TILE_ROWS + PMH_MAX_RUNS;

@@ -13,7 +13,7 @@
 // loads, LDS staging, grid-stride launches.
 //
 // Capacity limits (checked host-side): runs per section <= PMH_MAX_RUNS,
-// rows per run < 2^28 (packed winner format run:4 | row:28).
+// rows per run < 2^PMH_ROW_BITS (packed winner format run:5 | row:27).
 
 #include <hip/hip_runtime.h>
 
@@ -446,7 +446,14 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                     if (!sm.head[i]) continue;
                     int32_t tail = i;
                     while (tail + 1 < M && !sm.head[tail + 1]) tail++;
-                    if (rrod && drop_delete) {
+                    if (tail == i) {
+                        // ReducerMergeFunctionWrapper singleton bypass
+                        // (ReducerMergeFunctionWrapper.java:53-73): the lone
+                        // record is served as-is (any kind, incl. retracts);
+                        // a retract result drops under drop-delete
+                        // (DropDeleteReader.java:53-61)
+                        if (drop_delete && !(sm.sseq[mo[i]] & 1)) continue;
+                    } else if (rrod && drop_delete) {
                         // result kind = last member's kind (max packed sseq);
                         // DELETE results drop here (DropDeleteReader)
                         int64_t mx = INT64_MIN;
@@ -476,14 +483,18 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                         }
                         for (int x = 0; x < gn; x++) {
                             uint16_t s = gm[x];
-                            if (!rrod && !(sm.sseq[s] & 1)) bad_kind = true;
+                            // retracts only fail MULTI-record groups: the
+                            // wrapper bypasses the merge function for
+                            // singletons (gn == 1), so a lone retract passes
+                            if (!rrod && gn > 1 && !(sm.sseq[s] & 1))
+                                bad_kind = true;
                             int r = 0;
                             while (r + 1 <= k - 1 &&
                                    sm.segoff[r + 1] <= (int32_t)s)
                                 r++;
                             uint32_t grow = (uint32_t)(
                                 c0[r] + ((int32_t)s - sm.segoff[r]));
-                            mout[m_off + nm + x] = ((uint32_t)r << 28) | grow;
+                            mout[m_off + nm + x] = ((uint32_t)r << PMH_ROW_BITS) | grow;
                         }
                     }
                     ng++;
@@ -587,7 +598,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                         r++;
                     uint32_t grow = (uint32_t)(
                         c0[r] + ((int32_t)s_best - sm.segoff[r]));
-                    wout[my_off + nloc] = ((uint32_t)r << 28) | grow;
+                    wout[my_off + nloc] = ((uint32_t)r << PMH_ROW_BITS) | grow;
                 }
                 nloc++;
             }
@@ -703,8 +714,8 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                 uint32_t packed =
                     winners[t * (tile_rows + PMH_MAX_RUNS) +
                             (i - tile_offsets[t])];
-                run[x] = packed >> 28;
-                row[x] = packed & 0x0fffffff;
+                run[x] = packed >> PMH_ROW_BITS;
+                row[x] = packed & PMH_ROW_MASK;
             } else {  // dead lane: duplicate x=0's gather, stores are guarded
                 run[x] = run[0];
                 row[x] = row[0];
@@ -892,12 +903,12 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
 #pragma unroll
             for (int x = 0; x < 4; x++)
                 vm[x] = x < gn
-                            ? run_masks[mc[x] >> 28][mc[x] & 0x0fffffff]
+                            ? run_masks[mc[x] >> PMH_ROW_BITS][mc[x] & PMH_ROW_MASK]
                             : 0;
         }
         uint32_t last = mc[0];
-        int lrun = last >> 28;
-        int64_t lrow = last & 0x0fffffff;
+        int lrun = last >> PMH_ROW_BITS;
+        int64_t lrow = last & PMH_ROW_MASK;
         // newest-first index of the last DELETE member; gn = none (also the
         // value when RROD is off, which neutralizes every bound below)
         int j_del = gn;
@@ -905,19 +916,26 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
 #pragma unroll
             for (int x = 0; x < 4; x++) {
                 if (x >= gn || j_del < gn) continue;
-                const DevCol &dc = cols[(mc[x] >> 28) * n_cols + kind_col];
-                if (col_load<int32_t>(dc, mc[x] & 0x0fffffff) == 3) j_del = x;
+                const DevCol &dc = cols[(mc[x] >> PMH_ROW_BITS) * n_cols + kind_col];
+                if (col_load<int32_t>(dc, mc[x] & PMH_ROW_MASK) == 3) j_del = x;
             }
             for (int x = 4; x < gn && j_del == gn; x++) {
                 uint32_t m = mem[me - 1 - x];
-                const DevCol &dc = cols[(m >> 28) * n_cols + kind_col];
-                if (col_load<int32_t>(dc, m & 0x0fffffff) == 3) j_del = x;
+                const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + kind_col];
+                if (col_load<int32_t>(dc, m & PMH_ROW_MASK) == 3) j_del = x;
             }
         }
         for (int c = 0; c < n_cols; c++) {
             if (c == kind_col) {
-                ((int8_t *)out_ptrs[c])[i] =
-                    (rrod && j_del == 0) ? 3 : 0;  // DELETE : INSERT
+                int32_t kout;
+                if (gn == 1) {  // wrapper bypass: record served as-is,
+                                // including its own RowKind
+                    const DevCol &dc = cols[lrun * n_cols + kind_col];
+                    kout = col_load<int32_t>(dc, lrow);
+                } else {
+                    kout = (rrod && j_del == 0) ? 3 : 0;  // DELETE : INSERT
+                }
+                ((int8_t *)out_ptrs[c])[i] = (int8_t)kout;
                 continue;
             }
             int64_t run = lrun, row = lrow;
@@ -932,14 +950,14 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                     if (MASKS) {
                         v = (uint8_t)((vm[x] >> c) & 1);
                     } else {
-                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                        const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + c];
                         v = dc.valid0
-                                ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                                ? ((const uint8_t *)dc.valid0)[m & PMH_ROW_MASK]
                                 : 1;
                     }
                     if (v) {
-                        run = m >> 28;
-                        row = m & 0x0fffffff;
+                        run = m >> PMH_ROW_BITS;
+                        row = m & PMH_ROW_MASK;
                         ok = 1;
                     }
                 }
@@ -949,16 +967,16 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                     uint8_t v;
                     if (MASKS) {
                         v = (uint8_t)(
-                            (run_masks[m >> 28][m & 0x0fffffff] >> c) & 1);
+                            (run_masks[m >> PMH_ROW_BITS][m & PMH_ROW_MASK] >> c) & 1);
                     } else {
-                        const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                        const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + c];
                         v = dc.valid0
-                                ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                                ? ((const uint8_t *)dc.valid0)[m & PMH_ROW_MASK]
                                 : 1;
                     }
                     if (v) {
-                        run = m >> 28;
-                        row = m & 0x0fffffff;
+                        run = m >> PMH_ROW_BITS;
+                        row = m & PMH_ROW_MASK;
                         ok = 1;
                     }
                 }
@@ -978,16 +996,16 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                 uint8_t v;
                 if (MASKS) {
                     v = (uint8_t)(
-                        (run_masks[m >> 28][m & 0x0fffffff] >> c) & 1);
+                        (run_masks[m >> PMH_ROW_BITS][m & PMH_ROW_MASK] >> c) & 1);
                 } else {
-                    const DevCol &dc = cols[(m >> 28) * n_cols + c];
+                    const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + c];
                     v = dc.valid0
-                            ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff]
+                            ? ((const uint8_t *)dc.valid0)[m & PMH_ROW_MASK]
                             : 1;
                 }
                 if (v) {
-                    run = m >> 28;
-                    row = m & 0x0fffffff;
+                    run = m >> PMH_ROW_BITS;
+                    row = m & PMH_ROW_MASK;
                     ok = 1;
                 }
             }
@@ -1062,9 +1080,9 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
     const bool rrod = (flags & 16) != 0;
     auto valid_of = [&](uint32_t m, int c) -> uint8_t {
         if (MASKS)
-            return (uint8_t)((run_masks[m >> 28][m & 0x0fffffff] >> c) & 1);
-        const DevCol &dc = cols[(m >> 28) * n_cols + c];
-        return dc.valid0 ? ((const uint8_t *)dc.valid0)[m & 0x0fffffff] : 1;
+            return (uint8_t)((run_masks[m >> PMH_ROW_BITS][m & PMH_ROW_MASK] >> c) & 1);
+        const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + c];
+        return dc.valid0 ? ((const uint8_t *)dc.valid0)[m & PMH_ROW_MASK] : 1;
     };
     const int64_t total = *total_out;
     const int64_t per_block =
@@ -1103,7 +1121,7 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
 #pragma unroll
             for (int x = 0; x < 4; x++)
                 vma[x] = x < gn
-                             ? run_masks[ma[x] >> 28][ma[x] & 0x0fffffff]
+                             ? run_masks[ma[x] >> PMH_ROW_BITS][ma[x] & PMH_ROW_MASK]
                              : 0;
         }
         uint32_t last = ma[0];
@@ -1116,10 +1134,10 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
             }
         if (gn > 4) {
             last = mem[me - 1];
-            if (MASKS) vlast = run_masks[last >> 28][last & 0x0fffffff];
+            if (MASKS) vlast = run_masks[last >> PMH_ROW_BITS][last & PMH_ROW_MASK];
         }
-        const int lrun = last >> 28;
-        const int64_t lrow = last & 0x0fffffff;
+        const int lrun = last >> PMH_ROW_BITS;
+        const int64_t lrow = last & PMH_ROW_MASK;
         // ascending index of the LAST DELETE member (-1 = none)
         int d_del = -1;
         uint32_t mdel = 0;
@@ -1128,23 +1146,30 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
 #pragma unroll
             for (int x = 0; x < 4; x++) {
                 if (x >= gn) continue;
-                const DevCol &dc = cols[(ma[x] >> 28) * n_cols + kind_col];
-                if (col_load<int32_t>(dc, ma[x] & 0x0fffffff) == 3) d_del = x;
+                const DevCol &dc = cols[(ma[x] >> PMH_ROW_BITS) * n_cols + kind_col];
+                if (col_load<int32_t>(dc, ma[x] & PMH_ROW_MASK) == 3) d_del = x;
             }
             for (int x = 4; x < gn; x++) {
                 uint32_t m = mem[ms + x];
-                const DevCol &dc = cols[(m >> 28) * n_cols + kind_col];
-                if (col_load<int32_t>(dc, m & 0x0fffffff) == 3) d_del = x;
+                const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + kind_col];
+                if (col_load<int32_t>(dc, m & PMH_ROW_MASK) == 3) d_del = x;
             }
             if (d_del >= 0) {
                 mdel = mem[ms + d_del];
-                if (MASKS) vdel = run_masks[mdel >> 28][mdel & 0x0fffffff];
+                if (MASKS) vdel = run_masks[mdel >> PMH_ROW_BITS][mdel & PMH_ROW_MASK];
             }
         }
         for (int c = 0; c < n_cols; c++) {
             if (c == kind_col) {
-                ((int8_t *)out_ptrs[c])[i] =
-                    (rrod && d_del == gn - 1) ? 3 : 0;  // DELETE : INSERT
+                int32_t kout;
+                if (gn == 1) {  // wrapper bypass: record served as-is,
+                                // including its own RowKind
+                    const DevCol &dc = cols[lrun * n_cols + kind_col];
+                    kout = col_load<int32_t>(dc, lrow);
+                } else {
+                    kout = (rrod && d_del == gn - 1) ? 3 : 0;
+                }
+                ((int8_t *)out_ptrs[c])[i] = (int8_t)kout;
                 continue;
             }
             const int dt = col_dtype[c];
@@ -1152,9 +1177,9 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
             auto del_valid = [&]() -> uint8_t {
                 if (!col_nullable[c]) return 1;
                 if (MASKS) return (uint8_t)((vdel >> c) & 1);
-                const DevCol &dc = cols[(mdel >> 28) * n_cols + c];
+                const DevCol &dc = cols[(mdel >> PMH_ROW_BITS) * n_cols + c];
                 return dc.valid0
-                           ? ((const uint8_t *)dc.valid0)[mdel & 0x0fffffff]
+                           ? ((const uint8_t *)dc.valid0)[mdel & PMH_ROW_MASK]
                            : 1;
             };
             int64_t run = lrun, row = lrow;
@@ -1170,8 +1195,8 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                                : valid_of(last, c);
                 break;
             case PMH_AGG_FIRST_VALUE: {
-                run = ma[0] >> 28;
-                row = ma[0] & 0x0fffffff;
+                run = ma[0] >> PMH_ROW_BITS;
+                row = ma[0] & PMH_ROW_MASK;
                 if (col_nullable[c])
                     ok = MASKS ? (uint8_t)((vma[0] >> c) & 1)
                                : valid_of(ma[0], c);
@@ -1186,8 +1211,8 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                         if (x <= d_del) break;
                         uint32_t m = mem[ms + x];
                         if (valid_of(m, c)) {
-                            run = m >> 28;
-                            row = m & 0x0fffffff;
+                            run = m >> PMH_ROW_BITS;
+                            row = m & PMH_ROW_MASK;
                             ok = 1;
                         }
                     }
@@ -1197,14 +1222,14 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                         uint8_t v = MASKS ? (uint8_t)((vma[x] >> c) & 1)
                                           : valid_of(ma[x], c);
                         if (v) {
-                            run = ma[x] >> 28;
-                            row = ma[x] & 0x0fffffff;
+                            run = ma[x] >> PMH_ROW_BITS;
+                            row = ma[x] & PMH_ROW_MASK;
                             ok = 1;
                         }
                     }
                     if (!ok && d_del >= 0 && del_valid()) {
-                        run = mdel >> 28;  // initRow: the DELETE's field
-                        row = mdel & 0x0fffffff;
+                        run = mdel >> PMH_ROW_BITS;  // initRow: the DELETE's field
+                        row = mdel & PMH_ROW_MASK;
                         ok = 1;
                     }
                 }
@@ -1218,22 +1243,22 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                         uint8_t v = MASKS ? (uint8_t)((vma[x] >> c) & 1)
                                           : valid_of(ma[x], c);
                         if (v) {
-                            run = ma[x] >> 28;
-                            row = ma[x] & 0x0fffffff;
+                            run = ma[x] >> PMH_ROW_BITS;
+                            row = ma[x] & PMH_ROW_MASK;
                             ok = 1;
                         }
                     }
                     for (int32_t x = 4; !ok && x < gn; x++) {
                         uint32_t m = mem[ms + x];
                         if (valid_of(m, c)) {
-                            run = m >> 28;
-                            row = m & 0x0fffffff;
+                            run = m >> PMH_ROW_BITS;
+                            row = m & PMH_ROW_MASK;
                             ok = 1;
                         }
                     }
                 } else {
-                    run = ma[0] >> 28;
-                    row = ma[0] & 0x0fffffff;
+                    run = ma[0] >> PMH_ROW_BITS;
+                    row = ma[0] & PMH_ROW_MASK;
                 }
                 break;
             default: {  // SUM / MAX / MIN: full fold, null inputs skipped
@@ -1243,8 +1268,8 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 float facc = 0.f;
                 double dacc = 0.0;
                 auto fold_one = [&](uint32_t m) {
-                    const DevCol &dc = cols[(m >> 28) * n_cols + c];
-                    const int64_t r = m & 0x0fffffff;
+                    const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + c];
+                    const int64_t r = m & PMH_ROW_MASK;
                     int64_t vb = (dt == 4 || dt == 6)
                                      ? col_load<int64_t>(dc, r)
                                      : (int64_t)col_load<int32_t>(dc, r);
@@ -1602,6 +1627,10 @@ hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
                            total_rows, cuts);
     else if (k <= 8)
         hipLaunchKernelGGL(k_partition<8>, dim3(blocks), dim3(threads), 0,
+                           stream, keys, lens, k, tile_rows, n_bounds,
+                           total_rows, cuts);
+    else if (k <= 16)
+        hipLaunchKernelGGL(k_partition<16>, dim3(blocks), dim3(threads), 0,
                            stream, keys, lens, k, tile_rows, n_bounds,
                            total_rows, cuts);
     else

@@ -919,6 +919,15 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
     run.cols.resize(cols.size());
     int64_t total_rows = 0;
     for (const auto &fd : files) total_rows += fd.row_count;
+    if (total_rows >= ((int64_t)1 << PMH_ROW_BITS)) {
+        // a run is the CONCATENATION of non-overlapping files; the winner
+        // packing (run | row) indexes into the whole run, so the cap applies
+        // to the run total, not per file (a >=2^row-bits run would silently
+        // overflow into the run bits and merge wrong)
+        set_error("run totals %lld rows >= 2^%d per-run limit (%zu files)",
+                  (long long)total_rows, PMH_ROW_BITS, files.size());
+        return false;
+    }
     for (size_t c = 0; c < cols.size(); c++) {
         RunCol &rc = run.cols[c];
         rc.contig = plan->bufs.alloc(total_rows * cols[c].stored_esize);
@@ -1543,9 +1552,9 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             fd.max_key = fj["maxKey"].as_i64();
             fd.level = (int)fj["level"].as_i64();
             fd.input_index = idx++;
-            if (fd.row_count >= (1 << 28)) {
-                set_error("file %s exceeds 2^28 rows per run limit",
-                          fd.path.c_str());
+            if (fd.row_count >= ((int64_t)1 << PMH_ROW_BITS)) {
+                set_error("file %s exceeds 2^%d rows-per-run limit",
+                          fd.path.c_str(), PMH_ROW_BITS);
                 return nullptr;
             }
             files.push_back(fd);

@@ -128,6 +128,36 @@ class TestAggregationEngine:
         self._run(tmp_path, runs, aggs, compression="zstd",
                   row_group_rows=16_384, data_page_rows=4_096)
 
+    def test_singleton_retract_bypass(self, tmp_path):
+        # singleton groups bypass the merge function (ReducerMergeFunction
+        # Wrapper.java:53-73): lone retracts are legal — dropped under
+        # drop-delete, served as-is (own RowKind) under keep-delete
+        for drop_delete in (True, False):
+            rng = np.random.default_rng(94)
+            runs = []
+            n = 6_000
+            for i in range(2):
+                key = np.arange(n, dtype=np.int64) * 2 + i  # disjoint
+                kind = np.where(rng.random(n) < 0.25, 3, 0).astype(np.int8)
+                vals = rng.integers(-1000, 1000, n).astype(np.int32)
+                msk = rng.random(n) > 0.3
+                runs.append({"key": key,
+                             "seq": np.arange(i * n, (i + 1) * n, np.int64),
+                             "kind": kind, "values": [key.copy(), vals],
+                             "valid": [np.ones(n, bool), msk]})
+            metas = write_runs(runs, str(tmp_path / str(drop_delete)),
+                               compression="NONE")
+            exp = aggregation_model(runs, ["last_non_null_value", "sum"],
+                                    drop_delete=drop_delete)
+            with Session(0) as s:
+                with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                                   _value_cols(1),
+                                   merge_engine="aggregation",
+                                   aggregations={"v_c0": "sum"},
+                                   drop_delete=drop_delete) as plan:
+                    got = _read_all(plan)
+            _check(got, exp, ["v_k", "v_c0"])
+
     def test_retract_rejected(self, tmp_path):
         runs = gen_runs_partial_update(2, 5_000, n_value_cols=2, seed=98)
         # turn some rows of run 1 into DELETEs

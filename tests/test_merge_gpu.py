@@ -92,6 +92,17 @@ class TestDedupParity:
         runs = gen_runs_dedup(16, 8_000, n_value_cols=2, seed=47)
         _run_and_compare(tmp_path, runs)
 
+    def test_twentyfour_runs(self, tmp_path):
+        # beyond the reference's default spill threshold shapes: the widened
+        # run:5|row:27 winner packing takes sections up to 32 runs
+        runs = gen_runs_dedup(24, 5_000, n_value_cols=2, seed=147,
+                              delete_frac=0.1)
+        _run_and_compare(tmp_path, runs)
+
+    def test_thirtytwo_runs(self, tmp_path):
+        runs = gen_runs_dedup(32, 3_000, n_value_cols=2, seed=148)
+        _run_and_compare(tmp_path, runs)
+
     def test_heavy_collision_tiny_keyspace(self, tmp_path):
         # many groups span tile boundaries relative to key space
         rng = np.random.default_rng(48)
@@ -416,11 +427,29 @@ class TestErrorPaths:
     # descriptive error so a Java-side provider can fall
     # back to the stock reader per split
     def test_too_many_overlapping_runs(self, tmp_path):
-        runs = gen_runs_dedup(17, 500, n_value_cols=1, seed=70,
+        runs = gen_runs_dedup(33, 500, n_value_cols=1, seed=70,
                               delete_frac=0.0)
         metas = write_runs(runs, str(tmp_path), compression="NONE")
         with Session(0) as s:
-            with pytest.raises(RuntimeError, match="16"):
+            with pytest.raises(RuntimeError, match="32"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(1))
+
+    def test_run_row_overflow_rejected(self, tmp_path):
+        # a run is the concatenation of its files: two non-overlapping files
+        # whose TOTAL claims >= 2^27 rows must be rejected before the packed
+        # (run | row) winner format could overflow (files are never opened —
+        # the guard fires on the declared row counts)
+        metas = [{"path": str(tmp_path / "a.parquet"), "rowCount": 70_000_000,
+                  "minKey": 0, "maxKey": 10, "level": 0},
+                 {"path": str(tmp_path / "b.parquet"), "rowCount": 70_000_000,
+                  "minKey": 20, "maxKey": 30, "level": 0},
+                 # bridge file: overlaps both, forcing one section where a+b
+                 # concatenate into a single run
+                 {"path": str(tmp_path / "c.parquet"), "rowCount": 1_000,
+                  "minKey": 5, "maxKey": 25, "level": 0}]
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="per-run limit"):
                 MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
                               _value_cols(1))
 

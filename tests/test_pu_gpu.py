@@ -67,6 +67,41 @@ class TestPartialUpdate:
                                        update_frac=0.0)
         self._run(tmp_path, runs)
 
+    def test_singleton_retract_bypass(self, tmp_path):
+        # ReducerMergeFunctionWrapper.java:53-73: singleton groups bypass the
+        # merge function entirely, so a lone retract is legal in PU mode —
+        # dropped under drop-delete, served with its own RowKind otherwise
+        for drop_delete in (True, False):
+            rng = np.random.default_rng(93)
+            runs = []
+            n = 8_000
+            for i in range(3):
+                # keys disjoint across runs: every group is a singleton
+                key = (np.arange(n, dtype=np.int64) * 3 + i)
+                kind = np.where(rng.random(n) < 0.2, 3, 0).astype(np.int8)
+                kind[rng.random(n) < 0.05] = 1  # lone UPDATE_BEFOREs too
+                vals = rng.integers(-1000, 1000, n).astype(np.int32)
+                msk = rng.random(n) > 0.3
+                runs.append({"key": key,
+                             "seq": np.arange(i * n, (i + 1) * n, np.int64),
+                             "kind": kind, "values": [key.copy(), vals],
+                             "valid": [np.ones(n, bool), msk]})
+            metas = write_runs(runs, str(tmp_path / str(drop_delete)),
+                               compression="NONE")
+            exp = partial_update_model(runs, drop_delete=drop_delete)
+            with Session(0) as s:
+                with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                                   _value_cols(1),
+                                   merge_engine="partial-update",
+                                   drop_delete=drop_delete) as plan:
+                    got = _read_all(plan)
+            assert (got["_KEY_k"] == exp["key"]).all()
+            assert (got["_VALUE_KIND"] == exp["kind"]).all()
+            em = exp["valid"][1]
+            gm = got["v_c0#valid"]
+            assert (gm == em).all()
+            assert (got["v_c0"][em] == exp["values"][1][em]).all()
+
     def test_pu_rejects_retracts(self, tmp_path):
         runs = gen_runs_dedup(2, 5_000, n_value_cols=2, seed=74,
                               delete_frac=0.2)
