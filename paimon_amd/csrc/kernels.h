@@ -108,6 +108,24 @@ hipError_t pmh_launch_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
                                  int64_t *tile_offsets, int64_t *total_out,
                                  hipStream_t stream);
 
+// FUSED merge + emit (deduplicate / first-row): ticket-ordered persistent
+// workgroups, decoupled-lookback output offsets (packed agent-scope atomic
+// per tile in `status`, zeroed before launch along with `ticket`), emission
+// from LDS-staged columns. Replaces merge_tiles + scan_tiles + emit for
+// non-member-list engines. key_col = -1 for composite keys (key columns
+// then emit through the generic column path).
+hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
+                                 const DevCol *kinds, const int64_t *lens,
+                                 int k, const int32_t *cuts, int64_t n_tiles,
+                                 int64_t tile_rows, int flags,
+                                 const DevCol *cols, const uint8_t *col_dtype,
+                                 const uint8_t *col_nullable, int n_cols,
+                                 int key_col, int seq_col, int kind_col,
+                                 uint64_t *status, uint64_t *ticket,
+                                 int64_t *total_out, void *const *out_ptrs,
+                                 uint8_t *const *out_valid,
+                                 uint32_t *err_flag, hipStream_t stream);
+
 hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
                            const uint8_t *col_nullable, int n_cols, int k,
                            const uint32_t *winners,

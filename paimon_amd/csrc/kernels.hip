@@ -625,6 +625,452 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
     }
 }
 
+// ------------------------------------------------------------ k_merge_emit
+//
+// FUSED single-pass merge + emit for the deduplicate / first-row engines:
+// the partition cuts in, the merged output columns out — one kernel.
+// Replaces the k_merge_tiles -> k_scan_tiles -> k_emit chain (and its
+// winners round-trip through HBM) for non-member-list engines:
+//
+//  - tiles are taken from a global TICKET (atomic counter), so tile t-1 is
+//    always owned by a workgroup that started no later than tile t's — the
+//    forward-progress precondition for a decoupled-lookback prefix without
+//    any dispatch-order assumption (HIP promises none, MI355X_MICROARCH.md
+//    §Workgroup dispatch);
+//  - each tile stages key/seq/kind in LDS, merges, walks group winners
+//    (exactly k_merge_tiles' order contract), then publishes its survivor
+//    count through a PACKED agent-scope atomic word (flag:2 | value:62) and
+//    resolves its global output offset by wave-parallel lookback — the
+//    payload rides in the atomic itself, so no separate fence choreography
+//    is needed (per-XCD L2s are not coherent; agent-scope rel/acq is);
+//  - emission happens from LDS: key/seq/kind come from the already-staged
+//    merge arrays (full RowKind is recoverable from the packed seq word),
+//    and each remaining output column is staged COALESCED per run into a
+//    double-buffered LDS slab that reuses the merge arrays' space, then
+//    scattered to the dense output by winner index — LDS gathers + coalesced
+//    stores instead of the 4-8 B HBM gathers that left k_emit 96% latency-
+//    parked (profiles/r01_final_pmc_sq.md).
+//
+// Packed seq word: (sequenceNumber << 2) | (isAdd << 1) | (kind >> 1).
+// Ascending order == ascending (seq, isAdd) — the merge-order tie-break of
+// SortMergeReaderWithLoserTree.java:53-63 (the kind>>1 bit only breaks ties
+// between records with equal (seq, isAdd), which valid buckets never have —
+// sequence numbers are unique). RowKind = ((w & 1) << 1) | (1 - ((w >> 1) & 1)).
+// Requires |seq| < 2^61 (Paimon sequence numbers are non-negative counters).
+
+DEV int64_t ps2_pack(int64_t seq, int32_t kd) {
+    return (seq << 2) | ((int64_t)(kd == 0 || kd == 2) << 1) |
+           (int64_t)((kd >> 1) & 1);
+}
+DEV bool ps2_isadd(int64_t w) { return (w >> 1) & 1; }
+DEV int32_t ps2_kind(int64_t w) {
+    return (int32_t)(((w & 1) << 1) | (1 - ((w >> 1) & 1)));
+}
+
+struct FusedSmem {
+    union {
+        struct {
+            int64_t skey[PMH_TILE_MAX];
+            int64_t sseq[PMH_TILE_MAX];
+        };
+        // emission phase (skey/sseq dead after the key/seq/kind emit):
+        // two column-staging slabs, PMH_TILE_MAX elements of up to 8 B
+        uint8_t vbuf[2][PMH_TILE_MAX * 8];
+    };
+    uint16_t perm[2][PMH_TILE_MAX];
+    uint8_t head[PMH_TILE_MAX];
+    int32_t wave_tot[PMH_TILE_THREADS / 64];
+    int32_t segoff[PMH_MAX_RUNS + 1];
+    int32_t seglen[PMH_MAX_RUNS];
+    int64_t predcand[PMH_MAX_RUNS];
+    int64_t predkey;
+    int64_t s_tile;
+    int64_t s_goff;
+    int32_t haspred;
+    int32_t mtotal;
+    int32_t mreal;
+};
+
+constexpr uint64_t LOOK_AGG = 1ull << 62;
+constexpr uint64_t LOOK_PREFIX = 2ull << 62;
+constexpr uint64_t LOOK_VAL = (1ull << 62) - 1;
+
+template <bool FR>
+__launch_bounds__(PMH_TILE_THREADS) __global__
+void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
+                  const int64_t *lens, int k, const int32_t *cuts,
+                  int64_t n_tiles, int64_t tile_rows, int flags,
+                  const DevCol *cols /* k * n_cols, run-major */,
+                  const uint8_t *col_dtype, const uint8_t *col_nullable,
+                  int n_cols, int key_col /* -1: composite */, int seq_col,
+                  int kind_col, uint64_t *status, uint64_t *ticket,
+                  int64_t *total_out, void *const *out_ptrs,
+                  uint8_t *const *out_valid, uint32_t *err_flag) {
+    const bool drop_delete = flags & 1;
+    const bool ignore_delete = flags & 2;
+    __shared__ FusedSmem sm;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wv = tid >> 6;
+    constexpr int NW = PMH_TILE_THREADS / 64;
+    for (;;) {
+        if (tid == 0)
+            sm.s_tile = (int64_t)__hip_atomic_fetch_add(
+                ticket, 1ull, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        __syncthreads();
+        const int64_t tile = sm.s_tile;
+        if (tile >= n_tiles) return;
+        const int32_t *c0 = &cuts[tile * k];
+        const int32_t *c1 = &cuts[(tile + 1) * k];
+
+        // --- segment setup (k_merge_tiles' protocol: +1 extra per run)
+        if (tid < k) {
+            int32_t a = c0[tid], b = c1[tid];
+            int32_t ext = (b < (int32_t)lens[tid]) ? 1 : 0;
+            sm.seglen[tid] = (b - a) + ext;
+            sm.predcand[tid] =
+                a > 0 ? key_at(keys[tid].addr0, a - 1, keys[tid].esize)
+                      : INT64_MIN;
+            sm.head[tid] = a > 0;
+        }
+        __syncthreads();
+        if (tid == 0) {
+            int32_t off = 0, real = 0;
+            int64_t pred = 0;
+            int hp = 0;
+            for (int r = 0; r < k; r++) {
+                sm.segoff[r] = off;
+                real += c1[r] - c0[r];
+                off += sm.seglen[r];
+                if (sm.head[r]) {
+                    if (!hp || sm.predcand[r] > pred) pred = sm.predcand[r];
+                    hp = 1;
+                }
+            }
+            sm.segoff[k] = off;
+            sm.mtotal = off;
+            sm.mreal = real;
+            sm.predkey = pred;
+            sm.haspred = hp;
+        }
+        __syncthreads();
+        const int32_t M = sm.mtotal;
+        const int32_t Mreal = sm.mreal;
+        int32_t C = 0;       // this tile's survivor count
+        int cur = 0;
+        if (Mreal > 0) {
+            // --- stage key / packed-seq segments (coalesced per run)
+            for (int r = 0; r < k; r++) {
+                int32_t off = sm.segoff[r], len = sm.seglen[r];
+                int64_t base = c0[r];
+                const int kes = keys[r].esize;
+                const uint64_t kaddr0 = keys[r].addr0 + (uint64_t)base * kes;
+                const int64_t *saddr =
+                    reinterpret_cast<const int64_t *>(seqs[r].addr0) + base;
+                const int32_t *daddr =
+                    reinterpret_cast<const int32_t *>(kinds[r].addr0) + base;
+                for (int32_t i = tid; i < len; i += blockDim.x) {
+                    sm.skey[off + i] = key_at(kaddr0, i, kes);
+                    sm.sseq[off + i] = ps2_pack(saddr[i], daddr[i]);
+                    sm.perm[0][off + i] = (uint16_t)(off + i);
+                }
+            }
+            __syncthreads();
+
+            // --- pairwise stable merge (identical to k_merge_tiles)
+            for (int width = 1; width < k; width <<= 1) {
+                const int nxt = cur ^ 1;
+                const int CH = 8;
+                int n_chunks = (M + CH - 1) / CH;
+                for (int ch = tid; ch < n_chunks; ch += blockDim.x) {
+                    int64_t o = (int64_t)ch * CH;
+                    int remaining = (int)(M - o < CH ? M - o : CH);
+                    int p = 0;
+                    while (remaining > 0) {
+                        while ((p + 1) * 2 * width < k &&
+                               sm.segoff[(p + 1) * 2 * width] <= o)
+                            p++;
+                        int a0 = p * 2 * width;
+                        int amid = a0 + width < k ? a0 + width : k;
+                        int b1 = a0 + 2 * width < k ? a0 + 2 * width : k;
+                        int32_t abase = sm.segoff[a0];
+                        int32_t la = sm.segoff[amid] - abase;
+                        int32_t lb = sm.segoff[b1] - sm.segoff[amid];
+                        int64_t d = o - abase;
+                        int32_t lim = la + lb - (int32_t)d;
+                        if (lim <= 0) break;
+                        int n_out = lim < remaining ? lim : remaining;
+                        const uint16_t *pa = &sm.perm[cur][abase];
+                        const uint16_t *pb = &sm.perm[cur][abase + la];
+                        int32_t ai = corank(d, pa, la, pb, lb, sm.skey);
+                        int32_t bi = (int32_t)d - ai;
+                        uint16_t *out = &sm.perm[nxt][abase + d];
+                        for (int x = 0; x < n_out; x++) {
+                            bool takeA;
+                            if (ai >= la) takeA = false;
+                            else if (bi >= lb) takeA = true;
+                            else takeA = !(sm.skey[pa[ai]] > sm.skey[pb[bi]]);
+                            out[x] = takeA ? pa[ai++] : pb[bi++];
+                        }
+                        o += n_out;
+                        remaining -= n_out;
+                    }
+                }
+                cur = nxt;
+                __syncthreads();
+            }
+            const uint16_t *mo = sm.perm[cur];
+
+            // --- group heads + previous-tile continuation skip
+            for (int32_t i = tid; i < M; i += blockDim.x) {
+                int64_t kk = sm.skey[mo[i]];
+                uint8_t h = (i == 0) ? 1 : (kk != sm.skey[mo[i - 1]]);
+                if (sm.haspred && kk == sm.predkey) h = 0;
+                sm.head[i] = h;
+            }
+            __syncthreads();
+
+            // --- winner walk (k_merge_tiles' dedup/first-row rules), but
+            // winners are SEG INDICES into the LDS arrays, compacted into
+            // the non-current perm plane in key order
+            uint16_t *wl = sm.perm[cur ^ 1];
+            const int32_t per =
+                (Mreal + (int32_t)blockDim.x - 1) / blockDim.x;
+            int32_t my_lo = tid * per;
+            int32_t my_hi = my_lo + per < Mreal ? my_lo + per : Mreal;
+            int32_t my_off = 0;
+            for (int pass = 0; pass < 2; pass++) {
+                int32_t nloc = 0;
+                for (int32_t i = my_lo; i < my_hi; i++) {
+                    if (!sm.head[i]) continue;
+                    int32_t tail = i;
+                    uint16_t s_best = mo[i];
+                    int64_t v_best = sm.sseq[s_best];
+                    bool e_best = !ignore_delete || ps2_isadd(v_best);
+                    bool any_retract = !ps2_isadd(v_best);
+                    while (tail + 1 < M && !sm.head[tail + 1]) {
+                        tail++;
+                        uint16_t s = mo[tail];
+                        int64_t v = sm.sseq[s];
+                        bool e = !ignore_delete || ps2_isadd(v);
+                        any_retract |= !ps2_isadd(v);
+                        bool take = (e && !e_best) ||
+                                    (e == e_best &&
+                                     (FR ? v < v_best : v > v_best));
+                        if (take) {
+                            s_best = s;
+                            v_best = v;
+                            e_best = e;
+                        }
+                    }
+                    int32_t gsize = tail - i + 1;
+                    if (FR && !ignore_delete && any_retract && gsize > 1 &&
+                        err_flag)
+                        atomicOr(err_flag, 2u);
+                    if (!e_best && gsize > 1) continue;
+                    if (drop_delete && !ps2_isadd(v_best)) continue;
+                    if (pass == 1) wl[my_off + nloc] = s_best;
+                    nloc++;
+                }
+                if (pass == 0) {
+                    int32_t incl = nloc;
+                    for (int off = 1; off < 64; off <<= 1) {
+                        int32_t up = __shfl_up(incl, off, 64);
+                        if (lane >= off) incl += up;
+                    }
+                    if (lane == 63) sm.wave_tot[wv] = incl;
+                    __syncthreads();
+                    int32_t add = 0;
+                    C = 0;
+#pragma unroll
+                    for (int w = 0; w < NW; w++) {
+                        if (w < wv) add += sm.wave_tot[w];
+                        C += sm.wave_tot[w];
+                    }
+                    my_off = add + incl - nloc;
+                }
+            }
+        }
+
+        // --- publish count + decoupled lookback for the global offset.
+        // The count rides in the packed atomic word, so a successor that
+        // acquires the flag also gets the payload — no separate fence.
+        if (tid == 0 && tile > 0)
+            __hip_atomic_store(&status[tile], LOOK_AGG | (uint64_t)C,
+                               __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+        if (wv == 0) {
+            int64_t excl = 0;
+            if (tile > 0) {
+                int64_t look = tile - 1;
+                bool done = false;
+                while (!done) {
+                    int64_t idx = look - lane;
+                    uint64_t st = LOOK_PREFIX;  // virtual prefix 0 before t0
+                    if (idx >= 0) {
+                        int64_t spins = 0;
+                        do {
+                            st = __hip_atomic_load(&status[idx],
+                                                   __ATOMIC_ACQUIRE,
+                                                   __HIP_MEMORY_SCOPE_AGENT);
+                        } while ((st >> 62) == 0 && ++spins < (1ll << 27));
+                        if ((st >> 62) == 0) {  // bounded spin: fail loudly
+                            if (err_flag) atomicOr(err_flag, 4u);
+                            st = LOOK_PREFIX;
+                        }
+                    }
+                    bool isp = (st >> 62) == 2;
+                    uint64_t pm = __ballot(isp);
+                    int first = pm ? (int)(__ffsll((unsigned long long)pm) - 1)
+                                   : 64;
+                    int64_t contrib =
+                        (lane <= first) ? (int64_t)(st & LOOK_VAL) : 0;
+                    for (int off = 32; off; off >>= 1)
+                        contrib += __shfl_down(contrib, off, 64);
+                    excl += __shfl(contrib, 0, 64);
+                    if (first < 64) done = true;
+                    else look -= 64;
+                }
+            }
+            if (lane == 0) {
+                __hip_atomic_store(&status[tile],
+                                   LOOK_PREFIX | (uint64_t)(excl + C),
+                                   __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_AGENT);
+                sm.s_goff = excl;
+                if (tile == n_tiles - 1) *total_out = excl + C;
+            }
+        }
+        __syncthreads();
+        if (C == 0) continue;  // nothing to emit (all threads agree: C is
+                               // uniform after the pass-0 block reduction)
+        const int64_t goff = sm.s_goff;
+        const uint16_t *wl = sm.perm[cur ^ 1];
+
+        // --- emit key / seq / kind straight from the merge arrays
+        if (key_col >= 0) {
+            const int kdt = col_dtype[key_col];
+            for (int32_t i = tid; i < C; i += blockDim.x) {
+                int64_t v = sm.skey[wl[i]];
+                switch (kdt) {
+                case 1: ((int8_t *)out_ptrs[key_col])[goff + i] =
+                            (int8_t)v; break;
+                case 2: ((int16_t *)out_ptrs[key_col])[goff + i] =
+                            (int16_t)v; break;
+                case 3: ((int32_t *)out_ptrs[key_col])[goff + i] =
+                            (int32_t)v; break;
+                default: ((int64_t *)out_ptrs[key_col])[goff + i] = v; break;
+                }
+            }
+        }
+        for (int32_t i = tid; i < C; i += blockDim.x) {
+            int64_t w = sm.sseq[wl[i]];
+            ((int64_t *)out_ptrs[seq_col])[goff + i] = w >> 2;
+            ((int8_t *)out_ptrs[kind_col])[goff + i] = (int8_t)ps2_kind(w);
+        }
+        __syncthreads();  // skey/sseq die; vbuf slabs take their space
+
+        // --- remaining columns: stage per-run segments coalesced into an
+        // LDS slab, scatter to the dense output by winner index.
+        // Double-buffered when no column is nullable (one barrier/column);
+        // nullable plans run single-buffered with the validity bytes in the
+        // second slab.
+        bool any_null = false;
+        for (int c = 0; c < n_cols; c++)
+            if (col_nullable[c]) any_null = true;
+
+        auto stage_col = [&](int c, uint8_t *slab) {
+            for (int r = 0; r < k; r++) {
+                int32_t off = sm.segoff[r], len = sm.seglen[r];
+                const DevCol &dc = cols[r * n_cols + c];
+                if (dc.esize == 8) {
+                    const int64_t *src =
+                        reinterpret_cast<const int64_t *>(dc.addr0) + c0[r];
+                    int64_t *dst = reinterpret_cast<int64_t *>(slab) + off;
+                    for (int32_t i = tid; i < len; i += blockDim.x)
+                        dst[i] = src[i];
+                } else {
+                    const int32_t *src =
+                        reinterpret_cast<const int32_t *>(dc.addr0) + c0[r];
+                    int32_t *dst = reinterpret_cast<int32_t *>(slab) + off;
+                    for (int32_t i = tid; i < len; i += blockDim.x)
+                        dst[i] = src[i];
+                }
+            }
+        };
+        auto stage_valid = [&](int c, uint8_t *slab) {
+            for (int r = 0; r < k; r++) {
+                int32_t off = sm.segoff[r], len = sm.seglen[r];
+                const DevCol &dc = cols[r * n_cols + c];
+                if (dc.valid0) {
+                    const uint8_t *src =
+                        reinterpret_cast<const uint8_t *>(dc.valid0) + c0[r];
+                    for (int32_t i = tid; i < len; i += blockDim.x)
+                        slab[off + i] = src[i];
+                } else {
+                    for (int32_t i = tid; i < len; i += blockDim.x)
+                        slab[off + i] = 1;
+                }
+            }
+        };
+        auto emit_col = [&](int c, const uint8_t *slab,
+                            const uint8_t *vslab) {
+            const int dt = col_dtype[c];
+            const bool wide = dt == 4 || dt == 6;
+            uint8_t *ov = (col_nullable[c] && out_valid[c]) ? out_valid[c]
+                                                            : nullptr;
+            for (int32_t i = tid; i < C; i += blockDim.x) {
+                uint16_t s = wl[i];
+                int64_t v = wide
+                    ? reinterpret_cast<const int64_t *>(slab)[s]
+                    : (int64_t)reinterpret_cast<const int32_t *>(slab)[s];
+                switch (dt) {
+                case 1: ((int8_t *)out_ptrs[c])[goff + i] = (int8_t)v; break;
+                case 2: ((int16_t *)out_ptrs[c])[goff + i] =
+                            (int16_t)v; break;
+                case 3:
+                case 5: ((int32_t *)out_ptrs[c])[goff + i] =
+                            (int32_t)v; break;
+                default: ((int64_t *)out_ptrs[c])[goff + i] = v; break;
+                }
+                if (ov) ov[goff + i] = vslab ? vslab[s] : 1;
+            }
+        };
+
+        if (any_null) {
+            // single-buffered: values in slab 0, validity bytes in slab 1
+            for (int c = 0; c < n_cols; c++) {
+                if (c == seq_col || c == kind_col || c == key_col) continue;
+                stage_col(c, sm.vbuf[0]);
+                const bool nul = col_nullable[c];
+                if (nul) stage_valid(c, sm.vbuf[1]);
+                __syncthreads();
+                emit_col(c, sm.vbuf[0], nul ? sm.vbuf[1] : nullptr);
+                __syncthreads();
+            }
+        } else {
+            int prev = -1, slot = 0;
+            for (int c = 0; c < n_cols; c++) {
+                if (c == seq_col || c == kind_col || c == key_col) continue;
+                if (prev < 0) {  // first column: fill slab 0
+                    stage_col(c, sm.vbuf[0]);
+                    prev = c;
+                    continue;
+                }
+                __syncthreads();  // slab `slot` ready; slab slot^1 drained
+                stage_col(c, sm.vbuf[slot ^ 1]);
+                emit_col(prev, sm.vbuf[slot], nullptr);
+                prev = c;
+                slot ^= 1;
+            }
+            if (prev >= 0) {
+                __syncthreads();
+                emit_col(prev, sm.vbuf[slot], nullptr);
+            }
+        }
+        __syncthreads();  // all slabs drained before the next tile
+    }
+}
+
 // ------------------------------------------------------------ k_scan_tiles
 // Exclusive scan of tile_counts (single workgroup, chunked).
 __global__ void k_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
@@ -1736,6 +2182,35 @@ hipError_t pmh_launch_level_scatter(const RleChunk *chunks, int64_t n_chunks,
     int blocks = n_chunks < 4096 ? (int)(n_chunks ? n_chunks : 1) : 4096;
     hipLaunchKernelGGL(k_level_scatter, dim3(blocks), dim3(PMH_TILE_THREADS),
                        0, stream, chunks, n_chunks);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
+                                 const DevCol *kinds, const int64_t *lens,
+                                 int k, const int32_t *cuts, int64_t n_tiles,
+                                 int64_t tile_rows, int flags,
+                                 const DevCol *cols, const uint8_t *col_dtype,
+                                 const uint8_t *col_nullable, int n_cols,
+                                 int key_col, int seq_col, int kind_col,
+                                 uint64_t *status, uint64_t *ticket,
+                                 int64_t *total_out, void *const *out_ptrs,
+                                 uint8_t *const *out_valid,
+                                 uint32_t *err_flag, hipStream_t stream) {
+    // persistent workgroups: 2 resident per CU (LDS-bound) x 256 CUs; the
+    // ticket hands out tiles in order, so any residency is safe
+    int blocks = n_tiles < 512 ? (int)n_tiles : 512;
+    const bool fr = flags & 8;
+    auto launch = [&](auto kern) {
+        hipLaunchKernelGGL(kern, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
+                           stream, keys, seqs, kinds, lens, k, cuts, n_tiles,
+                           tile_rows, flags, cols, col_dtype, col_nullable,
+                           n_cols, key_col, seq_col, kind_col, status, ticket,
+                           total_out, out_ptrs, out_valid, err_flag);
+    };
+    if (fr)
+        launch(k_merge_emit<true>);
+    else
+        launch(k_merge_emit<false>);
     return hipGetLastError();
 }
 

@@ -220,6 +220,10 @@ struct Section {
     int64_t *total_dev = nullptr;
     uint16_t *group_start = nullptr;  // partial-update member offsets
     uint32_t *err_dev = nullptr;
+    // fused merge+emit path (dedup / first-row): per-tile lookback words +
+    // the tile ticket, zeroed per read
+    uint64_t *status = nullptr;
+    uint64_t *ticket = nullptr;
     // packed per-row validity (PU/agg emit): one u64 per row per run,
     // bit c = column c non-null; built once per section by k_pack_valid
     std::vector<uint64_t *> row_masks;
@@ -251,6 +255,8 @@ struct pmh_plan_t {
     bool host_output = false;
     bool pu = false;         // partial-update merge engine
     bool first_row = false;  // first-row merge engine
+    bool fused = false;      // single-pass k_merge_emit (non-member-list
+                             // engines; PMH_FUSED=0 falls back for A/B)
     bool agg = false;        // aggregation merge engine (uses PU member lists)
     uint8_t *col_agg_dev = nullptr;  // per-column PMH_AGG_* codes
     // composite key (>1 key column): order-preserving packed comparand
@@ -1350,10 +1356,20 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
     sec.all_cols = (DevCol *)up(allv.data(), allv.size() * sizeof(DevCol));
     sec.lens_dev = (int64_t *)up(lens.data(), k * sizeof(int64_t));
     sec.cuts = (int32_t *)plan->bufs.alloc((sec.n_tiles + 1) * k * 4);
-    sec.winners = (uint32_t *)plan->bufs.alloc(
-        sec.n_tiles * (PMH_TILE_ROWS + PMH_MAX_RUNS) * 4);
-    sec.tile_counts = (int32_t *)plan->bufs.alloc(sec.n_tiles * 4);
-    sec.tile_offsets = (int64_t *)plan->bufs.alloc(sec.n_tiles * 8);
+    if (plan->fused) {
+        // single-pass path: no winners round-trip, no scan arrays — just
+        // the per-tile lookback words and the tile ticket
+        sec.status = (uint64_t *)plan->bufs.alloc(sec.n_tiles * 8);
+        sec.ticket = (uint64_t *)plan->bufs.alloc(8);
+        if (!sec.status || !sec.ticket) return false;
+    } else {
+        sec.winners = (uint32_t *)plan->bufs.alloc(
+            sec.n_tiles * (PMH_TILE_ROWS + PMH_MAX_RUNS) * 4);
+        sec.tile_counts = (int32_t *)plan->bufs.alloc(sec.n_tiles * 4);
+        sec.tile_offsets = (int64_t *)plan->bufs.alloc(sec.n_tiles * 8);
+        if (!sec.winners || !sec.tile_counts || !sec.tile_offsets)
+            return false;
+    }
     sec.total_dev = (int64_t *)plan->bufs.alloc(8);
     sec.err_dev = (uint32_t *)plan->bufs.alloc(4);
     if (sec.err_dev) (void)hipMemset(sec.err_dev, 0, 4);
@@ -1401,8 +1417,7 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
         if (!sec.def_all) return false;
     }
     return sec.key_cols && sec.seq_cols && sec.kind_cols && sec.all_cols &&
-           sec.lens_dev && sec.cuts && sec.winners && sec.tile_counts &&
-           sec.tile_offsets && sec.total_dev && sec.err_dev;
+           sec.lens_dev && sec.cuts && sec.total_dev && sec.err_dev;
 }
 
 }  // namespace pmh
@@ -1534,6 +1549,12 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         }
         plan->drop_delete = j["drop_delete"].as_bool(true);
         plan->ignore_delete = j["ignore_delete"].as_bool(false);
+        {
+            // fused single-pass merge+emit is the product path for the
+            // winner engines; PMH_FUSED=0 keeps the 3-kernel chain for A/B
+            const char *pf = getenv("PMH_FUSED");
+            plan->fused = !plan->pu && !(pf && pf[0] == '0');
+        }
         if (plan->rrod && plan->ignore_delete) {
             set_error("remove-record-on-delete cannot be used with "
                       "ignore-delete (PartialUpdateMergeFunction.java:"
@@ -1809,6 +1830,28 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                 (p->rrod ? 16 : 0);
     if (const char *ab = getenv("PMH_ABLATE"))  // profiling-only phase knob
         flags |= (atoi(ab) & 0xf) << 8;
+    if (p->fused) {
+        // single-pass merge + emit: zero the lookback words + ticket, then
+        // one kernel does merge, offsets and emission (scan/emit launches
+        // and the winners round-trip disappear)
+        e = hipMemsetAsync(sec.status, 0, sec.n_tiles * 8, st);
+        if (e != hipSuccess) return fail("status memset", e);
+        e = hipMemsetAsync(sec.ticket, 0, 8, st);
+        if (e != hipSuccess) return fail("ticket memset", e);
+        int key_col = p->composite_key ? -1 : 0;
+        e = pmh_launch_merge_emit(
+            sec.key_cols, sec.seq_cols, sec.kind_cols, sec.lens_dev, k,
+            sec.cuts, sec.n_tiles, PMH_TILE_ROWS, flags, sec.all_cols,
+            p->col_dtype_dev, p->col_nullable_dev, n_cols, key_col,
+            p->n_key_cols, p->n_key_cols + 1, sec.status, sec.ticket,
+            sec.total_dev, p->out_ptrs_dev, p->out_valid_dev, sec.err_dev,
+            st);
+        if (e != hipSuccess) return fail("merge_emit", e);
+        (void)hipEventRecord(ev[3], st);
+        (void)hipEventRecord(ev[4], st);
+        (void)hipEventRecord(ev[5], st);
+        goto collect;
+    }
     e = pmh_launch_merge_tiles(sec.key_cols, sec.seq_cols, sec.kind_cols,
                                sec.lens_dev, k, sec.cuts, sec.n_tiles,
                                PMH_TILE_ROWS, flags, sec.winners,
@@ -1844,6 +1887,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     if (e != hipSuccess) return fail("emit", e);
     (void)hipEventRecord(ev[5], st);
 
+collect:
     int64_t total = 0;
     uint32_t err_word = 0;
     e = hipMemcpyAsync(&total, sec.total_dev, 8, hipMemcpyDeviceToHost, st);
@@ -1870,6 +1914,11 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         set_error("first-row merge engine cannot accept DELETE/UPDATE_BEFORE "
                   "records; configure 'ignore-delete' to skip them "
                   "(FirstRowMergeFunction.java:49-59)");
+        return -1;
+    }
+    if (err_word & 4) {
+        set_error("internal: fused merge lookback timed out waiting for a "
+                  "predecessor tile");
         return -1;
     }
 
