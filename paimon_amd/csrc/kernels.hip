@@ -696,7 +696,7 @@ constexpr uint64_t LOOK_PREFIX = 2ull << 62;
 constexpr uint64_t LOOK_VAL = (1ull << 62) - 1;
 
 template <bool FR>
-__launch_bounds__(PMH_TILE_THREADS) __global__
+__launch_bounds__(PMH_TILE_THREADS, 2) __global__
 void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                   const int64_t *lens, int k, const int32_t *cuts,
                   int64_t n_tiles, int64_t tile_rows, int flags,
@@ -943,73 +943,71 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         __syncthreads();
         if (C == 0) continue;  // nothing to emit (all threads agree: C is
                                // uniform after the pass-0 block reduction)
+        const int ablate = (flags >> 12) & 0x3;  // profiling-only: 1 = skip
+                                                 // value loop, 2 = skip all
+                                                 // emission
+        if (ablate == 2) continue;
         const int64_t goff = sm.s_goff;
         const uint16_t *wl = sm.perm[cur ^ 1];
 
-        // --- emit key / seq / kind straight from the merge arrays
-        if (key_col >= 0) {
-            const int kdt = col_dtype[key_col];
-            for (int32_t i = tid; i < C; i += blockDim.x) {
-                int64_t v = sm.skey[wl[i]];
-                switch (kdt) {
-                case 1: ((int8_t *)out_ptrs[key_col])[goff + i] =
-                            (int8_t)v; break;
-                case 2: ((int16_t *)out_ptrs[key_col])[goff + i] =
-                            (int16_t)v; break;
-                case 3: ((int32_t *)out_ptrs[key_col])[goff + i] =
-                            (int32_t)v; break;
-                default: ((int64_t *)out_ptrs[key_col])[goff + i] = v; break;
-                }
+        // per-thread slot map over the flat element space (computed once,
+        // reused by every column): slot s covers element tid + s*T
+        constexpr int RMAX =
+            (PMH_TILE_MAX + PMH_TILE_THREADS - 1) / PMH_TILE_THREADS;
+        int32_t srun[RMAX];
+        int32_t srow[RMAX];
+#pragma unroll
+        for (int s = 0; s < RMAX; s++) {
+            int32_t e = tid + s * PMH_TILE_THREADS;
+            srun[s] = -1;
+            srow[s] = 0;
+            if (e < M) {
+                int r = 0;
+                while (r + 1 < k && sm.segoff[r + 1] <= e) r++;
+                srun[s] = r;
+                srow[s] = c0[r] + (e - sm.segoff[r]);
             }
         }
-        for (int32_t i = tid; i < C; i += blockDim.x) {
-            int64_t w = sm.sseq[wl[i]];
-            ((int64_t *)out_ptrs[seq_col])[goff + i] = w >> 2;
-            ((int8_t *)out_ptrs[kind_col])[goff + i] = (int8_t)ps2_kind(w);
-        }
-        __syncthreads();  // skey/sseq die; vbuf slabs take their space
 
-        // --- remaining columns: stage per-run segments coalesced into an
-        // LDS slab, scatter to the dense output by winner index.
-        // Double-buffered when no column is nullable (one barrier/column);
-        // nullable plans run single-buffered with the validity bytes in the
-        // second slab.
-        bool any_null = false;
-        for (int c = 0; c < n_cols; c++)
-            if (col_nullable[c]) any_null = true;
-
-        auto stage_col = [&](int c, uint8_t *slab) {
-            for (int r = 0; r < k; r++) {
-                int32_t off = sm.segoff[r], len = sm.seglen[r];
-                const DevCol &dc = cols[r * n_cols + c];
-                if (dc.esize == 8) {
-                    const int64_t *src =
-                        reinterpret_cast<const int64_t *>(dc.addr0) + c0[r];
-                    int64_t *dst = reinterpret_cast<int64_t *>(slab) + off;
-                    for (int32_t i = tid; i < len; i += blockDim.x)
-                        dst[i] = src[i];
-                } else {
-                    const int32_t *src =
-                        reinterpret_cast<const int32_t *>(dc.addr0) + c0[r];
-                    int32_t *dst = reinterpret_cast<int32_t *>(slab) + off;
-                    for (int32_t i = tid; i < len; i += blockDim.x)
-                        dst[i] = src[i];
-                }
+        // register staging: issue a column's gathers into registers FIRST
+        // (no LDS write yet, so the loads stay in flight), do independent
+        // work (emit the previous column from the other slab), then write
+        // the registers to LDS — the vmcnt wait lands after the overlap
+        // work instead of before it.
+        int64_t rg[RMAX];
+        auto stage_load = [&](int c, int64_t *regs) {
+#pragma unroll
+            for (int s = 0; s < RMAX; s++) {
+                if (srun[s] < 0) continue;
+                const DevCol &dc = cols[srun[s] * n_cols + c];
+                regs[s] = dc.esize == 8
+                    ? reinterpret_cast<const int64_t *>(dc.addr0)[srow[s]]
+                    : (int64_t)reinterpret_cast<const int32_t *>(
+                          dc.addr0)[srow[s]];
+            }
+        };
+        auto stage_write = [&](int c, const int64_t *regs, uint8_t *slab) {
+            const int dt = col_dtype[c];
+            const bool wide = dt == 4 || dt == 6;
+#pragma unroll
+            for (int s = 0; s < RMAX; s++) {
+                if (srun[s] < 0) continue;
+                int32_t e = tid + s * PMH_TILE_THREADS;
+                if (wide)
+                    reinterpret_cast<int64_t *>(slab)[e] = regs[s];
+                else
+                    reinterpret_cast<int32_t *>(slab)[e] = (int32_t)regs[s];
             }
         };
         auto stage_valid = [&](int c, uint8_t *slab) {
-            for (int r = 0; r < k; r++) {
-                int32_t off = sm.segoff[r], len = sm.seglen[r];
-                const DevCol &dc = cols[r * n_cols + c];
-                if (dc.valid0) {
-                    const uint8_t *src =
-                        reinterpret_cast<const uint8_t *>(dc.valid0) + c0[r];
-                    for (int32_t i = tid; i < len; i += blockDim.x)
-                        slab[off + i] = src[i];
-                } else {
-                    for (int32_t i = tid; i < len; i += blockDim.x)
-                        slab[off + i] = 1;
-                }
+#pragma unroll
+            for (int s = 0; s < RMAX; s++) {
+                if (srun[s] < 0) continue;
+                int32_t e = tid + s * PMH_TILE_THREADS;
+                const DevCol &dc = cols[srun[s] * n_cols + c];
+                slab[e] = dc.valid0
+                    ? reinterpret_cast<const uint8_t *>(dc.valid0)[srow[s]]
+                    : 1;
             }
         };
         auto emit_col = [&](int c, const uint8_t *slab,
@@ -1035,12 +1033,54 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                 if (ov) ov[goff + i] = vslab ? vslab[s] : 1;
             }
         };
+        auto next_col = [&](int c) -> int {
+            for (c++; c < n_cols; c++)
+                if (c != seq_col && c != kind_col && c != key_col) return c;
+            return -1;
+        };
 
+        bool any_null = false;
+        for (int c = 0; c < n_cols; c++)
+            if (col_nullable[c]) any_null = true;
+
+        // --- overlap window 1: first column's gathers fly while the
+        // key/seq/kind emit reads the merge arrays
+        int cfirst = ablate == 1 ? -1 : next_col(-1);
+        if (cfirst >= 0 && !any_null) stage_load(cfirst, rg);
+
+        // emit key / seq / kind straight from the merge arrays
+        if (key_col >= 0) {
+            const int kdt = col_dtype[key_col];
+            for (int32_t i = tid; i < C; i += blockDim.x) {
+                int64_t v = sm.skey[wl[i]];
+                switch (kdt) {
+                case 1: ((int8_t *)out_ptrs[key_col])[goff + i] =
+                            (int8_t)v; break;
+                case 2: ((int16_t *)out_ptrs[key_col])[goff + i] =
+                            (int16_t)v; break;
+                case 3: ((int32_t *)out_ptrs[key_col])[goff + i] =
+                            (int32_t)v; break;
+                default: ((int64_t *)out_ptrs[key_col])[goff + i] = v; break;
+                }
+            }
+        }
+        for (int32_t i = tid; i < C; i += blockDim.x) {
+            int64_t w = sm.sseq[wl[i]];
+            ((int64_t *)out_ptrs[seq_col])[goff + i] = w >> 2;
+            ((int8_t *)out_ptrs[kind_col])[goff + i] = (int8_t)ps2_kind(w);
+        }
+        __syncthreads();  // skey/sseq die; vbuf slabs take their space
+
+        if (cfirst < 0) {
+            __syncthreads();
+            continue;
+        }
         if (any_null) {
-            // single-buffered: values in slab 0, validity bytes in slab 1
-            for (int c = 0; c < n_cols; c++) {
-                if (c == seq_col || c == kind_col || c == key_col) continue;
-                stage_col(c, sm.vbuf[0]);
+            // nullable plans: values in slab 0, validity bytes in slab 1
+            // (single-buffered; the dedup headline configs carry no nulls)
+            for (int c = cfirst; c >= 0; c = next_col(c)) {
+                stage_load(c, rg);
+                stage_write(c, rg, sm.vbuf[0]);
                 const bool nul = col_nullable[c];
                 if (nul) stage_valid(c, sm.vbuf[1]);
                 __syncthreads();
@@ -1048,24 +1088,20 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                 __syncthreads();
             }
         } else {
-            int prev = -1, slot = 0;
-            for (int c = 0; c < n_cols; c++) {
-                if (c == seq_col || c == kind_col || c == key_col) continue;
-                if (prev < 0) {  // first column: fill slab 0
-                    stage_col(c, sm.vbuf[0]);
-                    prev = c;
-                    continue;
-                }
-                __syncthreads();  // slab `slot` ready; slab slot^1 drained
-                stage_col(c, sm.vbuf[slot ^ 1]);
+            // double slab + register pipeline: column c+1's gathers are in
+            // flight while column c is emitted; one barrier per column
+            stage_write(cfirst, rg, sm.vbuf[0]);
+            int prev = cfirst, slot = 0;
+            for (int c = next_col(cfirst); c >= 0; c = next_col(c)) {
+                stage_load(c, rg);  // loads fly across the barrier + emit
+                __syncthreads();  // slab `slot` ready; slot^1 drained
                 emit_col(prev, sm.vbuf[slot], nullptr);
+                stage_write(c, rg, sm.vbuf[slot ^ 1]);
                 prev = c;
                 slot ^= 1;
             }
-            if (prev >= 0) {
-                __syncthreads();
-                emit_col(prev, sm.vbuf[slot], nullptr);
-            }
+            __syncthreads();
+            emit_col(prev, sm.vbuf[slot], nullptr);
         }
         __syncthreads();  // all slabs drained before the next tile
     }

@@ -1830,6 +1830,8 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                 (p->rrod ? 16 : 0);
     if (const char *ab = getenv("PMH_ABLATE"))  // profiling-only phase knob
         flags |= (atoi(ab) & 0xf) << 8;
+    if (const char *ab = getenv("PMH_FABL"))  // fused-kernel phase knob
+        flags |= (atoi(ab) & 0x3) << 12;      // (profiling only)
     if (p->fused) {
         // single-pass merge + emit: zero the lookback words + ticket, then
         // one kernel does merge, offsets and emission (scan/emit launches

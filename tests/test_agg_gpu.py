@@ -142,7 +142,7 @@ class TestAggregationEngine:
                 vals = rng.integers(-1000, 1000, n).astype(np.int32)
                 msk = rng.random(n) > 0.3
                 runs.append({"key": key,
-                             "seq": np.arange(i * n, (i + 1) * n, np.int64),
+                             "seq": np.arange(i * n, (i + 1) * n, dtype=np.int64),
                              "kind": kind, "values": [key.copy(), vals],
                              "valid": [np.ones(n, bool), msk]})
             metas = write_runs(runs, str(tmp_path / str(drop_delete)),
