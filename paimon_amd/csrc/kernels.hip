@@ -1012,6 +1012,7 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         };
         auto emit_col = [&](int c, const uint8_t *slab,
                             const uint8_t *vslab) {
+            if (ablate == 3) return;  // profiling: staging without emission
             const int dt = col_dtype[c];
             const bool wide = dt == 4 || dt == 6;
             uint8_t *ov = (col_nullable[c] && out_valid[c]) ? out_valid[c]

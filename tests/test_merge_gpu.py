@@ -438,15 +438,22 @@ class TestErrorPaths:
     def test_run_row_overflow_rejected(self, tmp_path):
         # a run is the concatenation of its files: two non-overlapping files
         # whose TOTAL claims >= 2^27 rows must be rejected before the packed
-        # (run | row) winner format could overflow (files are never opened —
-        # the guard fires on the declared row counts)
+        # (run | row) winner format could overflow (the oversized files are
+        # never opened — the guard fires on the declared row counts; only
+        # the tiny bridge run, which stages first, exists on disk)
+        from paimon_amd.reader import write_parquet
+        keys = np.arange(5, 26, dtype=np.int64)
+        write_parquet(str(tmp_path / "c.parquet"), [
+            ("_KEY_k", keys), ("_SEQUENCE_NUMBER", keys),
+            ("_VALUE_KIND", np.zeros(len(keys), np.int8)),
+            ("v_k", keys), ("v_c0", np.zeros(len(keys), np.int32))])
         metas = [{"path": str(tmp_path / "a.parquet"), "rowCount": 70_000_000,
                   "minKey": 0, "maxKey": 10, "level": 0},
                  {"path": str(tmp_path / "b.parquet"), "rowCount": 70_000_000,
                   "minKey": 20, "maxKey": 30, "level": 0},
                  # bridge file: overlaps both, forcing one section where a+b
                  # concatenate into a single run
-                 {"path": str(tmp_path / "c.parquet"), "rowCount": 1_000,
+                 {"path": str(tmp_path / "c.parquet"), "rowCount": 21,
                   "minKey": 5, "maxKey": 25, "level": 0}]
         with Session(0) as s:
             with pytest.raises(RuntimeError, match="per-run limit"):
