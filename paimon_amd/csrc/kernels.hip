@@ -1043,11 +1043,18 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         bool any_null = false;
         for (int c = 0; c < n_cols; c++)
             if (col_nullable[c]) any_null = true;
+        // emission mode (flags bit 14): 0 = direct gather per winner (high
+        // MLP, no per-column barriers — winners are ~72% dense within their
+        // run segments, so the 64B gather lines are mostly reused and the
+        // working set stays XCD-L2-resident); 1 = LDS-staged columns
+        // (coalesced loads but burst-and-wait per column: measured slower,
+        // kept for A/B via PMH_FSTAGE)
+        const bool staged = (flags >> 14) & 1;
 
         // --- overlap window 1: first column's gathers fly while the
         // key/seq/kind emit reads the merge arrays
         int cfirst = ablate == 1 ? -1 : next_col(-1);
-        if (cfirst >= 0 && !any_null) stage_load(cfirst, rg);
+        if (staged && cfirst >= 0 && !any_null) stage_load(cfirst, rg);
 
         // emit key / seq / kind straight from the merge arrays
         if (key_col >= 0) {
@@ -1070,12 +1077,97 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             ((int64_t *)out_ptrs[seq_col])[goff + i] = w >> 2;
             ((int8_t *)out_ptrs[kind_col])[goff + i] = (int8_t)ps2_kind(w);
         }
-        __syncthreads();  // skey/sseq die; vbuf slabs take their space
-
         if (cfirst < 0) {
             __syncthreads();
             continue;
         }
+        if (!staged) {
+            // --- direct-gather emission: thread i owns winners i, i+T,
+            // i+2T, ... R at a time wave-strided (coalesced stores); per
+            // column R independent gathers stay in flight. No barriers, no
+            // LDS traffic — overlaps freely with the other workgroup's
+            // merge phase on the CU.
+            constexpr int R = 4;
+            for (int32_t base = 0; base < C;
+                 base += (int32_t)blockDim.x * R) {
+                int32_t i0 = base + tid;
+                if (i0 >= C) break;
+                int run[R];
+                int64_t row[R];
+                int32_t idx[R];
+                int nr = 0;
+#pragma unroll
+                for (int x = 0; x < R; x++) {
+                    int32_t i = i0 + x * (int32_t)blockDim.x;
+                    bool live = i < C;
+                    idx[x] = i;
+                    if (live) {
+                        nr = x + 1;
+                        uint16_t s = wl[i];
+                        int r = 0;
+                        while (r + 1 < k && sm.segoff[r + 1] <= (int32_t)s)
+                            r++;
+                        run[x] = r;
+                        row[x] = c0[r] + ((int32_t)s - sm.segoff[r]);
+                    } else {
+                        run[x] = run[0];
+                        row[x] = row[0];
+                    }
+                }
+                for (int c = cfirst; c >= 0; c = next_col(c)) {
+                    const int dt = col_dtype[c];
+                    uint8_t *ov = (col_nullable[c] && out_valid[c])
+                                      ? out_valid[c] : nullptr;
+                    if (ov) {
+                        uint8_t vv[R];
+#pragma unroll
+                        for (int x = 0; x < R; x++) {
+                            const DevCol &dc = cols[run[x] * n_cols + c];
+                            vv[x] = dc.valid0
+                                ? reinterpret_cast<const uint8_t *>(
+                                      dc.valid0)[row[x]]
+                                : 1;
+                        }
+#pragma unroll
+                        for (int x = 0; x < R; x++)
+                            if (x < nr) ov[goff + idx[x]] = vv[x];
+                    }
+                    if (dt == 4 || dt == 6) {
+                        int64_t v[R];
+#pragma unroll
+                        for (int x = 0; x < R; x++)
+                            v[x] = reinterpret_cast<const int64_t *>(
+                                cols[run[x] * n_cols + c].addr0)[row[x]];
+#pragma unroll
+                        for (int x = 0; x < R; x++)
+                            if (x < nr)
+                                ((int64_t *)out_ptrs[c])[goff + idx[x]] =
+                                    v[x];
+                    } else {
+                        int32_t v[R];
+#pragma unroll
+                        for (int x = 0; x < R; x++)
+                            v[x] = reinterpret_cast<const int32_t *>(
+                                cols[run[x] * n_cols + c].addr0)[row[x]];
+#pragma unroll
+                        for (int x = 0; x < R; x++) {
+                            if (x >= nr) continue;
+                            switch (dt) {
+                            case 1: ((int8_t *)out_ptrs[c])[goff + idx[x]] =
+                                        (int8_t)v[x]; break;
+                            case 2: ((int16_t *)out_ptrs[c])[goff + idx[x]] =
+                                        (int16_t)v[x]; break;
+                            default: ((int32_t *)out_ptrs[c])[goff + idx[x]] =
+                                        v[x]; break;
+                            }
+                        }
+                    }
+                }
+            }
+            __syncthreads();  // wl/segoff stay live until every thread done
+            continue;
+        }
+        __syncthreads();  // skey/sseq die; vbuf slabs take their space
         if (any_null) {
             // nullable plans: values in slab 0, validity bytes in slab 1
             // (single-buffered; the dedup headline configs carry no nulls)

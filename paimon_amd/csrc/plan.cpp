@@ -1832,6 +1832,8 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         flags |= (atoi(ab) & 0xf) << 8;
     if (const char *ab = getenv("PMH_FABL"))  // fused-kernel phase knob
         flags |= (atoi(ab) & 0x3) << 12;      // (profiling only)
+    if (const char *fs = getenv("PMH_FSTAGE"))  // A/B: LDS-staged emission
+        flags |= (atoi(fs) & 1) << 14;          // instead of direct gather
     if (p->fused) {
         // single-pass merge + emit: zero the lookback words + ticket, then
         // one kernel does merge, offsets and emission (scan/emit launches
