@@ -102,9 +102,14 @@ template <int KM>  // compile-time run-count bound: keeps the per-thread
                    // window arrays in registers (KM=16 spilled 80 VGPRs)
 __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
                             int64_t tile_rows, int64_t n_bounds,
-                            int64_t total_rows, int32_t *cuts /* n_bounds*k */) {
-    int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (b >= n_bounds) return;
+                            int64_t total_rows, int64_t stride,
+                            int32_t *cuts /* n_bounds*k */) {
+    // stride > 1: coarse pass of the two-level partition — only bounds
+    // {0, stride, 2*stride, ...} u {n_bounds-1} are computed here; the
+    // interior bounds follow in k_partition_refine with windows (and key
+    // domains) clamped by the enclosing coarse cuts.
+    int64_t b = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * stride;
+    if (b > n_bounds - 1) b = n_bounds - 1;
     int64_t D = b * tile_rows;
     if (D > total_rows) D = total_rows;
 
@@ -182,6 +187,95 @@ __global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
         if (r >= k) continue;
         int64_t c = pos[r];
         bool has = (c < len[r]) && (ukey(key_at(addr[r], c, kes)) == klo);
+        if (t > 0 && has) { c++; t--; }
+        cuts[b * k + r] = (int32_t)c;
+    }
+}
+
+// ------------------------------------------------------- k_partition_refine
+//
+// Level 2 of the two-level partition: each interior bound bisects inside
+// the windows of its enclosing coarse cuts. Correct because a coarse cut
+// is two-sided: every element below it keys <= its pivot and the pivot
+// keys of bounds >= tile_rows ranks apart differ strictly (an equal-key
+// group has <= k << tile_rows members, one per run), so no element outside
+// the window can tie with any probed pivot in a way that changes a count
+// comparison.
+template <int KM>
+__global__ void k_partition_refine(const DevCol *keys, const int64_t *lens,
+                                   int k, int64_t tile_rows,
+                                   int64_t n_bounds, int64_t total_rows,
+                                   int64_t G, int32_t *cuts) {
+    int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= n_bounds - 1) return;
+    if (b % G == 0) return;  // coarse pass computed it
+    int64_t D = b * tile_rows;  // b interior => 0 < D < total_rows
+    int64_t g0 = (b / G) * G;
+    int64_t g1 = g0 + G;
+    if (g1 > n_bounds - 1) g1 = n_bounds - 1;
+    uint64_t addr[KM];
+    int64_t wlo[KM], whi[KM], pos[KM];
+    const int kes = keys[0].esize;
+    uint64_t klo = ~0ull, khi = 0;
+#pragma unroll
+    for (int r = 0; r < KM; r++) {
+        if (r >= k) continue;
+        addr[r] = keys[r].addr0;
+        wlo[r] = cuts[g0 * k + r];
+        whi[r] = cuts[g1 * k + r];
+        if (wlo[r] < whi[r]) {
+            uint64_t a = ukey(key_at(addr[r], wlo[r], kes));
+            uint64_t z = ukey(key_at(addr[r], whi[r] - 1, kes));
+            if (a < klo) klo = a;
+            if (z > khi) khi = z;
+        }
+    }
+    if (khi < klo) {  // every window empty: the coarse cuts already sum to D
+#pragma unroll
+        for (int r = 0; r < KM; r++) {
+            if (r >= k) continue;
+            cuts[b * k + r] = (int32_t)wlo[r];
+        }
+        return;
+    }
+    while (klo < khi) {
+        uint64_t mid = klo + ((khi - klo) >> 1);
+        bound_multi<true, KM>(addr, wlo, whi, k, kes, mid, pos);
+        int64_t cnt = 0;
+#pragma unroll
+        for (int r = 0; r < KM; r++) {
+            if (r >= k) continue;
+            cnt += pos[r];
+        }
+        if (cnt >= D) {
+            khi = mid;
+#pragma unroll
+            for (int r = 0; r < KM; r++) {
+                if (r >= k) continue;
+                whi[r] = pos[r];
+            }
+        } else {
+            klo = mid + 1;
+#pragma unroll
+            for (int r = 0; r < KM; r++) {
+                if (r >= k) continue;
+                wlo[r] = pos[r];
+            }
+        }
+    }
+    bound_multi<false, KM>(addr, wlo, whi, k, kes, klo, pos);
+    int64_t base = 0;
+#pragma unroll
+    for (int r = 0; r < KM; r++) {
+        if (r >= k) continue;
+        base += pos[r];
+    }
+    int64_t t = D - base;
+#pragma unroll
+    for (int r = 0; r < KM; r++) {
+        if (r >= k) continue;
+        int64_t c = pos[r];
+        bool has = (c < lens[r]) && (ukey(key_at(addr[r], c, kes)) == klo);
         if (t > 0 && has) { c++; t--; }
         cuts[b * k + r] = (int32_t)c;
     }
@@ -696,7 +790,7 @@ constexpr uint64_t LOOK_PREFIX = 2ull << 62;
 constexpr uint64_t LOOK_VAL = (1ull << 62) - 1;
 
 template <bool FR>
-__launch_bounds__(PMH_TILE_THREADS, 2) __global__
+__launch_bounds__(PMH_TILE_THREADS, 4) __global__
 void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                   const int64_t *lens, int k, const int32_t *cuts,
                   int64_t n_tiles, int64_t tile_rows, int flags,
@@ -2194,24 +2288,31 @@ hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
                                 int64_t tile_rows, int64_t n_bounds,
                                 int64_t total_rows, int32_t *cuts,
                                 hipStream_t stream) {
-    int threads = 128;
-    int blocks = (int)((n_bounds + threads - 1) / threads);
+    const int threads = 128;
+    // two-level: coarse bounds every PMH_COARSE_G tiles bisect the full
+    // runs; the interior bounds then bisect only inside their enclosing
+    // coarse windows (fewer key-domain iterations, L2-hot probes)
+    const int64_t G = n_bounds > PMH_COARSE_G + 1 ? PMH_COARSE_G : 1;
+    const int64_t n_coarse = (n_bounds - 2) / G + 2;  // {0,G,..} u {last}
+    int cblocks = (int)((n_coarse + threads - 1) / threads);
+    int rblocks = (int)((n_bounds + threads - 1) / threads);
+    auto launch = [&](auto coarse, auto refine) {
+        hipLaunchKernelGGL(coarse, dim3(cblocks), dim3(threads), 0, stream,
+                           keys, lens, k, tile_rows, n_bounds, total_rows, G,
+                           cuts);
+        if (G > 1)
+            hipLaunchKernelGGL(refine, dim3(rblocks), dim3(threads), 0,
+                               stream, keys, lens, k, tile_rows, n_bounds,
+                               total_rows, G, cuts);
+    };
     if (k <= 4)
-        hipLaunchKernelGGL(k_partition<4>, dim3(blocks), dim3(threads), 0,
-                           stream, keys, lens, k, tile_rows, n_bounds,
-                           total_rows, cuts);
+        launch(k_partition<4>, k_partition_refine<4>);
     else if (k <= 8)
-        hipLaunchKernelGGL(k_partition<8>, dim3(blocks), dim3(threads), 0,
-                           stream, keys, lens, k, tile_rows, n_bounds,
-                           total_rows, cuts);
+        launch(k_partition<8>, k_partition_refine<8>);
     else if (k <= 16)
-        hipLaunchKernelGGL(k_partition<16>, dim3(blocks), dim3(threads), 0,
-                           stream, keys, lens, k, tile_rows, n_bounds,
-                           total_rows, cuts);
+        launch(k_partition<16>, k_partition_refine<16>);
     else
-        hipLaunchKernelGGL(k_partition<PMH_MAX_RUNS>, dim3(blocks),
-                           dim3(threads), 0, stream, keys, lens, k, tile_rows,
-                           n_bounds, total_rows, cuts);
+        launch(k_partition<PMH_MAX_RUNS>, k_partition_refine<PMH_MAX_RUNS>);
     return hipGetLastError();
 }
 
@@ -2325,9 +2426,9 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  int64_t *total_out, void *const *out_ptrs,
                                  uint8_t *const *out_valid,
                                  uint32_t *err_flag, hipStream_t stream) {
-    // persistent workgroups: 2 resident per CU (LDS-bound) x 256 CUs; the
+    // persistent workgroups: 4 resident per CU (LDS-bound) x 256 CUs; the
     // ticket hands out tiles in order, so any residency is safe
-    int blocks = n_tiles < 512 ? (int)n_tiles : 512;
+    int blocks = n_tiles < 1024 ? (int)n_tiles : 1024;
     const bool fr = flags & 8;
     auto launch = [&](auto kern) {
         hipLaunchKernelGGL(kern, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
