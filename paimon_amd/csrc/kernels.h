@@ -17,13 +17,14 @@ namespace pmh {
 constexpr int PMH_MAX_RUNS = 32;
 constexpr int PMH_ROW_BITS = 27;
 constexpr uint32_t PMH_ROW_MASK = ((uint32_t)1 << PMH_ROW_BITS) - 1;
-// 1792-row tiles with 256-thread workgroups: ~39 KB LDS per tile keeps 4
-// workgroups resident per CU (was 3584/512 = 76 KB = 2/CU) — the extra
-// independent tiles per CU hide the merge/emit latency chains. The finer
-// tile grid is affordable because the partition runs two-level (coarse
+// 3584-row tiles with 512-thread workgroups (~76 KB LDS, 2 workgroups/CU).
+// Measured on MI355X: halving to 1792/256 for 4 WGs/CU made every phase
+// SLOWER (fused 6.3 -> 7.8 ms; C3 emit +60%) — the doubled tile count costs
+// more in per-tile setup/barriers/lookback than the extra resident tiles
+// buy in latency hiding. The partition runs two-level regardless (coarse
 // every PMH_COARSE_G-th boundary, windowed refine for the rest).
-constexpr int PMH_TILE_THREADS = 256;
-constexpr int64_t PMH_TILE_ROWS = 1792;
+constexpr int PMH_TILE_THREADS = 512;
+constexpr int64_t PMH_TILE_ROWS = 3584;
 constexpr int PMH_COARSE_G = 16;
 constexpr int PMH_TILE_MAX = PMH_TILE_ROWS + PMH_MAX_RUNS;
 constexpr int PMH_TILE_ITER =

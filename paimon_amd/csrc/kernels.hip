@@ -790,7 +790,7 @@ constexpr uint64_t LOOK_PREFIX = 2ull << 62;
 constexpr uint64_t LOOK_VAL = (1ull << 62) - 1;
 
 template <bool FR>
-__launch_bounds__(PMH_TILE_THREADS, 4) __global__
+__launch_bounds__(PMH_TILE_THREADS, 2) __global__
 void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                   const int64_t *lens, int k, const int32_t *cuts,
                   int64_t n_tiles, int64_t tile_rows, int flags,
@@ -982,14 +982,23 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                         C += sm.wave_tot[w];
                     }
                     my_off = add + incl - nloc;
+                    // publish the aggregate NOW (pass 1 only writes the
+                    // winner list): successors' lookbacks unblock one
+                    // walk-pass earlier. The count rides in the packed
+                    // atomic word, so a successor that acquires the flag
+                    // also gets the payload — no separate fence.
+                    if (tid == 0 && tile > 0)
+                        __hip_atomic_store(&status[tile],
+                                           LOOK_AGG | (uint64_t)C,
+                                           __ATOMIC_RELEASE,
+                                           __HIP_MEMORY_SCOPE_AGENT);
                 }
             }
         }
 
-        // --- publish count + decoupled lookback for the global offset.
-        // The count rides in the packed atomic word, so a successor that
-        // acquires the flag also gets the payload — no separate fence.
-        if (tid == 0 && tile > 0)
+        // --- decoupled lookback for the global offset (empty tiles still
+        // publish their zero count here)
+        if (tid == 0 && tile > 0 && Mreal == 0)
             __hip_atomic_store(&status[tile], LOOK_AGG | (uint64_t)C,
                                __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
         if (wv == 0) {
@@ -2426,9 +2435,9 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  int64_t *total_out, void *const *out_ptrs,
                                  uint8_t *const *out_valid,
                                  uint32_t *err_flag, hipStream_t stream) {
-    // persistent workgroups: 4 resident per CU (LDS-bound) x 256 CUs; the
+    // persistent workgroups: 2 resident per CU (LDS-bound) x 256 CUs; the
     // ticket hands out tiles in order, so any residency is safe
-    int blocks = n_tiles < 1024 ? (int)n_tiles : 1024;
+    int blocks = n_tiles < 512 ? (int)n_tiles : 512;
     const bool fr = flags & 8;
     auto launch = [&](auto kern) {
         hipLaunchKernelGGL(kern, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
