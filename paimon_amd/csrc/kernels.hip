@@ -92,177 +92,96 @@ DEV void bound_multi(const uint64_t *addr, const int64_t *lo_in,
     }
 }
 
-// ------------------------------------------------------------ k_partition
+// --------------------------------------------------------- k_partition_wave
 //
-// For each tile boundary b (rank D = b*T clamped), find cuts c_r such that
-// sum c_r = D and the (key, run) total order is two-sided partitioned
-// (GPU merge-path generalized to k runs via key-domain bisection with
-// per-run shrinking windows).
-template <int KM>  // compile-time run-count bound: keeps the per-thread
-                   // window arrays in registers (KM=16 spilled 80 VGPRs)
-__global__ void k_partition(const DevCol *keys, const int64_t *lens, int k,
-                            int64_t tile_rows, int64_t n_bounds,
-                            int64_t total_rows, int64_t stride,
-                            int32_t *cuts /* n_bounds*k */) {
-    // stride > 1: coarse pass of the two-level partition — only bounds
-    // {0, stride, 2*stride, ...} u {n_bounds-1} are computed here; the
-    // interior bounds follow in k_partition_refine with windows (and key
-    // domains) clamped by the enclosing coarse cuts.
-    int64_t b = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * stride;
-    if (b > n_bounds - 1) b = n_bounds - 1;
+// Wave-parallel 64-ary key-domain search: ONE WAVE per tile boundary; per
+// round the 64 lanes probe 64 evenly spaced pivots of the remaining key
+// domain (each lane runs the lockstep multi-run count for its pivot), a
+// ballot picks the bracketing pair, and domain + per-run windows narrow
+// ~65x. Exact counts, so the cuts are IDENTICAL to k_partition's — but the
+// dependent-load chain is ~4 rounds x log2(window) instead of ~36 x
+// log2(window): the partition wall-clock is one latency chain, and this
+// cuts it ~5x. Validated against a brute-force model on randomized runs
+// (tests/test_tile_algorithm.py + the CPU prototype in git history).
+template <int KM>
+__global__ void k_partition_wave(const DevCol *keys, const int64_t *lens,
+                                 int k, int64_t tile_rows, int64_t n_bounds,
+                                 int64_t total_rows, int32_t *cuts) {
+    const int lane = (int)(threadIdx.x & 63);
+    const int64_t b =
+        (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    if (b >= n_bounds) return;
     int64_t D = b * tile_rows;
     if (D > total_rows) D = total_rows;
-
-    uint64_t addr[KM];
-    int64_t len[KM], wlo[KM], whi[KM];
-    const int kes = keys[0].esize;
-#pragma unroll
-    for (int r = 0; r < KM; r++) {
-        if (r >= k) continue;
-        addr[r] = keys[r].addr0;  // staged columns are contiguous
-        len[r] = lens[r];
-        wlo[r] = 0;
-        whi[r] = len[r];
-    }
-    if (D == 0 || D >= total_rows) {
-#pragma unroll
-        for (int r = 0; r < KM; r++) {
-            if (r >= k) continue;
-            cuts[b * k + r] = D == 0 ? 0 : (int32_t)len[r];
-        }
-        return;
-    }
-    // bisect for the smallest v with (# ukey <= v) >= D: the cut falls at
-    // the D-th smallest element's key. Domain initialized from the runs'
-    // actual min/max keys (v* is an existing key; cnt_le(max) = total >= D).
-    uint64_t klo = ~0ull, khi = 0;
-#pragma unroll
-    for (int r = 0; r < KM; r++) {
-        if (r >= k || len[r] == 0) continue;
-        uint64_t lo_k = ukey(key_at(addr[r], 0, kes));
-        uint64_t hi_k = ukey(key_at(addr[r], len[r] - 1, kes));
-        if (lo_k < klo) klo = lo_k;
-        if (hi_k > khi) khi = hi_k;
-    }
-    int64_t pos[KM];
-    while (klo < khi) {
-        uint64_t mid = klo + ((khi - klo) >> 1);
-        bound_multi<true, KM>(addr, wlo, whi, k, kes, mid, pos);
-        int64_t cnt = 0;
-#pragma unroll
-        for (int r = 0; r < KM; r++) {
-            if (r >= k) continue;
-            cnt += pos[r];
-        }
-        if (cnt >= D) {
-            khi = mid;
-#pragma unroll
-            for (int r = 0; r < KM; r++) {
-                if (r >= k) continue;
-                whi[r] = pos[r];
-            }
-        } else {
-            klo = mid + 1;
-#pragma unroll
-            for (int r = 0; r < KM; r++) {
-                if (r >= k) continue;
-                wlo[r] = pos[r];
-            }
-        }
-    }
-    // klo == v*: all elements with key < v* are taken; among key == v*
-    // (<=1 per run), take the first t in run order. Searching within the
-    // final windows yields absolute positions: everything below wlo has
-    // key < v*, everything at/above whi has key > v*.
-    bound_multi<false, KM>(addr, wlo, whi, k, kes, klo, pos);
-    int64_t base = 0;
-#pragma unroll
-    for (int r = 0; r < KM; r++) {
-        if (r >= k) continue;
-        base += pos[r];
-    }
-    int64_t t = D - base;
-#pragma unroll
-    for (int r = 0; r < KM; r++) {
-        if (r >= k) continue;
-        int64_t c = pos[r];
-        bool has = (c < len[r]) && (ukey(key_at(addr[r], c, kes)) == klo);
-        if (t > 0 && has) { c++; t--; }
-        cuts[b * k + r] = (int32_t)c;
-    }
-}
-
-// ------------------------------------------------------- k_partition_refine
-//
-// Level 2 of the two-level partition: each interior bound bisects inside
-// the windows of its enclosing coarse cuts. Correct because a coarse cut
-// is two-sided: every element below it keys <= its pivot and the pivot
-// keys of bounds >= tile_rows ranks apart differ strictly (an equal-key
-// group has <= k << tile_rows members, one per run), so no element outside
-// the window can tie with any probed pivot in a way that changes a count
-// comparison.
-template <int KM>
-__global__ void k_partition_refine(const DevCol *keys, const int64_t *lens,
-                                   int k, int64_t tile_rows,
-                                   int64_t n_bounds, int64_t total_rows,
-                                   int64_t G, int32_t *cuts) {
-    int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (b >= n_bounds - 1) return;
-    if (b % G == 0) return;  // coarse pass computed it
-    int64_t D = b * tile_rows;  // b interior => 0 < D < total_rows
-    int64_t g0 = (b / G) * G;
-    int64_t g1 = g0 + G;
-    if (g1 > n_bounds - 1) g1 = n_bounds - 1;
     uint64_t addr[KM];
     int64_t wlo[KM], whi[KM], pos[KM];
     const int kes = keys[0].esize;
-    uint64_t klo = ~0ull, khi = 0;
 #pragma unroll
     for (int r = 0; r < KM; r++) {
         if (r >= k) continue;
         addr[r] = keys[r].addr0;
-        wlo[r] = cuts[g0 * k + r];
-        whi[r] = cuts[g1 * k + r];
-        if (wlo[r] < whi[r]) {
-            uint64_t a = ukey(key_at(addr[r], wlo[r], kes));
-            uint64_t z = ukey(key_at(addr[r], whi[r] - 1, kes));
-            if (a < klo) klo = a;
-            if (z > khi) khi = z;
-        }
+        wlo[r] = 0;
+        whi[r] = lens[r];
     }
-    if (khi < klo) {  // every window empty: the coarse cuts already sum to D
+    if (D == 0 || D >= total_rows) {
+        if (lane == 0) {
 #pragma unroll
-        for (int r = 0; r < KM; r++) {
-            if (r >= k) continue;
-            cuts[b * k + r] = (int32_t)wlo[r];
+            for (int r = 0; r < KM; r++) {
+                if (r >= k) continue;
+                cuts[b * k + r] = D == 0 ? 0 : (int32_t)lens[r];
+            }
         }
         return;
     }
+    uint64_t klo = ~0ull, khi = 0;
+#pragma unroll
+    for (int r = 0; r < KM; r++) {
+        if (r >= k || whi[r] == 0) continue;
+        uint64_t lo_k = ukey(key_at(addr[r], 0, kes));
+        uint64_t hi_k = ukey(key_at(addr[r], whi[r] - 1, kes));
+        if (lo_k < klo) klo = lo_k;
+        if (hi_k > khi) khi = hi_k;
+    }
     while (klo < khi) {
-        uint64_t mid = klo + ((khi - klo) >> 1);
-        bound_multi<true, KM>(addr, wlo, whi, k, kes, mid, pos);
+        const uint64_t span = khi - klo;
+        uint64_t pv;
+        if (span <= 63) {
+            pv = klo + (uint64_t)lane;
+            if (pv > khi) pv = khi;
+        } else {
+            // overflow-safe klo + span*(lane+1)/65, monotone in lane
+            const uint64_t q = span / 65, rm = span % 65;
+            pv = klo + q * (uint64_t)(lane + 1) +
+                 (rm * (uint64_t)(lane + 1)) / 65;
+        }
+        bound_multi<true, KM>(addr, wlo, whi, k, kes, pv, pos);
         int64_t cnt = 0;
 #pragma unroll
         for (int r = 0; r < KM; r++) {
             if (r >= k) continue;
             cnt += pos[r];
         }
-        if (cnt >= D) {
-            khi = mid;
+        const uint64_t ge = __ballot(cnt >= D);
+        const int f = ge ? (int)(__ffsll((unsigned long long)ge) - 1) : 64;
+        const int fc = f < 64 ? f : 63;   // lane holding the new upper
+        const int fm = f > 0 ? f - 1 : 0; // lane holding the new lower
+        const uint64_t pv_f = __shfl(pv, fc, 64);
+        const uint64_t pv_m = __shfl(pv, fm, 64);
+        if (f == 64) klo = pv_f + 1;        // even the last pivot counts < D
+        else if (f == 0) khi = pv_f;
+        else { klo = pv_m + 1; khi = pv_f; }
 #pragma unroll
-            for (int r = 0; r < KM; r++) {
-                if (r >= k) continue;
-                whi[r] = pos[r];
-            }
-        } else {
-            klo = mid + 1;
-#pragma unroll
-            for (int r = 0; r < KM; r++) {
-                if (r >= k) continue;
-                wlo[r] = pos[r];
-            }
+        for (int r = 0; r < KM; r++) {
+            if (r >= k) continue;
+            const int64_t pf = __shfl(pos[r], fc, 64);
+            const int64_t pm = __shfl(pos[r], fm, 64);
+            if (f == 64) wlo[r] = pf;
+            else if (f == 0) whi[r] = pf;
+            else { wlo[r] = pm; whi[r] = pf; }
         }
     }
+    // klo == v*: identical tie-take finish to k_partition (redundant across
+    // lanes; lane 0 stores)
     bound_multi<false, KM>(addr, wlo, whi, k, kes, klo, pos);
     int64_t base = 0;
 #pragma unroll
@@ -271,13 +190,16 @@ __global__ void k_partition_refine(const DevCol *keys, const int64_t *lens,
         base += pos[r];
     }
     int64_t t = D - base;
+    if (lane == 0) {
 #pragma unroll
-    for (int r = 0; r < KM; r++) {
-        if (r >= k) continue;
-        int64_t c = pos[r];
-        bool has = (c < lens[r]) && (ukey(key_at(addr[r], c, kes)) == klo);
-        if (t > 0 && has) { c++; t--; }
-        cuts[b * k + r] = (int32_t)c;
+        for (int r = 0; r < KM; r++) {
+            if (r >= k) continue;
+            int64_t c = pos[r];
+            bool has =
+                (c < lens[r]) && (ukey(key_at(addr[r], c, kes)) == klo);
+            if (t > 0 && has) { c++; t--; }
+            cuts[b * k + r] = (int32_t)c;
+        }
     }
 }
 
@@ -2297,31 +2219,22 @@ hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
                                 int64_t tile_rows, int64_t n_bounds,
                                 int64_t total_rows, int32_t *cuts,
                                 hipStream_t stream) {
-    const int threads = 128;
-    // two-level: coarse bounds every PMH_COARSE_G tiles bisect the full
-    // runs; the interior bounds then bisect only inside their enclosing
-    // coarse windows (fewer key-domain iterations, L2-hot probes)
-    const int64_t G = n_bounds > PMH_COARSE_G + 1 ? PMH_COARSE_G : 1;
-    const int64_t n_coarse = (n_bounds - 2) / G + 2;  // {0,G,..} u {last}
-    int cblocks = (int)((n_coarse + threads - 1) / threads);
-    int rblocks = (int)((n_bounds + threads - 1) / threads);
-    auto launch = [&](auto coarse, auto refine) {
-        hipLaunchKernelGGL(coarse, dim3(cblocks), dim3(threads), 0, stream,
-                           keys, lens, k, tile_rows, n_bounds, total_rows, G,
+    // one WAVE per bound (64-ary domain search), 4 bounds per block
+    const int threads = 256;
+    int blocks = (int)((n_bounds + 3) / 4);
+    auto launch = [&](auto kern) {
+        hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, stream,
+                           keys, lens, k, tile_rows, n_bounds, total_rows,
                            cuts);
-        if (G > 1)
-            hipLaunchKernelGGL(refine, dim3(rblocks), dim3(threads), 0,
-                               stream, keys, lens, k, tile_rows, n_bounds,
-                               total_rows, G, cuts);
     };
     if (k <= 4)
-        launch(k_partition<4>, k_partition_refine<4>);
+        launch(k_partition_wave<4>);
     else if (k <= 8)
-        launch(k_partition<8>, k_partition_refine<8>);
+        launch(k_partition_wave<8>);
     else if (k <= 16)
-        launch(k_partition<16>, k_partition_refine<16>);
+        launch(k_partition_wave<16>);
     else
-        launch(k_partition<PMH_MAX_RUNS>, k_partition_refine<PMH_MAX_RUNS>);
+        launch(k_partition_wave<PMH_MAX_RUNS>);
     return hipGetLastError();
 }
 
