@@ -129,9 +129,23 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  const uint8_t *col_nullable, int n_cols,
                                  int key_col, int seq_col, int kind_col,
                                  uint64_t *status, uint64_t *ticket,
-                                 int64_t *total_out, void *const *out_ptrs,
+                                 int64_t *total_out, uint32_t *dense_winners,
+                                 void *const *out_ptrs,
                                  uint8_t *const *out_valid,
                                  uint32_t *err_flag, hipStream_t stream);
+
+// SPLIT-mode value emission: gathers value columns by the densely-written
+// packed winners (k_merge_emit with dense_winners != null emitted key/seq/
+// kind and the winner list; this kernel runs at full occupancy, no LDS).
+hipError_t pmh_launch_emit_dense(const DevCol *cols,
+                                 const uint8_t *col_dtype,
+                                 const uint8_t *col_nullable, int n_cols,
+                                 int key_col, int seq_col, int kind_col,
+                                 const uint32_t *winners,
+                                 const int64_t *total_out,
+                                 void *const *out_ptrs,
+                                 uint8_t *const *out_valid,
+                                 hipStream_t stream);
 
 hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
                            const uint8_t *col_nullable, int n_cols, int k,

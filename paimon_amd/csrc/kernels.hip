@@ -106,11 +106,20 @@ DEV void bound_multi(const uint64_t *addr, const int64_t *lo_in,
 template <int KM>
 __global__ void k_partition_wave(const DevCol *keys, const int64_t *lens,
                                  int k, int64_t tile_rows, int64_t n_bounds,
-                                 int64_t total_rows, int32_t *cuts) {
+                                 int64_t total_rows, int64_t stride,
+                                 int32_t *cuts) {
+    // stride > 1: coarse pass of the two-level partition — bounds
+    // {0, stride, 2*stride, ...} u {n_bounds-1}; interior bounds follow in
+    // k_partition_refine with windows clamped by the enclosing coarse cuts
+    // (their probes then stay inside small, L2-resident windows: the wave
+    // kernel's 64 probes/round cost ~12x the old kernel's probe TRAFFIC
+    // when used for every bound — measured 2.4 ms vs 1.2 — so it only runs
+    // where its short chain matters: the full-depth coarse cuts)
     const int lane = (int)(threadIdx.x & 63);
-    const int64_t b =
-        (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
-    if (b >= n_bounds) return;
+    int64_t b =
+        ((int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6)) *
+        stride;
+    if (b > n_bounds - 1) b = n_bounds - 1;
     int64_t D = b * tile_rows;
     if (D > total_rows) D = total_rows;
     uint64_t addr[KM];
@@ -202,6 +211,96 @@ __global__ void k_partition_wave(const DevCol *keys, const int64_t *lens,
         }
     }
 }
+
+// ------------------------------------------------------- k_partition_refine
+//
+// Level 2 of the two-level partition: each interior bound bisects inside
+// the windows of its enclosing coarse cuts. Correct because a coarse cut
+// is two-sided: every element below it keys <= its pivot and the pivot
+// keys of bounds >= tile_rows ranks apart differ strictly (an equal-key
+// group has <= k << tile_rows members, one per run), so no element outside
+// the window can tie with any probed pivot in a way that changes a count
+// comparison.
+template <int KM>
+__global__ void k_partition_refine(const DevCol *keys, const int64_t *lens,
+                                   int k, int64_t tile_rows,
+                                   int64_t n_bounds, int64_t total_rows,
+                                   int64_t G, int32_t *cuts) {
+    int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (b >= n_bounds - 1) return;
+    if (b % G == 0) return;  // coarse pass computed it
+    int64_t D = b * tile_rows;  // b interior => 0 < D < total_rows
+    int64_t g0 = (b / G) * G;
+    int64_t g1 = g0 + G;
+    if (g1 > n_bounds - 1) g1 = n_bounds - 1;
+    uint64_t addr[KM];
+    int64_t wlo[KM], whi[KM], pos[KM];
+    const int kes = keys[0].esize;
+    uint64_t klo = ~0ull, khi = 0;
+#pragma unroll
+    for (int r = 0; r < KM; r++) {
+        if (r >= k) continue;
+        addr[r] = keys[r].addr0;
+        wlo[r] = cuts[g0 * k + r];
+        whi[r] = cuts[g1 * k + r];
+        if (wlo[r] < whi[r]) {
+            uint64_t a = ukey(key_at(addr[r], wlo[r], kes));
+            uint64_t z = ukey(key_at(addr[r], whi[r] - 1, kes));
+            if (a < klo) klo = a;
+            if (z > khi) khi = z;
+        }
+    }
+    if (khi < klo) {  // every window empty: the coarse cuts already sum to D
+#pragma unroll
+        for (int r = 0; r < KM; r++) {
+            if (r >= k) continue;
+            cuts[b * k + r] = (int32_t)wlo[r];
+        }
+        return;
+    }
+    while (klo < khi) {
+        uint64_t mid = klo + ((khi - klo) >> 1);
+        bound_multi<true, KM>(addr, wlo, whi, k, kes, mid, pos);
+        int64_t cnt = 0;
+#pragma unroll
+        for (int r = 0; r < KM; r++) {
+            if (r >= k) continue;
+            cnt += pos[r];
+        }
+        if (cnt >= D) {
+            khi = mid;
+#pragma unroll
+            for (int r = 0; r < KM; r++) {
+                if (r >= k) continue;
+                whi[r] = pos[r];
+            }
+        } else {
+            klo = mid + 1;
+#pragma unroll
+            for (int r = 0; r < KM; r++) {
+                if (r >= k) continue;
+                wlo[r] = pos[r];
+            }
+        }
+    }
+    bound_multi<false, KM>(addr, wlo, whi, k, kes, klo, pos);
+    int64_t base = 0;
+#pragma unroll
+    for (int r = 0; r < KM; r++) {
+        if (r >= k) continue;
+        base += pos[r];
+    }
+    int64_t t = D - base;
+#pragma unroll
+    for (int r = 0; r < KM; r++) {
+        if (r >= k) continue;
+        int64_t c = pos[r];
+        bool has = (c < lens[r]) && (ukey(key_at(addr[r], c, kes)) == klo);
+        if (t > 0 && has) { c++; t--; }
+        cuts[b * k + r] = (int32_t)c;
+    }
+}
+
 
 // ------------------------------------------------------------ k_merge_tiles
 //
@@ -720,7 +819,8 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                   const uint8_t *col_dtype, const uint8_t *col_nullable,
                   int n_cols, int key_col /* -1: composite */, int seq_col,
                   int kind_col, uint64_t *status, uint64_t *ticket,
-                  int64_t *total_out, void *const *out_ptrs,
+                  int64_t *total_out, uint32_t *dense_winners,
+                  void *const *out_ptrs,
                   uint8_t *const *out_valid, uint32_t *err_flag) {
     const bool drop_delete = flags & 1;
     const bool ignore_delete = flags & 2;
@@ -1106,6 +1206,23 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             __syncthreads();
             continue;
         }
+        if (dense_winners) {
+            // SPLIT mode: publish packed winners densely at the global
+            // offset; the value columns are emitted by k_emit_dense (no
+            // LDS, full occupancy) right after this kernel — the merge
+            // phases here cap residency at 2 workgroups/CU, which starves
+            // the gather phase of waves.
+            for (int32_t i = tid; i < C; i += blockDim.x) {
+                uint16_t s = wl[i];
+                int r = 0;
+                while (r + 1 < k && sm.segoff[r + 1] <= (int32_t)s) r++;
+                uint32_t grow =
+                    (uint32_t)(c0[r] + ((int32_t)s - sm.segoff[r]));
+                dense_winners[goff + i] = ((uint32_t)r << PMH_ROW_BITS) | grow;
+            }
+            __syncthreads();
+            continue;
+        }
         if (!staged) {
             // --- direct-gather emission: thread i owns winners i, i+T,
             // i+2T, ... R at a time wave-strided (coalesced stores); per
@@ -1222,6 +1339,99 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             emit_col(prev, sm.vbuf[slot], nullptr);
         }
         __syncthreads();  // all slabs drained before the next tile
+    }
+}
+
+
+// ------------------------------------------------------------ k_emit_dense
+//
+// Value emission for the SPLIT fused path: k_merge_emit already wrote key/
+// seq/kind and the packed winners DENSELY in output order, so this kernel
+// is a pure gather — no LDS, no tile search, 60-VGPR occupancy (8 waves/
+// SIMD vs the merge kernel's 4). Winners are ~72% dense inside their run
+// segments, so gather lines are mostly reused and stay XCD-L2-resident.
+__global__ void k_emit_dense(const DevCol *cols, const uint8_t *col_dtype,
+                             const uint8_t *col_nullable, int n_cols,
+                             int key_col, int seq_col, int kind_col,
+                             const uint32_t *winners,
+                             const int64_t *total_out,
+                             void *const *out_ptrs,
+                             uint8_t *const *out_valid) {
+    constexpr int R = 4;
+    const int64_t total = *total_out;
+    const int64_t per_block =
+        (total + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
+    const int64_t slice_lo = (int64_t)blockIdx.x * per_block;
+    const int64_t slice_hi =
+        slice_lo + per_block < total ? slice_lo + per_block : total;
+    for (int64_t base = slice_lo; base < slice_hi;
+         base += (int64_t)blockDim.x * R) {
+        int64_t i0 = base + threadIdx.x;
+        if (i0 >= slice_hi) break;
+        int run[R];
+        int64_t row[R], idx[R];
+        int nr = 0;
+#pragma unroll
+        for (int x = 0; x < R; x++) {
+            int64_t i = i0 + (int64_t)x * blockDim.x;
+            bool live = i < slice_hi;
+            idx[x] = i;
+            if (live) {
+                nr = x + 1;
+                uint32_t packed = winners[i];
+                run[x] = packed >> PMH_ROW_BITS;
+                row[x] = packed & PMH_ROW_MASK;
+            } else {
+                run[x] = run[0];
+                row[x] = row[0];
+            }
+        }
+        for (int c = 0; c < n_cols; c++) {
+            if (c == key_col || c == seq_col || c == kind_col) continue;
+            const int dt = col_dtype[c];
+            if (col_nullable[c] && out_valid[c]) {
+                uint8_t vv[R];
+#pragma unroll
+                for (int x = 0; x < R; x++) {
+                    const DevCol &dc = cols[run[x] * n_cols + c];
+                    vv[x] = dc.valid0
+                        ? reinterpret_cast<const uint8_t *>(
+                              dc.valid0)[row[x]]
+                        : 1;
+                }
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    if (x < nr) out_valid[c][idx[x]] = vv[x];
+            }
+            if (dt == 4 || dt == 6) {
+                int64_t v[R];
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    v[x] = reinterpret_cast<const int64_t *>(
+                        cols[run[x] * n_cols + c].addr0)[row[x]];
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    if (x < nr) ((int64_t *)out_ptrs[c])[idx[x]] = v[x];
+            } else {
+                int32_t v[R];
+#pragma unroll
+                for (int x = 0; x < R; x++)
+                    v[x] = reinterpret_cast<const int32_t *>(
+                        cols[run[x] * n_cols + c].addr0)[row[x]];
+#pragma unroll
+                for (int x = 0; x < R; x++) {
+                    if (x >= nr) continue;
+                    switch (dt) {
+                    case 1: ((int8_t *)out_ptrs[c])[idx[x]] =
+                                (int8_t)v[x]; break;
+                    case 2: ((int16_t *)out_ptrs[c])[idx[x]] =
+                                (int16_t)v[x]; break;
+                    default: ((int32_t *)out_ptrs[c])[idx[x]] =
+                                 v[x]; break;
+                    }
+                }
+            }
+        }
     }
 }
 
@@ -2219,22 +2429,31 @@ hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
                                 int64_t tile_rows, int64_t n_bounds,
                                 int64_t total_rows, int32_t *cuts,
                                 hipStream_t stream) {
-    // one WAVE per bound (64-ary domain search), 4 bounds per block
-    const int threads = 256;
-    int blocks = (int)((n_bounds + 3) / 4);
-    auto launch = [&](auto kern) {
-        hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, stream,
-                           keys, lens, k, tile_rows, n_bounds, total_rows,
+    // two-level: wave-coarse cuts every PMH_COARSE_G tiles (64-ary domain
+    // search, ~4 probe rounds deep), then windowed refine for the interior
+    // bounds (probes confined to the enclosing coarse windows)
+    const int64_t G = n_bounds > PMH_COARSE_G + 1 ? PMH_COARSE_G : 1;
+    const int64_t n_coarse = (n_bounds - 2) / G + 2;  // {0,G,..} u {last}
+    int cblocks = (int)((n_coarse + 3) / 4);  // one wave per coarse bound
+    int rblocks = (int)((n_bounds + 127) / 128);
+    auto launch = [&](auto coarse, auto refiner) {
+        hipLaunchKernelGGL(coarse, dim3(cblocks), dim3(256), 0, stream,
+                           keys, lens, k, tile_rows, n_bounds, total_rows, G,
                            cuts);
+        if (G > 1)
+            hipLaunchKernelGGL(refiner, dim3(rblocks), dim3(128), 0, stream,
+                               keys, lens, k, tile_rows, n_bounds,
+                               total_rows, G, cuts);
     };
     if (k <= 4)
-        launch(k_partition_wave<4>);
+        launch(k_partition_wave<4>, k_partition_refine<4>);
     else if (k <= 8)
-        launch(k_partition_wave<8>);
+        launch(k_partition_wave<8>, k_partition_refine<8>);
     else if (k <= 16)
-        launch(k_partition_wave<16>);
+        launch(k_partition_wave<16>, k_partition_refine<16>);
     else
-        launch(k_partition_wave<PMH_MAX_RUNS>);
+        launch(k_partition_wave<PMH_MAX_RUNS>,
+               k_partition_refine<PMH_MAX_RUNS>);
     return hipGetLastError();
 }
 
@@ -2345,7 +2564,8 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  const uint8_t *col_nullable, int n_cols,
                                  int key_col, int seq_col, int kind_col,
                                  uint64_t *status, uint64_t *ticket,
-                                 int64_t *total_out, void *const *out_ptrs,
+                                 int64_t *total_out, uint32_t *dense_winners,
+                                 void *const *out_ptrs,
                                  uint8_t *const *out_valid,
                                  uint32_t *err_flag, hipStream_t stream) {
     // persistent workgroups: 2 resident per CU (LDS-bound) x 256 CUs; the
@@ -2357,12 +2577,28 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                            stream, keys, seqs, kinds, lens, k, cuts, n_tiles,
                            tile_rows, flags, cols, col_dtype, col_nullable,
                            n_cols, key_col, seq_col, kind_col, status, ticket,
-                           total_out, out_ptrs, out_valid, err_flag);
+                           total_out, dense_winners, out_ptrs, out_valid,
+                           err_flag);
     };
     if (fr)
         launch(k_merge_emit<true>);
     else
         launch(k_merge_emit<false>);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_emit_dense(const DevCol *cols,
+                                 const uint8_t *col_dtype,
+                                 const uint8_t *col_nullable, int n_cols,
+                                 int key_col, int seq_col, int kind_col,
+                                 const uint32_t *winners,
+                                 const int64_t *total_out,
+                                 void *const *out_ptrs,
+                                 uint8_t *const *out_valid,
+                                 hipStream_t stream) {
+    hipLaunchKernelGGL(k_emit_dense, dim3(2048), dim3(256), 0, stream, cols,
+                       col_dtype, col_nullable, n_cols, key_col, seq_col,
+                       kind_col, winners, total_out, out_ptrs, out_valid);
     return hipGetLastError();
 }
 

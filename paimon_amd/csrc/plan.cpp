@@ -224,6 +224,7 @@ struct Section {
     // the tile ticket, zeroed per read
     uint64_t *status = nullptr;
     uint64_t *ticket = nullptr;
+    uint32_t *dense_winners = nullptr;  // split mode (PMH_FSPLIT)
     // packed per-row validity (PU/agg emit): one u64 per row per run,
     // bit c = column c non-null; built once per section by k_pack_valid
     std::vector<uint64_t *> row_masks;
@@ -257,6 +258,7 @@ struct pmh_plan_t {
     bool first_row = false;  // first-row merge engine
     bool fused = false;      // single-pass k_merge_emit (non-member-list
                              // engines; PMH_FUSED=0 falls back for A/B)
+    bool fsplit = false;     // split value emission (PMH_FSPLIT=1 A/B)
     bool agg = false;        // aggregation merge engine (uses PU member lists)
     uint8_t *col_agg_dev = nullptr;  // per-column PMH_AGG_* codes
     // composite key (>1 key column): order-preserving packed comparand
@@ -1362,6 +1364,11 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
         sec.status = (uint64_t *)plan->bufs.alloc(sec.n_tiles * 8);
         sec.ticket = (uint64_t *)plan->bufs.alloc(8);
         if (!sec.status || !sec.ticket) return false;
+        if (plan->fsplit) {
+            sec.dense_winners =
+                (uint32_t *)plan->bufs.alloc(sec.total_rows * 4);
+            if (!sec.dense_winners) return false;
+        }
     } else {
         sec.winners = (uint32_t *)plan->bufs.alloc(
             sec.n_tiles * (PMH_TILE_ROWS + PMH_MAX_RUNS) * 4);
@@ -1554,6 +1561,8 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             // winner engines; PMH_FUSED=0 keeps the 3-kernel chain for A/B
             const char *pf = getenv("PMH_FUSED");
             plan->fused = !plan->pu && !(pf && pf[0] == '0');
+            const char *fs = getenv("PMH_FSPLIT");
+            plan->fsplit = plan->fused && fs && fs[0] == '1';
         }
         if (plan->rrod && plan->ignore_delete) {
             set_error("remove-record-on-delete cannot be used with "
@@ -1848,11 +1857,18 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             sec.cuts, sec.n_tiles, PMH_TILE_ROWS, flags, sec.all_cols,
             p->col_dtype_dev, p->col_nullable_dev, n_cols, key_col,
             p->n_key_cols, p->n_key_cols + 1, sec.status, sec.ticket,
-            sec.total_dev, p->out_ptrs_dev, p->out_valid_dev, sec.err_dev,
-            st);
+            sec.total_dev, sec.dense_winners, p->out_ptrs_dev,
+            p->out_valid_dev, sec.err_dev, st);
         if (e != hipSuccess) return fail("merge_emit", e);
         (void)hipEventRecord(ev[3], st);
         (void)hipEventRecord(ev[4], st);
+        if (sec.dense_winners) {
+            e = pmh_launch_emit_dense(
+                sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols,
+                key_col, p->n_key_cols, p->n_key_cols + 1, sec.dense_winners,
+                sec.total_dev, p->out_ptrs_dev, p->out_valid_dev, st);
+            if (e != hipSuccess) return fail("emit_dense", e);
+        }
         (void)hipEventRecord(ev[5], st);
         goto collect;
     }
