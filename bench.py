@@ -211,6 +211,8 @@ def main():
         elapsed = float(t.item())
 
     stats1 = plan.stats()
+    path_mode = int(stats1.pop("path_mode", 0))
+    stats0.pop("path_mode", None)
     d = {k: stats1[k] - stats0[k] for k in stats1}
     K = args.steps
     rows_in = args.runs * args.rows  # per step per rank, by construction
@@ -233,11 +235,18 @@ def main():
     # output — i.e. exactly the whole-pipeline algorithmic bytes.
     # Legacy 3-kernel path: per-kernel models as in round 1.
     n_tiles = (rows_in + 3583) // 3584
-    # (the fused path records the scan/emit events back-to-back, so those
-    # legs are ~0; a 2% threshold separates the paths robustly)
-    if (kms["merge_ms"] > 0
-            and kms["emit_ms"] < 0.02 * kms["total_device_ms"]):
+    if path_mode == 1:
+        # fused in-kernel emission: one kernel moves the whole pipeline
         kernels = {"merge_emit": (kms["merge_ms"], in_bytes + out_bytes)}
+    elif path_mode == 2:
+        # split: A = merge staging + key/seq/kind emit + dense winners,
+        # B = value gather by winner
+        a_bytes = rows_in * 20 + U * (17 + 4)
+        b_bytes = U * (4 + 2 * (8 + args.vals * 4))
+        kernels = {
+            "merge_emit": (kms["merge_ms"], a_bytes),
+            "emit_dense": (kms["emit_ms"], b_bytes),
+        }
     else:
         merge_bytes = rows_in * 20 + U * 4 + n_tiles * 4
         emit_bytes = U * (4 + (28 + args.vals * 4) + (25 + args.vals * 4))

@@ -83,6 +83,8 @@ typedef struct pmh_stats {
     double emit_ms;          /* device time in gather/emit kernels */
     double total_device_ms;  /* end-to-end device time of read_next calls */
     double h2d_ms;           /* untimed-region staging cost, informational */
+    int64_t path_mode;       /* 0 = 3-kernel chain, 1 = fused in-kernel
+                              * emission, 2 = fused + split value emission */
 } pmh_stats;
 
 /* Session: owns the device + stream pool. device < 0 opens a host-only

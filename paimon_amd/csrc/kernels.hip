@@ -838,6 +838,50 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         if (tile >= n_tiles) return;
         const int32_t *c0 = &cuts[tile * k];
         const int32_t *c1 = &cuts[(tile + 1) * k];
+        int32_t my_lo = 0, my_hi = 0, my_off = 0;
+        int32_t C = 0;  // this tile's survivor count
+        int cur = 0;
+        uint16_t *wl = nullptr;
+        // one group-walk body for both passes (count, then emit): walks the
+        // thread's head range applying Deduplicate/FirstRow + wrapper +
+        // drop-delete rules, calling emit(n-th-winner, seg-index)
+        auto walk_pass = [&](int32_t lo_i, int32_t hi_i, auto emit) {
+            const uint16_t *mo = sm.perm[cur];
+            const int32_t M = sm.mtotal;
+            int32_t nloc = 0;
+            for (int32_t i = lo_i; i < hi_i; i++) {
+                if (!sm.head[i]) continue;
+                int32_t tail = i;
+                uint16_t s_best = mo[i];
+                int64_t v_best = sm.sseq[s_best];
+                bool e_best = !ignore_delete || ps2_isadd(v_best);
+                bool any_retract = !ps2_isadd(v_best);
+                while (tail + 1 < M && !sm.head[tail + 1]) {
+                    tail++;
+                    uint16_t s = mo[tail];
+                    int64_t v = sm.sseq[s];
+                    bool e = !ignore_delete || ps2_isadd(v);
+                    any_retract |= !ps2_isadd(v);
+                    bool take = (e && !e_best) ||
+                                (e == e_best &&
+                                 (FR ? v < v_best : v > v_best));
+                    if (take) {
+                        s_best = s;
+                        v_best = v;
+                        e_best = e;
+                    }
+                }
+                int32_t gsize = tail - i + 1;
+                if (FR && !ignore_delete && any_retract && gsize > 1 &&
+                    err_flag)
+                    atomicOr(err_flag, 2u);
+                if (!e_best && gsize > 1) continue;
+                if (drop_delete && !ps2_isadd(v_best)) continue;
+                emit(nloc, s_best);
+                nloc++;
+            }
+            return nloc;
+        };
 
         // --- segment setup (k_merge_tiles' protocol: +1 extra per run)
         if (tid < k) {
@@ -872,8 +916,6 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         __syncthreads();
         const int32_t M = sm.mtotal;
         const int32_t Mreal = sm.mreal;
-        int32_t C = 0;       // this tile's survivor count
-        int cur = 0;
         if (Mreal > 0) {
             // --- stage key / packed-seq segments (coalesced per run)
             for (int r = 0; r < k; r++) {
@@ -946,75 +988,41 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             }
             __syncthreads();
 
-            // --- winner walk (k_merge_tiles' dedup/first-row rules), but
-            // winners are SEG INDICES into the LDS arrays, compacted into
-            // the non-current perm plane in key order
-            uint16_t *wl = sm.perm[cur ^ 1];
+            // --- winner walk (k_merge_tiles' dedup/first-row rules).
+            // Pass 0 counts winners per thread; the global offset then
+            // arrives via lookback, and pass 1 (below, after the barrier)
+            // emits — in split mode STRAIGHT to the output arrays.
+            wl = sm.perm[cur ^ 1];
             const int32_t per =
                 (Mreal + (int32_t)blockDim.x - 1) / blockDim.x;
-            int32_t my_lo = tid * per;
-            int32_t my_hi = my_lo + per < Mreal ? my_lo + per : Mreal;
-            int32_t my_off = 0;
-            for (int pass = 0; pass < 2; pass++) {
-                int32_t nloc = 0;
-                for (int32_t i = my_lo; i < my_hi; i++) {
-                    if (!sm.head[i]) continue;
-                    int32_t tail = i;
-                    uint16_t s_best = mo[i];
-                    int64_t v_best = sm.sseq[s_best];
-                    bool e_best = !ignore_delete || ps2_isadd(v_best);
-                    bool any_retract = !ps2_isadd(v_best);
-                    while (tail + 1 < M && !sm.head[tail + 1]) {
-                        tail++;
-                        uint16_t s = mo[tail];
-                        int64_t v = sm.sseq[s];
-                        bool e = !ignore_delete || ps2_isadd(v);
-                        any_retract |= !ps2_isadd(v);
-                        bool take = (e && !e_best) ||
-                                    (e == e_best &&
-                                     (FR ? v < v_best : v > v_best));
-                        if (take) {
-                            s_best = s;
-                            v_best = v;
-                            e_best = e;
-                        }
-                    }
-                    int32_t gsize = tail - i + 1;
-                    if (FR && !ignore_delete && any_retract && gsize > 1 &&
-                        err_flag)
-                        atomicOr(err_flag, 2u);
-                    if (!e_best && gsize > 1) continue;
-                    if (drop_delete && !ps2_isadd(v_best)) continue;
-                    if (pass == 1) wl[my_off + nloc] = s_best;
-                    nloc++;
+            my_lo = tid * per;
+            my_hi = my_lo + per < Mreal ? my_lo + per : Mreal;
+            int32_t nloc = walk_pass(my_lo, my_hi, [](int32_t, uint16_t) {});
+            {
+                int32_t incl = nloc;
+                for (int off = 1; off < 64; off <<= 1) {
+                    int32_t up = __shfl_up(incl, off, 64);
+                    if (lane >= off) incl += up;
                 }
-                if (pass == 0) {
-                    int32_t incl = nloc;
-                    for (int off = 1; off < 64; off <<= 1) {
-                        int32_t up = __shfl_up(incl, off, 64);
-                        if (lane >= off) incl += up;
-                    }
-                    if (lane == 63) sm.wave_tot[wv] = incl;
-                    __syncthreads();
-                    int32_t add = 0;
-                    C = 0;
+                if (lane == 63) sm.wave_tot[wv] = incl;
+                __syncthreads();
+                int32_t add = 0;
+                C = 0;
 #pragma unroll
-                    for (int w = 0; w < NW; w++) {
-                        if (w < wv) add += sm.wave_tot[w];
-                        C += sm.wave_tot[w];
-                    }
-                    my_off = add + incl - nloc;
-                    // publish the aggregate NOW (pass 1 only writes the
-                    // winner list): successors' lookbacks unblock one
-                    // walk-pass earlier. The count rides in the packed
-                    // atomic word, so a successor that acquires the flag
-                    // also gets the payload — no separate fence.
-                    if (tid == 0 && tile > 0)
-                        __hip_atomic_store(&status[tile],
-                                           LOOK_AGG | (uint64_t)C,
-                                           __ATOMIC_RELEASE,
-                                           __HIP_MEMORY_SCOPE_AGENT);
+                for (int w = 0; w < NW; w++) {
+                    if (w < wv) add += sm.wave_tot[w];
+                    C += sm.wave_tot[w];
                 }
+                my_off = add + incl - nloc;
+                // publish the aggregate NOW (the emit pass only writes):
+                // successors' lookbacks unblock one pass earlier. The count
+                // rides in the packed atomic word, so a successor that
+                // acquires the flag also gets the payload — no extra fence.
+                if (tid == 0 && tile > 0)
+                    __hip_atomic_store(&status[tile],
+                                       LOOK_AGG | (uint64_t)C,
+                                       __ATOMIC_RELEASE,
+                                       __HIP_MEMORY_SCOPE_AGENT);
             }
         }
 
@@ -1073,7 +1081,44 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                                                  // emission
         if (ablate == 2) continue;
         const int64_t goff = sm.s_goff;
-        const uint16_t *wl = sm.perm[cur ^ 1];
+
+        if (dense_winners) {
+            // SPLIT mode: the emit pass writes key/seq/kind AND the packed
+            // winner straight to the output arrays at the global offset —
+            // no LDS winner list, no separate emission loops. Value columns
+            // follow in k_emit_dense (full occupancy, no LDS).
+            const int kdt = key_col >= 0 ? col_dtype[key_col] : 0;
+            walk_pass(my_lo, my_hi, [&](int32_t n, uint16_t s_best) {
+                const int64_t o = goff + my_off + n;
+                if (key_col >= 0) {
+                    int64_t v = sm.skey[s_best];
+                    switch (kdt) {
+                    case 1: ((int8_t *)out_ptrs[key_col])[o] =
+                                (int8_t)v; break;
+                    case 2: ((int16_t *)out_ptrs[key_col])[o] =
+                                (int16_t)v; break;
+                    case 3: ((int32_t *)out_ptrs[key_col])[o] =
+                                (int32_t)v; break;
+                    default: ((int64_t *)out_ptrs[key_col])[o] = v; break;
+                    }
+                }
+                const int64_t w = sm.sseq[s_best];
+                ((int64_t *)out_ptrs[seq_col])[o] = w >> 2;
+                ((int8_t *)out_ptrs[kind_col])[o] = (int8_t)ps2_kind(w);
+                int r = 0;
+                while (r + 1 < k && sm.segoff[r + 1] <= (int32_t)s_best) r++;
+                dense_winners[o] =
+                    ((uint32_t)r << PMH_ROW_BITS) |
+                    (uint32_t)(c0[r] + ((int32_t)s_best - sm.segoff[r]));
+            });
+            __syncthreads();
+            continue;
+        }
+        // non-split modes: emit pass fills the LDS winner list in key order
+        walk_pass(my_lo, my_hi, [&](int32_t n, uint16_t s_best) {
+            wl[my_off + n] = s_best;
+        });
+        __syncthreads();
 
         // per-thread slot map over the flat element space (computed once,
         // reused by every column): slot s covers element tid + s*T
@@ -1203,23 +1248,6 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             ((int8_t *)out_ptrs[kind_col])[goff + i] = (int8_t)ps2_kind(w);
         }
         if (cfirst < 0) {
-            __syncthreads();
-            continue;
-        }
-        if (dense_winners) {
-            // SPLIT mode: publish packed winners densely at the global
-            // offset; the value columns are emitted by k_emit_dense (no
-            // LDS, full occupancy) right after this kernel — the merge
-            // phases here cap residency at 2 workgroups/CU, which starves
-            // the gather phase of waves.
-            for (int32_t i = tid; i < C; i += blockDim.x) {
-                uint16_t s = wl[i];
-                int r = 0;
-                while (r + 1 < k && sm.segoff[r + 1] <= (int32_t)s) r++;
-                uint32_t grow =
-                    (uint32_t)(c0[r] + ((int32_t)s - sm.segoff[r]));
-                dense_winners[goff + i] = ((uint32_t)r << PMH_ROW_BITS) | grow;
-            }
             __syncthreads();
             continue;
         }

@@ -2023,6 +2023,7 @@ int pmh_plan_close(pmh_plan_t *p) {
 }
 
 int pmh_stats_get(pmh_plan_t *p, pmh_stats *out) {
+    if (p) p->stats.path_mode = !p->fused ? 0 : (p->fsplit ? 2 : 1);
     if (!p || !out) return -1;
     p->stats.hbm_bytes_algo = 0;  // filled by bench from encoded+output sizes
     *out = p->stats;

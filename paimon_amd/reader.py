@@ -46,7 +46,8 @@ class _Stats(ctypes.Structure):
                 ("merge_ms", ctypes.c_double), ("scan_ms", ctypes.c_double),
                 ("emit_ms", ctypes.c_double),
                 ("total_device_ms", ctypes.c_double),
-                ("h2d_ms", ctypes.c_double)]
+                ("h2d_ms", ctypes.c_double),
+                ("path_mode", ctypes.c_int64)]
 
 
 _lib = None
