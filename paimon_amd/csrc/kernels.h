@@ -123,7 +123,9 @@ hipError_t pmh_launch_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
 // then emit through the generic column path).
 hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  const DevCol *kinds, const int64_t *lens,
-                                 int k, const int32_t *cuts, int64_t n_tiles,
+                                 int k, const int32_t *cuts,
+                                 int64_t tile_base, int64_t tile_limit,
+                                 int64_t n_tiles,
                                  int64_t tile_rows, int flags,
                                  const DevCol *cols, const uint8_t *col_dtype,
                                  const uint8_t *col_nullable, int n_cols,
@@ -142,7 +144,8 @@ hipError_t pmh_launch_emit_dense(const DevCol *cols,
                                  const uint8_t *col_nullable, int n_cols,
                                  int key_col, int seq_col, int kind_col,
                                  const uint32_t *winners,
-                                 const int64_t *total_out,
+                                 int64_t t0, int64_t t1,
+                                 const uint64_t *status,
                                  void *const *out_ptrs,
                                  uint8_t *const *out_valid,
                                  hipStream_t stream);

@@ -814,7 +814,8 @@ template <bool FR>
 __launch_bounds__(PMH_TILE_THREADS, 2) __global__
 void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                   const int64_t *lens, int k, const int32_t *cuts,
-                  int64_t n_tiles, int64_t tile_rows, int flags,
+                  int64_t tile_base, int64_t tile_limit, int64_t n_tiles,
+                  int64_t tile_rows, int flags,
                   const DevCol *cols /* k * n_cols, run-major */,
                   const uint8_t *col_dtype, const uint8_t *col_nullable,
                   int n_cols, int key_col /* -1: composite */, int seq_col,
@@ -831,11 +832,11 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
     constexpr int NW = PMH_TILE_THREADS / 64;
     for (;;) {
         if (tid == 0)
-            sm.s_tile = (int64_t)__hip_atomic_fetch_add(
+            sm.s_tile = tile_base + (int64_t)__hip_atomic_fetch_add(
                 ticket, 1ull, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
         __syncthreads();
         const int64_t tile = sm.s_tile;
-        if (tile >= n_tiles) return;
+        if (tile >= tile_limit) return;
         const int32_t *c0 = &cuts[tile * k];
         const int32_t *c1 = &cuts[(tile + 1) * k];
         int32_t my_lo = 0, my_hi = 0, my_off = 0;
@@ -1082,39 +1083,9 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         if (ablate == 2) continue;
         const int64_t goff = sm.s_goff;
 
-        if (dense_winners) {
-            // SPLIT mode: the emit pass writes key/seq/kind AND the packed
-            // winner straight to the output arrays at the global offset —
-            // no LDS winner list, no separate emission loops. Value columns
-            // follow in k_emit_dense (full occupancy, no LDS).
-            const int kdt = key_col >= 0 ? col_dtype[key_col] : 0;
-            walk_pass(my_lo, my_hi, [&](int32_t n, uint16_t s_best) {
-                const int64_t o = goff + my_off + n;
-                if (key_col >= 0) {
-                    int64_t v = sm.skey[s_best];
-                    switch (kdt) {
-                    case 1: ((int8_t *)out_ptrs[key_col])[o] =
-                                (int8_t)v; break;
-                    case 2: ((int16_t *)out_ptrs[key_col])[o] =
-                                (int16_t)v; break;
-                    case 3: ((int32_t *)out_ptrs[key_col])[o] =
-                                (int32_t)v; break;
-                    default: ((int64_t *)out_ptrs[key_col])[o] = v; break;
-                    }
-                }
-                const int64_t w = sm.sseq[s_best];
-                ((int64_t *)out_ptrs[seq_col])[o] = w >> 2;
-                ((int8_t *)out_ptrs[kind_col])[o] = (int8_t)ps2_kind(w);
-                int r = 0;
-                while (r + 1 < k && sm.segoff[r + 1] <= (int32_t)s_best) r++;
-                dense_winners[o] =
-                    ((uint32_t)r << PMH_ROW_BITS) |
-                    (uint32_t)(c0[r] + ((int32_t)s_best - sm.segoff[r]));
-            });
-            __syncthreads();
-            continue;
-        }
-        // non-split modes: emit pass fills the LDS winner list in key order
+        // emit pass fills the LDS winner list in key order (direct global
+        // writes from the walk measured SLOWER: per-thread ranges scatter,
+        // while wl-indexed loops below stream perfectly coalesced)
         walk_pass(my_lo, my_hi, [&](int32_t n, uint16_t s_best) {
             wl[my_off + n] = s_best;
         });
@@ -1246,6 +1217,21 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             int64_t w = sm.sseq[wl[i]];
             ((int64_t *)out_ptrs[seq_col])[goff + i] = w >> 2;
             ((int8_t *)out_ptrs[kind_col])[goff + i] = (int8_t)ps2_kind(w);
+        }
+        if (dense_winners) {
+            // SPLIT mode: publish packed winners densely; value columns are
+            // emitted by k_emit_dense (full occupancy, no LDS) right after
+            // this kernel
+            for (int32_t i = tid; i < C; i += blockDim.x) {
+                uint16_t s = wl[i];
+                int r = 0;
+                while (r + 1 < k && sm.segoff[r + 1] <= (int32_t)s) r++;
+                dense_winners[goff + i] =
+                    ((uint32_t)r << PMH_ROW_BITS) |
+                    (uint32_t)(c0[r] + ((int32_t)s - sm.segoff[r]));
+            }
+            __syncthreads();
+            continue;
         }
         if (cfirst < 0) {
             __syncthreads();
@@ -1382,14 +1368,21 @@ __global__ void k_emit_dense(const DevCol *cols, const uint8_t *col_dtype,
                              const uint8_t *col_nullable, int n_cols,
                              int key_col, int seq_col, int kind_col,
                              const uint32_t *winners,
-                             const int64_t *total_out,
+                             int64_t t0, int64_t t1,
+                             const uint64_t *status,
                              void *const *out_ptrs,
                              uint8_t *const *out_valid) {
+    // output range of tile chunk [t0, t1): the inclusive lookback prefixes
+    // of t0-1 / t1-1 (both PREFIX-published — the chunk's merge pass
+    // completed before this launch was released by its event)
     constexpr int R = 4;
-    const int64_t total = *total_out;
+    const int64_t lo =
+        t0 > 0 ? (int64_t)(status[t0 - 1] & LOOK_VAL) : 0;
+    const int64_t total = (int64_t)(status[t1 - 1] & LOOK_VAL);
+    const int64_t span = total - lo;
     const int64_t per_block =
-        (total + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
-    const int64_t slice_lo = (int64_t)blockIdx.x * per_block;
+        (span + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
+    const int64_t slice_lo = lo + (int64_t)blockIdx.x * per_block;
     const int64_t slice_hi =
         slice_lo + per_block < total ? slice_lo + per_block : total;
     for (int64_t base = slice_lo; base < slice_hi;
@@ -2586,7 +2579,9 @@ hipError_t pmh_launch_level_scatter(const RleChunk *chunks, int64_t n_chunks,
 
 hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  const DevCol *kinds, const int64_t *lens,
-                                 int k, const int32_t *cuts, int64_t n_tiles,
+                                 int k, const int32_t *cuts,
+                                 int64_t tile_base, int64_t tile_limit,
+                                 int64_t n_tiles,
                                  int64_t tile_rows, int flags,
                                  const DevCol *cols, const uint8_t *col_dtype,
                                  const uint8_t *col_nullable, int n_cols,
@@ -2597,13 +2592,16 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  uint8_t *const *out_valid,
                                  uint32_t *err_flag, hipStream_t stream) {
     // persistent workgroups: 2 resident per CU (LDS-bound) x 256 CUs; the
-    // ticket hands out tiles in order, so any residency is safe
-    int blocks = n_tiles < 512 ? (int)n_tiles : 512;
+    // ticket (zeroed before this launch) hands out the chunk's tiles in
+    // order, so any residency is safe
+    const int64_t chunk = tile_limit - tile_base;
+    int blocks = chunk < 512 ? (int)chunk : 512;
     const bool fr = flags & 8;
     auto launch = [&](auto kern) {
         hipLaunchKernelGGL(kern, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
-                           stream, keys, seqs, kinds, lens, k, cuts, n_tiles,
-                           tile_rows, flags, cols, col_dtype, col_nullable,
+                           stream, keys, seqs, kinds, lens, k, cuts,
+                           tile_base, tile_limit, n_tiles, tile_rows, flags,
+                           cols, col_dtype, col_nullable,
                            n_cols, key_col, seq_col, kind_col, status, ticket,
                            total_out, dense_winners, out_ptrs, out_valid,
                            err_flag);
@@ -2620,13 +2618,15 @@ hipError_t pmh_launch_emit_dense(const DevCol *cols,
                                  const uint8_t *col_nullable, int n_cols,
                                  int key_col, int seq_col, int kind_col,
                                  const uint32_t *winners,
-                                 const int64_t *total_out,
+                                 int64_t t0, int64_t t1,
+                                 const uint64_t *status,
                                  void *const *out_ptrs,
                                  uint8_t *const *out_valid,
                                  hipStream_t stream) {
     hipLaunchKernelGGL(k_emit_dense, dim3(2048), dim3(256), 0, stream, cols,
                        col_dtype, col_nullable, n_cols, key_col, seq_col,
-                       kind_col, winners, total_out, out_ptrs, out_valid);
+                       kind_col, winners, t0, t1, status, out_ptrs,
+                       out_valid);
     return hipGetLastError();
 }
 

@@ -248,6 +248,7 @@ struct pmh_session_t {
 struct pmh_plan_t {
     pmh_session_t *session = nullptr;
     hipStream_t stream = nullptr;
+    hipStream_t stream_b = nullptr;  // split-mode value emission overlap
     pmh::DeviceBufs bufs;
     std::vector<pmh::ColSpec> cols;  // key..., seq, kind, value...
     int n_key_cols = 1;
@@ -1598,6 +1599,11 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             set_error("hipStreamCreate failed");
             return nullptr;
         }
+        if (plan->fsplit &&
+            hipStreamCreate(&plan->stream_b) != hipSuccess) {
+            set_error("hipStreamCreate (b) failed");
+            return nullptr;
+        }
 
         auto sections = interval_partition(files);
         auto t0 = std::chrono::steady_clock::now();
@@ -1852,24 +1858,68 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         e = hipMemsetAsync(sec.ticket, 0, 8, st);
         if (e != hipSuccess) return fail("ticket memset", e);
         int key_col = p->composite_key ? -1 : 0;
-        e = pmh_launch_merge_emit(
-            sec.key_cols, sec.seq_cols, sec.kind_cols, sec.lens_dev, k,
-            sec.cuts, sec.n_tiles, PMH_TILE_ROWS, flags, sec.all_cols,
-            p->col_dtype_dev, p->col_nullable_dev, n_cols, key_col,
-            p->n_key_cols, p->n_key_cols + 1, sec.status, sec.ticket,
-            sec.total_dev, sec.dense_winners, p->out_ptrs_dev,
-            p->out_valid_dev, sec.err_dev, st);
-        if (e != hipSuccess) return fail("merge_emit", e);
-        (void)hipEventRecord(ev[3], st);
-        (void)hipEventRecord(ev[4], st);
-        if (sec.dense_winners) {
-            e = pmh_launch_emit_dense(
-                sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols,
-                key_col, p->n_key_cols, p->n_key_cols + 1, sec.dense_winners,
-                sec.total_dev, p->out_ptrs_dev, p->out_valid_dev, st);
-            if (e != hipSuccess) return fail("emit_dense", e);
+        {
+            // split mode: chunk the tile space and pipeline — kernel A
+            // (merge + key/seq/kind + dense winners) for chunk i+1 runs on
+            // the plan stream while kernel B (value gather) for chunk i
+            // runs on stream_b, gated by a per-chunk event. Non-split: one
+            // A launch does everything.
+            const int64_t want =
+                sec.dense_winners ? (sec.n_tiles + 4095) / 4096 : 1;
+            const int64_t n_chunks =
+                want < 1 ? 1 : (want > 8 ? 8 : want);
+            const int64_t per =
+                (sec.n_tiles + n_chunks - 1) / n_chunks;
+            hipEvent_t cev[8];
+            for (int64_t c = 0; c < n_chunks; c++)
+                (void)hipEventCreateWithFlags(&cev[c],
+                                              hipEventDisableTiming);
+            for (int64_t c = 0; c < n_chunks; c++) {
+                int64_t t0 = c * per;
+                int64_t t1 = t0 + per < sec.n_tiles ? t0 + per : sec.n_tiles;
+                if (t0 >= t1) { (void)hipEventRecord(cev[c], st); continue; }
+                if (c > 0) {  // ticket restarts per chunk
+                    e = hipMemsetAsync(sec.ticket, 0, 8, st);
+                    if (e != hipSuccess) return fail("ticket memset", e);
+                }
+                e = pmh_launch_merge_emit(
+                    sec.key_cols, sec.seq_cols, sec.kind_cols, sec.lens_dev,
+                    k, sec.cuts, t0, t1, sec.n_tiles, PMH_TILE_ROWS, flags,
+                    sec.all_cols, p->col_dtype_dev, p->col_nullable_dev,
+                    n_cols, key_col, p->n_key_cols, p->n_key_cols + 1,
+                    sec.status, sec.ticket, sec.total_dev, sec.dense_winners,
+                    p->out_ptrs_dev, p->out_valid_dev, sec.err_dev, st);
+                if (e != hipSuccess) return fail("merge_emit", e);
+                (void)hipEventRecord(cev[c], st);
+            }
+            (void)hipEventRecord(ev[3], st);
+            (void)hipEventRecord(ev[4], st);
+            if (sec.dense_winners) {
+                hipStream_t sb = p->stream_b;
+                for (int64_t c = 0; c < n_chunks; c++) {
+                    int64_t t0 = c * per;
+                    int64_t t1 =
+                        t0 + per < sec.n_tiles ? t0 + per : sec.n_tiles;
+                    if (t0 >= t1) continue;
+                    (void)hipStreamWaitEvent(sb, cev[c], 0);
+                    e = pmh_launch_emit_dense(
+                        sec.all_cols, p->col_dtype_dev, p->col_nullable_dev,
+                        n_cols, key_col, p->n_key_cols, p->n_key_cols + 1,
+                        sec.dense_winners, t0, t1, sec.status,
+                        p->out_ptrs_dev, p->out_valid_dev, sb);
+                    if (e != hipSuccess) return fail("emit_dense", e);
+                }
+                // rejoin: the plan stream waits for the last B chunk
+                hipEvent_t done;
+                (void)hipEventCreateWithFlags(&done, hipEventDisableTiming);
+                (void)hipEventRecord(done, sb);
+                (void)hipStreamWaitEvent(st, done, 0);
+                (void)hipEventDestroy(done);
+            }
+            (void)hipEventRecord(ev[5], st);
+            for (int64_t c = 0; c < n_chunks; c++)
+                (void)hipEventDestroy(cev[c]);
         }
-        (void)hipEventRecord(ev[5], st);
         goto collect;
     }
     e = pmh_launch_merge_tiles(sec.key_cols, sec.seq_cols, sec.kind_cols,
@@ -2018,6 +2068,7 @@ int pmh_plan_reset(pmh_plan_t *p) {
 int pmh_plan_close(pmh_plan_t *p) {
     if (!p) return 0;
     if (p->stream) (void)hipStreamDestroy(p->stream);
+    if (p->stream_b) (void)hipStreamDestroy(p->stream_b);
     delete p;
     return 0;
 }
