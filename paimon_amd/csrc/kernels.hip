@@ -810,7 +810,7 @@ constexpr uint64_t LOOK_AGG = 1ull << 62;
 constexpr uint64_t LOOK_PREFIX = 2ull << 62;
 constexpr uint64_t LOOK_VAL = (1ull << 62) - 1;
 
-template <bool FR, int KM>
+template <bool FR>
 __launch_bounds__(PMH_TILE_THREADS, 2) __global__
 void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                   const int64_t *lens, int k, const int32_t *cuts,
@@ -931,65 +931,53 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                 for (int32_t i = tid; i < len; i += blockDim.x) {
                     sm.skey[off + i] = key_at(kaddr0, i, kes);
                     sm.sseq[off + i] = ps2_pack(saddr[i], daddr[i]);
+                    sm.perm[0][off + i] = (uint16_t)(off + i);
                 }
             }
             __syncthreads();
 
-            // --- single-pass k-way rank merge: element e's merged rank =
-            // its own segment offset + the count of elements before it in
-            // every OTHER run's segment (ties by run order: upper bound
-            // for lower-numbered runs, lower bound for higher) — the k-1
-            // binary searches run LOCKSTEP so their probe loads issue
-            // together each round. Replaces the ceil(log2 k)-level pairwise
-            // merge: no barriers, no corank, no perm ping-pong (and no
-            // perm[0] identity init). Ranks are a permutation, so the
-            // scatter write has no conflicts.
-            cur = 1;
-            {
-                uint16_t *mo_out = sm.perm[1];
-                for (int32_t e = tid; e < M; e += blockDim.x) {
-                    int r = 0;
-                    while (r + 1 < k && sm.segoff[r + 1] <= e) r++;
-                    const int64_t ke = sm.skey[e];
-                    int32_t rank = e - sm.segoff[r];
-                    int32_t blo[KM], bhi[KM];
-#pragma unroll
-                    for (int r2 = 0; r2 < KM; r2++) {
-                        if (r2 >= k || r2 == r) {
-                            blo[r2] = 0;
-                            bhi[r2] = 0;
-                        } else {
-                            blo[r2] = sm.segoff[r2];
-                            bhi[r2] = sm.segoff[r2] + sm.seglen[r2];
+            // --- pairwise stable merge (identical to k_merge_tiles)
+            for (int width = 1; width < k; width <<= 1) {
+                const int nxt = cur ^ 1;
+                const int CH = 8;
+                int n_chunks = (M + CH - 1) / CH;
+                for (int ch = tid; ch < n_chunks; ch += blockDim.x) {
+                    int64_t o = (int64_t)ch * CH;
+                    int remaining = (int)(M - o < CH ? M - o : CH);
+                    int p = 0;
+                    while (remaining > 0) {
+                        while ((p + 1) * 2 * width < k &&
+                               sm.segoff[(p + 1) * 2 * width] <= o)
+                            p++;
+                        int a0 = p * 2 * width;
+                        int amid = a0 + width < k ? a0 + width : k;
+                        int b1 = a0 + 2 * width < k ? a0 + 2 * width : k;
+                        int32_t abase = sm.segoff[a0];
+                        int32_t la = sm.segoff[amid] - abase;
+                        int32_t lb = sm.segoff[b1] - sm.segoff[amid];
+                        int64_t d = o - abase;
+                        int32_t lim = la + lb - (int32_t)d;
+                        if (lim <= 0) break;
+                        int n_out = lim < remaining ? lim : remaining;
+                        const uint16_t *pa = &sm.perm[cur][abase];
+                        const uint16_t *pb = &sm.perm[cur][abase + la];
+                        int32_t ai = corank(d, pa, la, pb, lb, sm.skey);
+                        int32_t bi = (int32_t)d - ai;
+                        uint16_t *out = &sm.perm[nxt][abase + d];
+                        for (int x = 0; x < n_out; x++) {
+                            bool takeA;
+                            if (ai >= la) takeA = false;
+                            else if (bi >= lb) takeA = true;
+                            else takeA = !(sm.skey[pa[ai]] > sm.skey[pb[bi]]);
+                            out[x] = takeA ? pa[ai++] : pb[bi++];
                         }
+                        o += n_out;
+                        remaining -= n_out;
                     }
-                    bool any = true;
-                    while (any) {
-                        any = false;
-#pragma unroll
-                        for (int r2 = 0; r2 < KM; r2++) {
-                            if (r2 >= k) continue;
-                            if (blo[r2] < bhi[r2]) {
-                                int32_t mid =
-                                    blo[r2] + ((bhi[r2] - blo[r2]) >> 1);
-                                int64_t km2 = sm.skey[mid];
-                                bool go = (km2 < ke) ||
-                                          (km2 == ke && r2 < r);
-                                if (go) blo[r2] = mid + 1;
-                                else bhi[r2] = mid;
-                                any |= blo[r2] < bhi[r2];
-                            }
-                        }
-                    }
-#pragma unroll
-                    for (int r2 = 0; r2 < KM; r2++) {
-                        if (r2 >= k || r2 == r) continue;
-                        rank += blo[r2] - sm.segoff[r2];
-                    }
-                    mo_out[rank] = (uint16_t)e;
                 }
+                cur = nxt;
+                __syncthreads();
             }
-            __syncthreads();
             const uint16_t *mo = sm.perm[cur];
 
             // --- group heads + previous-tile continuation skip
@@ -2621,13 +2609,10 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                            total_out, dense_winners, out_ptrs, out_valid,
                            err_flag);
     };
-    if (k <= 8) {
-        if (fr) launch(k_merge_emit<true, 8>);
-        else launch(k_merge_emit<false, 8>);
-    } else {
-        if (fr) launch(k_merge_emit<true, PMH_MAX_RUNS>);
-        else launch(k_merge_emit<false, PMH_MAX_RUNS>);
-    }
+    if (fr)
+        launch(k_merge_emit<true>);
+    else
+        launch(k_merge_emit<false>);
     return hipGetLastError();
 }
 
