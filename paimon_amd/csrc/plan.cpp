@@ -1562,8 +1562,10 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             // winner engines; PMH_FUSED=0 keeps the 3-kernel chain for A/B
             const char *pf = getenv("PMH_FUSED");
             plan->fused = !plan->pu && !(pf && pf[0] == '0');
+            // split emission is the default product path (fastest
+            // measured); PMH_FSPLIT=0 keeps in-kernel emission for A/B
             const char *fs = getenv("PMH_FSPLIT");
-            plan->fsplit = plan->fused && fs && fs[0] == '1';
+            plan->fsplit = plan->fused && !(fs && fs[0] == '0');
         }
         if (plan->rrod && plan->ignore_delete) {
             set_error("remove-record-on-delete cannot be used with "
