@@ -55,14 +55,32 @@ enum pmh_dtype {
     PMH_DT_INT64 = 4,
     PMH_DT_FLOAT32 = 5,
     PMH_DT_FLOAT64 = 6,
+    /* Dictionary-backed string column (CHAR/VARCHAR -> parquet BYTE_ARRAY,
+     * ParquetSchemaConverter.java:120-128): data = int32 ids into the
+     * column's dictionary (dict_data/dict_offsets/dict_len) — the same
+     * shape the reference's WritableColumnVector dictionary support and
+     * parquet's own dictionary encoding use. DECIMAL(p,s) is not a
+     * separate dtype: p <= 9 rides INT32, p <= 18 INT64 (unscaled values,
+     * ParquetSchemaConverter.java:153-171) with precision/scale set. */
+    PMH_DT_STRING = 7,
 };
 
 typedef struct pmh_col {
     const char *name;
     int32_t dtype;         /* enum pmh_dtype */
-    const void *data;      /* columnar values; device ptr unless host batch */
+    const void *data;      /* columnar values (PMH_DT_STRING: int32 ids);
+                              device ptr unless host batch */
     const uint8_t *valid;  /* byte-per-row validity (1=non-null); NULL = all
                               valid. (Arrow bitmap export: later round.) */
+    /* PMH_DT_STRING: the column dictionary (HOST-resident even for device
+     * batches — it is plan-level metadata built at staging). Entry i =
+     * dict_data[dict_offsets[i] .. dict_offsets[i+1]). */
+    const void *dict_data;
+    const int32_t *dict_offsets;  /* dict_len + 1 entries */
+    int32_t dict_len;
+    /* DECIMAL annotation on INT32/INT64 columns (0 = plain integer). */
+    int32_t precision;
+    int32_t scale;
 } pmh_col;
 
 typedef struct pmh_batch {

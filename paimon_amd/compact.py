@@ -51,6 +51,15 @@ def rewrite(session: Session, file_metas, key_cols, value_cols, out_dir,
     # value cols (KeyValueSerializer.java:34-99)
     key_names = [kc["name"] for kc in key_cols]
     special = set(key_names) | {"_SEQUENCE_NUMBER", "_VALUE_KIND"}
+    # DECIMAL annotations round-trip from the declared read types
+    # (ParquetSchemaConverter.java:153-171); dictionary strings write back
+    # with the batch's global dictionary (ids + dict page)
+    decimals = {}
+    for vc in list(key_cols) + list(value_cols):
+        t = vc.get("type", "")
+        if t.startswith("decimal(") and t.endswith(")"):
+            p, s = t[8:-1].split(",")
+            decimals[vc["name"]] = (int(p), int(s))
     after = []
     file_idx = 0
     try:
@@ -67,15 +76,21 @@ def rewrite(session: Session, file_metas, key_cols, value_cols, out_dir,
                 cols = [(kn, batch[kn][s:e]) for kn in key_names]
                 cols += [("_SEQUENCE_NUMBER", seq[s:e]),
                          ("_VALUE_KIND", kind[s:e])]
+                dicts = {}
                 for name, arr in batch.items():
-                    if name in special or name.endswith("#valid"):
+                    if name in special or name.endswith("#valid") \
+                            or name.endswith("#dict"):
                         continue
                     valid = batch.get(name + "#valid")
+                    d = batch.get(name + "#dict")
+                    if d is not None:
+                        dicts[name] = d
                     cols.append((name, arr[s:e],
                                  valid[s:e] if valid is not None else None))
                 path = os.path.join(out_dir, f"{file_prefix}-{file_idx}.parquet")
                 file_idx += 1
-                write_parquet(path, cols, compression=compression)
+                write_parquet(path, cols, compression=compression,
+                              dicts=dicts, decimals=decimals)
                 sq = seq[s:e]
                 kd = kind[s:e]
                 # minKey/maxKey are the first/last merged rows' full key

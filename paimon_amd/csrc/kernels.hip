@@ -1171,7 +1171,8 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                 case 2: ((int16_t *)out_ptrs[c])[goff + i] =
                             (int16_t)v; break;
                 case 3:
-                case 5: ((int32_t *)out_ptrs[c])[goff + i] =
+                case 5:
+                case 7: ((int32_t *)out_ptrs[c])[goff + i] =
                             (int32_t)v; break;
                 default: ((int64_t *)out_ptrs[c])[goff + i] = v; break;
                 }
@@ -1590,7 +1591,8 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                 break;
             }
             case 3:
-            case 5: {
+            case 5:
+            case 7: {  // string ids ride as int32
                 int32_t v[R];
 #pragma unroll
                 for (int x = 0; x < R; x++)
@@ -1855,6 +1857,7 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                 break;
             case 3:
             case 5:
+            case 7:  // string ids ride as int32
                 ((int32_t *)out_ptrs[c])[i] =
                     ok ? col_load<int32_t>(dc, row) : 0;
                 break;
@@ -2175,7 +2178,8 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
             case 1: ((int8_t *)out_ptrs[c])[i] = (int8_t)bits; break;
             case 2: ((int16_t *)out_ptrs[c])[i] = (int16_t)bits; break;
             case 3:
-            case 5: ((int32_t *)out_ptrs[c])[i] = (int32_t)bits; break;
+            case 5:
+            case 7: ((int32_t *)out_ptrs[c])[i] = (int32_t)bits; break;
             case 4:
             case 6: ((int64_t *)out_ptrs[c])[i] = bits; break;
             default: break;

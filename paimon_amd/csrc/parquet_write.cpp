@@ -80,16 +80,69 @@ int phys_of(int dtype) {
     case 4: return PHYS_INT64;
     case 5: return PHYS_FLOAT;
     case 6: return PHYS_DOUBLE;
+    case 7: return PHYS_BYTE_ARRAY;  // dictionary string
     }
     return -1;
 }
 
-int converted_of(int dtype) {  // ConvertedType; -1 = none
-    switch (dtype) {
+int converted_of(const PwCol &c) {  // ConvertedType; -1 = none
+    if (c.precision > 0) return 5;  // DECIMAL (INT32/INT64 physical, as
+                                    // ParquetSchemaConverter.java:153-171)
+    switch (c.dtype) {
     case 1: return 15;  // INT_8
     case 2: return 16;  // INT_16
+    case 7: return 0;   // UTF8
     }
     return -1;
+}
+
+// RLE/bit-packed hybrid id stream for RLE_DICTIONARY pages: a leading
+// bit-width byte, then bit-packed groups of 8 (the format
+// VectorizedRleValuesReader.java:977-1018 consumes; nulls skipped)
+void encode_dict_ids(const PwCol &c, int64_t s, int64_t e,
+                     std::string &out) {
+    int bw = 1;
+    while ((1 << bw) < c.dict_len) bw++;
+    std::vector<int32_t> ids;
+    ids.reserve(e - s);
+    const int32_t *d = (const int32_t *)c.data;
+    for (int64_t i = s; i < e; i++) {
+        if (c.valid && !c.valid[i]) continue;
+        ids.push_back(d[i]);
+    }
+    out.push_back((char)bw);
+    const int64_t ngroups = ((int64_t)ids.size() + 7) / 8;
+    TC t;
+    t.uvarint(((uint64_t)ngroups << 1) | 1);
+    out.append(t.out);
+    for (int64_t g = 0; g < ngroups; g++) {
+        uint64_t word = 0;
+        for (int j = 0; j < 8; j++) {
+            int64_t i = g * 8 + j;
+            uint64_t v = i < (int64_t)ids.size() ? (uint64_t)ids[i] : 0;
+            word |= v << (j * bw);
+        }
+        for (int b = 0; b < bw; b++)
+            out.push_back((char)((word >> (8 * b)) & 0xFF));
+    }
+}
+
+void dict_page_header(int64_t n_vals, int32_t unc, int32_t comp,
+                      std::string &out) {
+    TC t;
+    int l0 = 0;
+    t.i32(l0, 1, 2);     // type = DICTIONARY_PAGE
+    t.i32(l0, 2, unc);
+    t.i32(l0, 3, comp);
+    t.field(l0, 7, 12);  // dictionary_page_header
+    {
+        int l1 = 0;
+        t.i32(l1, 1, (int32_t)n_vals);
+        t.i32(l1, 2, ENC_PLAIN);
+        t.stop();
+    }
+    t.stop();
+    out.append(t.out);
 }
 
 // RLE/bit-packed hybrid def levels (bit width 1), with the 4-byte LE length
@@ -153,7 +206,7 @@ void encode_values(const PwCol &c, int64_t s, int64_t e, std::string &out) {
     }
 }
 
-void page_header(int64_t n_vals, int32_t unc, int32_t comp,
+void page_header(int64_t n_vals, int32_t unc, int32_t comp, int encoding,
                  std::string &out) {
     TC t;
     int l0 = 0;
@@ -164,7 +217,7 @@ void page_header(int64_t n_vals, int32_t unc, int32_t comp,
     {
         int l1 = 0;
         t.i32(l1, 1, (int32_t)n_vals);
-        t.i32(l1, 2, ENC_PLAIN);  // encoding
+        t.i32(l1, 2, encoding);
         t.i32(l1, 3, ENC_RLE);    // definition_level_encoding
         t.i32(l1, 4, ENC_RLE);    // repetition_level_encoding
         t.stop();
@@ -183,12 +236,19 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
               " not supported (v1: UNCOMPRESSED, ZSTD)";
         return false;
     }
-    for (const auto &c : cols)
+    for (const auto &c : cols) {
         if (phys_of(c.dtype) < 0) {
             err = "unsupported dtype " + std::to_string(c.dtype) +
-                  " for parquet write (v1 matrix: int8..int64/float/double)";
+                  " for parquet write (v1 matrix: int8..int64/float/double/"
+                  "decimal<=18/dictionary string)";
             return false;
         }
+        if (c.dtype == 7 &&
+            (!c.dict_data || !c.dict_offsets || c.dict_len <= 0)) {
+            err = "string column '" + c.name + "' needs a dictionary";
+            return false;
+        }
+    }
     if (row_group_rows <= 0) row_group_rows = 1 << 20;
     if (page_rows <= 0) page_rows = 1 << 16;
     FILE *f = fopen(path.c_str(), "wb");
@@ -200,6 +260,7 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
 
     struct CcInfo {
         int64_t data_page_offset;
+        int64_t dict_page_offset;  // 0 = no dictionary page
         int64_t total_size;       // compressed (on-file) bytes
         int64_t total_unc_size;   // uncompressed payload + headers
         int64_t num_values;
@@ -219,15 +280,53 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
         rg.rows = rg1 - rg0;
         for (const auto &c : cols) {
             CcInfo cc{};
-            cc.data_page_offset = (int64_t)buf.size();
             cc.num_values = rg1 - rg0;
             int64_t unc_total = 0;
+            const bool dict = c.dtype == 7;
+            const int enc = dict ? ENC_RLE_DICTIONARY : ENC_PLAIN;
+            if (dict) {
+                // dictionary page: PLAIN byte-array entries ([len][bytes])
+                cc.dict_page_offset = (int64_t)buf.size();
+                std::string dp;
+                for (int32_t i = 0; i < c.dict_len; i++) {
+                    uint32_t len =
+                        (uint32_t)(c.dict_offsets[i + 1] - c.dict_offsets[i]);
+                    dp.append((const char *)&len, 4);
+                    dp.append(
+                        (const char *)c.dict_data + c.dict_offsets[i], len);
+                }
+                // dictionary pages compress with the chunk codec, like
+                // data pages (parquet spec)
+                if (codec == CODEC_ZSTD) {
+                    std::vector<uint8_t> compd;
+                    if (!zstd_compress_buf((const uint8_t *)dp.data(),
+                                           dp.size(), compd, err)) {
+                        fclose(f);
+                        return false;
+                    }
+                    size_t h0 = buf.size();
+                    dict_page_header(c.dict_len, (int32_t)dp.size(),
+                                     (int32_t)compd.size(), buf);
+                    unc_total +=
+                        (int64_t)(buf.size() - h0) + (int64_t)dp.size();
+                    buf.append((const char *)compd.data(), compd.size());
+                } else {
+                    size_t h0 = buf.size();
+                    dict_page_header(c.dict_len, (int32_t)dp.size(),
+                                     (int32_t)dp.size(), buf);
+                    unc_total +=
+                        (int64_t)(buf.size() - h0) + (int64_t)dp.size();
+                    buf.append(dp);
+                }
+            }
+            cc.data_page_offset = (int64_t)buf.size();
             for (int64_t p0 = rg0; p0 < rg1 || (rg1 == rg0 && p0 == rg0);
                  p0 += page_rows) {
                 int64_t p1 = p0 + page_rows < rg1 ? p0 + page_rows : rg1;
                 std::string payload;
                 if (c.valid) encode_def_levels(c.valid + p0, p1 - p0, payload);
-                encode_values(c, p0, p1, payload);
+                if (dict) encode_dict_ids(c, p0, p1, payload);
+                else encode_values(c, p0, p1, payload);
                 if (codec == CODEC_ZSTD) {
                     std::vector<uint8_t> comp;
                     if (!zstd_compress_buf((const uint8_t *)payload.data(),
@@ -237,21 +336,23 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
                     }
                     size_t h0 = buf.size();
                     page_header(p1 - p0, (int32_t)payload.size(),
-                                (int32_t)comp.size(), buf);
+                                (int32_t)comp.size(), enc, buf);
                     unc_total +=
                         (int64_t)(buf.size() - h0) + (int64_t)payload.size();
                     buf.append((const char *)comp.data(), comp.size());
                 } else {
                     size_t h0 = buf.size();
                     page_header(p1 - p0, (int32_t)payload.size(),
-                                (int32_t)payload.size(), buf);
+                                (int32_t)payload.size(), enc, buf);
                     unc_total +=
                         (int64_t)(buf.size() - h0) + (int64_t)payload.size();
                     buf.append(payload);
                 }
                 if (rg1 == rg0) break;  // single empty page for 0 rows
             }
-            cc.total_size = (int64_t)buf.size() - cc.data_page_offset;
+            cc.total_size = (int64_t)buf.size() -
+                            (dict ? cc.dict_page_offset
+                                  : cc.data_page_offset);
             cc.total_unc_size = unc_total;
             rg.bytes += cc.total_size;
             rg.ccs.push_back(cc);
@@ -277,8 +378,12 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
             g.i32(lf, 1, phys_of(c.dtype));
             g.i32(lf, 3, c.valid ? 1 : 0);  // OPTIONAL : REQUIRED
             g.str(lf, 4, c.name);
-            int ct = converted_of(c.dtype);
+            int ct = converted_of(c);
             if (ct >= 0) g.i32(lf, 6, ct);
+            if (c.precision > 0) {
+                g.i32(lf, 7, c.scale);
+                g.i32(lf, 8, c.precision);
+            }
             g.stop();
         }
     }
@@ -291,14 +396,22 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
             const CcInfo &cc = rg.ccs[ci];
             const PwCol &c = cols[ci];
             int lc = 0;
-            t.i64(lc, 2, cc.data_page_offset);  // file_offset
+            t.i64(lc, 2, cc.dict_page_offset ? cc.dict_page_offset
+                                              : cc.data_page_offset);
             t.field(lc, 3, 12);                 // meta_data: struct
             {
                 int lm = 0;
                 t.i32(lm, 1, phys_of(c.dtype));
-                t.list_begin(lm, 2, 5, 2);  // encodings
-                t.uvarint(TC::zz(ENC_PLAIN));
-                t.uvarint(TC::zz(ENC_RLE));
+                if (c.dtype == 7) {
+                    t.list_begin(lm, 2, 5, 3);  // encodings
+                    t.uvarint(TC::zz(ENC_PLAIN));
+                    t.uvarint(TC::zz(ENC_RLE));
+                    t.uvarint(TC::zz(ENC_RLE_DICTIONARY));
+                } else {
+                    t.list_begin(lm, 2, 5, 2);  // encodings
+                    t.uvarint(TC::zz(ENC_PLAIN));
+                    t.uvarint(TC::zz(ENC_RLE));
+                }
                 t.list_begin(lm, 3, 8, 1);  // path_in_schema
                 t.uvarint(c.name.size());
                 t.out.append(c.name);
@@ -307,6 +420,8 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
                 t.i64(lm, 6, cc.total_unc_size);  // total_uncompressed_size
                 t.i64(lm, 7, cc.total_size);      // total_compressed_size
                 t.i64(lm, 9, cc.data_page_offset);
+                if (cc.dict_page_offset)
+                    t.i64(lm, 11, cc.dict_page_offset);
                 t.stop();
             }
             t.stop();

@@ -11,8 +11,16 @@ namespace pmh {
 struct PwCol {
     std::string name;
     int dtype;            // pmh_dtype
-    const void *data;     // host pointer, output width (int8 = 1 B, ...)
+    const void *data;     // host pointer, output width (int8 = 1 B, ...);
+                          // dtype 7 (string): int32 dictionary ids
     const uint8_t *valid; // byte validity or null (REQUIRED)
+    // dtype 7: the dictionary (entry i = dict_data[off[i], off[i+1]))
+    const uint8_t *dict_data = nullptr;
+    const int32_t *dict_offsets = nullptr;
+    int32_t dict_len = 0;
+    // DECIMAL(p,s) annotation on INT32/INT64 (0 = plain integer)
+    int32_t precision = 0;
+    int32_t scale = 0;
 };
 
 // codec: CODEC_UNCOMPRESSED or CODEC_ZSTD (parquet_meta.h values)

@@ -18,6 +18,7 @@
 #include <memory>
 #include <queue>
 #include <string>
+#include <unordered_map>
 #include <vector>
 
 #include "../../include/paimon_hip.h"
@@ -137,6 +138,29 @@ struct ColSpec {
     int dtype;  // pmh_dtype
     int out_esize;
     int stored_esize;  // parquet physical width (TINYINT stored as INT32)
+    int precision = 0;  // DECIMAL(p,s) annotation on INT32/INT64
+    int scale = 0;
+};
+
+// Plan-level GLOBAL dictionary for one string column: per-file parquet
+// dictionaries remap into it at staging (id streams then decode to global
+// ids on device and flow through merge/emit as int32), and the output
+// batch exposes it — the same dictionary-vector shape the reference's
+// columnar batches use (paimon-common heap vectors with setDictionary).
+struct StrDict {
+    std::vector<uint8_t> bytes;
+    std::vector<int32_t> offsets{0};
+    std::unordered_map<std::string, int32_t> index;
+    int32_t add(const uint8_t *p, uint32_t len) {
+        std::string s((const char *)p, len);
+        auto it = index.find(s);
+        if (it != index.end()) return it->second;
+        int32_t id = (int32_t)offsets.size() - 1;
+        bytes.insert(bytes.end(), p, p + len);
+        offsets.push_back((int32_t)bytes.size());
+        index.emplace(std::move(s), id);
+        return id;
+    }
 };
 
 struct DeviceBufs {
@@ -285,6 +309,7 @@ struct pmh_plan_t {
     std::vector<std::vector<uint8_t>> out_host;
     std::vector<pmh_col> batch_cols;
     std::vector<std::string> col_names;
+    std::vector<std::unique_ptr<pmh::StrDict>> sdicts;  // per col; string only
     pmh_stats stats{};
     double h2d_ms = 0;
 };
@@ -298,7 +323,17 @@ static int dtype_from_str(const std::string &s) {
     if (s == "int64" || s == "bigint") return PMH_DT_INT64;
     if (s == "float" || s == "float32") return PMH_DT_FLOAT32;
     if (s == "double" || s == "float64") return PMH_DT_FLOAT64;
+    if (s == "string" || s == "varchar" || s == "char") return PMH_DT_STRING;
     return -1;
+}
+
+// "decimal(p,s)": p <= 9 rides INT32, p <= 18 INT64 (unscaled values) —
+// exactly the reference's physical mapping, ParquetSchemaConverter.java:
+// 153-171 + is32BitDecimal/is64BitDecimal :334-341. Wider decimals (FLBA)
+// are a later round.
+static bool parse_decimal(const std::string &s, int *p, int *sc) {
+    if (s.rfind("decimal(", 0) != 0 || s.back() != ')') return false;
+    return sscanf(s.c_str(), "decimal(%d,%d)", p, sc) == 2;
 }
 
 static int dtype_out_esize(int dt) {
@@ -309,6 +344,7 @@ static int dtype_out_esize(int dt) {
     case PMH_DT_INT64: return 8;
     case PMH_DT_FLOAT32: return 4;
     case PMH_DT_FLOAT64: return 8;
+    case PMH_DT_STRING: return 4;  // global dictionary ids
     }
     return 0;
 }
@@ -322,11 +358,25 @@ static int dtype_stored_esize(int dt) {
     case PMH_DT_INT64: return 8;
     case PMH_DT_FLOAT32: return 4;
     case PMH_DT_FLOAT64: return 8;
+    case PMH_DT_STRING: return 4;  // the RLE id streams decode to int32
     }
     return 0;
 }
 
 static inline int popcount8(uint8_t b) { return __builtin_popcount(b); }
+
+static int expected_phys(int dtype) {
+    switch (dtype) {
+    case PMH_DT_INT8:
+    case PMH_DT_INT16:
+    case PMH_DT_INT32: return PHYS_INT32;
+    case PMH_DT_INT64: return PHYS_INT64;
+    case PMH_DT_FLOAT32: return PHYS_FLOAT;
+    case PMH_DT_FLOAT64: return PHYS_DOUBLE;
+    case PMH_DT_STRING: return PHYS_BYTE_ARRAY;
+    }
+    return -1;
+}
 
 // Walk a def-level stream (bit width 1): emit device work chunks with
 // running dense offsets (aux) and report the number of non-null values.
@@ -766,6 +816,12 @@ static bool prescan_present(const uint8_t *s, int64_t len, int64_t n_rows,
 static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
                            const StagedFile &sf, Run &run, int64_t row_base) {
     const auto &cols = plan->cols;
+    for (const auto &cs : cols)
+        if (cs.dtype == PMH_DT_STRING) {
+            set_error("%s: ORC string columns are not on the GPU path yet "
+                      "(parquet dictionary strings are)", fd.path.c_str());
+            return false;
+        }
     const OrcFileMeta &om = sf.orc_meta;
     std::vector<int> col_id(cols.size(), -1);
     for (size_t c = 0; c < cols.size(); c++) {
@@ -922,6 +978,47 @@ static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
 }
 
 // Stage one run (list of files) for the required columns onto the device.
+// Upload one chunk's dictionary for k_dict_gather. Numeric columns upload
+// the raw fixed-width values; STRING columns remap the chunk's byte-array
+// entries into the column's plan-level GLOBAL dictionary and upload the
+// int32 remap table instead (ids then decode straight to global ids).
+static bool upload_chunk_dict(pmh_plan_t *plan, int c,
+                              const uint8_t *dict_host, int64_t dict_count,
+                              int stored, void **out) {
+    if (plan->cols[c].dtype == PMH_DT_STRING) {
+        if (!plan->sdicts[c]) plan->sdicts[c].reset(new StrDict());
+        StrDict *sd = plan->sdicts[c].get();
+        std::vector<int32_t> remap((size_t)dict_count);
+        const uint8_t *p = dict_host;
+        for (int64_t i = 0; i < dict_count; i++) {
+            uint32_t len;
+            memcpy(&len, p, 4);
+            p += 4;
+            remap[i] = sd->add(p, len);
+            p += len;
+        }
+        void *dev = plan->bufs.alloc((dict_count ? dict_count : 1) * 4);
+        if (!dev) return false;
+        if (dict_count &&
+            hipMemcpy(dev, remap.data(), dict_count * 4,
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_error("H2D string remap failed");
+            return false;
+        }
+        *out = dev;
+        return true;
+    }
+    void *dev = plan->bufs.alloc(dict_count * stored);
+    if (!dev) return false;
+    if (hipMemcpy(dev, dict_host, dict_count * stored,
+                  hipMemcpyHostToDevice) != hipSuccess) {
+        set_error("H2D dict failed");
+        return false;
+    }
+    *out = dev;
+    return true;
+}
+
 static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                       Run &run) {
     const auto &cols = plan->cols;
@@ -963,6 +1060,17 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
             if (leaf[c] < 0) {
                 set_error("%s: column %s not found", fd.path.c_str(),
                           cols[c].name.c_str());
+                return false;
+            }
+            if ((int)sf.meta.phys_types.size() > leaf[c] &&
+                sf.meta.phys_types[leaf[c]] != expected_phys(cols[c].dtype)) {
+                set_error("%s: column %s physical type %d does not match "
+                          "declared type (expect %d; decimals ride "
+                          "INT32/INT64 per ParquetSchemaConverter.java:"
+                          "153-171, strings BYTE_ARRAY)",
+                          fd.path.c_str(), cols[c].name.c_str(),
+                          sf.meta.phys_types[leaf[c]],
+                          expected_phys(cols[c].dtype));
                 return false;
             }
         }
@@ -1052,6 +1160,14 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                               fd.path.c_str(), cols[c].name.c_str());
                     return false;
                 }
+                if (cols[c].dtype == PMH_DT_STRING && has_plain) {
+                    set_error("%s col %s: PLAIN byte-array pages are not on "
+                              "the GPU path yet — string columns must be "
+                              "dictionary-encoded (the parquet writer "
+                              "default; C5's dictionary strings are)",
+                              fd.path.c_str(), cols[c].name.c_str());
+                    return false;
+                }
                 // per-page value offsets past def levels (+ null detection)
                 std::vector<int64_t> vpos(cc.pages.size(), 0);
                 bool chunk_nulls = false;
@@ -1122,13 +1238,10 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                                          ppo[pi] + pos + 1, rc.rle_host))
                             return false;
                     }
-                    void *dict_dev = plan->bufs.alloc(dict_count * stored);
-                    if (!dict_dev) return false;
-                    if (hipMemcpy(dict_dev, dict_host, dict_count * stored,
-                                  hipMemcpyHostToDevice) != hipSuccess) {
-                        set_error("H2D dict failed");
+                    void *dict_dev = nullptr;
+                    if (!upload_chunk_dict(plan, (int)c, dict_host,
+                                           dict_count, stored, &dict_dev))
                         return false;
-                    }
                     GatherTask gt;
                     gt.start = dense_start;
                     gt.n = rc.dense_before - dense_start;
@@ -1222,13 +1335,10 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                                          rc.rle_host))
                             return false;
                     }
-                    void *dict_dev = plan->bufs.alloc(dict_count * stored);
-                    if (!dict_dev) return false;
-                    if (hipMemcpy(dict_dev, dict_host, dict_count * stored,
-                                  hipMemcpyHostToDevice) != hipSuccess) {
-                        set_error("H2D dict failed");
+                    void *dict_dev = nullptr;
+                    if (!upload_chunk_dict(plan, (int)c, dict_host,
+                                           dict_count, stored, &dict_dev))
                         return false;
-                    }
                     GatherTask gt;
                     gt.start = chunk_row0;
                     gt.n = cc.num_values;
@@ -1471,7 +1581,21 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         auto add_col = [&](const Json &cj) -> bool {
             ColSpec cs;
             cs.name = cj["name"].as_str();
-            cs.dtype = dtype_from_str(cj["type"].as_str());
+            const std::string ts = cj["type"].as_str();
+            int dp = 0, ds = 0;
+            if (parse_decimal(ts, &dp, &ds)) {
+                if (dp <= 0 || dp > 18 || ds < 0 || ds > dp) {
+                    set_error("decimal(%d,%d): v1 supports precision 1..18 "
+                              "(INT32/INT64 physical per the reference; "
+                              "FLBA decimals are a later round)", dp, ds);
+                    return false;
+                }
+                cs.dtype = dp <= 9 ? PMH_DT_INT32 : PMH_DT_INT64;
+                cs.precision = dp;
+                cs.scale = ds;
+            } else {
+                cs.dtype = dtype_from_str(ts);
+            }
             if (cs.dtype < 0 || cs.name.empty()) {
                 set_error("bad column spec");
                 return false;
@@ -1692,6 +1816,7 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             return nullptr;
         }
         for (auto &cs : plan->cols) plan->col_names.push_back(cs.name);
+        plan->sdicts.resize(plan->cols.size());
         if (plan->agg) {
             // "aggregations": {"col": "sum", ...}; unnamed value columns get
             // last_non_null_value (AggregateMergeFunction.java:197-203);
@@ -1725,6 +1850,14 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                     set_error("aggregate function '%s' not on the GPU path "
                               "(v1: sum, max, min, last_value, first_value, "
                               "last_non_null_value, first_non_null_value)",
+                              kv.second.as_str().c_str());
+                    return nullptr;
+                }
+                if (plan->cols[idx].dtype == PMH_DT_STRING &&
+                    (code == PMH_AGG_SUM || code == PMH_AGG_MAX ||
+                     code == PMH_AGG_MIN)) {
+                    set_error("aggregate '%s' on a string column is not on "
+                              "the GPU path (ids are not ordered by value)",
                               kv.second.as_str().c_str());
                     return nullptr;
                 }
@@ -2051,6 +2184,17 @@ collect:
             pc.valid = p->host_output
                            ? (const uint8_t *)p->out_valid_host[c].data()
                            : (const uint8_t *)p->out_valid[c];
+        pc.dict_data = nullptr;
+        pc.dict_offsets = nullptr;
+        pc.dict_len = 0;
+        pc.precision = p->cols[c].precision;
+        pc.scale = p->cols[c].scale;
+        if (p->cols[c].dtype == PMH_DT_STRING && p->sdicts[c]) {
+            StrDict *sd = p->sdicts[c].get();
+            pc.dict_data = sd->bytes.data();
+            pc.dict_offsets = sd->offsets.data();
+            pc.dict_len = (int32_t)sd->offsets.size() - 1;
+        }
     }
     if (out) {
         out->n_rows = total;
@@ -2120,6 +2264,11 @@ int pmh_write_parquet(const pmh_col *cols, int32_t n_cols, int64_t n_rows,
         pc[i].dtype = cols[i].dtype;
         pc[i].data = cols[i].data;
         pc[i].valid = cols[i].valid;
+        pc[i].dict_data = (const uint8_t *)cols[i].dict_data;
+        pc[i].dict_offsets = cols[i].dict_offsets;
+        pc[i].dict_len = cols[i].dict_len;
+        pc[i].precision = cols[i].precision;
+        pc[i].scale = cols[i].scale;
         if (pc[i].name.empty() || (!pc[i].data && n_rows > 0)) {
             set_error("pmh_write_parquet: column %d missing name/data", i);
             return -1;

@@ -178,3 +178,113 @@ def write_runs(runs, out_dir, compression="NONE", row_group_rows=1 << 20,
     with open(manifest, "w") as f:
         json.dump(metas, f, indent=1)
     return metas
+
+
+def gen_runs_c5(n_runs: int, rows_per_run: int, seed: int = 45,
+                str_card: int = 1000, delete_frac: float = 0.05):
+    """Runs for the C5 compaction config (BASELINE configs[4]): mixed types —
+    int64 pk, 4 x int32, decimal(18,2) (unscaled int64) and a dictionary-
+    encoded string of cardinality `str_card`. The string column is carried
+    as int32 ids + a shared dictionary (runs[i]["str_dict"]); the decimal as
+    unscaled int64 (INT64 physical per ParquetSchemaConverter.java:153-171).
+    Column order: v_k(int64), v_c0..3(int32), v_dec(decimal 18,2),
+    v_str(string)."""
+    rng = np.random.default_rng(seed)
+    total = n_runs * rows_per_run
+    space = _key_space(total)
+    seqs_all = rng.permutation(total).astype(np.int64)
+    sdict = [f"s{i:04d}-{rng.integers(0, 1 << 30):08x}" for i in
+             range(str_card)]
+    runs = []
+    for r in range(n_runs):
+        keys = _sorted_unique_keys(rng, rows_per_run, space)
+        seq = seqs_all[r * rows_per_run:(r + 1) * rows_per_run]
+        kind = np.where(rng.random(rows_per_run) < delete_frac,
+                        KIND_DELETE, KIND_INSERT).astype(np.int8)
+        values = [keys.copy()]
+        for _ in range(4):
+            values.append(rng.integers(-2**31, 2**31, size=rows_per_run,
+                                       dtype=np.int64).astype(np.int32))
+        values.append(rng.integers(-10**12, 10**12, size=rows_per_run,
+                                   dtype=np.int64))  # unscaled decimal
+        values.append(rng.integers(0, str_card, size=rows_per_run,
+                                   dtype=np.int32))  # string ids
+        runs.append({"key": keys, "seq": seq, "kind": kind,
+                     "values": values, "str_dict": sdict})
+    return runs
+
+
+C5_VALUE_COLS = ([{"name": "v_k", "type": "int64"}] +
+                 [{"name": f"v_c{i}", "type": "int32"} for i in range(4)] +
+                 [{"name": "v_dec", "type": "decimal(18,2)"},
+                  {"name": "v_str", "type": "string"}])
+
+
+def write_runs_c5(runs, out_dir, writer="pyarrow", compression="NONE",
+                  row_group_rows=0, data_page_rows=0):
+    """Write C5 runs. writer="pyarrow" pins the on-disk form against an
+    independent implementation (decimal written via store_decimal_as_integer
+    so the physical type is INT64, as the reference writes it; the string
+    column dictionary-encoded). writer="native" uses the library's own
+    writer (fast path for bench-scale data)."""
+    import decimal as _dec
+    os.makedirs(out_dir, exist_ok=True)
+    metas = []
+    names = [c["name"] for c in C5_VALUE_COLS]
+    for i, r in enumerate(runs):
+        path = os.path.join(out_dir, f"run-{i}.parquet")
+        n = len(r["key"])
+        sdict = r["str_dict"]
+        if writer == "native":
+            from .reader import write_parquet as native_write
+            cols = [("_KEY_k", r["key"]),
+                    ("_SEQUENCE_NUMBER", r["seq"]),
+                    ("_VALUE_KIND", r["kind"])]
+            for c, nm in enumerate(names):
+                cols.append((nm, r["values"][c]))
+            native_write(path, cols, compression=compression,
+                         row_group_rows=row_group_rows,
+                         page_rows=data_page_rows,
+                         dicts={"v_str": sdict},
+                         decimals={"v_dec": (18, 2)})
+        else:
+            fields = [pa.field("_KEY_k", pa.int64(), nullable=False),
+                      pa.field("_SEQUENCE_NUMBER", pa.int64(),
+                               nullable=False),
+                      pa.field("_VALUE_KIND", pa.int8(), nullable=False),
+                      pa.field("v_k", pa.int64(), nullable=False)]
+            fields += [pa.field(f"v_c{j}", pa.int32(), nullable=False)
+                       for j in range(4)]
+            fields += [pa.field("v_dec", pa.decimal128(18, 2),
+                                nullable=False),
+                       pa.field("v_str", pa.dictionary(pa.int32(),
+                                                       pa.string()),
+                                nullable=False)]
+            dec_vals = [
+                _dec.Decimal(int(u)).scaleb(-2)
+                for u in r["values"][5]]
+            arrays = ([pa.array(r["key"]), pa.array(r["seq"]),
+                       pa.array(r["kind"]), pa.array(r["values"][0])] +
+                      [pa.array(r["values"][1 + j]) for j in range(4)] +
+                      [pa.array(dec_vals, pa.decimal128(18, 2)),
+                       pa.DictionaryArray.from_arrays(
+                           pa.array(r["values"][6]),
+                           pa.array(sdict))])
+            tbl = pa.Table.from_arrays(arrays, schema=pa.schema(fields))
+            kw = {}
+            if row_group_rows:
+                kw["row_group_size"] = row_group_rows
+            if data_page_rows:
+                kw["data_page_size"] = data_page_rows * 8
+            pq.write_table(tbl, path,
+                           compression=None if compression in ("NONE",
+                                                               None)
+                           else compression.lower(),
+                           use_dictionary=["v_str"],
+                           store_decimal_as_integer=True,
+                           data_page_version="1.0", store_schema=False,
+                           **kw)
+        metas.append({"path": path, "rowCount": n,
+                      "minKey": int(r["key"][0]),
+                      "maxKey": int(r["key"][-1]), "level": 0})
+    return metas

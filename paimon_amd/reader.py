@@ -25,12 +25,17 @@ _HERE = os.path.dirname(os.path.abspath(__file__))
 LIB_PATH = os.path.join(_HERE, "libpaimon_hip.so")
 
 _DT_NP = {1: np.int8, 2: np.int16, 3: np.int32, 4: np.int64,
-          5: np.float32, 6: np.float64}
+          5: np.float32, 6: np.float64,
+          7: np.int32}  # PMH_DT_STRING: global-dictionary ids
 
 
 class _Col(ctypes.Structure):
     _fields_ = [("name", ctypes.c_char_p), ("dtype", ctypes.c_int32),
-                ("data", ctypes.c_void_p), ("valid", ctypes.c_void_p)]
+                ("data", ctypes.c_void_p), ("valid", ctypes.c_void_p),
+                ("dict_data", ctypes.c_void_p),
+                ("dict_offsets", ctypes.POINTER(ctypes.c_int32)),
+                ("dict_len", ctypes.c_int32),
+                ("precision", ctypes.c_int32), ("scale", ctypes.c_int32)]
 
 
 class _Batch(ctypes.Structure):
@@ -113,12 +118,22 @@ _NP_DT = {np.dtype(np.int8): 1, np.dtype(np.int16): 2,
 
 
 def write_parquet(path, columns, row_group_rows=0, page_rows=0,
-                  compression="NONE"):
-    """Write a Parquet v1 data file (PLAIN, uncompressed) via the native
-    writer (pmh_write_parquet) — the compaction write-back path. `columns`
-    is an ordered list of (name, values[, valid]) with numpy arrays; valid
-    None/omitted = REQUIRED column."""
+                  compression="NONE", dicts=None, decimals=None):
+    """Write a Parquet v1 data file via the native writer
+    (pmh_write_parquet) — the compaction write-back path. `columns` is an
+    ordered list of (name, values[, valid]) with numpy arrays; valid
+    None/omitted = REQUIRED column.
+
+    dicts: {name: sequence of bytes/str} — the column's values are int32
+    ids into that dictionary; written as a BYTE_ARRAY column with a
+    dictionary page + RLE_DICTIONARY data pages (the reference writer's
+    default for strings).
+    decimals: {name: (precision, scale)} — int32/int64 unscaled values
+    written with the DECIMAL annotation (INT32 p<=9 / INT64 p<=18,
+    ParquetSchemaConverter.java:153-171)."""
     lib = load_lib()
+    dicts = dicts or {}
+    decimals = decimals or {}
     n_rows = len(columns[0][1]) if columns else 0
     arrs = []  # keep contiguous buffers alive
     cols = (_Col * len(columns))()
@@ -133,6 +148,26 @@ def write_parquet(path, columns, row_group_rows=0, page_rows=0,
         cols[i].name = name.encode()
         cols[i].dtype = _NP_DT[vals.dtype]
         cols[i].data = vals.ctypes.data_as(ctypes.c_void_p)
+        if name in dicts:
+            if vals.dtype != np.int32:
+                raise ValueError(f"{name}: dictionary ids must be int32")
+            entries = [e.encode() if isinstance(e, str) else bytes(e)
+                       for e in dicts[name]]
+            blob = b"".join(entries)
+            offs = np.zeros(len(entries) + 1, dtype=np.int32)
+            np.cumsum([len(e) for e in entries], out=offs[1:])
+            blob_arr = np.frombuffer(blob, dtype=np.uint8).copy() if blob \
+                else np.zeros(1, dtype=np.uint8)
+            arrs += [blob_arr, offs]
+            cols[i].dtype = 7
+            cols[i].dict_data = blob_arr.ctypes.data_as(ctypes.c_void_p)
+            cols[i].dict_offsets = offs.ctypes.data_as(
+                ctypes.POINTER(ctypes.c_int32))
+            cols[i].dict_len = len(entries)
+        elif name in decimals:
+            p, s = decimals[name]
+            cols[i].precision = int(p)
+            cols[i].scale = int(s)
         if valid is not None:
             v8 = np.ascontiguousarray(np.asarray(valid, dtype=np.uint8))
             if len(v8) != n_rows:
@@ -235,6 +270,18 @@ class MergeReadPlan:
             else:
                 arr = np.empty(0, dtype=dt)
             out[name] = arr
+            if col.dtype == 7 and col.dict_len > 0:
+                # dictionary string column: values are global ids; expose
+                # the dictionary as an object array of bytes
+                offs = np.ctypeslib.as_array(col.dict_offsets,
+                                             shape=(col.dict_len + 1,))
+                blob = ctypes.cast(
+                    col.dict_data,
+                    ctypes.POINTER(ctypes.c_uint8 * int(offs[-1])))
+                bb = bytes(blob.contents) if offs[-1] else b""
+                out[name + "#dict"] = np.array(
+                    [bb[offs[j]:offs[j + 1]] for j in range(col.dict_len)],
+                    dtype=object)
             if col.valid and b.n_rows:
                 vbuf = ctypes.cast(col.valid,
                                    ctypes.POINTER(ctypes.c_uint8 * b.n_rows))
