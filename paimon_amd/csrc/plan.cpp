@@ -1648,6 +1648,7 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         }
         for (const auto &c : j["value_cols"].arr)
             if (!add_col(c)) return nullptr;
+        plan->sdicts.resize(plan->cols.size());  // before staging touches it
         std::string engine = j["merge_engine"].as_str("deduplicate");
         if (engine == "partial-update") {
             plan->pu = true;  // INSERT-only unless remove-record-on-delete
@@ -1816,7 +1817,6 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             return nullptr;
         }
         for (auto &cs : plan->cols) plan->col_names.push_back(cs.name);
-        plan->sdicts.resize(plan->cols.size());
         if (plan->agg) {
             // "aggregations": {"col": "sum", ...}; unnamed value columns get
             // last_non_null_value (AggregateMergeFunction.java:197-203);
