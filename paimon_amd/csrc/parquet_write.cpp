@@ -115,15 +115,19 @@ void encode_dict_ids(const PwCol &c, int64_t s, int64_t e,
     TC t;
     t.uvarint(((uint64_t)ngroups << 1) | 1);
     out.append(t.out);
-    for (int64_t g = 0; g < ngroups; g++) {
-        uint64_t word = 0;
-        for (int j = 0; j < 8; j++) {
-            int64_t i = g * 8 + j;
-            uint64_t v = i < (int64_t)ids.size() ? (uint64_t)ids[i] : 0;
-            word |= v << (j * bw);
+    // streaming bit accumulator: a group is 8 values x bw bits = bw bytes,
+    // LSB-first (Packer.LITTLE_ENDIAN); bw can exceed 8 (large dicts)
+    uint64_t acc = 0;
+    int nbits = 0;
+    for (int64_t i = 0; i < ngroups * 8; i++) {
+        uint64_t v = i < (int64_t)ids.size() ? (uint64_t)ids[i] : 0;
+        acc |= v << nbits;
+        nbits += bw;
+        while (nbits >= 8) {
+            out.push_back((char)(acc & 0xFF));
+            acc >>= 8;
+            nbits -= 8;
         }
-        for (int b = 0; b < bw; b++)
-            out.push_back((char)((word >> (8 * b)) & 0xFF));
     }
 }
 
