@@ -111,8 +111,13 @@ def ensure_data(args, rank):
 
 
 def cpu_baseline(args, metas_dir):
-    """Time the C oracle loser-tree dedup on a bounded prefix sample."""
-    from oracle import merge_dedup
+    """Time the C oracle loser-tree dedup on a bounded prefix sample:
+    N-thread (key-space-sliced loser trees, BASELINE.md leg 2) headline,
+    single-thread rate in the sample note. The pypaimon leg (BASELINE.md
+    leg 1) cannot run on the GPU box (/root/reference does not travel);
+    its in-container measurement is committed in
+    profiles/pypaimon_baseline.json."""
+    from oracle import merge_dedup, merge_dedup_count_mt
     from paimon_amd.datagen import gen_runs_dedup
     rows = min(args.cpu_baseline_rows, args.rows)
     runs = gen_runs_dedup(args.runs, args.rows, n_value_cols=0,
@@ -122,15 +127,20 @@ def cpu_baseline(args, metas_dir):
     n = sum(len(r["key"]) for r in sample)
     t0 = time.perf_counter()
     merge_dedup(sample, drop_delete=True)
-    dt = time.perf_counter() - t0
+    dt1 = time.perf_counter() - t0
+    cores = os.cpu_count() or 1
+    t0 = time.perf_counter()
+    merge_dedup_count_mt(sample, cores)
+    dtn = time.perf_counter() - t0
     return {
-        "value": n / dt,
+        "value": n / dtn,
         "unit": "rows/s",
-        "cores": 1,
+        "cores": cores,
         "kind": "port",
         "sample": (f"{args.runs} runs x {rows}-row prefixes of the same "
-                   f"workload ({n} rows, {dt:.1f}s single-thread C "
-                   "loser-tree restatement)"),
+                   f"workload ({n} rows; {cores}-thread key-space-sliced C "
+                   f"loser-tree restatement {dtn:.2f}s; single-thread "
+                   f"{n / dt1 / 1e6:.1f} M rows/s in {dt1:.2f}s)"),
     }
 
 

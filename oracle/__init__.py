@@ -7,6 +7,7 @@ libpaimon_hip.so) must never route through it.
 from .oracle import (  # noqa: F401
     merge_order,
     merge_dedup,
+    merge_dedup_count_mt,
     merge_dedup_model,
     merge_first_row_model,
     partial_update_model,

@@ -19,6 +19,7 @@
 #include <stdint.h>
 #include <stdlib.h>
 #include <string.h>
+#include <pthread.h>
 
 /* ---------------------------------------------------------------- *
  * RowKind byte encoding: paimon-api/.../types/RowKind.java:35-56.
@@ -430,6 +431,92 @@ int64_t pmo_merge_dedup_count(int n_runs, const int64_t **keys,
     free(orun);
     free(orow);
     return n;
+}
+
+/* N-thread variant of the dedup merge for the cpu_baseline (BASELINE.md:
+ * "single-thread and N-thread (one thread per bucket)"; C2 is one bucket,
+ * so the threads split the KEY SPACE instead — per-thread two-sided cuts
+ * via upper_bound of split keys, so an equal-key group never spans a
+ * boundary and per-slice loser trees compose exactly). Reported baseline
+ * only; never on the product path. */
+typedef struct {
+    int n_runs;
+    const int64_t **keys;
+    const int64_t **seqs;
+    const int8_t **kinds;
+    int64_t lo[64];  /* per-run slice bounds */
+    int64_t hi[64];
+    int ignore_delete, drop_delete;
+    int64_t out;
+} mt_slice;
+
+static int64_t ub_key(const int64_t *a, int64_t n, int64_t v) {
+    int64_t lo = 0, hi = n;
+    while (lo < hi) {
+        int64_t mid = lo + ((hi - lo) >> 1);
+        if (a[mid] <= v) lo = mid + 1;
+        else hi = mid;
+    }
+    return lo;
+}
+
+static void *mt_worker(void *arg) {
+    mt_slice *s = (mt_slice *)arg;
+    const int64_t *keys[64];
+    const int64_t *seqs[64];
+    const int8_t *kinds[64];
+    int64_t lens[64];
+    for (int r = 0; r < s->n_runs; r++) {
+        keys[r] = s->keys[r] + s->lo[r];
+        seqs[r] = s->seqs[r] + s->lo[r];
+        kinds[r] = s->kinds[r] + s->lo[r];
+        lens[r] = s->hi[r] - s->lo[r];
+    }
+    s->out = pmo_merge_dedup_count(s->n_runs, keys, seqs, kinds, lens,
+                                   s->ignore_delete, s->drop_delete);
+    return NULL;
+}
+
+int64_t pmo_merge_dedup_count_mt(int n_runs, const int64_t **keys,
+                                 const int64_t **seqs, const int8_t **kinds,
+                                 const int64_t *lens, int ignore_delete,
+                                 int drop_delete, int n_threads) {
+    if (n_threads < 1) n_threads = 1;
+    if (n_threads > 64) n_threads = 64;
+    if (n_runs > 64) return -1;
+    /* split keys: quantiles of the largest run's key column */
+    int big = 0;
+    for (int r = 1; r < n_runs; r++)
+        if (lens[r] > lens[big]) big = r;
+    if (lens[big] == 0) return 0;
+    mt_slice *sl = calloc(n_threads, sizeof(mt_slice));
+    pthread_t *th = malloc(sizeof(pthread_t) * n_threads);
+    for (int t = 0; t < n_threads; t++) {
+        sl[t].n_runs = n_runs;
+        sl[t].keys = keys;
+        sl[t].seqs = seqs;
+        sl[t].kinds = kinds;
+        sl[t].ignore_delete = ignore_delete;
+        sl[t].drop_delete = drop_delete;
+        for (int r = 0; r < n_runs; r++) {
+            if (t == 0) sl[t].lo[r] = 0;
+            else sl[t].lo[r] = sl[t - 1].hi[r];
+            if (t == n_threads - 1) sl[t].hi[r] = lens[r];
+            else {
+                int64_t split = keys[big][(lens[big] * (t + 1)) / n_threads];
+                sl[t].hi[r] = ub_key(keys[r], lens[r], split);
+            }
+        }
+    }
+    for (int t = 0; t < n_threads; t++) pthread_create(&th[t], NULL, mt_worker, &sl[t]);
+    int64_t total = 0;
+    for (int t = 0; t < n_threads; t++) {
+        pthread_join(th[t], NULL);
+        total += sl[t].out;
+    }
+    free(th);
+    free(sl);
+    return total;
 }
 
 /* ================================================================ *

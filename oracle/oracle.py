@@ -37,6 +37,7 @@ def _get_lib():
         _lib.pmo_merge_order.restype = ctypes.c_int64
         _lib.pmo_merge_dedup.restype = ctypes.c_int64
         _lib.pmo_merge_dedup_count.restype = ctypes.c_int64
+        _lib.pmo_merge_dedup_count_mt.restype = ctypes.c_int64
         _lib.pmo_rle_bp_decode.restype = ctypes.c_int64
         _lib.pmo_rle_bp_decode.argtypes = [
             ctypes.c_char_p, ctypes.c_int64, ctypes.c_int, ctypes.c_int64,
@@ -113,6 +114,17 @@ def _sorted_stream(runs):
                           for r in runs])
     order = np.lexsort((_kind_is_add(kind).astype(np.int8), seq, key))
     return key[order], seq[order], kind[order], run[order], row[order]
+
+
+def merge_dedup_count_mt(runs, n_threads, ignore_delete=False,
+                         drop_delete=True):
+    """N-thread dedup merge COUNT (key-space sliced loser trees) — the
+    BASELINE.md N-thread CPU restatement leg. Timing baseline only."""
+    lib = _get_lib()
+    keys, seqs, kinds, lens, hold = _ptr_arrays(runs)
+    return lib.pmo_merge_dedup_count_mt(
+        len(runs), keys, seqs, kinds, lens, int(ignore_delete),
+        int(drop_delete), int(n_threads))
 
 
 def merge_dedup_model(runs, ignore_delete=False, drop_delete=True):
