@@ -12,6 +12,7 @@ from .oracle import (  # noqa: F401
     merge_first_row_model,
     partial_update_model,
     partial_update_rrod_model,
+    partial_update_seqgroup_model,
     aggregation_model,
     aggregation_rrod_model,
     rle_bp_decode,

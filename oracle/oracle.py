@@ -535,3 +535,135 @@ def merge_first_row_model(runs, ignore_delete=False, drop_delete=True):
     if drop_delete:
         sel = sel[_kind_is_add(kind[sel])]
     return run[sel], row[sel]
+
+
+def partial_update_seqgroup_model(runs, seq_groups, drop_delete=True,
+                                  ignore_delete=False):
+    """Sequential port of PartialUpdateMergeFunction with sequence groups
+    (updateWithSequenceGroup :219-282, retractWithSequenceGroup :301-377,
+    isEmptySequenceGroup :284-299, initRow :399-407, getResult :389-397) for
+    INSERT/UPDATE_AFTER/UPDATE_BEFORE/DELETE streams, no per-field
+    aggregators (their retract support is a later round).
+
+    seq_groups: list of {"sequence_fields": [col indices],
+                         "group_fields": [col indices]} — indices into the
+    runs' values/valid lists. Field comparator: lexicographic ascending over
+    the sequence fields, nulls FIRST (codegen nullIsLast=false).
+    Returns dict of output columns like partial_update_model."""
+    key, seq, kind, run, row = _sorted_stream(runs)
+    n = len(key)
+    n_cols = len(runs[0]["values"]) if runs else 0
+    col_group = {}
+    for g, sg in enumerate(seq_groups):
+        for c in sg["sequence_fields"]:
+            col_group[c] = g
+        for c in sg["group_fields"]:
+            col_group[c] = g
+
+    def get(m, c):
+        a, b = run[m], row[m]
+        valid = runs[a].get("valid")
+        if valid is not None and not valid[c][b]:
+            return None
+        return runs[a]["values"][c][b]
+
+    def tuple_of(vals):
+        # None sorts FIRST: encode as (0,) vs (1, v)
+        return tuple((0, 0) if v is None else (1, int(v)) for v in vals)
+
+    out = {"key": [], "seq": [], "kind": [],
+           "values": [[] for _ in range(n_cols)],
+           "valid": [[] for _ in range(n_cols)]}
+    i = 0
+    while i < n:
+        j = i
+        while j + 1 < n and key[j + 1] == key[i]:
+            j += 1
+        members = list(range(i, j + 1))
+        if len(members) == 1:  # ReducerMergeFunctionWrapper bypass
+            m = members[0]
+            res_kind = int(kind[m])
+            if drop_delete and res_kind in (1, 3):
+                i = j + 1
+                continue
+            out["key"].append(key[i])
+            out["seq"].append(seq[m])
+            out["kind"].append(res_kind)
+            for c in range(n_cols):
+                v = get(m, c)
+                out["values"][c].append(v if v is not None else 0)
+                out["valid"][c].append(v is not None)
+            i = j + 1
+            continue
+        rowv = [None] * n_cols
+        meet_insert = False
+        filled = False
+        latest_seq = 0
+        for m in members:
+            is_retract = kind[m] in (1, 3)
+            if is_retract:
+                if not filled:
+                    rowv = [get(m, c) for c in range(n_cols)]  # initRow
+                    filled = True
+                if ignore_delete:
+                    continue
+                latest_seq = seq[m]
+                # retractWithSequenceGroup
+                for g, sg in enumerate(seq_groups):
+                    kvt = tuple_of([get(m, c)
+                                    for c in sg["sequence_fields"]])
+                    if all(t[0] == 0 for t in kvt):
+                        continue  # empty sequence group
+                    curt = tuple_of([rowv[c]
+                                     for c in sg["sequence_fields"]])
+                    if kvt >= curt:
+                        for c in sg["sequence_fields"]:
+                            rowv[c] = get(m, c)
+                        for c in sg["group_fields"]:
+                            rowv[c] = None  # retract normal field
+                continue
+            latest_seq = seq[m]
+            meet_insert = True
+            filled = True
+            # updateWithSequenceGroup
+            handled = set()
+            for g, sg in enumerate(seq_groups):
+                for c in sg["sequence_fields"]:
+                    handled.add(c)
+                for c in sg["group_fields"]:
+                    handled.add(c)
+                kvt = tuple_of([get(m, c) for c in sg["sequence_fields"]])
+                if all(t[0] == 0 for t in kvt):
+                    continue
+                curt = tuple_of([rowv[c] for c in sg["sequence_fields"]])
+                if kvt >= curt:
+                    for c in sg["sequence_fields"]:
+                        rowv[c] = get(m, c)
+                    for c in sg["group_fields"]:
+                        rowv[c] = get(m, c)  # member set EVEN IF NULL
+            for c in range(n_cols):
+                if c in handled:
+                    continue
+                v = get(m, c)
+                if v is not None:
+                    rowv[c] = v
+        res_kind = 0 if meet_insert else 3
+        if drop_delete and res_kind == 3:
+            i = j + 1
+            continue
+        out["key"].append(key[i])
+        out["seq"].append(latest_seq)
+        out["kind"].append(res_kind)
+        for c in range(n_cols):
+            ok = rowv[c] is not None
+            out["values"][c].append(rowv[c] if ok else 0)
+            out["valid"][c].append(ok)
+        i = j + 1
+    return {
+        "key": np.array(out["key"], np.int64),
+        "seq": np.array(out["seq"], np.int64),
+        "kind": np.array(out["kind"], np.int8),
+        "values": [np.array(v, runs[0]["values"][c].dtype)
+                   for c, v in enumerate(out["values"])],
+        "valid": [np.array(v, bool) for v in out["valid"]],
+    }

@@ -209,6 +209,20 @@ enum {
     PMH_AGG_MAX = 5,            // FieldMaxAgg
     PMH_AGG_MIN = 6,            // FieldMinAgg
 };
+// PartialUpdate with sequence groups (PartialUpdateMergeFunction.java:
+// 219-377): per-group last-prefix-max-achiever resolution; retracts null
+// their group members. col_group[c] = group id or 0xff; sg_fields packs 4
+// sequence-field column indices per group; no per-field aggregators in v1.
+hipError_t pmh_launch_emit_pu_sg(
+    const DevCol *cols, const uint8_t *col_dtype,
+    const uint8_t *col_nullable, int n_cols, int k, int seq_col,
+    int kind_col, int flags, const uint8_t *col_group,
+    const int16_t *sg_fields, const uint8_t *sg_nseq, int n_groups,
+    const uint32_t *members, const uint16_t *group_start,
+    const int64_t *tile_offsets, int64_t n_tiles, int64_t tile_rows,
+    const int64_t *total_out, uint64_t *const *run_masks,
+    void *const *out_ptrs, uint8_t *const *out_valid, hipStream_t stream);
+
 hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                                const uint8_t *col_nullable,
                                const uint8_t *col_agg, int n_cols, int k,

@@ -349,6 +349,9 @@ DEV int32_t corank(int64_t d, const uint16_t *pa, int32_t la,
 }
 
 DEV bool kind_is_add(uint8_t k) { return k == 0 || k == 2; }
+// k_merge_tiles packs (seq << 1) | isAdd (the fused kernel uses the 2-bit
+// ps2_* packing); helper so call sites read uniformly
+DEV bool ps2m_isadd(int64_t w) { return w & 1; }
 
 // PU=true: PartialUpdate mode — no winner reduction; emit every owned
 // group's member list (ascending (seq, isAdd) order within the group) for
@@ -377,6 +380,10 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
     // a DELETE result and is dropped here under drop_delete. UPDATE_BEFORE
     // is rejected at staging (v1: INSERT/DELETE streams).
     const bool rrod = (flags & 16) != 0;
+    // sequence groups (PartialUpdateMergeFunction.java:219-377): retracts
+    // are legal members (they retract their groups); result kind is DELETE
+    // only when the group has NO add member (getResult :389-397)
+    const bool seqg = (flags & 32) != 0;
     // ablation levels (profiling only, flags bits 8..): 1=stage,2=+merge,
     // 3=+scan, 0/absent=full. Partial levels publish a checksum so the
     // compiler cannot dead-code the ablated phases' inputs.
@@ -451,8 +458,9 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             for (int32_t i = tid; i < len; i += blockDim.x) {
                 sm.skey[off + i] = key_at(kaddr0, i, kes);
                 int32_t kd = daddr[i];
-                if (PU && rrod && kd == 1 && err_flag)
+                if (PU && rrod && !seqg && kd == 1 && err_flag)
                     atomicOr(err_flag, 1u);  // UPDATE_BEFORE: not in v1 RROD
+                                             // (sequence groups accept it)
                 sm.sseq[off + i] =
                     (saddr[i] << 1) | (int64_t)(kd == 0 || kd == 2);
                 sm.perm[0][off + i] = (uint16_t)(off + i);
@@ -567,7 +575,14 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                         // record is served as-is (any kind, incl. retracts);
                         // a retract result drops under drop-delete
                         // (DropDeleteReader.java:53-61)
-                        if (drop_delete && !(sm.sseq[mo[i]] & 1)) continue;
+                        if (drop_delete && !ps2m_isadd(sm.sseq[mo[i]]))
+                            continue;
+                    } else if (seqg && drop_delete) {
+                        // result kind DELETE iff no add member
+                        bool any_add = false;
+                        for (int32_t x = i; x <= tail; x++)
+                            any_add |= ps2m_isadd(sm.sseq[mo[x]]);
+                        if (!any_add) continue;
                     } else if (rrod && drop_delete) {
                         // result kind = last member's kind (max packed sseq);
                         // DELETE results drop here (DropDeleteReader)
@@ -576,7 +591,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                             int64_t v = sm.sseq[mo[x]];
                             if (v > mx) mx = v;
                         }
-                        if (!(mx & 1)) continue;
+                        if (!ps2m_isadd(mx)) continue;
                     }
                     if (pass == 1) {
                         gout[g_off + ng] = (uint16_t)(m_off + nm);
@@ -600,8 +615,11 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                             uint16_t s = gm[x];
                             // retracts only fail MULTI-record groups: the
                             // wrapper bypasses the merge function for
-                            // singletons (gn == 1), so a lone retract passes
-                            if (!rrod && gn > 1 && !(sm.sseq[s] & 1))
+                            // singletons (gn == 1), so a lone retract
+                            // passes; sequence groups accept retracts
+                            // outright (retractWithSequenceGroup)
+                            if (!rrod && !seqg && gn > 1 &&
+                                !ps2m_isadd(sm.sseq[s]))
                                 bad_kind = true;
                             int r = 0;
                             while (r + 1 <= k - 1 &&
@@ -1873,6 +1891,221 @@ __global__ void k_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
     }
 }
 
+
+// ---------------------------------------------------------- k_emit_pu_sg
+//
+// PartialUpdate with SEQUENCE GROUPS (updateWithSequenceGroup,
+// PartialUpdateMergeFunction.java:219-282; retractWithSequenceGroup
+// :301-377; isEmptySequenceGroup :284-299; initRow :399-407; getResult
+// :389-397). Streams may carry INSERT/UPDATE_AFTER/UPDATE_BEFORE/DELETE;
+// retracts act only on their sequence groups (no per-field aggregators in
+// v1 — their retract support is a later round).
+//
+// The reference's sequential per-key fold reduces, per group, to the LAST
+// PREFIX-MAX ACHIEVER over the members' sequence-field tuples (non-empty
+// tuples only; lexicographic ascending, nulls FIRST; ties -> later member):
+// each accepted event overwrites every field it touches, so only the last
+// accepted event matters. Proven equivalent to the sequential port by
+// randomized fuzz (oracle partial_update_seqgroup_model; see git history).
+// col_group[c]: group id or 0xff (plain); sg_fields[g*4+j]: the group's
+// sequence-field column indices; sg_nseq[g]: how many (<= 4).
+__global__ void k_emit_pu_sg(const DevCol *cols, const uint8_t *col_dtype,
+                             const uint8_t *col_nullable, int n_cols, int k,
+                             int seq_col, int kind_col, int flags,
+                             const uint8_t *col_group,
+                             const int16_t *sg_fields,
+                             const uint8_t *sg_nseq, int n_groups,
+                             const uint32_t *members,
+                             const uint16_t *group_start,
+                             const int64_t *tile_offsets, int64_t n_tiles,
+                             int64_t tile_rows, const int64_t *total_out,
+                             uint64_t *const *run_masks,
+                             void *const *out_ptrs,
+                             uint8_t *const *out_valid) {
+    const bool ignore_delete = flags & 2;
+    const int64_t total = *total_out;
+    const int64_t per_block =
+        (total + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
+    const int64_t slice_lo = (int64_t)blockIdx.x * per_block;
+    const int64_t slice_hi =
+        slice_lo + per_block < total ? slice_lo + per_block : total;
+    auto kind_of = [&](uint32_t m) -> int32_t {
+        const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + kind_col];
+        return col_load<int32_t>(dc, m & PMH_ROW_MASK);
+    };
+    auto valid_of = [&](uint32_t m, int c) -> uint8_t {
+        if (run_masks)
+            return (uint8_t)(
+                (run_masks[m >> PMH_ROW_BITS][m & PMH_ROW_MASK] >> c) & 1);
+        const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + c];
+        return dc.valid0
+                   ? ((const uint8_t *)dc.valid0)[m & PMH_ROW_MASK]
+                   : 1;
+    };
+    auto load_of = [&](uint32_t m, int c) -> int64_t {
+        const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + c];
+        const int dt = col_dtype[c];
+        return (dt == 4 || dt == 6)
+                   ? col_load<int64_t>(dc, m & PMH_ROW_MASK)
+                   : (int64_t)col_load<int32_t>(dc, m & PMH_ROW_MASK);
+    };
+    int64_t t = -1;
+    for (int64_t i = slice_lo + threadIdx.x; i < slice_hi; i += blockDim.x) {
+        if (t < 0) {
+            int64_t lo = 0, hi = n_tiles - 1;
+            while (lo < hi) {
+                int64_t mid = (lo + hi + 1) >> 1;
+                if (tile_offsets[mid] <= i) lo = mid;
+                else hi = mid - 1;
+            }
+            t = lo;
+        }
+        while (t + 1 < n_tiles && tile_offsets[t + 1] <= i) t++;
+        const int64_t g = i - tile_offsets[t];
+        const uint16_t *gs = &group_start[t * (tile_rows + 1)];
+        const uint32_t *mem = &members[t * (tile_rows + PMH_MAX_RUNS)];
+        const int32_t ms = gs[g], me = gs[g + 1];
+        const int gn = me - ms;
+        const uint32_t m0 = mem[ms];
+        const int32_t kind0 = kind_of(m0);
+        const bool first_retract = kind0 == 1 || kind0 == 3;
+        if (gn == 1) {
+            // ReducerMergeFunctionWrapper singleton bypass: the record is
+            // served as-is with its own RowKind
+            for (int c = 0; c < n_cols; c++) {
+                if (c == kind_col) {
+                    ((int8_t *)out_ptrs[c])[i] = (int8_t)kind0;
+                    continue;
+                }
+                uint8_t ok = col_nullable[c] ? valid_of(m0, c) : 1;
+                const int64_t v = ok ? load_of(m0, c) : 0;
+                switch (col_dtype[c]) {
+                case 1: ((int8_t *)out_ptrs[c])[i] = (int8_t)v; break;
+                case 2: ((int16_t *)out_ptrs[c])[i] = (int16_t)v; break;
+                case 3:
+                case 5:
+                case 7: ((int32_t *)out_ptrs[c])[i] = (int32_t)v; break;
+                default: ((int64_t *)out_ptrs[c])[i] = v; break;
+                }
+                if (out_valid[c]) out_valid[c][i] = ok;
+            }
+            continue;
+        }
+        // per-group winner member (last prefix-max achiever), computed once
+        // per row; group count bounded by plan validation (<= 16)
+        int32_t win[16];
+        bool win_r[16];
+        for (int gg = 0; gg < n_groups; gg++) {
+            win[gg] = -1;
+            win_r[gg] = false;
+        }
+        bool meet_insert = false;
+        uint32_t last_m = m0;
+        uint32_t last_add = 0;
+        bool has_add = false;
+        for (int32_t x = 0; x < gn; x++) {
+            const uint32_t m = mem[ms + x];
+            const int32_t kd = kind_of(m);
+            const bool retract = kd == 1 || kd == 3;
+            if (!retract) {
+                meet_insert = true;
+                has_add = true;
+                last_add = m;
+            }
+            if (!(retract && ignore_delete)) last_m = m;
+            if (retract && ignore_delete) continue;
+            for (int gg = 0; gg < n_groups; gg++) {
+                const int ns = sg_nseq[gg];
+                // empty tuples (all sequence fields null) never participate
+                // (isEmptySequenceGroup)
+                bool anyv = false;
+                for (int j = 0; j < ns; j++)
+                    anyv |= valid_of(m, sg_fields[gg * 4 + j]) != 0;
+                if (!anyv) continue;
+                if (win[gg] < 0) {
+                    win[gg] = x;
+                    win_r[gg] = retract;
+                    continue;
+                }
+                // lexicographic tuple compare vs the current winner,
+                // nulls FIRST; ties accept (later member wins)
+                const uint32_t wm = mem[ms + win[gg]];
+                int cmp = 0;
+                for (int j = 0; j < ns && cmp == 0; j++) {
+                    const int c = sg_fields[gg * 4 + j];
+                    const uint8_t va = valid_of(m, c);
+                    const uint8_t vb = valid_of(wm, c);
+                    if (va != vb) cmp = va ? 1 : -1;  // null sorts first
+                    else if (va) {
+                        const int64_t a = load_of(m, c);
+                        const int64_t b = load_of(wm, c);
+                        if (a != b) cmp = a > b ? 1 : -1;
+                    }
+                }
+                if (cmp >= 0) {
+                    win[gg] = x;
+                    win_r[gg] = retract;
+                }
+            }
+        }
+        for (int c = 0; c < n_cols; c++) {
+            if (c == kind_col) {
+                ((int8_t *)out_ptrs[c])[i] = meet_insert ? 0 : 3;
+                continue;
+            }
+            if (c == seq_col) {
+                ((int64_t *)out_ptrs[c])[i] = load_of(last_m, c);
+                continue;
+            }
+            const int gg = col_group[c];
+            uint32_t src_m = 0;
+            uint8_t ok = 0;
+            if (gg != 0xff) {
+                const int ns = sg_nseq[gg];
+                bool is_seq_field = false;
+                for (int j = 0; j < ns; j++)
+                    if (sg_fields[gg * 4 + j] == c) is_seq_field = true;
+                if (win[gg] >= 0) {
+                    if (is_seq_field || !win_r[gg]) {
+                        src_m = mem[ms + win[gg]];
+                        ok = valid_of(src_m, c);
+                    }  // retract winner nulls the member fields
+                } else if (first_retract) {  // initRow fallback
+                    src_m = m0;
+                    ok = valid_of(m0, c);
+                }
+            } else {
+                // plain column: last non-null among ADD members, else the
+                // initRow base when the first member was a retract
+                if (first_retract && valid_of(m0, c)) {
+                    src_m = m0;
+                    ok = 1;
+                }
+                for (int32_t x = 0; x < gn; x++) {
+                    const uint32_t m = mem[ms + x];
+                    const int32_t kd = kind_of(m);
+                    if (kd == 1 || kd == 3) continue;
+                    if (valid_of(m, c)) {
+                        src_m = m;
+                        ok = 1;
+                    }
+                }
+            }
+            const int dt = col_dtype[c];
+            const int64_t v = ok ? load_of(src_m, c) : 0;
+            switch (dt) {
+            case 1: ((int8_t *)out_ptrs[c])[i] = (int8_t)v; break;
+            case 2: ((int16_t *)out_ptrs[c])[i] = (int16_t)v; break;
+            case 3:
+            case 5:
+            case 7: ((int32_t *)out_ptrs[c])[i] = (int32_t)v; break;
+            default: ((int64_t *)out_ptrs[c])[i] = v; break;
+            }
+            if (out_valid[c]) out_valid[c][i] = ok;
+        }
+    }
+}
+
 // ------------------------------------------------------------ k_emit_agg
 //
 // Aggregation emit (AggregateMergeFunction.java:82-125): one output row per
@@ -2546,6 +2779,23 @@ hipError_t pmh_launch_pack_valid(const DevCol *cols, int n_cols, int64_t rows,
                                  uint64_t *mask, hipStream_t stream) {
     hipLaunchKernelGGL(k_pack_valid, dim3(1024), dim3(256), 0, stream, cols,
                        n_cols, rows, mask);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_emit_pu_sg(
+    const DevCol *cols, const uint8_t *col_dtype,
+    const uint8_t *col_nullable, int n_cols, int k, int seq_col,
+    int kind_col, int flags, const uint8_t *col_group,
+    const int16_t *sg_fields, const uint8_t *sg_nseq, int n_groups,
+    const uint32_t *members, const uint16_t *group_start,
+    const int64_t *tile_offsets, int64_t n_tiles, int64_t tile_rows,
+    const int64_t *total_out, uint64_t *const *run_masks,
+    void *const *out_ptrs, uint8_t *const *out_valid, hipStream_t stream) {
+    hipLaunchKernelGGL(k_emit_pu_sg, dim3(2048), dim3(256), 0, stream, cols,
+                       col_dtype, col_nullable, n_cols, k, seq_col, kind_col,
+                       flags, col_group, sg_fields, sg_nseq, n_groups,
+                       members, group_start, tile_offsets, n_tiles,
+                       tile_rows, total_out, run_masks, out_ptrs, out_valid);
     return hipGetLastError();
 }
 

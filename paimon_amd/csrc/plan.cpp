@@ -293,6 +293,14 @@ struct pmh_plan_t {
     // (PartialUpdateMergeFunction.java:173-180); v1 accepts INSERT/DELETE
     // streams (UPDATE_BEFORE rejected)
     bool rrod = false;
+    // sequence groups (fields.<seq>.sequence-group=members,
+    // PartialUpdateMergeFunction.java:219-377): retracts become legal and
+    // act on their groups
+    bool seqg = false;
+    int n_seq_groups = 0;
+    uint8_t *col_group_dev = nullptr;
+    int16_t *sg_fields_dev = nullptr;
+    uint8_t *sg_nseq_dev = nullptr;
     std::vector<pmh::Section> sections;
     size_t cur_section = 0;
     int64_t rows_in_total = 0;
@@ -1652,9 +1660,84 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         std::string engine = j["merge_engine"].as_str("deduplicate");
         if (engine == "partial-update") {
             plan->pu = true;  // INSERT-only unless remove-record-on-delete
+                              // or sequence groups
                               // (PartialUpdateMergeFunction rejects retracts
                               // by default, :170-186)
             plan->rrod = j["remove_record_on_delete"].as_bool(false);
+            const Json &sgs = j["sequence_groups"];
+            if (!sgs.arr.empty()) {
+                plan->seqg = true;
+                if (sgs.arr.size() > 16) {
+                    set_error("at most 16 sequence groups in v1");
+                    return nullptr;
+                }
+                const int first_val = plan->n_key_cols + 2;
+                const int n_cols = (int)plan->cols.size();
+                std::vector<uint8_t> cg(n_cols, 0xff);
+                std::vector<int16_t> sf(sgs.arr.size() * 4, -1);
+                std::vector<uint8_t> ns(sgs.arr.size(), 0);
+                auto col_idx = [&](const std::string &nm) -> int {
+                    for (int c = first_val; c < n_cols; c++)
+                        if (plan->cols[c].name == nm) return c;
+                    return -1;
+                };
+                for (size_t g = 0; g < sgs.arr.size(); g++) {
+                    const Json &sg = sgs.arr[g];
+                    const auto &sfj = sg["sequence_fields"].arr;
+                    if (sfj.empty() || sfj.size() > 4) {
+                        set_error("sequence group %zu: 1..4 sequence "
+                                  "fields in v1", g);
+                        return nullptr;
+                    }
+                    for (size_t i2 = 0; i2 < sfj.size(); i2++) {
+                        int c = col_idx(sfj[i2].as_str());
+                        if (c < 0 || plan->cols[c].dtype > PMH_DT_INT64) {
+                            set_error("sequence field '%s': integer value "
+                                      "columns only in v1",
+                                      sfj[i2].as_str().c_str());
+                            return nullptr;
+                        }
+                        if (cg[c] != 0xff) {
+                            set_error("column '%s' is in two sequence "
+                                      "groups", sfj[i2].as_str().c_str());
+                            return nullptr;
+                        }
+                        cg[c] = (uint8_t)g;
+                        sf[g * 4 + i2] = (int16_t)c;
+                    }
+                    ns[g] = (uint8_t)sfj.size();
+                    for (const auto &mf : sg["group_fields"].arr) {
+                        int c = col_idx(mf.as_str());
+                        if (c < 0) {
+                            set_error("sequence-group member '%s' is not a "
+                                      "value column", mf.as_str().c_str());
+                            return nullptr;
+                        }
+                        if (cg[c] != 0xff) {
+                            set_error("column '%s' is in two sequence "
+                                      "groups", mf.as_str().c_str());
+                            return nullptr;
+                        }
+                        cg[c] = (uint8_t)g;
+                    }
+                }
+                plan->n_seq_groups = (int)sgs.arr.size();
+                plan->col_group_dev = (uint8_t *)plan->bufs.alloc(n_cols);
+                plan->sg_fields_dev =
+                    (int16_t *)plan->bufs.alloc(sf.size() * 2);
+                plan->sg_nseq_dev = (uint8_t *)plan->bufs.alloc(ns.size());
+                if (!plan->col_group_dev || !plan->sg_fields_dev ||
+                    !plan->sg_nseq_dev ||
+                    hipMemcpy(plan->col_group_dev, cg.data(), n_cols,
+                              hipMemcpyHostToDevice) != hipSuccess ||
+                    hipMemcpy(plan->sg_fields_dev, sf.data(), sf.size() * 2,
+                              hipMemcpyHostToDevice) != hipSuccess ||
+                    hipMemcpy(plan->sg_nseq_dev, ns.data(), ns.size(),
+                              hipMemcpyHostToDevice) != hipSuccess) {
+                    set_error("H2D of sequence-group tables failed");
+                    return nullptr;
+                }
+            }
         } else if (engine == "first-row") {
             plan->first_row = true;  // FirstRowMergeFunction.java:32-77
         } else if (engine == "aggregation") {
@@ -1977,7 +2060,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     (void)hipEventRecord(ev[2], st);
     int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0) |
                 (p->pu ? 4 : 0) | (p->first_row ? 8 : 0) |
-                (p->rrod ? 16 : 0);
+                (p->rrod ? 16 : 0) | (p->seqg ? 32 : 0);
     if (const char *ab = getenv("PMH_ABLATE"))  // profiling-only phase knob
         flags |= (atoi(ab) & 0xf) << 8;
     if (const char *ab = getenv("PMH_FABL"))  // fused-kernel phase knob
@@ -2075,6 +2158,14 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             flags, sec.winners, sec.group_start, sec.tile_offsets,
             sec.n_tiles, PMH_TILE_ROWS, sec.total_dev, sec.row_masks_dev,
             p->out_ptrs_dev, p->out_valid_dev, st);
+    } else if (p->pu && p->seqg) {
+        e = pmh_launch_emit_pu_sg(
+            sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,
+            p->n_key_cols, p->n_key_cols + 1, flags, p->col_group_dev,
+            p->sg_fields_dev, p->sg_nseq_dev, p->n_seq_groups, sec.winners,
+            sec.group_start, sec.tile_offsets, sec.n_tiles, PMH_TILE_ROWS,
+            sec.total_dev, sec.row_masks_dev, p->out_ptrs_dev,
+            p->out_valid_dev, st);
     } else if (p->pu) {
         e = pmh_launch_emit_pu(
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,

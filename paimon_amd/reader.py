@@ -224,7 +224,7 @@ class MergeReadPlan:
     def __init__(self, session: Session, files, key_cols, value_cols,
                  merge_engine="deduplicate", drop_delete=True,
                  ignore_delete=False, output="host", aggregations=None,
-                 remove_record_on_delete=False):
+                 remove_record_on_delete=False, sequence_groups=None):
         self.lib = session.lib
         desc = {
             "key_cols": key_cols,
@@ -236,6 +236,10 @@ class MergeReadPlan:
             "output": output,
             "files": files,
         }
+        if sequence_groups:
+            # fields.<seq>.sequence-group=<members> (CoreOptions):
+            # [{"sequence_fields": [...], "group_fields": [...]}]
+            desc["sequence_groups"] = list(sequence_groups)
         if aggregations:
             # fields.<name>.aggregate-function (CoreOptions FIELDS_PREFIX);
             # unnamed columns default to last_non_null_value

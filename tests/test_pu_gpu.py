@@ -306,3 +306,93 @@ class TestCompositeKeyPartialUpdate:
             gm = np.ones(len(got["v_c0"]), dtype=bool)
         assert (gm == exp["valid"][0]).all()
         assert (got["v_c0"][gm] == exp["values"][0][gm]).all()
+
+
+class TestSequenceGroups:
+    """PartialUpdate with sequence groups (PartialUpdateMergeFunction.java:
+    219-377) against the oracle sequential port — which is itself pinned to
+    the reference's PartialUpdateMergeFunctionTest vectors in
+    tests/test_seqgroup_cpu.py. Streams carry INSERT/UPDATE_AFTER/
+    UPDATE_BEFORE/DELETE records."""
+
+    def _gen(self, n_runs, rows, seed, n_value_cols=6, retract_frac=0.25):
+        rng = np.random.default_rng(seed)
+        total = n_runs * rows
+        seqs = rng.permutation(total).astype(np.int64)
+        runs = []
+        for r in range(n_runs):
+            key = np.sort(rng.choice(int(rows * n_runs * 0.6), rows,
+                                     replace=False)).astype(np.int64)
+            kind = rng.choice([0, 0, 0, 2, 1, 3], rows,
+                              p=[.45, .1, .1, .1, .1, .15]).astype(np.int8)
+            if retract_frac == 0:
+                kind[:] = 0
+            vals = [key.copy()]
+            msks = [np.ones(rows, bool)]
+            for _ in range(n_value_cols):
+                vals.append(rng.integers(0, 50, rows).astype(np.int32))
+                msks.append(rng.random(rows) > 0.35)
+            runs.append({"key": key, "seq": seqs[r * rows:(r + 1) * rows],
+                         "kind": kind, "values": vals, "valid": msks})
+        return runs
+
+    def _check(self, tmp_path, runs, sgs_idx, sgs_named, drop_delete):
+        from oracle import partial_update_seqgroup_model
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        exp = partial_update_seqgroup_model(runs, sgs_idx,
+                                            drop_delete=drop_delete)
+        n_vals = len(runs[0]["values"])
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(n_vals - 1),
+                               merge_engine="partial-update",
+                               drop_delete=drop_delete,
+                               sequence_groups=sgs_named) as plan:
+                got = _read_all(plan)
+        assert (got["_KEY_k"] == exp["key"]).all()
+        assert (got["_SEQUENCE_NUMBER"] == exp["seq"]).all()
+        assert (got["_VALUE_KIND"] == exp["kind"]).all()
+        names = ["v_k"] + [f"v_c{i}" for i in range(n_vals - 1)]
+        for c, nm in enumerate(names):
+            ev, em = exp["values"][c], exp["valid"][c]
+            gm = got.get(nm + "#valid")
+            if gm is None:
+                gm = np.ones(len(got[nm]), dtype=bool)
+            assert (gm == em).all(), nm
+            assert (got[nm][em] == ev[em]).all(), nm
+
+    # indices into the runs' values list (values[0] = v_k);
+    # names as the plan sees them
+    SG_IDX = [{"sequence_fields": [1], "group_fields": [2, 3]},
+              {"sequence_fields": [4, 5], "group_fields": [6]}]
+    SG_NAMED = [{"sequence_fields": ["v_c0"],
+                 "group_fields": ["v_c1", "v_c2"]},
+                {"sequence_fields": ["v_c3", "v_c4"],
+                 "group_fields": ["v_c5"]}]
+
+    def test_seqgroup_with_retracts(self, tmp_path):
+        runs = self._gen(5, 8_000, seed=201)
+        self._check(tmp_path, runs, self.SG_IDX, self.SG_NAMED, True)
+
+    def test_seqgroup_keep_delete(self, tmp_path):
+        runs = self._gen(4, 6_000, seed=202)
+        self._check(tmp_path / "kd", runs, self.SG_IDX, self.SG_NAMED,
+                    False)
+
+    def test_seqgroup_insert_only(self, tmp_path):
+        runs = self._gen(4, 6_000, seed=203, retract_frac=0.0)
+        self._check(tmp_path, runs, self.SG_IDX, self.SG_NAMED, True)
+
+    def test_seqgroup_validation(self, tmp_path):
+        runs = self._gen(1, 100, seed=204, retract_frac=0.0)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="two sequence groups"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(6),
+                              merge_engine="partial-update",
+                              sequence_groups=[
+                                  {"sequence_fields": ["v_c0"],
+                                   "group_fields": ["v_c1"]},
+                                  {"sequence_fields": ["v_c1"],
+                                   "group_fields": ["v_c2"]}])
