@@ -321,7 +321,8 @@ class TestSequenceGroups:
         seqs = rng.permutation(total).astype(np.int64)
         runs = []
         for r in range(n_runs):
-            key = np.sort(rng.choice(int(rows * n_runs * 0.6), rows,
+            space = max(int(rows * n_runs * 0.6), rows)
+            key = np.sort(rng.choice(space, rows,
                                      replace=False)).astype(np.int64)
             kind = rng.choice([0, 0, 0, 2, 1, 3], rows,
                               p=[.45, .1, .1, .1, .1, .15]).astype(np.int8)
