@@ -167,6 +167,14 @@ int pmh_write_parquet(const pmh_col *cols, int32_t n_cols, int64_t n_rows,
  * independent compressor. */
 int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap);
 
+/* Parse one deletion vector from a DV index file slice (DeletionFile
+ * {path, offset, length}; BitmapDeletionVector.java:98-112 wrapper around
+ * the portable Roaring serialization). Writes up to `cap` deleted
+ * positions ascending; returns the total count or -1. CPU-only debug
+ * entry pinning the parser against independently-serialized fixtures. */
+int64_t pmh_debug_parse_dv(const char *path, int64_t offset, int64_t length,
+                           int64_t *out, int64_t cap);
+
 /* Restatement of IntervalPartition.partition() for int64 keys
  * (mergetree/compact/IntervalPartition.java:67-125): given n files'
  * (minKey, maxKey), writes section id and run-within-section id per file

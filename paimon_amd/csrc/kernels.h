@@ -26,6 +26,11 @@ constexpr uint32_t PMH_ROW_MASK = ((uint32_t)1 << PMH_ROW_BITS) - 1;
 constexpr int PMH_TILE_THREADS = 512;
 constexpr int64_t PMH_TILE_ROWS = 3584;
 constexpr int PMH_COARSE_G = 16;
+// packed-seq sentinel for rows removed by a deletion vector: the merge
+// kernels treat such rows as nonexistent (ApplyDeletionVectorReader
+// semantics). Real packed values are >= 0 (sequence numbers are
+// non-negative counters).
+constexpr int64_t PMH_DEAD = INT64_MIN;
 constexpr int PMH_TILE_MAX = PMH_TILE_ROWS + PMH_MAX_RUNS;
 constexpr int PMH_TILE_ITER =
     (PMH_TILE_MAX + PMH_TILE_THREADS - 1) / PMH_TILE_THREADS;
@@ -107,6 +112,7 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
                                   const DevCol *kinds, const int64_t *lens,
                                   int k, const int32_t *cuts, int64_t n_tiles,
                                   int64_t tile_rows, int flags,
+                                  const uint64_t *tombs,
                                   uint32_t *winners, int32_t *tile_counts,
                                   uint16_t *group_start, uint32_t *err_flag,
                                   hipStream_t stream);
@@ -127,6 +133,7 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  int64_t tile_base, int64_t tile_limit,
                                  int64_t n_tiles,
                                  int64_t tile_rows, int flags,
+                                 const uint64_t *tombs,
                                  const DevCol *cols, const uint8_t *col_dtype,
                                  const uint8_t *col_nullable, int n_cols,
                                  int key_col, int seq_col, int kind_col,

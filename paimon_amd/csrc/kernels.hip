@@ -365,8 +365,9 @@ __launch_bounds__(PMH_TILE_THREADS) __global__
 void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    const int64_t *lens, int k, const int32_t *cuts,
                    int64_t n_tiles, int64_t tile_rows, int flags,
-                   uint32_t *winners, int32_t *tile_counts,
-                   uint16_t *group_start, uint32_t *err_flag) {
+                   const uint64_t *tombs, uint32_t *winners,
+                   int32_t *tile_counts, uint16_t *group_start,
+                   uint32_t *err_flag) {
     const bool drop_delete = flags & 1;
     const bool ignore_delete = flags & 2;
     // FR = first-row engine (FirstRowMergeFunction.java:32-77): keep the
@@ -455,14 +456,20 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                 reinterpret_cast<const int64_t *>(seqs[r].addr0) + base;
             const int32_t *daddr =
                 reinterpret_cast<const int32_t *>(kinds[r].addr0) + base;
+            const uint8_t *tb =
+                tombs && tombs[r]
+                    ? reinterpret_cast<const uint8_t *>(tombs[r]) + base
+                    : nullptr;
             for (int32_t i = tid; i < len; i += blockDim.x) {
                 sm.skey[off + i] = key_at(kaddr0, i, kes);
                 int32_t kd = daddr[i];
-                if (PU && rrod && !seqg && kd == 1 && err_flag)
+                const bool dead = tb && tb[i];
+                if (PU && rrod && !seqg && kd == 1 && !dead && err_flag)
                     atomicOr(err_flag, 1u);  // UPDATE_BEFORE: not in v1 RROD
                                              // (sequence groups accept it)
                 sm.sseq[off + i] =
-                    (saddr[i] << 1) | (int64_t)(kd == 0 || kd == 2);
+                    dead ? PMH_DEAD
+                         : (saddr[i] << 1) | (int64_t)(kd == 0 || kd == 2);
                 sm.perm[0][off + i] = (uint16_t)(off + i);
             }
         }
@@ -567,49 +574,63 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                 int32_t ng = 0, nm = 0;
                 for (int32_t i = my_lo; i < my_hi; i++) {
                     if (!sm.head[i]) continue;
+                    // pre-scan the group: LIVE members only (deletion-
+                    // vector tombstones stage as PMH_DEAD and never reach
+                    // the merge — ApplyDeletionVectorReader semantics)
                     int32_t tail = i;
-                    while (tail + 1 < M && !sm.head[tail + 1]) tail++;
-                    if (tail == i) {
+                    int32_t nlive = 0;
+                    int64_t mx = INT64_MIN;
+                    bool any_add = false;
+                    uint16_t live1 = 0;
+                    for (int32_t x = i;; x++) {
+                        int64_t v = sm.sseq[mo[x]];
+                        if (v != PMH_DEAD) {
+                            if (nlive == 0) live1 = mo[x];
+                            nlive++;
+                            if (v > mx) mx = v;
+                            any_add |= ps2m_isadd(v);
+                        }
+                        if (!(x + 1 < M && !sm.head[x + 1])) {
+                            tail = x;
+                            break;
+                        }
+                    }
+                    if (nlive == 0) continue;
+                    if (nlive == 1) {
                         // ReducerMergeFunctionWrapper singleton bypass
                         // (ReducerMergeFunctionWrapper.java:53-73): the lone
                         // record is served as-is (any kind, incl. retracts);
                         // a retract result drops under drop-delete
                         // (DropDeleteReader.java:53-61)
-                        if (drop_delete && !ps2m_isadd(sm.sseq[mo[i]]))
+                        if (drop_delete && !ps2m_isadd(sm.sseq[live1]))
                             continue;
                     } else if (seqg && drop_delete) {
                         // result kind DELETE iff no add member
-                        bool any_add = false;
-                        for (int32_t x = i; x <= tail; x++)
-                            any_add |= ps2m_isadd(sm.sseq[mo[x]]);
                         if (!any_add) continue;
                     } else if (rrod && drop_delete) {
-                        // result kind = last member's kind (max packed sseq);
-                        // DELETE results drop here (DropDeleteReader)
-                        int64_t mx = INT64_MIN;
-                        for (int32_t x = i; x <= tail; x++) {
-                            int64_t v = sm.sseq[mo[x]];
-                            if (v > mx) mx = v;
-                        }
+                        // result kind = last member's kind (max packed
+                        // sseq); DELETE results drop here (DropDeleteReader)
                         if (!ps2m_isadd(mx)) continue;
                     }
                     if (pass == 1) {
                         gout[g_off + ng] = (uint16_t)(m_off + nm);
                         // merged order within a group is (key, run); the
                         // overlay consumes records in ascending (seq, isAdd)
-                        // order — insertion-sort the <= k members by the
-                        // packed sseq.
+                        // order — insertion-sort the <= k LIVE members by
+                        // the packed sseq.
                         uint16_t gm[PMH_MAX_RUNS];
-                        int gn = tail - i + 1;
-                        for (int x = 0; x < gn; x++) {
-                            uint16_t s = mo[i + x];
-                            int y = x;
+                        int gn = 0;
+                        for (int32_t x2 = i; x2 <= tail; x2++) {
+                            uint16_t s = mo[x2];
+                            if (sm.sseq[s] == PMH_DEAD) continue;
+                            int y = gn;
                             while (y > 0 &&
                                    sm.sseq[gm[y - 1]] > sm.sseq[s]) {
                                 gm[y] = gm[y - 1];
                                 y--;
                             }
                             gm[y] = s;
+                            gn++;
                         }
                         for (int x = 0; x < gn; x++) {
                             uint16_t s = gm[x];
@@ -631,7 +652,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                         }
                     }
                     ng++;
-                    nm += tail - i + 1;
+                    nm += nlive;
                 }
                 if (pass == 0) {
                     // dual wave scan of (ng, nm)
@@ -696,33 +717,35 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             int32_t nloc = 0;
             for (int32_t i = my_lo; i < my_hi; i++) {
                 if (!sm.head[i]) continue;
-                // walk the group, tracking the winner by (elig, sseq):
+                // walk the group's LIVE members (deletion-vector tombstones
+                // stage as PMH_DEAD), tracking the winner by (elig, sseq):
                 // deduplicate keeps the LAST record, first-row the FIRST
-                int32_t tail = i;
-                uint16_t s_best = mo[i];
-                int64_t v_best = sm.sseq[s_best];
-                bool e_best = !ignore_delete || (v_best & 1);
-                bool any_retract = !(v_best & 1);
-                while (tail + 1 < M && !sm.head[tail + 1]) {
-                    tail++;
-                    uint16_t s = mo[tail];
-                    int64_t v = sm.sseq[s];
-                    bool e = !ignore_delete || (v & 1);
-                    any_retract |= !(v & 1);
-                    bool take = (e && !e_best) ||
-                                (e == e_best &&
-                                 (first_row ? v < v_best : v > v_best));
-                    if (take) {
-                        s_best = s;
-                        v_best = v;
-                        e_best = e;
+                uint16_t s_best = 0;
+                int64_t v_best = 0;
+                bool e_best = false, any_retract = false;
+                int32_t nlive = 0;
+                for (int32_t x = i;; x++) {
+                    int64_t v = sm.sseq[mo[x]];
+                    if (v != PMH_DEAD) {
+                        bool e = !ignore_delete || (v & 1);
+                        any_retract |= !(v & 1);
+                        bool take = nlive == 0 || (e && !e_best) ||
+                                    (e == e_best &&
+                                     (first_row ? v < v_best : v > v_best));
+                        if (take) {
+                            s_best = mo[x];
+                            v_best = v;
+                            e_best = e;
+                        }
+                        nlive++;
                     }
+                    if (!(x + 1 < M && !sm.head[x + 1])) break;
                 }
-                int32_t gsize = tail - i + 1;
-                if (first_row && !ignore_delete && any_retract && gsize > 1 &&
+                if (nlive == 0) continue;
+                if (first_row && !ignore_delete && any_retract && nlive > 1 &&
                     err_flag)
                     atomicOr(err_flag, 2u);  // FirstRow rejects retracts
-                if (!e_best && gsize > 1) continue;  // all records ignored
+                if (!e_best && nlive > 1) continue;  // all records ignored
                 if (drop_delete && !(v_best & 1)) continue;
                 if (pass == 1) {
                     int r = 0;  // map seg index -> (run, global row)
@@ -833,7 +856,7 @@ __launch_bounds__(PMH_TILE_THREADS, 2) __global__
 void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                   const int64_t *lens, int k, const int32_t *cuts,
                   int64_t tile_base, int64_t tile_limit, int64_t n_tiles,
-                  int64_t tile_rows, int flags,
+                  int64_t tile_rows, int flags, const uint64_t *tombs,
                   const DevCol *cols /* k * n_cols, run-major */,
                   const uint8_t *col_dtype, const uint8_t *col_nullable,
                   int n_cols, int key_col /* -1: composite */, int seq_col,
@@ -870,31 +893,36 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             int32_t nloc = 0;
             for (int32_t i = lo_i; i < hi_i; i++) {
                 if (!sm.head[i]) continue;
-                int32_t tail = i;
-                uint16_t s_best = mo[i];
-                int64_t v_best = sm.sseq[s_best];
-                bool e_best = !ignore_delete || ps2_isadd(v_best);
-                bool any_retract = !ps2_isadd(v_best);
-                while (tail + 1 < M && !sm.head[tail + 1]) {
-                    tail++;
-                    uint16_t s = mo[tail];
-                    int64_t v = sm.sseq[s];
-                    bool e = !ignore_delete || ps2_isadd(v);
-                    any_retract |= !ps2_isadd(v);
-                    bool take = (e && !e_best) ||
-                                (e == e_best &&
-                                 (FR ? v < v_best : v > v_best));
-                    if (take) {
-                        s_best = s;
-                        v_best = v;
-                        e_best = e;
+                // walk the group's members, skipping deletion-vector
+                // tombstones (PMH_DEAD) — the reference's reader never
+                // shows deleted rows to the merge, so LIVE members define
+                // group size for the wrapper's singleton bypass too
+                uint16_t s_best = 0;
+                int64_t v_best = 0;
+                bool e_best = false, any_retract = false;
+                int32_t nlive = 0;
+                for (int32_t x = i;; x++) {
+                    int64_t v = sm.sseq[mo[x]];
+                    if (v != PMH_DEAD) {
+                        bool e = !ignore_delete || ps2_isadd(v);
+                        any_retract |= !ps2_isadd(v);
+                        bool take =
+                            nlive == 0 || (e && !e_best) ||
+                            (e == e_best && (FR ? v < v_best : v > v_best));
+                        if (take) {
+                            s_best = mo[x];
+                            v_best = v;
+                            e_best = e;
+                        }
+                        nlive++;
                     }
+                    if (!(x + 1 < M && !sm.head[x + 1])) break;
                 }
-                int32_t gsize = tail - i + 1;
-                if (FR && !ignore_delete && any_retract && gsize > 1 &&
+                if (nlive == 0) continue;
+                if (FR && !ignore_delete && any_retract && nlive > 1 &&
                     err_flag)
                     atomicOr(err_flag, 2u);
-                if (!e_best && gsize > 1) continue;
+                if (!e_best && nlive > 1) continue;
                 if (drop_delete && !ps2_isadd(v_best)) continue;
                 emit(nloc, s_best);
                 nloc++;
@@ -936,7 +964,9 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         const int32_t M = sm.mtotal;
         const int32_t Mreal = sm.mreal;
         if (Mreal > 0) {
-            // --- stage key / packed-seq segments (coalesced per run)
+            // --- stage key / packed-seq segments (coalesced per run);
+            // deletion-vector tombstones stage as PMH_DEAD (the walk skips
+            // them — ApplyDeletionVectorReader semantics)
             for (int r = 0; r < k; r++) {
                 int32_t off = sm.segoff[r], len = sm.seglen[r];
                 int64_t base = c0[r];
@@ -946,9 +976,15 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                     reinterpret_cast<const int64_t *>(seqs[r].addr0) + base;
                 const int32_t *daddr =
                     reinterpret_cast<const int32_t *>(kinds[r].addr0) + base;
+                const uint8_t *tb =
+                    tombs && tombs[r]
+                        ? reinterpret_cast<const uint8_t *>(tombs[r]) + base
+                        : nullptr;
                 for (int32_t i = tid; i < len; i += blockDim.x) {
                     sm.skey[off + i] = key_at(kaddr0, i, kes);
-                    sm.sseq[off + i] = ps2_pack(saddr[i], daddr[i]);
+                    sm.sseq[off + i] = (tb && tb[i])
+                                           ? PMH_DEAD
+                                           : ps2_pack(saddr[i], daddr[i]);
                     sm.perm[0][off + i] = (uint16_t)(off + i);
                 }
             }
@@ -2722,6 +2758,7 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
                                   const DevCol *kinds, const int64_t *lens,
                                   int k, const int32_t *cuts, int64_t n_tiles,
                                   int64_t tile_rows, int flags,
+                                  const uint64_t *tombs,
                                   uint32_t *winners, int32_t *tile_counts,
                                   uint16_t *group_start, uint32_t *err_flag,
                                   hipStream_t stream) {
@@ -2730,7 +2767,7 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
     auto launch = [&](auto kern) {
         hipLaunchKernelGGL(kern, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
                            stream, keys, seqs, kinds, lens, k, cuts, n_tiles,
-                           tile_rows, flags, winners, tile_counts,
+                           tile_rows, flags, tombs, winners, tile_counts,
                            group_start, err_flag);
     };
     if (pu)
@@ -2840,6 +2877,7 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  int64_t tile_base, int64_t tile_limit,
                                  int64_t n_tiles,
                                  int64_t tile_rows, int flags,
+                                 const uint64_t *tombs,
                                  const DevCol *cols, const uint8_t *col_dtype,
                                  const uint8_t *col_nullable, int n_cols,
                                  int key_col, int seq_col, int kind_col,
@@ -2858,7 +2896,7 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
         hipLaunchKernelGGL(kern, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
                            stream, keys, seqs, kinds, lens, k, cuts,
                            tile_base, tile_limit, n_tiles, tile_rows, flags,
-                           cols, col_dtype, col_nullable,
+                           tombs, cols, col_dtype, col_nullable,
                            n_cols, key_col, seq_col, kind_col, status, ticket,
                            total_out, dense_winners, out_ptrs, out_valid,
                            err_flag);

@@ -61,9 +61,134 @@ static bool zstd_decompress(const uint8_t *src, size_t src_n, uint8_t *dst,
     return true;
 }
 
+
+// ---------------------------------------------------- deletion vectors
+//
+// Paimon deletion vectors: per data file, a RoaringBitmap32
+// of deleted row positions stored in a DV index file at (offset, length)
+// (DeletionFile; deletionvectors/BitmapDeletionVector.java:98-112 wrapper =
+// [i32 BE size][i32 BE magic 1581511376][roaring bytes][i32 BE crc];
+// DeletionVector.read :101-118). The bitmap itself is the portable Roaring
+// serialization (little-endian; cookie 12347 = no run containers, cookie
+// low-16 12346 = with runs + bitset of run containers).
+static bool parse_roaring32(const uint8_t *p, int64_t len,
+                            std::vector<uint32_t> &out, std::string &err) {
+    auto rd16 = [&](int64_t o) { return (uint32_t)p[o] | ((uint32_t)p[o + 1] << 8); };
+    auto rd32 = [&](int64_t o) {
+        return (uint32_t)p[o] | ((uint32_t)p[o + 1] << 8) |
+               ((uint32_t)p[o + 2] << 16) | ((uint32_t)p[o + 3] << 24);
+    };
+    if (len < 8) { err = "roaring: short"; return false; }
+    uint32_t cookie = rd32(0);
+    int64_t o = 4;
+    int32_t n_cont;
+    bool has_run = false;
+    std::vector<uint8_t> run_flags;
+    if ((cookie & 0xFFFF) == 12346) {
+        has_run = true;
+        n_cont = (int32_t)(cookie >> 16) + 1;
+        int64_t rb = (n_cont + 7) / 8;
+        run_flags.assign(p + o, p + o + rb);
+        o += rb;
+    } else if (cookie == 12347) {
+        n_cont = (int32_t)rd32(o);
+        o += 4;
+    } else {
+        err = "roaring: bad cookie";
+        return false;
+    }
+    std::vector<uint32_t> keys(n_cont), cards(n_cont);
+    for (int i = 0; i < n_cont; i++) {
+        keys[i] = rd16(o);
+        cards[i] = rd16(o + 2) + 1;
+        o += 4;
+    }
+    const bool has_offsets = !has_run || n_cont >= 4;
+    if (has_offsets) o += 4 * (int64_t)n_cont;  // container offsets (unused)
+    for (int i = 0; i < n_cont; i++) {
+        const uint32_t hi = keys[i] << 16;
+        const bool is_run =
+            has_run && (run_flags[i / 8] >> (i % 8)) & 1;
+        if (is_run) {
+            uint32_t n_runs = rd16(o);
+            o += 2;
+            for (uint32_t r = 0; r < n_runs; r++) {
+                uint32_t start = rd16(o), rl = rd16(o + 2);
+                o += 4;
+                for (uint32_t v = start; v <= start + rl; v++)
+                    out.push_back(hi | v);
+            }
+        } else if (cards[i] > 4096) {  // bitmap container: 8 KB
+            for (int w = 0; w < 1024; w++) {
+                uint64_t word = 0;
+                for (int b = 0; b < 8; b++)
+                    word |= (uint64_t)p[o + w * 8 + b] << (8 * b);
+                while (word) {
+                    int bit = __builtin_ctzll(word);
+                    out.push_back(hi | (uint32_t)(w * 64 + bit));
+                    word &= word - 1;
+                }
+            }
+            o += 8192;
+        } else {  // array container
+            for (uint32_t v = 0; v < cards[i]; v++)
+                out.push_back(hi | rd16(o + 2 * v));
+            o += 2 * (int64_t)cards[i];
+        }
+        if (o > len) { err = "roaring: overrun"; return false; }
+    }
+    return true;
+}
+
+// Read one file's DV from its index file slice; returns deleted positions.
+static bool load_deletion_vector(const std::string &path, int64_t offset,
+                                 int64_t length,
+                                 std::vector<uint32_t> &out) {
+    FILE *f = fopen(path.c_str(), "rb");
+    if (!f) {
+        set_error("cannot open deletion vector file %s", path.c_str());
+        return false;
+    }
+    if (length <= 0) {  // whole file
+        fseek(f, 0, SEEK_END);
+        length = ftell(f) - offset;
+    }
+    std::vector<uint8_t> buf(length);
+    fseek(f, offset, SEEK_SET);
+    bool ok = fread(buf.data(), 1, length, f) == (size_t)length;
+    fclose(f);
+    if (!ok || length < 12) {
+        set_error("deletion vector read failed (%s @%lld+%lld)",
+                  path.c_str(), (long long)offset, (long long)length);
+        return false;
+    }
+    auto be32 = [&](int64_t o) {
+        return ((uint32_t)buf[o] << 24) | ((uint32_t)buf[o + 1] << 16) |
+               ((uint32_t)buf[o + 2] << 8) | (uint32_t)buf[o + 3];
+    };
+    uint32_t size = be32(0);
+    if (be32(4) != 1581511376u) {
+        set_error("deletion vector magic mismatch in %s", path.c_str());
+        return false;
+    }
+    if ((int64_t)size + 8 > length) {
+        set_error("deletion vector truncated in %s", path.c_str());
+        return false;
+    }
+    std::string err;
+    if (!parse_roaring32(buf.data() + 8, size - 4, out, err)) {
+        set_error("%s (%s)", err.c_str(), path.c_str());
+        return false;
+    }
+    return true;
+}
+
 // ------------------------------------------- IntervalPartition (restated)
 
 struct FileDesc {
+    std::string dv_path;  // deletion vector (DeletionFile); empty = none
+    int64_t dv_offset = 0;
+    int64_t dv_length = 0;
     std::string path;
     int64_t row_count = 0;
     int64_t min_key = 0;
@@ -225,6 +350,7 @@ struct RunCol {
 struct Run {
     int64_t length = 0;
     std::vector<RunCol> cols;
+    uint8_t *tomb = nullptr;  // deletion-vector tombstones (1 = deleted)
 };
 
 struct Section {
@@ -249,6 +375,10 @@ struct Section {
     uint64_t *status = nullptr;
     uint64_t *ticket = nullptr;
     uint32_t *dense_winners = nullptr;  // split mode (PMH_FSPLIT)
+    // per-run tombstone byte arrays (deletion vectors; null entries = no
+    // DV) + the device pointer table the merge kernels consume
+    uint64_t *tombs_dev = nullptr;
+    bool any_tomb = false;
     // packed per-row validity (PU/agg emit): one u64 per row per run,
     // bit c = column c non-null; built once per section by k_pack_valid
     std::vector<uint64_t *> row_masks;
@@ -1052,6 +1182,37 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
         }
         plan->encoded_bytes_total += total_rows * cols[c].stored_esize;
     }
+    std::vector<uint8_t> tomb_host;
+    {
+        int64_t rb = 0;
+        for (const auto &fd : files) {
+            if (!fd.dv_path.empty()) {
+                // ApplyDeletionVectorReader semantics (io/
+                // KeyValueFileReaderFactory.java:139-143): deleted positions
+                // never reach the merge — staged as per-run tombstones the
+                // merge kernels treat as nonexistent rows
+                std::vector<uint32_t> pos;
+                if (!load_deletion_vector(fd.dv_path, fd.dv_offset,
+                                          fd.dv_length, pos))
+                    return false;
+                if (tomb_host.empty()) tomb_host.assign(total_rows, 0);
+                for (uint32_t pp : pos) {
+                    if ((int64_t)pp < fd.row_count)
+                        tomb_host[rb + pp] = 1;
+                }
+            }
+            rb += fd.row_count;
+        }
+    }
+    if (!tomb_host.empty()) {
+        run.tomb = (uint8_t *)plan->bufs.alloc(total_rows);
+        if (!run.tomb ||
+            hipMemcpy(run.tomb, tomb_host.data(), total_rows,
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_error("H2D tombstones failed");
+            return false;
+        }
+    }
     int64_t row_base = 0;
     for (const auto &fd : files) {
         StagedFile sf;
@@ -1499,6 +1660,17 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
     sec.total_dev = (int64_t *)plan->bufs.alloc(8);
     sec.err_dev = (uint32_t *)plan->bufs.alloc(4);
     if (sec.err_dev) (void)hipMemset(sec.err_dev, 0, 4);
+    {
+        std::vector<uint64_t> th(k, 0);
+        for (int r = 0; r < k; r++) {
+            th[r] = (uint64_t)sec.runs[r].tomb;
+            sec.any_tomb |= sec.runs[r].tomb != nullptr;
+        }
+        if (sec.any_tomb) {
+            sec.tombs_dev = (uint64_t *)up(th.data(), k * 8);
+            if (!sec.tombs_dev) return false;
+        }
+    }
     if (plan->pu) {
         sec.group_start = (uint16_t *)plan->bufs.alloc(
             sec.n_tiles * (PMH_TILE_ROWS + 1) * 2);
@@ -1793,6 +1965,12 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             fd.max_key = fj["maxKey"].as_i64();
             fd.level = (int)fj["level"].as_i64();
             fd.input_index = idx++;
+            const Json &dv = fj["deletionVector"];
+            if (!dv.obj.empty()) {
+                fd.dv_path = dv["file"].as_str();
+                fd.dv_offset = dv["offset"].as_i64();
+                fd.dv_length = dv["length"].as_i64();
+            }
             if (fd.row_count >= ((int64_t)1 << PMH_ROW_BITS)) {
                 set_error("file %s exceeds 2^%d rows-per-run limit",
                           fd.path.c_str(), PMH_ROW_BITS);
@@ -2103,6 +2281,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                 e = pmh_launch_merge_emit(
                     sec.key_cols, sec.seq_cols, sec.kind_cols, sec.lens_dev,
                     k, sec.cuts, t0, t1, sec.n_tiles, PMH_TILE_ROWS, flags,
+                    sec.tombs_dev,
                     sec.all_cols, p->col_dtype_dev, p->col_nullable_dev,
                     n_cols, key_col, p->n_key_cols, p->n_key_cols + 1,
                     sec.status, sec.ticket, sec.total_dev, sec.dense_winners,
@@ -2142,7 +2321,8 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     }
     e = pmh_launch_merge_tiles(sec.key_cols, sec.seq_cols, sec.kind_cols,
                                sec.lens_dev, k, sec.cuts, sec.n_tiles,
-                               PMH_TILE_ROWS, flags, sec.winners,
+                               PMH_TILE_ROWS, flags, sec.tombs_dev,
+                               sec.winners,
                                sec.tile_counts, sec.group_start, sec.err_dev,
                                st);
     if (e != hipSuccess) return fail("merge_tiles", e);
@@ -2320,6 +2500,15 @@ int pmh_stats_get(pmh_plan_t *p, pmh_stats *out) {
 }
 
 void pmh_free_string(char *s) { free(s); }
+
+int64_t pmh_debug_parse_dv(const char *path, int64_t offset, int64_t length,
+                           int64_t *out, int64_t cap) {
+    std::vector<uint32_t> pos;
+    if (!load_deletion_vector(path, offset, length, pos)) return -1;
+    int64_t n = (int64_t)pos.size() < cap ? (int64_t)pos.size() : cap;
+    for (int64_t i = 0; i < n; i++) out[i] = pos[i];
+    return (int64_t)pos.size();
+}
 
 int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap) {
     size_t got = 0;

@@ -89,6 +89,10 @@ def load_lib(required=True):
     lib.pmh_write_parquet.argtypes = [
         ctypes.POINTER(_Col), ctypes.c_int32, ctypes.c_int64,
         ctypes.c_char_p, ctypes.c_int64, ctypes.c_int64, ctypes.c_char_p]
+    lib.pmh_debug_parse_dv.restype = ctypes.c_int64
+    lib.pmh_debug_parse_dv.argtypes = [
+        ctypes.c_char_p, ctypes.c_int64, ctypes.c_int64,
+        ctypes.POINTER(ctypes.c_int64), ctypes.c_int64]
     lib.pmh_debug_interval_partition.restype = ctypes.c_int
     lib.pmh_debug_interval_partition.argtypes = [
         ctypes.c_int, ctypes.POINTER(ctypes.c_int64),
@@ -181,6 +185,17 @@ def write_parquet(path, columns, row_group_rows=0, page_rows=0,
                                compression.encode())
     if rc != 0:
         raise RuntimeError(f"pmh_write_parquet: {last_error()}")
+
+
+def debug_parse_dv(path, offset=0, length=0, cap=1 << 22):
+    lib = load_lib()
+    out = np.empty(cap, dtype=np.int64)
+    n = lib.pmh_debug_parse_dv(
+        path.encode(), offset, length,
+        out.ctypes.data_as(ctypes.POINTER(ctypes.c_int64)), cap)
+    if n < 0:
+        raise RuntimeError(last_error())
+    return out[:min(n, cap)].copy()
 
 
 def interval_partition(min_keys, max_keys):
@@ -317,6 +332,13 @@ class MergeReadPlan:
 
 
 def file_descs_from_metas(metas):
-    return [{"path": m["path"], "rowCount": m["rowCount"],
+    out = []
+    for m in metas:
+        d = {"path": m["path"], "rowCount": m["rowCount"],
              "minKey": m["minKey"], "maxKey": m["maxKey"],
-             "level": m.get("level", 0)} for m in metas]
+             "level": m.get("level", 0)}
+        if m.get("deletionVector"):
+            # DeletionFile {path, offset, length}
+            d["deletionVector"] = m["deletionVector"]
+        out.append(d)
+    return out
