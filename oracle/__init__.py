@@ -15,6 +15,7 @@ from .oracle import (  # noqa: F401
     partial_update_seqgroup_model,
     aggregation_model,
     aggregation_rrod_model,
+    aggregation_retract_model,
     rle_bp_decode,
     lib_path,
 )

@@ -215,7 +215,11 @@ enum {
     PMH_AGG_SUM = 4,            // FieldSumAgg
     PMH_AGG_MAX = 5,            // FieldMaxAgg
     PMH_AGG_MIN = 6,            // FieldMinAgg
+    PMH_AGG_PRIMARY_KEY = 7,    // FieldPrimaryKeyAgg (agg = retract = input)
 };
+// col_agg bit 0x80: FieldIgnoreRetractAgg wrapper (fields.<f>.ignore-retract
+// = true): retract records leave the accumulator untouched
+constexpr uint8_t PMH_AGG_IGNORE_RETRACT = 0x80;
 // PartialUpdate with sequence groups (PartialUpdateMergeFunction.java:
 // 219-377): per-group last-prefix-max-achiever resolution; retracts null
 // their group members. col_group[c] = group id or 0xff; sg_fields packs 4

@@ -385,6 +385,10 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
     // are legal members (they retract their groups); result kind is DELETE
     // only when the group has NO add member (getResult :389-397)
     const bool seqg = (flags & 32) != 0;
+    // aggregation with retract-capable aggregators (FieldSumAgg.retract /
+    // FieldPrimaryKeyAgg / ignore-retract wrappers): retract members are
+    // legal and fold through aggregator.retract in k_emit_agg
+    const bool aggr = (flags & 64) != 0;
     // ablation levels (profiling only, flags bits 8..): 1=stage,2=+merge,
     // 3=+scan, 0/absent=full. Partial levels publish a checksum so the
     // compiler cannot dead-code the ablated phases' inputs.
@@ -639,7 +643,7 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                             // singletons (gn == 1), so a lone retract
                             // passes; sequence groups accept retracts
                             // outright (retractWithSequenceGroup)
-                            if (!rrod && !seqg && gn > 1 &&
+                            if (!rrod && !seqg && !aggr && gn > 1 &&
                                 !ps2m_isadd(sm.sseq[s]))
                                 bad_kind = true;
                             int r = 0;
@@ -2265,6 +2269,10 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 if (MASKS) vdel = run_masks[mdel >> PMH_ROW_BITS][mdel & PMH_ROW_MASK];
             }
         }
+        // retract mode (flags bit 64): per-member kinds drive
+        // aggregator.retract; only reached when every column's aggregator
+        // is retract-capable (plan validation)
+        const bool aggr = (flags & 64) != 0;
         for (int c = 0; c < n_cols; c++) {
             if (c == kind_col) {
                 int32_t kout;
@@ -2279,7 +2287,13 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 continue;
             }
             const int dt = col_dtype[c];
-            const int agg = gn == 1 ? PMH_AGG_LAST_VALUE : col_agg[c];
+            const uint8_t rawagg = col_agg[c];
+            const bool ign_retract =
+                (rawagg & PMH_AGG_IGNORE_RETRACT) != 0;
+            int agg = rawagg & 0x7f;
+            if (agg == PMH_AGG_PRIMARY_KEY)
+                agg = PMH_AGG_LAST_VALUE;  // agg == retract == input
+            if (gn == 1) agg = PMH_AGG_LAST_VALUE;
             auto del_valid = [&]() -> uint8_t {
                 if (!col_nullable[c]) return 1;
                 if (MASKS) return (uint8_t)((vdel >> c) & 1);
@@ -2292,7 +2306,38 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
             uint8_t ok = 1;
             int64_t bits = 0;  // result payload (raw stored bits / int value)
             bool direct = true;  // bits not set; load from (run,row) at store
-            switch (agg) {
+            bool handled = false;
+            if (aggr && ign_retract && gn > 1 &&
+                (agg == PMH_AGG_LAST_VALUE || agg == PMH_AGG_FIRST_VALUE ||
+                 agg == PMH_AGG_LAST_NON_NULL ||
+                 agg == PMH_AGG_FIRST_NON_NULL)) {
+                // FieldIgnoreRetractAgg wrapper: retract members leave the
+                // accumulator untouched, so only ADD members qualify
+                auto is_rt = [&](uint32_t m) -> bool {
+                    const DevCol &dc =
+                        cols[(m >> PMH_ROW_BITS) * n_cols + kind_col];
+                    const int32_t kd =
+                        col_load<int32_t>(dc, m & PMH_ROW_MASK);
+                    return kd == 1 || kd == 3;
+                };
+                const bool want_first = agg == PMH_AGG_FIRST_VALUE ||
+                                        agg == PMH_AGG_FIRST_NON_NULL;
+                const bool need_nn = agg == PMH_AGG_LAST_NON_NULL ||
+                                     agg == PMH_AGG_FIRST_NON_NULL;
+                ok = 0;
+                for (int32_t x = 0; x < gn; x++) {
+                    uint32_t m = mem[ms + x];
+                    if (is_rt(m)) continue;
+                    uint8_t v = col_nullable[c] ? valid_of(m, c) : 1;
+                    if (need_nn && !v) continue;
+                    run = m >> PMH_ROW_BITS;
+                    row = m & PMH_ROW_MASK;
+                    ok = need_nn ? 1 : v;
+                    if (want_first) break;
+                }
+                handled = true;
+            }
+            if (!handled) switch (agg) {
             case PMH_AGG_LAST_VALUE:
                 // last member as-is, its own validity (also the singleton
                 // ReducerMergeFunctionWrapper bypass)
@@ -2373,7 +2418,7 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                 int64_t iacc = 0;
                 float facc = 0.f;
                 double dacc = 0.0;
-                auto fold_one = [&](uint32_t m) {
+                auto fold_one = [&](uint32_t m, bool retract) {
                     const DevCol &dc = cols[(m >> PMH_ROW_BITS) * n_cols + c];
                     const int64_t r = m & PMH_ROW_MASK;
                     int64_t vb = (dt == 4 || dt == 6)
@@ -2381,18 +2426,27 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                                      : (int64_t)col_load<int32_t>(dc, r);
                     if (!ok) {
                         ok = 1;
-                        if (agg == PMH_AGG_SUM && dt == 5)
+                        if (agg == PMH_AGG_SUM && dt == 5) {
                             facc = __int_as_float((int32_t)vb);
-                        else if (agg == PMH_AGG_SUM && dt == 6)
+                            if (retract) facc = -facc;  // null acc: negate
+                        } else if (agg == PMH_AGG_SUM && dt == 6) {
                             dacc = __longlong_as_double(vb);
-                        else
-                            iacc = vb;  // raw bits for float max/min
+                            if (retract) dacc = -dacc;
+                        } else {
+                            iacc = retract && agg == PMH_AGG_SUM ? -vb : vb;
+                        }
                         return;
                     }
                     if (agg == PMH_AGG_SUM) {
-                        if (dt == 5) facc += __int_as_float((int32_t)vb);
-                        else if (dt == 6) dacc += __longlong_as_double(vb);
-                        else iacc += vb;
+                        // FieldSumAgg.agg / .retract (:60-110): subtract on
+                        // retract, same width/precision rules
+                        const double sgn = retract ? -1.0 : 1.0;
+                        if (dt == 5)
+                            facc += (float)sgn * __int_as_float((int32_t)vb);
+                        else if (dt == 6)
+                            dacc += sgn * __longlong_as_double(vb);
+                        else
+                            iacc += retract ? -vb : vb;
                     } else {
                         bool take;
                         if (dt == 5) {
@@ -2409,8 +2463,16 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                         if (take) iacc = vb;
                     }
                 };
+                auto is_retract = [&](uint32_t m) -> bool {
+                    if (!aggr) return false;
+                    const DevCol &dc =
+                        cols[(m >> PMH_ROW_BITS) * n_cols + kind_col];
+                    const int32_t kd =
+                        col_load<int32_t>(dc, m & PMH_ROW_MASK);
+                    return kd == 1 || kd == 3;
+                };
                 if (d_del >= 0 && del_valid())
-                    fold_one(mdel);  // aggregators continue from initRow
+                    fold_one(mdel, false);  // initRow continuation
 #pragma unroll
                 for (int x = 0; x < 4; x++) {
                     if (x >= gn || x <= d_del) continue;
@@ -2418,13 +2480,17 @@ __global__ void k_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                         !(MASKS ? (uint8_t)((vma[x] >> c) & 1)
                                 : valid_of(ma[x], c)))
                         continue;
-                    fold_one(ma[x]);
+                    const bool rt = is_retract(ma[x]);
+                    if (rt && ign_retract) continue;  // FieldIgnoreRetractAgg
+                    fold_one(ma[x], rt);
                 }
                 for (int32_t x = 4; x < gn; x++) {
                     if (x <= d_del) continue;
                     uint32_t m = mem[ms + x];
                     if (col_nullable[c] && !valid_of(m, c)) continue;
-                    fold_one(m);
+                    const bool rt = is_retract(m);
+                    if (rt && ign_retract) continue;
+                    fold_one(m, rt);
                 }
                 if (agg == PMH_AGG_SUM && dt == 5)
                     bits = (int64_t)__float_as_int(facc);

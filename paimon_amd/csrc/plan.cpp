@@ -427,6 +427,7 @@ struct pmh_plan_t {
     // PartialUpdateMergeFunction.java:219-377): retracts become legal and
     // act on their groups
     bool seqg = false;
+    bool agg_retract = false;  // every aggregator retract-capable
     int n_seq_groups = 0;
     uint8_t *col_group_dev = nullptr;
     int16_t *sg_fields_dev = nullptr;
@@ -2094,6 +2095,7 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                 if (s == "sum") return PMH_AGG_SUM;
                 if (s == "max") return PMH_AGG_MAX;
                 if (s == "min") return PMH_AGG_MIN;
+                if (s == "primary_key") return PMH_AGG_PRIMARY_KEY;
                 return -1;
             };
             const int first_val = plan->n_key_cols + 2;
@@ -2132,6 +2134,35 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                     return nullptr;
                 }
                 ca[idx] = (uint8_t)code;
+            }
+            // fields.<f>.ignore-retract = true (FieldIgnoreRetractAgg):
+            // retract records leave that field's accumulator untouched
+            for (const auto &irj : j["ignore_retract"].arr) {
+                int idx = -1;
+                for (int c = first_val; c < n_cols; c++)
+                    if (plan->cols[c].name == irj.as_str()) idx = c;
+                if (idx < 0) {
+                    set_error("ignore_retract: '%s' is not a value column",
+                              irj.as_str().c_str());
+                    return nullptr;
+                }
+                ca[idx] |= PMH_AGG_IGNORE_RETRACT;
+            }
+            // retract records are accepted iff EVERY value column's
+            // aggregator supports retraction (FieldSumAgg.retract,
+            // FieldPrimaryKeyAgg) or ignores it (ignore-retract wrapper) —
+            // otherwise the reference throws on the first retract
+            // (FieldAggregator.retract :47-53) and so do we, at merge time
+            {
+                bool all_rt = true;
+                for (int c = first_val; c < n_cols; c++) {
+                    uint8_t a = ca[c];
+                    if (a & PMH_AGG_IGNORE_RETRACT) continue;
+                    if (a == PMH_AGG_SUM || a == PMH_AGG_PRIMARY_KEY)
+                        continue;
+                    all_rt = false;
+                }
+                plan->agg_retract = all_rt && !plan->rrod;
             }
             plan->col_agg_dev = (uint8_t *)plan->bufs.alloc(n_cols);
             if (!plan->col_agg_dev ||
@@ -2238,7 +2269,8 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     (void)hipEventRecord(ev[2], st);
     int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0) |
                 (p->pu ? 4 : 0) | (p->first_row ? 8 : 0) |
-                (p->rrod ? 16 : 0) | (p->seqg ? 32 : 0);
+                (p->rrod ? 16 : 0) | (p->seqg ? 32 : 0) |
+                (p->agg_retract ? 64 : 0);
     if (const char *ab = getenv("PMH_ABLATE"))  // profiling-only phase knob
         flags |= (atoi(ab) & 0xf) << 8;
     if (const char *ab = getenv("PMH_FABL"))  // fused-kernel phase knob

@@ -239,7 +239,8 @@ class MergeReadPlan:
     def __init__(self, session: Session, files, key_cols, value_cols,
                  merge_engine="deduplicate", drop_delete=True,
                  ignore_delete=False, output="host", aggregations=None,
-                 remove_record_on_delete=False, sequence_groups=None):
+                 remove_record_on_delete=False, sequence_groups=None,
+                 ignore_retract=None):
         self.lib = session.lib
         desc = {
             "key_cols": key_cols,
@@ -255,6 +256,9 @@ class MergeReadPlan:
             # fields.<seq>.sequence-group=<members> (CoreOptions):
             # [{"sequence_fields": [...], "group_fields": [...]}]
             desc["sequence_groups"] = list(sequence_groups)
+        if ignore_retract:
+            # fields.<f>.ignore-retract = true (FieldIgnoreRetractAgg)
+            desc["ignore_retract"] = list(ignore_retract)
         if aggregations:
             # fields.<name>.aggregate-function (CoreOptions FIELDS_PREFIX);
             # unnamed columns default to last_non_null_value

@@ -230,3 +230,80 @@ class TestAggRemoveRecordOnDelete:
                               _value_cols(2), merge_engine="aggregation",
                               aggregations={"v_c0": "first_value"},
                               remove_record_on_delete=True)
+
+
+class TestAggregationRetracts:
+    """Retract records through retract-capable aggregators
+    (AggregateMergeFunction.add :80-101; FieldSumAgg.retract;
+    FieldPrimaryKeyAgg; FieldIgnoreRetractAgg) vs the oracle port of
+    getExpectedForAggSum's non-RROD branch (MergeFunctionTestUtils.java:
+    99-110)."""
+
+    def _gen(self, n_runs, rows, seed, n_value_cols=4, nulls=True):
+        rng = np.random.default_rng(seed)
+        total = n_runs * rows
+        seqs = rng.permutation(total).astype(np.int64)
+        runs = []
+        for r in range(n_runs):
+            space = max(int(rows * n_runs * 0.6), rows)
+            key = np.sort(rng.choice(space, rows,
+                                     replace=False)).astype(np.int64)
+            kind = rng.choice([0, 0, 0, 2, 1, 3], rows).astype(np.int8)
+            vals = [key.copy()]
+            msks = [np.ones(rows, bool)]
+            for _ in range(n_value_cols):
+                vals.append(rng.integers(0, 1000, rows).astype(np.int32))
+                msks.append(rng.random(rows) > 0.25 if nulls
+                            else np.ones(rows, bool))
+            runs.append({"key": key, "seq": seqs[r * rows:(r + 1) * rows],
+                         "kind": kind, "values": vals, "valid": msks})
+        return runs
+
+    def test_sum_retracts(self, tmp_path):
+        from oracle import aggregation_retract_model
+        runs = self._gen(5, 8_000, seed=401)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        aggs = ["primary_key", "sum", "sum", "sum", "sum"]
+        exp = aggregation_retract_model(runs, aggs, drop_delete=True)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(4), merge_engine="aggregation",
+                               aggregations={"v_k": "primary_key",
+                                             "v_c0": "sum", "v_c1": "sum",
+                                             "v_c2": "sum", "v_c3": "sum"},
+                               ) as plan:
+                got = _read_all(plan)
+        _check(got, exp, ["v_k", "v_c0", "v_c1", "v_c2", "v_c3"])
+
+    def test_ignore_retract_wrappers(self, tmp_path):
+        from oracle import aggregation_retract_model
+        runs = self._gen(4, 6_000, seed=402)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        aggs = ["primary_key", "sum", "last_non_null_value", "max",
+                "first_value"]
+        ign = {2, 3, 4}  # value-list indices of wrapped columns
+        exp = aggregation_retract_model(runs, aggs, ignore_retract=ign,
+                                        drop_delete=True)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(4), merge_engine="aggregation",
+                               aggregations={"v_k": "primary_key",
+                                             "v_c0": "sum",
+                                             "v_c1": "last_non_null_value",
+                                             "v_c2": "max",
+                                             "v_c3": "first_value"},
+                               ignore_retract=["v_c1", "v_c2", "v_c3"],
+                               ) as plan:
+                got = _read_all(plan)
+        _check(got, exp, ["v_k", "v_c0", "v_c1", "v_c2", "v_c3"])
+
+    def test_retract_without_capability_still_rejected(self, tmp_path):
+        # default last_non_null (no wrapper) + retract data -> loud error
+        runs = self._gen(2, 3_000, seed=403)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(4), merge_engine="aggregation",
+                               aggregations={"v_c0": "sum"}) as plan:
+                with pytest.raises(RuntimeError, match="aggregation"):
+                    _read_all(plan)
