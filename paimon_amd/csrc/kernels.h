@@ -96,7 +96,51 @@ struct Rlev2Chunk {
     uint64_t out_addr;     // absolute output base (patched at finalize)
 };
 
+// Host-prescanned DELTA_BINARY_PACKED work unit: one parquet delta BLOCK
+// (VectorizedDeltaBinaryPackedReader.java; parquet-format encodings.md):
+// per block, <= 8 miniblocks of vpm values each, bit widths packed into
+// `widths` (8 bits each), LSB-first bit packing. Blocks chain: the value
+// base entering block b = page first_value + sum of all prior blocks'
+// delta sums — computed on device (k_delta_sum + per-stream scan).
+struct DeltaChunk {
+    uint64_t src;       // packed miniblock data (byte-aligned)
+    uint64_t out_addr;  // absolute output base (patched at finalize)
+    int64_t out_start;  // output index of this block's FIRST delta value
+    int64_t min_delta;
+    int32_t count;      // real delta values in this block
+    int32_t stream;     // delta-stream index (for base resolution)
+    uint64_t widths;    // miniblock bit widths, 8 bits each
+    int16_t vpm;        // values per miniblock
+    int16_t n_mini;
+    int32_t out_esize;  // 4 or 8
+};
+
+// One DELTA stream (= one column chunk's page chain? no — one PAGE: pages
+// are self-contained): chunk range + the page's first value and where it
+// lands in the output.
+struct DeltaStream {
+    int64_t chunk_lo;
+    int64_t chunk_hi;
+    int64_t first;       // header first value = output element out0
+    int64_t out0;        // output index of the first value
+    uint64_t out_addr;   // absolute output base (patched at finalize)
+    int32_t out_esize;
+    int32_t _pad;
+};
+
 extern "C" {
+
+// DELTA_BINARY_PACKED decode, three phases batched across all chunks:
+// per-block delta sums, per-stream base scan, then unpack + wave scan +
+// base add into the output column.
+hipError_t pmh_launch_delta_sum(const DeltaChunk *chunks, int64_t n_chunks,
+                                int64_t *sums, hipStream_t stream);
+hipError_t pmh_launch_delta_scan(const DeltaStream *streams,
+                                 int64_t n_streams, const int64_t *sums,
+                                 int64_t *bases, hipStream_t stream);
+hipError_t pmh_launch_delta_emit(const DeltaChunk *chunks, int64_t n_chunks,
+                                 const int64_t *bases, hipStream_t stream);
+
 
 // Decode ORC RLEv2 / byte-RLE work chunks into a dense typed column
 // (int32 or int64 elements per chunk.out_esize). One wave per chunk.

@@ -2742,6 +2742,124 @@ __global__ void k_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks) {
     }
 }
 
+
+// ------------------------------------------------------------ DELTA decode
+//
+// Parquet DELTA_BINARY_PACKED (VectorizedDeltaBinaryPackedReader.java /
+// parquet-format Encodings.md): header <block><miniblocks><count><first>,
+// then blocks of [min_delta zigzag][miniblock bit widths][LSB-first packed
+// deltas]. value_i = first + sum of deltas. Three batched phases: per-block
+// sums (one wave per block), per-page base scan (pages are self-contained
+// streams; one workgroup per page), then unpack + wave-scan + base add.
+
+DEV uint64_t le_bits(const uint8_t *src, int64_t bit_off, int width) {
+    // LSB-first window like the RLE bit-packed decoder, any width <= 64:
+    // two aligned 8-byte loads + shift-combine (uploads padded by 16 B)
+    uint64_t addr = (uint64_t)src + (uint64_t)(bit_off >> 3);
+    uint64_t base = addr & ~7ull;
+    int off = (int)((addr - base) * 8) + (int)(bit_off & 7);
+    uint64_t lo = *reinterpret_cast<const uint64_t *>(base);
+    uint64_t hi = *reinterpret_cast<const uint64_t *>(base + 8);
+    uint64_t window = off ? ((lo >> off) | (hi << (64 - off))) : lo;
+    return width == 64 ? window : window & ((1ull << width) - 1);
+}
+
+DEV int64_t delta_at(const DeltaChunk &ch, int d) {
+    // miniblock of delta d and its packed offset
+    const int j = d / ch.vpm;
+    int64_t byte_off = 0;
+    for (int q = 0; q < j; q++)
+        byte_off += (int64_t)ch.vpm * ((ch.widths >> (8 * q)) & 0xff) / 8;
+    const int w = (int)((ch.widths >> (8 * j)) & 0xff);
+    if (w == 0) return ch.min_delta;
+    const uint64_t raw = le_bits((const uint8_t *)ch.src,
+                                 (int64_t)(d % ch.vpm) * w + byte_off * 8,
+                                 w);
+    return ch.min_delta + (int64_t)raw;
+}
+
+__global__ void k_delta_sum(const DeltaChunk *chunks, int64_t n_chunks,
+                            int64_t *sums) {
+    const int lane = (int)(threadIdx.x & 63);
+    const int64_t wave =
+        (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    const int64_t n_waves = (int64_t)gridDim.x * (blockDim.x >> 6);
+    for (int64_t c = wave; c < n_chunks; c += n_waves) {
+        const DeltaChunk ch = chunks[c];
+        int64_t acc = 0;
+        for (int d = lane; d < ch.count; d += 64) acc += delta_at(ch, d);
+        for (int off = 32; off; off >>= 1)
+            acc += __shfl_down(acc, off, 64);
+        if (lane == 0) sums[c] = acc;
+    }
+}
+
+__global__ void k_delta_scan(const DeltaStream *streams, int64_t n_streams,
+                             const int64_t *sums, int64_t *bases) {
+    // one workgroup per stream (page): running exclusive scan over its
+    // blocks' sums, seeded by the page's first value; also stores the
+    // page's first value itself
+    __shared__ int64_t s[256];
+    __shared__ int64_t running;
+    for (int64_t si = blockIdx.x; si < n_streams; si += gridDim.x) {
+        const DeltaStream st = streams[si];
+        if (threadIdx.x == 0) {
+            running = st.first;
+            if (st.out_esize == 8)
+                ((int64_t *)st.out_addr)[st.out0] = st.first;
+            else
+                ((int32_t *)st.out_addr)[st.out0] = (int32_t)st.first;
+        }
+        __syncthreads();
+        for (int64_t b = st.chunk_lo; b < st.chunk_hi; b += blockDim.x) {
+            int64_t i = b + threadIdx.x;
+            int64_t v = i < st.chunk_hi ? sums[i] : 0;
+            s[threadIdx.x] = v;
+            __syncthreads();
+            for (int d = 1; d < (int)blockDim.x; d <<= 1) {
+                int64_t add = threadIdx.x >= d ? s[threadIdx.x - d] : 0;
+                __syncthreads();
+                s[threadIdx.x] += add;
+                __syncthreads();
+            }
+            if (i < st.chunk_hi) bases[i] = running + s[threadIdx.x] - v;
+            __syncthreads();
+            if (threadIdx.x == 0) running += s[blockDim.x - 1];
+            __syncthreads();
+        }
+    }
+}
+
+__global__ void k_delta_emit(const DeltaChunk *chunks, int64_t n_chunks,
+                             const int64_t *bases) {
+    const int lane = (int)(threadIdx.x & 63);
+    const int64_t wave =
+        (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+    const int64_t n_waves = (int64_t)gridDim.x * (blockDim.x >> 6);
+    for (int64_t c = wave; c < n_chunks; c += n_waves) {
+        const DeltaChunk ch = chunks[c];
+        int64_t run = bases[c];
+        for (int b = 0; b < ch.count; b += 64) {
+            const int d = b + lane;
+            int64_t x = d < ch.count ? delta_at(ch, d) : 0;
+            // inclusive wave scan
+            int64_t acc = x;
+            for (int off = 1; off < 64; off <<= 1) {
+                int64_t up = __shfl_up(acc, off, 64);
+                if (lane >= off) acc += up;
+            }
+            if (d < ch.count) {
+                if (ch.out_esize == 8)
+                    ((int64_t *)ch.out_addr)[ch.out_start + d] = run + acc;
+                else
+                    ((int32_t *)ch.out_addr)[ch.out_start + d] =
+                        (int32_t)(run + acc);
+            }
+            run += __shfl(acc, 63, 64);
+        }
+    }
+}
+
 // ------------------------------------------------------------ k_rle_decode
 // Decode Parquet RLE/bit-packed hybrid streams (dictionary ids, def levels)
 // from host-prescanned run chunks (VectorizedRleValuesReader.java:977-1018
@@ -3021,6 +3139,34 @@ hipError_t pmh_launch_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks,
     int blocks = want < 4096 ? (int)(want ? want : 1) : 4096;
     hipLaunchKernelGGL(k_rlev2, dim3(blocks), dim3(256), 0, stream, chunks,
                        n_chunks);
+    return hipGetLastError();
+}
+
+
+hipError_t pmh_launch_delta_sum(const DeltaChunk *chunks, int64_t n_chunks,
+                                int64_t *sums, hipStream_t stream) {
+    int64_t want = (n_chunks + 3) / 4;
+    int blocks = want < 4096 ? (int)(want ? want : 1) : 4096;
+    hipLaunchKernelGGL(k_delta_sum, dim3(blocks), dim3(256), 0, stream,
+                       chunks, n_chunks, sums);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_delta_scan(const DeltaStream *streams,
+                                 int64_t n_streams, const int64_t *sums,
+                                 int64_t *bases, hipStream_t stream) {
+    int blocks = n_streams < 2048 ? (int)(n_streams ? n_streams : 1) : 2048;
+    hipLaunchKernelGGL(k_delta_scan, dim3(blocks), dim3(256), 0, stream,
+                       streams, n_streams, sums, bases);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_delta_emit(const DeltaChunk *chunks, int64_t n_chunks,
+                                 const int64_t *bases, hipStream_t stream) {
+    int64_t want = (n_chunks + 3) / 4;
+    int blocks = want < 4096 ? (int)(want ? want : 1) : 4096;
+    hipLaunchKernelGGL(k_delta_emit, dim3(blocks), dim3(256), 0, stream,
+                       chunks, n_chunks, bases);
     return hipGetLastError();
 }
 

@@ -345,6 +345,9 @@ struct RunCol {
     bool orc_encoded = false;
     std::vector<Rlev2Chunk> rlev2_host;
     Rlev2Chunk *rlev2_dev = nullptr;
+    // parquet DELTA_BINARY_PACKED pages (decoded on GPU at read time)
+    std::vector<DeltaChunk> delta_host;
+    std::vector<DeltaStream> dstreams_host;
 };
 
 struct Run {
@@ -387,6 +390,12 @@ struct Section {
     // batched decode work (all run-columns in ONE launch each)
     Rlev2Chunk *rlev2_all = nullptr;
     int64_t n_rlev2 = 0;
+    DeltaChunk *delta_all = nullptr;
+    int64_t n_delta = 0;
+    DeltaStream *dstreams_all = nullptr;
+    int64_t n_dstreams = 0;
+    int64_t *delta_sums = nullptr;   // per-chunk block sums
+    int64_t *delta_bases = nullptr;  // per-chunk entering values
     RleChunk *def_all = nullptr;
     int64_t n_def = 0;
     bool any_dict = false;
@@ -676,6 +685,103 @@ static bool prescan_rle(const uint8_t *s, int64_t len, int bit_width,
             cnt += vals;
         }
     }
+    return true;
+}
+
+
+// Prescan one DELTA_BINARY_PACKED page (VectorizedDeltaBinaryPackedReader
+// .java / parquet-format Encodings.md): header <block_size><miniblocks_per_
+// block><total_count><first zigzag>, then blocks of [min_delta zigzag]
+// [miniblock widths][packed miniblocks]. Each PAGE is self-contained (its
+// own first value), so it forms one DeltaStream; each block becomes one
+// DeltaChunk. Miniblocks holding no real values carry a width byte but no
+// data (their widths zero out in the packed u64 so device byte offsets
+// skip nothing); partially-filled miniblocks are stored in full.
+static bool prescan_delta(const uint8_t *pp, int64_t plen, int64_t n_values,
+                          int64_t out_row0, uint64_t dev_base,
+                          int64_t rel_off, uint64_t out_addr, int out_esize,
+                          std::vector<DeltaChunk> &chunks,
+                          std::vector<DeltaStream> &streams,
+                          std::string &err) {
+    int64_t pos = 0;
+    auto uleb = [&](uint64_t *v) -> bool {
+        *v = 0;
+        int sh = 0;
+        for (;;) {
+            if (pos >= plen) return false;
+            uint8_t b = pp[pos++];
+            *v |= (uint64_t)(b & 0x7f) << sh;
+            if (!(b & 0x80)) return true;
+            sh += 7;
+        }
+    };
+    auto zz = [](uint64_t v) {
+        return (int64_t)(v >> 1) ^ -(int64_t)(v & 1);
+    };
+    uint64_t bs, mpb, total, zfirst;
+    if (!uleb(&bs) || !uleb(&mpb) || !uleb(&total) || !uleb(&zfirst)) {
+        err = "delta: truncated header";
+        return false;
+    }
+    if (mpb == 0 || mpb > 8 || bs % mpb != 0) {
+        err = "delta: unsupported miniblock layout (v1: <= 8 per block)";
+        return false;
+    }
+    if ((int64_t)total != n_values) {
+        err = "delta: header count disagrees with page";
+        return false;
+    }
+    const int vpm = (int)(bs / mpb);
+    if (n_values == 0) return true;
+    DeltaStream st{};
+    st.chunk_lo = (int64_t)chunks.size();
+    st.first = zz(zfirst);
+    st.out0 = out_row0;
+    st.out_addr = out_addr;
+    st.out_esize = out_esize;
+    int64_t remaining = n_values - 1;
+    int64_t out_i = out_row0 + 1;
+    while (remaining > 0) {
+        uint64_t zmd;
+        if (!uleb(&zmd) || pos + (int64_t)mpb > plen) {
+            err = "delta: truncated block";
+            return false;
+        }
+        DeltaChunk ch{};
+        ch.min_delta = zz(zmd);
+        ch.vpm = (int16_t)vpm;
+        ch.n_mini = (int16_t)mpb;
+        ch.out_addr = out_addr;
+        ch.out_start = out_i;
+        ch.out_esize = out_esize;
+        ch.count = (int32_t)(remaining < (int64_t)bs ? remaining
+                                                     : (int64_t)bs);
+        int64_t data = 0;
+        for (uint64_t j = 0; j < mpb; j++) {
+            const int w = pp[pos + j];
+            if (w > 64) {
+                err = "delta: miniblock width > 64";
+                return false;
+            }
+            const int64_t real = remaining - (int64_t)j * vpm;
+            if (real > 0) {
+                ch.widths |= (uint64_t)w << (8 * j);
+                data += (int64_t)vpm * w / 8;
+            }
+        }
+        pos += mpb;
+        ch.src = dev_base + (uint64_t)(rel_off + pos);
+        if (pos + data > plen) {
+            err = "delta: truncated miniblocks";
+            return false;
+        }
+        pos += data;
+        chunks.push_back(ch);
+        out_i += ch.count;
+        remaining -= ch.count;
+    }
+    st.chunk_hi = (int64_t)chunks.size();
+    streams.push_back(st);
     return true;
 }
 
@@ -1305,7 +1411,7 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     return false;
                 }
                 // classify data pages
-                bool has_plain = false, has_dict = false;
+                bool has_plain = false, has_dict = false, has_delta = false;
                 const uint8_t *dict_host = nullptr;
                 int64_t dict_count = 0;
                 for (size_t pi = 0; pi < cc.pages.size(); pi++) {
@@ -1318,6 +1424,8 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     } else if (pg.encoding == ENC_RLE_DICTIONARY ||
                                pg.encoding == ENC_PLAIN_DICTIONARY) {
                         has_dict = true;
+                    } else if (pg.encoding == ENC_DELTA_BINARY_PACKED) {
+                        has_delta = true;
                     } else {
                         set_error("%s: unsupported encoding %d",
                                   fd.path.c_str(), pg.encoding);
@@ -1328,6 +1436,12 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     set_error("%s col %s: mixed PLAIN/dictionary pages in one "
                               "chunk not supported yet",
                               fd.path.c_str(), cols[c].name.c_str());
+                    return false;
+                }
+                if (has_delta && (has_plain || has_dict)) {
+                    set_error("%s col %s: mixed DELTA/other pages in one "
+                              "chunk not supported", fd.path.c_str(),
+                              cols[c].name.c_str());
                     return false;
                 }
                 if (cols[c].dtype == PMH_DT_STRING && has_plain) {
@@ -1355,7 +1469,59 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     }
                     vpos[pi] = pos;
                 }
-                if (chunk_nulls && has_dict) {
+                if (has_delta) {
+                    if (chunk_nulls) {
+                        set_error("%s col %s: nullable DELTA columns are a "
+                                  "later round", fd.path.c_str(),
+                                  cols[c].name.c_str());
+                        return false;
+                    }
+                    if (cols[c].dtype == PMH_DT_STRING) {
+                        set_error("%s col %s: DELTA byte arrays are a later "
+                                  "round", fd.path.c_str(),
+                                  cols[c].name.c_str());
+                        return false;
+                    }
+                    // upload the encoded payload image (+16 B pad for the
+                    // device bit-window loads); blocks decode on GPU at
+                    // read time (k_delta_sum/scan/emit)
+                    int64_t payload_len = 0;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++)
+                        payload_len = std::max(
+                            payload_len,
+                            ppo[pi] + (cc.codec != CODEC_UNCOMPRESSED
+                                           ? cc.pages[pi].uncompressed_size
+                                           : cc.pages[pi].compressed_size));
+                    void *dev = plan->bufs.alloc(payload_len + 16);
+                    if (!dev) return false;
+                    if (hipMemcpy(dev, payload_base, payload_len,
+                                  hipMemcpyHostToDevice) != hipSuccess) {
+                        set_error("H2D failed");
+                        return false;
+                    }
+                    plan->encoded_bytes_total += payload_len;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                        auto &pg = cc.pages[pi];
+                        if (pg.page_type != 0) continue;
+                        const uint8_t *pp = payload_base + ppo[pi];
+                        int64_t pos = vpos[pi];
+                        int64_t plen = (cc.codec != CODEC_UNCOMPRESSED
+                                            ? pg.uncompressed_size
+                                            : pg.compressed_size);
+                        std::string derr;
+                        if (!prescan_delta(pp + pos, plen - pos,
+                                           pg.num_values,
+                                           chunk_row0 + pg.first_row,
+                                           (uint64_t)dev, ppo[pi] + pos,
+                                           (uint64_t)rc.contig, stored,
+                                           rc.delta_host, rc.dstreams_host,
+                                           derr)) {
+                            set_error("%s col %s: %s", fd.path.c_str(),
+                                      cols[c].name.c_str(), derr.c_str());
+                            return false;
+                        }
+                    }
+                } else if (chunk_nulls && has_dict) {
                     // dictionary chunk with nulls: def-level streams stay
                     // encoded; the id stream decodes to DENSE positions on
                     // the GPU (k_rle_decode), k_dict_gather fills the dense
@@ -1694,14 +1860,25 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
     }
     std::vector<Rlev2Chunk> all_v;
     std::vector<RleChunk> all_d;
+    std::vector<DeltaChunk> all_dc;
+    std::vector<DeltaStream> all_ds;
     for (auto &run : sec.runs)
         for (auto &rc : run.cols) {
             sec.any_dict |= rc.dict_encoded;
-            sec.any_decode |=
-                rc.dict_encoded || rc.has_nulls || !rc.rlev2_host.empty();
+            sec.any_decode |= rc.dict_encoded || rc.has_nulls ||
+                              !rc.rlev2_host.empty() ||
+                              !rc.delta_host.empty();
             all_v.insert(all_v.end(), rc.rlev2_host.begin(),
                          rc.rlev2_host.end());
             all_d.insert(all_d.end(), rc.def_host.begin(), rc.def_host.end());
+            const int64_t cbase = (int64_t)all_dc.size();
+            all_dc.insert(all_dc.end(), rc.delta_host.begin(),
+                          rc.delta_host.end());
+            for (DeltaStream st : rc.dstreams_host) {
+                st.chunk_lo += cbase;  // indices into the BATCHED arrays
+                st.chunk_hi += cbase;
+                all_ds.push_back(st);
+            }
         }
     sec.n_rlev2 = (int64_t)all_v.size();
     sec.n_def = (int64_t)all_d.size();
@@ -1714,6 +1891,19 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
         sec.def_all =
             (RleChunk *)up(all_d.data(), all_d.size() * sizeof(RleChunk));
         if (!sec.def_all) return false;
+    }
+    sec.n_delta = (int64_t)all_dc.size();
+    sec.n_dstreams = (int64_t)all_ds.size();
+    if (sec.n_delta) {
+        sec.delta_all = (DeltaChunk *)up(all_dc.data(),
+                                         all_dc.size() * sizeof(DeltaChunk));
+        sec.dstreams_all = (DeltaStream *)up(
+            all_ds.data(), all_ds.size() * sizeof(DeltaStream));
+        sec.delta_sums = (int64_t *)plan->bufs.alloc(sec.n_delta * 8);
+        sec.delta_bases = (int64_t *)plan->bufs.alloc(sec.n_delta * 8);
+        if (!sec.delta_all || !sec.dstreams_all || !sec.delta_sums ||
+            !sec.delta_bases)
+            return false;
     }
     return sec.key_cols && sec.seq_cols && sec.kind_cols && sec.all_cols &&
            sec.lens_dev && sec.cuts && sec.total_dev && sec.err_dev;
@@ -2212,6 +2402,18 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         if (sec.n_rlev2) {
             hipError_t e = pmh_launch_rlev2(sec.rlev2_all, sec.n_rlev2, st);
             if (e != hipSuccess) return fail("rlev2", e);
+        }
+        if (sec.n_delta) {
+            hipError_t e = pmh_launch_delta_sum(sec.delta_all, sec.n_delta,
+                                                sec.delta_sums, st);
+            if (e == hipSuccess)
+                e = pmh_launch_delta_scan(sec.dstreams_all, sec.n_dstreams,
+                                          sec.delta_sums, sec.delta_bases,
+                                          st);
+            if (e == hipSuccess)
+                e = pmh_launch_delta_emit(sec.delta_all, sec.n_delta,
+                                          sec.delta_bases, st);
+            if (e != hipSuccess) return fail("delta", e);
         }
         for (auto &run : sec.runs) {
             for (size_t c = 0; c < run.cols.size(); c++) {
