@@ -778,3 +778,60 @@ def aggregation_retract_model(runs, aggs, ignore_retract=(),
                    for c, v in enumerate(out["values"])],
         "valid": [np.array(v, bool) for v in out["valid"]],
     }
+
+
+def merge_dedup_useq_model(runs, useq_idx, drop_delete=True,
+                           ignore_delete=False, first_row=False):
+    """Deduplicate / FirstRow with user-defined sequence fields
+    (utils/UserDefinedSeqComparator.java:38-80; total order becomes
+    ascending (key, seq fields..., seq, isAdd), nulls FIRST). useq_idx:
+    indices into the runs' values lists. Returns (run, row)."""
+    key0 = np.concatenate([r["key"] for r in runs])
+    seq0 = np.concatenate([r["seq"] for r in runs])
+    kind0 = np.concatenate([r["kind"] for r in runs])
+    run0 = np.concatenate([np.full(len(r["key"]), i, dtype=np.int32)
+                           for i, r in enumerate(runs)])
+    row0 = np.concatenate([np.arange(len(r["key"]), dtype=np.int64)
+                           for r in runs])
+    sort_keys = [_kind_is_add(kind0).astype(np.int8), seq0]
+    for c in reversed(useq_idx):
+        vals = np.concatenate([r["values"][c] for r in runs])
+        if "valid" in runs[0]:
+            msk = np.concatenate([r["valid"][c] for r in runs])
+        else:
+            msk = np.ones(len(vals), dtype=bool)
+        sort_keys.append(np.where(msk, vals, 0))
+        sort_keys.append(msk.astype(np.int8))  # nulls first
+    sort_keys.append(key0)
+    order = np.lexsort(tuple(sort_keys))
+    key = key0[order]
+    kind = kind0[order]
+    run = run0[order]
+    row = row0[order]
+    n = len(key)
+    if n == 0:
+        return (np.empty(0, dtype=np.int32), np.empty(0, dtype=np.int64))
+    head = np.empty(n, dtype=bool)
+    head[0] = True
+    head[1:] = key[1:] != key[:-1]
+    group_id = np.cumsum(head) - 1
+    n_groups = group_id[-1] + 1
+    group_size = np.bincount(group_id, minlength=n_groups)
+    idx = np.arange(n)
+    eligible = _kind_is_add(kind) if ignore_delete \
+        else np.ones(n, dtype=bool)
+    if first_row:  # FIRST eligible per group
+        pick = np.full(n_groups, n, dtype=np.int64)
+        np.minimum.at(pick, group_id[eligible], idx[eligible])
+        pick[pick == n] = -1
+    else:  # LAST eligible per group
+        pick = np.full(n_groups, -1, dtype=np.int64)
+        np.maximum.at(pick, group_id[eligible], idx[eligible])
+    single_first = np.full(n_groups, -1, dtype=np.int64)
+    np.maximum.at(single_first, group_id, idx)
+    singles = group_size == 1
+    pick[singles] = single_first[singles]
+    sel = pick[pick >= 0]
+    if drop_delete:
+        sel = sel[_kind_is_add(kind[sel])]
+    return run[sel], row[sel]

@@ -181,6 +181,7 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  const DevCol *cols, const uint8_t *col_dtype,
                                  const uint8_t *col_nullable, int n_cols,
                                  int key_col, int seq_col, int kind_col,
+                                 const int16_t *useq_cols, int n_useq,
                                  uint64_t *status, uint64_t *ticket,
                                  int64_t *total_out, uint32_t *dense_winners,
                                  void *const *out_ptrs,

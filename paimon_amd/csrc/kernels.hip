@@ -349,6 +349,34 @@ DEV int32_t corank(int64_t d, const uint16_t *pa, int32_t la,
 }
 
 DEV bool kind_is_add(uint8_t k) { return k == 0 || k == 2; }
+
+// user-defined sequence fields (CoreOptions sequence.field;
+// utils/UserDefinedSeqComparator.java:38-80, wired at
+// MergeFileSplitRead.java:543-545): listed value columns compare BEFORE
+// the sequence number, ascending, nulls FIRST (codegen nullIsLast=false).
+DEV int useq_cmp(const DevCol *cols, const uint8_t *col_dtype, int n_cols,
+                 const int16_t *ucols, int nu, int runA, int64_t rowA,
+                 int runB, int64_t rowB) {
+    for (int j = 0; j < nu; j++) {
+        const int c = ucols[j];
+        const DevCol &a = cols[runA * n_cols + c];
+        const DevCol &b = cols[runB * n_cols + c];
+        const uint8_t va =
+            a.valid0 ? ((const uint8_t *)a.valid0)[rowA] : 1;
+        const uint8_t vb =
+            b.valid0 ? ((const uint8_t *)b.valid0)[rowB] : 1;
+        if (va != vb) return va ? 1 : -1;  // null sorts first
+        if (!va) continue;
+        const int dt = col_dtype[c];
+        const bool wide = dt == 4 || dt == 6;
+        const int64_t x = wide ? col_load<int64_t>(a, rowA)
+                               : (int64_t)col_load<int32_t>(a, rowA);
+        const int64_t y = wide ? col_load<int64_t>(b, rowB)
+                               : (int64_t)col_load<int32_t>(b, rowB);
+        if (x != y) return x > y ? 1 : -1;
+    }
+    return 0;
+}
 // k_merge_tiles packs (seq << 1) | isAdd (the fused kernel uses the 2-bit
 // ps2_* packing); helper so call sites read uniformly
 DEV bool ps2m_isadd(int64_t w) { return w & 1; }
@@ -864,7 +892,8 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                   const DevCol *cols /* k * n_cols, run-major */,
                   const uint8_t *col_dtype, const uint8_t *col_nullable,
                   int n_cols, int key_col /* -1: composite */, int seq_col,
-                  int kind_col, uint64_t *status, uint64_t *ticket,
+                  int kind_col, const int16_t *useq_cols, int n_useq,
+                  uint64_t *status, uint64_t *ticket,
                   int64_t *total_out, uint32_t *dense_winners,
                   void *const *out_ptrs,
                   uint8_t *const *out_valid, uint32_t *err_flag) {
@@ -891,6 +920,12 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         // one group-walk body for both passes (count, then emit): walks the
         // thread's head range applying Deduplicate/FirstRow + wrapper +
         // drop-delete rules, calling emit(n-th-winner, seg-index)
+        auto seg_rr = [&](uint16_t s, int *rr, int64_t *row) {
+            int r = 0;
+            while (r + 1 < k && sm.segoff[r + 1] <= (int32_t)s) r++;
+            *rr = r;
+            *row = c0[r] + ((int32_t)s - sm.segoff[r]);
+        };
         auto walk_pass = [&](int32_t lo_i, int32_t hi_i, auto emit) {
             const uint16_t *mo = sm.perm[cur];
             const int32_t M = sm.mtotal;
@@ -910,9 +945,26 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                     if (v != PMH_DEAD) {
                         bool e = !ignore_delete || ps2_isadd(v);
                         any_retract |= !ps2_isadd(v);
-                        bool take =
-                            nlive == 0 || (e && !e_best) ||
-                            (e == e_best && (FR ? v < v_best : v > v_best));
+                        bool take;
+                        if (nlive == 0) {
+                            take = true;
+                        } else if (e != e_best) {
+                            take = e;
+                        } else if (n_useq > 0) {
+                            // (seq fields..., seq, isAdd) ascending: the
+                            // fields compare before the sequence number
+                            int ra, rb;
+                            int64_t qa, qb;
+                            seg_rr(mo[x], &ra, &qa);
+                            seg_rr(s_best, &rb, &qb);
+                            const int cu =
+                                useq_cmp(cols, col_dtype, n_cols, useq_cols,
+                                         n_useq, ra, qa, rb, qb);
+                            take = FR ? (cu < 0 || (cu == 0 && v < v_best))
+                                      : (cu > 0 || (cu == 0 && v > v_best));
+                        } else {
+                            take = FR ? v < v_best : v > v_best;
+                        }
                         if (take) {
                             s_best = mo[x];
                             v_best = v;
@@ -3065,6 +3117,7 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                                  const DevCol *cols, const uint8_t *col_dtype,
                                  const uint8_t *col_nullable, int n_cols,
                                  int key_col, int seq_col, int kind_col,
+                                 const int16_t *useq_cols, int n_useq,
                                  uint64_t *status, uint64_t *ticket,
                                  int64_t *total_out, uint32_t *dense_winners,
                                  void *const *out_ptrs,
@@ -3081,7 +3134,8 @@ hipError_t pmh_launch_merge_emit(const DevCol *keys, const DevCol *seqs,
                            stream, keys, seqs, kinds, lens, k, cuts,
                            tile_base, tile_limit, n_tiles, tile_rows, flags,
                            tombs, cols, col_dtype, col_nullable,
-                           n_cols, key_col, seq_col, kind_col, status, ticket,
+                           n_cols, key_col, seq_col, kind_col, useq_cols,
+                           n_useq, status, ticket,
                            total_out, dense_winners, out_ptrs, out_valid,
                            err_flag);
     };

@@ -437,6 +437,10 @@ struct pmh_plan_t {
     // act on their groups
     bool seqg = false;
     bool agg_retract = false;  // every aggregator retract-capable
+    // sequence.field (user-defined sequence comparator): compare listed
+    // value columns before the sequence number
+    int n_useq = 0;
+    int16_t *useq_dev = nullptr;
     int n_seq_groups = 0;
     uint8_t *col_group_dev = nullptr;
     int16_t *sg_fields_dev = nullptr;
@@ -2126,6 +2130,46 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                 return nullptr;
             }
         }
+        {
+            const Json &uf = j["sequence_fields"];
+            if (!uf.arr.empty()) {
+                if (plan->pu) {
+                    set_error("sequence.field with partial-update/"
+                              "aggregation engines is a later round (v1: "
+                              "deduplicate / first-row)");
+                    return nullptr;
+                }
+                if (uf.arr.size() > 4) {
+                    set_error("at most 4 sequence fields in v1");
+                    return nullptr;
+                }
+                const int first_val = plan->n_key_cols + 2;
+                const int n_cols = (int)plan->cols.size();
+                std::vector<int16_t> uc;
+                for (const auto &fj : uf.arr) {
+                    int idx = -1;
+                    for (int c = first_val; c < n_cols; c++)
+                        if (plan->cols[c].name == fj.as_str()) idx = c;
+                    if (idx < 0 ||
+                        plan->cols[idx].dtype > PMH_DT_INT64) {
+                        set_error("sequence field '%s': integer value "
+                                  "columns only in v1",
+                                  fj.as_str().c_str());
+                        return nullptr;
+                    }
+                    uc.push_back((int16_t)idx);
+                }
+                plan->n_useq = (int)uc.size();
+                plan->useq_dev =
+                    (int16_t *)plan->bufs.alloc(uc.size() * 2);
+                if (!plan->useq_dev ||
+                    hipMemcpy(plan->useq_dev, uc.data(), uc.size() * 2,
+                              hipMemcpyHostToDevice) != hipSuccess) {
+                    set_error("H2D of sequence fields failed");
+                    return nullptr;
+                }
+            }
+        }
         plan->drop_delete = j["drop_delete"].as_bool(true);
         plan->ignore_delete = j["ignore_delete"].as_bool(false);
         {
@@ -2133,6 +2177,11 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             // winner engines; PMH_FUSED=0 keeps the 3-kernel chain for A/B
             const char *pf = getenv("PMH_FUSED");
             plan->fused = !plan->pu && !(pf && pf[0] == '0');
+            if (plan->n_useq > 0 && !plan->pu && !plan->fused) {
+                set_error("sequence.field needs the fused merge path "
+                          "(unset PMH_FUSED=0)");
+                return nullptr;
+            }
             // split emission is the default product path (fastest
             // measured); PMH_FSPLIT=0 keeps in-kernel emission for A/B
             const char *fs = getenv("PMH_FSPLIT");
@@ -2518,6 +2567,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                     sec.tombs_dev,
                     sec.all_cols, p->col_dtype_dev, p->col_nullable_dev,
                     n_cols, key_col, p->n_key_cols, p->n_key_cols + 1,
+                    p->useq_dev, p->n_useq,
                     sec.status, sec.ticket, sec.total_dev, sec.dense_winners,
                     p->out_ptrs_dev, p->out_valid_dev, sec.err_dev, st);
                 if (e != hipSuccess) return fail("merge_emit", e);

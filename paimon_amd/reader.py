@@ -240,7 +240,7 @@ class MergeReadPlan:
                  merge_engine="deduplicate", drop_delete=True,
                  ignore_delete=False, output="host", aggregations=None,
                  remove_record_on_delete=False, sequence_groups=None,
-                 ignore_retract=None):
+                 ignore_retract=None, sequence_fields=None):
         self.lib = session.lib
         desc = {
             "key_cols": key_cols,
@@ -259,6 +259,10 @@ class MergeReadPlan:
         if ignore_retract:
             # fields.<f>.ignore-retract = true (FieldIgnoreRetractAgg)
             desc["ignore_retract"] = list(ignore_retract)
+        if sequence_fields:
+            # sequence.field (UserDefinedSeqComparator): value columns that
+            # compare before the sequence number
+            desc["sequence_fields"] = list(sequence_fields)
         if aggregations:
             # fields.<name>.aggregate-function (CoreOptions FIELDS_PREFIX);
             # unnamed columns default to last_non_null_value
