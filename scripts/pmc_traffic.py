@@ -20,13 +20,23 @@ import json
 import sys
 
 # kernel-name substring -> (bench kernel key, fetch correction factor, note)
+# (round 2: the split fused path replaced merge_tiles/emit for the winner
+# engines — k_merge_emit stages coalesced (FETCH counted at half on gfx950,
+# guide §HBM -> x2) and writes key/seq/kind + dense winners; k_emit_dense
+# gathers value columns at 64B line granularity (FETCH raw) and writes the
+# value outputs)
 KERNELS = {
+    "k_merge_emit": ("merge_emit", 2.0,
+                     "FETCH x2 (wide coalesced key/seq/kind staging, guide "
+                     "§HBM) + WRITE raw"),
+    "k_emit_dense": ("emit_dense", 1.0,
+                     "FETCH raw (64B-granule value gathers) + WRITE raw"),
     "k_merge_tiles": ("merge", 2.0,
                       "FETCH x2 (wide coalesced staging, guide §HBM) + "
                       "WRITE raw"),
-    "k_emit": ("emit", 1.0,
-               "FETCH raw (64B-granule gathers counted exactly) + WRITE raw "
-               "(validated = algorithmic output bytes)"),
+    "k_emit<": ("emit", 1.0,
+                "FETCH raw (64B-granule gathers counted exactly) + WRITE "
+                "raw (validated = algorithmic output bytes)"),
 }
 
 
@@ -45,11 +55,18 @@ def per_dispatch(pattern, counter):
 
 
 def match(averages):
+    # ordered most-specific-first; first substring hit wins per kernel
+    order = ["k_merge_emit", "k_emit_dense", "k_merge_tiles", "k_emit"]
     out = {}
     for kn, v in averages.items():
-        for sub, (key, corr, note) in KERNELS.items():
-            if sub in kn and "_pu" not in kn and "_agg" not in kn:
+        if "_pu" in kn or "_agg" in kn:
+            continue
+        for sub in order:
+            if sub in kn:
+                key, corr, note = KERNELS[sub if sub != "k_emit"
+                                          else "k_emit<"]
                 out[key] = (v, corr, note)
+                break
     return out
 
 
