@@ -2543,8 +2543,10 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             // the plan stream while kernel B (value gather) for chunk i
             // runs on stream_b, gated by a per-chunk event. Non-split: one
             // A launch does everything.
-            const int64_t want =
+            int64_t want =
                 sec.dense_winners ? (sec.n_tiles + 4095) / 4096 : 1;
+            if (const char *fc = getenv("PMH_FCHUNKS"))
+                want = sec.dense_winners ? atoll(fc) : 1;
             const int64_t n_chunks =
                 want < 1 ? 1 : (want > 8 ? 8 : want);
             const int64_t per =

@@ -59,7 +59,7 @@ timeout 420 rocprofv3 --pmc WRITE_SIZE \
     python3 /root/repo/bench.py --steps 3 --warmup 1 --cpu-baseline-rows 0 \
     > /root/repo/gpurun_out/tprof/write.log 2>&1
 python3 /root/repo/scripts/pmc_traffic.py /root/repo/gpurun_out/tprof \
-    8 10000000 8 > /root/repo/gpurun_out/pmc_traffic_r02.json
+    8 10000000 8 4 > /root/repo/gpurun_out/pmc_traffic_r02.json
 echo TRAFFIC_RC=$?
 # reduce the SQ pass before discarding the raw CSVs
 python3 - << 'PYEOF'

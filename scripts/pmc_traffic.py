@@ -40,7 +40,7 @@ KERNELS = {
 }
 
 
-def per_dispatch(pattern, counter):
+def per_dispatch(pattern, counter, divisor=None):
     tot, disp = {}, {}
     for f in glob.glob(pattern):
         with open(f) as fh:
@@ -51,7 +51,8 @@ def per_dispatch(pattern, counter):
                 tot[kn] = tot.get(kn, 0.0) + float(row["Counter_Value"])
                 disp.setdefault(kn, set()).add(
                     (row.get("Dispatch_Id"), row.get("Correlation_Id")))
-    return {k: tot[k] / len(disp[k]) for k in tot}
+    return {k: tot[k] / (divisor if divisor else len(disp[k]))
+            for k in tot}
 
 
 def match(averages):
@@ -73,10 +74,14 @@ def match(averages):
 def main():
     prof_dir = sys.argv[1]
     runs, rows, vals = (int(x) for x in sys.argv[2:5])
+    # measured steps of the profiled command (steps + warmup): the split
+    # path launches the merge kernel once per tile CHUNK, so per-dispatch
+    # averages no longer equal per-step traffic — report PER-STEP totals
+    n_steps = int(sys.argv[5]) if len(sys.argv) > 5 else 4
     fetch = match(per_dispatch(f"{prof_dir}/*pmc_fetch*counter*.csv",
-                               "FETCH_SIZE"))
+                               "FETCH_SIZE", n_steps))
     write = match(per_dispatch(f"{prof_dir}/*pmc_write*counter*.csv",
-                               "WRITE_SIZE"))
+                               "WRITE_SIZE", n_steps))
     # FETCH_SIZE/WRITE_SIZE report kilobytes
     KB = 1024.0
     kernels = {}
@@ -87,7 +92,7 @@ def main():
             "fetch_raw_bytes": fv * KB,
             "write_raw_bytes": wv * KB,
             "fetch_correction": fcorr,
-            "bytes_per_launch": fv * KB * fcorr + wv * KB,
+            "bytes_per_step": fv * KB * fcorr + wv * KB,
             "note": note,
         }
     # sanity: the merge kernels' algorithmic read bytes = rows*(8+8+4)

@@ -17,7 +17,7 @@ timeout 420 rocprofv3 --pmc WRITE_SIZE \
     > /root/repo/gpurun_out/tprof/write.log 2>&1
 echo WRITE_RC=$?
 python3 /root/repo/scripts/pmc_traffic.py /root/repo/gpurun_out/tprof \
-    8 10000000 8 > /root/repo/gpurun_out/pmc_traffic.json
+    8 10000000 8 4 > /root/repo/gpurun_out/pmc_traffic.json
 echo PARSE_RC=$?
 head -c 2000 /root/repo/gpurun_out/pmc_traffic.json
 # keep only the reduced json + logs; the raw counter CSVs can exceed the
