@@ -2543,8 +2543,11 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             // the plan stream while kernel B (value gather) for chunk i
             // runs on stream_b, gated by a per-chunk event. Non-split: one
             // A launch does everything.
-            int64_t want =
-                sec.dense_winners ? (sec.n_tiles + 4095) / 4096 : 1;
+            // measured: A/B stream overlap is a wash — B's full-occupancy
+            // gathers stretch A by about what the B tail saves (ch=1
+            // 7.19 ms vs ch=8 7.57 on the same box) — so the default is
+            // ONE chunk (A then B); PMH_FCHUNKS re-enables the pipeline
+            int64_t want = 1;
             if (const char *fc = getenv("PMH_FCHUNKS"))
                 want = sec.dense_winners ? atoll(fc) : 1;
             const int64_t n_chunks =
