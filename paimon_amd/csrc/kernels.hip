@@ -335,8 +335,9 @@ struct TileSmem {
 // merge(A, B), ties take A first. A/B are perm-index sequences; key lookup
 // through skey[.]. For i in [max(0,d-lb), min(d,la)), j = d-1-i is always
 // in [0, lb). Advance while A[i] <= B[j] (A[i] belongs in the first d).
+template <typename KT>
 DEV int32_t corank(int64_t d, const uint16_t *pa, int32_t la,
-                   const uint16_t *pb, int32_t lb, const int64_t *skey) {
+                   const uint16_t *pb, int32_t lb, const KT *skey) {
     int64_t ilo = d > lb ? d - lb : 0;
     int64_t ihi = d < la ? d : la;
     while (ilo < ihi) {
@@ -877,6 +878,15 @@ struct FusedSmem {
     int32_t haspred;
     int32_t mtotal;
     int32_t mreal;
+    // 32-bit tile-relative key mode: real merge tiles span ~3.6k sorted
+    // keys, so (max-min) almost always fits 32 bits — staged keys repack
+    // in place to uint32 offsets from tile_kmin (ukey space), halving the
+    // LDS bank pressure of the merge's random key compares
+    uint64_t wkmin[PMH_TILE_THREADS / 64];
+    uint64_t wkmax[PMH_TILE_THREADS / 64];
+    uint64_t tile_kmin;
+    uint32_t predkey32;
+    int32_t narrow;
 };
 
 constexpr uint64_t LOOK_AGG = 1ull << 62;
@@ -884,7 +894,8 @@ constexpr uint64_t LOOK_PREFIX = 2ull << 62;
 constexpr uint64_t LOOK_VAL = (1ull << 62) - 1;
 
 template <bool FR>
-__launch_bounds__(PMH_TILE_THREADS, 2) __global__
+__attribute__((amdgpu_waves_per_eu(4))) __launch_bounds__(PMH_TILE_THREADS, 2)
+__global__
 void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                   const int64_t *lens, int k, const int32_t *cuts,
                   int64_t tile_base, int64_t tile_limit, int64_t n_tiles,
@@ -1023,6 +1034,7 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
             // --- stage key / packed-seq segments (coalesced per run);
             // deletion-vector tombstones stage as PMH_DEAD (the walk skips
             // them — ApplyDeletionVectorReader semantics)
+            uint64_t tkmin = ~0ull, tkmax = 0;
             for (int r = 0; r < k; r++) {
                 int32_t off = sm.segoff[r], len = sm.seglen[r];
                 int64_t base = c0[r];
@@ -1037,67 +1049,133 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                         ? reinterpret_cast<const uint8_t *>(tombs[r]) + base
                         : nullptr;
                 for (int32_t i = tid; i < len; i += blockDim.x) {
-                    sm.skey[off + i] = key_at(kaddr0, i, kes);
+                    const int64_t kv = key_at(kaddr0, i, kes);
+                    const uint64_t uk = ukey(kv);
+                    if (uk < tkmin) tkmin = uk;
+                    if (uk > tkmax) tkmax = uk;
+                    sm.skey[off + i] = kv;
                     sm.sseq[off + i] = (tb && tb[i])
                                            ? PMH_DEAD
                                            : ps2_pack(saddr[i], daddr[i]);
                     sm.perm[0][off + i] = (uint16_t)(off + i);
                 }
             }
+            // block-reduce the tile key range, then repack the staged keys
+            // to 32-bit tile-relative offsets when they fit
+            for (int off2 = 32; off2; off2 >>= 1) {
+                const uint64_t a = __shfl_down(tkmin, off2, 64);
+                const uint64_t b = __shfl_down(tkmax, off2, 64);
+                if (a < tkmin) tkmin = a;
+                if (b > tkmax) tkmax = b;
+            }
+            if (lane == 0) {
+                sm.wkmin[wv] = tkmin;
+                sm.wkmax[wv] = tkmax;
+            }
             __syncthreads();
-
-            // --- pairwise stable merge (identical to k_merge_tiles)
-            for (int width = 1; width < k; width <<= 1) {
-                const int nxt = cur ^ 1;
-                const int CH = 8;
-                int n_chunks = (M + CH - 1) / CH;
-                for (int ch = tid; ch < n_chunks; ch += blockDim.x) {
-                    int64_t o = (int64_t)ch * CH;
-                    int remaining = (int)(M - o < CH ? M - o : CH);
-                    int p = 0;
-                    while (remaining > 0) {
-                        while ((p + 1) * 2 * width < k &&
-                               sm.segoff[(p + 1) * 2 * width] <= o)
-                            p++;
-                        int a0 = p * 2 * width;
-                        int amid = a0 + width < k ? a0 + width : k;
-                        int b1 = a0 + 2 * width < k ? a0 + 2 * width : k;
-                        int32_t abase = sm.segoff[a0];
-                        int32_t la = sm.segoff[amid] - abase;
-                        int32_t lb = sm.segoff[b1] - sm.segoff[amid];
-                        int64_t d = o - abase;
-                        int32_t lim = la + lb - (int32_t)d;
-                        if (lim <= 0) break;
-                        int n_out = lim < remaining ? lim : remaining;
-                        const uint16_t *pa = &sm.perm[cur][abase];
-                        const uint16_t *pb = &sm.perm[cur][abase + la];
-                        int32_t ai = corank(d, pa, la, pb, lb, sm.skey);
-                        int32_t bi = (int32_t)d - ai;
-                        uint16_t *out = &sm.perm[nxt][abase + d];
-                        for (int x = 0; x < n_out; x++) {
-                            bool takeA;
-                            if (ai >= la) takeA = false;
-                            else if (bi >= lb) takeA = true;
-                            else takeA = !(sm.skey[pa[ai]] > sm.skey[pb[bi]]);
-                            out[x] = takeA ? pa[ai++] : pb[bi++];
-                        }
-                        o += n_out;
-                        remaining -= n_out;
-                    }
+            if (tid == 0) {
+                uint64_t mn = ~0ull, mx = 0;
+#pragma unroll
+                for (int w = 0; w < NW; w++) {
+                    if (sm.wkmin[w] < mn) mn = sm.wkmin[w];
+                    if (sm.wkmax[w] > mx) mx = sm.wkmax[w];
                 }
-                cur = nxt;
-                __syncthreads();
-            }
-            const uint16_t *mo = sm.perm[cur];
-
-            // --- group heads + previous-tile continuation skip
-            for (int32_t i = tid; i < M; i += blockDim.x) {
-                int64_t kk = sm.skey[mo[i]];
-                uint8_t h = (i == 0) ? 1 : (kk != sm.skey[mo[i - 1]]);
-                if (sm.haspred && kk == sm.predkey) h = 0;
-                sm.head[i] = h;
+                sm.tile_kmin = mn;
+                sm.narrow = (mx - mn) < 0xfffffffeull;
+                const uint64_t up = ukey(sm.predkey);
+                sm.predkey32 =
+                    (sm.haspred && up >= mn && up - mn < 0xfffffffeull)
+                        ? (uint32_t)(up - mn)
+                        : 0xffffffffu;
             }
             __syncthreads();
+            if (sm.narrow) {
+                // in-place 64->32-bit repack in two barriered halves (the
+                // destination bytes of each half never overlap the other
+                // half's pending reads); 4 held values per thread per half
+                uint32_t *k32 = reinterpret_cast<uint32_t *>(sm.skey);
+#pragma unroll
+                for (int half = 0; half < 2; half++) {
+                    int64_t hold[4];
+#pragma unroll
+                    for (int h = 0; h < 4; h++) {
+                        const int32_t i =
+                            tid + (half * 4 + h) * (int32_t)blockDim.x;
+                        hold[h] = i < M ? sm.skey[i] : 0;
+                    }
+                    __syncthreads();
+#pragma unroll
+                    for (int h = 0; h < 4; h++) {
+                        const int32_t i =
+                            tid + (half * 4 + h) * (int32_t)blockDim.x;
+                        if (i < M)
+                            k32[i] =
+                                (uint32_t)(ukey(hold[h]) - sm.tile_kmin);
+                    }
+                    __syncthreads();
+                }
+            }
+
+            // --- pairwise stable merge + group heads, generic over the
+            // staged key width (uint32 tile-relative fast path; int64
+            // fallback for tiles spanning >= 2^32 of key space)
+            auto merge_heads = [&](auto *skeyT, auto predk, bool haspred2) {
+                for (int width = 1; width < k; width <<= 1) {
+                    const int nxt = cur ^ 1;
+                    const int CH = 8;
+                    int n_chunks = (M + CH - 1) / CH;
+                    for (int ch = tid; ch < n_chunks; ch += blockDim.x) {
+                        int64_t o = (int64_t)ch * CH;
+                        int remaining = (int)(M - o < CH ? M - o : CH);
+                        int p = 0;
+                        while (remaining > 0) {
+                            while ((p + 1) * 2 * width < k &&
+                                   sm.segoff[(p + 1) * 2 * width] <= o)
+                                p++;
+                            int a0 = p * 2 * width;
+                            int amid = a0 + width < k ? a0 + width : k;
+                            int b1 = a0 + 2 * width < k ? a0 + 2 * width : k;
+                            int32_t abase = sm.segoff[a0];
+                            int32_t la = sm.segoff[amid] - abase;
+                            int32_t lb = sm.segoff[b1] - sm.segoff[amid];
+                            int64_t d = o - abase;
+                            int32_t lim = la + lb - (int32_t)d;
+                            if (lim <= 0) break;
+                            int n_out = lim < remaining ? lim : remaining;
+                            const uint16_t *pa = &sm.perm[cur][abase];
+                            const uint16_t *pb = &sm.perm[cur][abase + la];
+                            int32_t ai = corank(d, pa, la, pb, lb, skeyT);
+                            int32_t bi = (int32_t)d - ai;
+                            uint16_t *out = &sm.perm[nxt][abase + d];
+                            for (int x = 0; x < n_out; x++) {
+                                bool takeA;
+                                if (ai >= la) takeA = false;
+                                else if (bi >= lb) takeA = true;
+                                else takeA =
+                                    !(skeyT[pa[ai]] > skeyT[pb[bi]]);
+                                out[x] = takeA ? pa[ai++] : pb[bi++];
+                            }
+                            o += n_out;
+                            remaining -= n_out;
+                        }
+                    }
+                    cur = nxt;
+                    __syncthreads();
+                }
+                const uint16_t *mo2 = sm.perm[cur];
+                for (int32_t i = tid; i < M; i += blockDim.x) {
+                    auto kk = skeyT[mo2[i]];
+                    uint8_t h = (i == 0) ? 1 : (kk != skeyT[mo2[i - 1]]);
+                    if (haspred2 && kk == predk) h = 0;
+                    sm.head[i] = h;
+                }
+                __syncthreads();
+            };
+            if (sm.narrow)
+                merge_heads(reinterpret_cast<const uint32_t *>(sm.skey),
+                            sm.predkey32, sm.haspred != 0);
+            else
+                merge_heads(sm.skey, sm.predkey, sm.haspred != 0);
 
             // --- winner walk (k_merge_tiles' dedup/first-row rules).
             // Pass 0 counts winners per thread; the global offset then
@@ -1192,6 +1270,16 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                                                  // emission
         if (ablate == 2) continue;
         const int64_t goff = sm.s_goff;
+        // staged keys may be 32-bit tile-relative: reconstruct the original
+        // key as (tile_kmin + rel) mapped back from ukey space
+        auto key_of = [&](uint16_t s) -> int64_t {
+            if (sm.narrow)
+                return (int64_t)((sm.tile_kmin +
+                                  reinterpret_cast<const uint32_t *>(
+                                      sm.skey)[s]) ^
+                                 0x8000000000000000ull);
+            return sm.skey[s];
+        };
 
         // emit pass fills the LDS winner list in key order (direct global
         // writes from the walk measured SLOWER: per-thread ranges scatter,
@@ -1315,7 +1403,7 @@ void k_merge_emit(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         if (key_col >= 0) {
             const int kdt = col_dtype[key_col];
             for (int32_t i = tid; i < C; i += blockDim.x) {
-                int64_t v = sm.skey[wl[i]];
+                int64_t v = key_of(wl[i]);
                 switch (kdt) {
                 case 1: ((int8_t *)out_ptrs[key_col])[goff + i] =
                             (int8_t)v; break;
