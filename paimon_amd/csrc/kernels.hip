@@ -3045,6 +3045,17 @@ __global__ void k_dict_gather_t(const int32_t *ids, const T *dict, int64_t n,
     for (; i < n; i += stride) out[i] = dict[ids[i]];
 }
 
+// endpoint seed for the single-level (k > 8) partition: write cuts for
+// b = 0 (all zeros) and b = n_bounds-1 (run lengths) so k_partition_refine
+// can treat [0, lens) as one full-width coarse window.
+__global__ void k_partition_seed(const int64_t *lens, int k,
+                                 int64_t n_bounds, int32_t *cuts) {
+    int r = threadIdx.x;
+    if (r >= k) return;
+    cuts[r] = 0;
+    cuts[(n_bounds - 1) * k + r] = (int32_t)lens[r];
+}
+
 // ---------------------------------------------------------------- launchers
 
 extern "C" {
@@ -3053,17 +3064,28 @@ hipError_t pmh_launch_partition(const DevCol *keys, const int64_t *lens, int k,
                                 int64_t tile_rows, int64_t n_bounds,
                                 int64_t total_rows, int32_t *cuts,
                                 hipStream_t stream) {
-    // two-level: wave-coarse cuts every PMH_COARSE_G tiles (64-ary domain
-    // search, ~4 probe rounds deep), then windowed refine for the interior
-    // bounds (probes confined to the enclosing coarse windows)
-    const int64_t G = n_bounds > PMH_COARSE_G + 1 ? PMH_COARSE_G : 1;
+    // k <= 8 two-level: wave-coarse cuts every PMH_COARSE_G tiles (64-ary
+    // domain search, ~4 probe rounds deep), then windowed refine for the
+    // interior bounds (probes confined to the enclosing coarse windows).
+    // k > 8: the per-lane window state of the wave kernel spills at KM >= 16
+    // and measured SLOWER than one-level bisection (16x20M: 3.9ms two-level
+    // vs 3.2ms one-level, same box — DESIGN.md §7), so seed the endpoints
+    // and run the refine kernel with a single full-width window per bound.
+    const bool two_level = k <= 8;
+    const int64_t G = two_level
+                          ? (n_bounds > PMH_COARSE_G + 1 ? PMH_COARSE_G : 1)
+                          : (n_bounds > 1 ? n_bounds - 1 : 1);
     const int64_t n_coarse = (n_bounds - 2) / G + 2;  // {0,G,..} u {last}
     int cblocks = (int)((n_coarse + 3) / 4);  // one wave per coarse bound
     int rblocks = (int)((n_bounds + 127) / 128);
     auto launch = [&](auto coarse, auto refiner) {
-        hipLaunchKernelGGL(coarse, dim3(cblocks), dim3(256), 0, stream,
-                           keys, lens, k, tile_rows, n_bounds, total_rows, G,
-                           cuts);
+        if (two_level)
+            hipLaunchKernelGGL(coarse, dim3(cblocks), dim3(256), 0, stream,
+                               keys, lens, k, tile_rows, n_bounds, total_rows,
+                               G, cuts);
+        else
+            hipLaunchKernelGGL(k_partition_seed, dim3(1), dim3(64), 0,
+                               stream, lens, k, n_bounds, cuts);
         if (G > 1)
             hipLaunchKernelGGL(refiner, dim3(rblocks), dim3(128), 0, stream,
                                keys, lens, k, tile_rows, n_bounds,

@@ -2173,17 +2173,24 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         plan->drop_delete = j["drop_delete"].as_bool(true);
         plan->ignore_delete = j["ignore_delete"].as_bool(false);
         {
-            // fused single-pass merge+emit is the product path for the
-            // winner engines; PMH_FUSED=0 keeps the 3-kernel chain for A/B
+            // Same-box A/B (DESIGN.md §7 experiment log): the classic
+            // partition->merge->scan->emit chain with the two-level
+            // partition is the fastest winner-engine configuration
+            // (C2 6.4ms vs 7.1ms split-fused; 16x20M 31.6 vs 36.4ms), so
+            // the chain is the product default. The fused single-pass path
+            // stays as the path for user-defined sequence fields (its
+            // comparator lives there) and behind PMH_FUSED=1 for A/B.
             const char *pf = getenv("PMH_FUSED");
-            plan->fused = !plan->pu && !(pf && pf[0] == '0');
+            bool want_fused = pf ? pf[0] != '0' : (plan->n_useq > 0);
+            plan->fused = !plan->pu && want_fused;
             if (plan->n_useq > 0 && !plan->pu && !plan->fused) {
                 set_error("sequence.field needs the fused merge path "
                           "(unset PMH_FUSED=0)");
                 return nullptr;
             }
-            // split emission is the default product path (fastest
-            // measured); PMH_FSPLIT=0 keeps in-kernel emission for A/B
+            // within the fused path, split value emission (merge_emit +
+            // emit_dense) is the faster variant; PMH_FSPLIT=0 keeps
+            // in-kernel emission for A/B
             const char *fs = getenv("PMH_FSPLIT");
             plan->fsplit = plan->fused && !(fs && fs[0] == '0');
         }

@@ -4,6 +4,8 @@ over real Parquet files) against the CPU oracle, on a real MI355X.
 These tests call through the C-ABI only; /root/reference is NOT read (the
 oracle golden fixtures are committed)."""
 
+import os
+
 import numpy as np
 import pytest
 
@@ -641,3 +643,43 @@ class TestProjections:
         assert (got["v_c0"] == exp["v_c0"]).all()
         assert "v_c1" not in got and "v_k" not in got
         assert list(got)[:2] != []  # column order follows the read type
+
+
+class TestFusedPathAB:
+    """The fused single-pass merge path (k_merge_emit [+ k_emit_dense]) is
+    non-default for plain dedup/first-row since the same-box A/B found the
+    classic chain faster — it remains the sequence-fields
+    path and must stay parity-green. PMH_FUSED=1 selects it explicitly."""
+
+    def _with_env(self, tmp_path, env, seed):
+        runs = gen_runs_dedup(6, 60_000, n_value_cols=4, seed=seed,
+                              delete_frac=0.15)
+        saved = {k: os.environ.get(k) for k in env}
+        os.environ.update(env)
+        try:
+            _run_and_compare(tmp_path, runs)
+        finally:
+            for k, v in saved.items():
+                if v is None:
+                    os.environ.pop(k, None)
+                else:
+                    os.environ[k] = v
+
+    def test_fused_split_pair(self, tmp_path):
+        self._with_env(tmp_path, {"PMH_FUSED": "1"}, seed=311)
+
+    def test_fused_inkernel_emit(self, tmp_path):
+        self._with_env(tmp_path, {"PMH_FUSED": "1", "PMH_FSPLIT": "0"},
+                       seed=312)
+
+    def test_default_is_legacy_chain(self, tmp_path):
+        # path_mode 0 = chain, 1 = fused in-kernel, 2 = fused split pair
+        runs = gen_runs_dedup(3, 30_000, n_value_cols=2, seed=313)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        assert "PMH_FUSED" not in os.environ
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(2)) as plan:
+                while plan.read_next() is not None:
+                    pass
+                assert plan.stats().get("path_mode", 0) == 0
