@@ -167,6 +167,12 @@ int pmh_write_parquet(const pmh_col *cols, int32_t n_cols, int64_t n_rows,
  * independent compressor. */
 int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap);
 
+/* Decode one zstd frame with the from-scratch scalar core (zstd_core.h,
+ * RFC 8878 restatement) on the host — test/validation entry for the GPU
+ * page decoder which shares the same core. Returns bytes or < 0. */
+int64_t pmh_debug_zstd_cpu(const void *src, int64_t n, void *dst,
+                           int64_t cap);
+
 /* Parse one deletion vector from a DV index file slice (DeletionFile
  * {path, offset, length}; BitmapDeletionVector.java:98-112 wrapper around
  * the portable Roaring serialization). Writes up to `cap` deleted

@@ -3056,7 +3056,21 @@ __global__ void k_partition_seed(const int64_t *lens, int k,
     cuts[(n_bounds - 1) * k + r] = (int32_t)lens[r];
 }
 
+// emit-family x-block count: 4096 measured ~5% faster than 2048 at C2
+// (smaller contiguous output slices balance the tail); PMH_EMIT_BLOCKS
+// is the A/B override.
+static int emit_grid() {
+    static int xblocks = 0;
+    if (xblocks <= 0) {
+        const char *b = getenv("PMH_EMIT_BLOCKS");
+        xblocks = b ? atoi(b) : 0;
+        if (xblocks <= 0) xblocks = 4096;
+    }
+    return xblocks;
+}
+
 // ---------------------------------------------------------------- launchers
+
 
 extern "C" {
 
@@ -3139,13 +3153,13 @@ hipError_t pmh_launch_emit_pu(const DevCol *cols, const uint8_t *col_dtype,
                               void *const *out_ptrs,
                               uint8_t *const *out_valid, hipStream_t stream) {
     if (run_masks)
-        hipLaunchKernelGGL(k_emit_pu<true>, dim3(2048), dim3(256), 0, stream,
+        hipLaunchKernelGGL(k_emit_pu<true>, dim3(emit_grid()), dim3(256), 0, stream,
                            cols, col_dtype, col_nullable, n_cols, k, seq_col,
                            kind_col, flags, members, group_start,
                            tile_offsets, n_tiles, tile_rows, total_out,
                            run_masks, out_ptrs, out_valid);
     else
-        hipLaunchKernelGGL(k_emit_pu<false>, dim3(2048), dim3(256), 0, stream,
+        hipLaunchKernelGGL(k_emit_pu<false>, dim3(emit_grid()), dim3(256), 0, stream,
                            cols, col_dtype, col_nullable, n_cols, k, seq_col,
                            kind_col, flags, members, group_start,
                            tile_offsets, n_tiles, tile_rows, total_out,
@@ -3177,7 +3191,7 @@ hipError_t pmh_launch_emit_pu_sg(
     const int64_t *tile_offsets, int64_t n_tiles, int64_t tile_rows,
     const int64_t *total_out, uint64_t *const *run_masks,
     void *const *out_ptrs, uint8_t *const *out_valid, hipStream_t stream) {
-    hipLaunchKernelGGL(k_emit_pu_sg, dim3(2048), dim3(256), 0, stream, cols,
+    hipLaunchKernelGGL(k_emit_pu_sg, dim3(emit_grid()), dim3(256), 0, stream, cols,
                        col_dtype, col_nullable, n_cols, k, seq_col, kind_col,
                        flags, col_group, sg_fields, sg_nseq, n_groups,
                        members, group_start, tile_offsets, n_tiles,
@@ -3198,13 +3212,13 @@ hipError_t pmh_launch_emit_agg(const DevCol *cols, const uint8_t *col_dtype,
                                uint8_t *const *out_valid,
                                hipStream_t stream) {
     if (run_masks)
-        hipLaunchKernelGGL(k_emit_agg<true>, dim3(2048), dim3(256), 0, stream,
+        hipLaunchKernelGGL(k_emit_agg<true>, dim3(emit_grid()), dim3(256), 0, stream,
                            cols, col_dtype, col_nullable, col_agg, n_cols, k,
                            seq_col, kind_col, flags, members, group_start,
                            tile_offsets, n_tiles, tile_rows, total_out,
                            run_masks, out_ptrs, out_valid);
     else
-        hipLaunchKernelGGL(k_emit_agg<false>, dim3(2048), dim3(256), 0,
+        hipLaunchKernelGGL(k_emit_agg<false>, dim3(emit_grid()), dim3(256), 0,
                            stream, cols, col_dtype, col_nullable, col_agg,
                            n_cols, k, seq_col, kind_col, flags, members,
                            group_start, tile_offsets, n_tiles, tile_rows,
@@ -3269,7 +3283,7 @@ hipError_t pmh_launch_emit_dense(const DevCol *cols,
                                  void *const *out_ptrs,
                                  uint8_t *const *out_valid,
                                  hipStream_t stream) {
-    hipLaunchKernelGGL(k_emit_dense, dim3(2048), dim3(256), 0, stream, cols,
+    hipLaunchKernelGGL(k_emit_dense, dim3(emit_grid()), dim3(256), 0, stream, cols,
                        col_dtype, col_nullable, n_cols, key_col, seq_col,
                        kind_col, winners, t0, t1, status, out_ptrs,
                        out_valid);
@@ -3292,7 +3306,7 @@ hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
                            int64_t tile_rows, const int64_t *total_out,
                            void *const *out_ptrs, uint8_t *const *out_valid,
                            hipStream_t stream) {
-    hipLaunchKernelGGL(k_emit, dim3(2048), dim3(256), 0, stream, cols,
+    hipLaunchKernelGGL(k_emit, dim3(emit_grid()), dim3(256), 0, stream, cols,
                        col_dtype, col_nullable, n_cols, k, winners,
                        tile_counts, tile_offsets, n_tiles, tile_rows,
                        total_out, out_ptrs, out_valid);

@@ -23,6 +23,7 @@
 
 #include "../../include/paimon_hip.h"
 #include "codec.h"
+#include "zstd_core.h"
 #include "common.h"
 #include "json.h"
 #include "kernels.h"
@@ -2804,6 +2805,19 @@ int64_t pmh_debug_parse_dv(const char *path, int64_t offset, int64_t length,
     int64_t n = (int64_t)pos.size() < cap ? (int64_t)pos.size() : cap;
     for (int64_t i = 0; i < n; i++) out[i] = pos[i];
     return (int64_t)pos.size();
+}
+
+// CPU-side entry to the scalar zstd restatement (zstd_core.h): decode one
+// frame; the CPU tests fuzz this against libzstd before the GPU kernel
+// (which shares the same core) ever runs.
+int64_t pmh_debug_zstd_cpu(const void *src, int64_t n, void *dst,
+                           int64_t cap) {
+    std::vector<uint8_t> lit(PZ_BLOCK_MAX);
+    std::vector<PzCtx> cx(1);
+    int64_t r = pz_decode_frame((const uint8_t *)src, n, (uint8_t *)dst, cap,
+                                lit.data(), cx.data());
+    if (r < 0) set_error("pz_decode_frame: error %lld", (long long)r);
+    return r;
 }
 
 int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap) {

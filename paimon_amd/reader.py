@@ -198,6 +198,24 @@ def debug_parse_dv(path, offset=0, length=0, cap=1 << 22):
     return out[:min(n, cap)].copy()
 
 
+def debug_zstd_cpu(data, cap=None):
+    """Decode one zstd frame with the scalar zstd_core.h restatement on the
+    host (test entry; the GPU page decoder shares the same core)."""
+    lib = load_lib()
+    lib.pmh_debug_zstd_cpu.restype = ctypes.c_int64
+    lib.pmh_debug_zstd_cpu.argtypes = [ctypes.c_void_p, ctypes.c_int64,
+                                       ctypes.c_void_p, ctypes.c_int64]
+    data = bytes(data)
+    if cap is None:
+        cap = max(16, len(data) * 64)
+    out = np.empty(cap, dtype=np.uint8)
+    n = lib.pmh_debug_zstd_cpu(data, len(data),
+                               out.ctypes.data_as(ctypes.c_void_p), cap)
+    if n < 0:
+        raise RuntimeError(last_error())
+    return bytes(out[:n])
+
+
 def interval_partition(min_keys, max_keys):
     """Returns (section_id, run_id) per input file, per the IntervalPartition
     restatement in libpaimon_hip (plan.cpp)."""
