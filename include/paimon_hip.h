@@ -103,6 +103,7 @@ typedef struct pmh_stats {
     double h2d_ms;           /* untimed-region staging cost, informational */
     int64_t path_mode;       /* 0 = 3-kernel chain, 1 = fused in-kernel
                               * emission, 2 = fused + split value emission */
+    int64_t gpu_zstd_pages;  /* pages decompressed by k_zstd_pages */
 } pmh_stats;
 
 /* Session: owns the device + stream pool. device < 0 opens a host-only
@@ -172,6 +173,11 @@ int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap);
  * page decoder which shares the same core. Returns bytes or < 0. */
 int64_t pmh_debug_zstd_cpu(const void *src, int64_t n, void *dst,
                            int64_t cap);
+
+/* Decode one zstd frame ON THE GPU (k_zstd_pages batch of 1); `expected`
+ * is the known decompressed size. Returns expected or < 0. */
+int64_t pmh_debug_zstd_gpu(const void *src, int64_t n, void *dst,
+                           int64_t expected);
 
 /* Parse one deletion vector from a DV index file slice (DeletionFile
  * {path, offset, length}; BitmapDeletionVector.java:98-112 wrapper around

@@ -144,6 +144,18 @@ hipError_t pmh_launch_delta_emit(const DeltaChunk *chunks, int64_t n_chunks,
 
 // Decode ORC RLEv2 / byte-RLE work chunks into a dense typed column
 // (int32 or int64 elements per chunk.out_esize). One wave per chunk.
+// on-GPU zstd page decompression (k_zstd_pages): one job per parquet page.
+// scratch must hold n_jobs * PZ_SLOT bytes (PZ_SLOT in zstd_core.h);
+// status[j] = decompressed bytes or a PZ_ERR_* code.
+struct ZstdJob {
+    uint64_t src_off, dst_off;  // into the batch src / dst blobs
+    uint32_t src_len, dst_len;
+};
+hipError_t pmh_launch_zstd_pages(const uint8_t *src,
+                                 const struct ZstdJob *jobs, int n,
+                                 uint8_t *dst, uint8_t *scratch,
+                                 int64_t *status, hipStream_t stream);
+
 hipError_t pmh_launch_rlev2(const Rlev2Chunk *chunks, int64_t n_chunks,
                             hipStream_t stream);
 

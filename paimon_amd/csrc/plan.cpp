@@ -63,6 +63,86 @@ static bool zstd_decompress(const uint8_t *src, size_t src_n, uint8_t *dst,
 }
 
 
+
+// ------------------------------------------------ on-GPU zstd page batch
+//
+// Decode all zstd pages of one column chunk on the GPU (k_zstd_pages, one
+// wavefront per page frame) and copy the uncompressed image back to the
+// host staging buffer. v1 keeps the host round trip so every downstream
+// staging path (def-level peeks, dict pages, DELTA prescan) is unchanged;
+// the PLAIN fast path staying device-resident is the follow-up. Returns
+// false (WITHOUT set_error) to fall back to the host codec.
+static bool gpu_zstd_enabled() {
+    static int on = -1;
+    if (on < 0) {
+        const char *e = getenv("PMH_GPU_ZSTD");
+        on = !(e && e[0] == '0');
+        if (on) {
+            int ndev = 0;
+            if (hipGetDeviceCount(&ndev) != hipSuccess || ndev == 0) on = 0;
+        }
+    }
+    return on;
+}
+
+struct ZstdBatchPage {
+    int64_t src_off;  // into src_base
+    int64_t src_len;
+    int64_t dst_off;  // into the host image
+    int64_t dst_len;
+};
+
+static bool gpu_zstd_batch(const uint8_t *src_base, int64_t src_total,
+                           const std::vector<ZstdBatchPage> &pages,
+                           uint8_t *host_out, int64_t out_total) {
+    if (pages.empty()) return true;
+    int n = (int)pages.size();
+    uint8_t *d_src = nullptr, *d_dst = nullptr, *d_scr = nullptr;
+    ZstdJob *d_jobs = nullptr;
+    int64_t *d_st = nullptr;
+    std::vector<ZstdJob> jobs(n);
+    for (int i = 0; i < n; i++) {
+        jobs[i] = {(uint64_t)pages[i].src_off, (uint64_t)pages[i].dst_off,
+                   (uint32_t)pages[i].src_len, (uint32_t)pages[i].dst_len};
+    }
+    bool ok = false;
+    std::vector<int64_t> st(n);
+    do {
+        if (hipMalloc(&d_src, src_total) != hipSuccess) break;
+        if (hipMalloc(&d_dst, out_total) != hipSuccess) break;
+        if (hipMalloc(&d_scr, (size_t)n * PZ_SLOT) != hipSuccess) break;
+        if (hipMalloc(&d_jobs, n * sizeof(ZstdJob)) != hipSuccess) break;
+        if (hipMalloc(&d_st, n * 8) != hipSuccess) break;
+        if (hipMemcpy(d_src, src_base, src_total, hipMemcpyHostToDevice) !=
+            hipSuccess)
+            break;
+        if (hipMemcpy(d_jobs, jobs.data(), n * sizeof(ZstdJob),
+                      hipMemcpyHostToDevice) != hipSuccess)
+            break;
+        if (pmh_launch_zstd_pages(d_src, d_jobs, n, d_dst, d_scr, d_st,
+                                  nullptr) != hipSuccess)
+            break;
+        if (hipMemcpy(st.data(), d_st, n * 8, hipMemcpyDeviceToHost) !=
+            hipSuccess)
+            break;
+        bool all = true;
+        for (int i = 0; i < n; i++)
+            if (st[i] != pages[i].dst_len) all = false;
+        if (!all) break;
+        if (hipMemcpy(host_out, d_dst, out_total, hipMemcpyDeviceToHost) !=
+            hipSuccess)
+            break;
+        ok = true;
+    } while (0);
+    if (d_src) (void)hipFree(d_src);
+    if (d_dst) (void)hipFree(d_dst);
+    if (d_scr) (void)hipFree(d_scr);
+    if (d_jobs) (void)hipFree(d_jobs);
+    if (d_st) (void)hipFree(d_st);
+    return ok;
+}
+
+
 // ---------------------------------------------------- deletion vectors
 //
 // Paimon deletion vectors: per data file, a RoaringBitmap32
@@ -1375,8 +1455,33 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     int64_t total_unc = 0;
                     for (auto &pg : cc.pages) total_unc += pg.uncompressed_size;
                     packed.resize(total_unc);
+                    bool gpu_done = false;
+                    if (cc.codec == CODEC_ZSTD && gpu_zstd_enabled()) {
+                        // batch the chunk's page frames through k_zstd_pages;
+                        // any failure falls back to the host codec below
+                        std::vector<ZstdBatchPage> zp(cc.pages.size());
+                        int64_t lo = INT64_MAX, hi = 0, o = 0;
+                        for (auto &pg : cc.pages) {
+                            if (pg.data_off < lo) lo = pg.data_off;
+                            if (pg.data_off + pg.compressed_size > hi)
+                                hi = pg.data_off + pg.compressed_size;
+                        }
+                        for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                            auto &pg = cc.pages[pi];
+                            zp[pi] = {pg.data_off - lo, pg.compressed_size, o,
+                                      pg.uncompressed_size};
+                            ppo[pi] = o;
+                            o += pg.uncompressed_size;
+                        }
+                        gpu_done = gpu_zstd_batch(sf.data.data() + lo,
+                                                  hi - lo, zp, packed.data(),
+                                                  total_unc);
+                        plan->stats.gpu_zstd_pages +=
+                            gpu_done ? (int64_t)cc.pages.size() : 0;
+                    }
                     int64_t off = 0;
-                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                    for (size_t pi = 0;
+                         !gpu_done && pi < cc.pages.size(); pi++) {
                         auto &pg = cc.pages[pi];
                         std::string cerr;
                         size_t got = 0;
@@ -2818,6 +2923,20 @@ int64_t pmh_debug_zstd_cpu(const void *src, int64_t n, void *dst,
                                 lit.data(), cx.data());
     if (r < 0) set_error("pz_decode_frame: error %lld", (long long)r);
     return r;
+}
+
+// GPU round trip of the zstd page kernel: decode one frame whose
+// decompressed size the caller knows (parity tests vs libzstd on the box).
+int64_t pmh_debug_zstd_gpu(const void *src, int64_t n, void *dst,
+                           int64_t expected) {
+    std::vector<ZstdBatchPage> zp(1);
+    zp[0] = {0, n, 0, expected};
+    if (!gpu_zstd_batch((const uint8_t *)src, n, zp, (uint8_t *)dst,
+                        expected)) {
+        set_error("gpu zstd decode failed (batch of 1)");
+        return -1;
+    }
+    return expected;
 }
 
 int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap) {

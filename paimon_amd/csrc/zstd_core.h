@@ -650,6 +650,10 @@ typedef struct {
     uint64_t rep[3];
 } PzCtx;
 
+// per-page scratch slot for the GPU kernel: literals buffer + context
+#define PZ_SLOT \
+    ((size_t)PZ_BLOCK_MAX + ((sizeof(PzCtx) + 255) & ~(size_t)255))
+
 // Serial zstd frame decode (RFC 8878 subset: no dictionary; content
 // checksum skipped). Returns decompressed size or a PZ_ERR_* code.
 // litbuf must hold PZ_BLOCK_MAX bytes.

@@ -52,7 +52,8 @@ class _Stats(ctypes.Structure):
                 ("emit_ms", ctypes.c_double),
                 ("total_device_ms", ctypes.c_double),
                 ("h2d_ms", ctypes.c_double),
-                ("path_mode", ctypes.c_int64)]
+                ("path_mode", ctypes.c_int64),
+                ("gpu_zstd_pages", ctypes.c_int64)]
 
 
 _lib = None
@@ -211,6 +212,22 @@ def debug_zstd_cpu(data, cap=None):
     out = np.empty(cap, dtype=np.uint8)
     n = lib.pmh_debug_zstd_cpu(data, len(data),
                                out.ctypes.data_as(ctypes.c_void_p), cap)
+    if n < 0:
+        raise RuntimeError(last_error())
+    return bytes(out[:n])
+
+
+def debug_zstd_gpu(data, expected_size):
+    """Decode one zstd frame on the GPU (k_zstd_pages); requires a device."""
+    lib = load_lib()
+    lib.pmh_debug_zstd_gpu.restype = ctypes.c_int64
+    lib.pmh_debug_zstd_gpu.argtypes = [ctypes.c_void_p, ctypes.c_int64,
+                                       ctypes.c_void_p, ctypes.c_int64]
+    data = bytes(data)
+    out = np.empty(max(1, expected_size), dtype=np.uint8)
+    n = lib.pmh_debug_zstd_gpu(data, len(data),
+                               out.ctypes.data_as(ctypes.c_void_p),
+                               expected_size)
     if n < 0:
         raise RuntimeError(last_error())
     return bytes(out[:n])
