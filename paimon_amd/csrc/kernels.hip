@@ -3060,24 +3060,369 @@ __global__ void k_partition_seed(const int64_t *lens, int k,
 
 // ------------------------------------------------------------ k_zstd_pages
 //
-// On-GPU zstd page decompression: one wavefront per parquet
-// page; v0 runs the scalar RFC 8878 core (zstd_core.h — fuzz-pinned against
-// libzstd on the host) on lane 0, with the page-level parallelism carrying
-// the throughput (thousands of pages in flight across 256 CUs). Scratch
-// holds a 128 KB literals buffer + decode context per job slot.
-__global__ void k_zstd_pages(const uint8_t *src, const ZstdJob *jobs, int n,
-                             uint8_t *dst, uint8_t *scratch,
-                             int64_t *status) {
-    const int wave = (int)((blockIdx.x * blockDim.x + threadIdx.x) >> 6);
+// On-GPU zstd page decompression: one WAVEFRONT per parquet
+// page frame. The scalar RFC 8878 core (zstd_core.h — fuzz-pinned against
+// libzstd on the host) does all header/table parsing; the wave supplies the
+// parallelism where the bytes move:
+//  - raw/RLE blocks and raw/RLE literals: 64-lane copies/fills;
+//  - huffman literals: the 4 interleaved streams decode on 4 lanes in SIMD
+//    lockstep, each with a register 64-bit bit-window (one 8-byte refill
+//    per ~7 symbols instead of 6 dependent byte loads per symbol), LUT in
+//    LDS;
+//  - sequence execution: 64-lane literal copies and match copies (overlap
+//    handled with the period trick: byte j of a match reads
+//    dst[start - off + j % off]), __threadfence_block between dependent
+//    phases (same-CU L1 visibility);
+//  - sequence DECODE stays serial on lane 0 (FSE state chain), buffered 64
+//    sequences at a time through LDS.
+// Scratch per job: 128 KB literals buffer + decode context.
+
+// per-wave LDS workspace
+struct ZWaveLds {
+    PzHuf hlut[1 << PZ_HUF_LOG_MAX];  // 4 KB huffman LUT
+    uint32_t ll[64], ml[64];          // sequence ring (lane 0 -> wave)
+    uint64_t off[64];
+    int32_t stat[4];                  // per-stream literal decode status
+    int64_t flag;                     // lane-0 -> wave broadcast / error
+};
+
+// backward bit reader with a register window (device hot loops)
+struct ZBack {
+    const uint8_t *buf;
+    int64_t bp;    // absolute bit position of the next read's top
+    uint64_t win;  // bits [wlo, wlo+64)
+    int64_t wlo;
+    DEV void init(const uint8_t *b, int64_t sentinel_bp) {
+        buf = b;
+        bp = sentinel_bp;
+        wlo = INT64_MIN / 2;
+        win = 0;
+    }
+    DEV void refill(int64_t lo) {
+        int64_t byte0 = lo >> 3;  // floor, works for negative lo
+        wlo = byte0 << 3;
+        uint64_t v = 0;
+#pragma unroll
+        for (int i = 0; i < 8; i++) {
+            int64_t by = byte0 + i;
+            v |= (uint64_t)(by >= 0 ? buf[by] : (uint8_t)0) << (8 * i);
+        }
+        win = v;
+    }
+    DEV uint32_t peek(int n) {  // bits [bp-n, bp), n <= 57
+        int64_t lo = bp - n;
+        if (lo < wlo || bp > wlo + 64) refill(lo);
+        return (uint32_t)((win >> (lo - wlo)) &
+                          ((n >= 32) ? 0xFFFFFFFFu : ((1u << n) - 1u)));
+    }
+};
+
+static DEV void z_fence() { __threadfence_block(); }
+
+// decode one frame with one wave. Returns decompressed size or PZ_ERR_*.
+static DEV int64_t zdev_page(const uint8_t *src, int64_t slen, uint8_t *dst,
+                             int64_t dcap, uint8_t *lit, PzCtx *cx,
+                             ZWaveLds *L) {
     const int lane = threadIdx.x & 63;
+    PzFrame F;
+    int fh = pz_parse_frame(src, slen, &F);  // all lanes, redundant+uniform
+    if (fh < 0) return fh;
+    int64_t sp = fh, dp = 0;
+    int htl = -1;
+    PzSeqState st;  // lane 0 only (sequence states + repeat offsets)
+    st.rep[0] = 1;
+    st.rep[1] = 4;
+    st.rep[2] = 8;
+    if (lane == 0) {
+        cx->htl = -1;
+        cx->have_ll = cx->have_of = cx->have_ml = 0;
+    }
+    int last = 0;
+    while (!last) {
+        if (sp + 3 > slen) return PZ_ERR_SRC_SMALL;
+        uint32_t bh = (uint32_t)src[sp] | ((uint32_t)src[sp + 1] << 8) |
+                      ((uint32_t)src[sp + 2] << 16);
+        sp += 3;
+        last = bh & 1;
+        int btype = (bh >> 1) & 3;
+        int64_t bsize = bh >> 3;
+        if (btype == 0) {  // raw block
+            if (sp + bsize > slen || dp + bsize > dcap)
+                return PZ_ERR_SRC_SMALL;
+            for (int64_t j = lane; j < bsize; j += 64)
+                dst[dp + j] = src[sp + j];
+            z_fence();
+            sp += bsize;
+            dp += bsize;
+            continue;
+        }
+        if (btype == 1) {  // RLE block
+            if (sp + 1 > slen || dp + bsize > dcap) return PZ_ERR_SRC_SMALL;
+            uint8_t v = src[sp];
+            for (int64_t j = lane; j < bsize; j += 64) dst[dp + j] = v;
+            z_fence();
+            sp += 1;
+            dp += bsize;
+            continue;
+        }
+        if (btype != 2 || bsize > slen - sp) return PZ_ERR_BLOCK;
+        const uint8_t *bb = src + sp;
+        int64_t bn = bsize;
+        sp += bsize;
+        PzLits Lh;
+        int rc = pz_parse_lits(bb, bn, &Lh);  // uniform on all lanes
+        if (rc < 0) return rc;
+        int64_t pos = Lh.hdr;
+        int64_t nlit = Lh.regen;
+        if (nlit > PZ_BLOCK_MAX) return PZ_ERR_LITERALS;
+        if (Lh.type == 0) {  // raw literals
+            if (pos + nlit > bn) return PZ_ERR_SRC_SMALL;
+            for (int64_t j = lane; j < nlit; j += 64) lit[j] = bb[pos + j];
+            pos += nlit;
+        } else if (Lh.type == 1) {  // RLE literals
+            if (pos + 1 > bn) return PZ_ERR_SRC_SMALL;
+            uint8_t v = bb[pos];
+            for (int64_t j = lane; j < nlit; j += 64) lit[j] = v;
+            pos += 1;
+        } else {  // huffman literals (2 = new tree, 3 = treeless)
+            if (pos + Lh.comp > bn) return PZ_ERR_SRC_SMALL;
+            const uint8_t *hp = bb + pos;
+            int64_t hn = Lh.comp;
+            int64_t off = 0;
+            if (Lh.type == 2) {
+                if (lane == 0) {
+                    int nw;
+                    int used = pz_huf_read_weights(hp, hn, cx->weights, &nw,
+                                                   cx->wksp64, cx->norm);
+                    int tl = used < 0
+                                 ? used
+                                 : pz_huf_build(cx->weights, nw, L->hlut);
+                    L->flag = used < 0 ? used
+                                       : (tl < 0 ? tl
+                                                 : (((int64_t)tl << 32) |
+                                                    (uint32_t)used));
+                }
+                z_fence();
+                int64_t f = L->flag;
+                if (f < 0) return f;
+                htl = (int)(f >> 32);
+                off = (int64_t)(uint32_t)f;
+            }
+            if (htl < 0) return PZ_ERR_HUFFMAN;
+            // stream layout: 1 stream, or 4 with a 6-byte jump table
+            int ns = Lh.n_streams;
+            const uint8_t *s0 = hp + off;
+            int64_t szs[4], outs[4], oofs[4], sofs[4];
+            if (ns == 1) {
+                szs[0] = hn - off;
+                outs[0] = nlit;
+                oofs[0] = 0;
+                sofs[0] = 0;
+            } else {
+                if (hn - off < 6) return PZ_ERR_SRC_SMALL;
+                int64_t s1 = s0[0] | ((int64_t)s0[1] << 8);
+                int64_t s2 = s0[2] | ((int64_t)s0[3] << 8);
+                int64_t s3 = s0[4] | ((int64_t)s0[5] << 8);
+                int64_t s4 = (hn - off - 6) - s1 - s2 - s3;
+                if (s4 <= 0) return PZ_ERR_LITERALS;
+                int64_t q = (nlit + 3) / 4;
+                if (3 * q > nlit) return PZ_ERR_LITERALS;
+                s0 += 6;
+                szs[0] = s1;
+                szs[1] = s2;
+                szs[2] = s3;
+                szs[3] = s4;
+                outs[0] = outs[1] = outs[2] = q;
+                outs[3] = nlit - 3 * q;
+                sofs[0] = 0;
+                sofs[1] = s1;
+                sofs[2] = s1 + s2;
+                sofs[3] = s1 + s2 + s3;
+                oofs[0] = 0;
+                oofs[1] = q;
+                oofs[2] = 2 * q;
+                oofs[3] = 3 * q;
+            }
+            if (lane < ns) {
+                const uint8_t *sb = s0 + sofs[lane];
+                int64_t want = outs[lane];
+                int64_t bp0 = pz_back_init(sb, szs[lane]);
+                int rcs = 0;
+                if (bp0 < 0) {
+                    rcs = PZ_ERR_HUFFMAN;
+                } else {
+                    ZBack rb;
+                    rb.init(sb, bp0);
+                    uint8_t *out = lit + oofs[lane];
+                    for (int64_t i = 0; i < want; i++) {
+                        uint32_t idx = rb.peek(htl);
+                        PzHuf e = L->hlut[idx];
+                        out[i] = e.sym;
+                        rb.bp -= e.nbits;
+                        if (rb.bp < 0) {
+                            rcs = PZ_ERR_HUFFMAN;
+                            break;
+                        }
+                    }
+                    if (!rcs && rb.bp != 0) rcs = PZ_ERR_HUFFMAN;
+                }
+                L->stat[lane] = rcs;
+            }
+            z_fence();
+            for (int i = 0; i < ns; i++)
+                if (L->stat[i] < 0) return L->stat[i];
+            pos += hn;
+        }
+        z_fence();  // literals visible to the whole wave
+        // ---------------- sequences
+        if (pos >= bn) return PZ_ERR_SEQ;
+        const uint8_t *sq = bb + pos;
+        int64_t sn = bn - pos;
+        int b0 = sq[0];
+        int nseq;
+        int64_t so = 1;
+        if (b0 < 128) {
+            nseq = b0;
+        } else if (b0 < 255) {
+            if (sn < 2) return PZ_ERR_SRC_SMALL;
+            nseq = ((b0 - 128) << 8) + sq[1];
+            so = 2;
+        } else {
+            if (sn < 3) return PZ_ERR_SRC_SMALL;
+            nseq = sq[1] + (sq[2] << 8) + 0x7F00;
+            so = 3;
+        }
+        if (nseq == 0) {  // literals only
+            if (dp + nlit > dcap) return PZ_ERR_DST_SMALL;
+            for (int64_t j = lane; j < nlit; j += 64) dst[dp + j] = lit[j];
+            z_fence();
+            dp += nlit;
+            continue;
+        }
+        if (so >= sn) return PZ_ERR_SRC_SMALL;
+        int modes = sq[so++];
+        if (modes & 3) return PZ_ERR_SEQ;
+        if (lane == 0) {  // FSE table builds + stream init, serial
+            int64_t f = 0;
+            int used = pz_seq_table(sq + so, sn - so, (modes >> 6) & 3, 0, 9,
+                                    35, cx->llT, &cx->ll_al, cx->have_ll,
+                                    cx->norm);
+            if (used >= 0) {
+                cx->have_ll = 1;
+                int64_t so2 = so + used;
+                int u2 = pz_seq_table(sq + so2, sn - so2, (modes >> 4) & 3, 1,
+                                      8, 31, cx->ofT, &cx->of_al,
+                                      cx->have_of, cx->norm);
+                if (u2 >= 0) {
+                    cx->have_of = 1;
+                    so2 += u2;
+                    int u3 = pz_seq_table(sq + so2, sn - so2,
+                                          (modes >> 2) & 3, 2, 9, 52,
+                                          cx->mlT, &cx->ml_al, cx->have_ml,
+                                          cx->norm);
+                    if (u3 >= 0) {
+                        cx->have_ml = 1;
+                        so2 += u3;
+                        f = so2 - so;
+                    } else {
+                        f = u3;
+                    }
+                } else {
+                    f = u2;
+                }
+            } else {
+                f = used;
+            }
+            if (f >= 0) {
+                uint64_t k0 = st.rep[0], k1 = st.rep[1], k2 = st.rep[2];
+                int rci = pz_seq_init(sq + so + f, sn - so - f, cx->ll_al,
+                                      cx->of_al, cx->ml_al, &st);
+                st.rep[0] = k0;  // repeat offsets persist across blocks
+                st.rep[1] = k1;
+                st.rep[2] = k2;
+                if (rci < 0) f = rci;
+            }
+            L->flag = f;
+        }
+        z_fence();
+        {
+            int64_t f = L->flag;
+            if (f < 0) return f;
+            so += f;
+        }
+        const uint8_t *bs = sq + so;
+        int64_t lpos = 0;
+        int done = 0;
+        while (done < nseq) {
+            int cnt = nseq - done < 64 ? nseq - done : 64;
+            if (lane == 0) {
+                int64_t f = 0;
+                for (int i = 0; i < cnt; i++) {
+                    PzSeq q;
+                    int rcq = pz_seq_next(bs, cx->llT, cx->ofT, cx->mlT, &st,
+                                          done + i == nseq - 1, &q);
+                    if (rcq < 0) {
+                        f = rcq;
+                        break;
+                    }
+                    L->ll[i] = q.ll;
+                    L->ml[i] = q.ml;
+                    L->off[i] = q.off;
+                }
+                L->flag = f;
+            }
+            z_fence();
+            if (L->flag < 0) return L->flag;
+            for (int i = 0; i < cnt; i++) {
+                int64_t ll = L->ll[i], ml = L->ml[i];
+                int64_t mo = (int64_t)L->off[i];
+                if (lpos + ll > nlit) return PZ_ERR_SEQ;
+                if (dp + ll + ml > dcap) return PZ_ERR_DST_SMALL;
+                for (int64_t j = lane; j < ll; j += 64)
+                    dst[dp + j] = lit[lpos + j];
+                z_fence();
+                dp += ll;
+                lpos += ll;
+                if (mo > dp) return PZ_ERR_OFFSET;
+                if (mo >= ml) {  // non-overlapping: plain parallel copy
+                    for (int64_t j = lane; j < ml; j += 64)
+                        dst[dp + j] = dst[dp + j - mo];
+                } else {  // overlapping: period-mo pattern replication
+                    for (int64_t j = lane; j < ml; j += 64)
+                        dst[dp + j] = dst[dp - mo + (j % mo)];
+                }
+                z_fence();
+                dp += ml;
+            }
+            done += cnt;
+        }
+        if (lane == 0) L->flag = st.bp == 0 ? 0 : PZ_ERR_SEQ;
+        z_fence();
+        if (L->flag < 0) return L->flag;
+        int64_t rem = nlit - lpos;
+        if (rem < 0 || dp + rem > dcap) return PZ_ERR_DST_SMALL;
+        for (int64_t j = lane; j < rem; j += 64) dst[dp + j] = lit[lpos + j];
+        z_fence();
+        dp += rem;
+    }
+    if (F.content_size >= 0 && dp != F.content_size) return PZ_ERR_CORRUPT;
+    return dp;
+}
+
+__launch_bounds__(256) __global__
+void k_zstd_pages(const uint8_t *src, const ZstdJob *jobs, int n,
+                  uint8_t *dst, uint8_t *scratch, int64_t *status) {
+    __shared__ ZWaveLds lds[4];
+    const int wave = (int)((blockIdx.x * blockDim.x + threadIdx.x) >> 6);
+    const int wiw = (threadIdx.x >> 6) & 3;
     const int waves = (int)((gridDim.x * blockDim.x) >> 6);
     for (int j = wave; j < n; j += waves) {
-        if (lane != 0) continue;
         uint8_t *lit = scratch + (size_t)j * PZ_SLOT;
         PzCtx *cx = (PzCtx *)(lit + PZ_BLOCK_MAX);
-        status[j] = pz_decode_frame(src + jobs[j].src_off, jobs[j].src_len,
-                                    dst + jobs[j].dst_off, jobs[j].dst_len,
-                                    lit, cx);
+        int64_t r = zdev_page(src + jobs[j].src_off, jobs[j].src_len,
+                              dst + jobs[j].dst_off, jobs[j].dst_len, lit,
+                              cx, &lds[wiw]);
+        if ((threadIdx.x & 63) == 0) status[j] = r;
     }
 }
 

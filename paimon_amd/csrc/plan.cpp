@@ -24,6 +24,7 @@
 #include "../../include/paimon_hip.h"
 #include "codec.h"
 #include "zstd_core.h"
+#include <map>
 #include "common.h"
 #include "json.h"
 #include "kernels.h"
@@ -1435,6 +1436,43 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                 return false;
             }
         }
+        // file-level GPU zstd pre-pass: decode EVERY
+        // selected zstd chunk's pages in ONE k_zstd_pages batch — thousands
+        // of wavefronts in flight hide the per-page serial decode latency
+        // that per-chunk batches cannot. Failure falls back to the host
+        // codec per chunk.
+        std::vector<uint8_t> file_unc;
+        std::map<const void *, int64_t> zchunk_base;
+        if (gpu_zstd_enabled()) {
+            std::vector<ZstdBatchPage> zp;
+            int64_t lo = INT64_MAX, hi = 0, out = 0;
+            for (auto &rg : sf.meta.row_groups)
+                for (size_t c = 0; c < cols.size(); c++) {
+                    auto &cc = rg.columns[leaf[c]];
+                    if (cc.codec != CODEC_ZSTD || cc.pages.empty()) continue;
+                    if (zchunk_base.count(&cc)) continue;
+                    zchunk_base[&cc] = out;
+                    for (auto &pg : cc.pages) {
+                        if (pg.data_off < lo) lo = pg.data_off;
+                        if (pg.data_off + pg.compressed_size > hi)
+                            hi = pg.data_off + pg.compressed_size;
+                        zp.push_back({pg.data_off, pg.compressed_size, out,
+                                      pg.uncompressed_size});
+                        out += pg.uncompressed_size;
+                    }
+                }
+            if (!zp.empty()) {
+                for (auto &p : zp) p.src_off -= lo;
+                file_unc.resize(out);
+                if (gpu_zstd_batch(sf.data.data() + lo, hi - lo, zp,
+                                   file_unc.data(), out)) {
+                    plan->stats.gpu_zstd_pages += (int64_t)zp.size();
+                } else {
+                    file_unc.clear();
+                    zchunk_base.clear();
+                }
+            }
+        }
         int64_t rg_row = 0;
         for (auto &rg : sf.meta.row_groups) {
             for (size_t c = 0; c < cols.size(); c++) {
@@ -1450,38 +1488,23 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                 std::vector<uint8_t> packed;  // compressed codecs
                 std::vector<int64_t> ppo(cc.pages.size());
                 const uint8_t *payload_base = sf.data.data() + chunk_start;
-                if (cc.codec == CODEC_ZSTD || cc.codec == CODEC_GZIP ||
+                auto zit = zchunk_base.find((const void *)&cc);
+                if (zit != zchunk_base.end()) {
+                    // pages already decoded by the file-level k_zstd_pages
+                    // batch; the chunk's image starts at zit->second
+                    int64_t o = 0;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                        ppo[pi] = o;
+                        o += cc.pages[pi].uncompressed_size;
+                    }
+                    payload_base = file_unc.data() + zit->second;
+                } else if (cc.codec == CODEC_ZSTD || cc.codec == CODEC_GZIP ||
                     cc.codec == CODEC_SNAPPY) {
                     int64_t total_unc = 0;
                     for (auto &pg : cc.pages) total_unc += pg.uncompressed_size;
                     packed.resize(total_unc);
-                    bool gpu_done = false;
-                    if (cc.codec == CODEC_ZSTD && gpu_zstd_enabled()) {
-                        // batch the chunk's page frames through k_zstd_pages;
-                        // any failure falls back to the host codec below
-                        std::vector<ZstdBatchPage> zp(cc.pages.size());
-                        int64_t lo = INT64_MAX, hi = 0, o = 0;
-                        for (auto &pg : cc.pages) {
-                            if (pg.data_off < lo) lo = pg.data_off;
-                            if (pg.data_off + pg.compressed_size > hi)
-                                hi = pg.data_off + pg.compressed_size;
-                        }
-                        for (size_t pi = 0; pi < cc.pages.size(); pi++) {
-                            auto &pg = cc.pages[pi];
-                            zp[pi] = {pg.data_off - lo, pg.compressed_size, o,
-                                      pg.uncompressed_size};
-                            ppo[pi] = o;
-                            o += pg.uncompressed_size;
-                        }
-                        gpu_done = gpu_zstd_batch(sf.data.data() + lo,
-                                                  hi - lo, zp, packed.data(),
-                                                  total_unc);
-                        plan->stats.gpu_zstd_pages +=
-                            gpu_done ? (int64_t)cc.pages.size() : 0;
-                    }
                     int64_t off = 0;
-                    for (size_t pi = 0;
-                         !gpu_done && pi < cc.pages.size(); pi++) {
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
                         auto &pg = cc.pages[pi];
                         std::string cerr;
                         size_t got = 0;
