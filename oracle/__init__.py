@@ -9,7 +9,7 @@ from .oracle import (  # noqa: F401
     merge_dedup,
     merge_dedup_count_mt,
     merge_dedup_model,
-    merge_dedup_useq_model,
+    merge_dedup_useq_model, full_changelog_model,
     merge_first_row_model,
     partial_update_model,
     partial_update_rrod_model,

@@ -835,3 +835,93 @@ def merge_dedup_useq_model(runs, useq_idx, drop_delete=True,
     if drop_delete:
         sel = sel[_kind_is_add(kind[sel])]
     return run[sel], row[sel]
+
+
+def full_changelog_model(runs, levels, max_level, row_dedup=False,
+                         ignore_delete=False):
+    """Numpy restatement of FullChangelogMergeFunctionWrapper.java:74-130
+    over DeduplicateMergeFunction (getResult decision table; pinned to
+    FullChangelogMergeFunctionWrapperTestBase's vectors in
+    tests/test_changelog_cpu.py).
+
+    levels: per-run level ints; a member is the "top level kv" iff its run's
+    level == max_level (at most one per key — checkState :76-78).
+    row_dedup: the valueEqualiser (compares ALL value columns).
+
+    Returns (cl_run, cl_row, cl_kind): the changelog stream in key order
+    (UPDATE_BEFORE precedes UPDATE_AFTER), where cl_kind is the changelog
+    RowKind and (run,row) the source record whose key/seq/values it carries,
+    plus (res_run, res_row): the result stream (= merged records that are
+    adds, setResultIfNotRetract ChangelogResult.java:45).
+    """
+    key, seq, kind, run, row = _sorted_stream(runs)
+    n = len(key)
+    if n == 0:
+        e32 = np.empty(0, dtype=np.int32)
+        e64 = np.empty(0, dtype=np.int64)
+        e8 = np.empty(0, dtype=np.int8)
+        return (e32, e64, e8), (e32.copy(), e64.copy())
+    head = np.empty(n, dtype=bool)
+    head[0] = True
+    head[1:] = key[1:] != key[:-1]
+    gid = np.cumsum(head) - 1
+    ng = gid[-1] + 1
+    size = np.bincount(gid, minlength=ng)
+    idx = np.arange(n)
+    lvlarr = np.asarray(levels, dtype=np.int64)[run]
+    is_top = lvlarr == max_level
+    tops_per_group = np.bincount(gid[is_top], minlength=ng)
+    assert (tops_per_group <= 1).all(), \
+        "Top level key-value already exists (checkState :76-78)"
+    top_idx = np.full(ng, -1, dtype=np.int64)
+    np.maximum.at(top_idx, gid[is_top], idx[is_top])
+    # merged record = deduplicate result (last in (seq, isAdd) order among
+    # eligible members; ignore-delete restricts eligibility)
+    eligible = _kind_is_add(kind) if ignore_delete else np.ones(n, bool)
+    merged_idx = np.full(ng, -1, dtype=np.int64)
+    np.maximum.at(merged_idx, gid[eligible], idx[eligible])
+    single_last = np.full(ng, -1, dtype=np.int64)
+    np.maximum.at(single_last, gid, idx)
+    singles = size == 1
+    merged_idx[singles] = single_last[singles]  # singleton wrapper bypass
+    has_top = top_idx >= 0
+    has_merged = merged_idx >= 0
+    m_add = np.zeros(ng, dtype=bool)
+    m_add[has_merged] = _kind_is_add(kind[merged_idx[has_merged]])
+
+    # per-group changelog decision
+    cl_run, cl_row, cl_kind = [], [], []
+    # row_dedup equality of value columns between two flat indices
+    def _values_equal(ia, ib):
+        ra, xa = run[ia], row[ia]
+        rb, xb = run[ib], row[ib]
+        for c in range(len(runs[0]["values"])):
+            va = runs[ra]["values"][c][xa]
+            vb = runs[rb]["values"][c][xb]
+            if va != vb:
+                return False
+        return True
+
+    for g in np.flatnonzero(has_top | (has_merged & m_add)):
+        t, m = top_idx[g], merged_idx[g]
+        if size[g] == 1:
+            # wrapper not initialized: only "no top and initial is add"
+            # emits INSERT (:117-119)
+            if t < 0 and m_add[g]:
+                cl_run.append(run[m]); cl_row.append(row[m]); cl_kind.append(0)
+            continue
+        if t < 0:
+            if m >= 0 and m_add[g]:
+                cl_run.append(run[m]); cl_row.append(row[m]); cl_kind.append(0)
+        else:
+            if not (m >= 0 and m_add[g]):
+                cl_run.append(run[t]); cl_row.append(row[t]); cl_kind.append(3)
+            elif (not row_dedup) or (not _values_equal(t, m)):
+                cl_run.append(run[t]); cl_row.append(row[t]); cl_kind.append(1)
+                cl_run.append(run[m]); cl_row.append(row[m]); cl_kind.append(2)
+    # result stream: merged if add (ChangelogResult.setResultIfNotRetract)
+    sel = merged_idx[has_merged & m_add]
+    return ((np.array(cl_run, dtype=np.int32),
+             np.array(cl_row, dtype=np.int64),
+             np.array(cl_kind, dtype=np.int8)),
+            (run[sel], row[sel]))

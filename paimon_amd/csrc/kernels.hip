@@ -1690,6 +1690,7 @@ __global__ void k_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,
 // Phase 2: gather winner rows into contiguous output columns.
 // One workgroup per tile (grid-stride); thread per output row; inner loop
 // over columns. Output writes are coalesced per column.
+template <int R>
 __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
                        const uint8_t *col_dtype, const uint8_t *col_nullable,
                        int n_cols, int k, const uint32_t *winners,
@@ -1706,7 +1707,6 @@ __global__ void k_emit(const DevCol *cols /* n_runs * n_cols, run-major */,
     // Each block owns a CONTIGUOUS output slice so its gathers stay within a
     // few tiles' source segments (~130 KB working set -> XCD-L2 resident;
     // grid-striding spread every block over the whole output and thrashed L2).
-    constexpr int R = 4;
     const int64_t total = *total_out;
     const int64_t per_block =
         (total + (int64_t)gridDim.x - 1) / (int64_t)gridDim.x;
@@ -3676,10 +3676,26 @@ hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
                            int64_t tile_rows, const int64_t *total_out,
                            void *const *out_ptrs, uint8_t *const *out_valid,
                            hipStream_t stream) {
-    hipLaunchKernelGGL(k_emit, dim3(emit_grid()), dim3(256), 0, stream, cols,
-                       col_dtype, col_nullable, n_cols, k, winners,
-                       tile_counts, tile_offsets, n_tiles, tile_rows,
-                       total_out, out_ptrs, out_valid);
+    // R = rows per thread iteration (independent gathers in flight);
+    // PMH_EMIT_R sweeps 4/6/8 (4 = measured default)
+    static int r_rows = 0;
+    if (r_rows <= 0) {
+        const char *e = getenv("PMH_EMIT_R");
+        r_rows = e ? atoi(e) : 4;
+        if (r_rows != 6 && r_rows != 8) r_rows = 4;
+    }
+    auto launch = [&](auto kern) {
+        hipLaunchKernelGGL(kern, dim3(emit_grid()), dim3(256), 0, stream,
+                           cols, col_dtype, col_nullable, n_cols, k, winners,
+                           tile_counts, tile_offsets, n_tiles, tile_rows,
+                           total_out, out_ptrs, out_valid);
+    };
+    if (r_rows == 6)
+        launch(k_emit<6>);
+    else if (r_rows == 8)
+        launch(k_emit<8>);
+    else
+        launch(k_emit<4>);
     return hipGetLastError();
 }
 
