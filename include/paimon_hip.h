@@ -135,6 +135,14 @@ int pmh_plan_close(pmh_plan_t *p);
 /* Rewind the plan to its first section without restaging: the next
  * pmh_read_next re-runs the full device pipeline on the resident encoded
  * data. Used by benchmarks to repeat the timed region; stats accumulate. */
+/* Changelog batch of the LAST pmh_read_next (changelog_producer =
+ * "full-compaction" in the plan JSON, with "max_level" and optional
+ * "changelog_row_deduplicate"): the FullChangelogMergeFunctionWrapper
+ * stream (INSERT / UPDATE_BEFORE / UPDATE_AFTER / DELETE rows in key
+ * order, same schema as the main batch). Valid until the next read_next.
+ * Returns row count or < 0. */
+int64_t pmh_changelog_next(pmh_plan_t *plan, pmh_batch *out);
+
 int pmh_plan_reset(pmh_plan_t *p);
 
 int pmh_stats_get(pmh_plan_t *p, pmh_stats *out);

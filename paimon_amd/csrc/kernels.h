@@ -144,6 +144,25 @@ hipError_t pmh_launch_delta_emit(const DeltaChunk *chunks, int64_t n_chunks,
 
 // Decode ORC RLEv2 / byte-RLE work chunks into a dense typed column
 // (int32 or int64 elements per chunk.out_esize). One wave per chunk.
+
+// full-compaction changelog chain (FullChangelogMergeFunctionWrapper):
+// finalize compacts the merge pass's provisional per-group entries into
+// rows (evaluating row-deduplicate value equality); emit gathers them.
+hipError_t pmh_launch_cl_finalize(const DevCol *cols,
+                                  const uint8_t *col_dtype, int n_cols,
+                                  int first_val, int k,
+                                  const uint64_t *cl_entries,
+                                  uint64_t *cl_out, int32_t *cl_counts,
+                                  int64_t n_tiles, int64_t tile_rows,
+                                  hipStream_t stream);
+hipError_t pmh_launch_cl_emit(const DevCol *cols, const uint8_t *col_dtype,
+                              const uint8_t *col_nullable, int n_cols,
+                              int kind_col, const uint64_t *cl_rows,
+                              const int32_t *cl_counts,
+                              const int64_t *cl_offsets, int64_t n_tiles,
+                              int64_t tile_rows, void *const *out_ptrs,
+                              uint8_t *const *out_valid, hipStream_t stream);
+
 // on-GPU zstd page decompression (k_zstd_pages): one job per parquet page.
 // scratch must hold n_jobs * PZ_SLOT bytes (PZ_SLOT in zstd_core.h);
 // status[j] = decompressed bytes or a PZ_ERR_* code.
@@ -171,6 +190,8 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
                                   const uint64_t *tombs,
                                   uint32_t *winners, int32_t *tile_counts,
                                   uint16_t *group_start, uint32_t *err_flag,
+                                  const uint8_t *run_levels, int max_level,
+                                  uint64_t *cl_entries, int32_t *cl_counts,
                                   hipStream_t stream);
 
 hipError_t pmh_launch_scan_tiles(const int32_t *tile_counts, int64_t n_tiles,

@@ -397,7 +397,13 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                    int64_t n_tiles, int64_t tile_rows, int flags,
                    const uint64_t *tombs, uint32_t *winners,
                    int32_t *tile_counts, uint16_t *group_start,
-                   uint32_t *err_flag) {
+                   uint32_t *err_flag,
+                   // full-compaction changelog (cl_entries != nullptr):
+                   // per-run levels, the table's max level, and the
+                   // per-group provisional changelog entries (2 slots per
+                   // group; k_cl_finalize compacts them to rows)
+                   const uint8_t *run_levels, int max_level,
+                   uint64_t *cl_entries, int32_t *cl_counts) {
     const bool drop_delete = flags & 1;
     const bool ignore_delete = flags & 2;
     // FR = first-row engine (FirstRowMergeFunction.java:32-77): keep the
@@ -472,7 +478,10 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         const int32_t M = sm.mtotal;
         const int32_t Mreal = sm.mreal;
         if (Mreal == 0) {
-            if (tid == 0) tile_counts[tile] = 0;
+            if (tid == 0) {
+                tile_counts[tile] = 0;
+                if (cl_counts) cl_counts[tile] = 0;
+            }
             __syncthreads();
             continue;
         }
@@ -745,10 +754,27 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
         int32_t my_lo = tid * per;
         int32_t my_hi = my_lo + per < Mreal ? my_lo + per : Mreal;
         uint32_t *wout = &winners[tile * (tile_rows + PMH_MAX_RUNS)];
+        // full-compaction changelog (FullChangelogMergeFunctionWrapper.java:
+        // 96-126): per live group, up to two provisional entries, written
+        // INDEPENDENTLY of drop-delete (a dropped DELETE result still emits
+        // DELETE(topLevelKv)). flags bit 128 = row-deduplicate pending
+        // (k_cl_finalize compares values and may drop the UB/UA pair).
+        const bool cl = cl_entries != nullptr;
+        const bool cl_pend = (flags & 128) != 0;
+        uint64_t *clout =
+            cl ? &cl_entries[tile * 2 * (tile_rows + PMH_MAX_RUNS)] : nullptr;
+        // map seg index -> packed (run, global row)
+        auto packm = [&](uint16_t sx) -> uint64_t {
+            int r = 0;
+            while (r + 1 <= k - 1 && sm.segoff[r + 1] <= (int32_t)sx) r++;
+            return ((uint32_t)r << PMH_ROW_BITS) |
+                   (uint32_t)(c0[r] + ((int32_t)sx - sm.segoff[r]));
+        };
         int32_t total = 0;
         int32_t my_off = 0;
+        int32_t my_cl = 0;
         for (int pass = 0; pass < 2; pass++) {
-            int32_t nloc = 0;
+            int32_t nloc = 0, ncl = 0;
             for (int32_t i = my_lo; i < my_hi; i++) {
                 if (!sm.head[i]) continue;
                 // walk the group's LIVE members (deletion-vector tombstones
@@ -758,6 +784,8 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                 int64_t v_best = 0;
                 bool e_best = false, any_retract = false;
                 int32_t nlive = 0;
+                uint16_t s_top = 0;
+                int n_top = 0;
                 for (int32_t x = i;; x++) {
                     int64_t v = sm.sseq[mo[x]];
                     if (v != PMH_DEAD) {
@@ -771,43 +799,97 @@ void k_merge_tiles(const DevCol *keys, const DevCol *seqs, const DevCol *kinds,
                             v_best = v;
                             e_best = e;
                         }
+                        if (cl) {
+                            int r = 0;
+                            while (r + 1 <= k - 1 &&
+                                   sm.segoff[r + 1] <= (int32_t)mo[x])
+                                r++;
+                            if (run_levels[r] == max_level) {
+                                s_top = mo[x];
+                                n_top++;
+                            }
+                        }
                         nlive++;
                     }
                     if (!(x + 1 < M && !sm.head[x + 1])) break;
                 }
                 if (nlive == 0) continue;
+                if (cl) {
+                    if (n_top > 1 && err_flag)
+                        atomicOr(err_flag, 8u);  // checkState :76-78
+                    // merged = wrapper result; singletons bypass the merge
+                    // function (so an ignored lone record still serves)
+                    bool m_ok = nlive == 1 || e_best;
+                    bool m_add = m_ok && (v_best & 1);
+                    uint64_t e0 = 0, e1 = 0;
+                    int cl_n = 0;
+                    auto mk = [&](uint16_t sx, uint64_t kd,
+                                  bool pd) -> uint64_t {
+                        return packm(sx) | (kd << 32) | (1ull << 35) |
+                               (pd ? (1ull << 36) : 0);
+                    };
+                    if (n_top == 0) {
+                        if (m_add) {  // INSERT(merged) — also the singleton
+                                      // "initial is add" rule (:117-119)
+                            e0 = mk(s_best, 0, false);
+                            cl_n = 1;
+                        }
+                    } else if (nlive > 1) {
+                        if (!m_add) {  // DELETE(topLevelKv) (:106-107)
+                            e0 = mk(s_top, 3, false);
+                            cl_n = 1;
+                        } else {  // UPDATE_BEFORE(top) + UPDATE_AFTER(merged)
+                            e0 = mk(s_top, 1, cl_pend);
+                            e1 = mk(s_best, 2, cl_pend);
+                            cl_n = 2;
+                        }
+                    }  // singleton that IS the top level: no change (:120-123)
+                    if (cl_n) {
+                        if (pass == 1) {
+                            clout[2 * (my_cl + ncl)] = e0;
+                            clout[2 * (my_cl + ncl) + 1] = e1;
+                        }
+                        ncl++;
+                    }
+                }
                 if (first_row && !ignore_delete && any_retract && nlive > 1 &&
                     err_flag)
                     atomicOr(err_flag, 2u);  // FirstRow rejects retracts
                 if (!e_best && nlive > 1) continue;  // all records ignored
                 if (drop_delete && !(v_best & 1)) continue;
-                if (pass == 1) {
-                    int r = 0;  // map seg index -> (run, global row)
-                    while (r + 1 <= k - 1 &&
-                           sm.segoff[r + 1] <= (int32_t)s_best)
-                        r++;
-                    uint32_t grow = (uint32_t)(
-                        c0[r] + ((int32_t)s_best - sm.segoff[r]));
-                    wout[my_off + nloc] = ((uint32_t)r << PMH_ROW_BITS) | grow;
-                }
+                if (pass == 1)
+                    wout[my_off + nloc] = (uint32_t)packm(s_best);
                 nloc++;
             }
             if (pass == 0) {
-                int32_t incl = nloc;
+                int32_t incl = nloc, iccl = ncl;
                 for (int off = 1; off < 64; off <<= 1) {
                     int32_t up = __shfl_up(incl, off, 64);
-                    if (lane >= off) incl += up;
+                    int32_t uc = __shfl_up(iccl, off, 64);
+                    if (lane >= off) {
+                        incl += up;
+                        iccl += uc;
+                    }
                 }
-                if (lane == 63) sm.wave_tot[wv] = incl;
+                if (lane == 63) {
+                    sm.wave_tot[wv] = incl;
+                    sm.wave_tot[NW + wv] = iccl;
+                }
                 __syncthreads();
-                int32_t add = 0;
+                int32_t add = 0, addc = 0, totc = 0;
                 total = 0;
 #pragma unroll
                 for (int w = 0; w < NW; w++) {
-                    if (w < wv) add += sm.wave_tot[w];
+                    if (w < wv) {
+                        add += sm.wave_tot[w];
+                        addc += sm.wave_tot[NW + w];
+                    }
                     total += sm.wave_tot[w];
+                    totc += sm.wave_tot[NW + w];
                 }
                 my_off = add + incl - nloc;
+                my_cl = addc + iccl - ncl;
+                if (cl && tid == 0) cl_counts[tile] = totc;
             }
         }
         if (tid == 0) tile_counts[tile] = total;
@@ -3058,6 +3140,151 @@ __global__ void k_partition_seed(const int64_t *lens, int k,
 }
 
 
+
+// ------------------------------------------------------------ k_cl_finalize
+//
+// Compact the provisional changelog entries into dense per-tile rows.
+// Pending UPDATE_BEFORE/UPDATE_AFTER pairs (changelog-producer.
+// row-deduplicate) compare the two records' value columns (the reference's
+// RecordEqualiser over the read schema) and are dropped when equal.
+// Writes the compacted rows into cl_out (same per-tile stride) and
+// rewrites cl_counts[tile] from group count to ROW count.
+__global__ void k_cl_finalize(const DevCol *cols, const uint8_t *col_dtype,
+                              int n_cols, int first_val, int k,
+                              const uint64_t *cl_entries, uint64_t *cl_out,
+                              int32_t *cl_counts, int64_t n_tiles,
+                              int64_t tile_rows) {
+    __shared__ int32_t scan[257];
+    const int64_t stride2 = 2 * (tile_rows + PMH_MAX_RUNS);
+    for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+        const uint64_t *in = &cl_entries[tile * stride2];
+        uint64_t *out = &cl_out[tile * stride2];
+        int32_t ng = cl_counts[tile];
+        const int tid = threadIdx.x;
+        const int32_t per = (ng + (int32_t)blockDim.x - 1) / blockDim.x;
+        int32_t lo = tid * per;
+        int32_t hi = lo + per < ng ? lo + per : ng;
+        // verdict per group: 2 bits (number of surviving rows, 0/1/2)
+        uint32_t verd[8] = {0, 0, 0, 0, 0, 0, 0, 0};  // per <= 128 groups
+        int32_t nrows = 0;
+        for (int32_t g = lo; g < hi; g++) {
+            uint64_t e0 = in[2 * g], e1 = in[2 * g + 1];
+            int nv = ((e0 >> 35) & 1) + ((e1 >> 35) & 1);
+            if ((e0 >> 36) & 1) {  // pending pair: value-equality check
+                uint32_t a = (uint32_t)e0, b = (uint32_t)e1;
+                int ra = a >> PMH_ROW_BITS, rb = b >> PMH_ROW_BITS;
+                int64_t xa = a & PMH_ROW_MASK, xb = b & PMH_ROW_MASK;
+                bool eq = true;
+                for (int c = first_val; eq && c < n_cols; c++) {
+                    const DevCol &da = cols[ra * n_cols + c];
+                    const DevCol &db = cols[rb * n_cols + c];
+                    uint8_t va = da.valid0
+                                     ? ((const uint8_t *)da.valid0)[xa]
+                                     : 1;
+                    uint8_t vb = db.valid0
+                                     ? ((const uint8_t *)db.valid0)[xb]
+                                     : 1;
+                    if (va != vb) {
+                        eq = false;
+                    } else if (va) {
+                        if (da.esize == 8
+                                ? (col_load<int64_t>(da, xa) !=
+                                   col_load<int64_t>(db, xb))
+                                : (col_load<int32_t>(da, xa) !=
+                                   col_load<int32_t>(db, xb)))
+                            eq = false;
+                    }
+                }
+                if (eq) nv = 0;
+            }
+            int li = g - lo;
+            verd[li >> 4] |= (uint32_t)nv << (2 * (li & 15));
+            nrows += nv;
+        }
+        // block exclusive scan of nrows (group order preserved)
+        scan[tid + 1] = nrows;
+        if (tid == 0) scan[0] = 0;
+        __syncthreads();
+        for (int off = 1; off < (int)blockDim.x; off <<= 1) {
+            int32_t v = tid + 1 > off ? scan[tid + 1 - off] : 0;
+            __syncthreads();
+            scan[tid + 1] += v;
+            __syncthreads();
+        }
+        int32_t o = scan[tid];
+        for (int32_t g = lo; g < hi; g++) {
+            int li = g - lo;
+            int nv = (verd[li >> 4] >> (2 * (li & 15))) & 3;
+            if (nv >= 1) out[o++] = in[2 * g];
+            if (nv == 2) out[o++] = in[2 * g + 1];
+        }
+        __syncthreads();
+        if (tid == (int)blockDim.x - 1) cl_counts[tile] = o;
+        __syncthreads();
+    }
+}
+
+// ---------------------------------------------------------------- k_cl_emit
+//
+// Gather the compacted changelog rows into the changelog output columns:
+// key/seq/value columns come from the source record; the kind column is
+// the changelog RowKind carried in the entry (INSERT / UPDATE_BEFORE /
+// UPDATE_AFTER / DELETE).
+__global__ void k_cl_emit(const DevCol *cols, const uint8_t *col_dtype,
+                          const uint8_t *col_nullable, int n_cols,
+                          int kind_col, const uint64_t *cl_rows,
+                          const int32_t *cl_counts,
+                          const int64_t *cl_offsets, int64_t n_tiles,
+                          int64_t tile_rows, void *const *out_ptrs,
+                          uint8_t *const *out_valid) {
+    const int64_t stride2 = 2 * (tile_rows + PMH_MAX_RUNS);
+    for (int64_t tile = blockIdx.x; tile < n_tiles; tile += gridDim.x) {
+        const uint64_t *in = &cl_rows[tile * stride2];
+        const int32_t n = cl_counts[tile];
+        const int64_t ob = cl_offsets[tile];
+        for (int32_t i = threadIdx.x; i < n; i += blockDim.x) {
+            uint64_t e = in[i];
+            uint32_t m = (uint32_t)e;
+            int run = m >> PMH_ROW_BITS;
+            int64_t row = m & PMH_ROW_MASK;
+            int8_t kd = (int8_t)((e >> 32) & 7);
+            int64_t o = ob + i;
+            for (int c = 0; c < n_cols; c++) {
+                const DevCol &dc = cols[run * n_cols + c];
+                if (col_nullable[c] && out_valid[c])
+                    out_valid[c][o] =
+                        dc.valid0 ? ((const uint8_t *)dc.valid0)[row] : 1;
+                if (c == kind_col) {
+                    ((int8_t *)out_ptrs[c])[o] = kd;
+                    continue;
+                }
+                switch (col_dtype[c]) {
+                case 1:
+                    ((int8_t *)out_ptrs[c])[o] =
+                        (int8_t)col_load<int32_t>(dc, row);
+                    break;
+                case 2:
+                    ((int16_t *)out_ptrs[c])[o] =
+                        (int16_t)col_load<int32_t>(dc, row);
+                    break;
+                case 3:
+                case 5:
+                case 7:
+                    ((int32_t *)out_ptrs[c])[o] = col_load<int32_t>(dc, row);
+                    break;
+                case 4:
+                case 6:
+                    ((int64_t *)out_ptrs[c])[o] = col_load<int64_t>(dc, row);
+                    break;
+                default:
+                    break;
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
 // ------------------------------------------------------------ k_zstd_pages
 //
 // On-GPU zstd page decompression: one WAVEFRONT per parquet
@@ -3494,6 +3721,8 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
                                   const uint64_t *tombs,
                                   uint32_t *winners, int32_t *tile_counts,
                                   uint16_t *group_start, uint32_t *err_flag,
+                                  const uint8_t *run_levels, int max_level,
+                                  uint64_t *cl_entries, int32_t *cl_counts,
                                   hipStream_t stream) {
     int blocks = n_tiles < 4096 ? (int)n_tiles : 4096;
     const bool pu = flags & 4, fr = flags & 8;
@@ -3501,7 +3730,8 @@ hipError_t pmh_launch_merge_tiles(const DevCol *keys, const DevCol *seqs,
         hipLaunchKernelGGL(kern, dim3(blocks), dim3(PMH_TILE_THREADS), 0,
                            stream, keys, seqs, kinds, lens, k, cuts, n_tiles,
                            tile_rows, flags, tombs, winners, tile_counts,
-                           group_start, err_flag);
+                           group_start, err_flag, run_levels, max_level,
+                           cl_entries, cl_counts);
     };
     if (pu)
         launch(k_merge_tiles<true, false>);  // PU/agg never first-row
@@ -3699,6 +3929,37 @@ hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
     return hipGetLastError();
 }
 
+
+
+hipError_t pmh_launch_cl_finalize(const DevCol *cols,
+                                  const uint8_t *col_dtype, int n_cols,
+                                  int first_val, int k,
+                                  const uint64_t *cl_entries,
+                                  uint64_t *cl_out, int32_t *cl_counts,
+                                  int64_t n_tiles, int64_t tile_rows,
+                                  hipStream_t stream) {
+    int blocks = n_tiles < 4096 ? (int)n_tiles : 4096;
+    hipLaunchKernelGGL(k_cl_finalize, dim3(blocks), dim3(256), 0, stream,
+                       cols, col_dtype, n_cols, first_val, k, cl_entries,
+                       cl_out, cl_counts, n_tiles, tile_rows);
+    return hipGetLastError();
+}
+
+hipError_t pmh_launch_cl_emit(const DevCol *cols, const uint8_t *col_dtype,
+                              const uint8_t *col_nullable, int n_cols,
+                              int kind_col, const uint64_t *cl_rows,
+                              const int32_t *cl_counts,
+                              const int64_t *cl_offsets, int64_t n_tiles,
+                              int64_t tile_rows, void *const *out_ptrs,
+                              uint8_t *const *out_valid,
+                              hipStream_t stream) {
+    int blocks = n_tiles < 4096 ? (int)n_tiles : 4096;
+    hipLaunchKernelGGL(k_cl_emit, dim3(blocks), dim3(256), 0, stream, cols,
+                       col_dtype, col_nullable, n_cols, kind_col, cl_rows,
+                       cl_counts, cl_offsets, n_tiles, tile_rows, out_ptrs,
+                       out_valid);
+    return hipGetLastError();
+}
 
 hipError_t pmh_launch_zstd_pages(const uint8_t *src, const ZstdJob *jobs,
                                  int n, uint8_t *dst, uint8_t *scratch,

@@ -285,7 +285,7 @@ struct FileDesc {
 // pack files into the fewest sorted runs via a min-heap on each run's last
 // maxKey.
 static std::vector<std::vector<std::vector<FileDesc>>> interval_partition(
-    std::vector<FileDesc> files) {
+    std::vector<FileDesc> files, bool level_pure_runs = false) {
     std::sort(files.begin(), files.end(), [](const FileDesc &a, const FileDesc &b) {
         if (a.min_key != b.min_key) return a.min_key < b.min_key;
         return a.max_key < b.max_key;
@@ -295,7 +295,7 @@ static std::vector<std::vector<std::vector<FileDesc>>> interval_partition(
     int64_t bound = 0;
     bool has_bound = false;
 
-    auto pack = [](const std::vector<FileDesc> &metas) {
+    auto pack = [level_pure_runs](const std::vector<FileDesc> &metas) {
         // min-heap of runs keyed by last file's maxKey (IntervalPartition.java:93-125)
         auto cmp = [](const std::vector<FileDesc> &a, const std::vector<FileDesc> &b) {
             return a.back().max_key > b.back().max_key;  // min-heap
@@ -307,7 +307,13 @@ static std::vector<std::vector<std::vector<FileDesc>>> interval_partition(
         for (size_t i = 1; i < metas.size(); i++) {
             auto top = q.top();
             q.pop();
-            if (metas[i].min_key > top.back().max_key) {
+            if (metas[i].min_key > top.back().max_key &&
+                (!level_pure_runs ||
+                 metas[i].level == top.back().level)) {
+                // level_pure_runs (changelog mode): records carry the level
+                // of the FILE they were read from (KeyValue.setLevel per
+                // file reader), so a run must not chain files of different
+                // levels — start a new run instead (same merge semantics)
                 top.push_back(metas[i]);
             } else {
                 q.push({metas[i]});
@@ -436,6 +442,8 @@ struct Run {
     int64_t length = 0;
     std::vector<RunCol> cols;
     uint8_t *tomb = nullptr;  // deletion-vector tombstones (1 = deleted)
+    int level = 0;            // the SortedRun's level (files of one run
+                              // share it; changelog top-level detection)
 };
 
 struct Section {
@@ -468,6 +476,13 @@ struct Section {
     // bit c = column c non-null; built once per section by k_pack_valid
     std::vector<uint64_t *> row_masks;
     uint64_t **row_masks_dev = nullptr;  // [k] device array of the above
+    // full-compaction changelog chain
+    uint8_t *run_levels = nullptr;   // [k]
+    uint64_t *cl_entries = nullptr;  // 2 slots per group, provisional
+    uint64_t *cl_rows = nullptr;     // compacted rows (k_cl_finalize)
+    int32_t *cl_counts = nullptr;
+    int64_t *cl_offsets = nullptr;
+    int64_t *cl_total_dev = nullptr;
     std::vector<int64_t *> ckeys;  // per-run composite keys (k_composite)
     // batched decode work (all run-columns in ONE launch each)
     Rlev2Chunk *rlev2_all = nullptr;
@@ -524,6 +539,18 @@ struct pmh_plan_t {
     int n_useq = 0;
     int16_t *useq_dev = nullptr;
     int n_seq_groups = 0;
+    // changelog-producer = full-compaction (FullChangelogMergeFunction-
+    // Wrapper): a second output stream of changelog rows per read_next
+    bool changelog = false;
+    bool cl_row_dedup = false;
+    int max_level = 0;
+    int64_t cl_rows_last = 0;
+    std::vector<void *> out_dev2;
+    void **out_ptrs2_dev = nullptr;
+    std::vector<uint8_t *> out_valid2;
+    uint8_t **out_valid2_dev = nullptr;
+    std::vector<std::vector<uint8_t>> out_host2, out_valid_host2;
+    std::vector<pmh_col> batch_cols2;
     uint8_t *col_group_dev = nullptr;
     int16_t *sg_fields_dev = nullptr;
     uint8_t *sg_nseq_dev = nullptr;
@@ -1356,6 +1383,7 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
     run.cols.resize(cols.size());
     int64_t total_rows = 0;
     for (const auto &fd : files) total_rows += fd.row_count;
+    run.level = files.empty() ? 0 : files[0].level;
     if (total_rows >= ((int64_t)1 << PMH_ROW_BITS)) {
         // a run is the CONCATENATION of non-overlapping files; the winner
         // packing (run | row) indexes into the whole run, so the cap applies
@@ -1957,6 +1985,21 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
         if (!sec.winners || !sec.tile_counts || !sec.tile_offsets)
             return false;
     }
+    if (plan->changelog) {
+        std::vector<uint8_t> lv(k);
+        for (int r = 0; r < k; r++)
+            lv[r] = (uint8_t)sec.runs[r].level;
+        sec.run_levels = (uint8_t *)up(lv.data(), k);
+        int64_t slots2 = sec.n_tiles * 2 * (PMH_TILE_ROWS + PMH_MAX_RUNS);
+        sec.cl_entries = (uint64_t *)plan->bufs.alloc(slots2 * 8);
+        sec.cl_rows = (uint64_t *)plan->bufs.alloc(slots2 * 8);
+        sec.cl_counts = (int32_t *)plan->bufs.alloc(sec.n_tiles * 4);
+        sec.cl_offsets = (int64_t *)plan->bufs.alloc(sec.n_tiles * 8);
+        sec.cl_total_dev = (int64_t *)plan->bufs.alloc(8);
+        if (!sec.run_levels || !sec.cl_entries || !sec.cl_rows ||
+            !sec.cl_counts || !sec.cl_offsets || !sec.cl_total_dev)
+            return false;
+    }
     sec.total_dev = (int64_t *)plan->bufs.alloc(8);
     sec.err_dev = (uint32_t *)plan->bufs.alloc(4);
     if (sec.err_dev) (void)hipMemset(sec.err_dev, 0, 4);
@@ -2302,6 +2345,33 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         plan->drop_delete = j["drop_delete"].as_bool(true);
         plan->ignore_delete = j["ignore_delete"].as_bool(false);
         {
+            std::string cp = j["changelog_producer"].as_str("none");
+            if (cp == "full-compaction") {
+                plan->changelog = true;
+                plan->cl_row_dedup =
+                    j["changelog_row_deduplicate"].as_bool(false);
+                plan->max_level = (int)j["max_level"].as_i64(-1);
+                if (plan->max_level < 0) {
+                    set_error("changelog_producer=full-compaction needs "
+                              "max_level (the table's num-levels - 1; "
+                              "FullChangelogMergeFunctionWrapper maxLevel)");
+                    return nullptr;
+                }
+                if (plan->pu || plan->agg) {
+                    set_error("changelog_producer=full-compaction is "
+                              "supported for deduplicate/first-row in v1 "
+                              "(partial-update/aggregation changelog is a "
+                              "later round)");
+                    return nullptr;
+                }
+            } else if (cp != "none") {
+                set_error("changelog_producer '%s' not supported (none | "
+                          "full-compaction; lookup changelog is a later "
+                          "round)", cp.c_str());
+                return nullptr;
+            }
+        }
+        {
             // Same-box A/B (DESIGN.md §7 experiment log): the classic
             // partition->merge->scan->emit chain with the two-level
             // partition is the fastest winner-engine configuration
@@ -2312,6 +2382,15 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             const char *pf = getenv("PMH_FUSED");
             bool want_fused = pf ? pf[0] != '0' : (plan->n_useq > 0);
             plan->fused = !plan->pu && want_fused;
+            if (plan->changelog && plan->fused) {
+                if (plan->n_useq > 0 || (pf && pf[0] != '0')) {
+                    set_error("changelog_producer=full-compaction runs on "
+                              "the classic merge chain (no sequence.field / "
+                              "PMH_FUSED=1 combination in v1)");
+                    return nullptr;
+                }
+                plan->fused = false;
+            }
             if (plan->n_useq > 0 && !plan->pu && !plan->fused) {
                 set_error("sequence.field needs the fused merge path "
                           "(unset PMH_FUSED=0)");
@@ -2369,7 +2448,7 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             return nullptr;
         }
 
-        auto sections = interval_partition(files);
+        auto sections = interval_partition(files, plan->changelog);
         auto t0 = std::chrono::steady_clock::now();
         for (auto &sec_files : sections) {
             Section sec;
@@ -2452,6 +2531,42 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                       hipMemcpyHostToDevice) != hipSuccess) {
             set_error("H2D of validity descriptors failed");
             return nullptr;
+        }
+        if (plan->changelog) {
+            // changelog outputs: up to 2 rows per key group <= 2 * inputs
+            int64_t cap2 = 2 * max_rows;
+            plan->out_dev2.resize(n_cols);
+            plan->out_valid2.assign(n_cols, nullptr);
+            for (int c = 0; c < n_cols; c++) {
+                plan->out_dev2[c] =
+                    plan->bufs.alloc(cap2 * plan->cols[c].out_esize);
+                if (!plan->out_dev2[c]) {
+                    set_error("changelog output allocation failed");
+                    return nullptr;
+                }
+                if (plan->col_nullable[c]) {
+                    plan->out_valid2[c] =
+                        (uint8_t *)plan->bufs.alloc(cap2);
+                    if (!plan->out_valid2[c]) {
+                        set_error("changelog validity allocation failed");
+                        return nullptr;
+                    }
+                }
+            }
+            plan->out_ptrs2_dev =
+                (void **)plan->bufs.alloc(n_cols * sizeof(void *));
+            plan->out_valid2_dev =
+                (uint8_t **)plan->bufs.alloc(n_cols * sizeof(void *));
+            if (!plan->out_ptrs2_dev || !plan->out_valid2_dev ||
+                hipMemcpy(plan->out_ptrs2_dev, plan->out_dev2.data(),
+                          n_cols * sizeof(void *),
+                          hipMemcpyHostToDevice) != hipSuccess ||
+                hipMemcpy(plan->out_valid2_dev, plan->out_valid2.data(),
+                          n_cols * sizeof(void *),
+                          hipMemcpyHostToDevice) != hipSuccess) {
+                set_error("H2D of changelog descriptors failed");
+                return nullptr;
+            }
         }
         for (auto &cs : plan->cols) plan->col_names.push_back(cs.name);
         if (plan->agg) {
@@ -2746,10 +2861,12 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     }
     e = pmh_launch_merge_tiles(sec.key_cols, sec.seq_cols, sec.kind_cols,
                                sec.lens_dev, k, sec.cuts, sec.n_tiles,
-                               PMH_TILE_ROWS, flags, sec.tombs_dev,
-                               sec.winners,
+                               PMH_TILE_ROWS,
+                               flags | (p->cl_row_dedup ? 128 : 0),
+                               sec.tombs_dev, sec.winners,
                                sec.tile_counts, sec.group_start, sec.err_dev,
-                               st);
+                               sec.run_levels, p->max_level, sec.cl_entries,
+                               sec.cl_counts, st);
     if (e != hipSuccess) return fail("merge_tiles", e);
     (void)hipEventRecord(ev[3], st);
     e = pmh_launch_scan_tiles(sec.tile_counts, sec.n_tiles, sec.tile_offsets,
@@ -2786,13 +2903,35 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                             p->out_valid_dev, st);
     }
     if (e != hipSuccess) return fail("emit", e);
+    if (p->changelog) {
+        e = pmh_launch_cl_finalize(sec.all_cols, p->col_dtype_dev, n_cols,
+                                   p->n_key_cols + 2, k, sec.cl_entries,
+                                   sec.cl_rows, sec.cl_counts, sec.n_tiles,
+                                   PMH_TILE_ROWS, st);
+        if (e != hipSuccess) return fail("cl_finalize", e);
+        e = pmh_launch_scan_tiles(sec.cl_counts, sec.n_tiles, sec.cl_offsets,
+                                  sec.cl_total_dev, st);
+        if (e != hipSuccess) return fail("cl_scan", e);
+        e = pmh_launch_cl_emit(sec.all_cols, p->col_dtype_dev,
+                               p->col_nullable_dev, n_cols,
+                               p->n_key_cols + 1, sec.cl_rows, sec.cl_counts,
+                               sec.cl_offsets, sec.n_tiles, PMH_TILE_ROWS,
+                               p->out_ptrs2_dev, p->out_valid2_dev, st);
+        if (e != hipSuccess) return fail("cl_emit", e);
+    }
     (void)hipEventRecord(ev[5], st);
 
 collect:
     int64_t total = 0;
+    int64_t cl_total = 0;
     uint32_t err_word = 0;
     e = hipMemcpyAsync(&total, sec.total_dev, 8, hipMemcpyDeviceToHost, st);
     if (e != hipSuccess) return fail("total D2H", e);
+    if (p->changelog && sec.cl_total_dev) {
+        e = hipMemcpyAsync(&cl_total, sec.cl_total_dev, 8,
+                           hipMemcpyDeviceToHost, st);
+        if (e != hipSuccess) return fail("cl total D2H", e);
+    }
     e = hipMemcpyAsync(&err_word, sec.err_dev, 4, hipMemcpyDeviceToHost, st);
     if (e != hipSuccess) return fail("err D2H", e);
     e = hipStreamSynchronize(st);
@@ -2820,6 +2959,12 @@ collect:
     if (err_word & 4) {
         set_error("internal: fused merge lookback timed out waiting for a "
                   "predecessor tile");
+        return -1;
+    }
+    if (err_word & 8) {
+        set_error("Top level key-value already exists (two runs at "
+                  "max_level hold the same key; "
+                  "FullChangelogMergeFunctionWrapper.java:76-78 checkState)");
         return -1;
     }
 
@@ -2892,6 +3037,46 @@ collect:
             pc.dict_len = (int32_t)sd->offsets.size() - 1;
         }
     }
+    if (p->changelog) {
+        p->cl_rows_last = cl_total;
+        p->batch_cols2.resize(n_cols);
+        if (p->host_output) {
+            p->out_host2.resize(n_cols);
+            p->out_valid_host2.resize(n_cols);
+            for (int c = 0; c < n_cols; c++) {
+                p->out_host2[c].resize(cl_total * p->cols[c].out_esize);
+                if (cl_total > 0 &&
+                    hipMemcpy(p->out_host2[c].data(), p->out_dev2[c],
+                              cl_total * p->cols[c].out_esize,
+                              hipMemcpyDeviceToHost) != hipSuccess) {
+                    set_error("D2H changelog failed");
+                    return -1;
+                }
+                if (p->out_valid2[c]) {
+                    p->out_valid_host2[c].resize(cl_total);
+                    if (cl_total > 0 &&
+                        hipMemcpy(p->out_valid_host2[c].data(),
+                                  p->out_valid2[c], cl_total,
+                                  hipMemcpyDeviceToHost) != hipSuccess) {
+                        set_error("D2H changelog validity failed");
+                        return -1;
+                    }
+                }
+            }
+        }
+        for (int c = 0; c < n_cols; c++) {
+            pmh_col pc = p->batch_cols[c];  // share names/dict/decimal info
+            pc.data = p->host_output ? (const void *)p->out_host2[c].data()
+                                     : (const void *)p->out_dev2[c];
+            pc.valid = nullptr;
+            if (p->out_valid2[c])
+                pc.valid =
+                    p->host_output
+                        ? (const uint8_t *)p->out_valid_host2[c].data()
+                        : (const uint8_t *)p->out_valid2[c];
+            p->batch_cols2[c] = pc;
+        }
+    }
     if (out) {
         out->n_rows = total;
         out->n_cols = n_cols;
@@ -2899,6 +3084,23 @@ collect:
         out->cols = p->batch_cols.data();
     }
     return total;
+}
+
+// Changelog batch of the LAST pmh_read_next call (changelog_producer =
+// full-compaction). Valid until the next read_next, like the main batch.
+int64_t pmh_changelog_next(pmh_plan_t *p, pmh_batch *out) {
+    if (!p) return -1;
+    if (!p->changelog) {
+        set_error("plan has no changelog_producer configured");
+        return -1;
+    }
+    if (out) {
+        out->n_rows = p->cl_rows_last;
+        out->n_cols = (int32_t)p->cols.size();
+        out->device = p->host_output ? -1 : p->session->device;
+        out->cols = p->batch_cols2.data();
+    }
+    return p->cl_rows_last;
 }
 
 int pmh_plan_reset(pmh_plan_t *p) {

@@ -78,6 +78,9 @@ def load_lib(required=True):
     lib.pmh_plan_create.argtypes = [ctypes.c_void_p, ctypes.c_char_p]
     lib.pmh_read_next.restype = ctypes.c_int64
     lib.pmh_read_next.argtypes = [ctypes.c_void_p, ctypes.POINTER(_Batch)]
+    lib.pmh_changelog_next.restype = ctypes.c_int64
+    lib.pmh_changelog_next.argtypes = [ctypes.c_void_p,
+                                       ctypes.POINTER(_Batch)]
     lib.pmh_plan_close.argtypes = [ctypes.c_void_p]
     lib.pmh_plan_reset.argtypes = [ctypes.c_void_p]
     lib.pmh_plan_reset.restype = ctypes.c_int
@@ -275,7 +278,9 @@ class MergeReadPlan:
                  merge_engine="deduplicate", drop_delete=True,
                  ignore_delete=False, output="host", aggregations=None,
                  remove_record_on_delete=False, sequence_groups=None,
-                 ignore_retract=None, sequence_fields=None):
+                 ignore_retract=None, sequence_fields=None,
+                 changelog_producer=None, changelog_row_dedup=False,
+                 max_level=None):
         self.lib = session.lib
         desc = {
             "key_cols": key_cols,
@@ -302,6 +307,13 @@ class MergeReadPlan:
             # fields.<name>.aggregate-function (CoreOptions FIELDS_PREFIX);
             # unnamed columns default to last_non_null_value
             desc["aggregations"] = dict(aggregations)
+        if changelog_producer:
+            # changelog-producer = full-compaction
+            # (FullChangelogMergeFunctionWrapper; max_level = num-levels - 1)
+            desc["changelog_producer"] = changelog_producer
+            desc["changelog_row_deduplicate"] = bool(changelog_row_dedup)
+            desc["max_level"] = int(max_level if max_level is not None
+                                    else -1)
         self.h = self.lib.pmh_plan_create(session.h,
                                           json.dumps(desc).encode())
         if not self.h:
@@ -344,6 +356,37 @@ class MergeReadPlan:
                 out[name + "#dict"] = np.array(
                     [bb[offs[j]:offs[j + 1]] for j in range(col.dict_len)],
                     dtype=object)
+            if col.valid and b.n_rows:
+                vbuf = ctypes.cast(col.valid,
+                                   ctypes.POINTER(ctypes.c_uint8 * b.n_rows))
+                out[name + "#valid"] = np.frombuffer(
+                    bytes(vbuf.contents), dtype=np.uint8).astype(bool)
+        return out
+
+    def read_changelog(self):
+        """Changelog batch of the LAST read_next (changelog_producer =
+        "full-compaction"): dict name -> numpy array (host output), may be
+        empty. Same schema as the main batch; _VALUE_KIND carries the
+        changelog RowKind (0=+I, 1=-U, 2=+U, 3=-D)."""
+        b = _Batch()
+        n = self.lib.pmh_changelog_next(self.h, ctypes.byref(b))
+        if n < 0:
+            raise RuntimeError(f"pmh_changelog_next: {last_error()}")
+        if self.output == "device":
+            return b
+        out = {}
+        for c in range(b.n_cols):
+            col = b.cols[c]
+            dt = np.dtype(_DT_NP[col.dtype])
+            name = col.name.decode()
+            if b.n_rows and col.data:
+                buf = ctypes.cast(
+                    col.data,
+                    ctypes.POINTER(ctypes.c_uint8 * (b.n_rows * dt.itemsize)))
+                arr = np.frombuffer(bytes(buf.contents), dtype=dt)
+            else:
+                arr = np.empty(0, dtype=dt)
+            out[name] = arr
             if col.valid and b.n_rows:
                 vbuf = ctypes.cast(col.valid,
                                    ctypes.POINTER(ctypes.c_uint8 * b.n_rows))

@@ -1,0 +1,156 @@
+"""Full-compaction changelog producer on the GPU
+(FullChangelogMergeFunctionWrapper.java:74-130 through k_merge_tiles'
+changelog entries + k_cl_finalize + k_cl_emit) vs the oracle model that is
+itself pinned to the reference's test vectors (tests/test_changelog_cpu.py)."""
+import numpy as np
+import pytest
+
+from oracle import full_changelog_model
+from paimon_amd import Session, MergeReadPlan, file_descs_from_metas
+from paimon_amd.datagen import gen_runs_dedup, write_runs
+
+pytestmark = pytest.mark.gpu
+
+KEY_COLS = [{"name": "_KEY_k", "type": "int64"}]
+MAX_LEVEL = 5
+
+
+def _value_cols(n):
+    return ([{"name": "v_k", "type": "int64"}] +
+            [{"name": f"v_c{i}", "type": "int32"} for i in range(n)])
+
+
+def _read_all(plan):
+    main, cl = [], []
+    while True:
+        b = plan.read_next()
+        if b is None:
+            break
+        main.append({k: v.copy() for k, v in b.items()})
+        cl.append({k: v.copy() for k, v in plan.read_changelog().items()})
+    cat = lambda parts: {k: np.concatenate([p[k] for p in parts])
+                         for k in parts[0]} if parts else {}
+    return cat(main), cat(cl)
+
+
+def _expected(runs, levels, row_dedup=False):
+    (cr, cw, ck), (rr, rw) = full_changelog_model(
+        runs, levels, MAX_LEVEL, row_dedup=row_dedup)
+    exp_cl = {
+        "_KEY_k": np.array([runs[a]["key"][b] for a, b in zip(cr, cw)],
+                           np.int64),
+        "_SEQUENCE_NUMBER": np.array(
+            [runs[a]["seq"][b] for a, b in zip(cr, cw)], np.int64),
+        "_VALUE_KIND": ck.astype(np.int8),
+        "v_k": np.array([runs[a]["values"][0][b] for a, b in zip(cr, cw)]),
+    }
+    exp_res_keys = np.array([runs[a]["key"][b] for a, b in zip(rr, rw)],
+                            np.int64)
+    return exp_cl, exp_res_keys
+
+
+def _run_case(tmp_path, runs, levels, row_dedup=False):
+    metas = write_runs(runs, str(tmp_path), compression="NONE")
+    for m, lvl in zip(metas, levels):
+        m["level"] = lvl
+    exp_cl, exp_res_keys = _expected(runs, levels, row_dedup)
+    with Session(0) as s:
+        with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                           _value_cols(len(runs[0]["values"]) - 1),
+                           changelog_producer="full-compaction",
+                           changelog_row_dedup=row_dedup,
+                           max_level=MAX_LEVEL) as plan:
+            main, cl = _read_all(plan)
+    assert len(main["_KEY_k"]) == len(exp_res_keys)
+    assert (main["_KEY_k"] == exp_res_keys).all()
+    assert len(cl["_KEY_k"]) == len(exp_cl["_KEY_k"]), \
+        (len(cl["_KEY_k"]), len(exp_cl["_KEY_k"]))
+    for name, e in exp_cl.items():
+        g = cl[name]
+        assert (g == e).all(), \
+            f"changelog col {name}: {np.flatnonzero(g != e)[:10]}"
+
+
+def _with_top_run(n_runs, rows, seed, delete_frac=0.2, top_frac=0.6,
+                  n_value_cols=3, value_card=None):
+    """n_runs level-0 runs + one top-level run holding earlier compaction
+    results (INSERT only, lowest sequence numbers)."""
+    rng = np.random.default_rng(seed)
+    runs = gen_runs_dedup(n_runs, rows, n_value_cols=n_value_cols, seed=seed,
+                          delete_frac=delete_frac)
+    keyspace = int(max(r["key"].max() for r in runs)) + 1
+    n_top = max(1, int(rows * top_frac))
+    keys = np.sort(rng.choice(keyspace, n_top, replace=False))
+    top = {
+        "key": keys.astype(np.int64),
+        "seq": np.arange(n_top, dtype=np.int64) - n_top,  # oldest
+        "kind": np.zeros(n_top, dtype=np.int8),
+        "values": [rng.integers(0, value_card or 1 << 30, n_top
+                                ).astype(np.int64)] +
+                  [rng.integers(0, value_card or 1 << 30, n_top
+                                ).astype(np.int32)
+                   for _ in range(n_value_cols)],
+    }
+    if value_card:
+        for r in runs:
+            for c in range(len(r["values"])):
+                r["values"][c] = (r["values"][c] % value_card).astype(
+                    r["values"][c].dtype)
+    runs.append(top)
+    return runs, [0] * n_runs + [MAX_LEVEL]
+
+
+class TestFullCompactionChangelog:
+    def test_basic(self, tmp_path):
+        runs, levels = _with_top_run(4, 30_000, seed=501)
+        _run_case(tmp_path, runs, levels)
+
+    def test_no_top_level(self, tmp_path):
+        runs = gen_runs_dedup(4, 25_000, n_value_cols=3, seed=502,
+                              delete_frac=0.15)
+        _run_case(tmp_path, runs, [0, 0, 1, 1])
+
+    def test_row_dedup_suppression(self, tmp_path):
+        # tiny value cardinality: many top/merged pairs compare equal
+        runs, levels = _with_top_run(3, 20_000, seed=503, delete_frac=0.1,
+                                     value_card=2)
+        _run_case(tmp_path, runs, levels, row_dedup=True)
+        _run_case(tmp_path, runs, levels, row_dedup=False)
+
+    def test_delete_heavy(self, tmp_path):
+        runs, levels = _with_top_run(4, 20_000, seed=504, delete_frac=0.6)
+        _run_case(tmp_path, runs, levels)
+
+    def test_sixteen_runs(self, tmp_path):
+        runs, levels = _with_top_run(15, 8_000, seed=505)
+        _run_case(tmp_path, runs, levels)
+
+    def test_duplicate_top_rejected(self, tmp_path):
+        runs, levels = _with_top_run(2, 5_000, seed=506)
+        # second copy of the top run at max level -> same key in two
+        # max-level runs -> checkState error
+        runs.append({k: (v.copy() if not isinstance(v, list) else
+                         [x.copy() for x in v])
+                     for k, v in runs[-1].items()})
+        runs[-1]["seq"] = runs[-1]["seq"] - 100_000
+        levels = levels + [MAX_LEVEL]
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        for m, lvl in zip(metas, levels):
+            m["level"] = lvl
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(len(runs[0]["values"]) - 1),
+                               changelog_producer="full-compaction",
+                               max_level=MAX_LEVEL) as plan:
+                with pytest.raises(RuntimeError, match="Top level"):
+                    while plan.read_next() is not None:
+                        pass
+
+    def test_missing_max_level_rejected(self, tmp_path):
+        runs, levels = _with_top_run(2, 2_000, seed=507)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="max_level"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(len(runs[0]["values"]) - 1),
+                              changelog_producer="full-compaction")
