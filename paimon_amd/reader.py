@@ -280,7 +280,7 @@ class MergeReadPlan:
                  remove_record_on_delete=False, sequence_groups=None,
                  ignore_retract=None, sequence_fields=None,
                  changelog_producer=None, changelog_row_dedup=False,
-                 max_level=None):
+                 max_level=None, sort_engine="loser-tree"):
         self.lib = session.lib
         desc = {
             "key_cols": key_cols,
@@ -290,6 +290,7 @@ class MergeReadPlan:
             "ignore_delete": ignore_delete,
             "remove_record_on_delete": remove_record_on_delete,
             "output": output,
+            "sort_engine": sort_engine,
             "files": files,
         }
         if sequence_groups:

@@ -683,3 +683,36 @@ class TestFusedPathAB:
                 while plan.read_next() is not None:
                     pass
                 assert plan.stats().get("path_mode", 0) == 0
+
+
+class TestSortEngineOption:
+    """sort-engine = min-heap (SortMergeReader.java:41-57): both host
+    algorithms have identical merge semantics; the GPU merge-path replaces
+    them — the option is accepted and results are identical."""
+
+    def test_min_heap_identical(self, tmp_path):
+        runs = gen_runs_dedup(4, 40_000, n_value_cols=3, seed=601,
+                              delete_frac=0.2)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        outs = []
+        for se in ("loser-tree", "min-heap"):
+            with Session(0) as s:
+                with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                                   _value_cols(3), sort_engine=se) as plan:
+                    ks = []
+                    while True:
+                        b = plan.read_next()
+                        if b is None:
+                            break
+                        ks.append(np.concatenate(
+                            [b["_KEY_k"].copy(), b["_SEQUENCE_NUMBER"].copy()]))
+                    outs.append(np.concatenate(ks))
+        assert (outs[0] == outs[1]).all()
+
+    def test_unknown_engine_rejected(self, tmp_path):
+        runs = gen_runs_dedup(1, 1_000, n_value_cols=1, seed=602)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="sort_engine"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(1), sort_engine="quick-sort")
