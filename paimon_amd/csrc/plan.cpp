@@ -1631,12 +1631,6 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     vpos[pi] = pos;
                 }
                 if (has_delta) {
-                    if (chunk_nulls) {
-                        set_error("%s col %s: nullable DELTA columns are a "
-                                  "later round", fd.path.c_str(),
-                                  cols[c].name.c_str());
-                        return false;
-                    }
                     if (cols[c].dtype == PMH_DT_STRING) {
                         set_error("%s col %s: DELTA byte arrays are a later "
                                   "round", fd.path.c_str(),
@@ -1661,6 +1655,7 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                         return false;
                     }
                     plan->encoded_bytes_total += payload_len;
+                    if (chunk_nulls) rc.has_nulls = true;
                     for (size_t pi = 0; pi < cc.pages.size(); pi++) {
                         auto &pg = cc.pages[pi];
                         if (pg.page_type != 0) continue;
@@ -1670,6 +1665,35 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                                             ? pg.uncompressed_size
                                             : pg.compressed_size);
                         std::string derr;
+                        if (chunk_nulls && max_def > 0) {
+                            // nullable DELTA: the stream encodes only the
+                            // NON-NULL values — decode them to the dense
+                            // buffer (out_addr 0 patched to dense_dev at
+                            // finalize) and let k_level_scatter position
+                            // them like dense PLAIN/dictionary values
+                            uint32_t dl_len;
+                            memcpy(&dl_len, pp, 4);
+                            int64_t rel = (int64_t)rc.levels_host.size();
+                            rc.levels_host.insert(rc.levels_host.end(),
+                                                  pp + 4, pp + 4 + dl_len);
+                            int64_t before = rc.dense_before;
+                            if (!prescan_def(pp + 4, dl_len, pg.num_values,
+                                             chunk_row0 + pg.first_row, rel,
+                                             &rc.dense_before, rc.def_host))
+                                return false;
+                            int64_t nvalid = rc.dense_before - before;
+                            if (!prescan_delta(pp + pos, plen - pos, nvalid,
+                                               before, (uint64_t)dev,
+                                               ppo[pi] + pos, 0ull, stored,
+                                               rc.delta_host,
+                                               rc.dstreams_host, derr)) {
+                                set_error("%s col %s: %s", fd.path.c_str(),
+                                          cols[c].name.c_str(),
+                                          derr.c_str());
+                                return false;
+                            }
+                            continue;
+                        }
                         if (!prescan_delta(pp + pos, plen - pos,
                                            pg.num_values,
                                            chunk_row0 + pg.first_row,
@@ -1901,6 +1925,12 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                 dc.dense_addr = (uint64_t)rc.dense_dev;
                 dc.esize = plan->cols[c].stored_esize;
             }
+            // nullable DELTA pages decode into the dense buffer
+            // (k_delta_emit runs before k_level_scatter)
+            for (auto &ch : rc.delta_host)
+                if (!ch.out_addr) ch.out_addr = (uint64_t)rc.dense_dev;
+            for (auto &ds : rc.dstreams_host)
+                if (!ds.out_addr) ds.out_addr = (uint64_t)rc.dense_dev;
             plan->encoded_bytes_total += rc.levels_host.size();
             rc.dense_host.clear();
             rc.dense_host.shrink_to_fit();
