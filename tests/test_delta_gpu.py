@@ -101,7 +101,9 @@ class TestDeltaBinaryPacked:
                               .astype(np.int64))
         _run(tmp_path, runs, ["v_c0"], vtype="int64")
 
-    def test_delta_nullable_rejected(self, tmp_path):
+    def test_delta_nullable_single_run(self, tmp_path):
+        # nullable DELTA columns are supported (dense decode + def scatter);
+        # the richer multi-run cases live in TestNullableDelta below
         n = 1000
         tbl = pa.table({
             "_KEY_k": pa.array(np.arange(n, dtype=np.int64)),
@@ -116,9 +118,15 @@ class TestDeltaBinaryPacked:
         metas = [{"path": path, "rowCount": n, "minKey": 0,
                   "maxKey": n - 1, "level": 0}]
         with Session(0) as s:
-            with pytest.raises(RuntimeError, match="DELTA"):
-                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
-                              [{"name": "v_c0", "type": "int32"}])
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               [{"name": "v_c0", "type": "int32"}]) as plan:
+                b = plan.read_next()
+                assert len(b["_KEY_k"]) == n
+                valid = b["v_c0#valid"]
+                exp_null = np.arange(n) % 7 == 0
+                assert (valid == ~exp_null).all()
+                assert (b["v_c0"][valid] ==
+                        np.arange(n, dtype=np.int32)[valid]).all()
 
 
 class TestNullableDelta:
