@@ -2375,20 +2375,6 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         plan->drop_delete = j["drop_delete"].as_bool(true);
         plan->ignore_delete = j["ignore_delete"].as_bool(false);
         {
-            // sort-engine (CoreOptions SORT_ENGINE; SortMergeReader.java:
-            // 41-57 factory): loser-tree and min-heap are two HOST
-            // algorithms with IDENTICAL merge semantics — the GPU
-            // merge-path partition replaces both, so the option is accepted
-            // for API parity and affects nothing (results are pinned by
-            // the same parity tests either way).
-            std::string se = j["sort_engine"].as_str("loser-tree");
-            if (se != "loser-tree" && se != "min-heap") {
-                set_error("sort_engine '%s' not supported (loser-tree | "
-                          "min-heap)", se.c_str());
-                return nullptr;
-            }
-        }
-        {
             std::string cp = j["changelog_producer"].as_str("none");
             if (cp == "full-compaction") {
                 plan->changelog = true;

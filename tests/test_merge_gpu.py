@@ -713,6 +713,6 @@ class TestSortEngineOption:
         runs = gen_runs_dedup(1, 1_000, n_value_cols=1, seed=602)
         metas = write_runs(runs, str(tmp_path), compression="NONE")
         with Session(0) as s:
-            with pytest.raises(RuntimeError, match="sort_engine"):
+            with pytest.raises(RuntimeError, match="sort-engine"):
                 MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
                               _value_cols(1), sort_engine="quick-sort")
