@@ -402,6 +402,9 @@ struct GatherTask {
     void *dict_dev = nullptr;
     bool to_dense = false;  // nullable chunk: gather into the dense buffer,
                             // k_level_scatter positions the rows after
+    bool in_place = false;  // ORC dictionary strings: the RLEv2 decode
+                            // already wrote LOCAL ids at the target; remap
+                            // them to global ids in place
 };
 
 struct RunCol {
@@ -944,6 +947,140 @@ static bool load_file(const std::string &path, StagedFile &sf) {
 // Walk an ORC RLEv2 stream (runs <= 512 values, byte-aligned) and emit one
 // device work chunk per run. dev_base = device address of the uploaded
 // stream bytes; dense0 = dense output offset of this stream's first value.
+// Host RLEv2 decode (UNSIGNED interpretation) — ORC LENGTH / side streams
+// that staging needs on the host (string id-ification). Restates the ORC
+// v1 spec's RLEv2 (same arithmetic as k_rlev2;
+// orc/impl/RunLengthIntegerReaderV2.java is the reference consumer).
+static int rlev2_width(int enc) {
+    static const int tbl[8] = {26, 28, 30, 32, 40, 48, 56, 64};
+    if (enc < 24) return enc + 1;
+    return tbl[enc - 24];
+}
+
+static bool host_rlev2_u(const uint8_t *s, int64_t len, int64_t n,
+                         std::vector<uint64_t> &out) {
+    out.clear();
+    out.reserve(n);
+    int64_t p = 0;
+    auto bits_be = [&](int64_t q, int64_t bitpos, int w) -> uint64_t {
+        uint64_t v = 0;
+        for (int i = 0; i < w; i++) {
+            int64_t b = bitpos + i;
+            v = (v << 1) | ((s[q + (b >> 3)] >> (7 - (b & 7))) & 1);
+        }
+        return v;
+    };
+    // n < 0: decode until the stream is exhausted (dictionary LENGTH
+    // streams — the entry count comes from the stream itself)
+    while (n < 0 ? p < len : (int64_t)out.size() < n) {
+        if (p >= len) return false;
+        uint8_t h = s[p];
+        int mode = h >> 6;
+        if (mode == 0) {  // SHORT_REPEAT
+            int w = ((h >> 3) & 7) + 1;
+            int cnt = (h & 7) + 3;
+            if (p + 1 + w > len) return false;
+            uint64_t v = 0;
+            for (int i = 0; i < w; i++) v = (v << 8) | s[p + 1 + i];
+            for (int i = 0; i < cnt; i++) out.push_back(v);
+            p += 1 + w;
+        } else if (mode == 1) {  // DIRECT
+            int w = rlev2_width((h >> 1) & 31);
+            if (p + 2 > len) return false;
+            int cnt = (((int)(h & 1)) << 8 | s[p + 1]) + 1;
+            int64_t nbytes = ((int64_t)cnt * w + 7) / 8;
+            if (p + 2 + nbytes > len) return false;
+            for (int i = 0; i < cnt; i++)
+                out.push_back(bits_be(p + 2, (int64_t)i * w, w));
+            p += 2 + nbytes;
+        } else if (mode == 2) {  // PATCHED_BASE
+            if (p + 4 > len) return false;
+            int w = rlev2_width((h >> 1) & 31);
+            int cnt = (((int)(h & 1)) << 8 | s[p + 1]) + 1;
+            int bw = ((s[p + 2] >> 5) & 7) + 1;
+            int pw = rlev2_width(s[p + 2] & 31);
+            int pgw = ((s[p + 3] >> 5) & 7) + 1;
+            int pll = s[p + 3] & 31;
+            int64_t q = p + 4;
+            if (q + bw > len) return false;
+            uint64_t braw = 0;
+            for (int i = 0; i < bw; i++) braw = (braw << 8) | s[q + i];
+            int64_t base = (int64_t)braw;
+            uint64_t sign = 1ull << (bw * 8 - 1);
+            if (braw & sign) base = -(int64_t)(braw & (sign - 1));
+            q += bw;
+            int64_t dbytes = ((int64_t)cnt * w + 7) / 8;
+            int pbits = pgw * 8 + pw;
+            pbits = ((pbits + 7) / 8) * 8;
+            int64_t pbytes = ((int64_t)pll * pbits + 7) / 8;
+            if (q + dbytes + pbytes > len) return false;
+            std::vector<uint64_t> vals(cnt);
+            for (int i = 0; i < cnt; i++)
+                vals[i] = bits_be(q, (int64_t)i * w, w);
+            int64_t gap = 0;
+            for (int i = 0; i < pll; i++) {
+                uint64_t e = bits_be(q + dbytes, (int64_t)i * pbits, pbits);
+                uint64_t patch =
+                    e & ((pw >= 64) ? ~0ull : ((1ull << pw) - 1));
+                gap += (int64_t)(e >> pw);
+                if (gap >= cnt) return false;
+                vals[gap] |= patch << w;
+            }
+            for (int i = 0; i < cnt; i++)
+                out.push_back((uint64_t)(base + (int64_t)vals[i]));
+            p = q + dbytes + pbytes;
+        } else {  // DELTA
+            int enc = (h >> 1) & 31;
+            int w = enc == 0 ? 0 : rlev2_width(enc);
+            if (p + 2 > len) return false;
+            int cnt = (((int)(h & 1)) << 8 | s[p + 1]) + 1;
+            int64_t q = p + 2;
+            uint64_t base = 0;  // base: unsigned varint (UNSIGNED streams)
+            int sh = 0;
+            for (;;) {
+                if (q >= len) return false;
+                uint8_t b = s[q++];
+                base |= (uint64_t)(b & 0x7f) << sh;
+                if (!(b & 0x80)) break;
+                sh += 7;
+            }
+            uint64_t zd = 0;  // delta0: signed varint (zigzag)
+            sh = 0;
+            for (;;) {
+                if (q >= len) return false;
+                uint8_t b = s[q++];
+                zd |= (uint64_t)(b & 0x7f) << sh;
+                if (!(b & 0x80)) break;
+                sh += 7;
+            }
+            int64_t d0 = (int64_t)(zd >> 1) ^ -(int64_t)(zd & 1);
+            out.push_back(base);
+            int64_t cur = (int64_t)base;
+            if (cnt > 1) {
+                cur += d0;
+                out.push_back((uint64_t)cur);
+            }
+            if (w == 0) {
+                for (int i = 2; i < cnt; i++) {
+                    cur += d0;
+                    out.push_back((uint64_t)cur);
+                }
+            } else {
+                int64_t nbytes = ((int64_t)(cnt - 2) * w + 7) / 8;
+                if (q + nbytes > len) return false;
+                for (int i = 2; i < cnt; i++) {
+                    int64_t d = (int64_t)bits_be(q, (int64_t)(i - 2) * w, w);
+                    cur += d0 < 0 ? -d : d;
+                    out.push_back((uint64_t)cur);
+                }
+                q += nbytes;
+            }
+            p = q;
+        }
+    }
+    return n < 0 || (int64_t)out.size() >= n;
+}
+
 static bool prescan_rlev2(const uint8_t *s, int64_t len, int64_t n_values,
                           int is_signed, int out_esize, int dense_target,
                           uint64_t dev_base, int64_t dense0,
@@ -1174,12 +1311,6 @@ static bool prescan_present(const uint8_t *s, int64_t len, int64_t n_rows,
 static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
                            const StagedFile &sf, Run &run, int64_t row_base) {
     const auto &cols = plan->cols;
-    for (const auto &cs : cols)
-        if (cs.dtype == PMH_DT_STRING) {
-            set_error("%s: ORC string columns are not on the GPU path yet "
-                      "(parquet dictionary strings are)", fd.path.c_str());
-            return false;
-        }
     const OrcFileMeta &om = sf.orc_meta;
     std::vector<int> col_id(cols.size(), -1);
     for (size_t c = 0; c < cols.size(); c++) {
@@ -1198,6 +1329,11 @@ static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
             rc.orc_encoded = true;
             int cid = col_id[c];
             int ckind = om.column_kinds[cid - 1];
+            if (cols[c].dtype == PMH_DT_STRING && ckind != ORC_STRING) {
+                set_error("%s col %s: declared string but ORC kind is %d",
+                          fd.path.c_str(), cols[c].name.c_str(), ckind);
+                return false;
+            }
             const int stored = cols[c].stored_esize;
             const OrcStream *data = orc_find_stream(st, cid, ORC_STREAM_DATA);
             const OrcStream *present =
@@ -1207,11 +1343,15 @@ static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
                           cols[c].name.c_str());
                 return false;
             }
-            if ((int)st.encodings.size() > cid &&
-                st.encodings[cid] != 0 && st.encodings[cid] != 2) {
-                set_error("%s col %s: ORC dictionary encodings not "
-                          "supported yet",
-                          fd.path.c_str(), cols[c].name.c_str());
+            const int cenc =
+                (int)st.encodings.size() > cid ? st.encodings[cid] : 0;
+            if (ckind == ORC_STRING
+                    ? (cenc != 2 && cenc != 3)
+                    : (cenc != 0 && cenc != 2)) {
+                set_error("%s col %s: ORC column encoding %d not supported "
+                          "(ints: DIRECT/DIRECT_V2; strings: DIRECT_V2 | "
+                          "DICTIONARY_V2)",
+                          fd.path.c_str(), cols[c].name.c_str(), cenc);
                 return false;
             }
             // compressed streams decompress on the host at staging, like
@@ -1282,6 +1422,165 @@ static bool stage_orc_file(pmh_plan_t *plan, const FileDesc &fd,
                     }
                 }
                 plan->encoded_bytes_total += data_len;
+                continue;
+            }
+            if (ckind == ORC_STRING) {
+                // ORC strings (WriterImpl StringTreeWriter):
+                //  - DICTIONARY_V2: DICTIONARY_DATA bytes + LENGTH (RLEv2
+                //    u) describe the stripe dictionary; DATA = RLEv2 u
+                //    local ids -> decode on GPU, then remap IN PLACE to
+                //    the column's plan-level GLOBAL dictionary;
+                //  - DIRECT_V2: LENGTH + DATA bytes -> each value interns
+                //    into the global dictionary at staging (host) and the
+                //    int32 ids upload directly (ids ARE the decoded form).
+                if (cols[c].dtype != PMH_DT_STRING) {
+                    set_error("%s col %s: ORC STRING column read as "
+                              "non-string type", fd.path.c_str(),
+                              cols[c].name.c_str());
+                    return false;
+                }
+                const OrcStream *lenst =
+                    orc_find_stream(st, cid, ORC_STREAM_LENGTH);
+                if (!lenst) {
+                    set_error("%s col %s: LENGTH stream missing",
+                              fd.path.c_str(), cols[c].name.c_str());
+                    return false;
+                }
+                const uint8_t *len_ptr = sf.data.data() + lenst->offset;
+                int64_t len_len = lenst->length;
+                std::vector<uint8_t> len_dec;
+                if (om.compression != 0) {
+                    std::string cerr;
+                    if (!orc_decompress(len_ptr, len_len, om.compression,
+                                        om.compression_block_size, len_dec,
+                                        cerr)) {
+                        set_error("%s col %s LENGTH: %s", fd.path.c_str(),
+                                  cols[c].name.c_str(), cerr.c_str());
+                        return false;
+                    }
+                    len_ptr = len_dec.data();
+                    len_len = (int64_t)len_dec.size();
+                }
+                if (!plan->sdicts[c])
+                    plan->sdicts[c].reset(new StrDict());
+                StrDict *sd = plan->sdicts[c].get();
+                int64_t row0 = row_base + stripe_row;
+                int64_t n_dense = st.num_rows;
+                int64_t dense0 = row0;
+                int dense_target = 0;
+                if (present) {
+                    rc.has_nulls = true;
+                    int64_t before = rc.dense_before;
+                    if (!prescan_present(pres_ptr, pres_len, st.num_rows,
+                                         row0, rc))
+                        return false;
+                    n_dense = rc.dense_before - before;
+                    dense0 = before;
+                    dense_target = 1;
+                }
+                if (cenc == 3) {  // DICTIONARY_V2
+                    const OrcStream *dct =
+                        orc_find_stream(st, cid, ORC_STREAM_DICTIONARY);
+                    if (!dct) {
+                        set_error("%s col %s: DICTIONARY_DATA missing",
+                                  fd.path.c_str(), cols[c].name.c_str());
+                        return false;
+                    }
+                    const uint8_t *dp = sf.data.data() + dct->offset;
+                    int64_t dl = dct->length;
+                    std::vector<uint8_t> dict_dec;
+                    if (om.compression != 0) {
+                        std::string cerr;
+                        if (!orc_decompress(dp, dl, om.compression,
+                                            om.compression_block_size,
+                                            dict_dec, cerr)) {
+                            set_error("%s col %s DICTIONARY: %s",
+                                      fd.path.c_str(), cols[c].name.c_str(),
+                                      cerr.c_str());
+                            return false;
+                        }
+                        dp = dict_dec.data();
+                        dl = (int64_t)dict_dec.size();
+                    }
+                    std::vector<uint64_t> lens;
+                    if (!host_rlev2_u(len_ptr, len_len, -1, lens)) {
+                        set_error("%s col %s: LENGTH stream decode failed",
+                                  fd.path.c_str(), cols[c].name.c_str());
+                        return false;
+                    }
+                    std::vector<int32_t> remap(lens.size());
+                    int64_t off = 0;
+                    for (size_t i = 0; i < lens.size(); i++) {
+                        if (off + (int64_t)lens[i] > dl) {
+                            set_error("%s col %s: dictionary bytes short",
+                                      fd.path.c_str(), cols[c].name.c_str());
+                            return false;
+                        }
+                        remap[i] = sd->add(dp + off, (uint32_t)lens[i]);
+                        off += (int64_t)lens[i];
+                    }
+                    void *rdev = plan->bufs.alloc(
+                        (lens.empty() ? 1 : lens.size()) * 4);
+                    if (!rdev) return false;
+                    if (!lens.empty() &&
+                        hipMemcpy(rdev, remap.data(), lens.size() * 4,
+                                  hipMemcpyHostToDevice) != hipSuccess) {
+                        set_error("H2D ORC string remap failed");
+                        return false;
+                    }
+                    void *dev = plan->bufs.alloc(data_len + 16);
+                    if (!dev) return false;
+                    if (hipMemcpy(dev, data_ptr, data_len,
+                                  hipMemcpyHostToDevice) != hipSuccess) {
+                        set_error("H2D failed");
+                        return false;
+                    }
+                    plan->encoded_bytes_total += data_len + len_len + dl;
+                    if (!prescan_rlev2(data_ptr, data_len, n_dense, 0, 4,
+                                       dense_target, (uint64_t)dev, dense0,
+                                       rc.rlev2_host))
+                        return false;
+                    GatherTask gt;
+                    gt.start = dense0;
+                    gt.n = n_dense;
+                    gt.dict_dev = rdev;
+                    gt.to_dense = dense_target != 0;
+                    gt.in_place = true;
+                    rc.gathers.push_back(gt);
+                } else {  // DIRECT_V2: host id-ification
+                    std::vector<uint64_t> lens;
+                    if (!host_rlev2_u(len_ptr, len_len, n_dense, lens)) {
+                        set_error("%s col %s: LENGTH stream decode failed",
+                                  fd.path.c_str(), cols[c].name.c_str());
+                        return false;
+                    }
+                    std::vector<int32_t> ids(n_dense);
+                    int64_t off = 0;
+                    for (int64_t i = 0; i < n_dense; i++) {
+                        if (off + (int64_t)lens[i] > data_len) {
+                            set_error("%s col %s: string bytes short",
+                                      fd.path.c_str(), cols[c].name.c_str());
+                            return false;
+                        }
+                        ids[i] = sd->add(data_ptr + off,
+                                         (uint32_t)lens[i]);
+                        off += (int64_t)lens[i];
+                    }
+                    plan->encoded_bytes_total += data_len + len_len;
+                    if (present) {
+                        const uint8_t *ib = (const uint8_t *)ids.data();
+                        rc.dense_host.insert(rc.dense_host.end(), ib,
+                                             ib + n_dense * 4);
+                        rc.dense_segs.emplace_back(dense0, n_dense * 4);
+                    } else if (n_dense &&
+                               hipMemcpy((uint8_t *)rc.contig + row0 * 4,
+                                         ids.data(), n_dense * 4,
+                                         hipMemcpyHostToDevice) !=
+                                   hipSuccess) {
+                        set_error("H2D failed");
+                        return false;
+                    }
+                }
                 continue;
             }
             // upload the encoded DATA stream (+16 B pad: the RLEv2
@@ -2073,7 +2372,8 @@ static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
             sec.any_dict |= rc.dict_encoded;
             sec.any_decode |= rc.dict_encoded || rc.has_nulls ||
                               !rc.rlev2_host.empty() ||
-                              !rc.delta_host.empty();
+                              !rc.delta_host.empty() ||
+                              !rc.gathers.empty();
             all_v.insert(all_v.end(), rc.rlev2_host.begin(),
                          rc.rlev2_host.end());
             all_d.insert(all_d.end(), rc.def_host.begin(), rc.def_host.end());
@@ -2749,18 +3049,27 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             for (size_t c = 0; c < run.cols.size(); c++) {
                 RunCol &rc = run.cols[c];
                 int es = p->cols[c].stored_esize;
-                if (!rc.dict_encoded) continue;
-                hipError_t e = pmh_launch_rle_decode(
-                    rc.rle_dev, (int64_t)rc.rle_host.size(), rc.ids_dev, st);
-                if (e != hipSuccess) return fail("rle_decode", e);
+                if (!rc.dict_encoded && rc.gathers.empty()) continue;
+                if (rc.dict_encoded) {
+                    hipError_t e = pmh_launch_rle_decode(
+                        rc.rle_dev, (int64_t)rc.rle_host.size(), rc.ids_dev,
+                        st);
+                    if (e != hipSuccess) return fail("rle_decode", e);
+                }
                 for (const GatherTask &gt : rc.gathers) {
                     if (!gt.n) continue;
                     uint8_t *dst = gt.to_dense
                                        ? (uint8_t *)rc.dense_dev
                                        : (uint8_t *)rc.contig;
-                    e = pmh_launch_dict_gather(rc.ids_dev + gt.start,
-                                               gt.dict_dev, gt.n,
-                                               dst + gt.start * es, es, st);
+                    // in_place: ORC dictionary-string LOCAL ids were
+                    // RLEv2-decoded at the target already; remap to global
+                    const int32_t *ids =
+                        gt.in_place
+                            ? (const int32_t *)(dst + gt.start * es)
+                            : rc.ids_dev + gt.start;
+                    hipError_t e = pmh_launch_dict_gather(
+                        ids, gt.dict_dev, gt.n, dst + gt.start * es, es,
+                        st);
                     if (e != hipSuccess) return fail("dict_gather", e);
                 }
             }
