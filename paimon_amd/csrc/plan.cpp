@@ -960,7 +960,7 @@ static int rlev2_width(int enc) {
 static bool host_rlev2_u(const uint8_t *s, int64_t len, int64_t n,
                          std::vector<uint64_t> &out) {
     out.clear();
-    out.reserve(n);
+    if (n > 0) out.reserve(n);
     int64_t p = 0;
     auto bits_be = [&](int64_t q, int64_t bitpos, int w) -> uint64_t {
         uint64_t v = 0;
