@@ -480,6 +480,20 @@ struct Section {
     std::vector<uint64_t *> row_masks;
     uint64_t **row_masks_dev = nullptr;  // [k] device array of the above
     // full-compaction changelog chain
+    // hierarchical sections (> PMH_MAX_RUNS runs): the reference spills
+    // to disk (MergeSorter.spillMergeSort); here the winner fold is
+    // ASSOCIATIVE for deduplicate/first-row (max/min by (seq, isAdd)), so
+    // run batches merge on-device into VIRTUAL runs first (keeping delete
+    // winners), then the normal chain merges the virtual runs.
+    bool hier = false;
+    DevCol *rkeys = nullptr, *rseqs = nullptr, *rkinds = nullptr,
+           *rall = nullptr;          // descriptors over the REAL runs
+    int64_t *rlens = nullptr;
+    uint64_t *rtombs = nullptr;
+    std::vector<int> blo, bhi;       // batch run ranges
+    std::vector<int64_t> brows, bntiles;
+    std::vector<void **> bout_ptrs;      // [n_cols] device arrays per batch
+    std::vector<uint8_t **> bout_valid;
     uint8_t *run_levels = nullptr;   // [k]
     uint64_t *cl_entries = nullptr;  // 2 slots per group, provisional
     uint64_t *cl_rows = nullptr;     // compacted rows (k_cl_finalize)
@@ -525,6 +539,9 @@ struct pmh_plan_t {
     bool fsplit = false;     // split value emission (PMH_FSPLIT=1 A/B)
     bool agg = false;        // aggregation merge engine (uses PU member lists)
     uint8_t *col_agg_dev = nullptr;  // per-column PMH_AGG_* codes
+    uint8_t *hier_dtype_dev = nullptr;  // batch-pass dtype map (int8/16 ->
+                                        // int32 so virtual runs hold the
+                                        // STORED width)
     // composite key (>1 key column): order-preserving packed comparand
     bool composite_key = false;
     uint64_t key_shifts = 0, key_bits = 0;  // 8 bits per sub-key
@@ -2251,6 +2268,130 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
     return true;
 }
 
+// Hierarchical section setup: group real runs into batches (<= PMH_MAX_RUNS
+// runs AND < 2^PMH_ROW_BITS input rows each, so every virtual run respects
+// the winner row packing), allocate the virtual runs' columns, and swap the
+// section's chain descriptors to point at them. The batch pass runs in
+// pmh_read_next (it needs decoded columns).
+static bool build_hier_section(pmh_plan_t *plan, Section &sec);
+
+static bool build_hier_section(pmh_plan_t *plan, Section &sec) {
+    const int n_real = (int)sec.runs.size();
+    const int n_cols = (int)plan->cols.size();
+    // keep the real-run descriptors for the batch pass
+    sec.rkeys = sec.key_cols;
+    sec.rseqs = sec.seq_cols;
+    sec.rkinds = sec.kind_cols;
+    sec.rall = sec.all_cols;
+    sec.rlens = sec.lens_dev;
+    sec.rtombs = sec.tombs_dev;
+    sec.tombs_dev = nullptr;  // tombstones are consumed by the batch pass
+    // batches: <= PMH_MAX_RUNS runs and < 2^PMH_ROW_BITS input rows each
+    {
+        int lo = 0;
+        while (lo < n_real) {
+            int hi = lo;
+            int64_t rows = 0;
+            while (hi < n_real && hi - lo < PMH_MAX_RUNS &&
+                   rows + sec.runs[hi].length < ((int64_t)1 << PMH_ROW_BITS))
+                rows += sec.runs[hi++].length;
+            if (hi == lo) {
+                set_error("run of %lld rows cannot batch under the 2^%d "
+                          "winner row limit",
+                          (long long)sec.runs[lo].length, PMH_ROW_BITS);
+                return false;
+            }
+            sec.blo.push_back(lo);
+            sec.bhi.push_back(hi);
+            sec.brows.push_back(rows);
+            sec.bntiles.push_back((rows + PMH_TILE_ROWS - 1) / PMH_TILE_ROWS);
+            lo = hi;
+        }
+        if ((int)sec.blo.size() > PMH_MAX_RUNS) {
+            set_error("section needs %zu virtual runs > %d (two-level "
+                      "hierarchy is a later round)",
+                      sec.blo.size(), PMH_MAX_RUNS);
+            return false;
+        }
+    }
+    const int nb = (int)sec.blo.size();
+    // batch-pass dtype map: int8/int16 emit as int32 so the virtual runs
+    // hold the STORED width the second merge/emit pass reads
+    if (!plan->hier_dtype_dev) {
+        std::vector<uint8_t> hd(n_cols);
+        for (int c = 0; c < n_cols; c++) {
+            uint8_t d = (uint8_t)plan->cols[c].dtype;
+            hd[c] = (d == 1 || d == 2) ? 3 : d;
+        }
+        plan->hier_dtype_dev = (uint8_t *)plan->bufs.alloc(n_cols);
+        if (!plan->hier_dtype_dev ||
+            hipMemcpy(plan->hier_dtype_dev, hd.data(), n_cols,
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_error("H2D hier dtype map failed");
+            return false;
+        }
+    }
+    // virtual runs: one per batch; columns at stored width
+    std::vector<DevCol> vkeys(nb), vseqs(nb), vkinds(nb), vall(nb * n_cols);
+    int seq_idx = plan->n_key_cols;
+    int kind_idx = plan->n_key_cols + 1;
+    for (int b = 0; b < nb; b++) {
+        std::vector<void *> outs(n_cols);
+        std::vector<uint8_t *> valids(n_cols, nullptr);
+        for (int c = 0; c < n_cols; c++) {
+            int es = plan->cols[c].stored_esize;
+            outs[c] = plan->bufs.alloc((size_t)sec.brows[b] * es);
+            if (!outs[c]) return false;
+            // nullability decided per plan at create end; allocate byte
+            // validity for every value column that CAN be null in any run
+            bool nullable = false;
+            for (auto &run : sec.runs)
+                if (run.cols[c].has_nulls) nullable = true;
+            if (nullable) {
+                valids[c] = (uint8_t *)plan->bufs.alloc(sec.brows[b]);
+                if (!valids[c]) return false;
+            }
+            DevCol dc{(uint64_t)outs[c], (uint64_t)valids[c], nullptr, 1,
+                      es};
+            vall[b * n_cols + c] = dc;
+            if (c == 0) vkeys[b] = dc;
+            if (c == seq_idx) vseqs[b] = dc;
+            if (c == kind_idx) vkinds[b] = dc;
+        }
+        void **op = (void **)plan->bufs.alloc(n_cols * sizeof(void *));
+        uint8_t **vp =
+            (uint8_t **)plan->bufs.alloc(n_cols * sizeof(void *));
+        if (!op || !vp ||
+            hipMemcpy(op, outs.data(), n_cols * sizeof(void *),
+                      hipMemcpyHostToDevice) != hipSuccess ||
+            hipMemcpy(vp, valids.data(), n_cols * sizeof(void *),
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            set_error("H2D hier batch outputs failed");
+            return false;
+        }
+        sec.bout_ptrs.push_back(op);
+        sec.bout_valid.push_back(vp);
+    }
+    auto up = [&](const void *host, size_t n) -> void * {
+        void *d = plan->bufs.alloc(n);
+        if (d && host &&
+            hipMemcpy(d, host, n, hipMemcpyHostToDevice) != hipSuccess)
+            return nullptr;
+        return d;
+    };
+    sec.key_cols = (DevCol *)up(vkeys.data(), nb * sizeof(DevCol));
+    sec.seq_cols = (DevCol *)up(vseqs.data(), nb * sizeof(DevCol));
+    sec.kind_cols = (DevCol *)up(vkinds.data(), nb * sizeof(DevCol));
+    sec.all_cols = (DevCol *)up(vall.data(), vall.size() * sizeof(DevCol));
+    sec.lens_dev = (int64_t *)plan->bufs.alloc(nb * 8);
+    if (!sec.key_cols || !sec.seq_cols || !sec.kind_cols || !sec.all_cols ||
+        !sec.lens_dev) {
+        set_error("hier descriptor allocation failed");
+        return false;
+    }
+    return true;
+}
+
 static bool build_section_descriptors(pmh_plan_t *plan, Section &sec) {
     int k = (int)sec.runs.size();
     int n_cols = (int)plan->cols.size();
@@ -2783,10 +2924,18 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
         for (auto &sec_files : sections) {
             Section sec;
             if ((int)sec_files.size() > PMH_MAX_RUNS) {
-                set_error("section has %zu runs > %d (spill path is a later "
-                          "round, MergeSorter.java:112-125)",
-                          sec_files.size(), PMH_MAX_RUNS);
-                return nullptr;
+                if (plan->pu || plan->agg || plan->ignore_delete ||
+                    plan->changelog || plan->fused || plan->composite_key) {
+                    set_error(
+                        "section has %zu runs > %d: the hierarchical "
+                        "winner merge serves deduplicate/first-row without "
+                        "ignore-delete / changelog / sequence-fields / "
+                        "composite keys in v1 (the reference spills to "
+                        "disk here, MergeSorter.java:112-125)",
+                        sec_files.size(), PMH_MAX_RUNS);
+                    return nullptr;
+                }
+                sec.hier = true;
             }
             sec.runs.resize(sec_files.size());
             int64_t run_rows = 0;
@@ -2799,6 +2948,8 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                 set_error("device allocation failed for section");
                 return nullptr;
             }
+            if (sec.hier && !build_hier_section(plan.get(), sec))
+                return nullptr;
             plan->rows_in_total += run_rows;
             plan->sections.push_back(std::move(sec));
         }
@@ -3103,9 +3254,54 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         }
     }
     (void)hipEventRecord(ev[1], st);
+    int64_t n_tiles_eff = sec.n_tiles;
+    int64_t rows_eff = sec.total_rows;
+    if (sec.hier) {
+        // batch pass: merge run batches into the virtual runs
+        // (winner-of-winners — the dedup/first-row fold is associative;
+        // deletes are KEPT here and drop only in the final chain)
+        const int nb = (int)sec.blo.size();
+        k = nb;
+        const int bflags = p->first_row ? 8 : 0;  // no drop, no ignore
+        for (int b = 0; b < nb; b++) {
+            int lo = sec.blo[b];
+            int kb = sec.bhi[b] - lo;
+            int64_t btiles = sec.bntiles[b];
+            hipError_t hb = pmh_launch_partition(
+                sec.rkeys + lo, sec.rlens + lo, kb, PMH_TILE_ROWS,
+                btiles + 1, sec.brows[b], sec.cuts, st);
+            if (hb != hipSuccess) return fail("hier partition", hb);
+            hb = pmh_launch_merge_tiles(
+                sec.rkeys + lo, sec.rseqs + lo, sec.rkinds + lo,
+                sec.rlens + lo, kb, sec.cuts, btiles, PMH_TILE_ROWS, bflags,
+                sec.rtombs ? sec.rtombs + lo : nullptr, sec.winners,
+                sec.tile_counts, sec.group_start, sec.err_dev, nullptr, 0,
+                nullptr, nullptr, st);
+            if (hb != hipSuccess) return fail("hier merge", hb);
+            hb = pmh_launch_scan_tiles(sec.tile_counts, btiles,
+                                       sec.tile_offsets, sec.lens_dev + b,
+                                       st);
+            if (hb != hipSuccess) return fail("hier scan", hb);
+            hb = pmh_launch_emit(sec.rall + (size_t)lo * n_cols,
+                                 p->hier_dtype_dev, p->col_nullable_dev,
+                                 n_cols, kb, sec.winners, sec.tile_counts,
+                                 sec.tile_offsets, btiles, PMH_TILE_ROWS,
+                                 sec.lens_dev + b, sec.bout_ptrs[b],
+                                 sec.bout_valid[b], st);
+            if (hb != hipSuccess) return fail("hier emit", hb);
+        }
+        std::vector<int64_t> vl(nb);
+        if (hipMemcpy(vl.data(), sec.lens_dev, nb * 8,
+                      hipMemcpyDeviceToHost) != hipSuccess)
+            return fail("hier lens D2H", hipErrorUnknown);
+        rows_eff = 0;
+        for (int b = 0; b < nb; b++) rows_eff += vl[b];
+        n_tiles_eff = (rows_eff + PMH_TILE_ROWS - 1) / PMH_TILE_ROWS;
+        if (n_tiles_eff == 0) n_tiles_eff = 1;
+    }
     hipError_t e = pmh_launch_partition(sec.key_cols, sec.lens_dev, k,
-                                        PMH_TILE_ROWS, sec.n_tiles + 1,
-                                        sec.total_rows, sec.cuts, st);
+                                        PMH_TILE_ROWS, n_tiles_eff + 1,
+                                        rows_eff, sec.cuts, st);
     if (e != hipSuccess) return fail("partition", e);
     (void)hipEventRecord(ev[2], st);
     int flags = (p->drop_delete ? 1 : 0) | (p->ignore_delete ? 2 : 0) |
@@ -3122,7 +3318,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         // single-pass merge + emit: zero the lookback words + ticket, then
         // one kernel does merge, offsets and emission (scan/emit launches
         // and the winners round-trip disappear)
-        e = hipMemsetAsync(sec.status, 0, sec.n_tiles * 8, st);
+        e = hipMemsetAsync(sec.status, 0, n_tiles_eff * 8, st);
         if (e != hipSuccess) return fail("status memset", e);
         e = hipMemsetAsync(sec.ticket, 0, 8, st);
         if (e != hipSuccess) return fail("ticket memset", e);
@@ -3143,14 +3339,14 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             const int64_t n_chunks =
                 want < 1 ? 1 : (want > 8 ? 8 : want);
             const int64_t per =
-                (sec.n_tiles + n_chunks - 1) / n_chunks;
+                (n_tiles_eff + n_chunks - 1) / n_chunks;
             hipEvent_t cev[8];
             for (int64_t c = 0; c < n_chunks; c++)
                 (void)hipEventCreateWithFlags(&cev[c],
                                               hipEventDisableTiming);
             for (int64_t c = 0; c < n_chunks; c++) {
                 int64_t t0 = c * per;
-                int64_t t1 = t0 + per < sec.n_tiles ? t0 + per : sec.n_tiles;
+                int64_t t1 = t0 + per < n_tiles_eff ? t0 + per : n_tiles_eff;
                 if (t0 >= t1) { (void)hipEventRecord(cev[c], st); continue; }
                 if (c > 0) {  // ticket restarts per chunk
                     e = hipMemsetAsync(sec.ticket, 0, 8, st);
@@ -3158,7 +3354,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                 }
                 e = pmh_launch_merge_emit(
                     sec.key_cols, sec.seq_cols, sec.kind_cols, sec.lens_dev,
-                    k, sec.cuts, t0, t1, sec.n_tiles, PMH_TILE_ROWS, flags,
+                    k, sec.cuts, t0, t1, n_tiles_eff, PMH_TILE_ROWS, flags,
                     sec.tombs_dev,
                     sec.all_cols, p->col_dtype_dev, p->col_nullable_dev,
                     n_cols, key_col, p->n_key_cols, p->n_key_cols + 1,
@@ -3175,7 +3371,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                 for (int64_t c = 0; c < n_chunks; c++) {
                     int64_t t0 = c * per;
                     int64_t t1 =
-                        t0 + per < sec.n_tiles ? t0 + per : sec.n_tiles;
+                        t0 + per < n_tiles_eff ? t0 + per : n_tiles_eff;
                     if (t0 >= t1) continue;
                     (void)hipStreamWaitEvent(sb, cev[c], 0);
                     e = pmh_launch_emit_dense(
@@ -3199,7 +3395,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         goto collect;
     }
     e = pmh_launch_merge_tiles(sec.key_cols, sec.seq_cols, sec.kind_cols,
-                               sec.lens_dev, k, sec.cuts, sec.n_tiles,
+                               sec.lens_dev, k, sec.cuts, n_tiles_eff,
                                PMH_TILE_ROWS,
                                flags | (p->cl_row_dedup ? 128 : 0),
                                sec.tombs_dev, sec.winners,
@@ -3208,7 +3404,7 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
                                sec.cl_counts, st);
     if (e != hipSuccess) return fail("merge_tiles", e);
     (void)hipEventRecord(ev[3], st);
-    e = pmh_launch_scan_tiles(sec.tile_counts, sec.n_tiles, sec.tile_offsets,
+    e = pmh_launch_scan_tiles(sec.tile_counts, n_tiles_eff, sec.tile_offsets,
                               sec.total_dev, st);
     if (e != hipSuccess) return fail("scan_tiles", e);
     (void)hipEventRecord(ev[4], st);
@@ -3217,27 +3413,27 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev,
             p->col_agg_dev, n_cols, k, p->n_key_cols, p->n_key_cols + 1,
             flags, sec.winners, sec.group_start, sec.tile_offsets,
-            sec.n_tiles, PMH_TILE_ROWS, sec.total_dev, sec.row_masks_dev,
+            n_tiles_eff, PMH_TILE_ROWS, sec.total_dev, sec.row_masks_dev,
             p->out_ptrs_dev, p->out_valid_dev, st);
     } else if (p->pu && p->seqg) {
         e = pmh_launch_emit_pu_sg(
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,
             p->n_key_cols, p->n_key_cols + 1, flags, p->col_group_dev,
             p->sg_fields_dev, p->sg_nseq_dev, p->n_seq_groups, sec.winners,
-            sec.group_start, sec.tile_offsets, sec.n_tiles, PMH_TILE_ROWS,
+            sec.group_start, sec.tile_offsets, n_tiles_eff, PMH_TILE_ROWS,
             sec.total_dev, sec.row_masks_dev, p->out_ptrs_dev,
             p->out_valid_dev, st);
     } else if (p->pu) {
         e = pmh_launch_emit_pu(
             sec.all_cols, p->col_dtype_dev, p->col_nullable_dev, n_cols, k,
             p->n_key_cols, p->n_key_cols + 1, flags, sec.winners,
-            sec.group_start, sec.tile_offsets, sec.n_tiles, PMH_TILE_ROWS,
+            sec.group_start, sec.tile_offsets, n_tiles_eff, PMH_TILE_ROWS,
             sec.total_dev, sec.row_masks_dev, p->out_ptrs_dev,
             p->out_valid_dev, st);
     } else {
         e = pmh_launch_emit(sec.all_cols, p->col_dtype_dev,
                             p->col_nullable_dev, n_cols, k, sec.winners,
-                            sec.tile_counts, sec.tile_offsets, sec.n_tiles,
+                            sec.tile_counts, sec.tile_offsets, n_tiles_eff,
                             PMH_TILE_ROWS, sec.total_dev, p->out_ptrs_dev,
                             p->out_valid_dev, st);
     }
@@ -3245,16 +3441,16 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
     if (p->changelog) {
         e = pmh_launch_cl_finalize(sec.all_cols, p->col_dtype_dev, n_cols,
                                    p->n_key_cols + 2, k, sec.cl_entries,
-                                   sec.cl_rows, sec.cl_counts, sec.n_tiles,
+                                   sec.cl_rows, sec.cl_counts, n_tiles_eff,
                                    PMH_TILE_ROWS, st);
         if (e != hipSuccess) return fail("cl_finalize", e);
-        e = pmh_launch_scan_tiles(sec.cl_counts, sec.n_tiles, sec.cl_offsets,
+        e = pmh_launch_scan_tiles(sec.cl_counts, n_tiles_eff, sec.cl_offsets,
                                   sec.cl_total_dev, st);
         if (e != hipSuccess) return fail("cl_scan", e);
         e = pmh_launch_cl_emit(sec.all_cols, p->col_dtype_dev,
                                p->col_nullable_dev, n_cols,
                                p->n_key_cols + 1, sec.cl_rows, sec.cl_counts,
-                               sec.cl_offsets, sec.n_tiles, PMH_TILE_ROWS,
+                               sec.cl_offsets, n_tiles_eff, PMH_TILE_ROWS,
                                p->out_ptrs2_dev, p->out_valid2_dev, st);
         if (e != hipSuccess) return fail("cl_emit", e);
     }

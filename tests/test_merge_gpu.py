@@ -428,14 +428,17 @@ class TestErrorPaths:
     # everything outside the matrix fails loudly at plan create with a
     # descriptive error so a Java-side provider can fall
     # back to the stock reader per split
-    def test_too_many_overlapping_runs(self, tmp_path):
+    def test_too_many_runs_unsupported_engine(self, tmp_path):
+        # >32 runs merge hierarchically for deduplicate/first-row
+        # (TestHierarchicalSections); engines whose fold is NOT associative
+        # still reject (the reference spills to disk, MergeSorter.java)
         runs = gen_runs_dedup(33, 500, n_value_cols=1, seed=70,
                               delete_frac=0.0)
         metas = write_runs(runs, str(tmp_path), compression="NONE")
         with Session(0) as s:
             with pytest.raises(RuntimeError, match="32"):
                 MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
-                              _value_cols(1))
+                              _value_cols(1), merge_engine="partial-update")
 
     def test_run_row_overflow_rejected(self, tmp_path):
         # a run is the concatenation of its files: two non-overlapping files
@@ -716,3 +719,94 @@ class TestSortEngineOption:
             with pytest.raises(RuntimeError, match="sort-engine"):
                 MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
                               _value_cols(1), sort_engine="quick-sort")
+
+
+class TestHierarchicalSections:
+    """> PMH_MAX_RUNS (32) overlapping runs: batches of runs merge into
+    VIRTUAL runs first (winner-of-winners — the deduplicate/first-row fold
+    is associative; deletes survive the batch pass and drop only at the
+    final chain), then the normal chain merges the virtual runs. The
+    reference spills to disk here (MergeSorter.spillMergeSort)."""
+
+    def test_40_runs_dedup(self, tmp_path):
+        runs = gen_runs_dedup(40, 4_000, n_value_cols=3, seed=901,
+                              delete_frac=0.25)
+        _run_and_compare(tmp_path, runs)
+
+    def test_64_runs_dedup(self, tmp_path):
+        runs = gen_runs_dedup(64, 1_500, n_value_cols=2, seed=902,
+                              delete_frac=0.3)
+        _run_and_compare(tmp_path, runs)
+
+    def test_33_runs_keep_delete(self, tmp_path):
+        runs = gen_runs_dedup(33, 2_000, n_value_cols=2, seed=903,
+                              delete_frac=0.4)
+        _run_and_compare(tmp_path, runs, drop_delete=False)
+
+    def test_first_row_40_runs(self, tmp_path):
+        from oracle import merge_first_row_model
+        runs = gen_runs_dedup(40, 2_000, n_value_cols=2, seed=904,
+                              delete_frac=0.0)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        r, w = merge_first_row_model(runs)
+        exp = np.array([runs[a]["key"][b] for a, b in zip(r, w)], np.int64)
+        exp_s = np.array([runs[a]["seq"][b] for a, b in zip(r, w)], np.int64)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(2),
+                               merge_engine="first-row") as plan:
+                got_k, got_s = [], []
+                while True:
+                    b = plan.read_next()
+                    if b is None:
+                        break
+                    got_k.append(b["_KEY_k"].copy())
+                    got_s.append(b["_SEQUENCE_NUMBER"].copy())
+        assert (np.concatenate(got_k) == exp).all()
+        assert (np.concatenate(got_s) == exp_s).all()
+
+    def test_hier_with_nullable_columns(self, tmp_path):
+        # nulls flow through the virtual runs' validity bytes
+        import pyarrow as pa
+        import pyarrow.parquet as pq
+        rng = np.random.default_rng(905)
+        runs = gen_runs_dedup(36, 3_000, n_value_cols=2, seed=905,
+                              delete_frac=0.1)
+        masks = [rng.random(len(r["key"])) < 0.3 for r in runs]
+        metas = []
+        for i, (r, m) in enumerate(zip(runs, masks)):
+            tbl = pa.table({
+                "_KEY_k": pa.array(r["key"]),
+                "_SEQUENCE_NUMBER": pa.array(r["seq"]),
+                "_VALUE_KIND": pa.array(r["kind"]),
+                "v_k": pa.array(r["values"][0]),
+                "v_c0": pa.array(r["values"][1], mask=m),
+                "v_c1": pa.array(r["values"][2]),
+            })
+            path = str(tmp_path / f"run-{i}.parquet")
+            pq.write_table(tbl, path, compression=None,
+                           use_dictionary=False, data_page_version="1.0",
+                           store_schema=False)
+            metas.append({"path": path, "rowCount": len(r["key"]),
+                          "minKey": int(r["key"][0]),
+                          "maxKey": int(r["key"][-1]), "level": 0})
+        rr, ww = merge_dedup(runs)
+        exp_k = np.array([runs[a]["key"][b] for a, b in zip(rr, ww)],
+                         np.int64)
+        exp_v = np.array([runs[a]["values"][1][b] for a, b in zip(rr, ww)])
+        exp_null = np.array([masks[a][b] for a, b in zip(rr, ww)], bool)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(2)) as plan:
+                got = {}
+                while True:
+                    b = plan.read_next()
+                    if b is None:
+                        break
+                    for kk, v in b.items():
+                        got.setdefault(kk, []).append(v.copy())
+                got = {kk: np.concatenate(v) for kk, v in got.items()}
+        assert (got["_KEY_k"] == exp_k).all()
+        assert (got["v_c0#valid"] == ~exp_null).all()
+        live = ~exp_null
+        assert (got["v_c0"][live] == exp_v[live]).all()
