@@ -154,3 +154,54 @@ class TestFullCompactionChangelog:
                 MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
                               _value_cols(len(runs[0]["values"]) - 1),
                               changelog_producer="full-compaction")
+
+
+class TestChangelogWithDeletionVectors:
+    """Cross-feature: DV tombstones remove members BEFORE the wrapper sees
+    them (ApplyDeletionVectorReader feeds the merge), so a deleted top-level
+    record means "no top" for the changelog decision."""
+
+    def test_dv_filtered_top(self, tmp_path):
+        from scripts.gen_dv_golden import serialize_roaring32, wrap_dv
+        rng = np.random.default_rng(911)
+        runs, levels = _with_top_run(3, 15_000, seed=911, delete_frac=0.15)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        for m, lvl in zip(metas, levels):
+            m["level"] = lvl
+        # delete ~20% of the TOP run's rows and ~10% of run 0's
+        dels = {
+            len(runs) - 1: sorted(rng.choice(
+                len(runs[-1]["key"]),
+                len(runs[-1]["key"]) // 5, replace=False).tolist()),
+            0: sorted(rng.choice(
+                len(runs[0]["key"]),
+                len(runs[0]["key"]) // 10, replace=False).tolist()),
+        }
+        blob = b""
+        for fi, pos in dels.items():
+            ser = wrap_dv(serialize_roaring32(pos))
+            metas[fi]["deletionVector"] = {
+                "file": str(tmp_path / "index.dv"),
+                "offset": len(blob), "length": len(ser)}
+            blob += ser
+        (tmp_path / "index.dv").write_bytes(blob)
+        # expected: run the oracle over the FILTERED runs
+        fruns = []
+        for i, r in enumerate(runs):
+            keep = np.ones(len(r["key"]), dtype=bool)
+            if i in dels:
+                keep[np.array(dels[i], dtype=np.int64)] = False
+            fruns.append({"key": r["key"][keep], "seq": r["seq"][keep],
+                          "kind": r["kind"][keep],
+                          "values": [v[keep] for v in r["values"]]})
+        exp_cl, exp_res_keys = _expected(fruns, levels)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(len(runs[0]["values"]) - 1),
+                               changelog_producer="full-compaction",
+                               max_level=MAX_LEVEL) as plan:
+                main, cl = _read_all(plan)
+        assert (main["_KEY_k"] == exp_res_keys).all()
+        assert len(cl["_KEY_k"]) == len(exp_cl["_KEY_k"])
+        for name, e in exp_cl.items():
+            assert (cl[name] == e).all(), name
