@@ -78,3 +78,29 @@ class TestScalarZstd:
         comp[0] ^= 0xFF  # break the magic
         with pytest.raises(RuntimeError):
             debug_zstd_cpu(bytes(comp), cap=1 << 16)
+
+
+class TestNoContentSizeFrames:
+    """Streaming-written frames omit the frame content size (FCS) —
+    pz_decode_frame must decode them (parquet pages always carry FCS, but
+    the decoder follows RFC 8878, not the writer's habits)."""
+
+    def test_streaming_frame(self):
+        import io
+        raw = (b"streaming frame payload " * 4000)
+        buf = pa.BufferOutputStream()
+        with pa.CompressedOutputStream(buf, "zstd") as z:
+            z.write(raw)
+        comp = buf.getvalue().to_pybytes()
+        got = debug_zstd_cpu(comp, cap=len(raw) + 64)
+        assert got == raw
+
+    def test_streaming_incompressible(self):
+        rng = np.random.default_rng(9)
+        raw = rng.integers(0, 256, 300_000, dtype=np.uint8).tobytes()
+        buf = pa.BufferOutputStream()
+        with pa.CompressedOutputStream(buf, "zstd") as z:
+            z.write(raw)
+        comp = buf.getvalue().to_pybytes()
+        got = debug_zstd_cpu(comp, cap=len(raw) + 64)
+        assert got == raw
