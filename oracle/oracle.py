@@ -838,7 +838,7 @@ def merge_dedup_useq_model(runs, useq_idx, drop_delete=True,
 
 
 def full_changelog_model(runs, levels, max_level, row_dedup=False,
-                         ignore_delete=False):
+                         ignore_delete=False, masks=None):
     """Numpy restatement of FullChangelogMergeFunctionWrapper.java:74-130
     over DeduplicateMergeFunction (getResult decision table; pinned to
     FullChangelogMergeFunctionWrapperTestBase's vectors in
@@ -896,9 +896,12 @@ def full_changelog_model(runs, levels, max_level, row_dedup=False,
         ra, xa = run[ia], row[ia]
         rb, xb = run[ib], row[ib]
         for c in range(len(runs[0]["values"])):
-            va = runs[ra]["values"][c][xa]
-            vb = runs[rb]["values"][c][xb]
-            if va != vb:
+            na = masks[ra][c][xa] if masks else False
+            nb = masks[rb][c][xb] if masks else False
+            if na != nb:
+                return False
+            if not na and runs[ra]["values"][c][xa] != \
+                    runs[rb]["values"][c][xb]:
                 return False
         return True
 

@@ -205,3 +205,55 @@ class TestChangelogWithDeletionVectors:
         assert len(cl["_KEY_k"]) == len(exp_cl["_KEY_k"])
         for name, e in exp_cl.items():
             assert (cl[name] == e).all(), name
+
+
+class TestRowDedupWithNulls:
+    """changelog-producer.row-deduplicate over NULLABLE value columns: the
+    equaliser treats null==null as equal and null!=value as different
+    (k_cl_finalize compares validity bytes before values)."""
+
+    def test_nullable_equality(self, tmp_path):
+        import pyarrow as pa
+        import pyarrow.parquet as pq
+        rng = np.random.default_rng(921)
+        runs, levels = _with_top_run(2, 12_000, seed=921, delete_frac=0.05,
+                                     value_card=3)
+        masks = []
+        for r in runs:
+            m = [np.zeros(len(r["key"]), bool)]  # v_k non-null
+            for _ in range(len(r["values"]) - 1):
+                m.append(rng.random(len(r["key"])) < 0.4)
+            masks.append(m)
+        metas = []
+        for i, r in enumerate(runs):
+            cols = {"_KEY_k": pa.array(r["key"]),
+                    "_SEQUENCE_NUMBER": pa.array(r["seq"]),
+                    "_VALUE_KIND": pa.array(r["kind"]),
+                    "v_k": pa.array(r["values"][0])}
+            for c in range(1, len(r["values"])):
+                cols[f"v_c{c-1}"] = pa.array(r["values"][c],
+                                             mask=masks[i][c])
+            tbl = pa.table(cols)
+            path = str(tmp_path / f"run-{i}.parquet")
+            pq.write_table(tbl, path, compression=None,
+                           use_dictionary=False, data_page_version="1.0",
+                           store_schema=False)
+            metas.append({"path": path, "rowCount": len(r["key"]),
+                          "minKey": int(r["key"][0]),
+                          "maxKey": int(r["key"][-1]),
+                          "level": levels[i]})
+        (cr, cw, ck), (rr2, rw2) = full_changelog_model(
+            runs, levels, MAX_LEVEL, row_dedup=True, masks=masks)
+        exp_keys = np.array([runs[a]["key"][b] for a, b in zip(cr, cw)],
+                            np.int64)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(len(runs[0]["values"]) - 1),
+                               changelog_producer="full-compaction",
+                               changelog_row_dedup=True,
+                               max_level=MAX_LEVEL) as plan:
+                main, cl = _read_all(plan)
+        assert len(cl["_KEY_k"]) == len(exp_keys), \
+            (len(cl["_KEY_k"]), len(exp_keys))
+        assert (cl["_KEY_k"] == exp_keys).all()
+        assert (cl["_VALUE_KIND"] == ck).all()
