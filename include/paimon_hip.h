@@ -187,6 +187,11 @@ int64_t pmh_debug_zstd_cpu(const void *src, int64_t n, void *dst,
 int64_t pmh_debug_zstd_gpu(const void *src, int64_t n, void *dst,
                            int64_t expected);
 
+/* Compress one frame with the from-scratch encoder (scalar core; the GPU
+ * page compressor shares it). Returns compressed bytes or < 0. */
+int64_t pmh_debug_zstd_enc_cpu(const void *src, int64_t n, void *dst,
+                               int64_t cap);
+
 /* Parse one deletion vector from a DV index file slice (DeletionFile
  * {path, offset, length}; BitmapDeletionVector.java:98-112 wrapper around
  * the portable Roaring serialization). Writes up to `cap` deleted

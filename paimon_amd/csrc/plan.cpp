@@ -3699,6 +3699,18 @@ int64_t pmh_debug_zstd_gpu(const void *src, int64_t n, void *dst,
     return expected;
 }
 
+// CPU-side entry to the from-scratch zstd ENCODER (zstd_core.h): compress
+// one frame; tests round-trip it through libzstd/pyarrow to prove the
+// frames are spec-valid (the GPU kernel shares the same core).
+int64_t pmh_debug_zstd_enc_cpu(const void *src, int64_t n, void *dst,
+                               int64_t cap) {
+    std::vector<PzEnc> e(1);
+    int64_t r = pz_encode_frame((const uint8_t *)src, n, (uint8_t *)dst,
+                                cap, e.data());
+    if (r < 0) set_error("pz_encode_frame: error %lld", (long long)r);
+    return r;
+}
+
 int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap) {
     size_t got = 0;
     std::string err;

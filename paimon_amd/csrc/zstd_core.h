@@ -831,3 +831,386 @@ PZHD int64_t pz_decode_frame(const uint8_t *src, int64_t slen, uint8_t *dst,
     if (F.content_size >= 0 && dp != F.content_size) return PZ_ERR_CORRUPT;
     return dp;
 }
+
+// ================================================================ ENCODER
+//
+// Simplified zstd COMPRESSOR producing spec-valid frames any zstd decoder
+// reads (RFC 8878): greedy hash-table LZ matching, RAW literals, sequences
+// entropy-coded with the PREDEFINED FSE distributions (no table
+// descriptions, no huffman literals — simplicity and wave-friendliness
+// over the last few percent of ratio). The write-side counterpart of the
+// decoder above; parquet page compression for the compaction write-back
+// (the reference delegates to parquet-java's zstd codec — the contract is
+// only "a valid zstd frame of these bytes").
+
+// forward LSB-first bit writer; closing appends the '1' sentinel bit
+typedef struct {
+    uint8_t *out;
+    int64_t cap;
+    int64_t pos;   // bytes written
+    uint64_t acc;
+    int nbits;
+} PzBitW;
+
+PZHD void pzw_init(PzBitW *w, uint8_t *out, int64_t cap) {
+    w->out = out;
+    w->cap = cap;
+    w->pos = 0;
+    w->acc = 0;
+    w->nbits = 0;
+}
+
+PZHD int pzw_add(PzBitW *w, uint64_t v, int n) {
+    if (n == 0) return 0;
+    w->acc |= (v & ((n >= 64) ? ~0ull : ((1ull << n) - 1))) << w->nbits;
+    w->nbits += n;
+    while (w->nbits >= 8) {
+        if (w->pos >= w->cap) return PZ_ERR_DST_SMALL;
+        w->out[w->pos++] = (uint8_t)w->acc;
+        w->acc >>= 8;
+        w->nbits -= 8;
+    }
+    return 0;
+}
+
+PZHD int64_t pzw_close(PzBitW *w) {  // returns bytes or error
+    if (pzw_add(w, 1, 1) < 0) return PZ_ERR_DST_SMALL;
+    if (w->nbits > 0) {
+        if (w->pos >= w->cap) return PZ_ERR_DST_SMALL;
+        w->out[w->pos++] = (uint8_t)w->acc;
+        w->acc = 0;
+        w->nbits = 0;
+    }
+    return w->pos;
+}
+
+// FSE encode tables (FSE_buildCTable over the predefined distributions)
+typedef struct {
+    int32_t deltaNbBits;    // (maxBitsOut << 16) - minStatePlus
+    int32_t deltaFindState;
+} PzFseCSym;
+
+typedef struct {
+    uint16_t stateTable[512];  // max predefined table size is 64; 512 is
+                               // roomy for any AL <= 9
+    PzFseCSym sym[64];
+    int tableLog;
+} PzFseC;
+
+PZHD int pz_fse_build_ctable(const int16_t *norm, int nsym, int al,
+                             PzFseC *ct) {
+    const int size = 1 << al;
+    ct->tableLog = al;
+    // spread (same placement as the decode table)
+    uint8_t spread[512];
+    int high = size - 1;
+    for (int s = 0; s < nsym; s++)
+        if (norm[s] == -1) spread[high--] = (uint8_t)s;
+    int step = (size >> 1) + (size >> 3) + 3;
+    int pos = 0;
+    for (int s = 0; s < nsym; s++) {
+        for (int i = 0; i < norm[s]; i++) {
+            spread[pos] = (uint8_t)s;
+            do {
+                pos = (pos + step) & (size - 1);
+            } while (pos > high);
+        }
+    }
+    if (pos != 0) return PZ_ERR_FSE;
+    // cumulative start per symbol
+    int cumul[64];
+    int total = 0;
+    for (int s = 0; s < nsym; s++) {
+        cumul[s] = total;
+        total += norm[s] == -1 ? 1 : norm[s];
+    }
+    if (total != size) return PZ_ERR_FSE;
+    // state table: cells in table order fill each symbol's slots
+    for (int u = 0; u < size; u++) {
+        int s = spread[u];
+        ct->stateTable[cumul[s]++] = (uint16_t)(size + u);
+    }
+    // per-symbol transforms
+    total = 0;
+    for (int s = 0; s < nsym; s++) {
+        int freq = norm[s] == -1 ? 1 : norm[s];
+        if (freq == 0) {
+            ct->sym[s].deltaNbBits = ((al + 1) << 16) - (1 << al);
+            ct->sym[s].deltaFindState = 0;
+            continue;
+        }
+        if (freq == 1) {
+            ct->sym[s].deltaNbBits = (al << 16) - (1 << al);
+            ct->sym[s].deltaFindState = total - 1;
+            total += 1;
+        } else {
+            int maxBitsOut = al - pz_highbit((uint32_t)(freq - 1));
+            int minStatePlus = freq << maxBitsOut;
+            ct->sym[s].deltaNbBits = (maxBitsOut << 16) - minStatePlus;
+            ct->sym[s].deltaFindState = total - freq;
+            total += freq;
+        }
+    }
+    return 0;
+}
+
+typedef struct {
+    uint32_t value;
+} PzFseCState;
+
+PZHD void pz_fse_cinit(PzFseCState *st, const PzFseC *ct, int sym) {
+    uint32_t nbBitsOut =
+        (uint32_t)(ct->sym[sym].deltaNbBits + (1 << 15)) >> 16;
+    uint32_t v = (nbBitsOut << 16) - (uint32_t)ct->sym[sym].deltaNbBits;
+    st->value =
+        ct->stateTable[(v >> nbBitsOut) + ct->sym[sym].deltaFindState];
+}
+
+PZHD int pz_fse_cenc(PzBitW *w, PzFseCState *st, const PzFseC *ct,
+                     int sym) {
+    uint32_t nbBitsOut =
+        (uint32_t)(st->value + (uint32_t)ct->sym[sym].deltaNbBits) >> 16;
+    int rc = pzw_add(w, st->value, (int)nbBitsOut);
+    st->value = ct->stateTable[(st->value >> nbBitsOut) +
+                               ct->sym[sym].deltaFindState];
+    return rc;
+}
+
+PZHD int pz_fse_cflush(PzBitW *w, const PzFseCState *st, const PzFseC *ct) {
+    return pzw_add(w, st->value, ct->tableLog);
+}
+
+// code mapping (inverse of the baseline tables above)
+PZHD int pz_ll_code(uint32_t ll) {
+    if (ll <= 15) return (int)ll;
+    for (int c = 35; c >= 16; c--)
+        if (ll >= pz_ll_base(c)) return c;
+    return 16;
+}
+
+PZHD int pz_ml_code(uint32_t ml) {  // ml >= 3
+    if (ml <= 34) return (int)(ml - 3);
+    for (int c = 52; c >= 32; c--)
+        if (ml >= pz_ml_base(c)) return c;
+    return 32;
+}
+
+// per-call encoder scratch (one per wave/job): hash table + sequence lists
+#define PZ_ENC_HLOG 14
+#define PZ_ENC_MAXSEQ (PZ_BLOCK_MAX / 4 + 16)
+typedef struct {
+    int32_t htab[1 << PZ_ENC_HLOG];
+    uint32_t s_ll[PZ_ENC_MAXSEQ];
+    uint32_t s_ml[PZ_ENC_MAXSEQ];
+    uint32_t s_of[PZ_ENC_MAXSEQ];
+    PzFseC ct_ll, ct_of, ct_ml;
+    int16_t norm[64];
+} PzEnc;
+
+PZHD uint32_t pz_read32(const uint8_t *p) {
+    return (uint32_t)p[0] | ((uint32_t)p[1] << 8) | ((uint32_t)p[2] << 16) |
+           ((uint32_t)p[3] << 24);
+}
+
+PZHD uint32_t pz_hash32(uint32_t v) {
+    return (v * 2654435761u) >> (32 - PZ_ENC_HLOG);
+}
+
+// Encode one frame: single-segment, 4-byte content size, blocks of
+// <= PZ_BLOCK_MAX. Returns compressed size or PZ_ERR_*. Worst case output
+// is n + 3 bytes/block + 9 header bytes — pass cap >= n + (n>>10) + 64.
+PZHD int64_t pz_encode_frame(const uint8_t *src, int64_t n, uint8_t *dst,
+                             int64_t cap, PzEnc *e) {
+    // frame header: magic, FHD = single_segment | fcs_flag 2 (4-byte FCS)
+    if (cap < 16) return PZ_ERR_DST_SMALL;
+    int64_t dp = 0;
+    dst[dp++] = 0x28;
+    dst[dp++] = 0xB5;
+    dst[dp++] = 0x2F;
+    dst[dp++] = 0xFD;
+    dst[dp++] = (2 << 6) | (1 << 5);  // FCS 4 bytes, single segment
+    dst[dp++] = (uint8_t)n;
+    dst[dp++] = (uint8_t)(n >> 8);
+    dst[dp++] = (uint8_t)(n >> 16);
+    dst[dp++] = (uint8_t)(n >> 24);
+    // build the predefined encode tables once per frame
+    int nsym, al, rc;
+    pz_fse_predef(0, e->norm, &nsym, &al);
+    if ((rc = pz_fse_build_ctable(e->norm, nsym, al, &e->ct_ll)) < 0)
+        return rc;
+    pz_fse_predef(1, e->norm, &nsym, &al);
+    if ((rc = pz_fse_build_ctable(e->norm, nsym, al, &e->ct_of)) < 0)
+        return rc;
+    pz_fse_predef(2, e->norm, &nsym, &al);
+    if ((rc = pz_fse_build_ctable(e->norm, nsym, al, &e->ct_ml)) < 0)
+        return rc;
+    for (int i = 0; i < (1 << PZ_ENC_HLOG); i++) e->htab[i] = -1;
+    int64_t bp = 0;  // block start
+    if (n == 0) {
+        // empty frame: one empty RAW last block
+        if (dp + 3 > cap) return PZ_ERR_DST_SMALL;
+        dst[dp++] = 1;  // last=1, type raw, size 0
+        dst[dp++] = 0;
+        dst[dp++] = 0;
+        return dp;
+    }
+    while (bp < n) {
+        int64_t bn = n - bp < PZ_BLOCK_MAX ? n - bp : PZ_BLOCK_MAX;
+        int last = bp + bn >= n;
+        // ---- greedy match pass over [bp, bp+bn)
+        int nseq = 0;
+        int64_t lit_start = bp;   // start of pending literals
+        int64_t lit_total = 0;    // literal bytes of this block
+        int64_t p = bp;
+        const int64_t limit = bp + bn - 4;  // need 4 bytes to match
+        while (p <= limit) {
+            uint32_t v = pz_read32(src + p);
+            uint32_t h = pz_hash32(v);
+            int64_t cand = e->htab[h];
+            e->htab[h] = (int32_t)p;
+            if (cand >= 0 && pz_read32(src + (int64_t)cand) == v &&
+                p - cand <= (1 << 27)) {
+                // extend
+                int64_t m = 4;
+                int64_t maxm = bp + bn - p;
+                while (m < maxm && src[cand + m] == src[p + m]) m++;
+                uint32_t ll = (uint32_t)(p - lit_start);
+                if (nseq >= PZ_ENC_MAXSEQ - 1) break;  // overflow: literals
+                e->s_ll[nseq] = ll;
+                e->s_ml[nseq] = (uint32_t)m;
+                e->s_of[nseq] = (uint32_t)(p - cand);
+                nseq++;
+                lit_total += ll;
+                // seed a few hashes inside the match (cheap index upkeep)
+                int64_t q = p + 1;
+                int64_t qe = p + m - 3;
+                for (; q < qe; q += 13)
+                    e->htab[pz_hash32(pz_read32(src + q))] = (int32_t)q;
+                p += m;
+                lit_start = p;
+            } else {
+                p++;
+            }
+        }
+        int64_t tail_lits = bp + bn - lit_start;
+        lit_total += tail_lits;
+        // ---- size the compressed form: lit header (<=3) + lits + seq
+        // header (<=3) + modes 1 + bitstream (bounded below); fall back to
+        // a RAW block unless compressed is smaller
+        // literals section header: raw literals, size_format by size
+        uint8_t lhdr[3];
+        int lhn;
+        if (lit_total < 32) {
+            lhdr[0] = (uint8_t)(lit_total << 3);
+            lhn = 1;
+        } else if (lit_total < 4096) {
+            lhdr[0] = (uint8_t)(((lit_total & 0xF) << 4) | (1 << 2));
+            lhdr[1] = (uint8_t)(lit_total >> 4);
+            lhn = 2;
+        } else {
+            lhdr[0] = (uint8_t)(((lit_total & 0xF) << 4) | (3 << 2));
+            lhdr[1] = (uint8_t)(lit_total >> 4);
+            lhdr[2] = (uint8_t)(lit_total >> 12);
+            lhn = 3;
+        }
+        // assemble into a bounded region after the 3-byte block header;
+        // if anything overflows the raw size, emit RAW instead
+        int64_t bh_pos = dp;
+        if (dp + 3 > cap) return PZ_ERR_DST_SMALL;
+        dp += 3;
+        int64_t body = dp;
+        int64_t raw_budget = bn;  // compressed must beat raw
+        int ok = 1;
+        if (body + lhn + lit_total + 4 - dp <= raw_budget &&
+            body + lhn + lit_total + 16 <= cap) {
+            // literals
+            for (int i = 0; i < lhn; i++) dst[dp++] = lhdr[i];
+            int64_t ls = bp;
+            for (int s = 0; s < nseq; s++) {
+                for (uint32_t i = 0; i < e->s_ll[s]; i++)
+                    dst[dp++] = src[ls + i];
+                ls += e->s_ll[s] + e->s_ml[s];
+            }
+            for (int64_t i = 0; i < tail_lits; i++)
+                dst[dp++] = src[lit_start + i];
+            // sequences header
+            if (nseq < 128) {
+                dst[dp++] = (uint8_t)nseq;
+            } else if (nseq < 0x7F00) {
+                dst[dp++] = (uint8_t)((nseq >> 8) + 128);
+                dst[dp++] = (uint8_t)nseq;
+            } else {
+                dst[dp++] = 255;
+                dst[dp++] = (uint8_t)(nseq - 0x7F00);
+                dst[dp++] = (uint8_t)((nseq - 0x7F00) >> 8);
+            }
+            if (nseq > 0) {
+                dst[dp++] = 0;  // modes: predefined / predefined / predef
+                PzBitW w;
+                pzw_init(&w, dst + dp, cap - dp);
+                int lastq = nseq - 1;
+                int llc = pz_ll_code(e->s_ll[lastq]);
+                int mlc = pz_ml_code(e->s_ml[lastq]);
+                uint32_t ofb = e->s_of[lastq] + 3;
+                int ofc = pz_highbit(ofb);
+                PzFseCState sml, sof, sll;
+                pz_fse_cinit(&sml, &e->ct_ml, mlc);
+                pz_fse_cinit(&sof, &e->ct_of, ofc);
+                pz_fse_cinit(&sll, &e->ct_ll, llc);
+                rc = 0;
+                rc |= pzw_add(&w, e->s_ll[lastq] - pz_ll_base(llc),
+                              pz_ll_bits(llc));
+                rc |= pzw_add(&w, e->s_ml[lastq] - pz_ml_base(mlc),
+                              pz_ml_bits(mlc));
+                rc |= pzw_add(&w, ofb - (1u << ofc), ofc);
+                for (int s = nseq - 2; s >= 0 && rc == 0; s--) {
+                    llc = pz_ll_code(e->s_ll[s]);
+                    mlc = pz_ml_code(e->s_ml[s]);
+                    ofb = e->s_of[s] + 3;
+                    ofc = pz_highbit(ofb);
+                    rc |= pz_fse_cenc(&w, &sof, &e->ct_of, ofc);
+                    rc |= pz_fse_cenc(&w, &sml, &e->ct_ml, mlc);
+                    rc |= pz_fse_cenc(&w, &sll, &e->ct_ll, llc);
+                    rc |= pzw_add(&w, e->s_ll[s] - pz_ll_base(llc),
+                                  pz_ll_bits(llc));
+                    rc |= pzw_add(&w, e->s_ml[s] - pz_ml_base(mlc),
+                                  pz_ml_bits(mlc));
+                    rc |= pzw_add(&w, ofb - (1u << ofc), ofc);
+                }
+                rc |= pz_fse_cflush(&w, &sml, &e->ct_ml);
+                rc |= pz_fse_cflush(&w, &sof, &e->ct_of);
+                rc |= pz_fse_cflush(&w, &sll, &e->ct_ll);
+                int64_t wbytes = rc == 0 ? pzw_close(&w) : -1;
+                if (wbytes < 0) {
+                    ok = 0;
+                } else {
+                    dp += wbytes;
+                }
+            }
+            if (ok && dp - body < raw_budget) {
+                uint32_t bh = (uint32_t)(last ? 1 : 0) | (2u << 1) |
+                              ((uint32_t)(dp - body) << 3);
+                dst[bh_pos] = (uint8_t)bh;
+                dst[bh_pos + 1] = (uint8_t)(bh >> 8);
+                dst[bh_pos + 2] = (uint8_t)(bh >> 16);
+            } else {
+                ok = 0;
+            }
+        } else {
+            ok = 0;
+        }
+        if (!ok) {  // RAW block
+            dp = body;
+            if (dp + bn > cap) return PZ_ERR_DST_SMALL;
+            for (int64_t i = 0; i < bn; i++) dst[dp + i] = src[bp + i];
+            dp += bn;
+            uint32_t bh =
+                (uint32_t)(last ? 1 : 0) | (0u << 1) | ((uint32_t)bn << 3);
+            dst[bh_pos] = (uint8_t)bh;
+            dst[bh_pos + 1] = (uint8_t)(bh >> 8);
+            dst[bh_pos + 2] = (uint8_t)(bh >> 16);
+        }
+        bp += bn;
+    }
+    return dp;
+}
