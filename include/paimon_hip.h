@@ -192,6 +192,10 @@ int64_t pmh_debug_zstd_gpu(const void *src, int64_t n, void *dst,
 int64_t pmh_debug_zstd_enc_cpu(const void *src, int64_t n, void *dst,
                                int64_t cap);
 
+/* Same, on the GPU (k_zstd_compress batch of 1). */
+int64_t pmh_debug_zstd_enc_gpu(const void *src, int64_t n, void *dst,
+                               int64_t cap);
+
 /* Parse one deletion vector from a DV index file slice (DeletionFile
  * {path, offset, length}; BitmapDeletionVector.java:98-112 wrapper around
  * the portable Roaring serialization). Writes up to `cap` deleted

@@ -170,6 +170,13 @@ struct ZstdJob {
     uint64_t src_off, dst_off;  // into the batch src / dst blobs
     uint32_t src_len, dst_len;
 };
+// on-GPU zstd page COMPRESSION (k_zstd_compress): one job per page;
+// scratch holds n * sizeof(PzEnc) bytes; status[j] = compressed size or
+// a PZ_ERR_* code.
+hipError_t pmh_launch_zstd_compress(const uint8_t *src,
+                                    const struct ZstdJob *jobs, int n,
+                                    uint8_t *dst, uint8_t *scratch,
+                                    int64_t *status, hipStream_t stream);
 hipError_t pmh_launch_zstd_pages(const uint8_t *src,
                                  const struct ZstdJob *jobs, int n,
                                  uint8_t *dst, uint8_t *scratch,

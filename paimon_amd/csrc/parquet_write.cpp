@@ -285,12 +285,16 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
         for (const auto &c : cols) {
             CcInfo cc{};
             cc.num_values = rg1 - rg0;
+            cc.data_page_offset = -1;  // set at the first data page
             int64_t unc_total = 0;
             const bool dict = c.dtype == 7;
             const int enc = dict ? ENC_RLE_DICTIONARY : ENC_PLAIN;
+            // pass 1: build every page payload of this chunk (dictionary
+            // page first), so ZSTD chunks can compress as ONE GPU batch
+            // (k_zstd_compress — §8f.1; host libzstd is the fallback)
+            std::vector<std::string> payloads;
+            std::vector<int64_t> prows;  // -1 marks the dictionary page
             if (dict) {
-                // dictionary page: PLAIN byte-array entries ([len][bytes])
-                cc.dict_page_offset = (int64_t)buf.size();
                 std::string dp;
                 for (int32_t i = 0; i < c.dict_len; i++) {
                     uint32_t len =
@@ -299,31 +303,9 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
                     dp.append(
                         (const char *)c.dict_data + c.dict_offsets[i], len);
                 }
-                // dictionary pages compress with the chunk codec, like
-                // data pages (parquet spec)
-                if (codec == CODEC_ZSTD) {
-                    std::vector<uint8_t> compd;
-                    if (!zstd_compress_buf((const uint8_t *)dp.data(),
-                                           dp.size(), compd, err)) {
-                        fclose(f);
-                        return false;
-                    }
-                    size_t h0 = buf.size();
-                    dict_page_header(c.dict_len, (int32_t)dp.size(),
-                                     (int32_t)compd.size(), buf);
-                    unc_total +=
-                        (int64_t)(buf.size() - h0) + (int64_t)dp.size();
-                    buf.append((const char *)compd.data(), compd.size());
-                } else {
-                    size_t h0 = buf.size();
-                    dict_page_header(c.dict_len, (int32_t)dp.size(),
-                                     (int32_t)dp.size(), buf);
-                    unc_total +=
-                        (int64_t)(buf.size() - h0) + (int64_t)dp.size();
-                    buf.append(dp);
-                }
+                payloads.push_back(std::move(dp));
+                prows.push_back(-1);
             }
-            cc.data_page_offset = (int64_t)buf.size();
             for (int64_t p0 = rg0; p0 < rg1 || (rg1 == rg0 && p0 == rg0);
                  p0 += page_rows) {
                 int64_t p1 = p0 + page_rows < rg1 ? p0 + page_rows : rg1;
@@ -331,28 +313,50 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
                 if (c.valid) encode_def_levels(c.valid + p0, p1 - p0, payload);
                 if (dict) encode_dict_ids(c, p0, p1, payload);
                 else encode_values(c, p0, p1, payload);
-                if (codec == CODEC_ZSTD) {
-                    std::vector<uint8_t> comp;
-                    if (!zstd_compress_buf((const uint8_t *)payload.data(),
-                                           payload.size(), comp, err)) {
-                        fclose(f);
-                        return false;
-                    }
-                    size_t h0 = buf.size();
-                    page_header(p1 - p0, (int32_t)payload.size(),
-                                (int32_t)comp.size(), enc, buf);
-                    unc_total +=
-                        (int64_t)(buf.size() - h0) + (int64_t)payload.size();
-                    buf.append((const char *)comp.data(), comp.size());
-                } else {
-                    size_t h0 = buf.size();
-                    page_header(p1 - p0, (int32_t)payload.size(),
-                                (int32_t)payload.size(), enc, buf);
-                    unc_total +=
-                        (int64_t)(buf.size() - h0) + (int64_t)payload.size();
-                    buf.append(payload);
-                }
+                payloads.push_back(std::move(payload));
+                prows.push_back(p1 - p0);
                 if (rg1 == rg0) break;  // single empty page for 0 rows
+            }
+            // pass 2: compress (ZSTD) — GPU batch, host fallback
+            std::vector<std::vector<uint8_t>> comp;
+            if (codec == CODEC_ZSTD) {
+                if (!pw_gpu_zstd_compress(payloads, comp)) {
+                    comp.resize(payloads.size());
+                    for (size_t i = 0; i < payloads.size(); i++) {
+                        if (!zstd_compress_buf(
+                                (const uint8_t *)payloads[i].data(),
+                                payloads[i].size(), comp[i], err)) {
+                            fclose(f);
+                            return false;
+                        }
+                    }
+                }
+            }
+            // pass 3: assemble headers + page bytes
+            for (size_t i = 0; i < payloads.size(); i++) {
+                const std::string &payload = payloads[i];
+                const bool is_dict = prows[i] < 0;
+                int32_t on_file = codec == CODEC_ZSTD
+                                      ? (int32_t)comp[i].size()
+                                      : (int32_t)payload.size();
+                if (is_dict)
+                    cc.dict_page_offset = (int64_t)buf.size();
+                else if (cc.data_page_offset < 0)
+                    cc.data_page_offset = (int64_t)buf.size();
+                size_t h0 = buf.size();
+                if (is_dict)
+                    dict_page_header(c.dict_len, (int32_t)payload.size(),
+                                     on_file, buf);
+                else
+                    page_header(prows[i], (int32_t)payload.size(), on_file,
+                                enc, buf);
+                unc_total +=
+                    (int64_t)(buf.size() - h0) + (int64_t)payload.size();
+                if (codec == CODEC_ZSTD)
+                    buf.append((const char *)comp[i].data(),
+                               comp[i].size());
+                else
+                    buf.append(payload);
             }
             cc.total_size = (int64_t)buf.size() -
                             (dict ? cc.dict_page_offset

@@ -29,3 +29,10 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
                    int64_t page_rows, int codec, std::string &err);
 
 }  // namespace pmh
+
+namespace pmh {
+// GPU batch page compression (implemented in plan.cpp over
+// k_zstd_compress); false = use the host codec.
+bool pw_gpu_zstd_compress(const std::vector<std::string> &payloads,
+                          std::vector<std::vector<uint8_t>> &outs);
+}  // namespace pmh

@@ -144,6 +144,77 @@ static bool gpu_zstd_batch(const uint8_t *src_base, int64_t src_total,
 }
 
 
+// GPU batch page compression for the parquet write-back (k_zstd_compress).
+// Returns false (no set_error) to fall back to the host codec.
+bool pw_gpu_zstd_compress(const std::vector<std::string> &payloads,
+                          std::vector<std::vector<uint8_t>> &outs) {
+    if (!gpu_zstd_enabled() || payloads.empty()) return false;
+    int n = (int)payloads.size();
+    int64_t src_total = 0, dst_total = 0;
+    std::vector<ZstdJob> jobs(n);
+    for (int i = 0; i < n; i++) {
+        int64_t sz = (int64_t)payloads[i].size();
+        int64_t cap = sz + (sz >> 8) + 1024;
+        jobs[i] = {(uint64_t)src_total, (uint64_t)dst_total, (uint32_t)sz,
+                   (uint32_t)cap};
+        src_total += sz;
+        dst_total += cap;
+    }
+    std::vector<uint8_t> src_host(src_total ? src_total : 1);
+    {
+        int64_t o = 0;
+        for (int i = 0; i < n; i++) {
+            memcpy(src_host.data() + o, payloads[i].data(),
+                   payloads[i].size());
+            o += (int64_t)payloads[i].size();
+        }
+    }
+    uint8_t *d_src = nullptr, *d_dst = nullptr, *d_scr = nullptr;
+    ZstdJob *d_jobs = nullptr;
+    int64_t *d_st = nullptr;
+    bool ok = false;
+    std::vector<int64_t> st(n);
+    do {
+        if (hipMalloc(&d_src, src_host.size()) != hipSuccess) break;
+        if (hipMalloc(&d_dst, dst_total ? dst_total : 1) != hipSuccess)
+            break;
+        if (hipMalloc(&d_scr, (size_t)n * sizeof(PzEnc)) != hipSuccess)
+            break;
+        if (hipMalloc(&d_jobs, n * sizeof(ZstdJob)) != hipSuccess) break;
+        if (hipMalloc(&d_st, n * 8) != hipSuccess) break;
+        if (hipMemcpy(d_src, src_host.data(), src_host.size(),
+                      hipMemcpyHostToDevice) != hipSuccess)
+            break;
+        if (hipMemcpy(d_jobs, jobs.data(), n * sizeof(ZstdJob),
+                      hipMemcpyHostToDevice) != hipSuccess)
+            break;
+        if (pmh_launch_zstd_compress(d_src, d_jobs, n, d_dst, d_scr, d_st,
+                                     nullptr) != hipSuccess)
+            break;
+        if (hipMemcpy(st.data(), d_st, n * 8, hipMemcpyDeviceToHost) !=
+            hipSuccess)
+            break;
+        bool all = true;
+        for (int i = 0; i < n; i++)
+            if (st[i] <= 0) all = false;
+        if (!all) break;
+        outs.resize(n);
+        ok = true;
+        for (int i = 0; i < n && ok; i++) {
+            outs[i].resize(st[i]);
+            if (hipMemcpy(outs[i].data(), d_dst + jobs[i].dst_off, st[i],
+                          hipMemcpyDeviceToHost) != hipSuccess)
+                ok = false;
+        }
+    } while (0);
+    if (d_src) (void)hipFree(d_src);
+    if (d_dst) (void)hipFree(d_dst);
+    if (d_scr) (void)hipFree(d_scr);
+    if (d_jobs) (void)hipFree(d_jobs);
+    if (d_st) (void)hipFree(d_st);
+    return ok;
+}
+
 // ---------------------------------------------------- deletion vectors
 //
 // Paimon deletion vectors: per data file, a RoaringBitmap32
@@ -3709,6 +3780,21 @@ int64_t pmh_debug_zstd_enc_cpu(const void *src, int64_t n, void *dst,
                                 cap, e.data());
     if (r < 0) set_error("pz_encode_frame: error %lld", (long long)r);
     return r;
+}
+
+// GPU round trip of the zstd COMPRESS kernel (batch of 1).
+int64_t pmh_debug_zstd_enc_gpu(const void *src, int64_t n, void *dst,
+                               int64_t cap) {
+    std::vector<std::string> in(1);
+    in[0].assign((const char *)src, (size_t)n);
+    std::vector<std::vector<uint8_t>> outs;
+    if (!pw_gpu_zstd_compress(in, outs)) {
+        set_error("gpu zstd compress failed (batch of 1)");
+        return -1;
+    }
+    if ((int64_t)outs[0].size() > cap) return PZ_ERR_DST_SMALL;
+    memcpy(dst, outs[0].data(), outs[0].size());
+    return (int64_t)outs[0].size();
 }
 
 int64_t pmh_debug_snappy(const void *src, int64_t n, void *dst, int64_t cap) {

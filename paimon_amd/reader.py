@@ -254,6 +254,23 @@ def debug_zstd_enc_cpu(data, cap=None):
     return bytes(out[:n])
 
 
+def debug_zstd_enc_gpu(data, cap=None):
+    """Compress one zstd frame ON THE GPU (k_zstd_compress batch of 1)."""
+    lib = load_lib()
+    lib.pmh_debug_zstd_enc_gpu.restype = ctypes.c_int64
+    lib.pmh_debug_zstd_enc_gpu.argtypes = [ctypes.c_void_p, ctypes.c_int64,
+                                           ctypes.c_void_p, ctypes.c_int64]
+    data = bytes(data)
+    if cap is None:
+        cap = len(data) + (len(data) >> 8) + 1024
+    out = np.empty(cap, dtype=np.uint8)
+    n = lib.pmh_debug_zstd_enc_gpu(data, len(data),
+                                   out.ctypes.data_as(ctypes.c_void_p), cap)
+    if n < 0:
+        raise RuntimeError(last_error())
+    return bytes(out[:n])
+
+
 def interval_partition(min_keys, max_keys):
     """Returns (section_id, run_id) per input file, per the IntervalPartition
     restatement in libpaimon_hip (plan.cpp)."""
