@@ -3151,15 +3151,19 @@ __global__ void k_partition_seed(const int64_t *lens, int k,
 __global__ void k_zstd_compress(const uint8_t *src, const ZstdJob *jobs,
                                 int n, uint8_t *dst, uint8_t *scratch,
                                 int64_t *status) {
+    // job = one 128 KB zstd BLOCK (matches are block-confined, so blocks
+    // compress independently — the host stitches them into frames);
+    // dst_len bit 31 carries the last-block flag
     const int wave = (int)((blockIdx.x * blockDim.x + threadIdx.x) >> 6);
     const int lane = threadIdx.x & 63;
     const int waves = (int)((gridDim.x * blockDim.x) >> 6);
     for (int j = wave; j < n; j += waves) {
         if (lane != 0) continue;
         PzEnc *e = (PzEnc *)(scratch + (size_t)j * sizeof(PzEnc));
-        status[j] = pz_encode_frame(src + jobs[j].src_off, jobs[j].src_len,
-                                    dst + jobs[j].dst_off, jobs[j].dst_len,
-                                    e);
+        int last = (int)(jobs[j].dst_len >> 31);
+        int64_t cap = jobs[j].dst_len & 0x7FFFFFFF;
+        status[j] = pz_encode_block(src + jobs[j].src_off, jobs[j].src_len,
+                                    last, dst + jobs[j].dst_off, cap, e);
     }
 }
 

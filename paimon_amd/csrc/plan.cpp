@@ -149,21 +149,40 @@ static bool gpu_zstd_batch(const uint8_t *src_base, int64_t src_total,
 bool pw_gpu_zstd_compress(const std::vector<std::string> &payloads,
                           std::vector<std::vector<uint8_t>> &outs) {
     if (!gpu_zstd_enabled() || payloads.empty()) return false;
-    int n = (int)payloads.size();
+    // one job per 128 KB BLOCK (pages can be multi-MB: block jobs keep
+    // thousands of waves busy); the host stitches blocks into frames
+    struct BlockRef {
+        int page;
+        int64_t dst_off;
+    };
+    std::vector<ZstdJob> jobs;
+    std::vector<BlockRef> refs;
     int64_t src_total = 0, dst_total = 0;
-    std::vector<ZstdJob> jobs(n);
-    for (int i = 0; i < n; i++) {
+    for (size_t i = 0; i < payloads.size(); i++) {
         int64_t sz = (int64_t)payloads[i].size();
-        int64_t cap = sz + (sz >> 8) + 1024;
-        jobs[i] = {(uint64_t)src_total, (uint64_t)dst_total, (uint32_t)sz,
-                   (uint32_t)cap};
+        int64_t off = 0;
+        while (off < sz) {
+            int64_t bn = sz - off < PZ_BLOCK_MAX ? sz - off : PZ_BLOCK_MAX;
+            int last = off + bn >= sz;
+            int64_t cap = bn + (bn >> 8) + 256;
+            ZstdJob j{(uint64_t)(src_total + off), (uint64_t)dst_total,
+                      (uint32_t)bn,
+                      (uint32_t)cap | (last ? 0x80000000u : 0u)};
+            jobs.push_back(j);
+            refs.push_back({(int)i, dst_total});
+            dst_total += cap;
+            off += bn;
+        }
         src_total += sz;
-        dst_total += cap;
+    }
+    int n = (int)jobs.size();
+    if (n == 0) {  // all payloads empty: frames assemble host-side below
+        n = 0;
     }
     std::vector<uint8_t> src_host(src_total ? src_total : 1);
     {
         int64_t o = 0;
-        for (int i = 0; i < n; i++) {
+        for (size_t i = 0; i < payloads.size(); i++) {
             memcpy(src_host.data() + o, payloads[i].data(),
                    payloads[i].size());
             o += (int64_t)payloads[i].size();
@@ -198,14 +217,28 @@ bool pw_gpu_zstd_compress(const std::vector<std::string> &payloads,
         for (int i = 0; i < n; i++)
             if (st[i] <= 0) all = false;
         if (!all) break;
-        outs.resize(n);
-        ok = true;
-        for (int i = 0; i < n && ok; i++) {
-            outs[i].resize(st[i]);
-            if (hipMemcpy(outs[i].data(), d_dst + jobs[i].dst_off, st[i],
-                          hipMemcpyDeviceToHost) != hipSuccess)
-                ok = false;
+        // stitch blocks into frames: header + concatenated block outputs
+        std::vector<uint8_t> blob(dst_total ? dst_total : 1);
+        if (dst_total &&
+            hipMemcpy(blob.data(), d_dst, dst_total,
+                      hipMemcpyDeviceToHost) != hipSuccess)
+            break;
+        outs.assign(payloads.size(), {});
+        for (size_t i = 0; i < payloads.size(); i++) {
+            uint8_t hdr[16];
+            int hn = pz_frame_header(hdr, (int64_t)payloads[i].size());
+            outs[i].insert(outs[i].end(), hdr, hdr + hn);
+            if (payloads[i].empty()) {
+                uint8_t raw0[3] = {1, 0, 0};  // empty raw last block
+                outs[i].insert(outs[i].end(), raw0, raw0 + 3);
+            }
         }
+        for (int j = 0; j < n; j++) {
+            auto &o = outs[refs[j].page];
+            o.insert(o.end(), blob.data() + refs[j].dst_off,
+                     blob.data() + refs[j].dst_off + st[j]);
+        }
+        ok = true;
     } while (0);
     if (d_src) (void)hipFree(d_src);
     if (d_dst) (void)hipFree(d_dst);

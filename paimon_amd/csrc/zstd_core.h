@@ -1016,24 +1016,12 @@ PZHD uint32_t pz_hash32(uint32_t v) {
     return (v * 2654435761u) >> (32 - PZ_ENC_HLOG);
 }
 
-// Encode one frame: single-segment, 4-byte content size, blocks of
-// <= PZ_BLOCK_MAX. Returns compressed size or PZ_ERR_*. Worst case output
-// is n + 3 bytes/block + 9 header bytes — pass cap >= n + (n>>10) + 64.
-PZHD int64_t pz_encode_frame(const uint8_t *src, int64_t n, uint8_t *dst,
-                             int64_t cap, PzEnc *e) {
-    // frame header: magic, FHD = single_segment | fcs_flag 2 (4-byte FCS)
-    if (cap < 16) return PZ_ERR_DST_SMALL;
-    int64_t dp = 0;
-    dst[dp++] = 0x28;
-    dst[dp++] = 0xB5;
-    dst[dp++] = 0x2F;
-    dst[dp++] = 0xFD;
-    dst[dp++] = (2 << 6) | (1 << 5);  // FCS 4 bytes, single segment
-    dst[dp++] = (uint8_t)n;
-    dst[dp++] = (uint8_t)(n >> 8);
-    dst[dp++] = (uint8_t)(n >> 16);
-    dst[dp++] = (uint8_t)(n >> 24);
-    // build the predefined encode tables once per frame
+// Encode ONE block (3-byte header + body) of <= PZ_BLOCK_MAX input bytes.
+// Matches are confined to the block (its own hash table), so blocks
+// compress INDEPENDENTLY — the GPU kernel runs one wave per block and the
+// host concatenates them into a frame. Returns block bytes or PZ_ERR_*.
+PZHD int64_t pz_encode_block(const uint8_t *src, int64_t bn, int last,
+                             uint8_t *dst, int64_t cap, PzEnc *e) {
     int nsym, al, rc;
     pz_fse_predef(0, e->norm, &nsym, &al);
     if ((rc = pz_fse_build_ctable(e->norm, nsym, al, &e->ct_ll)) < 0)
@@ -1045,18 +1033,9 @@ PZHD int64_t pz_encode_frame(const uint8_t *src, int64_t n, uint8_t *dst,
     if ((rc = pz_fse_build_ctable(e->norm, nsym, al, &e->ct_ml)) < 0)
         return rc;
     for (int i = 0; i < (1 << PZ_ENC_HLOG); i++) e->htab[i] = -1;
-    int64_t bp = 0;  // block start
-    if (n == 0) {
-        // empty frame: one empty RAW last block
-        if (dp + 3 > cap) return PZ_ERR_DST_SMALL;
-        dst[dp++] = 1;  // last=1, type raw, size 0
-        dst[dp++] = 0;
-        dst[dp++] = 0;
-        return dp;
-    }
-    while (bp < n) {
-        int64_t bn = n - bp < PZ_BLOCK_MAX ? n - bp : PZ_BLOCK_MAX;
-        int last = bp + bn >= n;
+    const int64_t bp = 0;
+    int64_t dp = 0;
+    {
         // ---- greedy match pass over [bp, bp+bn)
         int nseq = 0;
         int64_t lit_start = bp;   // start of pending literals
@@ -1210,6 +1189,46 @@ PZHD int64_t pz_encode_frame(const uint8_t *src, int64_t n, uint8_t *dst,
             dst[bh_pos + 1] = (uint8_t)(bh >> 8);
             dst[bh_pos + 2] = (uint8_t)(bh >> 16);
         }
+    }
+    return dp;
+}
+
+// frame header for an n-byte single-segment frame (magic + FHD + 4-byte
+// content size); returns header bytes
+PZHD int pz_frame_header(uint8_t *dst, int64_t n) {
+    dst[0] = 0x28;
+    dst[1] = 0xB5;
+    dst[2] = 0x2F;
+    dst[3] = 0xFD;
+    dst[4] = (2 << 6) | (1 << 5);  // FCS 4 bytes, single segment
+    dst[5] = (uint8_t)n;
+    dst[6] = (uint8_t)(n >> 8);
+    dst[7] = (uint8_t)(n >> 16);
+    dst[8] = (uint8_t)(n >> 24);
+    return 9;
+}
+
+// Encode one frame serially (host path / tests): header + independent
+// blocks. Worst case output n + (n >> 10) + 64.
+PZHD int64_t pz_encode_frame(const uint8_t *src, int64_t n, uint8_t *dst,
+                             int64_t cap, PzEnc *e) {
+    if (cap < 16) return PZ_ERR_DST_SMALL;
+    int64_t dp = pz_frame_header(dst, n);
+    if (n == 0) {
+        if (dp + 3 > cap) return PZ_ERR_DST_SMALL;
+        dst[dp++] = 1;  // last, raw, size 0
+        dst[dp++] = 0;
+        dst[dp++] = 0;
+        return dp;
+    }
+    int64_t bp = 0;
+    while (bp < n) {
+        int64_t bn = n - bp < PZ_BLOCK_MAX ? n - bp : PZ_BLOCK_MAX;
+        int last = bp + bn >= n;
+        int64_t r = pz_encode_block(src + bp, bn, last, dst + dp, cap - dp,
+                                    e);
+        if (r < 0) return r;
+        dp += r;
         bp += bn;
     }
     return dp;
