@@ -3154,12 +3154,15 @@ __global__ void k_zstd_compress(const uint8_t *src, const ZstdJob *jobs,
     // job = one 128 KB zstd BLOCK (matches are block-confined, so blocks
     // compress independently — the host stitches them into frames);
     // dst_len bit 31 carries the last-block flag
+    __shared__ int32_t lhtab[4][1 << PZ_ENC_HLOG];  // 16 KB per wave
     const int wave = (int)((blockIdx.x * blockDim.x + threadIdx.x) >> 6);
+    const int wiw = (threadIdx.x >> 6) & 3;
     const int lane = threadIdx.x & 63;
     const int waves = (int)((gridDim.x * blockDim.x) >> 6);
     for (int j = wave; j < n; j += waves) {
         if (lane != 0) continue;
         PzEnc *e = (PzEnc *)(scratch + (size_t)j * sizeof(PzEnc));
+        e->htab = lhtab[wiw];
         int last = (int)(jobs[j].dst_len >> 31);
         int64_t cap = jobs[j].dst_len & 0x7FFFFFFF;
         status[j] = pz_encode_block(src + jobs[j].src_off, jobs[j].src_len,

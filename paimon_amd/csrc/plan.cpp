@@ -3809,6 +3809,8 @@ int64_t pmh_debug_zstd_gpu(const void *src, int64_t n, void *dst,
 int64_t pmh_debug_zstd_enc_cpu(const void *src, int64_t n, void *dst,
                                int64_t cap) {
     std::vector<PzEnc> e(1);
+    std::vector<int32_t> ht(1 << PZ_ENC_HLOG);
+    e[0].htab = ht.data();
     int64_t r = pz_encode_frame((const uint8_t *)src, n, (uint8_t *)dst,
                                 cap, e.data());
     if (r < 0) set_error("pz_encode_frame: error %lld", (long long)r);

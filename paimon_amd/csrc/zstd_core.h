@@ -995,11 +995,14 @@ PZHD int pz_ml_code(uint32_t ml) {  // ml >= 3
     return 32;
 }
 
-// per-call encoder scratch (one per wave/job): hash table + sequence lists
-#define PZ_ENC_HLOG 14
+// per-call encoder scratch (one per wave/job): sequence lists + tables;
+// the hash table is caller-provided (the GPU kernel places it in LDS —
+// the match loop is a serial dependent-load chain and LDS latency is the
+// difference between ~0.3 and several MB/s per wave)
+#define PZ_ENC_HLOG 12
 #define PZ_ENC_MAXSEQ (PZ_BLOCK_MAX / 4 + 16)
 typedef struct {
-    int32_t htab[1 << PZ_ENC_HLOG];
+    int32_t *htab;  // [1 << PZ_ENC_HLOG]
     uint32_t s_ll[PZ_ENC_MAXSEQ];
     uint32_t s_ml[PZ_ENC_MAXSEQ];
     uint32_t s_of[PZ_ENC_MAXSEQ];
