@@ -320,7 +320,8 @@ bool write_parquet(const std::vector<PwCol> &cols, int64_t n_rows,
             // pass 2: compress (ZSTD) — GPU batch, host fallback
             std::vector<std::vector<uint8_t>> comp;
             if (codec == CODEC_ZSTD) {
-                if (!pw_gpu_zstd_compress(payloads, comp)) {
+                if (!pw_gpu_zstd_enc_enabled() ||
+                    !pw_gpu_zstd_compress(payloads, comp)) {
                     comp.resize(payloads.size());
                     for (size_t i = 0; i < payloads.size(); i++) {
                         if (!zstd_compress_buf(

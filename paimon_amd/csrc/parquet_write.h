@@ -35,4 +35,5 @@ namespace pmh {
 // k_zstd_compress); false = use the host codec.
 bool pw_gpu_zstd_compress(const std::vector<std::string> &payloads,
                           std::vector<std::vector<uint8_t>> &outs);
+bool pw_gpu_zstd_enc_enabled();  // PMH_GPU_ZSTD_ENC=1 opt-in
 }  // namespace pmh

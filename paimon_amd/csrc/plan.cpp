@@ -146,6 +146,18 @@ static bool gpu_zstd_batch(const uint8_t *src_base, int64_t src_total,
 
 // GPU batch page compression for the parquet write-back (k_zstd_compress).
 // Returns false (no set_error) to fall back to the host codec.
+// Opt-in (PMH_GPU_ZSTD_ENC=1): the v0 block compressor is spec-valid and
+// wave-parallel, but the measured write-side A/B (profiles/ r02 zenc_ab)
+// has host libzstd 4x faster AND ~2.5x smaller (huffman literals + a real
+// parser beat predefined-FSE + greedy matching on typical column data) —
+// so the HOST codec stays the write-side default. The debug entry and
+// tests exercise the GPU path regardless.
+bool pw_gpu_zstd_enc_enabled() {
+    // read per call (cheap; tests toggle it within one process)
+    const char *e = getenv("PMH_GPU_ZSTD_ENC");
+    return e && e[0] == '1' && gpu_zstd_enabled();
+}
+
 bool pw_gpu_zstd_compress(const std::vector<std::string> &payloads,
                           std::vector<std::vector<uint8_t>> &outs) {
     if (!gpu_zstd_enabled() || payloads.empty()) return false;

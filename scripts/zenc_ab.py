@@ -19,7 +19,7 @@ cols = [("_KEY_k", k), ("_SEQUENCE_NUMBER", k.copy()),
        [(f"v_c{i}", ((k // (i + 2)) % 97).astype(np.int32))
         for i in range(4)]
 for tag in ("gpu", "host"):
-    os.environ["PMH_GPU_ZSTD"] = "1" if tag == "gpu" else "0"
+    os.environ["PMH_GPU_ZSTD_ENC"] = "1" if tag == "gpu" else "0"
     # env read once per process; fork a child per leg
     pid = os.fork()
     if pid == 0:

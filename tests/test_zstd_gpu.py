@@ -1,6 +1,8 @@
 """GPU zstd page decoding (k_zstd_pages): device round trips vs libzstd
 frames, and the staging integration (zstd parquet chunks decode on the GPU —
 stats.gpu_zstd_pages counts them)."""
+import os
+
 import numpy as np
 import pyarrow as pa
 import pytest
@@ -91,10 +93,12 @@ class TestCompressKernel:
         self._rt(rng.integers(0, 256, 300_000, dtype=np.uint8).tobytes())
 
     def test_write_back_pyarrow_reads(self, tmp_path):
-        # write a zstd parquet file (pages compressed by the GPU batch
-        # path) and read it back with PYARROW — full interop proof
+        # write a zstd parquet file with pages compressed by the GPU batch
+        # path (PMH_GPU_ZSTD_ENC=1 opt-in — host libzstd is the measured
+        # write-side default) and read it back with PYARROW
         import pyarrow.parquet as pq
         from paimon_amd.reader import write_parquet
+        os.environ["PMH_GPU_ZSTD_ENC"] = "1"
         n = 300_000
         k = np.arange(n, dtype=np.int64)
         v = (k * 31 + 7).astype(np.int32)
