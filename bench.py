@@ -158,11 +158,16 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     n_gpus = world if world > 1 else args.gpus
     if world == 1 and args.gpus > 1:
-        # standalone multi-GPU invocation: run sequential replicas is wrong;
-        # the driver launches via torch.distributed.run. Run rank 0 only.
-        print("WARNING: --gpus>1 without torchrun; running 1 rank",
-              file=sys.stderr)
-        n_gpus = 1
+        # first-class launcher: re-exec under torch.distributed.run so a
+        # direct `bench.py --gpus N` measures N ranks (the driver's own
+        # torchrun invocation takes the world>1 path and never gets here)
+        import subprocess
+        cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+               f"--nproc-per-node={args.gpus}", "--master-addr=127.0.0.1",
+               "--master-port=29517", os.path.abspath(__file__)] +               sys.argv[1:]
+        print(f"[bench] relaunching {args.gpus} ranks via "
+              "torch.distributed.run", file=sys.stderr)
+        raise SystemExit(subprocess.call(cmd))
 
     import torch
     dist = None

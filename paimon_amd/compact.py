@@ -35,7 +35,7 @@ def _roll_slices(n, target_rows):
 def rewrite(session: Session, file_metas, key_cols, value_cols, out_dir,
             output_level, drop_delete=False, merge_engine="deduplicate",
             target_file_rows=20_000_000, compression="NONE",
-            file_prefix="compact"):
+            file_prefix="compact", schema_id=0):
     """Merge `file_metas` (one bucket's sorted runs) and write the result as
     rolling Parquet data files. Returns CompactResult-shaped dict:
     {"before": file_metas, "after": [DataFileMeta...]} with per-file stats
@@ -102,6 +102,39 @@ def rewrite(session: Session, file_metas, key_cols, value_cols, out_dir,
                 else:
                     min_key = [int(batch[kn][s]) for kn in key_names]
                     max_key = [int(batch[kn][e - 1]) for kn in key_names]
+                # per-column value stats (SimpleStats valueStats +
+                # nullCounts) and schemaId, completing the DataFileMeta
+                # surface (io/DataFileMeta.java:124-190)
+                vstats = {}
+                for name, arr in batch.items():
+                    if name in special or name.endswith("#valid") \
+                            or name.endswith("#dict"):
+                        continue
+                    valid = batch.get(name + "#valid")
+                    sl = arr[s:e]
+                    if valid is not None:
+                        lv = valid[s:e]
+                        nn = sl[lv]
+                        nulls = int(len(sl) - len(nn))
+                    else:
+                        nn = sl
+                        nulls = 0
+                    d = batch.get(name + "#dict")
+                    if d is not None:
+                        # string columns: min/max over the decoded strings
+                        vals = (d[nn] if len(nn) else
+                                np.empty(0, dtype=object))
+                        vstats[name] = {
+                            "min": (vals.min().decode()
+                                    if len(vals) else None),
+                            "max": (vals.max().decode()
+                                    if len(vals) else None),
+                            "nullCount": nulls}
+                    else:
+                        vstats[name] = {
+                            "min": (nn.min().item() if len(nn) else None),
+                            "max": (nn.max().item() if len(nn) else None),
+                            "nullCount": nulls}
                 after.append({
                     "path": path,
                     "fileName": os.path.basename(path),
@@ -114,6 +147,8 @@ def rewrite(session: Session, file_metas, key_cols, value_cols, out_dir,
                     "deleteRowCount": int(np.count_nonzero(
                         ~np.isin(kd, KIND_IS_ADD))),
                     "level": int(output_level),
+                    "schemaId": int(schema_id),
+                    "valueStats": vstats,
                 })
     finally:
         plan.close()
