@@ -145,6 +145,23 @@ hipError_t pmh_launch_delta_emit(const DeltaChunk *chunks, int64_t n_chunks,
 // Decode ORC RLEv2 / byte-RLE work chunks into a dense typed column
 // (int32 or int64 elements per chunk.out_esize). One wave per chunk.
 
+// value-filter term (single-run sections only — MergeFileSplitRead.java:
+// 227-239: value filters must not push into overlapping sections).
+// op: 0 eq, 1 ne, 2 lt, 3 le, 4 gt, 5 ge, 6 is_null, 7 is_not_null
+struct FilterTerm {
+    int32_t col;     // index into the plan's column list
+    int32_t op;
+    int64_t ilit;    // integer/string-id literal
+    double dlit;     // float literal
+    int32_t is_fp;   // compare as double
+    int32_t pad;
+};
+// mark rows FAILING the conjunction as dead in the tombstone array
+hipError_t pmh_launch_filter(const DevCol *cols, int n_cols,
+                             const struct FilterTerm *terms, int n_terms,
+                             int64_t rows, uint8_t *tomb,
+                             hipStream_t stream);
+
 // full-compaction changelog chain (FullChangelogMergeFunctionWrapper):
 // finalize compacts the merge pass's provisional per-group entries into
 // rows (evaluating row-deduplicate value equality); emit gathers them.

@@ -3170,6 +3170,63 @@ __global__ void k_zstd_compress(const uint8_t *src, const ZstdJob *jobs,
     }
 }
 
+// ------------------------------------------------------------- k_filter
+//
+// Value-filter pushdown for SINGLE-RUN sections (MergeFileSplitRead.java:
+// 227-239: each key appears once, so dropping rows is safe; overlapping
+// sections get no value filters). Failing rows mark the run's tombstone
+// byte — the same "never reaches the merge" mechanism deletion vectors
+// use. Comparison semantics follow LeafPredicate: a NULL field fails
+// every comparison; is_null/is_not_null test validity.
+__global__ void k_filter(const DevCol *cols, int n_cols,
+                         const FilterTerm *terms, int n_terms, int64_t rows,
+                         uint8_t *tomb) {
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i < rows; i += stride) {
+        bool pass = true;
+        for (int t = 0; t < n_terms && pass; t++) {
+            const FilterTerm &ft = terms[t];
+            const DevCol &dc = cols[ft.col];
+            bool valid =
+                !dc.valid0 || ((const uint8_t *)dc.valid0)[i] != 0;
+            if (ft.op == 6) {
+                pass = !valid;
+                continue;
+            }
+            if (ft.op == 7) {
+                pass = valid;
+                continue;
+            }
+            if (!valid) {
+                pass = false;  // NULL fails comparisons
+                continue;
+            }
+            int cmp;
+            if (ft.is_fp) {
+                double v = dc.esize == 8
+                               ? *(const double *)(dc.addr0 + i * 8)
+                               : (double)*(const float *)(dc.addr0 + i * 4);
+                cmp = v < ft.dlit ? -1 : (v > ft.dlit ? 1 : 0);
+            } else {
+                int64_t v = dc.esize == 8 ? col_load<int64_t>(dc, i)
+                                          : (int64_t)col_load<int32_t>(dc, i);
+                cmp = v < ft.ilit ? -1 : (v > ft.ilit ? 1 : 0);
+            }
+            switch (ft.op) {
+            case 0: pass = cmp == 0; break;
+            case 1: pass = cmp != 0; break;
+            case 2: pass = cmp < 0; break;
+            case 3: pass = cmp <= 0; break;
+            case 4: pass = cmp > 0; break;
+            case 5: pass = cmp >= 0; break;
+            default: pass = false; break;
+            }
+        }
+        if (!pass) tomb[i] = 1;
+    }
+}
+
 // ------------------------------------------------------------ k_cl_finalize
 //
 // Compact the provisional changelog entries into dense per-tile rows.
@@ -3959,6 +4016,17 @@ hipError_t pmh_launch_emit(const DevCol *cols, const uint8_t *col_dtype,
 }
 
 
+
+hipError_t pmh_launch_filter(const DevCol *cols, int n_cols,
+                             const FilterTerm *terms, int n_terms,
+                             int64_t rows, uint8_t *tomb,
+                             hipStream_t stream) {
+    int64_t want = (rows + 255) / 256;
+    int blocks = want < 4096 ? (int)(want ? want : 1) : 4096;
+    hipLaunchKernelGGL(k_filter, dim3(blocks), dim3(256), 0, stream, cols,
+                       n_cols, terms, n_terms, rows, tomb);
+    return hipGetLastError();
+}
 
 hipError_t pmh_launch_cl_finalize(const DevCol *cols,
                                   const uint8_t *col_dtype, int n_cols,

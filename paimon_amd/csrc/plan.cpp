@@ -655,6 +655,10 @@ struct pmh_plan_t {
     bool fsplit = false;     // split value emission (PMH_FSPLIT=1 A/B)
     bool agg = false;        // aggregation merge engine (uses PU member lists)
     uint8_t *col_agg_dev = nullptr;  // per-column PMH_AGG_* codes
+    // value filters (conjunction; applied to single-run sections only,
+    // MergeFileSplitRead.java:227-239)
+    std::vector<pmh::FilterTerm> filters;
+    pmh::FilterTerm *filters_dev = nullptr;
     uint8_t *hier_dtype_dev = nullptr;  // batch-pass dtype map (int8/16 ->
                                         // int32 so virtual runs hold the
                                         // STORED width)
@@ -2996,6 +3000,67 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
             return nullptr;
         }
         plan->host_output = j["output"].as_str("device") == "host";
+        {
+            // "filters": [{"field": "...", "op": "eq|ne|lt|le|gt|ge|
+            // is_null|is_not_null", "literal": x}] — a conjunction; the
+            // reference pushes VALUE filters only into single-run
+            // sections (overlapping runs would lose newer records,
+            // MergeFileSplitRead.java:227-239). Key filters prune by file
+            // stats before planning (caller side, as DataSplit already
+            // carries the pruned file list).
+            const Json &fj = j["filters"];
+            for (const auto &f : fj.arr) {
+                std::string field = f["field"].as_str("");
+                std::string op = f["op"].as_str("");
+                int idx = -1;
+                for (size_t c = 0; c < plan->cols.size(); c++)
+                    if (plan->cols[c].name == field) idx = (int)c;
+                if (idx < 0) {
+                    set_error("filters: unknown field '%s'", field.c_str());
+                    return nullptr;
+                }
+                static const char *ops[] = {"eq", "ne", "lt", "le",
+                                            "gt", "ge", "is_null",
+                                            "is_not_null"};
+                int opc = -1;
+                for (int o = 0; o < 8; o++)
+                    if (op == ops[o]) opc = o;
+                if (opc < 0) {
+                    set_error("filters: unknown op '%s'", op.c_str());
+                    return nullptr;
+                }
+                FilterTerm ft{};
+                ft.col = idx;
+                ft.op = opc;
+                int dt = plan->cols[idx].dtype;
+                if (dt == PMH_DT_STRING && opc < 6 && opc > 1) {
+                    set_error("filters: order comparisons on string "
+                              "columns are not supported (dictionary ids "
+                              "are unordered); eq/ne/is_null only");
+                    return nullptr;
+                }
+                ft.is_fp = (dt == PMH_DT_FLOAT32 || dt == PMH_DT_FLOAT64);
+                if (opc < 6) {
+                    if (ft.is_fp)
+                        ft.dlit = f["literal"].type == Json::NUM
+                                      ? f["literal"].num
+                                      : 0.0;
+                    else ft.ilit = f["literal"].as_i64(0);
+                }
+                plan->filters.push_back(ft);
+            }
+            if (!plan->filters.empty()) {
+                plan->filters_dev = (FilterTerm *)plan->bufs.alloc(
+                    plan->filters.size() * sizeof(FilterTerm));
+                if (!plan->filters_dev ||
+                    hipMemcpy(plan->filters_dev, plan->filters.data(),
+                              plan->filters.size() * sizeof(FilterTerm),
+                              hipMemcpyHostToDevice) != hipSuccess) {
+                    set_error("H2D filters failed");
+                    return nullptr;
+                }
+            }
+        }
 
         std::vector<FileDesc> files;
         int idx = 0;
@@ -3059,6 +3124,19 @@ pmh_plan_t *pmh_plan_create(pmh_session_t *s, const char *plan_json) {
                 if (!stage_run(plan.get(), sec_files[r], sec.runs[r]))
                     return nullptr;
                 run_rows += sec.runs[r].length;
+            }
+            if (!plan->filters.empty() && sec.runs.size() == 1 &&
+                !sec.runs[0].tomb && sec.runs[0].length > 0) {
+                // single-run section: value filters apply via the
+                // tombstone mechanism (k_filter marks failing rows)
+                sec.runs[0].tomb =
+                    (uint8_t *)plan->bufs.alloc(sec.runs[0].length);
+                if (!sec.runs[0].tomb ||
+                    hipMemset(sec.runs[0].tomb, 0,
+                              sec.runs[0].length) != hipSuccess) {
+                    set_error("filter tombstone allocation failed");
+                    return nullptr;
+                }
             }
             if (!build_section_descriptors(plan.get(), sec)) {
                 set_error("device allocation failed for section");
@@ -3370,6 +3448,16 @@ int64_t pmh_read_next(pmh_plan_t *p, pmh_batch *out) {
         }
     }
     (void)hipEventRecord(ev[1], st);
+    if (p->filters_dev && k == 1 && !sec.hier && sec.runs[0].tomb &&
+        sec.runs[0].length > 0) {
+        // value-filter pushdown: single-run sections only (overlapping
+        // sections would lose newer records — MergeFileSplitRead.java:
+        // 227-239); idempotent across repeats (OR into the tombstones)
+        hipError_t fe = pmh_launch_filter(
+            sec.all_cols, n_cols, p->filters_dev, (int)p->filters.size(),
+            sec.runs[0].length, sec.runs[0].tomb, st);
+        if (fe != hipSuccess) return fail("filter", fe);
+    }
     int64_t n_tiles_eff = sec.n_tiles;
     int64_t rows_eff = sec.total_rows;
     if (sec.hier) {

@@ -315,7 +315,7 @@ class MergeReadPlan:
                  remove_record_on_delete=False, sequence_groups=None,
                  ignore_retract=None, sequence_fields=None,
                  changelog_producer=None, changelog_row_dedup=False,
-                 max_level=None, sort_engine="loser-tree"):
+                 max_level=None, sort_engine="loser-tree", filters=None):
         self.lib = session.lib
         desc = {
             "key_cols": key_cols,
@@ -343,6 +343,11 @@ class MergeReadPlan:
             # fields.<name>.aggregate-function (CoreOptions FIELDS_PREFIX);
             # unnamed columns default to last_non_null_value
             desc["aggregations"] = dict(aggregations)
+        if filters:
+            # value-filter conjunction; pushed into SINGLE-RUN sections
+            # only (MergeFileSplitRead.java:227-239) — overlapping
+            # sections emit unfiltered, as the reference reader does
+            desc["filters"] = list(filters)
         if changelog_producer:
             # changelog-producer = full-compaction
             # (FullChangelogMergeFunctionWrapper; max_level = num-levels - 1)

@@ -1,0 +1,133 @@
+"""Value-filter pushdown (MergeFileSplitRead.java:227-239): filters apply
+ONLY to single-run sections (each key appears once, dropping rows is safe);
+overlapping sections emit unfiltered — the engine re-filters downstream."""
+import numpy as np
+import pytest
+
+from oracle import merge_dedup
+from paimon_amd import Session, MergeReadPlan, file_descs_from_metas
+from paimon_amd.datagen import gen_runs_dedup, write_runs
+
+pytestmark = pytest.mark.gpu
+
+KEY_COLS = [{"name": "_KEY_k", "type": "int64"}]
+
+
+def _value_cols(n):
+    return ([{"name": "v_k", "type": "int64"}] +
+            [{"name": f"v_c{i}", "type": "int32"} for i in range(n)])
+
+
+def _read_keys(plan, col="_KEY_k"):
+    out = []
+    while True:
+        b = plan.read_next()
+        if b is None:
+            break
+        out.append(b[col].copy())
+    return np.concatenate(out) if out else np.empty(0, np.int64)
+
+
+class TestValueFilters:
+    def test_single_run_filtered(self, tmp_path):
+        runs = gen_runs_dedup(1, 50_000, n_value_cols=2, seed=951)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        v = runs[0]["values"][1]
+        exp = runs[0]["key"][v >= 1000]
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(2),
+                               filters=[{"field": "v_c0", "op": "ge",
+                                         "literal": 1000}]) as plan:
+                got = _read_keys(plan)
+        assert (got == exp).all(), (len(got), len(exp))
+
+    def test_conjunction(self, tmp_path):
+        runs = gen_runs_dedup(1, 40_000, n_value_cols=2, seed=952)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        v0 = runs[0]["values"][1]
+        v1 = runs[0]["values"][2]
+        keep = (v0 >= 500) & (v1 < 2_000_000_000)
+        exp = runs[0]["key"][keep]
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(2),
+                               filters=[
+                                   {"field": "v_c0", "op": "ge",
+                                    "literal": 500},
+                                   {"field": "v_c1", "op": "lt",
+                                    "literal": 2_000_000_000}]) as plan:
+                got = _read_keys(plan)
+        assert (got == exp).all()
+
+    def test_overlapping_sections_unfiltered(self, tmp_path):
+        # 4 overlapping runs: the filter must NOT apply (reference comment:
+        # pushing value filters into overlapping runs loses records)
+        runs = gen_runs_dedup(4, 20_000, n_value_cols=2, seed=953,
+                              delete_frac=0.1)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        r, w = merge_dedup(runs)
+        exp = np.array([runs[a]["key"][b] for a, b in zip(r, w)], np.int64)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(2),
+                               filters=[{"field": "v_c0", "op": "ge",
+                                         "literal": 10**9}]) as plan:
+                got = _read_keys(plan)
+        assert (got == exp).all()  # unfiltered
+
+    def test_null_semantics(self, tmp_path):
+        import pyarrow as pa
+        import pyarrow.parquet as pq
+        n = 30_000
+        rng = np.random.default_rng(954)
+        k = np.arange(n, dtype=np.int64)
+        v = rng.integers(0, 1000, n).astype(np.int32)
+        mask = rng.random(n) < 0.3
+        tbl = pa.table({"_KEY_k": pa.array(k),
+                        "_SEQUENCE_NUMBER": pa.array(k),
+                        "_VALUE_KIND": pa.array(np.zeros(n, np.int8)),
+                        "v_c0": pa.array(v, mask=mask)})
+        path = str(tmp_path / "f.parquet")
+        pq.write_table(tbl, path, compression=None, use_dictionary=False,
+                       data_page_version="1.0", store_schema=False)
+        metas = [{"path": path, "rowCount": n, "minKey": 0,
+                  "maxKey": n - 1, "level": 0}]
+        vc = [{"name": "v_c0", "type": "int32"}]
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS, vc,
+                               filters=[{"field": "v_c0", "op": "ge",
+                                         "literal": 500}]) as plan:
+                got = _read_keys(plan)
+            exp = k[~mask & (v >= 500)]  # NULL fails comparisons
+            assert (got == exp).all()
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS, vc,
+                               filters=[{"field": "v_c0",
+                                         "op": "is_null"}]) as plan:
+                got = _read_keys(plan)
+            assert (got == k[mask]).all()
+
+    def test_seq_filter_and_key_col(self, tmp_path):
+        # filters may reference any plan column, including the key
+        runs = gen_runs_dedup(1, 20_000, n_value_cols=1, seed=955)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        kk = runs[0]["key"]
+        exp = kk[kk > int(kk[len(kk) // 2])]
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(1),
+                               filters=[{"field": "_KEY_k", "op": "gt",
+                                         "literal": int(kk[len(kk) // 2])}]
+                               ) as plan:
+                got = _read_keys(plan)
+        assert (got == exp).all()
+
+    def test_bad_filter_rejected(self, tmp_path):
+        runs = gen_runs_dedup(1, 1_000, n_value_cols=1, seed=956)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        with Session(0) as s:
+            with pytest.raises(RuntimeError, match="unknown field"):
+                MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                              _value_cols(1),
+                              filters=[{"field": "nope", "op": "eq",
+                                        "literal": 1}])
