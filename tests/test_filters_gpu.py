@@ -30,7 +30,8 @@ def _read_keys(plan, col="_KEY_k"):
 
 class TestValueFilters:
     def test_single_run_filtered(self, tmp_path):
-        runs = gen_runs_dedup(1, 50_000, n_value_cols=2, seed=951)
+        runs = gen_runs_dedup(1, 50_000, n_value_cols=2, seed=951,
+                              delete_frac=0.0)
         metas = write_runs(runs, str(tmp_path), compression="NONE")
         v = runs[0]["values"][1]
         exp = runs[0]["key"][v >= 1000]
@@ -43,7 +44,8 @@ class TestValueFilters:
         assert (got == exp).all(), (len(got), len(exp))
 
     def test_conjunction(self, tmp_path):
-        runs = gen_runs_dedup(1, 40_000, n_value_cols=2, seed=952)
+        runs = gen_runs_dedup(1, 40_000, n_value_cols=2, seed=952,
+                              delete_frac=0.0)
         metas = write_runs(runs, str(tmp_path), compression="NONE")
         v0 = runs[0]["values"][1]
         v1 = runs[0]["values"][2]
@@ -109,7 +111,8 @@ class TestValueFilters:
 
     def test_seq_filter_and_key_col(self, tmp_path):
         # filters may reference any plan column, including the key
-        runs = gen_runs_dedup(1, 20_000, n_value_cols=1, seed=955)
+        runs = gen_runs_dedup(1, 20_000, n_value_cols=1, seed=955,
+                              delete_frac=0.0)
         metas = write_runs(runs, str(tmp_path), compression="NONE")
         kk = runs[0]["key"]
         exp = kk[kk > int(kk[len(kk) // 2])]
