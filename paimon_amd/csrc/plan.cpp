@@ -950,6 +950,66 @@ static bool prescan_rle(const uint8_t *s, int64_t len, int bit_width,
 // DeltaChunk. Miniblocks holding no real values carry a width byte but no
 // data (their widths zero out in the packed u64 so device byte offsets
 // skip nothing); partially-filled miniblocks are stored in full.
+// Host DELTA_BINARY_PACKED decode of one complete stream (parquet
+// delta-encoding spec; the GPU path decodes these on device — this host
+// version serves DELTA_BYTE_ARRAY string staging, where values intern into
+// the global dictionary anyway). Returns bytes consumed or -1.
+static int64_t host_delta_i64(const uint8_t *pp, int64_t plen,
+                              std::vector<int64_t> &out) {
+    int64_t pos = 0;
+    auto uleb = [&](uint64_t *v) -> bool {
+        *v = 0;
+        int sh = 0;
+        for (;;) {
+            if (pos >= plen) return false;
+            uint8_t b = pp[pos++];
+            *v |= (uint64_t)(b & 0x7f) << sh;
+            if (!(b & 0x80)) return true;
+            sh += 7;
+        }
+    };
+    auto zz = [](uint64_t v) {
+        return (int64_t)(v >> 1) ^ -(int64_t)(v & 1);
+    };
+    uint64_t bs, mpb, total, zfirst;
+    if (!uleb(&bs) || !uleb(&mpb) || !uleb(&total) || !uleb(&zfirst))
+        return -1;
+    if (mpb == 0 || bs % mpb != 0) return -1;
+    const int64_t vpm = (int64_t)(bs / mpb);
+    out.clear();
+    out.reserve(total);
+    if (total == 0) return pos;
+    int64_t cur = zz(zfirst);
+    out.push_back(cur);
+    int64_t remaining = (int64_t)total - 1;
+    while (remaining > 0) {
+        uint64_t zmd;
+        if (!uleb(&zmd) || pos + (int64_t)mpb > plen) return -1;
+        int64_t mind = zz(zmd);
+        const uint8_t *widths = pp + pos;
+        pos += mpb;
+        for (uint64_t m = 0; m < mpb && remaining > 0; m++) {
+            int w = widths[m];
+            if (w > 64) return -1;
+            int64_t take = remaining < vpm ? remaining : vpm;
+            if (pos + (vpm * w + 7) / 8 > plen) return -1;
+            for (int64_t j = 0; j < take; j++) {
+                uint64_t d = 0;  // LSB-first bit-packed
+                for (int b = 0; b < w; b++) {
+                    int64_t bit = j * w + b;
+                    d |= (uint64_t)((pp[pos + (bit >> 3)] >> (bit & 7)) & 1)
+                         << b;
+                }
+                cur += mind + (int64_t)d;
+                out.push_back(cur);
+            }
+            pos += (vpm * w + 7) / 8;  // full miniblock advances
+            remaining -= take;
+        }
+    }
+    return pos;
+}
+
 static bool prescan_delta(const uint8_t *pp, int64_t plen, int64_t n_values,
                           int64_t out_row0, uint64_t dev_base,
                           int64_t rel_off, uint64_t out_addr, int out_esize,
@@ -2008,7 +2068,8 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     return false;
                 }
                 // classify data pages
-                bool has_plain = false, has_dict = false, has_delta = false;
+                bool has_plain = false, has_dict = false, has_delta = false,
+                     has_dba = false;
                 const uint8_t *dict_host = nullptr;
                 int64_t dict_count = 0;
                 for (size_t pi = 0; pi < cc.pages.size(); pi++) {
@@ -2023,6 +2084,9 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                         has_dict = true;
                     } else if (pg.encoding == ENC_DELTA_BINARY_PACKED) {
                         has_delta = true;
+                    } else if (pg.encoding == ENC_DELTA_BYTE_ARRAY &&
+                               cols[c].dtype == PMH_DT_STRING) {
+                        has_dba = true;
                     } else {
                         set_error("%s: unsupported encoding %d",
                                   fd.path.c_str(), pg.encoding);
@@ -2035,7 +2099,8 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                               fd.path.c_str(), cols[c].name.c_str());
                     return false;
                 }
-                if (has_delta && (has_plain || has_dict)) {
+                if ((has_delta || has_dba) &&
+                    (has_plain || has_dict || (has_delta && has_dba))) {
                     set_error("%s col %s: mixed DELTA/other pages in one "
                               "chunk not supported", fd.path.c_str(),
                               cols[c].name.c_str());
@@ -2066,7 +2131,105 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     }
                     vpos[pi] = pos;
                 }
-                if (has_delta) {
+                if (has_dba) {
+                    // DELTA_BYTE_ARRAY strings (VectorizedDeltaByteArray-
+                    // Reader.java): per page, prefix lengths (DELTA) +
+                    // suffix lengths (DELTA) + suffix bytes; values share
+                    // prefixes with their predecessor. They intern into
+                    // the plan-level GLOBAL dictionary at staging (the ids
+                    // are the device representation, like ORC DIRECT_V2).
+                    if (!plan->sdicts[c])
+                        plan->sdicts[c].reset(new StrDict());
+                    StrDict *sd = plan->sdicts[c].get();
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                        auto &pg = cc.pages[pi];
+                        if (pg.page_type != 0) continue;
+                        const uint8_t *pp = payload_base + ppo[pi];
+                        int64_t plen2 = (cc.codec != CODEC_UNCOMPRESSED
+                                             ? pg.uncompressed_size
+                                             : pg.compressed_size);
+                        int64_t row0 = chunk_row0 + pg.first_row;
+                        int64_t n_dense = pg.num_values;
+                        int64_t dense0 = row0;
+                        bool to_dense = false;
+                        if (max_def > 0) {
+                            uint32_t dl_len;
+                            memcpy(&dl_len, pp, 4);
+                            if (chunk_nulls) {
+                                rc.has_nulls = true;
+                                int64_t rel =
+                                    (int64_t)rc.levels_host.size();
+                                rc.levels_host.insert(rc.levels_host.end(),
+                                                      pp + 4,
+                                                      pp + 4 + dl_len);
+                                int64_t before = rc.dense_before;
+                                if (!prescan_def(pp + 4, dl_len,
+                                                 pg.num_values, row0, rel,
+                                                 &rc.dense_before,
+                                                 rc.def_host))
+                                    return false;
+                                n_dense = rc.dense_before - before;
+                                dense0 = before;
+                                to_dense = true;
+                            }
+                        }
+                        int64_t pos = vpos[pi];
+                        std::vector<int64_t> pre, suf;
+                        int64_t used = host_delta_i64(pp + pos, plen2 - pos,
+                                                      pre);
+                        if (used < 0 ||
+                            (int64_t)pre.size() < n_dense) {
+                            set_error("%s col %s: DELTA_BYTE_ARRAY prefix "
+                                      "stream bad", fd.path.c_str(),
+                                      cols[c].name.c_str());
+                            return false;
+                        }
+                        pos += used;
+                        used = host_delta_i64(pp + pos, plen2 - pos, suf);
+                        if (used < 0 ||
+                            (int64_t)suf.size() < n_dense) {
+                            set_error("%s col %s: DELTA_BYTE_ARRAY suffix "
+                                      "stream bad", fd.path.c_str(),
+                                      cols[c].name.c_str());
+                            return false;
+                        }
+                        pos += used;
+                        std::string prev;
+                        std::vector<int32_t> ids(n_dense);
+                        for (int64_t i = 0; i < n_dense; i++) {
+                            int64_t pl = pre[i], sl = suf[i];
+                            if (pl < 0 || sl < 0 ||
+                                pl > (int64_t)prev.size() ||
+                                pos + sl > plen2) {
+                                set_error("%s col %s: DELTA_BYTE_ARRAY "
+                                          "lengths bad", fd.path.c_str(),
+                                          cols[c].name.c_str());
+                                return false;
+                            }
+                            std::string v = prev.substr(0, pl);
+                            v.append((const char *)pp + pos, (size_t)sl);
+                            pos += sl;
+                            ids[i] = sd->add((const uint8_t *)v.data(),
+                                             (uint32_t)v.size());
+                            prev = std::move(v);
+                        }
+                        plan->encoded_bytes_total += plen2;
+                        if (to_dense) {
+                            const uint8_t *ib = (const uint8_t *)ids.data();
+                            rc.dense_host.insert(rc.dense_host.end(), ib,
+                                                 ib + n_dense * 4);
+                            rc.dense_segs.emplace_back(dense0, n_dense * 4);
+                        } else if (n_dense &&
+                                   hipMemcpy((uint8_t *)rc.contig +
+                                                 row0 * 4,
+                                             ids.data(), n_dense * 4,
+                                             hipMemcpyHostToDevice) !=
+                                       hipSuccess) {
+                            set_error("H2D failed");
+                            return false;
+                        }
+                    }
+                } else if (has_delta) {
                     if (cols[c].dtype == PMH_DT_STRING) {
                         set_error("%s col %s: DELTA byte arrays are a later "
                                   "round", fd.path.c_str(),
