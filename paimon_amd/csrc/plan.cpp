@@ -2084,8 +2084,12 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                         has_dict = true;
                     } else if (pg.encoding == ENC_DELTA_BINARY_PACKED) {
                         has_delta = true;
-                    } else if (pg.encoding == ENC_DELTA_BYTE_ARRAY &&
+                    } else if ((pg.encoding == ENC_DELTA_BYTE_ARRAY ||
+                                pg.encoding ==
+                                    ENC_DELTA_LENGTH_BYTE_ARRAY) &&
                                cols[c].dtype == PMH_DT_STRING) {
+                        // DELTA_LENGTH_BYTE_ARRAY = DELTA_BYTE_ARRAY with
+                        // no prefix stream (lengths + bytes)
                         has_dba = true;
                     } else {
                         set_error("%s: unsupported encoding %d",
@@ -2175,16 +2179,24 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                         }
                         int64_t pos = vpos[pi];
                         std::vector<int64_t> pre, suf;
-                        int64_t used = host_delta_i64(pp + pos, plen2 - pos,
-                                                      pre);
-                        if (used < 0 ||
-                            (int64_t)pre.size() < n_dense) {
-                            set_error("%s col %s: DELTA_BYTE_ARRAY prefix "
-                                      "stream bad", fd.path.c_str(),
-                                      cols[c].name.c_str());
-                            return false;
+                        const bool dlba =
+                            pg.encoding == ENC_DELTA_LENGTH_BYTE_ARRAY;
+                        int64_t used;
+                        if (dlba) {
+                            pre.assign((size_t)n_dense, 0);  // no prefixes
+                        } else {
+                            used = host_delta_i64(pp + pos, plen2 - pos,
+                                                  pre);
+                            if (used < 0 ||
+                                (int64_t)pre.size() < n_dense) {
+                                set_error("%s col %s: DELTA_BYTE_ARRAY "
+                                          "prefix stream bad",
+                                          fd.path.c_str(),
+                                          cols[c].name.c_str());
+                                return false;
+                            }
+                            pos += used;
                         }
-                        pos += used;
                         used = host_delta_i64(pp + pos, plen2 - pos, suf);
                         if (used < 0 ||
                             (int64_t)suf.size() < n_dense) {

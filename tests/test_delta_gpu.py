@@ -291,3 +291,29 @@ class TestDeltaByteArray:
 
     def test_multi_page(self, tmp_path):
         self._check(tmp_path, rows=120_000, page_kb=16, seed=973)
+
+
+class TestDeltaLengthByteArray:
+    def test_basic(self, tmp_path):
+        # DELTA_LENGTH_BYTE_ARRAY: lengths (DELTA) + raw bytes, no prefixes
+        n = 40_000
+        rng = np.random.default_rng(981)
+        k = np.arange(n, dtype=np.int64)
+        strs = [f"value-{int(x) % 500:03d}-{'y' * (int(x) % 5)}" for x in k]
+        tbl = pa.table({"_KEY_k": pa.array(k),
+                        "_SEQUENCE_NUMBER": pa.array(k),
+                        "_VALUE_KIND": pa.array(np.zeros(n, np.int8)),
+                        "v_s": pa.array(strs)})
+        path = os.path.join(str(tmp_path), "dlba.parquet")
+        pq.write_table(tbl, path, compression=None, use_dictionary=False,
+                       column_encoding={"v_s": "DELTA_LENGTH_BYTE_ARRAY"},
+                       data_page_version="1.0", store_schema=False,
+                       data_page_size=32 * 1024)
+        metas = [{"path": path, "rowCount": n, "minKey": 0,
+                  "maxKey": n - 1, "level": 0}]
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               [{"name": "v_s", "type": "string"}]) as plan:
+                b = plan.read_next()
+                dec = [b["v_s#dict"][i].decode() for i in b["v_s"]]
+        assert dec == strs
