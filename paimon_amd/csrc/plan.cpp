@@ -93,6 +93,14 @@ struct ZstdBatchPage {
     int64_t dst_len;
 };
 
+static bool gpu_zstd_batch_dev(
+    const uint8_t *src_base, int64_t src_total,
+    const std::vector<ZstdBatchPage> &pages, ::pmh_plan_t *plan,
+    uint8_t **dev_out, int64_t out_total,
+    const std::vector<std::pair<int64_t, int64_t>> &host_rngs,
+    uint8_t *host_out);
+
+
 static bool gpu_zstd_batch(const uint8_t *src_base, int64_t src_total,
                            const std::vector<ZstdBatchPage> &pages,
                            uint8_t *host_out, int64_t out_total) {
@@ -1967,8 +1975,13 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
         // codec per chunk.
         std::vector<uint8_t> file_unc;
         std::map<const void *, int64_t> zchunk_base;
+        // chunks whose decoded image can stay DEVICE-RESIDENT (simple
+        // PLAIN, no def levels, no dictionary page): their uncompressed
+        // bytes never cross PCIe — the PLAIN staging copies D2D
+        std::map<const void *, uint8_t *> zchunk_dev;
         if (gpu_zstd_enabled()) {
             std::vector<ZstdBatchPage> zp;
+            std::vector<std::pair<int64_t, int64_t>> host_rngs;
             int64_t lo = INT64_MAX, hi = 0, out = 0;
             for (auto &rg : sf.meta.row_groups)
                 for (size_t c = 0; c < cols.size(); c++) {
@@ -1976,7 +1989,15 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                     if (cc.codec != CODEC_ZSTD || cc.pages.empty()) continue;
                     if (zchunk_base.count(&cc)) continue;
                     zchunk_base[&cc] = out;
+                    bool needs_host =
+                        sf.meta.max_def_levels[leaf[c]] > 0 ||
+                        cols[c].dtype == PMH_DT_STRING;
+                    int64_t c0 = out;
                     for (auto &pg : cc.pages) {
+                        if (pg.page_type == 2 ||
+                            (pg.page_type == 0 &&
+                             pg.encoding != ENC_PLAIN))
+                            needs_host = true;
                         if (pg.data_off < lo) lo = pg.data_off;
                         if (pg.data_off + pg.compressed_size > hi)
                             hi = pg.data_off + pg.compressed_size;
@@ -1984,16 +2005,23 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                                       pg.uncompressed_size});
                         out += pg.uncompressed_size;
                     }
+                    if (needs_host) host_rngs.push_back({c0, out});
+                    else zchunk_dev[&cc] = nullptr;  // patched below
                 }
             if (!zp.empty()) {
                 for (auto &p : zp) p.src_off -= lo;
                 file_unc.resize(out);
-                if (gpu_zstd_batch(sf.data.data() + lo, hi - lo, zp,
-                                   file_unc.data(), out)) {
+                uint8_t *dev_blob = nullptr;
+                if (gpu_zstd_batch_dev(sf.data.data() + lo, hi - lo, zp,
+                                       plan, &dev_blob, out, host_rngs,
+                                       file_unc.data())) {
                     plan->stats.gpu_zstd_pages += (int64_t)zp.size();
+                    for (auto &kv : zchunk_dev)
+                        kv.second = dev_blob + zchunk_base[kv.first];
                 } else {
                     file_unc.clear();
                     zchunk_base.clear();
+                    zchunk_dev.clear();
                 }
             }
         }
@@ -2012,16 +2040,21 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                 std::vector<uint8_t> packed;  // compressed codecs
                 std::vector<int64_t> ppo(cc.pages.size());
                 const uint8_t *payload_base = sf.data.data() + chunk_start;
+                const uint8_t *zdev_base = nullptr;
                 auto zit = zchunk_base.find((const void *)&cc);
                 if (zit != zchunk_base.end()) {
                     // pages already decoded by the file-level k_zstd_pages
-                    // batch; the chunk's image starts at zit->second
+                    // batch; the chunk's image starts at zit->second.
+                    // Device-resident chunks (zchunk_dev) never copied
+                    // back — the PLAIN staging below goes D2D.
                     int64_t o = 0;
                     for (size_t pi = 0; pi < cc.pages.size(); pi++) {
                         ppo[pi] = o;
                         o += cc.pages[pi].uncompressed_size;
                     }
                     payload_base = file_unc.data() + zit->second;
+                    auto zdv = zchunk_dev.find((const void *)&cc);
+                    if (zdv != zchunk_dev.end()) zdev_base = zdv->second;
                 } else if (cc.codec == CODEC_ZSTD || cc.codec == CODEC_GZIP ||
                     cc.codec == CODEC_SNAPPY) {
                     int64_t total_unc = 0;
@@ -2405,6 +2438,26 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                         rc.dense_segs.emplace_back(before,
                                                    nvalid * stored);
                     }
+                } else if ((has_plain || (!has_dict && cc.num_values > 0))
+                           && zdev_base) {
+                    // PLAIN pages already decoded on DEVICE: stage D2D
+                    int64_t row_at = chunk_row0;
+                    for (size_t pi = 0; pi < cc.pages.size(); pi++) {
+                        auto &pg = cc.pages[pi];
+                        if (pg.page_type != 0) continue;
+                        int64_t nbytes = (int64_t)pg.num_values * stored;
+                        if (nbytes &&
+                            hipMemcpy((uint8_t *)rc.contig +
+                                          row_at * stored,
+                                      zdev_base + ppo[pi] + vpos[pi],
+                                      nbytes,
+                                      hipMemcpyDeviceToDevice) !=
+                                hipSuccess) {
+                            set_error("D2D staging failed");
+                            return false;
+                        }
+                        row_at += pg.num_values;
+                    }
                 } else if (has_plain || (!has_dict && cc.num_values > 0)) {
                     // pack PLAIN value payloads and copy into contig
                     std::vector<uint8_t> pack(cc.num_values * stored);
@@ -2561,6 +2614,69 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
             return false;
     }
     return true;
+}
+
+// Like gpu_zstd_batch, but the decoded blob STAYS on device (allocated
+// from the plan arena; *dev_out) and only `host_rngs` (chunks whose
+// staging needs host peeks) copy back. Simple PLAIN chunks then stage
+// with D2D copies — no PCIe round trip for their bytes.
+static bool gpu_zstd_batch_dev(
+    const uint8_t *src_base, int64_t src_total,
+    const std::vector<ZstdBatchPage> &pages, pmh_plan_t *plan,
+    uint8_t **dev_out, int64_t out_total,
+    const std::vector<std::pair<int64_t, int64_t>> &host_rngs,
+    uint8_t *host_out) {
+    if (pages.empty()) return true;
+    int n = (int)pages.size();
+    uint8_t *d_src = nullptr, *d_scr = nullptr;
+    ZstdJob *d_jobs = nullptr;
+    int64_t *d_st = nullptr;
+    uint8_t *d_dst = (uint8_t *)plan->bufs.alloc(out_total ? out_total : 1);
+    if (!d_dst) return false;
+    std::vector<ZstdJob> jobs(n);
+    for (int i = 0; i < n; i++)
+        jobs[i] = {(uint64_t)pages[i].src_off, (uint64_t)pages[i].dst_off,
+                   (uint32_t)pages[i].src_len, (uint32_t)pages[i].dst_len};
+    bool ok = false;
+    std::vector<int64_t> st(n);
+    do {
+        if (hipMalloc(&d_src, src_total) != hipSuccess) break;
+        if (hipMalloc(&d_scr, (size_t)n * PZ_SLOT) != hipSuccess) break;
+        if (hipMalloc(&d_jobs, n * sizeof(ZstdJob)) != hipSuccess) break;
+        if (hipMalloc(&d_st, n * 8) != hipSuccess) break;
+        if (hipMemcpy(d_src, src_base, src_total, hipMemcpyHostToDevice) !=
+            hipSuccess)
+            break;
+        if (hipMemcpy(d_jobs, jobs.data(), n * sizeof(ZstdJob),
+                      hipMemcpyHostToDevice) != hipSuccess)
+            break;
+        if (pmh_launch_zstd_pages(d_src, d_jobs, n, d_dst, d_scr, d_st,
+                                  nullptr) != hipSuccess)
+            break;
+        if (hipMemcpy(st.data(), d_st, n * 8, hipMemcpyDeviceToHost) !=
+            hipSuccess)
+            break;
+        bool all = true;
+        for (int i = 0; i < n; i++)
+            if (st[i] != pages[i].dst_len) all = false;
+        if (!all) break;
+        ok = true;
+        for (auto &r : host_rngs) {
+            if (r.second > r.first &&
+                hipMemcpy(host_out + r.first, d_dst + r.first,
+                          r.second - r.first,
+                          hipMemcpyDeviceToHost) != hipSuccess) {
+                ok = false;
+                break;
+            }
+        }
+    } while (0);
+    if (d_src) (void)hipFree(d_src);
+    if (d_scr) (void)hipFree(d_scr);
+    if (d_jobs) (void)hipFree(d_jobs);
+    if (d_st) (void)hipFree(d_st);
+    if (ok) *dev_out = d_dst;
+    return ok;
 }
 
 // Hierarchical section setup: group real runs into batches (<= PMH_MAX_RUNS

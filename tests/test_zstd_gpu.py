@@ -107,3 +107,39 @@ class TestCompressKernel:
         t = pq.read_table(path)
         assert (t["_KEY_k"].to_numpy() == k).all()
         assert (t["v"].to_numpy() == v).all()
+
+
+class TestDeviceResidentStaging:
+    def test_mixed_nullable_and_plain_chunks(self, tmp_path):
+        # nullable chunk (host-peeked def levels) and simple PLAIN chunks
+        # (device-resident, D2D-staged) in the same zstd file
+        import pyarrow.parquet as pq
+        rng = np.random.default_rng(41)
+        n = 120_000
+        k = np.arange(n, dtype=np.int64) * 2
+        v0 = rng.integers(0, 1 << 30, n).astype(np.int32)
+        mask = rng.random(n) < 0.25
+        tbl = pa.table({
+            "_KEY_k": pa.array(k),
+            "_SEQUENCE_NUMBER": pa.array(k.copy()),
+            "_VALUE_KIND": pa.array(np.zeros(n, np.int8)),
+            "v_c0": pa.array(v0, mask=mask),
+            "v_c1": pa.array((k * 3).astype(np.int64)),
+        })
+        path = str(tmp_path / "mix.parquet")
+        pq.write_table(tbl, path, compression="zstd",
+                       use_dictionary=False, data_page_version="1.0",
+                       store_schema=False, data_page_size=64 * 1024)
+        metas = [{"path": path, "rowCount": n, "minKey": 0,
+                  "maxKey": int(k[-1]), "level": 0}]
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               [{"name": "v_c0", "type": "int32"},
+                                {"name": "v_c1", "type": "int64"}]) as plan:
+                b = plan.read_next()
+                st = plan.stats()
+        assert st["gpu_zstd_pages"] > 0
+        assert (b["_KEY_k"] == k).all()
+        assert (b["v_c1"] == k * 3).all()
+        assert (b["v_c0#valid"] == ~mask).all()
+        assert (b["v_c0"][~mask] == v0[~mask]).all()
