@@ -1983,15 +1983,27 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
             std::vector<ZstdBatchPage> zp;
             std::vector<std::pair<int64_t, int64_t>> host_rngs;
             int64_t lo = INT64_MAX, hi = 0, out = 0;
+            // chunk classes: HOST (dict pages / non-PLAIN / strings:
+            // full D2H), PREFIX (PLAIN under a nullable schema: D2H only
+            // each page's def-level prefix; values stay on device unless
+            // the prefix shows REAL nulls), DEV (required PLAIN: nothing
+            // copies back)
+            struct PfxChunk {
+                const void *cc;
+                int64_t base;
+                int leafc;
+            };
+            std::vector<PfxChunk> pfx_chunks;
+            const int64_t PFX = 16384;
             for (auto &rg : sf.meta.row_groups)
                 for (size_t c = 0; c < cols.size(); c++) {
                     auto &cc = rg.columns[leaf[c]];
                     if (cc.codec != CODEC_ZSTD || cc.pages.empty()) continue;
                     if (zchunk_base.count(&cc)) continue;
                     zchunk_base[&cc] = out;
-                    bool needs_host =
-                        sf.meta.max_def_levels[leaf[c]] > 0 ||
-                        cols[c].dtype == PMH_DT_STRING;
+                    bool needs_host = cols[c].dtype == PMH_DT_STRING;
+                    const bool has_def =
+                        sf.meta.max_def_levels[leaf[c]] > 0;
                     int64_t c0 = out;
                     for (auto &pg : cc.pages) {
                         if (pg.page_type == 2 ||
@@ -2005,19 +2017,79 @@ static bool stage_run(pmh_plan_t *plan, const std::vector<FileDesc> &files,
                                       pg.uncompressed_size});
                         out += pg.uncompressed_size;
                     }
-                    if (needs_host) host_rngs.push_back({c0, out});
-                    else zchunk_dev[&cc] = nullptr;  // patched below
+                    if (needs_host)
+                        host_rngs.push_back({c0, out});
+                    else if (has_def)
+                        pfx_chunks.push_back({&cc, c0, leaf[c]});
+                    else
+                        zchunk_dev[&cc] = nullptr;  // patched below
                 }
             if (!zp.empty()) {
                 for (auto &p : zp) p.src_off -= lo;
                 file_unc.resize(out);
                 uint8_t *dev_blob = nullptr;
+                // page-prefix ranges for PREFIX chunks (def levels live
+                // at page start)
+                for (const auto &pc : pfx_chunks) {
+                    const auto &cc =
+                        *(const ColumnChunkMeta *)pc.cc;
+                    int64_t o = pc.base;
+                    for (auto &pg : cc.pages) {
+                        int64_t take = pg.uncompressed_size < PFX
+                                           ? pg.uncompressed_size
+                                           : PFX;
+                        host_rngs.push_back({o, o + take});
+                        o += pg.uncompressed_size;
+                    }
+                }
                 if (gpu_zstd_batch_dev(sf.data.data() + lo, hi - lo, zp,
                                        plan, &dev_blob, out, host_rngs,
                                        file_unc.data())) {
                     plan->stats.gpu_zstd_pages += (int64_t)zp.size();
+                    // PREFIX chunks: stay on device unless a page's def
+                    // stream shows real nulls (or overflows the prefix) —
+                    // then the chunk needs its full host image
+                    for (const auto &pc : pfx_chunks) {
+                        const auto &cc =
+                            *(const ColumnChunkMeta *)pc.cc;
+                        bool full = false;
+                        int64_t o = pc.base, cend = pc.base;
+                        for (auto &pg : cc.pages)
+                            cend += pg.uncompressed_size;
+                        for (auto &pg : cc.pages) {
+                            if (pg.page_type == 0) {
+                                uint32_t dl;
+                                memcpy(&dl, file_unc.data() + o, 4);
+                                if (4 + (int64_t)dl > PFX ||
+                                    def_levels_have_nulls(
+                                        file_unc.data() + o + 4, dl,
+                                        pg.num_values))
+                                    full = true;
+                            }
+                            o += pg.uncompressed_size;
+                            if (full) break;
+                        }
+                        if (full) {
+                            if (hipMemcpy(file_unc.data() + pc.base,
+                                          dev_blob + pc.base,
+                                          cend - pc.base,
+                                          hipMemcpyDeviceToHost) !=
+                                hipSuccess) {
+                                set_error("zstd prefix D2H failed");
+                                return false;
+                            }
+                        } else {
+                            zchunk_dev[pc.cc] = nullptr;
+                        }
+                    }
                     for (auto &kv : zchunk_dev)
                         kv.second = dev_blob + zchunk_base[kv.first];
+                    if (getenv("PMH_DEBUG_ZSTD"))
+                        fprintf(stderr,
+                                "[zstd] file chunks=%zu dev-resident=%zu "
+                                "host-rngs=%zu\n",
+                                zchunk_base.size(), zchunk_dev.size(),
+                                host_rngs.size());
                 } else {
                     file_unc.clear();
                     zchunk_base.clear();
