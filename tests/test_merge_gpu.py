@@ -810,3 +810,54 @@ class TestHierarchicalSections:
         assert (got["v_c0#valid"] == ~exp_null).all()
         live = ~exp_null
         assert (got["v_c0"][live] == exp_v[live]).all()
+
+
+class TestHierarchicalCrosses:
+    """Hierarchical sections x other features: deletion vectors consume in
+    the batch pass (rtombs sub-array), zstd inputs decode through
+    k_zstd_pages before the batch merges."""
+
+    def test_hier_with_deletion_vectors(self, tmp_path):
+        from scripts.gen_dv_golden import serialize_roaring32, wrap_dv
+        rng = np.random.default_rng(991)
+        runs = gen_runs_dedup(36, 3_000, n_value_cols=2, seed=991,
+                              delete_frac=0.2)
+        metas = write_runs(runs, str(tmp_path), compression="NONE")
+        dels = {}
+        for fi in (0, 7, 35):
+            n = len(runs[fi]["key"])
+            dels[fi] = sorted(rng.choice(n, n // 6, replace=False).tolist())
+        blob = b""
+        for fi, pos in dels.items():
+            ser = wrap_dv(serialize_roaring32(pos))
+            metas[fi]["deletionVector"] = {
+                "file": str(tmp_path / "idx.dv"),
+                "offset": len(blob), "length": len(ser)}
+            blob += ser
+        (tmp_path / "idx.dv").write_bytes(blob)
+        fruns = []
+        for i, r in enumerate(runs):
+            keep = np.ones(len(r["key"]), dtype=bool)
+            if i in dels:
+                keep[np.array(dels[i], dtype=np.int64)] = False
+            fruns.append({"key": r["key"][keep], "seq": r["seq"][keep],
+                          "kind": r["kind"][keep],
+                          "values": [v[keep] for v in r["values"]]})
+        rr, ww = merge_dedup(fruns)
+        exp = np.array([fruns[a]["key"][b] for a, b in zip(rr, ww)],
+                       np.int64)
+        with Session(0) as s:
+            with MergeReadPlan(s, file_descs_from_metas(metas), KEY_COLS,
+                               _value_cols(2)) as plan:
+                got = []
+                while True:
+                    b = plan.read_next()
+                    if b is None:
+                        break
+                    got.append(b["_KEY_k"].copy())
+        assert (np.concatenate(got) == exp).all()
+
+    def test_hier_with_zstd_inputs(self, tmp_path):
+        runs = gen_runs_dedup(34, 4_000, n_value_cols=2, seed=992,
+                              delete_frac=0.15)
+        _run_and_compare(tmp_path, runs, compression="zstd")
