@@ -3428,7 +3428,79 @@ struct ZBack {
         return (uint32_t)((win >> (lo - wlo)) &
                           ((n >= 32) ? 0xFFFFFFFFu : ((1u << n) - 1u)));
     }
+    DEV uint32_t peek_abs(int n) {  // bits [bp, bp+n) — bp pre-decremented
+        if (n == 0) return 0;
+        if (bp < wlo || bp + n > wlo + 64) refill(bp);
+        return (uint32_t)((win >> (bp - wlo)) &
+                          ((n >= 32) ? 0xFFFFFFFFu : ((1u << n) - 1u)));
+    }
 };
+
+// lane-0 sequence decode over the register bit-window (the scalar
+// pz_seq_next re-gathers 6 bytes per field read; ZBack refills once per
+// ~56 bits — the "next lever" for k_zstd_pages' match-heavy frames)
+static DEV int zdev_seq_next(ZBack &rb, const PzFse *llT, const PzFse *ofT,
+                             const PzFse *mlT, uint32_t &sll, uint32_t &sof,
+                             uint32_t &sml, uint64_t rep[3], int last,
+                             PzSeq *out) {
+    int llc = llT[sll].sym;
+    int ofc = ofT[sof].sym;
+    int mlc = mlT[sml].sym;
+    if (llc > 35 || mlc > 52 || ofc > 31) return PZ_ERR_SEQ;
+    uint64_t ofv;
+    if (ofc > 0) {
+        rb.bp -= ofc;
+        ofv = ((uint64_t)1 << ofc) + rb.peek_abs(ofc);
+    } else {
+        ofv = 1;
+    }
+    int mb = pz_ml_bits(mlc);
+    rb.bp -= mb;
+    uint32_t ml = pz_ml_base(mlc) + rb.peek_abs(mb);
+    int lb = pz_ll_bits(llc);
+    rb.bp -= lb;
+    uint32_t ll = pz_ll_base(llc) + rb.peek_abs(lb);
+    uint64_t off;
+    if (ofv > 3) {
+        off = ofv - 3;
+        rep[2] = rep[1];
+        rep[1] = rep[0];
+        rep[0] = off;
+    } else {
+        int idx = (int)ofv - 1 + (ll == 0 ? 1 : 0);
+        if (idx == 0) {
+            off = rep[0];
+        } else if (idx == 1) {
+            off = rep[1];
+            rep[1] = rep[0];
+            rep[0] = off;
+        } else if (idx == 2) {
+            off = rep[2];
+            rep[2] = rep[1];
+            rep[1] = rep[0];
+            rep[0] = off;
+        } else {
+            off = rep[0] - 1;
+            if (off == 0) return PZ_ERR_OFFSET;
+            rep[2] = rep[1];
+            rep[1] = rep[0];
+            rep[0] = off;
+        }
+    }
+    out->ll = ll;
+    out->ml = ml;
+    out->off = off;
+    if (!last) {
+        rb.bp -= llT[sll].nbits;
+        sll = llT[sll].base + rb.peek_abs(llT[sll].nbits);
+        rb.bp -= mlT[sml].nbits;
+        sml = mlT[sml].base + rb.peek_abs(mlT[sml].nbits);
+        rb.bp -= ofT[sof].nbits;
+        sof = ofT[sof].base + rb.peek_abs(ofT[sof].nbits);
+        if (rb.bp < 0) return PZ_ERR_SEQ;
+    }
+    return 0;
+}
 
 static DEV void z_fence() { __threadfence_block(); }
 
@@ -3666,14 +3738,21 @@ static DEV int64_t zdev_page(const uint8_t *src, int64_t slen, uint8_t *dst,
         const uint8_t *bs = sq + so;
         int64_t lpos = 0;
         int done = 0;
+        ZBack rb;  // lane-0 register bit-window over the sequence stream
+        if (lane == 0) {
+            rb.init(bs, st.bp);
+            rb.bp = st.bp;
+        }
         while (done < nseq) {
             int cnt = nseq - done < 64 ? nseq - done : 64;
             if (lane == 0) {
                 int64_t f = 0;
                 for (int i = 0; i < cnt; i++) {
                     PzSeq q;
-                    int rcq = pz_seq_next(bs, cx->llT, cx->ofT, cx->mlT, &st,
-                                          done + i == nseq - 1, &q);
+                    int rcq = zdev_seq_next(rb, cx->llT, cx->ofT, cx->mlT,
+                                            st.s_ll, st.s_of, st.s_ml,
+                                            st.rep,
+                                            done + i == nseq - 1, &q);
                     if (rcq < 0) {
                         f = rcq;
                         break;
@@ -3682,6 +3761,7 @@ static DEV int64_t zdev_page(const uint8_t *src, int64_t slen, uint8_t *dst,
                     L->ml[i] = q.ml;
                     L->off[i] = q.off;
                 }
+                if (done + cnt >= nseq) st.bp = rb.bp;  // final check uses it
                 L->flag = f;
             }
             z_fence();
