@@ -169,3 +169,42 @@ class TestFuzzVsSequentialPort:
             gres = sorted((int(runs[a]["key"][b]), int(runs[a]["seq"][b]))
                           for a, b in zip(*res))
             assert gres == sorted((t[0], t[1]) for t in eres)
+
+
+class TestFuzzTopRunPlacement:
+    """The top-level run's position among the runs must not matter (top
+    detection is by LEVEL, not run order) — fuzz with the max-level run
+    first, middle, and last, plus null-aware row_dedup."""
+
+    @pytest.mark.parametrize("top_pos", [0, 2, 4])
+    def test_top_position(self, top_pos):
+        rng = np.random.default_rng(300 + top_pos)
+        for trial in range(10):
+            k = 5
+            runs, levels = [], []
+            seqbase = 0
+            for ri in range(k):
+                nrows = int(rng.integers(1, 300))
+                keys = np.sort(rng.choice(500, nrows, replace=False))
+                kind = np.where(rng.random(nrows) < 0.3, D, I
+                                ).astype(np.int8)
+                lvl = MAX_LEVEL if ri == top_pos else 0
+                if lvl == MAX_LEVEL:
+                    kind[:] = I
+                    seq = np.arange(nrows, dtype=np.int64) - 10_000
+                else:
+                    seq = np.arange(nrows, dtype=np.int64) + seqbase
+                    seqbase += nrows
+                runs.append({"key": keys.astype(np.int64), "seq": seq,
+                             "kind": kind,
+                             "values": [rng.integers(0, 4, nrows
+                                                     ).astype(np.int32)]})
+                levels.append(lvl)
+            cl, res = full_changelog_model(runs, levels, MAX_LEVEL,
+                                           row_dedup=bool(trial % 2))
+            ecl, eres = _seq_port(runs, levels, MAX_LEVEL,
+                                  bool(trial % 2))
+            got = [(int(runs[a]["key"][b]), int(kk))
+                   for a, b, kk in zip(*cl)]
+            exp = [(t[0], kk) for t, kk in ecl]
+            assert got == exp, (top_pos, trial)
